@@ -1,0 +1,381 @@
+"""Shared search engine: task-graph construction, dispatch, cv_results_ assembly.
+
+MI355X-native re-implementation of the behavior of the reference's shared
+``_fit`` (spark-sklearn ``grid_search.py::_fit`` ~L110-300 [R], see
+SURVEY.md section 3.1).  The reference builds a (candidate x fold) task list,
+ships it to Spark executors via ``sc.parallelize(...).map(fun).collect()``
+with the training matrix broadcast once per executor, then assembles the
+sklearn ``cv_results_`` dict on the driver.
+
+Here the same task graph is scheduled onto a pluggable :class:`Backend`:
+
+* ``LocalBackend``       -- CPU thread pool (the moral equivalent of Spark
+                            local mode); real sklearn fits inside tasks.
+* ``DistributedBackend`` -- one process per GPU over ``torch.distributed``
+                            (RCCL on ROCm; gloo for CPU tests), the dataset
+                            broadcast once into each GPU's HBM, fold scores
+                            all-gathered, refit on rank 0.
+* ``HipEngineBackend``   -- batched candidate fits through the hand-written
+                            CDNA4 HIP kernels (spark_sklearn_amd.ops).
+
+The ``cv_results_`` layout is the conformance contract -- identical key set,
+shapes, masked ``param_*`` arrays, tie-ranking and ``best_index_`` semantics
+as ``sklearn.model_selection.GridSearchCV`` (oracle: installed sklearn),
+which the reference deliberately mirrors.
+"""
+
+from __future__ import annotations
+
+import numbers
+import time
+import warnings
+from collections import defaultdict
+from functools import partial
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+from numpy.ma import MaskedArray
+from scipy.stats import rankdata
+
+from sklearn.base import BaseEstimator, MetaEstimatorMixin, clone, is_classifier
+from sklearn.metrics import check_scoring
+from sklearn.model_selection import check_cv
+from sklearn.utils.validation import indexable
+
+__all__ = ["Task", "TaskResult", "fit_and_score_task", "BaseSearchCV"]
+
+
+class Task:
+    """One (candidate, fold) unit of work.
+
+    Plain-old-data by design: the reference ships ``(params, train_idx,
+    test_idx)`` tuples tagged with a submission index so results can be
+    re-ordered deterministically after an unordered ``collect()``
+    (reference ``grid_search.py`` ~L120 [R]).  We keep the same tagging so
+    every backend (threadpool, torch.distributed, HIP engine) is free to
+    return results in any order.
+    """
+
+    __slots__ = ("index", "cand_index", "fold_index", "params", "train", "test")
+
+    def __init__(self, index, cand_index, fold_index, params, train, test):
+        self.index = index
+        self.cand_index = cand_index
+        self.fold_index = fold_index
+        self.params = params
+        self.train = train
+        self.test = test
+
+
+class TaskResult:
+    """Fixed-width per-task result record.
+
+    Mirrors the tuple returned by the closure the reference ships to
+    executors (``_fit_and_score`` output: train_score?, test_score,
+    n_test_samples, fit_time, score_time -- SURVEY.md section 3.1).  This is
+    also the RCCL all-gather wire format (6 scalars / task, see
+    parallel/distributed.py).
+    """
+
+    __slots__ = ("index", "test_score", "train_score", "n_test", "fit_time", "score_time")
+
+    def __init__(self, index, test_score, train_score, n_test, fit_time, score_time):
+        self.index = index
+        self.test_score = test_score
+        self.train_score = train_score
+        self.n_test = n_test
+        self.fit_time = fit_time
+        self.score_time = score_time
+
+
+def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
+                       fit_params=None, return_train_score=True,
+                       error_score="raise") -> TaskResult:
+    """Clone-fit-score one task on the host (CPU tier).
+
+    Behavioral equivalent of the worker closure in the reference
+    (``grid_search.py`` ~L155 [R]): ``clone`` -> ``set_params`` ->
+    ``fit(X[train], y[train])`` -> score on the test fold, with
+    ``error_score`` swallow-or-raise semantics for robustness on large
+    grids (one bad candidate must not kill a 10k-task job).
+    """
+    est = clone(base_estimator)
+    if task.params:
+        est.set_params(**task.params)
+    fit_params = fit_params or {}
+
+    X_train, y_train = _safe_index(X, task.train), _safe_index(y, task.train)
+    X_test, y_test = _safe_index(X, task.test), _safe_index(y, task.test)
+
+    t0 = time.perf_counter()
+    try:
+        if y_train is None:
+            est.fit(X_train, **fit_params)
+        else:
+            est.fit(X_train, y_train, **fit_params)
+        fit_time = time.perf_counter() - t0
+        t1 = time.perf_counter()
+        test_score = scorer(est, X_test, y_test)
+        score_time = time.perf_counter() - t1
+        train_score = scorer(est, X_train, y_train) if return_train_score else np.nan
+    except Exception as exc:  # noqa: BLE001 - error_score contract
+        fit_time = time.perf_counter() - t0
+        if error_score == "raise":
+            raise
+        if not isinstance(error_score, numbers.Number):
+            raise ValueError(
+                "error_score must be the string 'raise' or a numeric value"
+            ) from exc
+        warnings.warn(
+            f"Estimator fit failed for params {task.params!r}; the score on "
+            f"this train-test partition will be set to {error_score}. "
+            f"Details: {exc!r}",
+            stacklevel=2,
+        )
+        test_score = float(error_score)
+        train_score = float(error_score) if return_train_score else np.nan
+        score_time = 0.0
+
+    n_test = len(task.test)
+    return TaskResult(task.index, float(test_score), float(train_score),
+                      n_test, fit_time, score_time)
+
+
+def _safe_index(data, indices):
+    if data is None:
+        return None
+    if hasattr(data, "iloc"):
+        return data.iloc[indices]
+    if isinstance(data, np.ndarray):
+        return data[indices]
+    if hasattr(data, "__getitem__") and hasattr(data, "shape"):  # scipy sparse
+        return data[indices]
+    return [data[i] for i in indices]
+
+
+class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
+    """Common machinery for :class:`GridSearchCV` / :class:`RandomizedSearchCV`.
+
+    Constructor surface follows the reference era (spark-sklearn 0.3.0
+    tracking sklearn 0.18/0.19 [R]) with a leading context argument in the
+    subclasses; ``n_jobs``/``pre_dispatch`` are accepted-but-ignored for
+    drop-in compatibility, exactly as the reference ignores them
+    (SURVEY.md C2).
+
+    Divergences from the 0.18-era surface (documented per SURVEY.md 3.1):
+
+    * ``iid`` defaults to ``False`` (modern sklearn semantics -- unweighted
+      mean across folds) so that ``cv_results_`` matches the installed
+      sklearn oracle bit-for-bit; pass ``iid=True`` for the reference's
+      n_test-weighted means.
+    * ``return_train_score`` keeps the reference default ``True``.
+    """
+
+    def __init__(self, estimator, scoring=None, fit_params=None, n_jobs=1,
+                 iid=False, refit=True, cv=None, verbose=0,
+                 pre_dispatch="2*n_jobs", error_score="raise",
+                 return_train_score=True):
+        self.estimator = estimator
+        self.scoring = scoring
+        self.fit_params = fit_params
+        self.n_jobs = n_jobs          # accepted-but-ignored (reference parity)
+        self.iid = iid
+        self.refit = refit
+        self.cv = cv
+        self.verbose = verbose
+        self.pre_dispatch = pre_dispatch  # accepted-but-ignored
+        self.error_score = error_score
+        self.return_train_score = return_train_score
+
+    # -- subclass hooks ----------------------------------------------------
+    def _get_param_iterator(self):
+        raise NotImplementedError
+
+    def _get_backend(self):
+        """Resolve the execution backend from the context handle."""
+        ctx = getattr(self, "ctx", None)
+        from .parallel import resolve_backend
+
+        return resolve_backend(ctx)
+
+    # -- the engine --------------------------------------------------------
+    def fit(self, X, y=None, groups=None, **fit_params):
+        """Run the full search: build tasks, dispatch, aggregate, refit."""
+        estimator = self.estimator
+        cv = check_cv(self.cv, y, classifier=is_classifier(estimator))
+        self.scorer_ = check_scoring(estimator, scoring=self.scoring)
+
+        X, y, groups = indexable(X, y, groups)
+        merged_fit_params = dict(self.fit_params or {})
+        merged_fit_params.update(fit_params)
+
+        candidates = list(self._get_param_iterator())
+        n_candidates = len(candidates)
+        splits = list(cv.split(X, y, groups))
+        n_splits = len(splits)
+
+        if self.verbose > 0:
+            print(f"Fitting {n_splits} folds for each of {n_candidates} "
+                  f"candidates, totalling {n_candidates * n_splits} fits")
+
+        tasks = []
+        idx = 0
+        for cand_i, params in enumerate(candidates):
+            for fold_i, (train, test) in enumerate(splits):
+                tasks.append(Task(idx, cand_i, fold_i, params,
+                                  np.asarray(train), np.asarray(test)))
+                idx += 1
+
+        backend = self._get_backend()
+        results = backend.run_search(
+            estimator, X, y, tasks,
+            scorer=self.scorer_,
+            fit_params=merged_fit_params,
+            return_train_score=self.return_train_score,
+            error_score=self.error_score,
+            verbose=self.verbose,
+        )
+        # restore deterministic submit order regardless of backend ordering
+        by_index = {r.index: r for r in results}
+        missing = [t.index for t in tasks if t.index not in by_index]
+        if missing:
+            raise RuntimeError(
+                f"backend returned {len(results)} results for {len(tasks)} "
+                f"tasks; missing indices {missing[:5]}..."
+            )
+        ordered = [by_index[i] for i in range(len(tasks))]
+
+        self.cv_results_ = self._assemble_cv_results(
+            candidates, ordered, n_candidates, n_splits)
+        self.n_splits_ = n_splits
+        self.multimetric_ = False
+        self.best_index_ = int(
+            np.flatnonzero(self.cv_results_["rank_test_score"] == 1)[0])
+        self.best_params_ = self.cv_results_["params"][self.best_index_]
+        self.best_score_ = float(
+            self.cv_results_["mean_test_score"][self.best_index_])
+
+        if self.refit:
+            best_estimator = clone(estimator).set_params(**self.best_params_)
+            t0 = time.perf_counter()
+            refit_fn = getattr(backend, "refit", None)
+            if refit_fn is not None:
+                best_estimator = refit_fn(best_estimator, X, y,
+                                          merged_fit_params)
+            elif y is not None:
+                best_estimator.fit(X, y, **merged_fit_params)
+            else:
+                best_estimator.fit(X, **merged_fit_params)
+            self.refit_time_ = time.perf_counter() - t0
+            self.best_estimator_ = best_estimator
+        return self
+
+    # -- cv_results_ assembly ---------------------------------------------
+    def _assemble_cv_results(self, candidates, ordered_results,
+                             n_candidates, n_splits) -> Dict[str, Any]:
+        """Build the sklearn-layout ``cv_results_`` dict.
+
+        Exact key set: ``mean_fit_time``,
+        ``std_fit_time``, ``mean_score_time``, ``std_score_time``,
+        ``split{k}_test_score``, ``mean_test_score``, ``std_test_score``,
+        ``rank_test_score``, train-score variants iff
+        ``return_train_score``, masked ``param_<p>`` arrays, ``params``.
+        """
+        test_scores = np.array(
+            [r.test_score for r in ordered_results], dtype=np.float64
+        ).reshape(n_candidates, n_splits)
+        fit_times = np.array(
+            [r.fit_time for r in ordered_results], dtype=np.float64
+        ).reshape(n_candidates, n_splits)
+        score_times = np.array(
+            [r.score_time for r in ordered_results], dtype=np.float64
+        ).reshape(n_candidates, n_splits)
+        n_test = np.array(
+            [r.n_test for r in ordered_results], dtype=np.float64
+        ).reshape(n_candidates, n_splits)
+
+        results: Dict[str, Any] = {}
+
+        def _store(key, array, splits=False, rank=False, weights=None):
+            if splits:
+                for k in range(n_splits):
+                    results[f"split{k}_{key}"] = array[:, k]
+            means = np.average(array, axis=1, weights=weights)
+            results[f"mean_{key}"] = means
+            # weighted std around the (possibly weighted) mean, as sklearn does
+            results[f"std_{key}"] = np.sqrt(
+                np.average((array - means[:, np.newaxis]) ** 2,
+                           axis=1, weights=weights))
+            if rank:
+                results[f"rank_{key}"] = np.asarray(
+                    rankdata(-means, method="min"), dtype=np.int32)
+
+        score_weights = n_test if self.iid else None
+        _store("fit_time", fit_times)
+        _store("score_time", score_times)
+        _store("test_score", test_scores, splits=True, rank=True,
+               weights=score_weights)
+        if self.return_train_score:
+            train_scores = np.array(
+                [r.train_score for r in ordered_results], dtype=np.float64
+            ).reshape(n_candidates, n_splits)
+            _store("train_score", train_scores, splits=True)
+
+        # masked param_* arrays: mask=True where a candidate lacks that param
+        # (heterogeneous grids) -- reference grid_search.py ~L270 [R]
+        param_results: Dict[str, MaskedArray] = defaultdict(
+            partial(MaskedArray, np.empty(n_candidates), mask=True,
+                    dtype=object))
+        for cand_i, params in enumerate(candidates):
+            for name, value in params.items():
+                param_results[f"param_{name}"][cand_i] = value
+        results.update(param_results)
+        results["params"] = list(candidates)
+        return results
+
+    # -- post-fit surface delegated to best_estimator_ ---------------------
+    def _check_is_fitted(self, method_name):
+        if not self.refit:
+            raise AttributeError(
+                f"This {type(self).__name__} instance was initialized with "
+                f"refit=False; {method_name} is only available after "
+                f"refitting on the best parameters.")
+        if not hasattr(self, "best_estimator_"):
+            raise AttributeError(
+                f"This {type(self).__name__} instance is not fitted yet.")
+
+    def predict(self, X):
+        self._check_is_fitted("predict")
+        return self.best_estimator_.predict(X)
+
+    def predict_proba(self, X):
+        self._check_is_fitted("predict_proba")
+        return self.best_estimator_.predict_proba(X)
+
+    def predict_log_proba(self, X):
+        self._check_is_fitted("predict_log_proba")
+        return self.best_estimator_.predict_log_proba(X)
+
+    def decision_function(self, X):
+        self._check_is_fitted("decision_function")
+        return self.best_estimator_.decision_function(X)
+
+    def transform(self, X):
+        self._check_is_fitted("transform")
+        return self.best_estimator_.transform(X)
+
+    def inverse_transform(self, X):
+        self._check_is_fitted("inverse_transform")
+        return self.best_estimator_.inverse_transform(X)
+
+    def score(self, X, y=None):
+        if hasattr(self, "scorer_") and self.scorer_ is not None and \
+                hasattr(self, "best_estimator_"):
+            return self.scorer_(self.best_estimator_, X, y)
+        self._check_is_fitted("score")
+        return self.best_estimator_.score(X, y)
+
+    @property
+    def classes_(self):
+        self._check_is_fitted("classes_")
+        return self.best_estimator_.classes_

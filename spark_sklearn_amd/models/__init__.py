@@ -1,0 +1,7 @@
+"""MI355X-native estimator families (batched CDNA4 kernel fits)."""
+
+from .linear import (LinearRegression, LinearSVC, LogisticRegression, Ridge,
+                     SGDClassifier)
+
+__all__ = ["LogisticRegression", "LinearSVC", "SGDClassifier",
+           "LinearRegression", "Ridge"]

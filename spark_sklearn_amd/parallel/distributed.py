@@ -1,0 +1,152 @@
+"""torch.distributed backend: one process per GPU over RCCL/xGMI.
+
+Replaces the reference's Spark runtime with the
+MI355X-native communication plan of SURVEY.md section 5.8:
+
+* RC1 -- the dataset is broadcast ONCE from rank 0 into every GPU's 288 GB
+  HBM (``dist.broadcast`` -> rcclBroadcast over xGMI; tree/direct p2p fans
+  out over the root's 7 links).  This replaces Spark's TorrentBroadcast.
+* task dispatch -- needs no network at all on one node: tasks are
+  partitioned statically, candidate-round-robin, so every rank derives its
+  share from the task list deterministically (replaces the Spark scheduler).
+* RC2 -- fold scores are fixed-width 6-float records summed into a dense
+  (n_tasks, 6) table with one all-reduce (each rank fills only its rows;
+  zeros elsewhere), so every rank assembles identical ``cv_results_``.
+  This replaces ``collect()``.
+* refit -- rank 0 refits on the full resident matrix (it already holds all
+  of X after RC1; RC4 in SURVEY.md needs no collective) and broadcasts the
+  fitted parameters so every rank's ``best_estimator_`` is usable.
+
+On CPU (backend "gloo") the same code path runs with sklearn host fits --
+this is the multi-process CI tier (world_size=2, no GPU).
+"""
+
+from __future__ import annotations
+
+import pickle
+from typing import List, Optional
+
+import numpy as np
+
+from ..base_search import Task, TaskResult
+from .backend import Backend, execute_local_tasks
+
+__all__ = ["DistributedBackend"]
+
+
+class DistributedBackend(Backend):
+    """Task-parallel search over an initialized torch.distributed group.
+
+    Parameters
+    ----------
+    device : torch.device or None
+        This rank's device.  ``None`` auto-selects ``cuda:LOCAL_RANK`` when
+        a GPU is visible, else CPU (gloo tier).
+    n_threads : int or None
+        Thread-pool width for host fits on each rank.
+    """
+
+    def __init__(self, device=None, n_threads: Optional[int] = None):
+        import torch
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            raise RuntimeError(
+                "DistributedBackend requires torch.distributed to be "
+                "initialized (launch with torchrun / init_process_group)")
+        self.rank = dist.get_rank()
+        self.world_size = dist.get_world_size()
+        self.n_threads = n_threads
+        if device is None and torch.cuda.is_available():
+            import os
+
+            local = int(os.environ.get("LOCAL_RANK", self.rank))
+            device = torch.device("cuda", local % torch.cuda.device_count())
+        self.device = device
+        self._torch = torch
+        self._dist = dist
+
+    # -- RC1: dataset broadcast -------------------------------------------
+    def broadcast_data(self, X, y):
+        """Broadcast (X, y) from rank 0 to all ranks.
+
+        Ranks other than 0 may pass ``X=None``; they receive rank 0's
+        arrays.  GPU path: tensors land directly in HBM and stay resident
+        for the whole search.  Numpy-in, numpy-out on the gloo tier.
+        """
+        torch, dist = self._torch, self._dist
+        comm_dev = self.device if (self.device is not None and
+                                   self.device.type == "cuda") else "cpu"
+
+        def _bcast_array(arr):
+            # shape/dtype header first (tiny object bcast), then payload
+            if self.rank == 0:
+                t = torch.as_tensor(np.ascontiguousarray(arr)).to(comm_dev)
+                header = [tuple(t.shape), str(t.dtype)]
+            else:
+                header = [None, None]
+            dist.broadcast_object_list(header, src=0)
+            if self.rank != 0:
+                dtype = getattr(torch, header[1].replace("torch.", ""))
+                t = torch.empty(header[0], dtype=dtype, device=comm_dev)
+            dist.broadcast(t, src=0)
+            return t
+
+        X_t = _bcast_array(X) if (self.rank == 0 and X is not None) or \
+            self.rank != 0 else None
+        if X_t is None:
+            raise ValueError("rank 0 must provide X")
+        has_y = [y is not None if self.rank == 0 else None]
+        dist.broadcast_object_list(has_y, src=0)
+        y_t = _bcast_array(y) if has_y[0] else None
+        return X_t, y_t
+
+    # -- partition + execute + RC2 gather ---------------------------------
+    def local_task_filter(self, tasks: List[Task]) -> List[Task]:
+        """Candidate-round-robin partition: rank r owns candidates with
+        ``cand_index % world_size == r``, keeping all folds of a candidate
+        on one rank so batched kernels see whole candidates."""
+        return [t for t in tasks if t.cand_index % self.world_size == self.rank]
+
+    def run_search(self, estimator, X, y, tasks, *, scorer, fit_params,
+                   return_train_score, error_score, verbose):
+        torch, dist = self._torch, self._dist
+        local = self.local_task_filter(tasks)
+        if verbose > 0 and self.rank == 0:
+            print(f"[dist] {self.world_size} ranks, "
+                  f"{len(tasks)} tasks (~{len(local)} per rank)")
+        results = execute_local_tasks(
+            estimator, X, y, local, scorer=scorer, fit_params=fit_params,
+            return_train_score=return_train_score, error_score=error_score,
+            verbose=verbose, n_threads=self.n_threads, device=self.device)
+
+        # RC2: dense (n_tasks, 6) float64 table, one sum-all-reduce.
+        comm_dev = self.device if (self.device is not None and
+                                   self.device.type == "cuda" and
+                                   dist.get_backend() == "nccl") else "cpu"
+        table = torch.zeros((len(tasks), 6), dtype=torch.float64,
+                            device=comm_dev)
+        for r in results:
+            table[r.index] = torch.tensor(
+                [r.index, r.test_score, r.train_score, r.n_test,
+                 r.fit_time, r.score_time], dtype=torch.float64)
+        dist.all_reduce(table, op=dist.ReduceOp.SUM)
+        table = table.cpu().numpy()
+        return [TaskResult(int(row[0]), row[1], row[2], int(row[3]),
+                           row[4], row[5]) for row in table]
+
+    # -- rank-0 refit + parameter broadcast --------------------------------
+    def refit(self, estimator, X, y, fit_params):
+        dist = self._dist
+        if self.rank == 0:
+            if y is not None:
+                estimator.fit(X, y, **(fit_params or {}))
+            else:
+                estimator.fit(X, **(fit_params or {}))
+            payload = [pickle.dumps(estimator)]
+        else:
+            payload = [None]
+        dist.broadcast_object_list(payload, src=0)
+        if self.rank != 0:
+            estimator = pickle.loads(payload[0])
+        return estimator

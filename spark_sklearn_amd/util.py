@@ -1,0 +1,102 @@
+"""Context and convenience helpers.
+
+Behavioral counterpart of the reference's ``util.py``
+(``createLocalSparkSession(appName=...)`` ~L10 [R]): where the
+reference spins up a local SparkSession, this module discovers HIP devices,
+wires up torch.distributed (RCCL over xGMI when launched one-process-per-GPU
+with torchrun; gloo on CPU) and hands back a context object that the search
+classes accept as their leading argument.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+__all__ = ["HipContext", "createLocalContext", "init_distributed"]
+
+
+class HipContext:
+    """Execution context: device discovery + backend + engine knobs.
+
+    Engine-level knobs live here rather than on the search objects so the
+    search constructor surface stays sklearn-compatible (SURVEY.md
+    section 5.6): ``candidate_batch`` (candidates folded into one kernel
+    launch), ``n_streams`` (HIP streams per GPU), ``dtype`` for the
+    resident matrix.
+    """
+
+    def __init__(self, backend=None, device=None, candidate_batch: int = 128,
+                 n_streams: int = 2, dtype: str = "bfloat16",
+                 n_threads: Optional[int] = None):
+        from .parallel import resolve_backend
+
+        self.candidate_batch = candidate_batch
+        self.n_streams = n_streams
+        self.dtype = dtype
+        self.n_threads = n_threads
+        self.device = device if device is not None else _default_device()
+        self.backend = resolve_backend(backend)
+        # push engine knobs down where backends look for them
+        for attr in ("candidate_batch", "n_streams", "dtype"):
+            setattr(self.backend, attr, getattr(self, attr))
+
+    @property
+    def n_devices(self) -> int:
+        try:
+            import torch
+
+            return torch.cuda.device_count() if torch.cuda.is_available() else 0
+        except Exception:
+            return 0
+
+    def __repr__(self):
+        return (f"HipContext(device={self.device}, "
+                f"n_devices={self.n_devices}, "
+                f"backend={type(self.backend).__name__}, "
+                f"candidate_batch={self.candidate_batch})")
+
+
+def _default_device():
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", 0))
+            return torch.device("cuda", local % torch.cuda.device_count())
+    except Exception:
+        pass
+    return None
+
+
+def createLocalContext(appName: str = "spark-sklearn-amd",
+                       **kwargs) -> HipContext:
+    """Local-mode context (name kept parallel to the reference's
+    ``createLocalSparkSession``).  Uses the local GPU when present, else a
+    CPU thread pool."""
+    return HipContext(**kwargs)
+
+
+def init_distributed(backend: Optional[str] = None,
+                     timeout_seconds: int = 1800) -> HipContext:
+    """Initialize torch.distributed from torchrun env vars and return a
+    distributed context.
+
+    On ROCm the ``"nccl"`` backend IS RCCL; collectives run over xGMI.
+    Falls back to gloo when no GPU is visible (the CPU CI tier).
+    """
+    import datetime
+
+    import torch
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", 0))
+            torch.cuda.set_device(local % torch.cuda.device_count())
+        dist.init_process_group(
+            backend=backend,
+            timeout=datetime.timedelta(seconds=timeout_seconds))
+    return HipContext(backend="dist")

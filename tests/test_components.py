@@ -1,0 +1,197 @@
+"""Parity-component tests: Converter, KeyedEstimator/KeyedModel, gapply,
+CSRVectorUDT (reference test files test_converter.py / test_keyed_models.py
+/ test_gapply.py)."""
+
+import numpy as np
+import pandas as pd
+import pytest
+import scipy.sparse as sp
+from sklearn.cluster import KMeans
+from sklearn.linear_model import LogisticRegression as SKLogReg
+from sklearn.preprocessing import StandardScaler
+
+from spark_sklearn_amd import (Converter, CSRVectorUDT, KeyedEstimator,
+                               SparkSklearnEstimator, gapply)
+from spark_sklearn_amd.models import LinearRegression, LogisticRegression
+from spark_sklearn_amd.udt import DeviceCSRMatrix, stack_csr_rows
+
+
+# ---------------------------- Converter -----------------------------------
+
+def test_converter_roundtrip_logreg(binary_blobs):
+    X, y = binary_blobs
+    ours = LogisticRegression(max_iter=100).fit(X, y)
+    conv = Converter()
+    sk = conv.toSKLearn(ours)
+    assert isinstance(sk, SKLogReg)
+    # predictions must agree (same linear decision function)
+    np.testing.assert_array_equal(sk.predict(X), ours.predict(X))
+    back = conv.toHip(sk)
+    np.testing.assert_allclose(back.coef_, ours.coef_, rtol=1e-12)
+    np.testing.assert_array_equal(back.predict(X), ours.predict(X))
+
+
+def test_converter_linear_regression():
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 5))
+    y = X @ rng.standard_normal(5) + 1.0
+    ours = LinearRegression(max_iter=400).fit(X, y)
+    sk = Converter().toSKLearn(ours)
+    np.testing.assert_allclose(sk.predict(X), ours.predict(X), rtol=1e-5)
+
+
+def test_converter_rejects_unsupported():
+    with pytest.raises(ValueError):
+        Converter().toSKLearn(KMeans())
+    with pytest.raises(ValueError):
+        Converter().toHip(StandardScaler())
+
+
+def test_converter_unfitted_raises(binary_blobs):
+    with pytest.raises(ValueError, match="not fitted"):
+        Converter().toSKLearn(LogisticRegression())
+
+
+# ------------------------- KeyedEstimator ---------------------------------
+
+def _keyed_df():
+    rng = np.random.default_rng(0)
+    rows = []
+    for k in range(3):
+        center = np.array([k * 10.0, -k * 5.0])
+        for _ in range(30):
+            x = center + rng.standard_normal(2)
+            rows.append({"k": k, "features": x, "label": float(x[0] > k * 10)})
+    return pd.DataFrame(rows)
+
+
+def test_keyed_clusterer():
+    df = _keyed_df()
+    ke = KeyedEstimator(sklearnEstimator=KMeans(n_clusters=2, n_init=2,
+                                                random_state=0),
+                        keyCols=["k"])
+    model = ke.fit(df)
+    assert ke.estimatorType == "clusterer"
+    assert len(model.keyedModels) == 3
+    out = model.transform(df)
+    assert "prediction" in out.columns
+    assert len(out) == len(df)
+    assert list(out.index) == list(df.index)
+
+
+def test_keyed_predictor_matches_per_key_sklearn():
+    df = _keyed_df()
+    ke = KeyedEstimator(sklearnEstimator=SKLogReg(max_iter=200),
+                        keyCols=["k"], xCol="features", yCol="label")
+    model = ke.fit(df)
+    assert ke.estimatorType == "predictor"
+    out = model.transform(df)
+    # oracle: per-key local sklearn fit must give identical predictions
+    for k, group in df.groupby("k"):
+        X = np.stack(group["features"].values)
+        y = group["label"].values
+        ref = SKLogReg(max_iter=200).fit(X, y)
+        got = out.loc[group.index, "prediction"].values
+        np.testing.assert_array_equal(got, ref.predict(X))
+
+
+def test_keyed_transformer():
+    df = _keyed_df()
+    ke = KeyedEstimator(sklearnEstimator=StandardScaler(), keyCols=["k"])
+    model = ke.fit(df)
+    assert ke.estimatorType == "transformer"
+    out = model.transform(df)
+    assert "output" in out.columns
+    for k, group in out.groupby("k"):
+        vals = np.stack(group["output"].values)
+        np.testing.assert_allclose(vals.mean(axis=0), 0.0, atol=1e-9)
+
+
+def test_keyed_unseen_key():
+    df = _keyed_df()
+    model = KeyedEstimator(sklearnEstimator=KMeans(n_clusters=2, n_init=2),
+                           keyCols=["k"]).fit(df)
+    extra = pd.DataFrame(
+        [{"k": 99, "features": np.zeros(2)}])
+    with pytest.raises(KeyError):
+        model.transform(extra)
+    dropped = model.transform(extra, unseenKeys="drop")
+    assert len(dropped) == 0
+
+
+def test_keyed_validation():
+    with pytest.raises(ValueError):
+        KeyedEstimator(sklearnEstimator=None)
+    with pytest.raises(ValueError):
+        KeyedEstimator(sklearnEstimator=KMeans(), estimatorType="predictor")
+    with pytest.raises(ValueError):
+        KeyedEstimator(sklearnEstimator=KMeans(),
+                       estimatorType="bogus")
+    wrapper = SparkSklearnEstimator(KMeans(n_clusters=4))
+    assert wrapper.n_clusters == 4
+
+
+# ------------------------------ gapply ------------------------------------
+
+def test_gapply_matches_pandas_groupby():
+    df = pd.DataFrame({"g": [0, 0, 1, 1, 1], "v": [1.0, 2.0, 3.0, 4.0, 5.0],
+                       "w": [10.0, 20.0, 30.0, 40.0, 50.0]})
+
+    def summarize(key, pdf):
+        return pd.DataFrame({"v_sum": [pdf["v"].sum()],
+                             "w_mean": [pdf["w"].mean()]})
+
+    out = gapply((df, "g"), summarize, ["v_sum", "w_mean"], "v", "w")
+    oracle = df.groupby("g").agg(v_sum=("v", "sum"), w_mean=("w", "mean"))
+    np.testing.assert_allclose(out["v_sum"].values, oracle["v_sum"].values)
+    np.testing.assert_allclose(out["w_mean"].values, oracle["w_mean"].values)
+    assert list(out["g"]) == [0, 1]
+
+
+def test_gapply_schema_mismatch_and_empty():
+    df = pd.DataFrame({"g": [0, 1], "v": [1.0, 2.0]})
+    with pytest.raises(ValueError, match="schema"):
+        gapply((df, "g"), lambda k, p: pd.DataFrame({"x": [1]}),
+               ["wrong"], "v")
+    out = gapply((df, "g"), lambda k, p: None, ["x"], "v")
+    assert len(out) == 0 and list(out.columns) == ["g", "x"]
+
+
+def test_gapply_no_retain_group_columns():
+    df = pd.DataFrame({"g": [0, 0, 1], "v": [1.0, 2.0, 3.0]})
+    out = gapply((df, "g"), lambda k, p: pd.DataFrame({"s": [p["v"].sum()]}),
+                 ["s"], "v", retainGroupColumns=False)
+    assert list(out.columns) == ["s"]
+    assert sorted(out["s"]) == [3.0, 3.0]
+
+
+# ------------------------------- UDT --------------------------------------
+
+def test_csr_vector_udt_roundtrip():
+    row = sp.csr_matrix(np.array([[0.0, 1.5, 0.0, -2.0]]))
+    rec = CSRVectorUDT.serialize(row)
+    assert rec["size"] == 4
+    back = CSRVectorUDT.deserialize(rec)
+    np.testing.assert_allclose(back.toarray(), row.toarray())
+    with pytest.raises(ValueError):
+        CSRVectorUDT.serialize(sp.csr_matrix(np.eye(2)))
+
+
+def test_stack_csr_rows_mixed():
+    rows = [sp.csr_matrix(np.array([[1.0, 0.0]])),
+            {"indices": np.array([1]), "values": np.array([2.0]), "size": 2},
+            np.array([3.0, 4.0])]
+    out = stack_csr_rows(rows)
+    np.testing.assert_allclose(out.toarray(),
+                               [[1, 0], [0, 2], [3, 4]])
+
+
+def test_device_csr_cpu_roundtrip():
+    mat = sp.random(50, 20, density=0.1, format="csr", random_state=0)
+    dev = DeviceCSRMatrix.from_scipy(mat, device="cpu")
+    back = dev.to_scipy()
+    assert back.shape == mat.shape
+    assert dev.nnz == mat.nnz
+    # bf16 storage tolerance
+    np.testing.assert_allclose(back.toarray(), mat.toarray(),
+                               atol=0.01, rtol=0.01)
