@@ -1,0 +1,179 @@
+"""Device-side execution of candidate batches through the CDNA4 kernels.
+
+Mirrors engine._HostLinearExecutor's interface (fit / scores / msq) with
+the compute running in spark_sklearn_amd._hip_ops.  The dataset is uploaded
+ONCE per (array, device) and stays resident in HBM for the whole search
+(module-level cache) -- the single-GPU analog of the RC1 broadcast.
+
+MFMA fragment-layout probe: the A/B operand k-mapping of
+v_mfma_f32_16x16x32_bf16 cannot be verified without hardware, so on first
+use all 4 (AMAP, BMAP) combinations run against a random asymmetric host
+reference (transpose-detecting by construction, guide G9) and the matching
+one is installed with set_layout().
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import numpy as np
+
+from . import load_extension
+
+_METRIC_ID = {"accuracy": 0, "neg_log_loss": 1, "r2": 2}
+_layout_probed = False
+_data_cache: dict = {}
+
+
+def _roundup(x: int, m: int) -> int:
+    return ((x + m - 1) // m) * m
+
+
+def ensure_layout(device) -> int:
+    """Probe and install the MFMA fragment layout (idempotent)."""
+    global _layout_probed
+    ext = load_extension()
+    if _layout_probed and ext.get_layout() >= 0:
+        return ext.get_layout()
+    import torch
+
+    rng = np.random.default_rng(1234)
+    A = rng.standard_normal((16, 32)).astype(np.float32)
+    B = rng.standard_normal((32, 16)).astype(np.float32)
+    ref = (A.astype(np.float64) @ B.astype(np.float64))
+    At = torch.as_tensor(A).to(device).to(torch.bfloat16)
+    Bt = torch.as_tensor(np.ascontiguousarray(B.T)).to(device).to(
+        torch.bfloat16)
+    # recompute the reference at bf16 input precision
+    ref_bf = (At.float().cpu().numpy().astype(np.float64) @
+              Bt.float().cpu().numpy().T.astype(np.float64))
+    out = ext.gemm_probe(At.contiguous(), Bt.contiguous()).cpu().numpy()
+    errs = [float(np.abs(out[p] - ref_bf).max()) for p in range(4)]
+    best = int(np.argmin(errs))
+    if errs[best] > 0.05 * float(np.abs(ref_bf).max()):
+        raise RuntimeError(
+            f"MFMA layout probe failed: no candidate matches "
+            f"(errors {errs}, ref scale {np.abs(ref).max():.3f})")
+    ext.set_layout(best)
+    _layout_probed = True
+    return best
+
+
+def _as_device_bf16(X, device):
+    """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array."""
+    import torch
+
+    if isinstance(X, torch.Tensor):
+        key = (X.data_ptr(), X.shape, str(device))
+    else:
+        key = (id(X), np.asarray(X).shape, str(device))
+    hit = _data_cache.get(key)
+    if hit is not None:
+        return hit
+    t = X if isinstance(X, torch.Tensor) else torch.as_tensor(
+        np.ascontiguousarray(np.asarray(X, dtype=np.float32)))
+    n, F = t.shape
+    Fp = max(64, _roundup(F, 64))
+    dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
+    dev_t[:, :F] = t.to(device).to(torch.bfloat16)
+    if len(_data_cache) > 4:
+        _data_cache.clear()
+    _data_cache[key] = (dev_t, F)
+    return dev_t, F
+
+
+class DeviceLinearExecutor:
+    """Executes batched linear fits/scores on one MI355X GPU."""
+
+    def __init__(self, X, y_enc, classes, device=None):
+        import torch
+
+        self.torch = torch
+        self.device = device if device is not None else torch.device(
+            "cuda", torch.cuda.current_device())
+        self.ext = load_extension()
+        ensure_layout(self.device)
+        self.X, self.F = _as_device_bf16(X, self.device)
+        self.Fp = self.X.shape[1]
+        self.y = torch.as_tensor(
+            np.asarray(y_enc, dtype=np.float32)).to(self.device)
+        self.classes = classes
+        self.msq = float(self.ext.msq_rows(self.X)) * \
+            (1.0 if self.F == self.Fp else 1.0)  # zero-pad adds nothing
+        self._fit_state = None  # (Wt_dev, bias_dev) for the scores call
+
+    # -- fit ---------------------------------------------------------------
+    def fit(self, train_idx, plan):
+        torch = self.torch
+        nt = len(train_idx)
+        NTp = _roundup(max(nt, 64), 64)
+        idx_np = np.zeros(NTp, dtype=np.int32)
+        idx_np[:nt] = np.asarray(train_idx, dtype=np.int32)
+        idx = torch.as_tensor(idx_np).to(self.device)
+
+        B = plan.B
+        Bp = max(64, _roundup(B, 64))
+
+        def pad_i32(a, fill=0):
+            out = np.full(Bp, fill, dtype=np.int32)
+            out[:B] = np.asarray(a, dtype=np.int32)
+            return torch.as_tensor(out).to(self.device)
+
+        def pad_f32(a, fill=0.0):
+            out = np.full(Bp, fill, dtype=np.float32)
+            out[:B] = np.asarray(a, dtype=np.float32)
+            return torch.as_tensor(out).to(self.device)
+
+        loss_t = pad_i32(plan.loss)
+        lr_t = pad_f32(plan.lr)
+        l1_t = pad_f32(plan.l1)
+        l2_t = pad_f32(plan.l2)
+        mom_t = pad_f32(plan.momentum)
+        tol_t = pad_f32(plan.tol)
+        mi_t = pad_i32(plan.max_iter)      # pads: max_iter 0 -> frozen
+        tgt_t = pad_i32(plan.target_class)
+        ic_np = np.zeros(Bp, dtype=np.uint8)
+        ic_np[:B] = np.asarray(plan.fit_intercept, dtype=np.uint8)
+        ic_t = torch.as_tensor(ic_np).to(self.device)
+
+        max_epochs = int(plan.max_iter.max()) if B else 0
+        check_every = 32 if max_epochs > 48 else 0
+        W, bias, n_iter, cur_loss, Wt = self.ext.linear_fit(
+            self.X, self.y, idx, loss_t, lr_t, l1_t, l2_t, mom_t, tol_t,
+            mi_t, tgt_t, ic_t, nt, max_epochs, check_every)
+        torch.cuda.synchronize(self.device)
+        self._fit_state = (Wt, bias, loss_t, tgt_t)
+        W_np = W[:B, :self.F].float().cpu().numpy().T.copy()   # [F, B]
+        return (W_np, bias[:B].float().cpu().numpy(),
+                n_iter[:B].cpu().numpy(),
+                cur_loss[:B].float().cpu().numpy())
+
+    # -- scores ------------------------------------------------------------
+    def scores(self, idx, W, b, plan, metric):
+        torch = self.torch
+        if self._fit_state is None:
+            raise RuntimeError("scores() requires a preceding fit()")
+        Wt, bias, loss_t, tgt_t = self._fit_state
+        Bp = Wt.shape[0]
+        if Bp not in (64, 128, 256):
+            raise RuntimeError(f"unsupported column batch {Bp} for scoring")
+        n_cand = int(plan.col_to_cand.max()) + 1
+        C = plan.n_classes
+        idx_t = torch.as_tensor(
+            np.asarray(idx, dtype=np.int64)).to(self.device)
+        sums = self.ext.linear_scores(
+            self.X, self.y, idx_t, Wt, bias, loss_t, tgt_t,
+            C, _METRIC_ID[metric], n_cand).cpu().numpy()
+        torch.cuda.synchronize(self.device)
+        n = len(idx)
+        if metric == "accuracy":
+            return sums / n
+        if metric == "neg_log_loss":
+            return sums / n
+        # r2: 1 - ss_res / ss_tot
+        y_sel = self.y[idx_t]
+        ss_tot = float(((y_sel - y_sel.mean()) ** 2).sum().item())
+        if ss_tot <= 0:
+            return np.zeros(n_cand)
+        return 1.0 - sums / ss_tot
