@@ -615,7 +615,8 @@ double msq_rows(torch::Tensor X) {
 }
 
 // linear_fit: the full epoch loop on device.
-// Returns (W [Bp][Fp] f32, bias [Bp], n_iter [Bp] i32, cur_loss [Bp] f32).
+// Returns (W [Bp][Fp] f32, bias [Bp], n_iter [Bp] i32, cur_loss [Bp] f32,
+//          Wt [Bp][Fp] bf16 -- kept on device for the scores kernel).
 std::vector<torch::Tensor> linear_fit(
     torch::Tensor X, torch::Tensor y, torch::Tensor idx,
     torch::Tensor loss_id, torch::Tensor lr, torch::Tensor l1,
@@ -723,7 +724,7 @@ std::vector<torch::Tensor> linear_fit(
       hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s);
     }
   }
-  return {W, bias, n_iter, cur_loss};
+  return {W, bias, n_iter, cur_loss, Wt};
 }
 
 // linear_scores: per-candidate sums; host finalizes means/r2.
