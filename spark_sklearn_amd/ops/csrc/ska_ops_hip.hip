@@ -90,6 +90,43 @@ __device__ __forceinline__ bf16x8 lds_frag(const __bf16* base, int lane) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// Permuted + swizzled LDS tiles (fast path for the HW-probed layout P=3).
+//
+// The probe shows gfx950's mfma_f32_16x16x32_bf16 wants per-lane operand
+// elements k = (lane>>4)*4 + (i&3) + (i>>2)*16 (two 4-blocks, +16 apart).
+// Generic staging leaves those as 2 x 8B LDS reads with ~4-way bank
+// aliasing (PMC: 25.6M SQ_LDS_BANK_CONFLICT/dispatch).  Fix: store each
+// 64-k row PERMUTED so one lane's 8 fragment elements are contiguous
+// (single ds_read_b128), row stride exactly 64 elements (128 B), and an
+// XOR of the 16B-group index by (row&7) (guide T2, st-swizzle) so the 8
+// groups x 4 banks tile all 32 banks with balanced load.
+//   pos(k)  = (k>>5)*32 + (((k&15)>>2)<<3) + (((k>>4)&1)<<2) + (k&3)
+//   addr(r,k) = r*64 + (pos(k) ^ ((r&7)<<3))
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int perm_pos(int k) {
+  return ((k >> 5) << 5) + (((k & 15) >> 2) << 3) + (((k >> 4) & 1) << 2) +
+         (k & 3);
+}
+__device__ __forceinline__ int swz8(int p, int r) {
+  return p ^ ((r & 7) << 3);
+}
+// store 8 globally-contiguous elements (k0 % 8 == 0) into row r
+__device__ __forceinline__ void store_row8(__bf16* row, int r, int k0,
+                                           bf16x8 v) {
+  bf16x4 lo = {v[0], v[1], v[2], v[3]};
+  bf16x4 hi = {v[4], v[5], v[6], v[7]};
+  *reinterpret_cast<bf16x4*>(row + swz8(perm_pos(k0), r)) = lo;
+  *reinterpret_cast<bf16x4*>(row + swz8(perm_pos(k0 + 4), r)) = hi;
+}
+// read one lane's fragment: row pointer (r*64 into the tile), row index r,
+// k-group g = lane>>4, half h = ks/32
+__device__ __forceinline__ bf16x8 lds_frag_p(const __bf16* row, int r,
+                                             int g, int h) {
+  return *reinterpret_cast<const bf16x8*>(row + swz8(h * 32 + g * 8, r));
+}
+
 // ---------------------------------------------------------------------------
 // layout probe: D = A(16x32) . B(32x16), one wave, LDS-staged like the real
 // kernels.  Host test compares all 4 (AMAP, BMAP) combos vs reference.
@@ -180,189 +217,737 @@ __device__ __forceinline__ void loss_grad(int lid, float z, float yv,
 }
 
 // ---------------------------------------------------------------------------
-// fwd_d_kernel: Z = X[idx] . Wt^T + b, fused d/loss epilogue.
-// grid (NTp/64, Bp/64), block 256 (4 waves as 2x2 of 32x32 wave-tiles).
+// fwd_d2_kernel: v2 of fwd_d -- block tile 64 rows x Bp cols (Bp<=256).
+// 4 waves side-by-side over columns, each wave computes 64 x (Bp/4):
+// 4 m-frags x nfrag n-frags -> 4x the MFMA work per staged byte of the
+// 64x64 v1 tile, and W is staged once per k-chunk for ALL columns (no
+// cross-block X re-reads within a row tile).
 // ---------------------------------------------------------------------------
-template <int P>
+template <int P, int NF>
 __global__ __launch_bounds__(256)
-void fwd_d_kernel(const __bf16* __restrict__ X,       // [n][Fp]
-                  const float* __restrict__ y,        // [n]
-                  const int* __restrict__ idx,        // [NTp]
-                  const __bf16* __restrict__ Wt,      // [Bp][Fp]
-                  const float* __restrict__ bias,     // [Bp]
-                  const int* __restrict__ loss_id,    // [Bp]
-                  const int* __restrict__ target,     // [Bp]
-                  __bf16* __restrict__ Dt,            // [Bp][NTp]
-                  float* __restrict__ loss_sum,       // [Bp]
-                  float* __restrict__ db_sum,         // [Bp]
-                  int nt, int NTp, int Fp) {
+void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
+                   const float* __restrict__ y,        // [n]
+                   const int* __restrict__ idx,        // [NTp]
+                   const __bf16* __restrict__ Wt,      // [Bp][Fp]
+                   const float* __restrict__ bias,     // [Bp]
+                   const int* __restrict__ loss_id,    // [Bp]
+                   const int* __restrict__ target,     // [Bp]
+                   __bf16* __restrict__ Dt,            // [Bp][NTp]
+                   float* __restrict__ loss_sum,       // [Bp]
+                   float* __restrict__ db_sum,         // [Bp]
+                   int nt, int NTp, int Fp, int Bp, int want_loss) {
   constexpr int AMAP = P >> 1, BMAP = P & 1;
+  (void)want_loss;  // generic path always computes loss
   __shared__ __bf16 sX[64][72];
-  __shared__ __bf16 sW[64][72];
-  __shared__ float sLoss[64];
-  __shared__ float sDb[64];
+  __shared__ __bf16 sW[256][72];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
   __shared__ int sIdx[64];
   __shared__ float sY[64];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wm = wave & 1, wn = wave >> 1;
-  const int rt = blockIdx.x, ct = blockIdx.y;
+  const int rt = blockIdx.x;
+  constexpr int cpw = NF * 16;        // cols per wave: 16/32/64
 
   if (tid < 64) {
     int r = rt * 64 + tid;
-    int gi = idx[r < NTp ? r : NTp - 1];
+    int gi = idx[r];
     sIdx[tid] = gi;
     sY[tid] = y[gi];
-    sLoss[tid] = 0.f;
-    sDb[tid] = 0.f;
   }
+  for (int i = tid; i < Bp; i += 256) { sLoss[i] = 0.f; sDb[i] = 0.f; }
   __syncthreads();
 
-  f32x4 acc[2][2];
-  for (int i = 0; i < 2; ++i)
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[4][4];
+  for (int i = 0; i < 4; ++i)
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int srow = tid >> 2;          // staging: 64 rows x 4 segs of 16
+  const int srow = tid >> 2;
   const int sseg = tid & 3;
   for (int kc = 0; kc < Fp; kc += 64) {
-    // stage X rows (gathered) and W rows (candidate-major): 16B vectors
     {
       const uint4* gx = reinterpret_cast<const uint4*>(
           X + (long)sIdx[srow] * Fp + kc + sseg * 16);
       uint4* lx = reinterpret_cast<uint4*>(&sX[srow][sseg * 16]);
       lx[0] = gx[0];
       lx[1] = gx[1];
+    }
+    for (int e = tid; e < Bp * 4; e += 256) {
+      int c = e >> 2, seg = e & 3;
       const uint4* gw = reinterpret_cast<const uint4*>(
-          Wt + (long)(ct * 64 + srow) * Fp + kc + sseg * 16);
-      uint4* lw = reinterpret_cast<uint4*>(&sW[srow][sseg * 16]);
+          Wt + (long)c * Fp + kc + seg * 16);
+      uint4* lw = reinterpret_cast<uint4*>(&sW[c][seg * 16]);
       lw[0] = gw[0];
       lw[1] = gw[1];
     }
     __syncthreads();
     for (int ks = 0; ks < 64; ks += 32) {
-      bf16x8 a0 = lds_frag<AMAP>(&sX[wm * 32 + (lane & 15)][ks], lane);
-      bf16x8 a1 = lds_frag<AMAP>(&sX[wm * 32 + 16 + (lane & 15)][ks], lane);
-      bf16x8 b0 = lds_frag<BMAP>(&sW[wn * 32 + (lane & 15)][ks], lane);
-      bf16x8 b1 = lds_frag<BMAP>(&sW[wn * 32 + 16 + (lane & 15)][ks], lane);
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        a[m] = lds_frag<AMAP>(&sX[m * 16 + (lane & 15)][ks], lane);
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        bf16x8 b = lds_frag<BMAP>(
+            &sW[wave * cpw + fn * 16 + (lane & 15)][ks], lane);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], b, acc[m][fn], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
-  // epilogue: z -> d, loss; transposed D write (4 consecutive rows / 8B)
-  for (int fn = 0; fn < 2; ++fn) {
-    int col_in = wn * 32 + fn * 16 + (lane & 15);
-    int c = ct * 64 + col_in;
-    float bia = bias[c];
-    int lid = loss_id[c];
-    int tgt = target[c];
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = wave * cpw + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
     float lsum = 0.f, dsum = 0.f;
-    for (int fm = 0; fm < 2; ++fm) {
-      int row_base = wm * 32 + fm * 16 + ((lane >> 4) << 2);
+    for (int m = 0; m < 4; ++m) {
+      int row_base = m * 16 + ((lane >> 4) << 2);
       bf16x4 dpack;
       for (int r = 0; r < 4; ++r) {
-        int row_in = row_base + r;
-        int gr = rt * 64 + row_in;
-        float z = acc[fm][fn][r] + bia;
+        int gr = rt * 64 + row_base + r;
+        float z = acc[m][fn][r] + bia;
         float d, ls;
-        loss_grad(lid, z, sY[row_in], tgt, &d, &ls);
+        loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
         if (gr >= nt) { d = 0.f; ls = 0.f; }
         dpack[r] = (__bf16)d;
         lsum += ls;
         dsum += d;
       }
-      bf16x4* dst = reinterpret_cast<bf16x4*>(
-          Dt + (long)c * NTp + rt * 64 + row_base);
-      *dst = dpack;
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 64 + row_base) = dpack;
     }
     atomicAdd(&sLoss[col_in], lsum);
     atomicAdd(&sDb[col_in], dsum);
   }
   __syncthreads();
-  if (tid < 64) {
-    atomicAdd(&loss_sum[ct * 64 + tid], sLoss[tid]);
-    atomicAdd(&db_sum[ct * 64 + tid], sDb[tid]);
+  for (int i = tid; i < Bp; i += 256) {
+    atomicAdd(&loss_sum[i], sLoss[i]);
+    atomicAdd(&db_sum[i], sDb[i]);
   }
 }
 
 // ---------------------------------------------------------------------------
-// bwd_g_kernel: G^T[c][f] += sum_r D^T[c][r] * X[idx[r]][f], split-K over
-// row segments.  grid (Bp/64, Fp/64, nseg), block 256.
+// bwd_g2_kernel: v2 of bwd_g -- output tile Bp cands x 64 features, reading
+// the RESIDENT TRANSPOSE Xt [Fp][n] so the B-operand stages coalesced
+// (contiguity fast path: CV folds give contiguous index runs, so 16-element
+// idx chunks are usually vector loads; scalar gather otherwise).
+// grid (Fp/64, nseg).
 // ---------------------------------------------------------------------------
-template <int P>
+template <int P, int NF>
 __global__ __launch_bounds__(256)
-void bwd_g_kernel(const __bf16* __restrict__ X,     // [n][Fp]
-                  const int* __restrict__ idx,      // [NTp]
-                  const __bf16* __restrict__ Dt,    // [Bp][NTp]
-                  float* __restrict__ Gt,           // [Bp][Fp]
-                  int NTp, int Fp, int seg_rows) {
+void bwd_g2_kernel(const __bf16* __restrict__ Xt,    // [Fp][n]
+                   const int* __restrict__ idx,      // [NTp]
+                   const __bf16* __restrict__ Dt,    // [Bp][NTp]
+                   float* __restrict__ Gt,           // [Bp][Fp]
+                   long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
   constexpr int AMAP = P >> 1, BMAP = P & 1;
-  __shared__ __bf16 sD[64][72];    // candidate x row-chunk
-  __shared__ __bf16 sXt[64][72];   // feature x row-chunk (transposed)
-  __shared__ int sRid[64];
+  __shared__ __bf16 sD[256][72];
+  __shared__ __bf16 sXt[64][72];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wm = wave & 1, wn = wave >> 1;
-  const int ct = blockIdx.x, ft = blockIdx.y;
-  const int r0 = blockIdx.z * seg_rows;
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
   const int r1 = min(r0 + seg_rows, NTp);
+  constexpr int cpw = NF * 16;
 
-  f32x4 acc[2][2];
-  for (int i = 0; i < 2; ++i)
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[4][4];
+  for (int i = 0; i < 4; ++i)
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int srow = tid >> 2;
-  const int sseg = tid & 3;
   for (int rc = r0; rc < r1; rc += 64) {
-    if (tid < 64) sRid[tid] = idx[rc + tid];
-    __syncthreads();
-    // stage D^T tile: candidate-major, contiguous in r
-    {
+    for (int e = tid; e < Bp * 4; e += 256) {
+      int c = e >> 2, seg = e & 3;
       const uint4* gd = reinterpret_cast<const uint4*>(
-          Dt + (long)(ct * 64 + srow) * NTp + rc + sseg * 16);
-      uint4* ld = reinterpret_cast<uint4*>(&sD[srow][sseg * 16]);
+          Dt + (long)c * NTp + rc + seg * 16);
+      uint4* ld = reinterpret_cast<uint4*>(&sD[c][seg * 16]);
       ld[0] = gd[0];
       ld[1] = gd[1];
     }
-    // stage X transposed: read 8 contiguous f per thread, scatter rows
     {
-      int r = tid & 63;
-      int fs = tid >> 6;                 // 4 threads-groups x 2 iters = 8 segs
-      for (int it = 0; it < 2; ++it) {
-        int fseg = fs + it * 4;
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(
-            X + (long)sRid[r] * Fp + ft * 64 + fseg * 8);
-        for (int j = 0; j < 8; ++j) sXt[fseg * 8 + j][r] = v[j];
+      // stage Xt rows (features) gathered over idx; 16-chunk contiguity
+      // fast path (KFold folds are contiguous runs)
+      int f = tid >> 2, seg = tid & 3;
+      int rbase = rc + seg * 16;
+      int i0 = idx[rbase];
+      int i15 = idx[rbase + 15];
+      const __bf16* src = Xt + (long)(ft * 64 + f) * n_rows;
+      if (i15 - i0 == 15) {
+        const uint4* gx = reinterpret_cast<const uint4*>(src + i0);
+        // i0 may be unaligned for 16B vectors; use 8x2B fallback if so
+        if (((unsigned long)(src + i0) & 15) == 0) {
+          uint4* lx = reinterpret_cast<uint4*>(&sXt[f][seg * 16]);
+          lx[0] = gx[0];
+          lx[1] = gx[1];
+        } else {
+          for (int j = 0; j < 16; ++j) sXt[f][seg * 16 + j] = src[i0 + j];
+        }
+      } else {
+        for (int j = 0; j < 16; ++j)
+          sXt[f][seg * 16 + j] = src[idx[rbase + j]];
       }
     }
     __syncthreads();
     for (int ks = 0; ks < 64; ks += 32) {
-      bf16x8 a0 = lds_frag<AMAP>(&sD[wm * 32 + (lane & 15)][ks], lane);
-      bf16x8 a1 = lds_frag<AMAP>(&sD[wm * 32 + 16 + (lane & 15)][ks], lane);
-      bf16x8 b0 = lds_frag<BMAP>(&sXt[wn * 32 + (lane & 15)][ks], lane);
-      bf16x8 b1 = lds_frag<BMAP>(&sXt[wn * 32 + 16 + (lane & 15)][ks], lane);
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      bf16x8 b[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        b[fn] = lds_frag<BMAP>(&sXt[fn * 16 + (lane & 15)][ks], lane);
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        bf16x8 a = lds_frag<AMAP>(
+            &sD[wave * cpw + m * 16 + (lane & 15)][ks], lane);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b[fn], acc[m][fn], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
 
-  for (int fm = 0; fm < 2; ++fm) {
-    for (int fn = 0; fn < 2; ++fn) {
-      int f_in = wn * 32 + fn * 16 + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fn * 16 + (lane & 15);
       for (int r = 0; r < 4; ++r) {
-        int c_in = wm * 32 + fm * 16 + ((lane >> 4) << 2) + r;
-        atomicAdd(&Gt[(long)(ct * 64 + c_in) * Fp + ft * 64 + f_in],
-                  acc[fm][fn][r]);
+        int c_in = wave * cpw + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 64 + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// fwd_d3_kernel: P=3 fast path -- permuted/swizzled LDS + 2-phase
+// double-buffered staging (issue next k-chunk's global loads into
+// registers while MFMA consumes the current LDS buffer; guide T3 minimum
+// recipe) + want_loss gating.
+// ---------------------------------------------------------------------------
+template <int NF>
+__global__ __launch_bounds__(256)
+void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                   const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                   const float* __restrict__ bias,
+                   const int* __restrict__ loss_id,
+                   const int* __restrict__ target, __bf16* __restrict__ Dt,
+                   float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                   int nt, int NTp, int Fp, int Bp, int want_loss) {
+  __shared__ __bf16 sX[2][64 * 64];
+  __shared__ __bf16 sW[2][NF * 64 * 64];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[64];
+  __shared__ float sY[64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rt = blockIdx.x;
+  constexpr int cpw = NF * 16;
+  const int g = lane >> 4;
+  const int srow = tid >> 2;
+  const int sseg = tid & 3;
+  const int wrow = tid >> 2;        // W staging: base col row
+
+  if (tid < 64) {
+    int gi = idx[rt * 64 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+  }
+  for (int i = tid; i < Bp; i += 256) { sLoss[i] = 0.f; sDb[i] = 0.f; }
+  __syncthreads();
+
+  bf16x8 rx[2], rw[2 * NF];
+  auto issue = [&](int kc) {
+    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
+    rx[0] = gx[0];
+    rx[1] = gx[1];
+#pragma unroll
+    for (int i = 0; i < NF; ++i) {
+      int c = wrow + i * 64;
+      const bf16x8* gw = reinterpret_cast<const bf16x8*>(
+          Wt + (long)c * Fp + kc + sseg * 16);
+      rw[2 * i] = gw[0];
+      rw[2 * i + 1] = gw[1];
+    }
+  };
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+#pragma unroll
+    for (int i = 0; i < NF; ++i) {
+      int c = wrow + i * 64;
+      store_row8(&sW[b][c * 64], c, sseg * 16, rw[2 * i]);
+      store_row8(&sW[b][c * 64], c, sseg * 16 + 8, rw[2 * i + 1]);
+    }
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  issue(0);
+  commit(0);
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = wave * cpw + fn * 16 + (lane & 15);
+        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (kc + 1 < nkc) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = wave * cpw + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 64 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = 1.f / (1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 64 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+    atomicAdd(&sDb[col_in], dsum);
+  }
+  __syncthreads();
+  for (int i = tid; i < Bp; i += 256) {
+    if (want_loss) atomicAdd(&loss_sum[i], sLoss[i]);
+    atomicAdd(&db_sum[i], sDb[i]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bwd_g3_kernel: P=3 fast path -- permuted/swizzled LDS + 2-phase
+// double-buffered staging; reads the resident transpose Xt with a
+// contiguity fast path for CV-fold index runs.
+// ---------------------------------------------------------------------------
+template <int NF>
+__global__ __launch_bounds__(256)
+void bwd_g3_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                   const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                   long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  __shared__ __bf16 sD[2][NF * 64 * 64];
+  __shared__ __bf16 sXt[2][64 * 64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  constexpr int cpw = NF * 16;
+  const int g = lane >> 4;
+  const int drow = tid >> 2;
+  const int sseg = tid & 3;
+  const int xf = tid >> 2;           // Xt staging: feature row
+
+  f32x4 acc[NF][4];
+#pragma unroll
+  for (int i = 0; i < NF; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 rd[2 * NF], rxa, rxb;
+  auto issue = [&](int rc) {
+#pragma unroll
+    for (int i = 0; i < NF; ++i) {
+      int c = drow + i * 64;
+      const bf16x8* gd = reinterpret_cast<const bf16x8*>(
+          Dt + (long)c * NTp + rc + sseg * 16);
+      rd[2 * i] = gd[0];
+      rd[2 * i + 1] = gd[1];
+    }
+    int rbase = rc + sseg * 16;
+    const __bf16* src = Xt + (long)(ft * 64 + xf) * n_rows;
+    int i0 = idx[rbase];
+    if (idx[rbase + 15] - i0 == 15 &&
+        (((unsigned long)(src + i0) & 15) == 0)) {
+      const bf16x8* gx = reinterpret_cast<const bf16x8*>(src + i0);
+      rxa = gx[0];
+      rxb = gx[1];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxa[j] = src[idx[rbase + j]];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxb[j] = src[idx[rbase + 8 + j]];
+    }
+  };
+  auto commit = [&](int b) {
+#pragma unroll
+    for (int i = 0; i < NF; ++i) {
+      int c = drow + i * 64;
+      store_row8(&sD[b][c * 64], c, sseg * 16, rd[2 * i]);
+      store_row8(&sD[b][c * 64], c, sseg * 16 + 8, rd[2 * i + 1]);
+    }
+    store_row8(&sXt[b][xf * 64], xf, sseg * 16, rxa);
+    store_row8(&sXt[b][xf * 64], xf, sseg * 16 + 8, rxb);
+  };
+
+  issue(r0);
+  commit(0);
+  __syncthreads();
+  for (int rc = r0; rc < r1; rc += 64) {
+    int b = (int)(((rc - r0) >> 6) & 1);
+    if (rc + 64 < r1) issue(rc + 64);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fn * 16 + (lane & 15);
+        bfr[fn] = lds_frag_p(&sXt[b][f * 64], f, g, h);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = wave * cpw + m * 16 + (lane & 15);
+        bf16x8 a = lds_frag_p(&sD[b][c * 64], c, g, h);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (rc + 64 < r1) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = wave * cpw + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 64 + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// fwd_d5_kernel: Bp=256 wide-block variant -- 512 threads / 8 waves,
+// block tile 128 rows x 256 cols (waves 0-3: rows 0-63, waves 4-7:
+// rows 64-127; each wave 64x64 output).  Same permuted LDS + 2-phase
+// pipeline as fwd_d3 but W staging is amortized over twice the rows.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512)
+void fwd_d5_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                   const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                   const float* __restrict__ bias,
+                   const int* __restrict__ loss_id,
+                   const int* __restrict__ target, __bf16* __restrict__ Dt,
+                   float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                   int nt, int NTp, int Fp, int Bp, int want_loss) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][128 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[128];
+  __shared__ float sY[128];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;        // column group
+  const int rhalf = wave >> 2;       // row half (0/1)
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int srow = tid >> 2;         // X staging row 0..127
+  const int sseg = tid & 3;
+  const int wrow = tid >> 3;         // W staging row 0..63 (x4 = 256)
+  const int wseg2 = (tid >> 2) & 1;  // which 32-elem half
+
+  if (tid < 128) {
+    int gi = idx[rt * 128 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+  }
+  if (tid < 256) { sLoss[tid] = 0.f; sDb[tid] = 0.f; }
+  __syncthreads();
+
+  // X: 128x64 elems / 512 thr = 16 each; W: 256x64 / 512 = 32 each
+  bf16x8 rx[2], rw[4];
+  auto issue = [&](int kc) {
+    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
+    rx[0] = gx[0];
+    rx[1] = gx[1];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = wrow + i * 128;
+      const bf16x8* gw = reinterpret_cast<const bf16x8*>(
+          Wt + (long)c * Fp + kc + wseg2 * 32);
+      rw[2 * i] = gw[0];
+      rw[2 * i + 1] = gw[1];
+    }
+  };
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = wrow + i * 128;
+      store_row8(&sW[b][c * 64], c, wseg2 * 32, rw[2 * i]);
+      store_row8(&sW[b][c * 64], c, wseg2 * 32 + 8, rw[2 * i + 1]);
+    }
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  issue(0);
+  commit(0);
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rhalf * 64 + m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (kc + 1 < nkc) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rhalf * 64 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 128 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = 1.f / (1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 128 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+    atomicAdd(&sDb[col_in], dsum);
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bwd_g5_kernel: Bp=256 wide-block variant -- 512 threads / 8 waves,
+// output tile 256 cands x 128 features (waves 0-3: f 0-63, waves 4-7:
+// f 64-127).  grid (Fp/128, nseg).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512)
+void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                   const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                   long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sD[2][256 * 64];
+  __shared__ __bf16 sXt[2][128 * 64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int fhalf = wave >> 2;
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  const int g = lane >> 4;
+  const int drow = tid >> 3;          // D staging: 64 base rows x4
+  const int dseg2 = (tid >> 2) & 1;
+  const int sseg = tid & 3;
+  const int xf = tid >> 2;            // Xt staging: feature 0..127
+
+  f32x4 acc[NF][4];
+#pragma unroll
+  for (int i = 0; i < NF; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 rd[4], rxa, rxb;
+  auto issue = [&](int rc) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = drow + i * 128;
+      const bf16x8* gd = reinterpret_cast<const bf16x8*>(
+          Dt + (long)c * NTp + rc + dseg2 * 32);
+      rd[2 * i] = gd[0];
+      rd[2 * i + 1] = gd[1];
+    }
+    int rbase = rc + sseg * 16;
+    const __bf16* src = Xt + (long)(ft * 128 + xf) * n_rows;
+    int i0 = idx[rbase];
+    if (idx[rbase + 15] - i0 == 15 &&
+        (((unsigned long)(src + i0) & 15) == 0)) {
+      const bf16x8* gx = reinterpret_cast<const bf16x8*>(src + i0);
+      rxa = gx[0];
+      rxb = gx[1];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxa[j] = src[idx[rbase + j]];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxb[j] = src[idx[rbase + 8 + j]];
+    }
+  };
+  auto commit = [&](int b) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = drow + i * 128;
+      store_row8(&sD[b][c * 64], c, dseg2 * 32, rd[2 * i]);
+      store_row8(&sD[b][c * 64], c, dseg2 * 32 + 8, rd[2 * i + 1]);
+    }
+    store_row8(&sXt[b][xf * 64], xf, sseg * 16, rxa);
+    store_row8(&sXt[b][xf * 64], xf, sseg * 16 + 8, rxb);
+  };
+
+  issue(r0);
+  commit(0);
+  __syncthreads();
+  for (int rc = r0; rc < r1; rc += 64) {
+    int b = (int)(((rc - r0) >> 6) & 1);
+    if (rc + 64 < r1) issue(rc + 64);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fhalf * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = lds_frag_p(&sXt[b][f * 64], f, g, h);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = cwave * 64 + m * 16 + (lane & 15);
+        bf16x8 a = lds_frag_p(&sD[b][c * 64], c, g, h);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (rc + 64 < r1) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fhalf * 64 + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cwave * 64 + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 128 + f_in], acc[m][fn][r]);
       }
     }
   }
@@ -380,7 +965,8 @@ __global__ void col_finalize_kernel(
     const int* __restrict__ max_iter, const float* __restrict__ lr,
     const float* __restrict__ mom, const float* __restrict__ tol,
     const unsigned char* __restrict__ fit_ic,
-    float inv_n, int epoch, int Bp, int* __restrict__ n_active) {
+    float inv_n, int epoch, int Bp, int want_loss,
+    int* __restrict__ n_active) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= Bp) return;
   float cur = loss_sum[c] * inv_n;
@@ -388,13 +974,16 @@ __global__ void col_finalize_kernel(
   epoch_mask[c] = ea;
   if (ea) {
     n_iter[c] = epoch + 1;
-    cur_loss[c] = cur;
     float gb = db_sum[c] * inv_n * (float)fit_ic[c];
     float vb = mom[c] * Vb[c] - lr[c] * gb;
     Vb[c] = vb;
     bias[c] += vb;
-    if (fabsf(prev_loss[c] - cur) <= tol[c] * fmaxf(cur, 1.f)) active[c] = 0;
-    prev_loss[c] = cur;
+    if (want_loss) {
+      cur_loss[c] = cur;
+      if (fabsf(prev_loss[c] - cur) <= tol[c] * fmaxf(cur, 1.f))
+        active[c] = 0;
+      prev_loss[c] = cur;
+    }
   }
   loss_sum[c] = 0.f;
   db_sum[c] = 0.f;
@@ -435,7 +1024,7 @@ void update_w_kernel(float* __restrict__ W, float* __restrict__ V,
 // grid ceil(n_idx/32); block 256 (4 waves side by side over Bp cols).
 // dynamic LDS: zbuf f32[32][Bp+8] + sX[32][72] + sW[Bp][72] + scand[ncand]
 // ---------------------------------------------------------------------------
-template <int P>
+template <int P, int NF>
 __global__ __launch_bounds__(256)
 void scores_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                    const long* __restrict__ idx,   // [n_idx] (no padding)
@@ -457,8 +1046,7 @@ void scores_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int rt = blockIdx.x;
-  const int colsPerWave = Bp / 4;            // Bp in {64,128,256}
-  const int nfrag = colsPerWave / 16;
+  constexpr int colsPerWave = NF * 16;       // Bp in {64,128,256}
 
   if (tid < 32) {
     long r = (long)rt * 32 + tid;
@@ -493,7 +1081,8 @@ void scores_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     for (int ks = 0; ks < 64; ks += 32) {
       bf16x8 a0 = lds_frag<AMAP>(&sX[(lane & 15) * 72 + ks], lane);
       bf16x8 a1 = lds_frag<AMAP>(&sX[(16 + (lane & 15)) * 72 + ks], lane);
-      for (int fn = 0; fn < nfrag; ++fn) {
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
         int colbase = wave * colsPerWave + fn * 16 + (lane & 15);
         bf16x8 b = lds_frag<BMAP>(&sW[colbase * 72 + ks], lane);
         acc[0][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc[0][fn], 0, 0, 0);
@@ -504,8 +1093,10 @@ void scores_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   }
 
   // spill z to LDS (add bias)
+#pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
-    for (int fn = 0; fn < nfrag; ++fn) {
+#pragma unroll
+    for (int fn = 0; fn < NF; ++fn) {
       int col = wave * colsPerWave + fn * 16 + (lane & 15);
       float bia = bias[col];
       for (int r = 0; r < 4; ++r) {
@@ -618,7 +1209,7 @@ double msq_rows(torch::Tensor X) {
 // Returns (W [Bp][Fp] f32, bias [Bp], n_iter [Bp] i32, cur_loss [Bp] f32,
 //          Wt [Bp][Fp] bf16 -- kept on device for the scores kernel).
 std::vector<torch::Tensor> linear_fit(
-    torch::Tensor X, torch::Tensor y, torch::Tensor idx,
+    torch::Tensor X, torch::Tensor Xt, torch::Tensor y, torch::Tensor idx,
     torch::Tensor loss_id, torch::Tensor lr, torch::Tensor l1,
     torch::Tensor l2, torch::Tensor mom, torch::Tensor tol,
     torch::Tensor max_iter, torch::Tensor target, torch::Tensor fit_ic,
@@ -626,11 +1217,15 @@ std::vector<torch::Tensor> linear_fit(
   TORCH_CHECK(g_layout >= 0, "MFMA layout not probed; call ensure_layout()");
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kBFloat16 && X.dim() == 2);
   TORCH_CHECK(X.stride(1) == 1 && X.stride(0) == X.size(1));
+  TORCH_CHECK(Xt.is_cuda() && Xt.dtype() == torch::kBFloat16 &&
+              Xt.size(0) == X.size(1) && Xt.size(1) == X.size(0) &&
+              Xt.stride(1) == 1);
   const at::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
   int Fp = (int)X.size(1);
   TORCH_CHECK(Fp % 64 == 0, "Fp must be a multiple of 64");
   int Bp = (int)loss_id.size(0);
-  TORCH_CHECK(Bp % 64 == 0, "Bp must be a multiple of 64");
+  TORCH_CHECK(Bp % 64 == 0 && Bp <= 256,
+              "Bp must be 64, 128 or 256 (chunk larger batches)");
   long NTp = idx.size(0);
   TORCH_CHECK(NTp % 64 == 0 && nt <= NTp);
   TORCH_CHECK(idx.dtype() == torch::kInt32);
@@ -659,21 +1254,25 @@ std::vector<torch::Tensor> linear_fit(
   hipStream_t s = cur_stream();
 
   // split-K segmentation: target ~1024 blocks for the bwd launch
-  int ctiles = Bp / 64, ftiles = Fp / 64;
+  int ftiles = Fp / 64;
   long rtiles = NTp / 64;
-  long want_nseg = std::max<long>(1, 1024 / (ctiles * ftiles));
+  long want_nseg = std::max<long>(1, 1024 / ftiles);
   long seg_tiles = std::max<long>(1, (rtiles + want_nseg - 1) / want_nseg);
   long nseg = (rtiles + seg_tiles - 1) / seg_tiles;
   int seg_rows = (int)(seg_tiles * 64);
 
-  dim3 fwd_grid((unsigned)(NTp / 64), (unsigned)(Bp / 64));
-  dim3 bwd_grid((unsigned)ctiles, (unsigned)ftiles, (unsigned)nseg);
+  dim3 fwd_grid((unsigned)(NTp / 64));
+  dim3 bwd_grid((unsigned)ftiles, (unsigned)nseg);
   dim3 fin_grid((unsigned)((Bp + 255) / 256));
   TORCH_CHECK(Fp % 256 == 0 || Fp == 64 || Fp == 128,
               "Fp must be 64, 128 or a multiple of 256");
   int upd_threads = Fp >= 256 ? 256 : Fp;
   dim3 upd_grid2((unsigned)(Fp / upd_threads), (unsigned)Bp);
 
+  const __bf16* Xtp = reinterpret_cast<const __bf16*>(Xt.data_ptr());
+  long n_rows = X.size(0);
+  bool any_tol = tol.max().item<float>() > 0.f;
+  int want_loss_cur = 1;
   auto launch_epoch = [&](int epoch, auto fwd_k, auto bwd_k) {
     hipLaunchKernelGGL(fwd_k, fwd_grid, dim3(256), 0, s, Xp,
                        y.data_ptr<float>(), idx.data_ptr<int>(),
@@ -682,11 +1281,12 @@ std::vector<torch::Tensor> linear_fit(
                        target.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(Dt.data_ptr()),
                        loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
-                       (int)nt, (int)NTp, Fp);
-    hipLaunchKernelGGL(bwd_k, bwd_grid, dim3(256), 0, s, Xp,
+                       (int)nt, (int)NTp, Fp, Bp, want_loss_cur);
+    hipLaunchKernelGGL(bwd_k, bwd_grid, dim3(256), 0, s, Xtp,
                        idx.data_ptr<int>(),
                        reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                       Gt.data_ptr<float>(), (int)NTp, Fp, seg_rows);
+                       Gt.data_ptr<float>(), n_rows, (int)NTp, Fp, Bp,
+                       seg_rows);
     hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
                        loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
                        bias.data_ptr<float>(), Vb.data_ptr<float>(),
@@ -696,7 +1296,7 @@ std::vector<torch::Tensor> linear_fit(
                        lr.data_ptr<float>(), mom.data_ptr<float>(),
                        tol.data_ptr<float>(),
                        fit_ic.data_ptr<unsigned char>(), inv_n, epoch, Bp,
-                       n_active.data_ptr<int>());
+                       want_loss_cur, n_active.data_ptr<int>());
     hipLaunchKernelGGL(update_w_kernel, upd_grid2, dim3(upd_threads), 0, s,
                        W.data_ptr<float>(), V.data_ptr<float>(),
                        Gt.data_ptr<float>(),
@@ -706,12 +1306,74 @@ std::vector<torch::Tensor> linear_fit(
                        l2.data_ptr<float>(), inv_n, Fp);
   };
 
+  int nf = Bp / 64;   // fragments per wave: 1, 2 or 4
   for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
-    switch (g_layout) {
-      case 0: launch_epoch(epoch, fwd_d_kernel<0>, bwd_g_kernel<0>); break;
-      case 1: launch_epoch(epoch, fwd_d_kernel<1>, bwd_g_kernel<1>); break;
-      case 2: launch_epoch(epoch, fwd_d_kernel<2>, bwd_g_kernel<2>); break;
-      default: launch_epoch(epoch, fwd_d_kernel<3>, bwd_g_kernel<3>); break;
+    want_loss_cur = (any_tol || epoch == (int)max_epochs - 1) ? 1 : 0;
+    if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
+      if (nf == 4 && NTp % 128 == 0 && Fp % 128 == 0) {
+        // wide 512-thread blocks (128-row / 128-feature tiles)
+        hipLaunchKernelGGL(fwd_d5_kernel, dim3((unsigned)(NTp / 128)),
+                           dim3(512), 0, s, Xp, y.data_ptr<float>(),
+                           idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                           bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+                           target.data_ptr<int>(),
+                           reinterpret_cast<__bf16*>(Dt.data_ptr()),
+                           loss_sum.data_ptr<float>(),
+                           db_sum.data_ptr<float>(), (int)nt, (int)NTp, Fp,
+                           Bp, want_loss_cur);
+        hipLaunchKernelGGL(bwd_g5_kernel,
+                           dim3((unsigned)(Fp / 128), (unsigned)nseg),
+                           dim3(512), 0, s, Xtp, idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Dt.data_ptr()),
+                           Gt.data_ptr<float>(), n_rows, (int)NTp, Fp, Bp,
+                           seg_rows);
+        hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
+                           loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
+                           bias.data_ptr<float>(), Vb.data_ptr<float>(),
+                           prev_loss.data_ptr<float>(), cur_loss.data_ptr<float>(),
+                           active.data_ptr<int>(), epoch_mask.data_ptr<int>(),
+                           n_iter.data_ptr<int>(), max_iter.data_ptr<int>(),
+                           lr.data_ptr<float>(), mom.data_ptr<float>(),
+                           tol.data_ptr<float>(),
+                           fit_ic.data_ptr<unsigned char>(), inv_n, epoch, Bp,
+                           want_loss_cur, n_active.data_ptr<int>());
+        hipLaunchKernelGGL(update_w_kernel, upd_grid2, dim3(upd_threads), 0, s,
+                           W.data_ptr<float>(), V.data_ptr<float>(),
+                           Gt.data_ptr<float>(),
+                           reinterpret_cast<__bf16*>(Wt.data_ptr()),
+                           epoch_mask.data_ptr<int>(), lr.data_ptr<float>(),
+                           mom.data_ptr<float>(), l1.data_ptr<float>(),
+                           l2.data_ptr<float>(), inv_n, Fp);
+      } else switch (nf) {
+        case 1: launch_epoch(epoch, fwd_d3_kernel<1>, bwd_g3_kernel<1>); break;
+        case 2: launch_epoch(epoch, fwd_d3_kernel<2>, bwd_g3_kernel<2>); break;
+        default: launch_epoch(epoch, fwd_d3_kernel<4>, bwd_g3_kernel<4>); break;
+      }
+      HIP_CHECK(hipGetLastError());
+      if (check_every > 0 && (epoch + 1) % check_every == 0 &&
+          epoch + 1 < max_epochs) {
+        int na = n_active.item<int>();
+        if (na == 0) break;
+        n_active.zero_();
+      } else {
+        HIP_CHECK(hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s));
+      }
+      continue;
+    }
+    switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
+      case 0:  launch_epoch(epoch, fwd_d2_kernel<0, 1>, bwd_g2_kernel<0, 1>); break;
+      case 1:  launch_epoch(epoch, fwd_d2_kernel<0, 2>, bwd_g2_kernel<0, 2>); break;
+      case 2:  launch_epoch(epoch, fwd_d2_kernel<0, 4>, bwd_g2_kernel<0, 4>); break;
+      case 4:  launch_epoch(epoch, fwd_d2_kernel<1, 1>, bwd_g2_kernel<1, 1>); break;
+      case 5:  launch_epoch(epoch, fwd_d2_kernel<1, 2>, bwd_g2_kernel<1, 2>); break;
+      case 6:  launch_epoch(epoch, fwd_d2_kernel<1, 4>, bwd_g2_kernel<1, 4>); break;
+      case 8:  launch_epoch(epoch, fwd_d2_kernel<2, 1>, bwd_g2_kernel<2, 1>); break;
+      case 9:  launch_epoch(epoch, fwd_d2_kernel<2, 2>, bwd_g2_kernel<2, 2>); break;
+      case 10: launch_epoch(epoch, fwd_d2_kernel<2, 4>, bwd_g2_kernel<2, 4>); break;
+      case 12: launch_epoch(epoch, fwd_d2_kernel<3, 1>, bwd_g2_kernel<3, 1>); break;
+      case 13: launch_epoch(epoch, fwd_d2_kernel<3, 2>, bwd_g2_kernel<3, 2>); break;
+      default: launch_epoch(epoch, fwd_d2_kernel<3, 4>, bwd_g2_kernel<3, 4>); break;
     }
     HIP_CHECK(hipGetLastError());
     if (check_every > 0 && (epoch + 1) % check_every == 0 &&
@@ -761,11 +1423,20 @@ torch::Tensor linear_scores(torch::Tensor X, torch::Tensor y,
                        target.data_ptr<int>(), (int)C, (int)metric, n_idx,
                        Fp, Bp, (int)n_cand, out.data_ptr<double>());
   };
-  switch (g_layout) {
-    case 0: launch(scores_kernel<0>); break;
-    case 1: launch(scores_kernel<1>); break;
-    case 2: launch(scores_kernel<2>); break;
-    default: launch(scores_kernel<3>); break;
+  int nf = Bp / 64;
+  switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
+    case 0:  launch(scores_kernel<0, 1>); break;
+    case 1:  launch(scores_kernel<0, 2>); break;
+    case 2:  launch(scores_kernel<0, 4>); break;
+    case 4:  launch(scores_kernel<1, 1>); break;
+    case 5:  launch(scores_kernel<1, 2>); break;
+    case 6:  launch(scores_kernel<1, 4>); break;
+    case 8:  launch(scores_kernel<2, 1>); break;
+    case 9:  launch(scores_kernel<2, 2>); break;
+    case 10: launch(scores_kernel<2, 4>); break;
+    case 12: launch(scores_kernel<3, 1>); break;
+    case 13: launch(scores_kernel<3, 2>); break;
+    default: launch(scores_kernel<3, 4>); break;
   }
   HIP_CHECK(hipGetLastError());
   return out;
