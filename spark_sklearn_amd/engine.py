@@ -2,9 +2,9 @@
 
 The key idiomatic-GPU decision with no reference analogue (SURVEY.md
 section 7.1): the reference runs 1 fit per CPU task; a lone 256-feature fit
-cannot feed 256 CUs.  Here B candidates (x OVR classes) fold into ONE
-batched kernel launch -- W becomes [F, B] -- turning a 10k-candidate grid
-into ~10k/B large MFMA launches per GPU.
+cannot feed 256 CUs.  Here B candidates (x OVR classes / centroid blocks)
+fold into ONE batched kernel launch, turning a 10k-candidate grid into
+~10k/B large MFMA launches per GPU.
 
 Execution tiers:
 
@@ -12,12 +12,15 @@ Execution tiers:
   (ops/csrc/*.hip, gfx950 MFMA kernels).  The extension is REQUIRED on a
   GPU box -- a missing .so raises instead of silently falling back, so GPU
   test runs genuinely exercise the native path.
-* CPU: the numpy reference solver (models/_solver_ref.py), which defines
-  the numerics the kernels must match.
+* CPU: the numpy reference solvers (models/_solver_ref.py, _kmeans_ref.py),
+  which define the numerics the kernels must match.
 
-Both tiers share plan construction, scoring semantics and timing
-attribution, so the CPU tier is an honest behavioral twin (BASELINE
-config 1).
+Estimator families plug in through a small adapter protocol (``FAMILIES``):
+encode targets, build a batch plan, create host/device executors, map
+sklearn scorers to device-computable metrics, and install fitted parameters
+back onto estimator objects.  ``run_batched_tasks`` itself is family-
+agnostic: fold grouping, candidate chunking, timing attribution and the
+error_score contract are shared.
 """
 
 from __future__ import annotations
@@ -31,115 +34,141 @@ import numpy as np
 from sklearn.base import clone, is_classifier
 
 from .base_search import Task, TaskResult
-from .models._solver_ref import (BatchPlan, linear_fit_ref,
-                                 linear_scores_ref)
 
-__all__ = ["run_batched_tasks", "fit_single"]
+__all__ = ["run_batched_tasks", "fit_single", "FAMILIES"]
 
-DEFAULT_COLUMN_BATCH = 256   # columns per kernel launch (candidates x classes)
+DEFAULT_COLUMN_BATCH = 256   # columns per kernel launch
 
 
-# --------------------------------------------------------------------------
-# plan construction (shared by CPU and GPU tiers)
-# --------------------------------------------------------------------------
+def _device_or_none(device):
+    """Resolve execution device; native extension mandatory on GPU."""
+    try:
+        import torch
 
-def _encode_targets(estimator, y):
-    """Classifier: labels -> class indices + classes_.  Regressor: float y."""
-    y = np.asarray(y)
-    if is_classifier(estimator):
-        classes, y_enc = np.unique(y, return_inverse=True)
-        return classes, y_enc.astype(np.int32)
-    return None, np.asarray(y, dtype=np.float32)
-
-
-def _build_plan(estimator, params_list: Sequence[dict], n_samples: int,
-                mean_sq_row_norm: float, n_classes: int,
-                error_score) -> tuple:
-    """Expand candidates into per-column kernel parameters.
-
-    Returns (plan, bad) where ``bad`` maps candidate-slot -> exception for
-    candidates whose parameters failed validation (error_score contract:
-    one bad candidate must not kill the batch).
-    """
-    cols_per_cand = n_classes if n_classes > 2 else 1
-    specs, bad = [], {}
-    # one prototype clone + plain dict merge per candidate: sklearn clone()
-    # costs ~30us each, which at 10k candidates is seconds of pure host
-    # overhead between kernel launches
-    proto = clone(estimator)
-    base_params = proto.get_params()
-    spec_cls = type(proto)
-    for slot, params in enumerate(params_list):
-        try:
-            unknown = set(params) - set(base_params)
-            if unknown:
-                raise ValueError(
-                    f"invalid parameters {sorted(unknown)} for "
-                    f"{spec_cls.__name__}")
-            proto.__dict__.update(base_params)
-            proto.__dict__.update(params)
-            spec = proto._column_spec(n_samples, mean_sq_row_norm)
-        except Exception as exc:  # noqa: BLE001 - error_score contract
-            if error_score == "raise":
-                raise
-            bad[slot] = exc
-            spec = dict(loss=0, lr=0.0, l1=0.0, l2=0.0, momentum=0.0,
-                        tol=0.0, max_iter=0, fit_intercept=0)
-        specs.append(spec)
-
-    B = len(specs) * cols_per_cand
-    plan = BatchPlan(
-        loss=np.empty(B, np.int32), lr=np.empty(B, np.float32),
-        l1=np.empty(B, np.float32), l2=np.empty(B, np.float32),
-        momentum=np.empty(B, np.float32), tol=np.empty(B, np.float32),
-        max_iter=np.empty(B, np.int32), target_class=np.empty(B, np.int32),
-        col_to_cand=np.empty(B, np.int32), n_classes=cols_per_cand,
-        fit_intercept=np.empty(B, np.uint8))
-    for slot, spec in enumerate(specs):
-        for c in range(cols_per_cand):
-            j = slot * cols_per_cand + c
-            plan.loss[j] = spec["loss"]
-            plan.lr[j] = spec["lr"]
-            plan.l1[j] = spec["l1"]
-            plan.l2[j] = spec["l2"]
-            plan.momentum[j] = spec["momentum"]
-            plan.tol[j] = spec["tol"]
-            plan.max_iter[j] = spec["max_iter"]
-            plan.fit_intercept[j] = spec["fit_intercept"]
-            if n_classes == 0:          # regression
-                plan.target_class[j] = -1
-            elif cols_per_cand == 1:    # binary: discriminate class index 1
-                plan.target_class[j] = 1
-            else:
-                plan.target_class[j] = c
-            plan.col_to_cand[j] = slot
-    return plan, bad
+        if device is not None and getattr(device, "type", None) == "cuda":
+            return device
+        if device is None and torch.cuda.is_available():
+            return torch.device("cuda", torch.cuda.current_device())
+    except ImportError:
+        pass
+    return None
 
 
-def _scorer_metric(scorer, estimator) -> Optional[str]:
-    """Map an sklearn scorer onto a device-computable metric name."""
-    if scorer is None:
-        return "accuracy" if is_classifier(estimator) else "r2"
-    if type(scorer).__name__ == "_PassthroughScorer":
-        return "accuracy" if is_classifier(estimator) else "r2"
-    fn = getattr(scorer, "_score_func", None)
-    name = getattr(fn, "__name__", "")
-    sign = getattr(scorer, "_sign", 1)
-    if name == "accuracy_score":
-        return "accuracy"
-    if name == "r2_score":
-        return "r2"
-    if name == "log_loss" and sign == -1:
-        return "neg_log_loss"
-    return None   # unsupported -> host-scoring fallback
+class LinearFamily:
+    """Adapter for the linear estimator family (logistic/hinge/squared)."""
 
+    name = "linear"
 
-# --------------------------------------------------------------------------
-# CPU tier
-# --------------------------------------------------------------------------
+    @staticmethod
+    def encode_targets(estimator, y):
+        y = np.asarray(y)
+        if is_classifier(estimator):
+            classes, y_enc = np.unique(y, return_inverse=True)
+            return classes, y_enc.astype(np.int32)
+        return None, np.asarray(y, dtype=np.float32)
+
+    @staticmethod
+    def cols_per_cand(n_classes):
+        return n_classes if n_classes > 2 else 1
+
+    @staticmethod
+    def cand_chunk(estimator, params_list, column_batch):
+        # resolved against the label count at run time by the caller
+        return None
+
+    @staticmethod
+    def make_executor(X, y_enc, classes, device):
+        if device is not None:
+            from .ops import DeviceLinearExecutor
+
+            return DeviceLinearExecutor(X, y_enc, classes, device)
+        return _HostLinearExecutor(X, y_enc, classes)
+
+    @staticmethod
+    def build_plan(estimator, params_list, n_samples, executor, n_classes,
+                   error_score):
+        from .models._solver_ref import BatchPlan
+
+        cols_per_cand = LinearFamily.cols_per_cand(n_classes)
+        specs, bad = [], {}
+        # one prototype + dict merge per candidate (sklearn clone costs
+        # ~30us each: seconds of host overhead at 10k candidates)
+        proto = clone(estimator)
+        base_params = proto.get_params()
+        for slot, params in enumerate(params_list):
+            try:
+                unknown = set(params) - set(base_params)
+                if unknown:
+                    raise ValueError(
+                        f"invalid parameters {sorted(unknown)} for "
+                        f"{type(proto).__name__}")
+                proto.__dict__.update(base_params)
+                proto.__dict__.update(params)
+                spec = proto._column_spec(n_samples, executor.msq)
+            except Exception as exc:  # noqa: BLE001 - error_score contract
+                if error_score == "raise":
+                    raise
+                bad[slot] = exc
+                spec = dict(loss=0, lr=0.0, l1=0.0, l2=0.0, momentum=0.0,
+                            tol=0.0, max_iter=0, fit_intercept=0)
+            specs.append(spec)
+
+        B = len(specs) * cols_per_cand
+        plan = BatchPlan(
+            loss=np.empty(B, np.int32), lr=np.empty(B, np.float32),
+            l1=np.empty(B, np.float32), l2=np.empty(B, np.float32),
+            momentum=np.empty(B, np.float32), tol=np.empty(B, np.float32),
+            max_iter=np.empty(B, np.int32),
+            target_class=np.empty(B, np.int32),
+            col_to_cand=np.empty(B, np.int32), n_classes=cols_per_cand,
+            fit_intercept=np.empty(B, np.uint8))
+        for slot, spec in enumerate(specs):
+            for c in range(cols_per_cand):
+                j = slot * cols_per_cand + c
+                plan.loss[j] = spec["loss"]
+                plan.lr[j] = spec["lr"]
+                plan.l1[j] = spec["l1"]
+                plan.l2[j] = spec["l2"]
+                plan.momentum[j] = spec["momentum"]
+                plan.tol[j] = spec["tol"]
+                plan.max_iter[j] = spec["max_iter"]
+                plan.fit_intercept[j] = spec["fit_intercept"]
+                if n_classes == 0:
+                    plan.target_class[j] = -1
+                elif cols_per_cand == 1:
+                    plan.target_class[j] = 1
+                else:
+                    plan.target_class[j] = c
+                plan.col_to_cand[j] = slot
+        return plan, bad
+
+    @staticmethod
+    def metric_for(scorer, estimator) -> Optional[str]:
+        if scorer is None or type(scorer).__name__ == "_PassthroughScorer":
+            return "accuracy" if is_classifier(estimator) else "r2"
+        fn = getattr(scorer, "_score_func", None)
+        name = getattr(fn, "__name__", "")
+        sign = getattr(scorer, "_sign", 1)
+        if name == "accuracy_score":
+            return "accuracy"
+        if name == "r2_score":
+            return "r2"
+        if name == "log_loss" and sign == -1:
+            return "neg_log_loss"
+        return None
+
+    @staticmethod
+    def install_fitted(est, fitted, slot, plan, classes):
+        W, b = fitted
+        C = plan.n_classes
+        cols = slice(slot * C, (slot + 1) * C)
+        est._finalize_fit(np.ascontiguousarray(W[:, cols]),
+                          np.ascontiguousarray(b[cols]), None, classes)
+        return est
+
 
 class _HostLinearExecutor:
-    """Numpy execution of a candidate batch (reference semantics)."""
+    """Numpy execution of a linear candidate batch (reference semantics)."""
 
     def __init__(self, X, y_enc, classes):
         self.X = np.ascontiguousarray(np.asarray(X), dtype=np.float32)
@@ -149,38 +178,31 @@ class _HostLinearExecutor:
                                         axis=1)))
 
     def fit(self, train_idx, plan):
-        return linear_fit_ref(self.X[train_idx], self.y[train_idx], plan)
+        from .models._solver_ref import linear_fit_ref
 
-    def scores(self, idx, W, b, plan, metric):
-        # labels are already encoded to class indices, so scoring happens
-        # in encoded space (binary: target class 1 vs 0)
+        W, b, n_iter, loss = linear_fit_ref(
+            self.X[train_idx], self.y[train_idx], plan)
+        return (W, b), n_iter, loss
+
+    def scores(self, idx, fitted, plan, metric):
+        from .models._solver_ref import linear_scores_ref
+
+        W, b = fitted
+        # labels already encoded -> scoring in encoded space
         return linear_scores_ref(self.X[idx], self.y[idx], W, b, plan,
                                  metric=metric, classes=None)
 
 
-def _get_executor(estimator, X, y_enc, classes, device):
-    """Pick the device or host executor.  On a CUDA device the native
-    extension is mandatory (fail loudly, never a silent eager fallback)."""
-    use_gpu = False
-    try:
-        import torch
-
-        if device is not None and getattr(device, "type", None) == "cuda":
-            use_gpu = True
-        elif device is None and torch.cuda.is_available():
-            use_gpu = True
-            device = torch.device("cuda", torch.cuda.current_device())
-    except ImportError:
-        pass
-    if use_gpu:
-        from .ops import DeviceLinearExecutor
-
-        return DeviceLinearExecutor(X, y_enc, classes, device)
-    return _HostLinearExecutor(X, y_enc, classes)
+def get_family(estimator):
+    fam = getattr(type(estimator), "_engine_family", None)
+    if fam is None:
+        raise TypeError(
+            f"{type(estimator).__name__} is not a batched-engine family")
+    return FAMILIES[fam]
 
 
 # --------------------------------------------------------------------------
-# the batched task runner
+# the generic batched task runner
 # --------------------------------------------------------------------------
 
 def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
@@ -197,17 +219,22 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     if fit_params:
         warnings.warn("fit_params are ignored by the batched device path",
                       stacklevel=2)
-    classes, y_enc = _encode_targets(estimator, y)
+    family = get_family(estimator)
+    classes, y_enc = family.encode_targets(estimator, y)
     n_classes = len(classes) if classes is not None else 0
-    executor = _get_executor(estimator, X, y_enc, classes, device)
-    metric = _scorer_metric(scorer, estimator)
+    device = _device_or_none(device)
+    executor = family.make_executor(X, y_enc, classes, device)
+    metric = family.metric_for(scorer, estimator)
 
     by_fold: Dict[int, List[Task]] = {}
     for t in tasks:
         by_fold.setdefault(t.fold_index, []).append(t)
 
-    cols_per_cand = n_classes if n_classes > 2 else 1
-    cand_chunk = max(1, column_batch // cols_per_cand)
+    cols_per_cand = family.cols_per_cand(n_classes)
+    cand_chunk = family.cand_chunk(estimator, [t.params for t in tasks],
+                                   column_batch)
+    if cand_chunk is None:
+        cand_chunk = max(1, column_batch // max(cols_per_cand, 1))
     results: List[TaskResult] = []
 
     for fold_index in sorted(by_fold):
@@ -217,21 +244,21 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             chunk = fold_tasks[start:start + cand_chunk]
             params_list = [t.params for t in chunk]
             t0 = time.perf_counter()
-            plan, bad = _build_plan(
-                estimator, params_list, len(train), executor.msq,
+            plan, bad = family.build_plan(
+                estimator, params_list, len(train), executor,
                 n_classes, error_score)
-            W, b, n_iter, _ = executor.fit(train, plan)
+            fitted, n_iter, _ = executor.fit(train, plan)
             fit_time = time.perf_counter() - t0
 
             t1 = time.perf_counter()
             if metric is not None:
-                test_scores = executor.scores(test, W, b, plan, metric)
-                train_scores = (executor.scores(train, W, b, plan, metric)
+                test_scores = executor.scores(test, fitted, plan, metric)
+                train_scores = (executor.scores(train, fitted, plan, metric)
                                 if return_train_score else
                                 np.full(len(chunk), np.nan))
             else:
                 test_scores, train_scores = _host_score_fallback(
-                    estimator, executor, X, y, W, b, plan, chunk, scorer,
+                    family, estimator, X, y, fitted, plan, chunk, scorer,
                     classes, train, test, return_train_score)
             score_time = time.perf_counter() - t1
 
@@ -244,32 +271,30 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                     ts = tr = float(error_score)
                     warnings.warn(
                         f"candidate {task.params!r} failed plan validation "
-                        f"({bad[slot]!r}); scoring error_score={error_score}",
-                        stacklevel=2)
+                        f"({bad[slot]!r}); scoring error_score="
+                        f"{error_score}", stacklevel=2)
                 else:
                     ts = float(test_scores[slot])
                     tr = float(train_scores[slot])
                 results.append(TaskResult(task.index, ts, tr,
-                                          len(task.test), per_fit, per_score))
+                                          len(task.test), per_fit,
+                                          per_score))
         if verbose > 1:
             print(f"[engine] fold {fold_index}: "
                   f"{len(fold_tasks)} candidates done")
     return results
 
 
-def _host_score_fallback(estimator, executor, X, y, W, b, plan, chunk,
+def _host_score_fallback(family, estimator, X, y, fitted, plan, chunk,
                          scorer, classes, train, test, return_train_score):
     """Custom scorer: build fitted clones on the host and score each."""
-    X = np.asarray(X)
+    X = np.asarray(X) if not hasattr(X, "cpu") else X.cpu().float().numpy()
     y = np.asarray(y)
-    C = plan.n_classes
     test_scores = np.empty(len(chunk))
     train_scores = np.full(len(chunk), np.nan)
     for slot, task in enumerate(chunk):
-        cols = slice(slot * C, (slot + 1) * C)
         est = clone(estimator).set_params(**task.params)
-        est._finalize_fit(np.asarray(W[:, cols]), np.asarray(b[cols]),
-                          None, classes)
+        est = family.install_fitted(est, fitted, slot, plan, classes)
         test_scores[slot] = scorer(est, X[test], y[test])
         if return_train_score:
             train_scores[slot] = scorer(est, X[train], y[train])
@@ -282,13 +307,23 @@ def _host_score_fallback(estimator, executor, X, y, W, b, plan, chunk,
 
 def fit_single(estimator, X, y, device=None):
     """Fit one estimator (its OVR classes still batch into one launch)."""
-    classes, y_enc = _encode_targets(estimator, y)
+    family = get_family(estimator)
+    classes, y_enc = family.encode_targets(estimator, y)
     n_classes = len(classes) if classes is not None else 0
-    executor = _get_executor(estimator, X, y_enc, classes, device)
-    n = np.asarray(X).shape[0] if not hasattr(X, "shape") else X.shape[0]
-    plan, bad = _build_plan(estimator, [{}], n, executor.msq, n_classes,
-                            "raise")
+    device = _device_or_none(device)
+    executor = family.make_executor(X, y_enc, classes, device)
+    n = X.shape[0] if hasattr(X, "shape") else len(X)
+    plan, _ = family.build_plan(estimator, [{}], n, executor, n_classes,
+                                "raise")
     all_idx = np.arange(n, dtype=np.int64)
-    W, b, n_iter, _ = executor.fit(all_idx, plan)
-    return estimator._finalize_fit(np.asarray(W), np.asarray(b), n_iter,
-                                   classes)
+    fitted, n_iter, _ = executor.fit(all_idx, plan)
+    family.install_fitted(estimator, fitted, 0, plan, classes)
+    if n_iter is not None:
+        estimator.n_iter_ = n_iter
+    return estimator
+
+
+from .models.kmeans import KMeansFamily  # noqa: E402  (no import cycle:
+# models.kmeans only imports engine lazily inside methods)
+
+FAMILIES = {"linear": LinearFamily, "kmeans": KMeansFamily}

@@ -1,7 +1,8 @@
 """MI355X-native estimator families (batched CDNA4 kernel fits)."""
 
+from .kmeans import KMeans
 from .linear import (LinearRegression, LinearSVC, LogisticRegression, Ridge,
                      SGDClassifier)
 
 __all__ = ["LogisticRegression", "LinearSVC", "SGDClassifier",
-           "LinearRegression", "Ridge"]
+           "LinearRegression", "Ridge", "KMeans"]

@@ -1,0 +1,244 @@
+"""MI355X-native KMeans (Lloyd) with centroid-block candidate batching.
+
+sklearn-compatible KMeans whose inner loop is the K2 HIP kernel family
+(assign: MFMA GEMM S = X . C_all^T + per-block argmin; update: LDS-tiled
+centroid accumulation) on GPU and models/_kmeans_ref.py on CPU.  Replaces
+the sklearn Cython lloyd iteration the reference borrows (SURVEY.md C12,
+section 2.5 K2).
+
+The batching story mirrors the linear family: every (candidate, n_init
+restart) pair contributes one centroid block to a single batched C_all
+matrix, so a whole grid-search chunk advances with one GEMM per Lloyd
+iteration.  ``n_init`` restarts are just extra blocks; the best restart by
+train inertia represents the candidate (exactly sklearn's n_init
+semantics).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+
+from sklearn.base import BaseEstimator, ClusterMixin
+
+from ._kmeans_ref import (KMeansPlan, kmeans_fit_ref, kmeans_init_centers,
+                          kmeans_score_ref)
+
+__all__ = ["KMeans"]
+
+INIT_SAMPLE_ROWS = 50_000   # k-means++ seeding runs on a sample this size
+
+
+class KMeans(ClusterMixin, BaseEstimator):
+    """Lloyd KMeans on CDNA4 MFMA kernels.
+
+    Parameters follow sklearn: ``n_clusters``, ``init`` in {"k-means++",
+    "random"}, ``n_init``, ``max_iter``, ``tol`` (scaled by the mean
+    feature variance, as sklearn does), ``random_state``.
+
+    Documented divergences from sklearn (SURVEY.md section 7.3 policy --
+    compare scores, not internals): empty clusters keep their previous
+    center instead of being relocated; k-means++ seeds from a <=50k row
+    sample; tol uses the full-matrix variance rather than per-fold.
+    """
+
+    _hip_batched = True
+    _engine_family = "kmeans"
+
+    def __init__(self, n_clusters=8, init="k-means++", n_init=1,
+                 max_iter=300, tol=1e-4, random_state=None):
+        self.n_clusters = n_clusters
+        self.init = init
+        self.n_init = n_init
+        self.max_iter = max_iter
+        self.tol = tol
+        self.random_state = random_state
+
+    # -- standalone fit ----------------------------------------------------
+    def fit(self, X, y=None):
+        from ..engine import _device_or_none, get_family
+
+        family = get_family(self)
+        device = _device_or_none(None)
+        executor = family.make_executor(X, None, None, device)
+        n = X.shape[0]
+        plan = family.build_plan(self, [{}], n, executor, 0, "raise")[0]
+        all_idx = np.arange(n, dtype=np.int64)
+        fitted, n_iter, _ = executor.fit(all_idx, plan)
+        family.install_fitted(self, fitted, 0, plan, None)
+        C_all, best_block, _ = fitted
+        self.labels_ = executor.assign_labels(all_idx, self.cluster_centers_)
+        self.n_iter_ = int(n_iter[int(best_block[0])])
+        self.n_features_in_ = X.shape[1]
+        return self
+
+    def predict(self, X):
+        self._check_fitted()
+        X = np.asarray(X, dtype=np.float64)
+        C = self.cluster_centers_.astype(np.float64)
+        d = (np.sum(X ** 2, axis=1)[:, None] - 2.0 * X @ C.T +
+             np.sum(C ** 2, axis=1)[None, :])
+        return np.argmin(d, axis=1).astype(np.int32)
+
+    def transform(self, X):
+        """Distances to each cluster center (sklearn contract)."""
+        self._check_fitted()
+        X = np.asarray(X, dtype=np.float64)
+        C = self.cluster_centers_.astype(np.float64)
+        d2 = (np.sum(X ** 2, axis=1)[:, None] - 2.0 * X @ C.T +
+              np.sum(C ** 2, axis=1)[None, :])
+        return np.sqrt(np.maximum(d2, 0.0))
+
+    def fit_predict(self, X, y=None):
+        return self.fit(X, y).labels_
+
+    def score(self, X, y=None):
+        """Negative inertia of X under the fitted centers (sklearn)."""
+        self._check_fitted()
+        X = np.asarray(X, dtype=np.float64)
+        C = self.cluster_centers_.astype(np.float64)
+        d2 = (np.sum(X ** 2, axis=1)[:, None] - 2.0 * X @ C.T +
+              np.sum(C ** 2, axis=1)[None, :])
+        return -float(np.sum(np.min(d2, axis=1)))
+
+    def _check_fitted(self):
+        if not hasattr(self, "cluster_centers_"):
+            raise AttributeError("this KMeans instance is not fitted yet")
+
+
+class KMeansFamily:
+    """Engine adapter: centroid-block batching."""
+
+    name = "kmeans"
+
+    @staticmethod
+    def encode_targets(estimator, y):
+        return None, None          # unsupervised
+
+    @staticmethod
+    def cols_per_cand(n_classes):
+        return 1
+
+    @staticmethod
+    def cand_chunk(estimator, params_list, column_batch):
+        worst = 1
+        for p in params_list:
+            k = p.get("n_clusters", estimator.n_clusters)
+            ni = p.get("n_init", estimator.n_init)
+            worst = max(worst, int(k) * int(ni))
+        return max(1, column_batch // worst)
+
+    @staticmethod
+    def make_executor(X, y_enc, classes, device):
+        if device is not None:
+            from ..ops.kmeans_executor import DeviceKMeansExecutor
+
+            return DeviceKMeansExecutor(X, device)
+        return _HostKMeansExecutor(X)
+
+    @staticmethod
+    def build_plan(estimator, params_list, n_samples, executor, n_classes,
+                   error_score):
+        from sklearn.base import clone
+
+        proto = clone(estimator)
+        base_params = proto.get_params()
+        blocks, bad = [], {}
+        for slot, params in enumerate(params_list):
+            try:
+                unknown = set(params) - set(base_params)
+                if unknown:
+                    raise ValueError(
+                        f"invalid parameters {sorted(unknown)} for KMeans")
+                merged = dict(base_params, **params)
+                k = int(merged["n_clusters"])
+                ni = int(merged["n_init"])
+                if k < 1 or ni < 1:
+                    raise ValueError("n_clusters and n_init must be >= 1")
+                blocks.append((slot, k, ni, merged))
+            except Exception as exc:  # noqa: BLE001 - error_score contract
+                if error_score == "raise":
+                    raise
+                bad[slot] = exc
+                blocks.append((slot, 1, 1, dict(base_params, max_iter=0,
+                                                init="random",
+                                                random_state=0)))
+
+        cstart = [0]
+        block_to_cand, max_iters, tols = [], [], []
+        C0_parts = []
+        mean_var = executor.feature_mean_var
+        for slot, k, ni, merged in blocks:
+            seed = merged.get("random_state")
+            base_seed = seed if seed is not None else 1234 + slot
+            for r in range(ni):
+                rng = np.random.default_rng(int(base_seed) + r)
+                sample = executor.init_sample(rng)
+                C0_parts.append(
+                    kmeans_init_centers(sample, k, merged["init"], rng))
+                cstart.append(cstart[-1] + k)
+                block_to_cand.append(slot)
+                max_iters.append(int(merged["max_iter"]))
+                tols.append(float(merged["tol"]) * mean_var)
+        plan = KMeansPlan(
+            cstart=np.asarray(cstart, np.int32),
+            block_to_cand=np.asarray(block_to_cand, np.int32),
+            max_iter=np.asarray(max_iters, np.int32),
+            tol_scaled=np.asarray(tols, np.float32),
+            n_cand=len(params_list))
+        plan.C0 = np.concatenate(C0_parts, axis=0).astype(np.float32)
+        return plan, bad
+
+    @staticmethod
+    def metric_for(scorer, estimator) -> Optional[str]:
+        if scorer is None or type(scorer).__name__ == "_PassthroughScorer":
+            return "neg_inertia"
+        return None
+
+    @staticmethod
+    def install_fitted(est, fitted, slot, plan, classes):
+        C_all, best_block, inertia = fitted
+        b = int(best_block[slot])
+        lo, hi = int(plan.cstart[b]), int(plan.cstart[b + 1])
+        est.cluster_centers_ = np.array(C_all[lo:hi], dtype=np.float64)
+        est.inertia_ = float(inertia[b])
+        est.n_features_in_ = C_all.shape[1]
+        return est
+
+
+class _HostKMeansExecutor:
+    """Numpy execution of a centroid-block batch (reference semantics)."""
+
+    def __init__(self, X):
+        self.X = np.ascontiguousarray(np.asarray(X), dtype=np.float32)
+        self.feature_mean_var = float(np.mean(np.var(self.X, axis=0)))
+        self.msq = float(np.mean(np.sum(self.X.astype(np.float64) ** 2,
+                                        axis=1)))
+
+    def init_sample(self, rng) -> np.ndarray:
+        n = self.X.shape[0]
+        if n <= INIT_SAMPLE_ROWS:
+            return self.X
+        rows = rng.choice(n, size=INIT_SAMPLE_ROWS, replace=False)
+        return self.X[rows]
+
+    def fit(self, train_idx, plan):
+        C_all, n_iter, inertia = kmeans_fit_ref(
+            self.X[train_idx], plan.C0, plan)
+        best_block = np.empty(plan.n_cand, np.int32)
+        for c in range(plan.n_cand):
+            blocks = np.flatnonzero(plan.block_to_cand == c)
+            best_block[c] = blocks[np.argmin(inertia[blocks])]
+        return (C_all, best_block, inertia), n_iter, inertia
+
+    def scores(self, idx, fitted, plan, metric):
+        assert metric == "neg_inertia"
+        C_all, best_block, _ = fitted
+        return kmeans_score_ref(self.X[idx], C_all, plan, best_block)
+
+    def assign_labels(self, idx, centers) -> np.ndarray:
+        X = self.X[idx].astype(np.float64)
+        C = np.asarray(centers, dtype=np.float64)
+        d = (-2.0 * X @ C.T + np.sum(C ** 2, axis=1)[None, :])
+        return np.argmin(d, axis=1).astype(np.int32)

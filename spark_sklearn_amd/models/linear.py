@@ -35,6 +35,7 @@ class _LinearBase(BaseEstimator):
     """Shared plumbing for the batched linear family."""
 
     _hip_batched = True          # engine routes fits through ops/ kernels
+    _engine_family = "linear"
     _loss_id: int = LOSS_LOGISTIC
 
     # -- per-estimator plan hooks (engine + local fit share these) ---------

@@ -148,12 +148,12 @@ class DeviceLinearExecutor:
         torch.cuda.synchronize(self.device)
         self._fit_state = (Wt, bias, loss_t, tgt_t)
         W_np = W[:B, :self.F].float().cpu().numpy().T.copy()   # [F, B]
-        return (W_np, bias[:B].float().cpu().numpy(),
+        return ((W_np, bias[:B].float().cpu().numpy()),
                 n_iter[:B].cpu().numpy(),
                 cur_loss[:B].float().cpu().numpy())
 
     # -- scores ------------------------------------------------------------
-    def scores(self, idx, W, b, plan, metric):
+    def scores(self, idx, fitted, plan, metric):
         torch = self.torch
         if self._fit_state is None:
             raise RuntimeError("scores() requires a preceding fit()")

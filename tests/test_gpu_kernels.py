@@ -89,7 +89,7 @@ def test_linear_fit_matches_ref_logistic(dev, problem):
     train = np.arange(0, 6000, dtype=np.int64)
 
     ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
-    W_d, b_d, ni_d, loss_d = ex.fit(train, plan)
+    (W_d, b_d), ni_d, loss_d = ex.fit(train, plan)
     W_r, b_r, ni_r, loss_r = linear_fit_ref(Xb[train], y[train], plan)
 
     assert W_d.shape == W_r.shape == (X.shape[1], B)
@@ -113,7 +113,7 @@ def test_linear_fit_l1_and_hinge(dev, problem):
     train = np.arange(100, 7000, dtype=np.int64)
 
     ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
-    W_d, b_d, _, _ = ex.fit(train, plan)
+    (W_d, b_d), _, _ = ex.fit(train, plan)
     W_r, b_r, _, _ = linear_fit_ref(Xb[train], y[train], plan)
     np.testing.assert_allclose(W_d, W_r, rtol=0.08, atol=1e-3)
     np.testing.assert_allclose(b_d, b_r, rtol=0.08, atol=1e-3)
@@ -128,7 +128,7 @@ def test_early_stop_n_iter(dev, problem):
     plan.tol[:] = [2e-3, 0.0]
     train = np.arange(4096, dtype=np.int64)
     ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
-    _, _, ni_d, _ = ex.fit(train, plan)
+    _, ni_d, _ = ex.fit(train, plan)
     assert ni_d[0] < ni_d[1]
     assert ni_d[1] == 200
 
@@ -146,9 +146,9 @@ def test_scores_match_ref(dev, problem):
     test = np.arange(6000, 8192, dtype=np.int64)
 
     ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
-    W_d, b_d, _, _ = ex.fit(train, plan)
+    (W_d, b_d), _, _ = ex.fit(train, plan)
     for metric in ("accuracy", "neg_log_loss"):
-        s_d = ex.scores(test, W_d, b_d, plan, metric)
+        s_d = ex.scores(test, (W_d, b_d), plan, metric)
         s_r = linear_scores_ref(Xb[test], y[test],
                                 W_d, b_d, plan, metric)
         np.testing.assert_allclose(s_d, s_r, rtol=5e-3, atol=5e-3)
@@ -167,8 +167,8 @@ def test_r2_scores(dev):
     train = np.arange(0, 3000, dtype=np.int64)
     test = np.arange(3000, 4096, dtype=np.int64)
     ex = DeviceLinearExecutor(X, yr, None, dev)
-    W_d, b_d, _, _ = ex.fit(train, plan)
-    s_d = ex.scores(test, W_d, b_d, plan, "r2")
+    (W_d, b_d), _, _ = ex.fit(train, plan)
+    s_d = ex.scores(test, (W_d, b_d), plan, "r2")
     s_r = linear_scores_ref(Xb[test], yr[test], W_d, b_d, plan, "r2")
     np.testing.assert_allclose(s_d, s_r, rtol=1e-3, atol=1e-3)
     assert s_d[0] > 0.9
@@ -191,10 +191,10 @@ def test_multiclass_ovr_device(dev):
     train = np.arange(0, 4500, dtype=np.int64)
     test = np.arange(4500, 6000, dtype=np.int64)
     ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
-    W_d, b_d, _, _ = ex.fit(train, plan)
+    (W_d, b_d), _, _ = ex.fit(train, plan)
     W_r, b_r, _, _ = linear_fit_ref(Xb[train], y[train], plan)
     np.testing.assert_allclose(W_d, W_r, rtol=0.08, atol=1e-3)
-    acc_d = ex.scores(test, W_d, b_d, plan, "accuracy")
+    acc_d = ex.scores(test, (W_d, b_d), plan, "accuracy")
     acc_r = linear_scores_ref(Xb[test], y[test], W_d, b_d, plan, "accuracy")
     np.testing.assert_allclose(acc_d, acc_r, atol=5e-3)
     assert acc_d[0] > 0.9
@@ -212,7 +212,7 @@ def test_full_grid_search_on_device(dev, problem):
     # CPU tier on the same machine: force host execution via device-less run
     import spark_sklearn_amd.engine as eng
 
-    classes, y_enc = eng._encode_targets(
+    classes, y_enc = eng.LinearFamily.encode_targets(
         LogisticRegression(), y)
     host = eng._HostLinearExecutor(_bf16_round(X), y_enc, classes)
     assert gs_gpu.best_score_ > 0.85
@@ -220,13 +220,13 @@ def test_full_grid_search_on_device(dev, problem):
     from sklearn.model_selection import KFold
 
     cv = list(KFold(3).split(X))
-    from spark_sklearn_amd.engine import _build_plan
+    from spark_sklearn_amd.engine import LinearFamily
 
-    plan, _ = _build_plan(LogisticRegression(max_iter=15, tol=0.0),
-                          gs_gpu.cv_results_["params"],
-                          len(cv[0][0]), host.msq, 2, "raise")
-    W, b, _, _ = host.fit(cv[0][0], plan)
-    s_host = host.scores(cv[0][1], W, b, plan, "accuracy")
+    plan, _ = LinearFamily.build_plan(LogisticRegression(max_iter=15, tol=0.0),
+                                      gs_gpu.cv_results_["params"],
+                                      len(cv[0][0]), host, 2, "raise")
+    fitted, _, _ = host.fit(cv[0][0], plan)
+    s_host = host.scores(cv[0][1], fitted, plan, "accuracy")
     np.testing.assert_allclose(
         gs_gpu.cv_results_["split0_test_score"], s_host, atol=0.02)
 
