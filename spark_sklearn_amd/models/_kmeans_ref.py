@@ -83,8 +83,21 @@ def kmeans_init_centers(X_sample: np.ndarray, k: int, init: str,
     return centers
 
 
+def bf16_round_arr(a: np.ndarray) -> np.ndarray:
+    """Round fp32 to bf16-representable values (round-to-nearest-even).
+
+    Distances are DEFINED against bf16-rounded centers in both tiers: the
+    device GEMM consumes the bf16 centroid mirror, so the reference rounds
+    identically -- keeps the two bit-consistent instead of hiding a
+    systematic 0.2%-of-|S| skew inside test tolerances."""
+    u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+    r = ((u + 0x7FFF + ((u >> 16) & 1)) >> 16) << 16
+    return r.astype(np.uint32).view(np.float32)
+
+
 def _assign(X, C_all, plan):
     """labels [n, n_blocks] int32 + per-block inertia (float64)."""
+    C_all = bf16_round_arr(C_all)
     S = X @ C_all.T                                   # [n, K_total]
     cn = np.sum(C_all.astype(np.float64) ** 2, axis=1)
     xn = np.sum(X.astype(np.float64) ** 2, axis=1)

@@ -1154,6 +1154,257 @@ void scores_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     atomicAdd(&out_sum[cand], (double)scand[cand]);
 }
 
+
+// ===========================================================================
+// K2: KMeans (Lloyd) -- centroid-block batched.  P=3 permuted LDS only
+// (the HW-probed layout; host refuses other layouts for this family).
+// ===========================================================================
+
+// per-row squared norms (assign-phase distance completion)
+__global__ void row_norms_kernel(const __bf16* __restrict__ X, long n,
+                                 int Fp, float* __restrict__ out) {
+  long row = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; row < n; row += (long)gridDim.x * blockDim.x) {
+    const __bf16* xr = X + row * Fp;
+    float s = 0.f;
+    for (int f = 0; f < Fp; f += 8) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(xr + f);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = (float)v[j];
+        s += x * x;
+      }
+    }
+    out[row] = s;
+  }
+}
+
+// assign: S = X[idx] . Ct^T; per (row, block) argmin of cn - 2S -> labels,
+// inertia partials.  32-row tiles, Kp <= 256 centroid columns, NF = Kp/64.
+template <int NF>
+__global__ __launch_bounds__(256)
+void kmeans_assign_kernel(const __bf16* __restrict__ X,
+                          const int* __restrict__ idx,      // [NTp]
+                          const __bf16* __restrict__ Ct,    // [Kp][Fp]
+                          const float* __restrict__ cnorm,  // [Kp]
+                          const float* __restrict__ xnorm,  // [n]
+                          const int* __restrict__ cstart,   // [nb+1]
+                          unsigned char* __restrict__ labels,  // [nb][NTp]
+                          float* __restrict__ inertia,      // [nb]
+                          int nt, int NTp, int Fp, int Kp, int nb,
+                          int want_labels) {
+  constexpr int cpw = NF * 16;
+  const int Kpad = NF * 64;
+  extern __shared__ char km_smem[];
+  float* zbuf = reinterpret_cast<float*>(km_smem);          // [32][Kpad+8]
+  float* sIn = zbuf + 32 * (Kpad + 8);                      // [nb<=256]
+  int* sIdx = reinterpret_cast<int*>(sIn + 256);            // [32]
+  __bf16* sX = reinterpret_cast<__bf16*>(sIdx + 32);        // [32][64]
+  __bf16* sC = sX + 32 * 64;                                // [Kpad][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+
+  if (tid < 32) sIdx[tid] = idx[rt * 32 + tid];
+  for (int i = tid; i < nb; i += 256) sIn[i] = 0.f;
+  __syncthreads();
+
+  f32x4 acc[2][NF];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int kc = 0; kc < Fp; kc += 64) {
+    {
+      int r = tid >> 3, seg = tid & 7;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          X + (long)sIdx[r] * Fp + kc + seg * 8);
+      store_row8(&sX[r * 64], r, seg * 8, v);
+    }
+    for (int e = tid; e < Kpad * 8; e += 256) {
+      int c = e >> 3, seg = e & 7;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          Ct + (long)c * Fp + kc + seg * 8);
+      store_row8(&sC[c * 64], c, seg * 8, v);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a0 = lds_frag_p(&sX[(lane & 15) * 64], lane & 15, g, h);
+      bf16x8 a1 = lds_frag_p(&sX[(16 + (lane & 15)) * 64],
+                             16 + (lane & 15), g, h);
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = wave * cpw + fn * 16 + (lane & 15);
+        bf16x8 b = lds_frag_p(&sC[c * 64], c, g, h);
+        acc[0][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a0, b, acc[0][fn], 0, 0, 0);
+        acc[1][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a1, b, acc[1][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // spill q = cn - 2S to zbuf
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < NF; ++fn) {
+      int col = wave * cpw + fn * 16 + (lane & 15);
+      float cn = cnorm[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = fm * 16 + ((lane >> 4) << 2) + r;
+        zbuf[row * (Kpad + 8) + col] = cn - 2.f * acc[fm][fn][r];
+      }
+    }
+  }
+  __syncthreads();
+
+  for (int p = tid; p < 32 * nb; p += 256) {
+    int row = p & 31;
+    int b = p >> 5;
+    int gr = rt * 32 + row;
+    if (gr >= nt) continue;
+    int lo = cstart[b], hi = cstart[b + 1];
+    const float* q = &zbuf[row * (Kpad + 8)];
+    float best = q[lo];
+    int bj = 0;
+    for (int j = lo + 1; j < hi; ++j)
+      if (q[j] < best) { best = q[j]; bj = j - lo; }
+    if (want_labels)
+      labels[(long)b * NTp + gr] = (unsigned char)bj;
+    atomicAdd(&sIn[b], xnorm[sIdx[row]] + best);
+  }
+  __syncthreads();
+  for (int b = tid; b < nb; b += 256) atomicAdd(&inertia[b], sIn[b]);
+}
+
+// update: accumulate per-centroid sums/counts in an LDS tile, flush with
+// global atomics.  grid (row_chunks, Fp/64); counts only on f-chunk 0.
+__global__ __launch_bounds__(256)
+void kmeans_update_kernel(const __bf16* __restrict__ X,
+                          const int* __restrict__ idx,
+                          const unsigned char* __restrict__ labels,
+                          const int* __restrict__ cstart,
+                          const int* __restrict__ block_active,
+                          float* __restrict__ sums,      // [Kp][Fp]
+                          float* __restrict__ counts,    // [Kp]
+                          int nt, int NTp, int Fp, int Kp, int nb,
+                          int rows_per_block) {
+  extern __shared__ char km2_smem[];
+  float* acc = reinterpret_cast<float*>(km2_smem);   // [Kp][68]
+  float* cnt = acc + Kp * 68;                        // [Kp]
+
+  const int tid = threadIdx.x;
+  const int ft = blockIdx.y;
+  const int r0 = blockIdx.x * rows_per_block;
+  const int r1 = min(r0 + rows_per_block, nt);
+
+  for (int i = tid; i < Kp * 68; i += 256) acc[i] = 0.f;
+  for (int i = tid; i < Kp; i += 256) cnt[i] = 0.f;
+  __syncthreads();
+
+  // 32 rows in flight: thread = (row 0..31) x (fseg 0..7)
+  const int rsub = tid >> 3;
+  const int fseg = tid & 7;
+  for (int rc = r0; rc < r1; rc += 32) {
+    int r = rc + rsub;
+    if (r < r1) {
+      long gi = idx[r];
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          X + gi * Fp + ft * 64 + fseg * 8);
+      for (int b = 0; b < nb; ++b) {
+        if (!block_active[b]) continue;
+        int base = cstart[b] + labels[(long)b * NTp + r];
+        float* dst = &acc[base * 68 + fseg * 8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) atomicAdd(&dst[j], (float)v[j]);
+        if (ft == 0 && fseg == 0) atomicAdd(&cnt[base], 1.f);
+      }
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < Kp * 64; e += 256) {
+    int c = e >> 6, f = e & 63;
+    float v = acc[c * 68 + f];
+    if (v != 0.f) atomicAdd(&sums[(long)c * Fp + ft * 64 + f], v);
+  }
+  if (ft == 0)
+    for (int c = tid; c < Kp; c += 256)
+      if (cnt[c] != 0.f) atomicAdd(&counts[c], cnt[c]);
+}
+
+// center update: new = count>0 ? sum/count : old; accumulate shift^2 per
+// block and refresh the bf16 mirror + cnorm; zero sums.
+__global__ void kmeans_center_kernel(float* __restrict__ C,      // [Kp][Fp]
+                                     __bf16* __restrict__ Ct,    // [Kp][Fp]
+                                     float* __restrict__ cnorm,  // [Kp]
+                                     float* __restrict__ sums,
+                                     float* __restrict__ counts,
+                                     const int* __restrict__ cen_block,
+                                     const int* __restrict__ block_active,
+                                     float* __restrict__ shift2,  // [nb]
+                                     int Fp, int Kp) {
+  int c = blockIdx.x;
+  int b = cen_block[c];
+  float cntv = counts[c];
+  __shared__ float sh[256];
+  __shared__ float nrm[256];
+  float local_sh = 0.f, local_nrm = 0.f;
+  bool upd = block_active[b] && cntv > 0.f;
+  for (int f = threadIdx.x; f < Fp; f += blockDim.x) {
+    long off = (long)c * Fp + f;
+    float oldv = C[off];
+    float newv = upd ? sums[off] / cntv : oldv;
+    sums[off] = 0.f;
+    float d = newv - oldv;
+    local_sh += d * d;
+    __bf16 nb16 = (__bf16)newv;
+    float rv = (float)nb16;       // cnorm must match the bf16 GEMM operand
+    local_nrm += rv * rv;
+    C[off] = newv;
+    Ct[off] = nb16;
+  }
+  sh[threadIdx.x] = local_sh;
+  nrm[threadIdx.x] = local_nrm;
+  __syncthreads();
+  for (int s2 = blockDim.x / 2; s2 > 0; s2 >>= 1) {
+    if (threadIdx.x < s2) {
+      sh[threadIdx.x] += sh[threadIdx.x + s2];
+      nrm[threadIdx.x] += nrm[threadIdx.x + s2];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    cnorm[c] = nrm[0];
+    if (block_active[b]) atomicAdd(&shift2[b], sh[0]);
+    counts[c] = 0.f;
+  }
+}
+
+// per-block convergence bookkeeping
+__global__ void kmeans_block_finalize_kernel(
+    float* __restrict__ shift2, int* __restrict__ active,
+    int* __restrict__ n_iter, const int* __restrict__ max_iter,
+    const float* __restrict__ tol, int epoch, int nb,
+    int* __restrict__ n_active) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= nb) return;
+  int ea = active[b] && (epoch < max_iter[b]);
+  if (ea) {
+    n_iter[b] = epoch + 1;
+    if (shift2[b] <= tol[b]) active[b] = 0;
+  }
+  shift2[b] = 0.f;
+  if (active[b] && (epoch + 1) < max_iter[b]) atomicAdd(n_active, 1);
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -1442,6 +1693,182 @@ torch::Tensor linear_scores(torch::Tensor X, torch::Tensor y,
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// KMeans host wrappers
+// ---------------------------------------------------------------------------
+
+static void launch_kmeans_assign(const __bf16* X, const int* idx,
+                                 const __bf16* Ct, const float* cnorm,
+                                 const float* xnorm, const int* cstart,
+                                 unsigned char* labels, float* inertia,
+                                 int nt, int NTp, int Fp, int Kp, int nb,
+                                 int want_labels, hipStream_t s) {
+  size_t smem = (size_t)32 * (Kp + 8) * 4 + 256 * 4 + 32 * 4 +
+                (size_t)32 * 64 * 2 + (size_t)Kp * 64 * 2;
+  dim3 grid((unsigned)(NTp / 32));
+  auto go = [&](auto k) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(k),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem));
+    hipLaunchKernelGGL(k, grid, dim3(256), smem, s, X, idx, Ct, cnorm,
+                       xnorm, cstart, labels, inertia, nt, NTp, Fp, Kp, nb,
+                       want_labels);
+  };
+  switch (Kp / 64) {
+    case 1: go(kmeans_assign_kernel<1>); break;
+    case 2: go(kmeans_assign_kernel<2>); break;
+    case 3: go(kmeans_assign_kernel<3>); break;
+    default: go(kmeans_assign_kernel<4>); break;
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// Full Lloyd loop on device.  Returns (C [Kp][Fp] f32, n_iter [nb],
+// inertia_final [nb] f32, Ct bf16, cnorm f32) -- Ct/cnorm stay resident for
+// the scores pass.
+std::vector<torch::Tensor> kmeans_fit(
+    torch::Tensor X, torch::Tensor idx, torch::Tensor C0,
+    torch::Tensor cstart, torch::Tensor cen_block, torch::Tensor max_iter,
+    torch::Tensor tol, torch::Tensor xnorm, int64_t nt, int64_t max_epochs,
+    int64_t check_every) {
+  TORCH_CHECK(g_layout == 3,
+              "kmeans kernels require the probed layout P=3 (MI355X)");
+  const at::cuda::CUDAGuard guard(X.device());
+  int Fp = (int)X.size(1);
+  int Kp = (int)C0.size(0);
+  int nb = (int)max_iter.size(0);
+  long NTp = idx.size(0);
+  TORCH_CHECK(Kp % 64 == 0 && Kp <= 256);
+  TORCH_CHECK(NTp % 32 == 0);
+  TORCH_CHECK(C0.dtype() == torch::kFloat32 && C0.size(1) == Fp);
+
+  auto optF = X.options().dtype(torch::kFloat32);
+  auto optI = X.options().dtype(torch::kInt32);
+  auto C = C0.clone();
+  auto Ct = torch::empty({Kp, Fp}, X.options());
+  auto cnorm = torch::empty({Kp}, optF);
+  auto sums = torch::zeros({Kp, Fp}, optF);
+  auto counts = torch::zeros({Kp}, optF);
+  auto labels = torch::zeros({(long)nb, NTp},
+                             X.options().dtype(torch::kUInt8));
+  auto shift2 = torch::zeros({nb + 1}, optF);
+  auto active = torch::ones({nb + 1}, optI);
+  active[nb] = 0;
+  auto n_iter = torch::zeros({nb}, optI);
+  auto inertia = torch::zeros({nb}, optF);
+  auto n_active = torch::zeros({1}, optI);
+
+  hipStream_t s = cur_stream();
+  const __bf16* Xp = reinterpret_cast<const __bf16*>(X.data_ptr());
+
+  // init pass: counts==0 everywhere -> keeps C, fills Ct + cnorm
+  hipLaunchKernelGGL(kmeans_center_kernel, dim3(Kp), dim3(256), 0, s,
+                     C.data_ptr<float>(),
+                     reinterpret_cast<__bf16*>(Ct.data_ptr()),
+                     cnorm.data_ptr<float>(), sums.data_ptr<float>(),
+                     counts.data_ptr<float>(), cen_block.data_ptr<int>(),
+                     active.data_ptr<int>(), shift2.data_ptr<float>(),
+                     Fp, Kp);
+
+  int ftiles = Fp / 64;
+  int gridx = std::max(1, 1024 / ftiles);
+  int rows_per_block = (int)(((nt + gridx - 1) / gridx + 31) / 32 * 32);
+  gridx = (int)((nt + rows_per_block - 1) / rows_per_block);
+  size_t usmem = (size_t)Kp * 68 * 4 + (size_t)Kp * 4;
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(kmeans_update_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, (int)usmem));
+
+  for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
+    HIP_CHECK(hipMemsetAsync(inertia.data_ptr(), 0, nb * sizeof(float), s));
+    launch_kmeans_assign(Xp, idx.data_ptr<int>(),
+                         reinterpret_cast<const __bf16*>(Ct.data_ptr()),
+                         cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
+                         cstart.data_ptr<int>(),
+                         labels.data_ptr<unsigned char>(),
+                         inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp,
+                         Kp, nb, 1, s);
+    hipLaunchKernelGGL(kmeans_update_kernel, dim3(gridx, ftiles), dim3(256),
+                       usmem, s, Xp, idx.data_ptr<int>(),
+                       labels.data_ptr<unsigned char>(),
+                       cstart.data_ptr<int>(), active.data_ptr<int>(),
+                       sums.data_ptr<float>(), counts.data_ptr<float>(),
+                       (int)nt, (int)NTp, Fp, Kp, nb, rows_per_block);
+    hipLaunchKernelGGL(kmeans_center_kernel, dim3(Kp), dim3(256), 0, s,
+                       C.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(Ct.data_ptr()),
+                       cnorm.data_ptr<float>(), sums.data_ptr<float>(),
+                       counts.data_ptr<float>(), cen_block.data_ptr<int>(),
+                       active.data_ptr<int>(), shift2.data_ptr<float>(),
+                       Fp, Kp);
+    hipLaunchKernelGGL(kmeans_block_finalize_kernel,
+                       dim3((nb + 255) / 256), dim3(256), 0, s,
+                       shift2.data_ptr<float>(), active.data_ptr<int>(),
+                       n_iter.data_ptr<int>(), max_iter.data_ptr<int>(),
+                       tol.data_ptr<float>(), epoch, nb,
+                       n_active.data_ptr<int>());
+    HIP_CHECK(hipGetLastError());
+    if (check_every > 0 && (epoch + 1) % check_every == 0 &&
+        epoch + 1 < max_epochs) {
+      int na = n_active.item<int>();
+      if (na == 0) break;
+      n_active.zero_();
+    } else {
+      HIP_CHECK(hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s));
+    }
+  }
+  // final inertia under the final centers
+  HIP_CHECK(hipMemsetAsync(inertia.data_ptr(), 0, nb * sizeof(float), s));
+  launch_kmeans_assign(Xp, idx.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(Ct.data_ptr()),
+                       cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
+                       cstart.data_ptr<int>(),
+                       labels.data_ptr<unsigned char>(),
+                       inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp,
+                       Kp, nb, 1, s);
+  return {C, n_iter, inertia, Ct, cnorm, labels};
+}
+
+// assignment / inertia of arbitrary rows under given centers (scores +
+// labels_ paths)
+std::vector<torch::Tensor> kmeans_assign(
+    torch::Tensor X, torch::Tensor idx, torch::Tensor Ct,
+    torch::Tensor cnorm, torch::Tensor xnorm, torch::Tensor cstart,
+    int64_t nt, int64_t want_labels) {
+  TORCH_CHECK(g_layout == 3, "kmeans kernels require layout P=3");
+  const at::cuda::CUDAGuard guard(X.device());
+  int Fp = (int)X.size(1);
+  int Kp = (int)Ct.size(0);
+  int nb = (int)cstart.size(0) - 1;
+  long NTp = idx.size(0);
+  auto labels = torch::zeros({(long)nb, NTp},
+                             X.options().dtype(torch::kUInt8));
+  auto inertia = torch::zeros({nb}, X.options().dtype(torch::kFloat32));
+  launch_kmeans_assign(
+      reinterpret_cast<const __bf16*>(X.data_ptr()), idx.data_ptr<int>(),
+      reinterpret_cast<const __bf16*>(Ct.data_ptr()),
+      cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
+      cstart.data_ptr<int>(), labels.data_ptr<unsigned char>(),
+      inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp, Kp, nb,
+      (int)want_labels, cur_stream());
+  return {inertia, labels};
+}
+
+torch::Tensor row_norms(torch::Tensor X) {
+  const at::cuda::CUDAGuard guard(X.device());
+  long n = X.size(0);
+  int Fp = (int)X.size(1);
+  auto out = torch::empty({n}, X.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<long>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(row_norms_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(),
+                     reinterpret_cast<const __bf16*>(X.data_ptr()), n, Fp,
+                     out.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
@@ -1449,4 +1876,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("msq_rows", &msq_rows, "mean squared row norm");
   m.def("linear_fit", &linear_fit, "batched linear fit (device epoch loop)");
   m.def("linear_scores", &linear_scores, "batched per-candidate scores");
+  m.def("kmeans_fit", &kmeans_fit, "batched Lloyd loop (centroid blocks)");
+  m.def("kmeans_assign", &kmeans_assign, "assignment/inertia under centers");
+  m.def("row_norms", &row_norms, "per-row squared norms");
 }

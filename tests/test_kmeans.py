@@ -32,7 +32,9 @@ def test_matches_sklearn_inertia(blobs):
     from sklearn.metrics import adjusted_rand_score
 
     assert adjusted_rand_score(ours.labels_, sk.labels_) > 0.99
-    assert abs(ours.score(X) + ours.inertia_) < 1e-6 * ours.inertia_
+    # score() uses exact fp64 distances; inertia_ comes from the batched
+    # solver whose distances are defined against bf16-rounded centers
+    assert abs(ours.score(X) + ours.inertia_) < 1e-4 * ours.inertia_
 
 
 def test_predict_transform(blobs):
