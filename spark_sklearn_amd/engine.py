@@ -168,14 +168,22 @@ class LinearFamily:
 
 
 class _HostLinearExecutor:
-    """Numpy execution of a linear candidate batch (reference semantics)."""
+    """Numpy/scipy execution of a linear candidate batch (reference
+    semantics); accepts dense arrays or scipy CSR matrices."""
 
     def __init__(self, X, y_enc, classes):
-        self.X = np.ascontiguousarray(np.asarray(X), dtype=np.float32)
+        import scipy.sparse as sp
+
+        if sp.issparse(X):
+            self.X = sp.csr_matrix(X, dtype=np.float32)
+            sq = self.X.multiply(self.X).sum(axis=1)
+            self.msq = float(np.mean(np.asarray(sq)))
+        else:
+            self.X = np.ascontiguousarray(np.asarray(X), dtype=np.float32)
+            self.msq = float(np.mean(np.sum(self.X.astype(np.float64) ** 2,
+                                            axis=1)))
         self.y = y_enc
         self.classes = classes
-        self.msq = float(np.mean(np.sum(self.X.astype(np.float64) ** 2,
-                                        axis=1)))
 
     def fit(self, train_idx, plan):
         from .models._solver_ref import linear_fit_ref

@@ -95,8 +95,13 @@ def _column_targets(y: np.ndarray, plan: BatchPlan) -> np.ndarray:
 def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
                    sample_stride: int = 1):
     """Fit B columns simultaneously.  Returns (W [F,B], b [B], n_iter [B],
-    final_loss [B])."""
-    X = np.ascontiguousarray(X, dtype=np.float32)
+    final_loss [B]).  X may be dense or scipy CSR."""
+    import scipy.sparse as sp
+
+    if sp.issparse(X):
+        X = sp.csr_matrix(X, dtype=np.float32)
+    else:
+        X = np.ascontiguousarray(X, dtype=np.float32)
     n, F = X.shape
     B = plan.B
     T = _column_targets(np.asarray(y), plan)
@@ -113,7 +118,7 @@ def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
     max_epochs = int(plan.max_iter.max())
     inv_n = np.float32(1.0 / n)
     for epoch in range(max_epochs):
-        Z = X @ W + b[np.newaxis, :]                    # [n, B]
+        Z = np.asarray(X @ W) + b[np.newaxis, :]        # [n, B]
         D = np.empty_like(Z)
         loss_sum = np.zeros(B, dtype=np.float64)
         for lid, cols in _loss_groups(plan):
@@ -142,7 +147,8 @@ def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
         cur_loss = np.where(epoch_active, cur, cur_loss)
         n_iter = np.where(epoch_active, epoch + 1, n_iter).astype(np.int32)
 
-        G = X.T @ D * inv_n + plan.l2 * W + plan.l1 * np.sign(W)   # [F, B]
+        G = np.asarray(X.T @ D) * inv_n + plan.l2 * W \
+            + plan.l1 * np.sign(W)                           # [F, B]
         gb = D.mean(axis=0) * plan.fit_intercept
 
         mask = epoch_active.astype(np.float32)
@@ -176,8 +182,11 @@ def linear_scores_ref(X: np.ndarray, y: np.ndarray, W: np.ndarray,
 
     metric: 'accuracy' (OVR argmax / sign), 'neg_log_loss', or 'r2'.
     """
-    X = np.ascontiguousarray(X, dtype=np.float32)
-    Z = X @ W + b[np.newaxis, :]
+    import scipy.sparse as sp
+
+    if not sp.issparse(X):
+        X = np.ascontiguousarray(X, dtype=np.float32)
+    Z = np.asarray(X @ W) + b[np.newaxis, :]
     n_cand = int(plan.col_to_cand.max()) + 1
     C = plan.n_classes
     scores = np.zeros(n_cand, dtype=np.float64)

@@ -85,9 +85,12 @@ class _LinearBase(BaseEstimator):
 
     def decision_function(self, X):
         self._check_fitted()
-        X = np.asarray(X, dtype=np.float64)
+        import scipy.sparse as sp
+
+        if not sp.issparse(X):
+            X = np.asarray(X, dtype=np.float64)
         coef = np.atleast_2d(self.coef_)
-        scores = X @ coef.T + np.atleast_1d(self.intercept_)
+        scores = np.asarray(X @ coef.T) + np.atleast_1d(self.intercept_)
         return scores.ravel() if scores.shape[1] == 1 else scores
 
     def _check_fitted(self):

@@ -61,22 +61,42 @@ def ensure_layout(device) -> int:
 
 
 def _as_device_bf16(X, device):
-    """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array."""
+    """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array.
+
+    scipy CSR matrices are ingested chunk-wise into a dense bf16 resident
+    matrix: 288 GB of HBM3E makes even a 10M x 1024 matrix (20 GB) cheap
+    to hold dense, and the MFMA fit kernels then run at dense speed.
+    (A dedicated LDS-staged CSR kernel path -- SURVEY.md K5 -- is the
+    planned upgrade for matrices that exceed dense residency.)"""
+    import scipy.sparse as sp
     import torch
 
     if isinstance(X, torch.Tensor):
         key = (X.data_ptr(), X.shape, str(device))
     else:
-        key = (id(X), np.asarray(X).shape, str(device))
+        key = (id(X), X.shape, str(device))
     hit = _data_cache.get(key)
     if hit is not None:
         return hit
-    t = X if isinstance(X, torch.Tensor) else torch.as_tensor(
-        np.ascontiguousarray(np.asarray(X, dtype=np.float32)))
-    n, F = t.shape
-    Fp = max(64, _roundup(F, 64))
-    dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
-    dev_t[:, :F] = t.to(device).to(torch.bfloat16)
+    if sp.issparse(X):
+        Xc = sp.csr_matrix(X)
+        n, F = Xc.shape
+        Fp = max(64, _roundup(F, 64))
+        dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
+        step = max(1, (1 << 28) // max(F, 1))   # ~256M elems per chunk
+        for lo in range(0, n, step):
+            hi = min(lo + step, n)
+            dense = torch.as_tensor(
+                Xc[lo:hi].toarray().astype(np.float32))
+            dev_t[lo:hi, :F] = dense.to(device).to(torch.bfloat16)
+        t = None
+    else:
+        t = X if isinstance(X, torch.Tensor) else torch.as_tensor(
+            np.ascontiguousarray(np.asarray(X, dtype=np.float32)))
+        n, F = t.shape
+        Fp = max(64, _roundup(F, 64))
+        dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
+        dev_t[:, :F] = t.to(device).to(torch.bfloat16)
     # resident transpose: the bwd kernel's B-operand stages coalesced from
     # X^T rows (288 GB HBM makes the copy cheap; one-time data prep)
     dev_tt = dev_t.t().contiguous()
