@@ -331,7 +331,9 @@ def fit_single(estimator, X, y, device=None):
     return estimator
 
 
+from .models.gbt import GBTFamily  # noqa: E402
 from .models.kmeans import KMeansFamily  # noqa: E402  (no import cycle:
-# models.kmeans only imports engine lazily inside methods)
+# model modules only import engine lazily inside methods)
 
-FAMILIES = {"linear": LinearFamily, "kmeans": KMeansFamily}
+FAMILIES = {"linear": LinearFamily, "kmeans": KMeansFamily,
+            "gbt": GBTFamily}
