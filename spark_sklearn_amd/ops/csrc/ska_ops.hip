@@ -1405,6 +1405,262 @@ __global__ void kmeans_block_finalize_kernel(
   if (active[b] && (epoch + 1) < max_iter[b]) atomicAdd(n_active, 1);
 }
 
+
+// ===========================================================================
+// K3: histogram gradient boosting (depth-wise heap trees)
+// ===========================================================================
+
+__global__ void gbt_grad_kernel(const float* __restrict__ raw,
+                                const float* __restrict__ y,
+                                float* __restrict__ g, float* __restrict__ h,
+                                long nt, int loss) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    if (loss == LOSS_LOGISTIC) {
+      float p = 1.f / (1.f + __expf(-raw[r]));
+      g[r] = p - y[r];
+      h[r] = p * (1.f - p);
+    } else {
+      g[r] = raw[r] - y[r];
+      h[r] = 1.f;
+    }
+  }
+}
+
+// per-(node, feature, bin) g/h/count histograms in LDS, flushed with
+// global atomics.  Rows keep a heap node_id; settled rows (node at an
+// earlier level) are skipped by the level range check.
+__global__ __launch_bounds__(256)
+void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
+                     const int* __restrict__ idxmap,          // [nt]
+                     const int* __restrict__ node_id,         // [nt]
+                     const float* __restrict__ g,
+                     const float* __restrict__ h,
+                     float* __restrict__ hist,  // [lvl_nodes][F][256][3]
+                     long nt, int F, int lvl_lo, int lvl_nodes, int f_pb,
+                     long rows_pb) {
+  extern __shared__ float lh[];     // [lvl_nodes][f_pb][256][3]
+  const int tid = threadIdx.x;
+  const int f0 = blockIdx.y * f_pb;
+  const long r0 = (long)blockIdx.x * rows_pb;
+  const long r1 = min(r0 + rows_pb, nt);
+  const int cells = lvl_nodes * f_pb * 256 * 3;
+
+  for (int i = tid; i < cells; i += 256) lh[i] = 0.f;
+  __syncthreads();
+
+  for (long r = r0 + tid; r < r1; r += 256) {
+    int nd = node_id[r];
+    if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
+    long gi = idxmap[r];
+    float gr = g[r], hr = h[r];
+    const unsigned char* xrow = Xb + gi * F + f0;
+    for (int ff = 0; ff < f_pb; ++ff) {
+      int bin = xrow[ff];
+      float* cell = &lh[(((nd - lvl_lo) * f_pb + ff) * 256 + bin) * 3];
+      atomicAdd(cell, gr);
+      atomicAdd(cell + 1, hr);
+      atomicAdd(cell + 2, 1.f);
+    }
+  }
+  __syncthreads();
+  for (int i = tid; i < cells; i += 256) {
+    float v = lh[i];
+    if (v == 0.f) continue;
+    int c = i / 3, ch = i % 3;
+    int bin = c % 256;
+    int ff = (c / 256) % f_pb;
+    int ndl = c / (256 * f_pb);
+    atomicAdd(&hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3 + ch], v);
+  }
+}
+
+// best split per node: 256 threads scan features; LDS argmax reduce.
+__global__ __launch_bounds__(256)
+void gbt_split_find_kernel(const float* __restrict__ hist, int F,
+                           int lvl_lo, float lam, float lr, int min_leaf,
+                           int force_leaf, int* __restrict__ feat,
+                           int* __restrict__ thr, float* __restrict__ val) {
+  const int nd_local = blockIdx.x;
+  const int nd = lvl_lo + nd_local;
+  const int tid = threadIdx.x;
+  const float* nh = hist + (long)nd_local * F * 256 * 3;
+
+  __shared__ float sGain[256];
+  __shared__ int sF[256], sT[256];
+  __shared__ float sG, sH;
+  float bgain = 0.f;
+  int bf = -1, bt = 0;
+  float Gtot = 0.f, Htot = 0.f;
+
+  for (int f = tid; f < F; f += 256) {
+    const float* fh = nh + (long)f * 256 * 3;
+    float G = 0.f, H = 0.f, C = 0.f;
+    for (int b2 = 0; b2 < 256; ++b2) {
+      G += fh[b2 * 3];
+      H += fh[b2 * 3 + 1];
+      C += fh[b2 * 3 + 2];
+    }
+    if (f == tid) { Gtot = G; Htot = H; }
+    if (!force_leaf) {
+      float ps = G * G / (H + lam + 1e-12f);
+      float gl = 0.f, hl = 0.f, cl = 0.f;
+      for (int b2 = 0; b2 < 255; ++b2) {
+        gl += fh[b2 * 3];
+        hl += fh[b2 * 3 + 1];
+        cl += fh[b2 * 3 + 2];
+        float cr = C - cl;
+        if (cl < (float)min_leaf || cr < (float)min_leaf) continue;
+        float gr2 = G - gl, hr2 = H - hl;
+        float gain = gl * gl / (hl + lam + 1e-12f) +
+                     gr2 * gr2 / (hr2 + lam + 1e-12f) - ps;
+        if (gain > bgain + 1e-12f) { bgain = gain; bf = f; bt = b2; }
+      }
+    }
+  }
+  sGain[tid] = bgain;
+  sF[tid] = bf;
+  sT[tid] = bt;
+  if (tid == 0) { sG = Gtot; sH = Htot; }
+  __syncthreads();
+  for (int s2 = 128; s2 > 0; s2 >>= 1) {
+    if (tid < s2) {
+      // deterministic tie-break: keep the LOWER feature index
+      if (sGain[tid + s2] > sGain[tid] + 1e-12f ||
+          (sGain[tid + s2] > sGain[tid] - 1e-12f &&
+           sF[tid + s2] >= 0 &&
+           (sF[tid] < 0 || sF[tid + s2] < sF[tid]))) {
+        sGain[tid] = sGain[tid + s2];
+        sF[tid] = sF[tid + s2];
+        sT[tid] = sT[tid + s2];
+      }
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    if (sF[0] < 0 || sGain[0] <= 1e-12f) {
+      feat[nd] = -1;
+      if (sH > 0.f || sG != 0.f)
+        val[nd] = -sG / (sH + lam + 1e-12f) * lr;
+    } else {
+      feat[nd] = sF[0];
+      thr[nd] = sT[0];
+      val[nd] = 0.f;
+    }
+  }
+}
+
+__global__ void gbt_split_apply_kernel(const unsigned char* __restrict__ Xb,
+                                       const int* __restrict__ idxmap,
+                                       int* __restrict__ node_id,
+                                       const int* __restrict__ feat,
+                                       const int* __restrict__ thr,
+                                       long nt, int F, int lvl_lo,
+                                       int lvl_hi) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    int nd = node_id[r];
+    if (nd < lvl_lo || nd >= lvl_hi) continue;
+    int f = feat[nd];
+    if (f < 0) continue;
+    int bin = Xb[(long)idxmap[r] * F + f];
+    node_id[r] = 2 * nd + 1 + (bin > thr[nd] ? 1 : 0);
+  }
+}
+
+// bottom-level leaf values: per-node g/h sums via atomics
+__global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
+                                     const float* __restrict__ g,
+                                     const float* __restrict__ h,
+                                     float* __restrict__ gsum,
+                                     float* __restrict__ hsum,
+                                     long nt, int lvl_lo, int lvl_hi) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    int nd = node_id[r];
+    if (nd < lvl_lo || nd >= lvl_hi) continue;
+    atomicAdd(&gsum[nd], g[r]);
+    atomicAdd(&hsum[nd], h[r]);
+  }
+}
+
+__global__ void gbt_leaf_vals_kernel(const float* __restrict__ gsum,
+                                     const float* __restrict__ hsum,
+                                     float* __restrict__ val,
+                                     int* __restrict__ feat, float lam,
+                                     float lr, int lvl_lo, int lvl_hi) {
+  int nd = lvl_lo + blockIdx.x * blockDim.x + threadIdx.x;
+  if (nd >= lvl_hi) return;
+  if (hsum[nd] > 0.f) {
+    feat[nd] = -1;
+    val[nd] = -gsum[nd] / (hsum[nd] + lam + 1e-12f) * lr;
+  }
+}
+
+__global__ void gbt_leaf_apply_kernel(float* __restrict__ raw,
+                                      int* __restrict__ node_id,
+                                      const float* __restrict__ val,
+                                      long nt) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    raw[r] += val[node_id[r]];
+    node_id[r] = 0;
+  }
+}
+
+// predict: walk all trees for arbitrary rows
+__global__ void gbt_predict_kernel(const unsigned char* __restrict__ Xb,
+                                   const int* __restrict__ idx,  // [n_idx]
+                                   const int* __restrict__ feats,  // [T][NN]
+                                   const int* __restrict__ thrs,
+                                   const float* __restrict__ vals,
+                                   int T, int NN, int F, long n_idx,
+                                   float f0, float* __restrict__ out) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < n_idx; r += (long)gridDim.x * blockDim.x) {
+    const unsigned char* xrow = Xb + (long)idx[r] * F;
+    float acc = f0;
+    for (int t = 0; t < T; ++t) {
+      const int* tf = feats + (long)t * NN;
+      const int* tt = thrs + (long)t * NN;
+      int nd = 0;
+      int guard = 0;
+      while (tf[nd] >= 0 && guard++ < 16) {
+        nd = 2 * nd + 1 + (xrow[tf[nd]] > tt[nd] ? 1 : 0);
+        if (nd >= NN) { nd = (nd - 1) / 2; break; }
+      }
+      acc += vals[(long)t * NN + nd];
+    }
+    out[r] = acc;
+  }
+}
+
+// metric sums: [0] = correct count (accuracy), [1] = squared residual
+__global__ void gbt_metric_kernel(const float* __restrict__ raw,
+                                  const float* __restrict__ y,
+                                  const int* __restrict__ idx, long n_idx,
+                                  int metric, double* __restrict__ out) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  double local = 0.0;
+  for (; r < n_idx; r += (long)gridDim.x * blockDim.x) {
+    float yv = y[idx[r]];
+    if (metric == 0) {
+      local += ((raw[r] > 0.f) == (yv > 0.5f)) ? 1.0 : 0.0;
+    } else {
+      float d = yv - raw[r];
+      local += (double)d * d;
+    }
+  }
+  __shared__ double red[256];
+  red[threadIdx.x] = local;
+  __syncthreads();
+  for (int s2 = blockDim.x / 2; s2 > 0; s2 >>= 1) {
+    if (threadIdx.x < s2) red[threadIdx.x] += red[threadIdx.x + s2];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(out, red[0]);
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -1869,6 +2125,139 @@ torch::Tensor row_norms(torch::Tensor X) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// GBT host wrappers
+// ---------------------------------------------------------------------------
+
+// One full boosting fit on device.  Returns (feats [T][NN] i32,
+// thrs [T][NN] i32, vals [T][NN] f32).
+std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
+                                   torch::Tensor y_tr, int64_t loss,
+                                   int64_t n_estimators, double lrate,
+                                   int64_t max_depth, int64_t min_leaf,
+                                   double lam, double f0) {
+  const at::cuda::CUDAGuard guard(Xb.device());
+  TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 6,
+              "device GBT supports max_depth in [1, 6]");
+  int F = (int)Xb.size(1);
+  long nt = idxmap.size(0);
+  int T = (int)n_estimators;
+  int NN = (1 << (max_depth + 1)) - 1;
+
+  auto optF = y_tr.options().dtype(torch::kFloat32);
+  auto optI = y_tr.options().dtype(torch::kInt32);
+  auto raw = torch::full({nt}, f0, optF);
+  auto g = torch::empty({nt}, optF);
+  auto h = torch::empty({nt}, optF);
+  auto node_id = torch::zeros({nt}, optI);
+  auto feats = torch::full({(long)T, NN}, -1,
+                           optI);
+  auto thrs = torch::zeros({(long)T, NN}, optI);
+  auto vals = torch::zeros({(long)T, NN}, optF);
+  int max_lvl_nodes = 1 << (max_depth - 1);
+  auto hist = torch::empty({(long)max_lvl_nodes * F * 256 * 3}, optF);
+  auto gsum = torch::zeros({NN}, optF);
+  auto hsum = torch::zeros({NN}, optF);
+
+  hipStream_t s = cur_stream();
+  const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+
+  for (int t = 0; t < T; ++t) {
+    int* feat_t = feats.data_ptr<int>() + (long)t * NN;
+    int* thr_t = thrs.data_ptr<int>() + (long)t * NN;
+    float* val_t = vals.data_ptr<float>() + (long)t * NN;
+    hipLaunchKernelGGL(gbt_grad_kernel, dim3(blocks), dim3(256), 0, s,
+                       raw.data_ptr<float>(), y_tr.data_ptr<float>(),
+                       g.data_ptr<float>(), h.data_ptr<float>(), nt,
+                       (int)loss);
+    for (int d = 0; d < (int)max_depth; ++d) {
+      int lvl_lo = (1 << d) - 1;
+      int lvl_nodes = 1 << d;
+      int f_pb = std::min(8, std::max(1, (96 * 1024) / (3072 * lvl_nodes)));
+      while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
+      int fchunks = F / f_pb;
+      long rowsegs = std::max<long>(1, 2048 / fchunks);
+      long rows_pb = (nt + rowsegs - 1) / rowsegs;
+      rowsegs = (nt + rows_pb - 1) / rows_pb;
+      size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 3 * 4;
+      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0,
+                               (size_t)lvl_nodes * F * 256 * 3 * 4, s));
+      HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(gbt_hist_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
+      hipLaunchKernelGGL(gbt_hist_kernel,
+                         dim3((unsigned)rowsegs, (unsigned)fchunks),
+                         dim3(256), lsm, s, Xp, idxmap.data_ptr<int>(),
+                         node_id.data_ptr<int>(), g.data_ptr<float>(),
+                         h.data_ptr<float>(), hist.data_ptr<float>(), nt,
+                         F, lvl_lo, lvl_nodes, f_pb, rows_pb);
+      hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
+                         0, s, hist.data_ptr<float>(), F, lvl_lo,
+                         (float)lam, (float)lrate, (int)min_leaf, 0,
+                         feat_t, thr_t, val_t);
+      hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
+                         0, s, Xp, idxmap.data_ptr<int>(),
+                         node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
+                         lvl_lo, lvl_lo + lvl_nodes);
+      HIP_CHECK(hipGetLastError());
+    }
+    // bottom-level leaves
+    int bl_lo = (1 << max_depth) - 1;
+    HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
+    HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
+    hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0, s,
+                       node_id.data_ptr<int>(), g.data_ptr<float>(),
+                       h.data_ptr<float>(), gsum.data_ptr<float>(),
+                       hsum.data_ptr<float>(), nt, bl_lo, NN);
+    hipLaunchKernelGGL(gbt_leaf_vals_kernel,
+                       dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
+                       gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                       val_t, feat_t, (float)lam, (float)lrate, bl_lo, NN);
+    hipLaunchKernelGGL(gbt_leaf_apply_kernel, dim3(blocks), dim3(256), 0, s,
+                       raw.data_ptr<float>(), node_id.data_ptr<int>(),
+                       val_t, nt);
+    HIP_CHECK(hipGetLastError());
+  }
+  return {feats, thrs, vals};
+}
+
+torch::Tensor gbt_predict(torch::Tensor Xb, torch::Tensor idx,
+                          torch::Tensor feats, torch::Tensor thrs,
+                          torch::Tensor vals, double f0) {
+  const at::cuda::CUDAGuard guard(Xb.device());
+  int T = (int)feats.size(0);
+  int NN = (int)feats.size(1);
+  int F = (int)Xb.size(1);
+  long n_idx = idx.size(0);
+  auto out = torch::empty({n_idx},
+                          vals.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<long>((n_idx + 255) / 256, 2048);
+  hipLaunchKernelGGL(gbt_predict_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), Xb.data_ptr<unsigned char>(),
+                     idx.data_ptr<int>(), feats.data_ptr<int>(),
+                     thrs.data_ptr<int>(), vals.data_ptr<float>(), T, NN,
+                     F, n_idx, (float)f0, out.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+torch::Tensor gbt_metric(torch::Tensor raw, torch::Tensor y,
+                         torch::Tensor idx, int64_t metric) {
+  const at::cuda::CUDAGuard guard(raw.device());
+  long n_idx = idx.size(0);
+  auto out = torch::zeros({1}, raw.options().dtype(torch::kFloat64));
+  int blocks = (int)std::min<long>((n_idx + 255) / 256, 2048);
+  hipLaunchKernelGGL(gbt_metric_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), raw.data_ptr<float>(),
+                     y.data_ptr<float>(), idx.data_ptr<int>(), n_idx,
+                     (int)metric, out.data_ptr<double>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
@@ -1879,4 +2268,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_fit", &kmeans_fit, "batched Lloyd loop (centroid blocks)");
   m.def("kmeans_assign", &kmeans_assign, "assignment/inertia under centers");
   m.def("row_norms", &row_norms, "per-row squared norms");
+  m.def("gbt_fit", &gbt_fit, "histogram boosting fit (depth-wise trees)");
+  m.def("gbt_predict", &gbt_predict, "tree-walk raw predictions");
+  m.def("gbt_metric", &gbt_metric, "accuracy/ss_res sums");
 }
