@@ -1661,6 +1661,285 @@ __global__ void gbt_metric_kernel(const float* __restrict__ raw,
   if (threadIdx.x == 0) atomicAdd(out, red[0]);
 }
 
+
+// ---------------------------------------------------------------------------
+// fwd_d7_kernel: 1024-thread fwd for Bp=256 -- block tile 256 rows x 256
+// cols, 16 waves (4 row-quads x 4 col-waves, each 64x64 like fwd_d5's
+// waves).  16 waves/CU at 1 block/CU doubles occupancy over fwd_d5 and
+// halves the per-row staging overhead.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                   const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                   const float* __restrict__ bias,
+                   const int* __restrict__ loss_id,
+                   const int* __restrict__ target, __bf16* __restrict__ Dt,
+                   float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                   int nt, int NTp, int Fp, int Bp, int want_loss) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[256];
+  __shared__ float sY[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;        // row quad 0..3
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int srow = tid >> 2;          // staging row/col 0..255
+  const int sseg = tid & 3;
+
+  if (tid < 256) {
+    int gi = idx[rt * 256 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+  __syncthreads();
+
+  bf16x8 rx[2], rw[2];
+  auto issue = [&](int kc) {
+    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
+    rx[0] = gx[0];
+    rx[1] = gx[1];
+    const bf16x8* gw = reinterpret_cast<const bf16x8*>(
+        Wt + (long)srow * Fp + kc + sseg * 16);
+    rw[0] = gw[0];
+    rw[1] = gw[1];
+  };
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  issue(0);
+  commit(0);
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rquad * 64 + m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (kc + 1 < nkc) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 256 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = 1.f / (1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 256 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+    atomicAdd(&sDb[col_in], dsum);
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
+// 22M bank conflicts per dispatch of the generic [72]-stride tiles.
+// ---------------------------------------------------------------------------
+template <int NF>
+__global__ __launch_bounds__(256)
+void scores3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                    const long* __restrict__ idx,
+                    const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    const int* __restrict__ loss_id,
+                    const int* __restrict__ target,
+                    int C, int metric, long n_idx, int Fp, int Bp,
+                    int n_cand, double* __restrict__ out_sum) {
+  constexpr int colsPerWave = NF * 16;
+  const int scand_pad = (n_cand + 3) & ~3;
+  extern __shared__ char smem[];
+  long* sRow = reinterpret_cast<long*>(smem);                // [32]
+  float* zbuf = reinterpret_cast<float*>(sRow + 32);         // [32][Bp+8]
+  float* scand = zbuf + 32 * (Bp + 8);                       // [scand_pad]
+  __bf16* sX = reinterpret_cast<__bf16*>(scand + scand_pad); // [32][64]
+  __bf16* sW = sX + 32 * 64;                                 // [Bp][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+
+  if (tid < 32) {
+    long r = (long)rt * 32 + tid;
+    sRow[tid] = (r < n_idx) ? idx[r] : idx[n_idx - 1];
+  }
+  for (int i = tid; i < n_cand; i += 256) scand[i] = 0.f;
+  __syncthreads();
+
+  f32x4 acc[2][NF];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int kc = 0; kc < Fp; kc += 64) {
+    {
+      int r = tid >> 3, seg = tid & 7;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          X + sRow[r] * Fp + kc + seg * 8);
+      store_row8(&sX[r * 64], r, seg * 8, v);
+    }
+    for (int e = tid; e < Bp * 8; e += 256) {
+      int c = e >> 3, seg = e & 7;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          Wt + (long)c * Fp + kc + seg * 8);
+      store_row8(&sW[c * 64], c, seg * 8, v);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a0 = lds_frag_p(&sX[(lane & 15) * 64], lane & 15, g, h);
+      bf16x8 a1 = lds_frag_p(&sX[(16 + (lane & 15)) * 64],
+                             16 + (lane & 15), g, h);
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = wave * colsPerWave + fn * 16 + (lane & 15);
+        bf16x8 b = lds_frag_p(&sW[c * 64], c, g, h);
+        acc[0][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a0, b, acc[0][fn], 0, 0, 0);
+        acc[1][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a1, b, acc[1][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < NF; ++fn) {
+      int col = wave * colsPerWave + fn * 16 + (lane & 15);
+      float bia = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = fm * 16 + ((lane >> 4) << 2) + r;
+        zbuf[row * (Bp + 8) + col] = acc[fm][fn][r] + bia;
+      }
+    }
+  }
+  __syncthreads();
+
+  const float eps = 1e-12f;
+  for (int p = tid; p < 32 * n_cand; p += 256) {
+    int row = p & 31;
+    int cand = p >> 5;
+    long gr = (long)rt * 32 + row;
+    if (gr >= n_idx) continue;
+    float yv = y[sRow[row]];
+    const float* zr = &zbuf[row * (Bp + 8) + cand * C];
+    float val = 0.f;
+    if (metric == 2) {
+      float r = yv - zr[0];
+      val = r * r;
+    } else if (C == 1) {
+      float z = zr[0];
+      if (metric == 0) {
+        int pred = (z > 0.f) ? 1 : 0;
+        val = (pred == (int)yv) ? 1.f : 0.f;
+      } else {
+        float pr = 1.f / (1.f + __expf(-z));
+        float t = (yv == 1.f) ? 1.f : 0.f;
+        val = t * __logf(pr + eps) + (1.f - t) * __logf(1.f - pr + eps);
+      }
+    } else {
+      if (metric == 0) {
+        int best = 0;
+        float bz = zr[0];
+        for (int c = 1; c < C; ++c)
+          if (zr[c] > bz) { bz = zr[c]; best = c; }
+        val = (best == (int)yv) ? 1.f : 0.f;
+      } else {
+        float ssum = 0.f, sy = 0.f;
+        for (int c = 0; c < C; ++c) {
+          float s2 = 1.f / (1.f + __expf(-zr[c]));
+          ssum += s2;
+          if (c == (int)yv) sy = s2;
+        }
+        val = __logf(sy / fmaxf(ssum, eps) + eps);
+      }
+    }
+    atomicAdd(&scand[cand], val);
+  }
+  __syncthreads();
+  for (int cand = tid; cand < n_cand; cand += 256)
+    atomicAdd(&out_sum[cand], (double)scand[cand]);
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -1817,10 +2096,10 @@ std::vector<torch::Tensor> linear_fit(
   for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
     want_loss_cur = (any_tol || epoch == (int)max_epochs - 1) ? 1 : 0;
     if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
-      if (nf == 4 && NTp % 128 == 0 && Fp % 128 == 0) {
-        // wide 512-thread blocks (128-row / 128-feature tiles)
-        hipLaunchKernelGGL(fwd_d5_kernel, dim3((unsigned)(NTp / 128)),
-                           dim3(512), 0, s, Xp, y.data_ptr<float>(),
+      if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
+        // 1024-thread fwd (256x256 tile, 16 waves/CU)
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+                           dim3(1024), 0, s, Xp, y.data_ptr<float>(),
                            idx.data_ptr<int>(),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                            bias.data_ptr<float>(), loss_id.data_ptr<int>(),
@@ -1930,7 +2209,30 @@ torch::Tensor linear_scores(torch::Tensor X, torch::Tensor y,
                        target.data_ptr<int>(), (int)C, (int)metric, n_idx,
                        Fp, Bp, (int)n_cand, out.data_ptr<double>());
   };
+  size_t smem3 = 32 * 8 + 32 * (Bp + 8) * 4 + scand_pad * 4 +
+                 32 * 64 * 2 + (size_t)Bp * 64 * 2;
+  auto launch3 = [&](auto k) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(k),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem3));
+    hipLaunchKernelGGL(k, grid, dim3(256), smem3, s,
+                       reinterpret_cast<const __bf16*>(X.data_ptr()),
+                       y.data_ptr<float>(), idx.data_ptr<long>(),
+                       reinterpret_cast<const __bf16*>(Wt_bf16.data_ptr()),
+                       bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+                       target.data_ptr<int>(), (int)C, (int)metric, n_idx,
+                       Fp, Bp, (int)n_cand, out.data_ptr<double>());
+  };
   int nf = Bp / 64;
+  if (g_layout == 3) {
+    switch (nf) {
+      case 1: launch3(scores3_kernel<1>); break;
+      case 2: launch3(scores3_kernel<2>); break;
+      default: launch3(scores3_kernel<4>); break;
+    }
+    HIP_CHECK(hipGetLastError());
+    return out;
+  }
   switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
     case 0:  launch(scores_kernel<0, 1>); break;
     case 1:  launch(scores_kernel<0, 2>); break;

@@ -130,7 +130,7 @@ class DeviceLinearExecutor:
     def fit(self, train_idx, plan):
         torch = self.torch
         nt = len(train_idx)
-        NTp = _roundup(max(nt, 128), 128)   # 128: wide-block kernel tiles
+        NTp = _roundup(max(nt, 256), 256)   # 256: widest fwd kernel tile
         idx_np = np.zeros(NTp, dtype=np.int32)
         idx_np[:nt] = np.asarray(train_idx, dtype=np.int32)
         idx = torch.as_tensor(idx_np).to(self.device)
