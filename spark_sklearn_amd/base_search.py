@@ -227,6 +227,7 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 idx += 1
 
         backend = self._get_backend()
+        t_run0 = time.perf_counter()
         results = backend.run_search(
             estimator, X, y, tasks,
             scorer=self.scorer_,
@@ -235,6 +236,15 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             error_score=self.error_score,
             verbose=self.verbose,
         )
+        run_elapsed = time.perf_counter() - t_run0
+        # observability counters: candidate-fits/sec
+        # is the north-star metric; exposed on the fitted object
+        self.n_fits_ = len(tasks)
+        self.fit_duration_ = run_elapsed
+        self.fits_per_sec_ = len(tasks) / max(run_elapsed, 1e-9)
+        if self.verbose > 0:
+            print(f"[search] {self.n_fits_} fits in {run_elapsed:.2f}s "
+                  f"({self.fits_per_sec_:.1f} candidate-fits/s)")
         # restore deterministic submit order regardless of backend ordering
         by_index = {r.index: r for r in results}
         missing = [t.index for t in tasks if t.index not in by_index]
@@ -244,6 +254,10 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 f"tasks; missing indices {missing[:5]}..."
             )
         ordered = [by_index[i] for i in range(len(tasks))]
+
+        score_log = getattr(backend, "score_log", None)
+        if score_log and getattr(backend, "rank", 0) == 0:
+            self._append_score_log(score_log, tasks, ordered)
 
         self.cv_results_ = self._assemble_cv_results(
             candidates, ordered, n_candidates, n_splits)
@@ -269,6 +283,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self.refit_time_ = time.perf_counter() - t0
             self.best_estimator_ = best_estimator
         return self
+
+    def _append_score_log(self, path, tasks, ordered):
+        """Append-only JSONL score log: one record
+        per (candidate, fold) fit so a killed 100k-candidate sweep leaves
+        its completed work on disk."""
+        import json
+
+        with open(path, "a") as f:
+            for t, r in zip(tasks, ordered):
+                f.write(json.dumps({
+                    "cand": t.cand_index, "fold": t.fold_index,
+                    "params": {k: repr(v) for k, v in t.params.items()},
+                    "test_score": r.test_score,
+                    "fit_time": r.fit_time}) + "\n")
 
     # -- cv_results_ assembly ---------------------------------------------
     def _assemble_cv_results(self, candidates, ordered_results,

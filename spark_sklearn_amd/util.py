@@ -28,17 +28,21 @@ class HipContext:
 
     def __init__(self, backend=None, device=None, candidate_batch: int = 128,
                  n_streams: int = 2, dtype: str = "bfloat16",
-                 n_threads: Optional[int] = None):
+                 n_threads: Optional[int] = None, score_log: str = None,
+                 progress=None):
         from .parallel import resolve_backend
 
         self.candidate_batch = candidate_batch
         self.n_streams = n_streams
         self.dtype = dtype
         self.n_threads = n_threads
+        self.score_log = score_log
+        self.progress = progress
         self.device = device if device is not None else _default_device()
         self.backend = resolve_backend(backend)
         # push engine knobs down where backends look for them
-        for attr in ("candidate_batch", "n_streams", "dtype"):
+        for attr in ("candidate_batch", "n_streams", "dtype", "score_log",
+                     "progress"):
             setattr(self.backend, attr, getattr(self, attr))
 
     @property
