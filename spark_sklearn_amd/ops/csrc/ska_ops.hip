@@ -953,6 +953,109 @@ void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// bwd_g8_kernel: 1024-thread bwd for Bp=256, Fp multiple of 256 -- each
+// block owns the WHOLE 256x256 G tile for one split-K row segment
+// (16 waves = 4 cand-quads x 4 feature-waves), so D and X^T are staged
+// once per row chunk for all of G.  grid (Fp/256, nseg).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void bwd_g8_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                   const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                   long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sD[2][256 * 64];
+  __shared__ __bf16 sXt[2][256 * 64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cquad = wave >> 2;       // candidate quad 0..3
+  const int fwave = wave & 3;        // feature wave 0..3
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  const int g = lane >> 4;
+  const int srow = tid >> 2;         // staging row 0..255
+  const int sseg = tid & 3;
+
+  f32x4 acc[NF][4];
+#pragma unroll
+  for (int i = 0; i < NF; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  bf16x8 rd[2], rxa, rxb;
+  auto issue = [&](int rc) {
+    const bf16x8* gd = reinterpret_cast<const bf16x8*>(
+        Dt + (long)srow * NTp + rc + sseg * 16);
+    rd[0] = gd[0];
+    rd[1] = gd[1];
+    int rbase = rc + sseg * 16;
+    const __bf16* src = Xt + (long)(ft * 256 + srow) * n_rows;
+    int i0 = idx[rbase];
+    if (idx[rbase + 15] - i0 == 15 &&
+        (((unsigned long)(src + i0) & 15) == 0)) {
+      const bf16x8* gx = reinterpret_cast<const bf16x8*>(src + i0);
+      rxa = gx[0];
+      rxb = gx[1];
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxa[j] = src[idx[rbase + j]];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) rxb[j] = src[idx[rbase + 8 + j]];
+    }
+  };
+  auto commit = [&](int b) {
+    store_row8(&sD[b][srow * 64], srow, sseg * 16, rd[0]);
+    store_row8(&sD[b][srow * 64], srow, sseg * 16 + 8, rd[1]);
+    store_row8(&sXt[b][srow * 64], srow, sseg * 16, rxa);
+    store_row8(&sXt[b][srow * 64], srow, sseg * 16 + 8, rxb);
+  };
+
+  issue(r0);
+  commit(0);
+  __syncthreads();
+  for (int rc = r0; rc < r1; rc += 64) {
+    int b = (int)(((rc - r0) >> 6) & 1);
+    if (rc + 64 < r1) issue(rc + 64);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fwave * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = lds_frag_p(&sXt[b][f * 64], f, g, h);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = cquad * 64 + m * 16 + (lane & 15);
+        bf16x8 a = lds_frag_p(&sD[b][c * 64], c, g, h);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (rc + 64 < r1) commit(b ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fwave * 64 + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cquad * 64 + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 256 + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // col_finalize_kernel: per-column bias update + convergence, on-device.
 // ---------------------------------------------------------------------------
@@ -2108,12 +2211,21 @@ std::vector<torch::Tensor> linear_fit(
                            loss_sum.data_ptr<float>(),
                            db_sum.data_ptr<float>(), (int)nt, (int)NTp, Fp,
                            Bp, want_loss_cur);
-        hipLaunchKernelGGL(bwd_g5_kernel,
-                           dim3((unsigned)(Fp / 128), (unsigned)nseg),
-                           dim3(512), 0, s, Xtp, idx.data_ptr<int>(),
-                           reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                           Gt.data_ptr<float>(), n_rows, (int)NTp, Fp, Bp,
-                           seg_rows);
+        if (Fp % 256 == 0) {
+          hipLaunchKernelGGL(bwd_g8_kernel,
+                             dim3((unsigned)(Fp / 256), (unsigned)nseg),
+                             dim3(1024), 0, s, Xtp, idx.data_ptr<int>(),
+                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
+                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                             Bp, seg_rows);
+        } else {
+          hipLaunchKernelGGL(bwd_g5_kernel,
+                             dim3((unsigned)(Fp / 128), (unsigned)nseg),
+                             dim3(512), 0, s, Xtp, idx.data_ptr<int>(),
+                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
+                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                             Bp, seg_rows);
+        }
         hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
                            loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
                            bias.data_ptr<float>(), Vb.data_ptr<float>(),

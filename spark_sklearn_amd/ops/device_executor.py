@@ -125,15 +125,34 @@ class DeviceLinearExecutor:
         self.msq = float(self.ext.msq_rows(self.X)) * \
             (1.0 if self.F == self.Fp else 1.0)  # zero-pad adds nothing
         self._fit_state = None  # (Wt_dev, bias_dev) for the scores call
+        self._idx_cache: dict = {}   # fold index uploads (reused per fold)
+
+    def _cached_idx(self, arr, pad, dtype):
+        """Device index upload, cached per fold array (folds repeat for
+        every candidate chunk of a search)."""
+        key = (id(arr), len(arr), pad, str(dtype))
+        hit = self._idx_cache.get(key)
+        if hit is not None:
+            return hit
+        n = len(arr)
+        if pad:
+            NTp = _roundup(max(n, pad), pad)
+            out = np.zeros(NTp, dtype=dtype)
+            out[:n] = np.asarray(arr, dtype=dtype)
+        else:
+            out = np.asarray(arr, dtype=dtype)
+        t = self.torch.as_tensor(out).to(self.device)
+        if len(self._idx_cache) > 64:
+            self._idx_cache.clear()
+        self._idx_cache[key] = t
+        return t
 
     # -- fit ---------------------------------------------------------------
     def fit(self, train_idx, plan):
         torch = self.torch
         nt = len(train_idx)
-        NTp = _roundup(max(nt, 256), 256)   # 256: widest fwd kernel tile
-        idx_np = np.zeros(NTp, dtype=np.int32)
-        idx_np[:nt] = np.asarray(train_idx, dtype=np.int32)
-        idx = torch.as_tensor(idx_np).to(self.device)
+        idx = self._cached_idx(train_idx, 256, np.int32)
+        NTp = idx.shape[0]
 
         B = plan.B
         Bp = max(64, _roundup(B, 64))
@@ -183,8 +202,7 @@ class DeviceLinearExecutor:
             raise RuntimeError(f"unsupported column batch {Bp} for scoring")
         n_cand = int(plan.col_to_cand.max()) + 1
         C = plan.n_classes
-        idx_t = torch.as_tensor(
-            np.asarray(idx, dtype=np.int64)).to(self.device)
+        idx_t = self._cached_idx(idx, 0, np.int64)
         sums = self.ext.linear_scores(
             self.X, self.y, idx_t, Wt, bias, loss_t, tgt_t,
             C, _METRIC_ID[metric], n_cand).cpu().numpy()
