@@ -1,0 +1,102 @@
+#!/usr/bin/env python3
+"""Secondary family benchmarks (BASELINE configs 3-5 evidence).
+
+Prints one JSON line per family.  Run on an MI355X box:
+    python benchmarks/bench_families.py
+"""
+
+import json
+import sys
+import time
+
+sys.path.insert(0, "/root/repo")
+
+import numpy as np
+
+
+def bench_kmeans():
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import KMeans
+
+    rng = np.random.default_rng(0)
+    n, f, true_k = 1_000_000, 256, 16
+    centers = rng.standard_normal((true_k, f)).astype(np.float32) * 3
+    lab = rng.integers(0, true_k, n)
+    X = centers[lab] + rng.standard_normal((n, f), dtype=np.float32)
+
+    grid = {"n_clusters": [4, 8, 12, 16, 24, 32], "n_init": [2]}
+    gs = ska.GridSearchCV(None, KMeans(random_state=0, max_iter=20, tol=0.0),
+                          grid, cv=5, return_train_score=False)
+    t0 = time.perf_counter()
+    gs.fit(X)
+    dt = time.perf_counter() - t0
+    return {"family": "kmeans", "config": "1Mx256, k in 4..32 x n_init=2, "
+            "cv=5, 20 Lloyd iters", "n_fits": gs.n_fits_,
+            "wall_s": round(dt, 2),
+            "fits_per_sec": round(gs.n_fits_ / dt, 2),
+            "best": gs.best_params_}
+
+
+def bench_gbt():
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import GradientBoostingClassifier
+
+    rng = np.random.default_rng(0)
+    n, f = 1_000_000, 128
+    X = rng.standard_normal((n, f), dtype=np.float32)
+    y = ((X[:, 0] * X[:, 1] + X[:, 2] + X[:, 3] ** 2 - 1 +
+          0.5 * rng.standard_normal(n)) > 0).astype(np.int64)
+
+    grid = {"max_depth": [3, 5], "learning_rate": [0.1, 0.3]}
+    gs = ska.GridSearchCV(None,
+                          GradientBoostingClassifier(n_estimators=50),
+                          grid, cv=3, return_train_score=False)
+    t0 = time.perf_counter()
+    gs.fit(X, y)
+    dt = time.perf_counter() - t0
+    return {"family": "gbt", "config": "1Mx128, 50 trees, depth 3/5, cv=3",
+            "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
+            "fits_per_sec": round(gs.n_fits_ / dt, 3),
+            "trees_per_sec": round(gs.n_fits_ * 50 / dt, 1),
+            "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
+
+
+def bench_sweep():
+    """Config 5: large randomized sweep with successive-halving pruning."""
+    from scipy.stats import loguniform
+
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    n, f = 1_000_000, 256
+    X = rng.standard_normal((n, f), dtype=np.float32)
+    w = rng.standard_normal(f, dtype=np.float32) / np.sqrt(f)
+    y = ((X @ w + 0.25 * rng.standard_normal(n, dtype=np.float32)) > 0
+         ).astype(np.int64)
+
+    rs = ska.RandomizedSearchCV(
+        None, LogisticRegression(max_iter=25, tol=0.0),
+        {"C": loguniform(1e-4, 1e3), "penalty": ["l1", "l2"]},
+        n_iter=10_000, cv=5, random_state=0,
+        halving_factor=4, min_resource_frac=0.12,
+        return_train_score=False)
+    t0 = time.perf_counter()
+    rs.fit(X, y)
+    dt = time.perf_counter() - t0
+    res = rs.cv_results_
+    return {"family": "randomized+halving",
+            "config": "10k candidates, 1Mx256, cv=5, factor 4",
+            "wall_s": round(dt, 2),
+            "final_rung_candidates": int(np.sum(
+                res["resource_iter"] == 25)),
+            "best_score": round(rs.best_score_, 4),
+            "candidates_per_sec": round(10_000 / dt, 1)}
+
+
+if __name__ == "__main__":
+    which = sys.argv[1:] or ["kmeans", "gbt", "sweep"]
+    for name in which:
+        fn = {"kmeans": bench_kmeans, "gbt": bench_gbt,
+              "sweep": bench_sweep}[name]
+        print(json.dumps(fn()), flush=True)

@@ -38,11 +38,27 @@ from numpy.ma import MaskedArray
 from scipy.stats import rankdata
 
 from sklearn.base import BaseEstimator, MetaEstimatorMixin, clone, is_classifier
+from sklearn.utils.metaestimators import available_if
 from sklearn.metrics import check_scoring
 from sklearn.model_selection import check_cv
 from sklearn.utils.validation import indexable
 
 __all__ = ["Task", "TaskResult", "fit_and_score_task", "BaseSearchCV"]
+
+
+def _search_estimator_has(attr):
+    """available_if check: delegate attribute availability to the
+    (fitted) estimator; refit=False raises the sklearn message chain."""
+
+    def check(self):
+        self._check_refit_for(attr)
+        if hasattr(self, "best_estimator_"):
+            getattr(self.best_estimator_, attr)
+            return True
+        getattr(self.estimator, attr)
+        return True
+
+    return check
 
 
 class Task:
@@ -77,15 +93,18 @@ class TaskResult:
     parallel/distributed.py).
     """
 
-    __slots__ = ("index", "test_score", "train_score", "n_test", "fit_time", "score_time")
+    __slots__ = ("index", "test_score", "train_score", "n_test",
+                 "fit_time", "score_time", "error")
 
-    def __init__(self, index, test_score, train_score, n_test, fit_time, score_time):
+    def __init__(self, index, test_score, train_score, n_test, fit_time,
+                 score_time, error=None):
         self.index = index
         self.test_score = test_score
         self.train_score = train_score
         self.n_test = n_test
         self.fit_time = fit_time
         self.score_time = score_time
+        self.error = error     # formatted exception text for failed fits
 
 
 def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
@@ -101,11 +120,29 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
     """
     est = clone(base_estimator)
     if task.params:
-        est.set_params(**task.params)
+        # clone parameter values that are themselves estimators, so grid
+        # entries are never fitted in place (sklearn contract)
+        est.set_params(**{k: clone(v, safe=False)
+                          for k, v in task.params.items()})
     fit_params = fit_params or {}
 
-    X_train, y_train = _safe_index(X, task.train), _safe_index(y, task.train)
-    X_test, y_test = _safe_index(X, task.test), _safe_index(y, task.test)
+    if _is_pairwise(est):
+        # precomputed kernels/distances: slice BOTH axes (train columns)
+        import scipy.sparse as _sp
+
+        if not (hasattr(X, "shape") or _sp.issparse(X)):
+            raise ValueError(
+                "Precomputed kernels or affinity matrices have to be "
+                "passed as arrays or sparse matrices.")
+        X = np.asarray(X)
+        if X.ndim != 2 or X.shape[0] != X.shape[1]:
+            raise ValueError(
+                "X should be a square kernel matrix")
+        X_train = X[np.ix_(task.train, task.train)]
+        X_test = X[np.ix_(task.test, task.train)]
+    else:
+        X_train, X_test = _safe_index(X, task.train), _safe_index(X, task.test)
+    y_train, y_test = _safe_index(y, task.train), _safe_index(y, task.test)
 
     t0 = time.perf_counter()
     try:
@@ -115,9 +152,16 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
             est.fit(X_train, y_train, **fit_params)
         fit_time = time.perf_counter() - t0
         t1 = time.perf_counter()
-        test_score = scorer(est, X_test, y_test)
-        score_time = time.perf_counter() - t1
-        train_score = scorer(est, X_train, y_train) if return_train_score else np.nan
+        if y_test is None:
+            test_score = scorer(est, X_test)
+            score_time = time.perf_counter() - t1
+            train_score = scorer(est, X_train) if return_train_score \
+                else np.nan
+        else:
+            test_score = scorer(est, X_test, y_test)
+            score_time = time.perf_counter() - t1
+            train_score = scorer(est, X_train, y_train) \
+                if return_train_score else np.nan
     except Exception as exc:  # noqa: BLE001 - error_score contract
         fit_time = time.perf_counter() - t0
         if error_score == "raise":
@@ -126,19 +170,26 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
             raise ValueError(
                 "error_score must be the string 'raise' or a numeric value"
             ) from exc
-        warnings.warn(
-            f"Estimator fit failed for params {task.params!r}; the score on "
-            f"this train-test partition will be set to {error_score}. "
-            f"Details: {exc!r}",
-            stacklevel=2,
-        )
-        test_score = float(error_score)
-        train_score = float(error_score) if return_train_score else np.nan
-        score_time = 0.0
+        import traceback
+
+        err = traceback.format_exc()
+        n_test = len(task.test)
+        return TaskResult(task.index, float(error_score),
+                          float(error_score) if return_train_score else np.nan,
+                          n_test, fit_time, 0.0, error=err)
 
     n_test = len(task.test)
     return TaskResult(task.index, float(test_score), float(train_score),
                       n_test, fit_time, score_time)
+
+
+def _is_pairwise(estimator) -> bool:
+    try:
+        from sklearn.utils import get_tags
+
+        return bool(get_tags(estimator).input_tags.pairwise)
+    except Exception:
+        return False
 
 
 def _safe_index(data, indices):
@@ -203,6 +254,18 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         """Run the full search: build tasks, dispatch, aggregate, refit."""
         estimator = self.estimator
         cv = check_cv(self.cv, y, classifier=is_classifier(estimator))
+        if isinstance(self.scoring, (dict, list, tuple, set)):
+            if self.refit:
+                raise ValueError(
+                    "For multi-metric scoring, the parameter refit must be "
+                    "set to a scorer key or a callable to refit an "
+                    "estimator with the best parameter setting on the "
+                    "whole data and make the best_* attributes available "
+                    "for that metric. If this is not needed, refit should "
+                    "be set to False explicitly.")
+            raise NotImplementedError(
+                "multi-metric scoring is not supported by this engine yet; "
+                "run one search per metric")
         self.scorer_ = check_scoring(estimator, scoring=self.scoring)
 
         X, y, groups = indexable(X, y, groups)
@@ -213,6 +276,16 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         n_candidates = len(candidates)
         splits = list(cv.split(X, y, groups))
         n_splits = len(splits)
+        expected_splits = cv.get_n_splits(X, y, groups)
+        if n_splits * n_candidates == 0:
+            raise ValueError(
+                "No fits were performed. Was the CV iterator empty? "
+                "Were there no candidates?")
+        if expected_splits != n_splits:
+            raise ValueError(
+                "cv.split and cv.get_n_splits returned inconsistent "
+                f"results. Expected {expected_splits} splits, got "
+                f"{n_splits}")
 
         if self.verbose > 0:
             print(f"Fitting {n_splits} folds for each of {n_candidates} "
@@ -254,6 +327,7 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 f"tasks; missing indices {missing[:5]}..."
             )
         ordered = [by_index[i] for i in range(len(tasks))]
+        self._handle_fit_failures(tasks, ordered)
 
         score_log = getattr(backend, "score_log", None)
         if score_log and getattr(backend, "rank", 0) == 0:
@@ -263,11 +337,21 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             candidates, ordered, n_candidates, n_splits)
         self.n_splits_ = n_splits
         self.multimetric_ = False
-        self.best_index_ = int(
-            np.flatnonzero(self.cv_results_["rank_test_score"] == 1)[0])
+        if callable(self.refit):
+            # sklearn contract: refit may be a callable selecting
+            # best_index_ from cv_results_; best_score_ is then undefined
+            best = self.refit(self.cv_results_)
+            if not isinstance(best, (int, np.integer)):
+                raise TypeError("best_index_ returned is not an integer")
+            if not 0 <= best < n_candidates:
+                raise IndexError("best_index_ index out of range")
+            self.best_index_ = int(best)
+        else:
+            self.best_index_ = int(
+                np.flatnonzero(self.cv_results_["rank_test_score"] == 1)[0])
+            self.best_score_ = float(
+                self.cv_results_["mean_test_score"][self.best_index_])
         self.best_params_ = self.cv_results_["params"][self.best_index_]
-        self.best_score_ = float(
-            self.cv_results_["mean_test_score"][self.best_index_])
 
         if self.refit:
             best_estimator = clone(estimator).set_params(**self.best_params_)
@@ -283,6 +367,39 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self.refit_time_ = time.perf_counter() - t0
             self.best_estimator_ = best_estimator
         return self
+
+    def _handle_fit_failures(self, tasks, ordered):
+        """sklearn failure contract: one aggregated FitFailedWarning when
+        some fits failed under a numeric error_score; ValueError when ALL
+        fits failed."""
+        failed = [r for r in ordered if r.error is not None]
+        if not failed:
+            return
+        from collections import Counter
+
+        from sklearn.exceptions import FitFailedWarning
+
+        n, nf = len(ordered), len(failed)
+        by_msg = Counter(r.error for r in failed)
+        details = "\n".join(
+            f"--------------------------------------------------------\n"
+            f"{cnt} fits failed with the following error:\n{msg}"
+            for msg, cnt in by_msg.items())
+        if nf == n:
+            raise ValueError(
+                f"\nAll the {n} fits failed.\n"
+                "It is very likely that your model is misconfigured.\n"
+                "You can try to debug the error by setting "
+                "error_score='raise'.\n\n"
+                f"Below are more details about the failures:\n{details}")
+        warnings.warn(
+            f"\n{nf} fits failed out of a total of {n}.\n"
+            "The score on these train-test partitions for these parameters"
+            f" will be set to {self.error_score}.\n"
+            "If these failures are not expected, you can try to debug them "
+            "by setting error_score='raise'.\n\n"
+            f"Below are more details about the failures:\n{details}",
+            FitFailedWarning)
 
     def _append_score_log(self, path, tasks, ordered):
         """Append-only JSONL score log: one record
@@ -334,9 +451,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             results[f"std_{key}"] = np.sqrt(
                 np.average((array - means[:, np.newaxis]) ** 2,
                            axis=1, weights=weights))
+            if key.startswith(("train_", "test_")) and \
+                    np.any(~np.isfinite(means)):
+                warnings.warn(
+                    f"One or more of the {key.split('_')[0]} scores "
+                    f"are non-finite: {means}", UserWarning)
             if rank:
-                results[f"rank_{key}"] = np.asarray(
-                    rankdata(-means, method="min"), dtype=np.int32)
+                if np.isnan(means).all():
+                    results[f"rank_{key}"] = np.ones_like(
+                        means, dtype=np.int32)
+                else:
+                    filled = np.nan_to_num(means,
+                                           nan=np.nanmin(means) - 1)
+                    results[f"rank_{key}"] = np.asarray(
+                        rankdata(-filled, method="min"), dtype=np.int32)
 
         score_weights = n_test if self.iid else None
         _store("fit_time", fit_times)
@@ -349,61 +477,93 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             ).reshape(n_candidates, n_splits)
             _store("train_score", train_scores, splits=True)
 
-        # masked param_* arrays: mask=True where a candidate lacks that param
-        # (heterogeneous grids) -- reference grid_search.py ~L270 [R]
-        param_results: Dict[str, MaskedArray] = defaultdict(
-            partial(MaskedArray, np.empty(n_candidates), mask=True,
-                    dtype=object))
+        # masked param_* arrays: mask=True where a candidate lacks that
+        # param (heterogeneous grids) -- reference grid_search.py ~L270 [R].
+        # dtype semantics follow the sklearn contract: inferred from the
+        # present values; strings and ragged sequences fall back to object.
+        param_values: Dict[str, dict] = defaultdict(dict)
         for cand_i, params in enumerate(candidates):
             for name, value in params.items():
-                param_results[f"param_{name}"][cand_i] = value
-        results.update(param_results)
+                param_values[f"param_{name}"][cand_i] = value
+        for key, present in param_values.items():
+            plist = list(present.values())
+            try:
+                arr = np.array(plist)
+                dtype = (arr.dtype if arr.dtype.kind != "U" and
+                         arr.ndim == 1 else object)
+            except ValueError:
+                dtype = object
+            ma = MaskedArray(np.empty(n_candidates, dtype=dtype), mask=True)
+            for i, v in present.items():
+                ma[i] = v
+            results[key] = ma
         results["params"] = list(candidates)
         return results
 
     # -- post-fit surface delegated to best_estimator_ ---------------------
-    def _check_is_fitted(self, method_name):
+    # available_if: the method exists on the search object only when it
+    # exists on the (fitted or prototype) estimator, and refit=False makes
+    # access raise AttributeError with sklearn's exact message chain.
+    def _check_refit_for(self, attr):
         if not self.refit:
             raise AttributeError(
                 f"This {type(self).__name__} instance was initialized with "
-                f"refit=False; {method_name} is only available after "
-                f"refitting on the best parameters.")
-        if not hasattr(self, "best_estimator_"):
-            raise AttributeError(
-                f"This {type(self).__name__} instance is not fitted yet.")
+                f"`refit=False`. {attr} is available only after refitting "
+                "on the best parameters. You can refit an estimator "
+                "manually using the `best_params_` attribute")
 
+    @available_if(_search_estimator_has("predict"))
     def predict(self, X):
-        self._check_is_fitted("predict")
         return self.best_estimator_.predict(X)
 
+    @available_if(_search_estimator_has("predict_proba"))
     def predict_proba(self, X):
-        self._check_is_fitted("predict_proba")
         return self.best_estimator_.predict_proba(X)
 
+    @available_if(_search_estimator_has("predict_log_proba"))
     def predict_log_proba(self, X):
-        self._check_is_fitted("predict_log_proba")
         return self.best_estimator_.predict_log_proba(X)
 
+    @available_if(_search_estimator_has("decision_function"))
     def decision_function(self, X):
-        self._check_is_fitted("decision_function")
         return self.best_estimator_.decision_function(X)
 
+    @available_if(_search_estimator_has("transform"))
     def transform(self, X):
-        self._check_is_fitted("transform")
         return self.best_estimator_.transform(X)
 
+    @available_if(_search_estimator_has("inverse_transform"))
     def inverse_transform(self, X):
-        self._check_is_fitted("inverse_transform")
         return self.best_estimator_.inverse_transform(X)
 
     def score(self, X, y=None):
         if hasattr(self, "scorer_") and self.scorer_ is not None and \
                 hasattr(self, "best_estimator_"):
             return self.scorer_(self.best_estimator_, X, y)
-        self._check_is_fitted("score")
+        self._check_refit_for("score")
         return self.best_estimator_.score(X, y)
 
     @property
     def classes_(self):
-        self._check_is_fitted("classes_")
+        self._check_refit_for("classes_")
         return self.best_estimator_.classes_
+
+    @property
+    def n_features_in_(self):
+        if not hasattr(self, "best_estimator_"):
+            raise AttributeError(
+                f"{type(self).__name__} object has no n_features_in_ "
+                "attribute (call fit first)")
+        return self.best_estimator_.n_features_in_
+
+    def __sklearn_tags__(self):
+        tags = super().__sklearn_tags__()
+        try:
+            from sklearn.utils import get_tags
+
+            sub = get_tags(self.estimator)
+            tags.input_tags.pairwise = sub.input_tags.pairwise
+            tags.estimator_type = sub.estimator_type
+        except Exception:
+            pass
+        return tags

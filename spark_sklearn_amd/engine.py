@@ -273,20 +273,20 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             per_fit = fit_time / len(chunk)
             per_score = score_time / len(chunk)
             for slot, task in enumerate(chunk):
+                err = None
                 if slot in bad:
                     if not isinstance(error_score, (int, float)):
                         raise bad[slot]
                     ts = tr = float(error_score)
-                    warnings.warn(
-                        f"candidate {task.params!r} failed plan validation "
-                        f"({bad[slot]!r}); scoring error_score="
-                        f"{error_score}", stacklevel=2)
+                    err = (f"{type(bad[slot]).__name__}: {bad[slot]}\n"
+                           f"(candidate {task.params!r} failed plan "
+                           f"validation)")
                 else:
                     ts = float(test_scores[slot])
                     tr = float(train_scores[slot])
                 results.append(TaskResult(task.index, ts, tr,
                                           len(task.test), per_fit,
-                                          per_score))
+                                          per_score, error=err))
         if verbose > 1:
             print(f"[engine] fold {fold_index}: "
                   f"{len(fold_tasks)} candidates done")

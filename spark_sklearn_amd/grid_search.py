@@ -71,21 +71,8 @@ class GridSearchCV(BaseSearchCV):
             return_train_score=return_train_score)
         self.ctx = ctx
         self.param_grid = param_grid
-        _check_param_grid(param_grid)
 
     def _get_param_iterator(self):
+        # sklearn's ParameterGrid performs the exact upstream validation
+        # (typed messages for scalars / empty sequences) at fit time
         return ParameterGrid(self.param_grid)
-
-
-def _check_param_grid(param_grid):
-    if hasattr(param_grid, "items"):
-        param_grid = [param_grid]
-    for p in param_grid:
-        for name, v in p.items():
-            if isinstance(v, str) or not hasattr(v, "__iter__"):
-                raise ValueError(
-                    f"Parameter grid value for '{name}' needs to be a "
-                    f"non-string sequence, got {v!r}")
-            if len(list(v)) == 0:
-                raise ValueError(
-                    f"Parameter grid for '{name}' cannot be empty")

@@ -124,16 +124,25 @@ class DistributedBackend(Backend):
         comm_dev = self.device if (self.device is not None and
                                    self.device.type == "cuda" and
                                    dist.get_backend() == "nccl") else "cpu"
-        table = torch.zeros((len(tasks), 6), dtype=torch.float64,
+        table = torch.zeros((len(tasks), 7), dtype=torch.float64,
                             device=comm_dev)
         for r in results:
             table[r.index] = torch.tensor(
                 [r.index, r.test_score, r.train_score, r.n_test,
-                 r.fit_time, r.score_time], dtype=torch.float64)
+                 r.fit_time, r.score_time,
+                 1.0 if r.error is not None else 0.0],
+                dtype=torch.float64)
         dist.all_reduce(table, op=dist.ReduceOp.SUM)
         table = table.cpu().numpy()
+        local_err = {r.index: r.error for r in results
+                     if r.error is not None}
         return [TaskResult(int(row[0]), row[1], row[2], int(row[3]),
-                           row[4], row[5]) for row in table]
+                           row[4], row[5],
+                           error=(local_err.get(int(row[0]),
+                                  "fit failed on a remote rank "
+                                  "(error text not gathered)")
+                                  if row[6] > 0 else None))
+                for row in table]
 
     # -- rank-0 refit + parameter broadcast --------------------------------
     def refit(self, estimator, X, y, fit_params):

@@ -119,6 +119,8 @@ def test_invalid_params_error_score(blobs):
     gs = ska.GridSearchCV(None, KMeans(random_state=0),
                           {"n_clusters": [-1, 4]}, cv=3,
                           error_score=-1e18, return_train_score=False)
-    with pytest.warns(UserWarning):
+    from sklearn.exceptions import FitFailedWarning
+
+    with pytest.warns(FitFailedWarning):
         gs.fit(X)
     assert gs.best_params_["n_clusters"] == 4

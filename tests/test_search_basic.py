@@ -37,9 +37,11 @@ def _xy(n=200, f=8, seed=0):
 
 def test_error_score_numeric_swallows():
     X, y = _xy()
+    from sklearn.exceptions import FitFailedWarning
+
     gs = ska.GridSearchCV(None, FailingEstimator(), {"C": [-1.0, 1.0]},
                           cv=3, error_score=-999.0, refit=True)
-    with pytest.warns(UserWarning):
+    with pytest.warns(FitFailedWarning, match="3 fits failed out of"):
         gs.fit(X, y)
     assert gs.cv_results_["mean_test_score"][0] == -999.0
     assert gs.best_params_ == {"C": 1.0}
@@ -63,10 +65,11 @@ def test_refit_false_blocks_predict():
 
 
 def test_param_grid_validation():
+    X, y = _xy()
+    with pytest.raises(TypeError):
+        ska.GridSearchCV(None, svm.SVC(), {"C": "not-a-sequence"}).fit(X, y)
     with pytest.raises(ValueError):
-        ska.GridSearchCV(None, svm.SVC(), {"C": "not-a-sequence"})
-    with pytest.raises(ValueError):
-        ska.GridSearchCV(None, svm.SVC(), {"C": []})
+        ska.GridSearchCV(None, svm.SVC(), {"C": []}).fit(X, y)
 
 
 def test_verbose_and_groups(capsys):
