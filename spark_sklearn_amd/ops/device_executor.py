@@ -76,8 +76,9 @@ def _as_device_bf16(X, device):
     else:
         key = (id(X), X.shape, str(device))
     hit = _data_cache.get(key)
-    if hit is not None:
-        return hit
+    if hit is not None and (isinstance(X, torch.Tensor) or
+                            hit[0] is X):
+        return hit[1]
     if sp.issparse(X):
         Xc = sp.csr_matrix(X)
         n, F = Xc.shape
@@ -102,7 +103,7 @@ def _as_device_bf16(X, device):
     dev_tt = dev_t.t().contiguous()
     if len(_data_cache) > 4:
         _data_cache.clear()
-    _data_cache[key] = (dev_t, dev_tt, F)
+    _data_cache[key] = (X, (dev_t, dev_tt, F))
     return dev_t, dev_tt, F
 
 
@@ -129,11 +130,15 @@ class DeviceLinearExecutor:
 
     def _cached_idx(self, arr, pad, dtype):
         """Device index upload, cached per fold array (folds repeat for
-        every candidate chunk of a search)."""
+        every candidate chunk of a search).
+
+        The cache entry HOLDS the source array: keying on id() alone is
+        unsound because Python recycles ids after garbage collection --
+        a fresh fold array could silently alias a dead one's entry."""
         key = (id(arr), len(arr), pad, str(dtype))
         hit = self._idx_cache.get(key)
-        if hit is not None:
-            return hit
+        if hit is not None and hit[0] is arr:
+            return hit[1]
         n = len(arr)
         if pad:
             NTp = _roundup(max(n, pad), pad)
@@ -144,7 +149,7 @@ class DeviceLinearExecutor:
         t = self.torch.as_tensor(out).to(self.device)
         if len(self._idx_cache) > 64:
             self._idx_cache.clear()
-        self._idx_cache[key] = t
+        self._idx_cache[key] = (arr, t)
         return t
 
     # -- fit ---------------------------------------------------------------

@@ -25,7 +25,7 @@ class DeviceGBTExecutor:
         self.ext = load_extension()
         key = (id(X), getattr(X, "shape", None), str(device))
         hit = _bin_cache.get(key)
-        if hit is None:
+        if hit is None or hit[0] is not X:
             Xnp = np.asarray(X) if not hasattr(X, "cpu") else \
                 X.cpu().float().numpy()
             edges = quantile_bins(Xnp)
@@ -33,9 +33,10 @@ class DeviceGBTExecutor:
             Xb_t = torch.as_tensor(Xb).to(device)
             if len(_bin_cache) > 2:
                 _bin_cache.clear()
-            _bin_cache[key] = (Xb_t, edges)
+            # the entry HOLDS X: id() alone is unsound after GC recycling
+            _bin_cache[key] = (X, (Xb_t, edges))
             hit = _bin_cache[key]
-        self.Xb, self.edges = hit
+        self.Xb, self.edges = hit[1]
         self.F = self.Xb.shape[1]
         self.y = torch.as_tensor(
             np.asarray(y_enc, dtype=np.float32)).to(device)
