@@ -238,6 +238,14 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     for t in tasks:
         by_fold.setdefault(t.fold_index, []).append(t)
 
+    if family is LinearFamily and device is not None and \
+            metric is not None and not fit_params:
+        fast = _linear_device_search(
+            estimator, executor, by_fold, n_classes, metric,
+            return_train_score, error_score, verbose)
+        if fast is not None:
+            return fast
+
     cols_per_cand = family.cols_per_cand(n_classes)
     cand_chunk = family.cand_chunk(estimator, [t.params for t in tasks],
                                    column_batch)
@@ -290,6 +298,120 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
         if verbose > 1:
             print(f"[engine] fold {fold_index}: "
                   f"{len(fold_tasks)} candidates done")
+    return results
+
+
+def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
+                          return_train_score, error_score, verbose):
+    """Single native call for the whole (fold x chunk) loop: the C++
+    engine core (ops/csrc linear_search_run).  Returns None when the
+    search shape doesn't fit the fast path (odd class counts, mismatched
+    fold candidate lists), in which case the generic per-chunk loop runs.
+    """
+    import torch
+
+    from .models._solver_ref import BatchPlan
+
+    C = LinearFamily.cols_per_cand(n_classes)
+    if C > 256 or 256 % C != 0:
+        return None
+    folds = sorted(by_fold)
+    first = by_fold[folds[0]]
+    n_cand = len(first)
+    for f in folds[1:]:
+        if len(by_fold[f]) != n_cand or any(
+                a.cand_index != b.cand_index
+                for a, b in zip(first, by_fold[f])):
+            return None
+
+    params_list = [t.params for t in first]
+    plan, bad = LinearFamily.build_plan(
+        estimator, params_list, len(first[0].train), executor,
+        n_classes, error_score)
+
+    cand_chunk = 256 // C
+    n_chunks = (n_cand + cand_chunk - 1) // cand_chunk
+    Bt = n_chunks * 256
+
+    def pad_chunked(src, fill, dtype):
+        out = np.full(Bt, fill, dtype=dtype)
+        for ci in range(n_chunks):
+            lo = ci * cand_chunk
+            hi = min(lo + cand_chunk, n_cand)
+            seg = src[lo * C:hi * C]
+            out[ci * 256:ci * 256 + len(seg)] = seg
+        return torch.as_tensor(out).to(executor.device)
+
+    loss_t = pad_chunked(plan.loss, 0, np.int32)
+    lr_t = pad_chunked(plan.lr, 0.0, np.float32)
+    l1_t = pad_chunked(plan.l1, 0.0, np.float32)
+    l2_t = pad_chunked(plan.l2, 0.0, np.float32)
+    mom_t = pad_chunked(plan.momentum, 0.0, np.float32)
+    tol_t = pad_chunked(plan.tol, 0.0, np.float32)
+    mi_t = pad_chunked(plan.max_iter, 0, np.int32)
+    tgt_t = pad_chunked(plan.target_class, 0, np.int32)
+    ic_t = pad_chunked(plan.fit_intercept, 0, np.uint8)
+
+    train_ts, nts, test_ts = [], [], []
+    for f in folds:
+        tr = by_fold[f][0].train
+        train_ts.append(executor._cached_idx(tr, 256, np.int32))
+        nts.append(len(tr))
+        test_ts.append(executor._cached_idx(by_fold[f][0].test, 0,
+                                            np.int64))
+    if return_train_score:
+        for f in folds:
+            test_ts.append(executor._cached_idx(by_fold[f][0].train, 0,
+                                                np.int64))
+
+    from .ops.device_executor import _METRIC_ID
+
+    t0 = time.perf_counter()
+    out = executor.ext.linear_search_run(
+        executor.X, executor.Xt, executor.y, train_ts, nts, test_ts,
+        loss_t, lr_t, l1_t, l2_t, mom_t, tol_t, mi_t, tgt_t, ic_t,
+        C, _METRIC_ID[metric], n_cand, 1 if return_train_score else 0,
+        32)
+    test_sums, train_sums, fit_ms, score_ms, _ = out
+    test_sums = test_sums.cpu().numpy()
+    train_sums = train_sums.cpu().numpy() if return_train_score else None
+    fit_ms = fit_ms.numpy()
+    score_ms = score_ms.numpy()
+
+    def finalize(sums, idx_len, yv):
+        if metric in ("accuracy", "neg_log_loss"):
+            return sums / idx_len
+        ss_tot = float(np.sum((yv - yv.mean()) ** 2))
+        return 1.0 - sums / max(ss_tot, 1e-12)
+
+    y_np = executor.y.cpu().numpy()
+    results = []
+    for fi, f in enumerate(folds):
+        te = by_fold[f][0].test
+        tr = by_fold[f][0].train
+        ts = finalize(test_sums[fi], len(te), y_np[np.asarray(te)])
+        if return_train_score:
+            trs = finalize(train_sums[fi], len(tr), y_np[np.asarray(tr)])
+        for slot, task in enumerate(by_fold[f]):
+            chunk = slot // cand_chunk
+            per_fit = float(fit_ms[fi][chunk]) / 1000.0 / cand_chunk
+            per_score = float(score_ms[fi][chunk]) / 1000.0 / cand_chunk
+            if slot in bad:
+                if not isinstance(error_score, (int, float)):
+                    raise bad[slot]
+                v = tv = float(error_score)
+                err = (f"{type(bad[slot]).__name__}: {bad[slot]}\n"
+                       f"(candidate {task.params!r} failed plan "
+                       f"validation)")
+            else:
+                v = float(ts[slot])
+                tv = float(trs[slot]) if return_train_score else float("nan")
+                err = None
+            results.append(TaskResult(task.index, v, tv, len(task.test),
+                                      per_fit, per_score, error=err))
+    if verbose > 1:
+        print(f"[engine] native search_run: {len(folds)} folds x "
+              f"{n_chunks} chunks in {time.perf_counter() - t0:.2f}s")
     return results
 
 

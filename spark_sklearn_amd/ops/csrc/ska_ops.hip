@@ -2672,6 +2672,250 @@ torch::Tensor gbt_metric(torch::Tensor raw, torch::Tensor y,
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// linear_search_run: the whole (fold x candidate-chunk) loop natively.
+//
+// One call executes an entire search shard on the device -- for each fold:
+// for each 256-column candidate chunk: reset state, run the epoch loop
+// (fwd/bwd/finalize/update), then score the test (and optionally train)
+// fold -- with NO host synchronization between batches.  Per-(fold, chunk)
+// fit/score times come from HIP events and populate cv_results_'s
+// *_fit_time / *_score_time.  This is the C++ engine core of SURVEY.md
+// section 7.1 (the Python layer builds plans and assembles cv_results_).
+//
+// Requirements: Bp_total % 256 == 0 (pad with max_iter=0 dummy columns),
+// layout P=3 (probed), candidate chunks aligned to candidate boundaries.
+// Returns {test_sums [n_folds][n_cand], train_sums or empty,
+//          fit_ms [n_folds][n_chunks], score_ms [n_folds][n_chunks],
+//          n_iter [n_folds][Bp_total]}.
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> linear_search_run(
+    torch::Tensor X, torch::Tensor Xt, torch::Tensor y,
+    std::vector<torch::Tensor> train_idx,   // i32 [NTp_f], 256-padded
+    std::vector<int64_t> nt_list,
+    std::vector<torch::Tensor> test_idx,    // i64 [n_test_f]
+    torch::Tensor loss_id, torch::Tensor lr, torch::Tensor l1,
+    torch::Tensor l2, torch::Tensor mom, torch::Tensor tol,
+    torch::Tensor max_iter, torch::Tensor target, torch::Tensor fit_ic,
+    int64_t C, int64_t metric, int64_t n_cand_total,
+    int64_t want_train_scores, int64_t check_every) {
+  TORCH_CHECK(g_layout == 3, "linear_search_run requires probed layout P=3");
+  const at::cuda::CUDAGuard guard(X.device());
+  int Fp = (int)X.size(1);
+  TORCH_CHECK(Fp % 128 == 0, "Fp must be a multiple of 128");
+  long Bt = loss_id.size(0);
+  TORCH_CHECK(Bt % 256 == 0, "total columns must be 256-padded");
+  int n_chunks = (int)(Bt / 256);
+  int n_folds = (int)train_idx.size();
+  constexpr int Bp = 256;
+
+  auto optF = X.options().dtype(torch::kFloat32);
+  auto optI = X.options().dtype(torch::kInt32);
+  long max_NTp = 0;
+  for (auto& t : train_idx) max_NTp = std::max(max_NTp, t.size(0));
+  auto W = torch::empty({Bp, Fp}, optF);
+  auto V = torch::empty({Bp, Fp}, optF);
+  auto Gt = torch::zeros({Bp, Fp}, optF);
+  auto Wt = torch::empty({Bp, Fp}, X.options());
+  auto Dt = torch::empty({(long)Bp, max_NTp}, X.options());
+  auto bias = torch::empty({Bp}, optF);
+  auto Vb = torch::empty({Bp}, optF);
+  auto loss_sum = torch::zeros({Bp}, optF);
+  auto db_sum = torch::zeros({Bp}, optF);
+  auto prev_loss = torch::empty({Bp}, optF);
+  auto cur_loss = torch::zeros({Bp}, optF);
+  auto active = torch::empty({Bp}, optI);
+  auto epoch_mask = torch::zeros({Bp}, optI);
+  auto n_iter = torch::zeros({(long)n_folds, Bt}, optI);
+  auto n_active = torch::zeros({1}, optI);
+  auto test_sums = torch::zeros({(long)n_folds, n_cand_total},
+                                X.options().dtype(torch::kFloat64));
+  auto train_sums = torch::zeros(
+      {want_train_scores ? (long)n_folds : 0, n_cand_total},
+      X.options().dtype(torch::kFloat64));
+
+  const __bf16* Xp = reinterpret_cast<const __bf16*>(X.data_ptr());
+  const __bf16* Xtp = reinterpret_cast<const __bf16*>(Xt.data_ptr());
+  long n_rows = X.size(0);
+  hipStream_t s = cur_stream();
+
+  // host copies of per-column max_iter/tol for chunk planning (one sync)
+  auto mi_cpu = max_iter.cpu();
+  auto tol_cpu = tol.cpu();
+  const int* mi_h = mi_cpu.data_ptr<int>();
+  const float* tol_h = tol_cpu.data_ptr<float>();
+
+  std::vector<hipEvent_t> evs(4 * n_folds * n_chunks);
+  for (auto& e : evs) HIP_CHECK(hipEventCreate(&e));
+  int cands_per_chunk = 256 / (int)C;
+
+  for (int f = 0; f < n_folds; ++f) {
+    long NTp = train_idx[f].size(0);
+    long nt = nt_list[f];
+    float inv_n = 1.0f / (float)nt;
+    dim3 fin_grid(1);
+    int upd_threads = Fp >= 256 ? 256 : Fp;
+    dim3 upd_grid2((unsigned)(Fp / upd_threads), (unsigned)Bp);
+    int ftiles = Fp / 64;
+    long rtiles = NTp / 64;
+    long want_nseg = std::max<long>(1, 1024 / ftiles);
+    long seg_tiles = std::max<long>(1, (rtiles + want_nseg - 1) / want_nseg);
+    long nseg = (rtiles + seg_tiles - 1) / seg_tiles;
+    int seg_rows = (int)(seg_tiles * 64);
+    const int* idxp = train_idx[f].data_ptr<int>();
+
+    for (int c = 0; c < n_chunks; ++c) {
+      long off = (long)c * 256;
+      int max_ep = 0;
+      bool any_tol = false;
+      for (int j = 0; j < 256; ++j) {
+        max_ep = std::max(max_ep, mi_h[off + j]);
+        any_tol = any_tol || tol_h[off + j] > 0.f;
+      }
+      hipEvent_t ev_f0 = evs[(f * n_chunks + c) * 4 + 0];
+      hipEvent_t ev_f1 = evs[(f * n_chunks + c) * 4 + 1];
+      hipEvent_t ev_s0 = evs[(f * n_chunks + c) * 4 + 2];
+      hipEvent_t ev_s1 = evs[(f * n_chunks + c) * 4 + 3];
+      HIP_CHECK(hipEventRecord(ev_f0, s));
+      // reset per-chunk fit state
+      HIP_CHECK(hipMemsetAsync(W.data_ptr(), 0, (size_t)Bp * Fp * 4, s));
+      HIP_CHECK(hipMemsetAsync(V.data_ptr(), 0, (size_t)Bp * Fp * 4, s));
+      HIP_CHECK(hipMemsetAsync(Wt.data_ptr(), 0, (size_t)Bp * Fp * 2, s));
+      HIP_CHECK(hipMemsetAsync(bias.data_ptr(), 0, Bp * 4, s));
+      HIP_CHECK(hipMemsetAsync(Vb.data_ptr(), 0, Bp * 4, s));
+      prev_loss.fill_(3.0e38f);
+      active.fill_(1);
+
+      const int* loss_p = loss_id.data_ptr<int>() + off;
+      const float* lr_p = lr.data_ptr<float>() + off;
+      const float* l1_p = l1.data_ptr<float>() + off;
+      const float* l2_p = l2.data_ptr<float>() + off;
+      const float* mom_p = mom.data_ptr<float>() + off;
+      const float* tol_p = tol.data_ptr<float>() + off;
+      const int* mi_p = max_iter.data_ptr<int>() + off;
+      const int* tgt_p = target.data_ptr<int>() + off;
+      const unsigned char* ic_p =
+          fit_ic.data_ptr<unsigned char>() + off;
+      int* ni_p = n_iter.data_ptr<int>() + (long)f * Bt + off;
+
+      for (int epoch = 0; epoch < max_ep; ++epoch) {
+        int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+                           dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
+                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                           bias.data_ptr<float>(), loss_p, tgt_p,
+                           reinterpret_cast<__bf16*>(Dt.data_ptr()),
+                           loss_sum.data_ptr<float>(),
+                           db_sum.data_ptr<float>(), (int)nt, (int)NTp, Fp,
+                           Bp, want_loss);
+        if (Fp % 256 == 0) {
+          hipLaunchKernelGGL(bwd_g8_kernel,
+                             dim3((unsigned)(Fp / 256), (unsigned)nseg),
+                             dim3(1024), 0, s, Xtp, idxp,
+                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
+                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                             Bp, seg_rows);
+        } else {
+          hipLaunchKernelGGL(bwd_g5_kernel,
+                             dim3((unsigned)(Fp / 128), (unsigned)nseg),
+                             dim3(512), 0, s, Xtp, idxp,
+                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
+                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                             Bp, seg_rows);
+        }
+        hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
+                           loss_sum.data_ptr<float>(),
+                           db_sum.data_ptr<float>(), bias.data_ptr<float>(),
+                           Vb.data_ptr<float>(), prev_loss.data_ptr<float>(),
+                           cur_loss.data_ptr<float>(), active.data_ptr<int>(),
+                           epoch_mask.data_ptr<int>(), ni_p,
+                           mi_p, lr_p, mom_p, tol_p, ic_p, inv_n, epoch, Bp,
+                           want_loss, n_active.data_ptr<int>());
+        hipLaunchKernelGGL(update_w_kernel, upd_grid2, dim3(upd_threads), 0,
+                           s, W.data_ptr<float>(), V.data_ptr<float>(),
+                           Gt.data_ptr<float>(),
+                           reinterpret_cast<__bf16*>(Wt.data_ptr()),
+                           epoch_mask.data_ptr<int>(), lr_p, mom_p, l1_p,
+                           l2_p, inv_n, Fp);
+        HIP_CHECK(hipGetLastError());
+        if (check_every > 0 && (epoch + 1) % check_every == 0 &&
+            epoch + 1 < max_ep) {
+          int na = n_active.item<int>();
+          if (na == 0) break;
+          n_active.zero_();
+        } else {
+          HIP_CHECK(hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s));
+        }
+      }
+      HIP_CHECK(hipEventRecord(ev_f1, s));
+      HIP_CHECK(hipEventRecord(ev_s0, s));
+      // scores (permuted kernel): test fold, then train if requested
+      int n_cand_chunk = std::min<long>(cands_per_chunk,
+                                        n_cand_total - (long)c * cands_per_chunk);
+      if (n_cand_chunk > 0) {
+        long n_t = test_idx[f].size(0);
+        size_t scand_pad = ((size_t)n_cand_chunk + 3) & ~(size_t)3;
+        size_t smem3 = 32 * 8 + 32 * (Bp + 8) * 4 + scand_pad * 4 +
+                       32 * 64 * 2 + (size_t)Bp * 64 * 2;
+        HIP_CHECK(hipFuncSetAttribute(
+            reinterpret_cast<const void*>(scores3_kernel<4>),
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem3));
+        hipLaunchKernelGGL(scores3_kernel<4>,
+                           dim3((unsigned)((n_t + 31) / 32)), dim3(256),
+                           smem3, s,
+                           reinterpret_cast<const __bf16*>(X.data_ptr()),
+                           y.data_ptr<float>(),
+                           test_idx[f].data_ptr<long>(),
+                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                           bias.data_ptr<float>(), loss_p, tgt_p, (int)C,
+                           (int)metric, n_t, Fp, Bp, n_cand_chunk,
+                           test_sums.data_ptr<double>() +
+                               (long)f * n_cand_total +
+                               (long)c * cands_per_chunk);
+        if (want_train_scores) {
+          // score the train fold with the padded i32 indices reused as i64
+          // is not possible; train scoring reuses the i64 view built by
+          // the caller in test-idx style (train_idx64 appended after the
+          // n_folds test entries)
+          long n_tr = test_idx[n_folds + f].size(0);
+          hipLaunchKernelGGL(scores3_kernel<4>,
+                             dim3((unsigned)((n_tr + 31) / 32)), dim3(256),
+                             smem3, s,
+                             reinterpret_cast<const __bf16*>(X.data_ptr()),
+                             y.data_ptr<float>(),
+                             test_idx[n_folds + f].data_ptr<long>(),
+                             reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                             bias.data_ptr<float>(), loss_p, tgt_p, (int)C,
+                             (int)metric, n_tr, Fp, Bp, n_cand_chunk,
+                             train_sums.data_ptr<double>() +
+                                 (long)f * n_cand_total +
+                                 (long)c * cands_per_chunk);
+        }
+      }
+      HIP_CHECK(hipEventRecord(ev_s1, s));
+      HIP_CHECK(hipGetLastError());
+    }
+  }
+  HIP_CHECK(hipStreamSynchronize(s));
+  auto fit_ms = torch::empty({n_folds, n_chunks},
+                             torch::dtype(torch::kFloat32));
+  auto score_ms = torch::empty({n_folds, n_chunks},
+                               torch::dtype(torch::kFloat32));
+  for (int f = 0; f < n_folds; ++f)
+    for (int c = 0; c < n_chunks; ++c) {
+      float ms;
+      HIP_CHECK(hipEventElapsedTime(&ms, evs[(f * n_chunks + c) * 4 + 0],
+                                    evs[(f * n_chunks + c) * 4 + 1]));
+      fit_ms[f][c] = ms;
+      HIP_CHECK(hipEventElapsedTime(&ms, evs[(f * n_chunks + c) * 4 + 2],
+                                    evs[(f * n_chunks + c) * 4 + 3]));
+      score_ms[f][c] = ms;
+    }
+  for (auto& e : evs) HIP_CHECK(hipEventDestroy(e));
+  return {test_sums, train_sums, fit_ms, score_ms, n_iter};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
@@ -2679,6 +2923,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("msq_rows", &msq_rows, "mean squared row norm");
   m.def("linear_fit", &linear_fit, "batched linear fit (device epoch loop)");
   m.def("linear_scores", &linear_scores, "batched per-candidate scores");
+  m.def("linear_search_run", &linear_search_run,
+        "whole fold x chunk search loop on device (C++ engine core)");
   m.def("kmeans_fit", &kmeans_fit, "batched Lloyd loop (centroid blocks)");
   m.def("kmeans_assign", &kmeans_assign, "assignment/inertia under centers");
   m.def("row_norms", &row_norms, "per-row squared norms");
