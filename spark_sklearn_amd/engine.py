@@ -372,7 +372,7 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
         loss_t, lr_t, l1_t, l2_t, mom_t, tol_t, mi_t, tgt_t, ic_t,
         C, _METRIC_ID[metric], n_cand, 1 if return_train_score else 0,
         32)
-    test_sums, train_sums, fit_ms, score_ms, _ = out
+    test_sums, train_sums, fit_ms, score_ms = out[0], out[1], out[2], out[3]
     test_sums = test_sums.cpu().numpy()
     train_sums = train_sums.cpu().numpy() if return_train_score else None
     fit_ms = fit_ms.numpy()

@@ -690,159 +690,6 @@ void bwd_g3_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 
 
 // ---------------------------------------------------------------------------
-// fwd_d5_kernel: Bp=256 wide-block variant -- 512 threads / 8 waves,
-// block tile 128 rows x 256 cols (waves 0-3: rows 0-63, waves 4-7:
-// rows 64-127; each wave 64x64 output).  Same permuted LDS + 2-phase
-// pipeline as fwd_d3 but W staging is amortized over twice the rows.
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(512)
-void fwd_d5_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
-                   const int* __restrict__ idx, const __bf16* __restrict__ Wt,
-                   const float* __restrict__ bias,
-                   const int* __restrict__ loss_id,
-                   const int* __restrict__ target, __bf16* __restrict__ Dt,
-                   float* __restrict__ loss_sum, float* __restrict__ db_sum,
-                   int nt, int NTp, int Fp, int Bp, int want_loss) {
-  constexpr int NF = 4;
-  __shared__ __bf16 sX[2][128 * 64];
-  __shared__ __bf16 sW[2][256 * 64];
-  __shared__ float sLoss[256];
-  __shared__ float sDb[256];
-  __shared__ int sIdx[128];
-  __shared__ float sY[128];
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int cwave = wave & 3;        // column group
-  const int rhalf = wave >> 2;       // row half (0/1)
-  const int rt = blockIdx.x;
-  const int g = lane >> 4;
-  const int srow = tid >> 2;         // X staging row 0..127
-  const int sseg = tid & 3;
-  const int wrow = tid >> 3;         // W staging row 0..63 (x4 = 256)
-  const int wseg2 = (tid >> 2) & 1;  // which 32-elem half
-
-  if (tid < 128) {
-    int gi = idx[rt * 128 + tid];
-    sIdx[tid] = gi;
-    sY[tid] = y[gi];
-  }
-  if (tid < 256) { sLoss[tid] = 0.f; sDb[tid] = 0.f; }
-  __syncthreads();
-
-  // X: 128x64 elems / 512 thr = 16 each; W: 256x64 / 512 = 32 each
-  bf16x8 rx[2], rw[4];
-  auto issue = [&](int kc) {
-    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
-        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
-    rx[0] = gx[0];
-    rx[1] = gx[1];
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int c = wrow + i * 128;
-      const bf16x8* gw = reinterpret_cast<const bf16x8*>(
-          Wt + (long)c * Fp + kc + wseg2 * 32);
-      rw[2 * i] = gw[0];
-      rw[2 * i + 1] = gw[1];
-    }
-  };
-  auto commit = [&](int b) {
-    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int c = wrow + i * 128;
-      store_row8(&sW[b][c * 64], c, wseg2 * 32, rw[2 * i]);
-      store_row8(&sW[b][c * 64], c, wseg2 * 32 + 8, rw[2 * i + 1]);
-    }
-  };
-
-  f32x4 acc[4][NF];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  issue(0);
-  commit(0);
-  __syncthreads();
-  const int nkc = Fp >> 6;
-  for (int kc = 0; kc < nkc; ++kc) {
-    int b = kc & 1;
-    if (kc + 1 < nkc) issue((kc + 1) << 6);
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      bf16x8 a[4];
-#pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        int r = rhalf * 64 + m * 16 + (lane & 15);
-        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
-      }
-#pragma unroll
-      for (int fn = 0; fn < NF; ++fn) {
-        int c = cwave * 64 + fn * 16 + (lane & 15);
-        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
-#pragma unroll
-        for (int m = 0; m < 4; ++m)
-          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[m], bb, acc[m][fn], 0, 0, 0);
-      }
-    }
-    if (kc + 1 < nkc) commit(b ^ 1);
-    __syncthreads();
-  }
-
-#pragma unroll
-  for (int fn = 0; fn < NF; ++fn) {
-    int col_in = cwave * 64 + fn * 16 + (lane & 15);
-    float bia = bias[col_in];
-    int lid = loss_id[col_in];
-    int tgt = target[col_in];
-    float lsum = 0.f, dsum = 0.f;
-#pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      int row_base = rhalf * 64 + m * 16 + ((lane >> 4) << 2);
-      bf16x4 dpack;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int gr = rt * 128 + row_base + r;
-        float z = acc[m][fn][r] + bia;
-        float d, ls = 0.f;
-        if (want_loss) {
-          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
-        } else {
-          float yv = sY[row_base + r];
-          if (lid == LOSS_LOGISTIC) {
-            float t = (yv == (float)tgt) ? 1.f : 0.f;
-            d = 1.f / (1.f + __expf(-z)) - t;
-          } else if (lid == LOSS_HINGE) {
-            float sg = (yv == (float)tgt) ? 1.f : -1.f;
-            d = (sg * z < 1.f) ? -sg : 0.f;
-          } else {
-            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
-            d = z - t;
-          }
-        }
-        if (gr >= nt) { d = 0.f; ls = 0.f; }
-        dpack[r] = (__bf16)d;
-        lsum += ls;
-        dsum += d;
-      }
-      *reinterpret_cast<bf16x4*>(
-          Dt + (long)col_in * NTp + rt * 128 + row_base) = dpack;
-    }
-    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
-    atomicAdd(&sDb[col_in], dsum);
-  }
-  __syncthreads();
-  if (tid < 256) {
-    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
-    atomicAdd(&db_sum[tid], sDb[tid]);
-  }
-}
-
-// ---------------------------------------------------------------------------
 // bwd_g5_kernel: Bp=256 wide-block variant -- 512 threads / 8 waves,
 // output tile 256 cands x 128 features (waves 0-3: f 0-63, waves 4-7:
 // f 64-127).  grid (Fp/128, nseg).
@@ -864,8 +711,6 @@ void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   const int r0 = blockIdx.y * seg_rows;
   const int r1 = min(r0 + seg_rows, NTp);
   const int g = lane >> 4;
-  const int drow = tid >> 3;          // D staging: 64 base rows x4
-  const int dseg2 = (tid >> 2) & 1;
   const int sseg = tid & 3;
   const int xf = tid >> 2;            // Xt staging: feature 0..127
 
@@ -877,11 +722,16 @@ void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 
   bf16x8 rd[4], rxa, rxb;
   auto issue = [&](int rc) {
+    // D tile [256 c][64 r]: 1024 16-elem loads over 512 threads --
+    // load l = tid + i*512 covers (c = l>>2, seg = l&3) COMPLETELY.
+    // (The original drow/dseg2 mapping staged half the rows twice and
+    // left the other half as LDS garbage.)
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      int c = drow + i * 128;
+      int l = tid + i * 512;
+      int c = l >> 2, seg = l & 3;
       const bf16x8* gd = reinterpret_cast<const bf16x8*>(
-          Dt + (long)c * NTp + rc + dseg2 * 32);
+          Dt + (long)c * NTp + rc + seg * 16);
       rd[2 * i] = gd[0];
       rd[2 * i + 1] = gd[1];
     }
@@ -903,9 +753,10 @@ void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   auto commit = [&](int b) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      int c = drow + i * 128;
-      store_row8(&sD[b][c * 64], c, dseg2 * 32, rd[2 * i]);
-      store_row8(&sD[b][c * 64], c, dseg2 * 32 + 8, rd[2 * i + 1]);
+      int l = tid + i * 512;
+      int c = l >> 2, seg = l & 3;
+      store_row8(&sD[b][c * 64], c, seg * 16, rd[2 * i]);
+      store_row8(&sD[b][c * 64], c, seg * 16 + 8, rd[2 * i + 1]);
     }
     store_row8(&sXt[b][xf * 64], xf, sseg * 16, rxa);
     store_row8(&sXt[b][xf * 64], xf, sseg * 16 + 8, rxb);
@@ -2913,7 +2764,8 @@ std::vector<torch::Tensor> linear_search_run(
       score_ms[f][c] = ms;
     }
   for (auto& e : evs) HIP_CHECK(hipEventDestroy(e));
-  return {test_sums, train_sums, fit_ms, score_ms, n_iter};
+  // W/bias of the LAST chunk fitted -- debugging/refit aid
+  return {test_sums, train_sums, fit_ms, score_ms, n_iter, W, bias};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -237,3 +237,77 @@ def test_extension_mandatory_on_gpu(dev):
 
     assert extension_available(), \
         "native extension missing on a GPU box -- hard failure by policy"
+
+
+@pytest.mark.parametrize("f", [96, 256])
+def test_wide_batch_bp256_matches_ref(dev, f):
+    """Bp=256 kernel paths (fwd_d7 + bwd_g5/g8) vs the numpy reference.
+
+    Regression guard: the wide-batch kernels the flagship bench runs were
+    originally only numerics-tested at Bp=64 (different kernels); a
+    staging-coverage bug in bwd_g5 shipped undetected until a full-width
+    test existed."""
+    rng = np.random.default_rng(3)
+    n = 4096
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    w = rng.standard_normal(f).astype(np.float32) / np.sqrt(f)
+    y = ((X @ w) > 0).astype(np.int64)
+    Xb = _bf16_round(X)
+    msq = float(np.mean(np.sum(Xb.astype(np.float64) ** 2, axis=1)))
+    lr = auto_lr(msq, LOSS_LOGISTIC, 0.0)
+    B = 256
+    plan = _mkplan(B, LOSS_LOGISTIC, lr,
+                   l2=np.logspace(-6, -2, B).astype(np.float32),
+                   max_iter=6)
+    ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
+    train = np.arange(0, 3500, dtype=np.int64)
+    (W_d, b_d), ni_d, _ = ex.fit(train, plan)
+    W_r, b_r, ni_r, _ = linear_fit_ref(Xb[train], y[train], plan)
+    assert np.isfinite(W_d).all()
+    np.testing.assert_allclose(W_d, W_r, rtol=0.05, atol=2e-4)
+    np.testing.assert_allclose(b_d, b_r, rtol=0.05, atol=2e-4)
+    assert list(ni_d) == list(ni_r)
+    s_d = ex.scores(train, (W_d, b_d), plan, "accuracy")
+    s_r = linear_scores_ref(Xb[train], y[train], W_d, b_d, plan, "accuracy")
+    np.testing.assert_allclose(s_d, s_r, atol=5e-3)
+
+
+def test_native_search_run_matches_per_chunk(dev):
+    """The C++ whole-search runner must equal the per-chunk path."""
+    import spark_sklearn_amd.engine as eng
+    from spark_sklearn_amd.base_search import Task
+    from spark_sklearn_amd.models import LogisticRegression
+    from sklearn.model_selection import KFold
+
+    rng = np.random.default_rng(11)
+    n, f = 8192, 96
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    w = rng.standard_normal(f).astype(np.float32) / np.sqrt(f)
+    y = ((X @ w + 0.2 * rng.standard_normal(n)) > 0).astype(np.int64)
+    est = LogisticRegression(max_iter=15, tol=0.0)
+    splits = list(KFold(3).split(X))
+    tasks = []
+    i = 0
+    for ci, params in enumerate([{"C": c} for c in (0.01, 0.1, 1.0, 10.0)]):
+        for fi, (tr, te) in enumerate(splits):
+            tasks.append(Task(i, ci, fi, params, np.asarray(tr),
+                              np.asarray(te)))
+            i += 1
+
+    def run(disable_fast):
+        orig = eng._linear_device_search
+        if disable_fast:
+            eng._linear_device_search = lambda *a, **k: None
+        try:
+            res = eng.run_batched_tasks(
+                est, X, y, tasks, scorer=None, fit_params=None,
+                return_train_score=True, error_score="raise", verbose=0)
+        finally:
+            eng._linear_device_search = orig
+        return sorted(res, key=lambda r: r.index)
+
+    slow, fast = run(True), run(False)
+    for s_, f_ in zip(slow, fast):
+        assert abs(s_.test_score - f_.test_score) < 2e-3, \
+            (s_.index, s_.test_score, f_.test_score)
+        assert abs(s_.train_score - f_.train_score) < 2e-3
