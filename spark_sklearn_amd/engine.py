@@ -378,20 +378,27 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
     fit_ms = fit_ms.numpy()
     score_ms = score_ms.numpy()
 
-    def finalize(sums, idx_len, yv):
+    def finalize(sums, idx_len, idx):
         if metric in ("accuracy", "neg_log_loss"):
             return sums / idx_len
-        ss_tot = float(np.sum((yv - yv.mean()) ** 2))
+        y_sel = _y_np()[np.asarray(idx)]
+        ss_tot = float(np.sum((y_sel - y_sel.mean()) ** 2))
         return 1.0 - sums / max(ss_tot, 1e-12)
 
-    y_np = executor.y.cpu().numpy()
+    _y_cache = []
+
+    def _y_np():
+        if not _y_cache:
+            _y_cache.append(executor.y.cpu().numpy())
+        return _y_cache[0]
+
     results = []
     for fi, f in enumerate(folds):
         te = by_fold[f][0].test
         tr = by_fold[f][0].train
-        ts = finalize(test_sums[fi], len(te), y_np[np.asarray(te)])
+        ts = finalize(test_sums[fi], len(te), te)
         if return_train_score:
-            trs = finalize(train_sums[fi], len(tr), y_np[np.asarray(tr)])
+            trs = finalize(train_sums[fi], len(tr), tr)
         for slot, task in enumerate(by_fold[f]):
             chunk = slot // cand_chunk
             per_fit = float(fit_ms[fi][chunk]) / 1000.0 / cand_chunk
