@@ -1145,6 +1145,7 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
                           const int* __restrict__ cstart,   // [nb+1]
                           unsigned char* __restrict__ labels,  // [nb][NTp]
                           float* __restrict__ inertia,      // [nb]
+                          float* __restrict__ counts,       // [Kp] or null
                           int nt, int NTp, int Fp, int Kp, int nb,
                           int want_labels) {
   constexpr int cpw = NF * 16;
@@ -1152,7 +1153,8 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
   extern __shared__ char km_smem[];
   float* zbuf = reinterpret_cast<float*>(km_smem);          // [32][Kpad+8]
   float* sIn = zbuf + 32 * (Kpad + 8);                      // [nb<=256]
-  int* sIdx = reinterpret_cast<int*>(sIn + 256);            // [32]
+  float* sCnt = sIn + 256;                                  // [Kpad]
+  int* sIdx = reinterpret_cast<int*>(sCnt + Kpad);          // [32]
   __bf16* sX = reinterpret_cast<__bf16*>(sIdx + 32);        // [32][64]
   __bf16* sC = sX + 32 * 64;                                // [Kpad][64]
 
@@ -1164,6 +1166,7 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
 
   if (tid < 32) sIdx[tid] = idx[rt * 32 + tid];
   for (int i = tid; i < nb; i += 256) sIn[i] = 0.f;
+  for (int i = tid; i < Kpad; i += 256) sCnt[i] = 0.f;
   __syncthreads();
 
   f32x4 acc[2][NF];
@@ -1233,65 +1236,30 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
       if (q[j] < best) { best = q[j]; bj = j - lo; }
     if (want_labels)
       labels[(long)b * NTp + gr] = (unsigned char)bj;
+    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], 1.f);
     atomicAdd(&sIn[b], xnorm[sIdx[row]] + best);
   }
   __syncthreads();
   for (int b = tid; b < nb; b += 256) atomicAdd(&inertia[b], sIn[b]);
+  if (counts != nullptr)
+    for (int i = tid; i < Kp; i += 256)
+      if (sCnt[i] != 0.f) atomicAdd(&counts[i], sCnt[i]);
 }
 
-// update: accumulate per-centroid sums/counts in an LDS tile, flush with
-// global atomics.  grid (row_chunks, Fp/64); counts only on f-chunk 0.
-__global__ __launch_bounds__(256)
-void kmeans_update_kernel(const __bf16* __restrict__ X,
-                          const int* __restrict__ idx,
-                          const unsigned char* __restrict__ labels,
-                          const int* __restrict__ cstart,
-                          const int* __restrict__ block_active,
-                          float* __restrict__ sums,      // [Kp][Fp]
-                          float* __restrict__ counts,    // [Kp]
-                          int nt, int NTp, int Fp, int Kp, int nb,
-                          int rows_per_block) {
-  extern __shared__ char km2_smem[];
-  float* acc = reinterpret_cast<float*>(km2_smem);   // [Kp][68]
-  float* cnt = acc + Kp * 68;                        // [Kp]
-
-  const int tid = threadIdx.x;
-  const int ft = blockIdx.y;
-  const int r0 = blockIdx.x * rows_per_block;
-  const int r1 = min(r0 + rows_per_block, nt);
-
-  for (int i = tid; i < Kp * 68; i += 256) acc[i] = 0.f;
-  for (int i = tid; i < Kp; i += 256) cnt[i] = 0.f;
-  __syncthreads();
-
-  // 32 rows in flight: thread = (row 0..31) x (fseg 0..7)
-  const int rsub = tid >> 3;
-  const int fseg = tid & 7;
-  for (int rc = r0; rc < r1; rc += 32) {
-    int r = rc + rsub;
-    if (r < r1) {
-      long gi = idx[r];
-      bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          X + gi * Fp + ft * 64 + fseg * 8);
-      for (int b = 0; b < nb; ++b) {
-        if (!block_active[b]) continue;
-        int base = cstart[b] + labels[(long)b * NTp + r];
-        float* dst = &acc[base * 68 + fseg * 8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) atomicAdd(&dst[j], (float)v[j]);
-        if (ft == 0 && fseg == 0) atomicAdd(&cnt[base], 1.f);
-      }
+// one-hot membership matrix: D[cstart[b]+label_b(r)][r] = 1 for train
+// rows.  Feeds the MFMA bwd kernel: sums = X^T . onehot -- replaces the
+// LDS-atomic update kernel (same-label rows serialized its atomics).
+__global__ void kmeans_onehot_kernel(const unsigned char* __restrict__ labels,
+                                     const int* __restrict__ cstart,
+                                     __bf16* __restrict__ Dt,  // [Kp][NTp]
+                                     long nt, long NTp, int nb) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    for (int b = 0; b < nb; ++b) {
+      int c = cstart[b] + labels[(long)b * NTp + r];
+      Dt[(long)c * NTp + r] = (__bf16)1.0f;
     }
   }
-  __syncthreads();
-  for (int e = tid; e < Kp * 64; e += 256) {
-    int c = e >> 6, f = e & 63;
-    float v = acc[c * 68 + f];
-    if (v != 0.f) atomicAdd(&sums[(long)c * Fp + ft * 64 + f], v);
-  }
-  if (ft == 0)
-    for (int c = tid; c < Kp; c += 256)
-      if (cnt[c] != 0.f) atomicAdd(&counts[c], cnt[c]);
 }
 
 // center update: new = count>0 ? sum/count : old; accumulate shift^2 per
@@ -2223,18 +2191,19 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
                                  const __bf16* Ct, const float* cnorm,
                                  const float* xnorm, const int* cstart,
                                  unsigned char* labels, float* inertia,
-                                 int nt, int NTp, int Fp, int Kp, int nb,
-                                 int want_labels, hipStream_t s) {
-  size_t smem = (size_t)32 * (Kp + 8) * 4 + 256 * 4 + 32 * 4 +
-                (size_t)32 * 64 * 2 + (size_t)Kp * 64 * 2;
+                                 float* counts, int nt, int NTp, int Fp,
+                                 int Kp, int nb, int want_labels,
+                                 hipStream_t s) {
+  size_t smem = (size_t)32 * (Kp + 8) * 4 + 256 * 4 + (size_t)Kp * 4 +
+                32 * 4 + (size_t)32 * 64 * 2 + (size_t)Kp * 64 * 2;
   dim3 grid((unsigned)(NTp / 32));
   auto go = [&](auto k) {
     HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(k),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem));
     hipLaunchKernelGGL(k, grid, dim3(256), smem, s, X, idx, Ct, cnorm,
-                       xnorm, cstart, labels, inertia, nt, NTp, Fp, Kp, nb,
-                       want_labels);
+                       xnorm, cstart, labels, inertia, counts, nt, NTp,
+                       Fp, Kp, nb, want_labels);
   };
   switch (Kp / 64) {
     case 1: go(kmeans_assign_kernel<1>); break;
@@ -2249,7 +2218,7 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
 // inertia_final [nb] f32, Ct bf16, cnorm f32) -- Ct/cnorm stay resident for
 // the scores pass.
 std::vector<torch::Tensor> kmeans_fit(
-    torch::Tensor X, torch::Tensor idx, torch::Tensor C0,
+    torch::Tensor X, torch::Tensor Xt, torch::Tensor idx, torch::Tensor C0,
     torch::Tensor cstart, torch::Tensor cen_block, torch::Tensor max_iter,
     torch::Tensor tol, torch::Tensor xnorm, int64_t nt, int64_t max_epochs,
     int64_t check_every) {
@@ -2282,6 +2251,9 @@ std::vector<torch::Tensor> kmeans_fit(
 
   hipStream_t s = cur_stream();
   const __bf16* Xp = reinterpret_cast<const __bf16*>(X.data_ptr());
+  const __bf16* Xtp = reinterpret_cast<const __bf16*>(Xt.data_ptr());
+  long n_rows = X.size(0);
+  auto Dhot = torch::zeros({Kp, NTp}, X.options());
 
   // init pass: counts==0 everywhere -> keeps C, fills Ct + cnorm
   hipLaunchKernelGGL(kmeans_center_kernel, dim3(Kp), dim3(256), 0, s,
@@ -2292,14 +2264,14 @@ std::vector<torch::Tensor> kmeans_fit(
                      active.data_ptr<int>(), shift2.data_ptr<float>(),
                      Fp, Kp);
 
+  // split-K geometry for the sums GEMM (bwd_g3 generic tiles)
   int ftiles = Fp / 64;
-  int gridx = std::max(1, 1024 / ftiles);
-  int rows_per_block = (int)(((nt + gridx - 1) / gridx + 31) / 32 * 32);
-  gridx = (int)((nt + rows_per_block - 1) / rows_per_block);
-  size_t usmem = (size_t)Kp * 68 * 4 + (size_t)Kp * 4;
-  HIP_CHECK(hipFuncSetAttribute(
-      reinterpret_cast<const void*>(kmeans_update_kernel),
-      hipFuncAttributeMaxDynamicSharedMemorySize, (int)usmem));
+  long rtiles = NTp / 64;
+  long want_nseg = std::max<long>(1, 1024 / ftiles);
+  long seg_tiles = std::max<long>(1, (rtiles + want_nseg - 1) / want_nseg);
+  long nseg = (rtiles + seg_tiles - 1) / seg_tiles;
+  int seg_rows = (int)(seg_tiles * 64);
+  int onehot_blocks = (int)std::min<long>((nt + 255) / 256, 2048);
 
   for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
     HIP_CHECK(hipMemsetAsync(inertia.data_ptr(), 0, nb * sizeof(float), s));
@@ -2308,14 +2280,48 @@ std::vector<torch::Tensor> kmeans_fit(
                          cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
                          cstart.data_ptr<int>(),
                          labels.data_ptr<unsigned char>(),
-                         inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp,
+                         inertia.data_ptr<float>(),
+                         counts.data_ptr<float>(), (int)nt, (int)NTp, Fp,
                          Kp, nb, 1, s);
-    hipLaunchKernelGGL(kmeans_update_kernel, dim3(gridx, ftiles), dim3(256),
-                       usmem, s, Xp, idx.data_ptr<int>(),
-                       labels.data_ptr<unsigned char>(),
-                       cstart.data_ptr<int>(), active.data_ptr<int>(),
-                       sums.data_ptr<float>(), counts.data_ptr<float>(),
-                       (int)nt, (int)NTp, Fp, Kp, nb, rows_per_block);
+    // sums = X^T . onehot(labels): the same MFMA bwd kernel as the
+    // linear family (one-hot rows are exact in bf16)
+    HIP_CHECK(hipMemsetAsync(Dhot.data_ptr(), 0,
+                             (size_t)Kp * NTp * 2, s));
+    hipLaunchKernelGGL(kmeans_onehot_kernel, dim3(onehot_blocks),
+                       dim3(256), 0, s, labels.data_ptr<unsigned char>(),
+                       cstart.data_ptr<int>(),
+                       reinterpret_cast<__bf16*>(Dhot.data_ptr()), nt,
+                       (long)NTp, nb);
+    switch (Kp / 64) {
+      case 1:
+        hipLaunchKernelGGL(bwd_g3_kernel<1>, dim3(ftiles, (unsigned)nseg),
+                           dim3(256), 0, s, Xtp, idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Dhot.data_ptr()),
+                           sums.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                           Kp, seg_rows);
+        break;
+      case 2:
+        hipLaunchKernelGGL(bwd_g3_kernel<2>, dim3(ftiles, (unsigned)nseg),
+                           dim3(256), 0, s, Xtp, idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Dhot.data_ptr()),
+                           sums.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                           Kp, seg_rows);
+        break;
+      case 3:
+        hipLaunchKernelGGL(bwd_g3_kernel<3>, dim3(ftiles, (unsigned)nseg),
+                           dim3(256), 0, s, Xtp, idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Dhot.data_ptr()),
+                           sums.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                           Kp, seg_rows);
+        break;
+      default:
+        hipLaunchKernelGGL(bwd_g3_kernel<4>, dim3(ftiles, (unsigned)nseg),
+                           dim3(256), 0, s, Xtp, idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Dhot.data_ptr()),
+                           sums.data_ptr<float>(), n_rows, (int)NTp, Fp,
+                           Kp, seg_rows);
+        break;
+    }
     hipLaunchKernelGGL(kmeans_center_kernel, dim3(Kp), dim3(256), 0, s,
                        C.data_ptr<float>(),
                        reinterpret_cast<__bf16*>(Ct.data_ptr()),
@@ -2346,8 +2352,8 @@ std::vector<torch::Tensor> kmeans_fit(
                        cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
                        cstart.data_ptr<int>(),
                        labels.data_ptr<unsigned char>(),
-                       inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp,
-                       Kp, nb, 1, s);
+                       inertia.data_ptr<float>(), nullptr, (int)nt,
+                       (int)NTp, Fp, Kp, nb, 1, s);
   return {C, n_iter, inertia, Ct, cnorm, labels};
 }
 
@@ -2371,7 +2377,7 @@ std::vector<torch::Tensor> kmeans_assign(
       reinterpret_cast<const __bf16*>(Ct.data_ptr()),
       cnorm.data_ptr<float>(), xnorm.data_ptr<float>(),
       cstart.data_ptr<int>(), labels.data_ptr<unsigned char>(),
-      inertia.data_ptr<float>(), (int)nt, (int)NTp, Fp, Kp, nb,
+      inertia.data_ptr<float>(), nullptr, (int)nt, (int)NTp, Fp, Kp, nb,
       (int)want_labels, cur_stream());
   return {inertia, labels};
 }

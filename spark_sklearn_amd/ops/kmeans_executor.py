@@ -84,8 +84,8 @@ class DeviceKMeansExecutor:
         max_epochs = int(plan.max_iter.max()) if nb else 0
         check_every = 8 if max_epochs > 12 else 0
         C, n_iter, inertia, Ct, cnorm, labels = self.ext.kmeans_fit(
-            self.X, idx_t, C0_t, cstart_t, cen_block_t, mi_t, tol_t,
-            self.xnorm, nt, max_epochs, check_every)
+            self.X, self.Xt, idx_t, C0_t, cstart_t, cen_block_t, mi_t,
+            tol_t, self.xnorm, nt, max_epochs, check_every)
         torch.cuda.synchronize(dev)
         self._score_state = (Ct, cnorm, cstart_t)
 
