@@ -79,6 +79,16 @@ class LinearFamily:
     @staticmethod
     def make_executor(X, y_enc, classes, device):
         if device is not None:
+            import scipy.sparse as sp
+
+            if sp.issparse(X) and (classes is None or len(classes) <= 2):
+                from .ops.sparse_executor import (CSR_DENSITY_LIMIT,
+                                                  SparseLinearExecutor)
+
+                density = X.nnz / max(1, X.shape[0] * X.shape[1])
+                if density <= CSR_DENSITY_LIMIT:
+                    # CSR-resident K5 kernels: never densified
+                    return SparseLinearExecutor(X, y_enc, classes, device)
             from .ops import DeviceLinearExecutor
 
             return DeviceLinearExecutor(X, y_enc, classes, device)
@@ -238,7 +248,10 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     for t in tasks:
         by_fold.setdefault(t.fold_index, []).append(t)
 
+    from .ops.device_executor import DeviceLinearExecutor as _DLE
+
     if family is LinearFamily and device is not None and \
+            isinstance(executor, _DLE) and \
             metric is not None and not fit_params:
         fast = _linear_device_search(
             estimator, executor, by_fold, n_classes, metric,

@@ -1862,6 +1862,234 @@ void scores3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     atomicAdd(&out_sum[cand], (double)scand[cand]);
 }
 
+
+// ===========================================================================
+// K5: CSR-resident linear fits -- for matrices too large or too sparse to
+// densify into HBM.  Layouts:
+//   X CSR:   indptr i32 [n+1], cols i32 [nnz], vals bf16 [nnz]
+//   bwd:     transpose entries sorted by (4096-row chunk, feature, row):
+//            consecutive entries share a feature (register-accumulated
+//            runs, ONE coalesced flush per run) and a row chunk (the dense
+//            D rows they touch stay L2-resident)
+//   Wf/Gf:   feature-major mirrors [Fp][Bp] -- a wave reads/flushes one
+//            contiguous 64*CPL-column slice per nnz
+//   D:       ROW-major bf16 [nt][Bp]; lane owns columns
+//            [lane*CPL, lane*CPL + CPL)
+// ===========================================================================
+
+template <int CPL>   // columns per lane; Bp = 64 * CPL
+__global__ __launch_bounds__(256)
+void csr_fwd_d_kernel(const int* __restrict__ indptr,
+                      const int* __restrict__ cols,
+                      const __bf16* __restrict__ vals,
+                      const float* __restrict__ y,
+                      const int* __restrict__ idx,     // [nt] train rows
+                      const __bf16* __restrict__ Wf,   // [Fp][Bp]
+                      const float* __restrict__ bias,
+                      const int* __restrict__ loss_id,
+                      const int* __restrict__ target,
+                      __bf16* __restrict__ D,          // [nt][Bp]
+                      float* __restrict__ loss_sum,
+                      float* __restrict__ db_sum,
+                      long nt, int Bp, int want_loss) {
+  const int lane = threadIdx.x & 63;
+  const long gwave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const long nwaves = ((long)gridDim.x * blockDim.x) >> 6;
+  const int c0 = lane * CPL;
+
+  float ls[CPL], db[CPL];
+#pragma unroll
+  for (int j = 0; j < CPL; ++j) { ls[j] = 0.f; db[j] = 0.f; }
+
+  for (long i = gwave; i < nt; i += nwaves) {
+    long r = idx[i];
+    float z[CPL];
+#pragma unroll
+    for (int j = 0; j < CPL; ++j) z[j] = 0.f;
+    int k1 = indptr[r + 1];
+    for (int k = indptr[r]; k < k1; ++k) {
+      float v = (float)vals[k];            // lane-uniform broadcast reads
+      const __bf16* wrow = Wf + (long)cols[k] * Bp + c0;
+#pragma unroll
+      for (int j = 0; j < CPL; ++j) z[j] += v * (float)wrow[j];
+    }
+    float yv = y[r];
+    __bf16* drow = D + i * Bp + c0;
+#pragma unroll
+    for (int j = 0; j < CPL; ++j) {
+      int c = c0 + j;
+      float zz = z[j] + bias[c];
+      float d, l = 0.f;
+      if (want_loss) {
+        loss_grad(loss_id[c], zz, yv, target[c], &d, &l);
+      } else {
+        int lid = loss_id[c];
+        int tgt = target[c];
+        if (lid == LOSS_LOGISTIC) {
+          float t = (yv == (float)tgt) ? 1.f : 0.f;
+          d = 1.f / (1.f + __expf(-zz)) - t;
+        } else if (lid == LOSS_HINGE) {
+          float sg = (yv == (float)tgt) ? 1.f : -1.f;
+          d = (sg * zz < 1.f) ? -sg : 0.f;
+        } else {
+          float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+          d = zz - t;
+        }
+      }
+      drow[j] = (__bf16)d;
+      ls[j] += l;
+      db[j] += d;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < CPL; ++j) {
+    if (want_loss) atomicAdd(&loss_sum[c0 + j], ls[j]);
+    atomicAdd(&db_sum[c0 + j], db[j]);
+  }
+}
+
+template <int CPL>
+__global__ __launch_bounds__(256)
+void csr_bwd_g_kernel(const int* __restrict__ tf,    // [nnz] feature
+                      const int* __restrict__ tr,    // [nnz] global row
+                      const __bf16* __restrict__ tv, // [nnz] value
+                      const long* __restrict__ chunk_off,  // [nchunks+1]
+                      const int* __restrict__ inv,   // [n] -> local / -1
+                      const __bf16* __restrict__ D,  // [nt][Bp]
+                      float* __restrict__ Gf,        // [Fp][Bp]
+                      int Bp) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int c0 = lane * CPL;
+  long lo = chunk_off[blockIdx.x], hi = chunk_off[blockIdx.x + 1];
+  long per = (hi - lo + 3) / 4;
+  long s0 = lo + wave * per;
+  long s1 = min(s0 + per, hi);
+
+  float acc[CPL];
+#pragma unroll
+  for (int j = 0; j < CPL; ++j) acc[j] = 0.f;
+  int curf = -1;
+  for (long k = s0; k < s1; ++k) {
+    int f = tf[k];
+    if (f != curf) {
+      if (curf >= 0) {
+#pragma unroll
+        for (int j = 0; j < CPL; ++j)
+          if (acc[j] != 0.f)
+            atomicAdd(&Gf[(long)curf * Bp + c0 + j], acc[j]);
+      }
+      curf = f;
+#pragma unroll
+      for (int j = 0; j < CPL; ++j) acc[j] = 0.f;
+    }
+    int dl = inv[tr[k]];
+    if (dl < 0) continue;
+    float v = (float)tv[k];
+    const __bf16* drow = D + (long)dl * Bp + c0;
+#pragma unroll
+    for (int j = 0; j < CPL; ++j) acc[j] += v * (float)drow[j];
+  }
+  if (curf >= 0) {
+#pragma unroll
+    for (int j = 0; j < CPL; ++j)
+      if (acc[j] != 0.f)
+        atomicAdd(&Gf[(long)curf * Bp + c0 + j], acc[j]);
+  }
+}
+
+// sparse-mode weight update: f-major gradient/mirror, c-major master
+__global__ void update_w_sparse_kernel(
+    float* __restrict__ W,    // [Bp][Fp]
+    float* __restrict__ V,    // [Bp][Fp]
+    float* __restrict__ Gf,   // [Fp][Bp]
+    __bf16* __restrict__ Wf,  // [Fp][Bp]
+    const int* __restrict__ epoch_mask, const float* __restrict__ lr,
+    const float* __restrict__ mom, const float* __restrict__ l1v,
+    const float* __restrict__ l2v, float inv_n, int Fp, int Bp) {
+  int c = blockIdx.y;
+  int f = blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= Fp) return;
+  long cf = (long)c * Fp + f;
+  long fc = (long)f * Bp + c;
+  float g = Gf[fc];
+  Gf[fc] = 0.f;
+  if (!epoch_mask[c]) return;
+  float w = W[cf];
+  float grad = g * inv_n + l2v[c] * w +
+               l1v[c] * ((w > 0.f) ? 1.f : ((w < 0.f) ? -1.f : 0.f));
+  float v = mom[c] * V[cf] - lr[c] * grad;
+  V[cf] = v;
+  w += v;
+  W[cf] = w;
+  Wf[fc] = (__bf16)w;
+}
+
+// scores over CSR rows: binary / regression metrics (C == 1 only; the
+// engine densifies multiclass sparse searches)
+template <int CPL>
+__global__ __launch_bounds__(256)
+void csr_scores_kernel(const int* __restrict__ indptr,
+                       const int* __restrict__ cols,
+                       const __bf16* __restrict__ vals,
+                       const float* __restrict__ y,
+                       const long* __restrict__ idx,  // [n_idx]
+                       const __bf16* __restrict__ Wf,
+                       const float* __restrict__ bias,
+                       const int* __restrict__ loss_id,
+                       const int* __restrict__ target,
+                       int metric, long n_idx, int Bp, int n_cand,
+                       double* __restrict__ out_sum) {
+  const int lane = threadIdx.x & 63;
+  const long gwave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const long nwaves = ((long)gridDim.x * blockDim.x) >> 6;
+  const int c0 = lane * CPL;
+  const float eps = 1e-12f;
+
+  float accm[CPL];
+#pragma unroll
+  for (int j = 0; j < CPL; ++j) accm[j] = 0.f;
+
+  for (long i = gwave; i < n_idx; i += nwaves) {
+    long r = idx[i];
+    float z[CPL];
+#pragma unroll
+    for (int j = 0; j < CPL; ++j) z[j] = 0.f;
+    int k1 = indptr[r + 1];
+    for (int k = indptr[r]; k < k1; ++k) {
+      float v = (float)vals[k];
+      const __bf16* wrow = Wf + (long)cols[k] * Bp + c0;
+#pragma unroll
+      for (int j = 0; j < CPL; ++j) z[j] += v * (float)wrow[j];
+    }
+    float yv = y[r];
+#pragma unroll
+    for (int j = 0; j < CPL; ++j) {
+      int c = c0 + j;
+      if (c >= n_cand) continue;
+      float zz = z[j] + bias[c];
+      if (metric == 2) {
+        float d = yv - zz;
+        accm[j] += d * d;
+      } else if (metric == 0) {
+        int pred = (zz > 0.f) ? 1 : 0;
+        accm[j] += (pred == (int)yv) ? 1.f : 0.f;
+      } else {
+        float pr = 1.f / (1.f + __expf(-zz));
+        float t = (yv == 1.f) ? 1.f : 0.f;
+        accm[j] += t * __logf(pr + eps) +
+                   (1.f - t) * __logf(1.f - pr + eps);
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < CPL; ++j) {
+    int c = c0 + j;
+    if (c < n_cand && accm[j] != 0.f)
+      atomicAdd(&out_sum[c], (double)accm[j]);
+  }
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -2774,6 +3002,140 @@ std::vector<torch::Tensor> linear_search_run(
   return {test_sums, train_sums, fit_ms, score_ms, n_iter, W, bias};
 }
 
+
+// ---------------------------------------------------------------------------
+// K5 host wrappers
+// ---------------------------------------------------------------------------
+
+std::vector<torch::Tensor> csr_linear_fit(
+    torch::Tensor indptr, torch::Tensor ccols, torch::Tensor vals,
+    torch::Tensor tf, torch::Tensor tr, torch::Tensor tv,
+    torch::Tensor chunk_off, torch::Tensor inv, torch::Tensor y,
+    torch::Tensor idx,   // i32 [nt], exact (no padding)
+    torch::Tensor loss_id, torch::Tensor lr, torch::Tensor l1,
+    torch::Tensor l2, torch::Tensor mom, torch::Tensor tol,
+    torch::Tensor max_iter, torch::Tensor target, torch::Tensor fit_ic,
+    int64_t Fp, int64_t max_epochs, int64_t check_every) {
+  const at::cuda::CUDAGuard guard(vals.device());
+  int Bp = (int)loss_id.size(0);
+  TORCH_CHECK(Bp == 64 || Bp == 128 || Bp == 256);
+  long nt = idx.size(0);
+  int nchunks = (int)chunk_off.size(0) - 1;
+
+  auto optF = y.options().dtype(torch::kFloat32);
+  auto optI = y.options().dtype(torch::kInt32);
+  auto W = torch::zeros({Bp, Fp}, optF);
+  auto V = torch::zeros({Bp, Fp}, optF);
+  auto Gf = torch::zeros({Fp, Bp}, optF);
+  auto Wf = torch::zeros({Fp, Bp}, vals.options());
+  auto D = torch::empty({nt, (long)Bp}, vals.options());
+  auto bias = torch::zeros({Bp}, optF);
+  auto Vb = torch::zeros({Bp}, optF);
+  auto loss_sum = torch::zeros({Bp}, optF);
+  auto db_sum = torch::zeros({Bp}, optF);
+  auto prev_loss = torch::full({Bp}, 3.0e38, optF);
+  auto cur_loss = torch::zeros({Bp}, optF);
+  auto active = torch::ones({Bp}, optI);
+  auto epoch_mask = torch::zeros({Bp}, optI);
+  auto n_iter = torch::zeros({Bp}, optI);
+  auto n_active = torch::zeros({1}, optI);
+
+  hipStream_t s = cur_stream();
+  float inv_n = 1.0f / (float)nt;
+  bool any_tol = tol.max().item<float>() > 0.f;
+  int fwd_blocks = (int)std::min<long>((nt + 3) / 4, 4096);
+  int upd_threads = Fp >= 256 ? 256 : (int)Fp;
+  dim3 upd_grid((unsigned)((Fp + upd_threads - 1) / upd_threads),
+                (unsigned)Bp);
+  int CPL = Bp / 64;
+
+  auto fwd = [&](auto k, int want_loss) {
+    hipLaunchKernelGGL(k, dim3(fwd_blocks), dim3(256), 0, s,
+                       indptr.data_ptr<int>(), ccols.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(vals.data_ptr()),
+                       y.data_ptr<float>(), idx.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(Wf.data_ptr()),
+                       bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+                       target.data_ptr<int>(),
+                       reinterpret_cast<__bf16*>(D.data_ptr()),
+                       loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
+                       nt, Bp, want_loss);
+  };
+  auto bwd = [&](auto k) {
+    hipLaunchKernelGGL(k, dim3(nchunks), dim3(256), 0, s,
+                       tf.data_ptr<int>(), tr.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(tv.data_ptr()),
+                       chunk_off.data_ptr<long>(), inv.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(D.data_ptr()),
+                       Gf.data_ptr<float>(), Bp);
+  };
+
+  for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
+    int want_loss = (any_tol || epoch == (int)max_epochs - 1) ? 1 : 0;
+    switch (CPL) {
+      case 1: fwd(csr_fwd_d_kernel<1>, want_loss); bwd(csr_bwd_g_kernel<1>); break;
+      case 2: fwd(csr_fwd_d_kernel<2>, want_loss); bwd(csr_bwd_g_kernel<2>); break;
+      default: fwd(csr_fwd_d_kernel<4>, want_loss); bwd(csr_bwd_g_kernel<4>); break;
+    }
+    hipLaunchKernelGGL(col_finalize_kernel, dim3((Bp + 255) / 256),
+                       dim3(256), 0, s, loss_sum.data_ptr<float>(),
+                       db_sum.data_ptr<float>(), bias.data_ptr<float>(),
+                       Vb.data_ptr<float>(), prev_loss.data_ptr<float>(),
+                       cur_loss.data_ptr<float>(), active.data_ptr<int>(),
+                       epoch_mask.data_ptr<int>(), n_iter.data_ptr<int>(),
+                       max_iter.data_ptr<int>(), lr.data_ptr<float>(),
+                       mom.data_ptr<float>(), tol.data_ptr<float>(),
+                       fit_ic.data_ptr<unsigned char>(), inv_n, epoch, Bp,
+                       want_loss, n_active.data_ptr<int>());
+    hipLaunchKernelGGL(update_w_sparse_kernel, upd_grid, dim3(upd_threads),
+                       0, s, W.data_ptr<float>(), V.data_ptr<float>(),
+                       Gf.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(Wf.data_ptr()),
+                       epoch_mask.data_ptr<int>(), lr.data_ptr<float>(),
+                       mom.data_ptr<float>(), l1.data_ptr<float>(),
+                       l2.data_ptr<float>(), inv_n, (int)Fp, Bp);
+    HIP_CHECK(hipGetLastError());
+    if (check_every > 0 && (epoch + 1) % check_every == 0 &&
+        epoch + 1 < max_epochs) {
+      int na = n_active.item<int>();
+      if (na == 0) break;
+      n_active.zero_();
+    } else {
+      HIP_CHECK(hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s));
+    }
+  }
+  return {W, bias, n_iter, cur_loss, Wf};
+}
+
+torch::Tensor csr_linear_scores(
+    torch::Tensor indptr, torch::Tensor ccols, torch::Tensor vals,
+    torch::Tensor y, torch::Tensor idx, torch::Tensor Wf,
+    torch::Tensor bias, torch::Tensor loss_id, torch::Tensor target,
+    int64_t metric, int64_t n_cand) {
+  const at::cuda::CUDAGuard guard(vals.device());
+  int Bp = (int)Wf.size(1);
+  long n_idx = idx.size(0);
+  auto out = torch::zeros({n_cand}, y.options().dtype(torch::kFloat64));
+  int blocks = (int)std::min<long>((n_idx + 3) / 4, 4096);
+  auto go = [&](auto k) {
+    hipLaunchKernelGGL(k, dim3(blocks), dim3(256), 0, cur_stream(),
+                       indptr.data_ptr<int>(), ccols.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(vals.data_ptr()),
+                       y.data_ptr<float>(), idx.data_ptr<long>(),
+                       reinterpret_cast<const __bf16*>(Wf.data_ptr()),
+                       bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+                       target.data_ptr<int>(), (int)metric, n_idx, Bp,
+                       (int)n_cand, out.data_ptr<double>());
+  };
+  switch (Bp / 64) {
+    case 1: go(csr_scores_kernel<1>); break;
+    case 2: go(csr_scores_kernel<2>); break;
+    default: go(csr_scores_kernel<4>); break;
+  }
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
@@ -2789,4 +3151,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gbt_fit", &gbt_fit, "histogram boosting fit (depth-wise trees)");
   m.def("gbt_predict", &gbt_predict, "tree-walk raw predictions");
   m.def("gbt_metric", &gbt_metric, "accuracy/ss_res sums");
+  m.def("csr_linear_fit", &csr_linear_fit,
+        "CSR-resident batched linear fit (K5)");
+  m.def("csr_linear_scores", &csr_linear_scores,
+        "CSR-resident per-candidate scores");
 }

@@ -1,0 +1,163 @@
+"""CSR-resident device execution for the linear family (K5 kernels).
+
+Used when the input is scipy-sparse and low-density: the matrix stays in
+HBM as CSR (indptr/indices/data) plus a (row-chunk, feature, row)-sorted
+transpose entry list for the backward pass -- never densified.  Binary /
+regression targets only (multiclass OVR sparse searches densify instead;
+the per-row argmax would span lanes).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import scipy.sparse as sp
+
+from . import load_extension
+from .device_executor import _METRIC_ID, _roundup, ensure_layout
+
+__all__ = ["SparseLinearExecutor", "CSR_DENSITY_LIMIT"]
+
+CSR_DENSITY_LIMIT = 0.05      # auto-pick the CSR path at or below this
+BWD_CHUNK_ROWS = 4096         # D rows touched by a chunk stay L2-resident
+
+_sparse_cache: dict = {}
+
+
+class SparseLinearExecutor:
+    """Executes batched linear fits/scores over a resident CSR matrix."""
+
+    def __init__(self, X, y_enc, classes, device):
+        import torch
+
+        self.torch = torch
+        self.device = device
+        self.ext = load_extension()
+        if ensure_layout(device) != 3:
+            raise RuntimeError("CSR kernels require the probed layout P=3")
+        if classes is not None and len(classes) > 2:
+            raise RuntimeError(
+                "sparse executor supports binary/regression targets")
+
+        key = (id(X), X.shape, str(device))
+        hit = _sparse_cache.get(key)
+        if hit is None or hit[0] is not X:
+            Xc = sp.csr_matrix(X, dtype=np.float32)
+            Xc.sort_indices()
+            n, F = Xc.shape
+            dev = device
+            indptr = torch.as_tensor(
+                Xc.indptr.astype(np.int32)).to(dev)
+            cols = torch.as_tensor(
+                Xc.indices.astype(np.int32)).to(dev)
+            vals = torch.as_tensor(Xc.data).to(dev).to(torch.bfloat16)
+            # bwd structure: entries sorted by (row chunk, feature, row)
+            coo = Xc.tocoo()
+            chunk = (coo.row // BWD_CHUNK_ROWS).astype(np.int64)
+            order = np.lexsort((coo.row, coo.col, chunk))
+            tf = torch.as_tensor(
+                coo.col[order].astype(np.int32)).to(dev)
+            tr = torch.as_tensor(
+                coo.row[order].astype(np.int32)).to(dev)
+            tv = torch.as_tensor(
+                coo.data[order]).to(dev).to(torch.bfloat16)
+            nchunks = int(chunk.max()) + 1 if len(chunk) else 1
+            co = np.searchsorted(chunk[order], np.arange(nchunks + 1))
+            chunk_off = torch.as_tensor(co.astype(np.int64)).to(dev)
+            sq = np.asarray(Xc.multiply(Xc).sum(axis=1)).ravel()
+            msq = float(np.mean(sq))
+            if len(_sparse_cache) > 2:
+                _sparse_cache.clear()
+            _sparse_cache[key] = (X, (indptr, cols, vals, tf, tr, tv,
+                                      chunk_off, n, F, msq))
+            hit = _sparse_cache[key]
+        (self.indptr, self.cols, self.vals, self.tf, self.tr, self.tv,
+         self.chunk_off, self.n, self.F, self.msq) = hit[1]
+        self.y = torch.as_tensor(
+            np.asarray(y_enc, dtype=np.float32)).to(device)
+        self._fit_state = None
+        self._inv_cache: dict = {}
+        self._idx_cache: dict = {}
+
+    def _cached(self, cache, arr, builder):
+        key = (id(arr), len(arr))
+        hit = cache.get(key)
+        if hit is not None and hit[0] is arr:
+            return hit[1]
+        t = builder(arr)
+        if len(cache) > 32:
+            cache.clear()
+        cache[key] = (arr, t)
+        return t
+
+    def fit(self, train_idx, plan):
+        torch = self.torch
+        nt = len(train_idx)
+        B = plan.B
+        Bp = max(64, _roundup(B, 64))
+        if Bp > 256:
+            raise RuntimeError("chunk larger batches to <= 256 columns")
+
+        def pad_i32(a, fill=0):
+            out = np.full(Bp, fill, dtype=np.int32)
+            out[:B] = np.asarray(a, dtype=np.int32)
+            return torch.as_tensor(out).to(self.device)
+
+        def pad_f32(a, fill=0.0):
+            out = np.full(Bp, fill, dtype=np.float32)
+            out[:B] = np.asarray(a, dtype=np.float32)
+            return torch.as_tensor(out).to(self.device)
+
+        idx_t = self._cached(
+            self._idx_cache, train_idx,
+            lambda a: torch.as_tensor(
+                np.asarray(a, dtype=np.int32)).to(self.device))
+
+        def build_inv(a):
+            inv = np.full(self.n, -1, dtype=np.int32)
+            inv[np.asarray(a, dtype=np.int64)] = np.arange(
+                len(a), dtype=np.int32)
+            return torch.as_tensor(inv).to(self.device)
+
+        inv_t = self._cached(self._inv_cache, train_idx, build_inv)
+
+        ic_np = np.zeros(Bp, dtype=np.uint8)
+        ic_np[:B] = np.asarray(plan.fit_intercept, dtype=np.uint8)
+        Fp = max(64, _roundup(self.F, 64))
+        max_epochs = int(plan.max_iter.max()) if B else 0
+        W, bias, n_iter, cur_loss, Wf = self.ext.csr_linear_fit(
+            self.indptr, self.cols, self.vals, self.tf, self.tr, self.tv,
+            self.chunk_off, inv_t, self.y, idx_t,
+            pad_i32(plan.loss), pad_f32(plan.lr), pad_f32(plan.l1),
+            pad_f32(plan.l2), pad_f32(plan.momentum), pad_f32(plan.tol),
+            pad_i32(plan.max_iter), pad_i32(plan.target_class),
+            torch.as_tensor(ic_np).to(self.device),
+            Fp, max_epochs, 32 if max_epochs > 48 else 0)
+        torch.cuda.synchronize(self.device)
+        self._fit_state = (Wf, bias, pad_i32(plan.loss),
+                           pad_i32(plan.target_class))
+        W_np = W[:B, :self.F].float().cpu().numpy().T.copy()
+        return ((W_np, bias[:B].float().cpu().numpy()),
+                n_iter[:B].cpu().numpy(),
+                cur_loss[:B].float().cpu().numpy())
+
+    def scores(self, idx, fitted, plan, metric):
+        torch = self.torch
+        if self._fit_state is None:
+            raise RuntimeError("scores() requires a preceding fit()")
+        Wf, bias, loss_t, tgt_t = self._fit_state
+        n_cand = int(plan.col_to_cand.max()) + 1
+        if plan.n_classes != 1:
+            raise RuntimeError("sparse scoring is binary/regression only")
+        idx_t = torch.as_tensor(
+            np.asarray(idx, dtype=np.int64)).to(self.device)
+        sums = self.ext.csr_linear_scores(
+            self.indptr, self.cols, self.vals, self.y, idx_t, Wf, bias,
+            loss_t, tgt_t, _METRIC_ID[metric], n_cand).cpu().numpy()
+        n = len(idx)
+        if metric in ("accuracy", "neg_log_loss"):
+            return sums / n
+        y_sel = self.y[idx_t].cpu().numpy()
+        ss_tot = float(np.sum((y_sel - y_sel.mean()) ** 2))
+        if ss_tot <= 0:
+            return np.zeros(n_cand)
+        return 1.0 - sums / ss_tot
