@@ -1360,7 +1360,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const float* __restrict__ h,
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
                      long nt, int F, int lvl_lo, int lvl_nodes, int f_pb,
-                     long rows_pb) {
+                     long rows_pb, int even_only) {
   extern __shared__ float lh[];     // [lvl_nodes][f_pb][256][3]
   const int tid = threadIdx.x;
   const int f0 = blockIdx.y * f_pb;
@@ -1374,6 +1374,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
   for (long r = r0 + tid; r < r1; r += 256) {
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
+    if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
     long gi = idxmap[r];
     float gr = g[r], hr = h[r];
     const unsigned char* xrow = Xb + gi * F + f0;
@@ -1394,6 +1395,24 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
     int ff = (c / 256) % f_pb;
     int ndl = c / (256 * f_pb);
     atomicAdd(&hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3 + ch], v);
+  }
+}
+
+// sibling subtraction: right-child histogram = parent - left.  Parents
+// that became leaves have no descending rows: both children zero.
+__global__ void gbt_sibling_kernel(const float* __restrict__ prev,
+                                   float* __restrict__ cur,
+                                   const int* __restrict__ feat,
+                                   int prev_lo, int F, int lvl_nodes) {
+  int pair = blockIdx.y;                   // local parent index
+  long cells = (long)F * 256 * 3;
+  long base_l = (long)(2 * pair) * cells;
+  long base_r = (long)(2 * pair + 1) * cells;
+  long base_p = (long)pair * cells;
+  bool split = feat[prev_lo + pair] >= 0;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < cells;
+       i += (long)gridDim.x * blockDim.x) {
+    cur[base_r + i] = split ? prev[base_p + i] - cur[base_l + i] : 0.f;
   }
 }
 
@@ -1490,19 +1509,33 @@ __global__ void gbt_split_apply_kernel(const unsigned char* __restrict__ Xb,
   }
 }
 
-// bottom-level leaf values: per-node g/h sums via atomics
+// bottom-level leaf values: per-node g/h sums.  LDS-staged per block --
+// direct global atomics serialized 1M adds onto <=64 node addresses
+// (4.8 ms/tree measured; this form is ~100x cheaper).
 __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
                                      const float* __restrict__ g,
                                      const float* __restrict__ h,
                                      float* __restrict__ gsum,
                                      float* __restrict__ hsum,
                                      long nt, int lvl_lo, int lvl_hi) {
+  __shared__ float lg[128], lh[128];
+  int nn = lvl_hi - lvl_lo;
+  for (int i = threadIdx.x; i < nn; i += blockDim.x) {
+    lg[i] = 0.f;
+    lh[i] = 0.f;
+  }
+  __syncthreads();
   long r = blockIdx.x * blockDim.x + threadIdx.x;
   for (; r < nt; r += (long)gridDim.x * blockDim.x) {
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_hi) continue;
-    atomicAdd(&gsum[nd], g[r]);
-    atomicAdd(&hsum[nd], h[r]);
+    atomicAdd(&lg[nd - lvl_lo], g[r]);
+    atomicAdd(&lh[nd - lvl_lo], h[r]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < nn; i += blockDim.x) {
+    if (lg[i] != 0.f) atomicAdd(&gsum[lvl_lo + i], lg[i]);
+    if (lh[i] != 0.f) atomicAdd(&hsum[lvl_lo + i], lh[i]);
   }
 }
 
@@ -2657,6 +2690,8 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
   auto vals = torch::zeros({(long)T, NN}, optF);
   int max_lvl_nodes = 1 << (max_depth - 1);
   auto hist = torch::empty({(long)max_lvl_nodes * F * 256 * 3}, optF);
+  auto hist_prev = torch::empty({(long)std::max(1, max_lvl_nodes / 2) *
+                                 F * 256 * 3}, optF);
   auto gsum = torch::zeros({NN}, optF);
   auto hsum = torch::zeros({NN}, optF);
 
@@ -2682,6 +2717,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
       long rows_pb = (nt + rowsegs - 1) / rowsegs;
       rowsegs = (nt + rows_pb - 1) / rows_pb;
       size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 3 * 4;
+      int even_only = (d > 0) ? 1 : 0;    // sibling subtraction
       HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0,
                                (size_t)lvl_nodes * F * 256 * 3 * 4, s));
       HIP_CHECK(hipFuncSetAttribute(
@@ -2692,11 +2728,25 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
                          dim3(256), lsm, s, Xp, idxmap.data_ptr<int>(),
                          node_id.data_ptr<int>(), g.data_ptr<float>(),
                          h.data_ptr<float>(), hist.data_ptr<float>(), nt,
-                         F, lvl_lo, lvl_nodes, f_pb, rows_pb);
+                         F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only);
+      if (even_only) {
+        int prev_lo = (1 << (d - 1)) - 1;
+        long cells = (long)F * 256 * 3;
+        unsigned gx = (unsigned)std::min<long>((cells + 255) / 256, 1024);
+        hipLaunchKernelGGL(gbt_sibling_kernel,
+                           dim3(gx, (unsigned)(lvl_nodes / 2)), dim3(256),
+                           0, s, hist_prev.data_ptr<float>(),
+                           hist.data_ptr<float>(), feat_t, prev_lo, F,
+                           lvl_nodes);
+      }
       hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
                          0, s, hist.data_ptr<float>(), F, lvl_lo,
                          (float)lam, (float)lrate, (int)min_leaf, 0,
                          feat_t, thr_t, val_t);
+      if (d + 1 < (int)max_depth)
+        hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
+            .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
+                   /*non_blocking=*/true);
       hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
                          0, s, Xp, idxmap.data_ptr<int>(),
                          node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
