@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--rows", type=int, default=1_000_000)
     p.add_argument("--features", type=int, default=256)
     p.add_argument("--cv", type=int, default=5)
-    p.add_argument("--cands-per-step", type=int, default=128)
+    p.add_argument("--cands-per-step", type=int, default=256)
     p.add_argument("--epochs", type=int, default=25)
     p.add_argument("--grid-size", type=int, default=10_000)
     p.add_argument("--cpu", action="store_true",

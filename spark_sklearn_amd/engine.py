@@ -328,6 +328,8 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
     C = LinearFamily.cols_per_cand(n_classes)
     if C > 256 or 256 % C != 0:
         return None
+    if executor.Fp % 128 != 0:
+        return None      # the native runner needs wide feature tiles
     folds = sorted(by_fold)
     first = by_fold[folds[0]]
     n_cand = len(first)
