@@ -127,7 +127,9 @@ class KMeansFamily:
             k = p.get("n_clusters", estimator.n_clusters)
             ni = p.get("n_init", estimator.n_init)
             worst = max(worst, int(k) * int(ni))
-        return max(1, column_batch // worst)
+        # cap total centroids per chunk at 128: keeps the wide assign
+        # kernel's zbuf within LDS (larger k falls back to 32-row tiles)
+        return max(1, min(column_batch, 128) // worst)
 
     @staticmethod
     def make_executor(X, y_enc, classes, device):
@@ -169,14 +171,26 @@ class KMeansFamily:
         block_to_cand, max_iters, tols = [], [], []
         C0_parts = []
         mean_var = executor.feature_mean_var
+        # centers are deterministic in (k, init, seed): cache them on the
+        # executor so the 5 CV folds (and repeated chunks) don't re-run
+        # host-side k-means++ seeding
+        c0_cache = getattr(executor, "_c0_cache", None)
+        if c0_cache is None:
+            c0_cache = executor._c0_cache = {}
         for slot, k, ni, merged in blocks:
             seed = merged.get("random_state")
             base_seed = seed if seed is not None else 1234 + slot
             for r in range(ni):
-                rng = np.random.default_rng(int(base_seed) + r)
-                sample = executor.init_sample(rng)
-                C0_parts.append(
-                    kmeans_init_centers(sample, k, merged["init"], rng))
+                ckey = (k, merged["init"], int(base_seed) + r)
+                C0 = c0_cache.get(ckey)
+                if C0 is None:
+                    rng = np.random.default_rng(int(base_seed) + r)
+                    sample = executor.init_sample(rng)
+                    C0 = kmeans_init_centers(sample, k, merged["init"], rng)
+                    if len(c0_cache) > 512:
+                        c0_cache.clear()
+                    c0_cache[ckey] = C0
+                C0_parts.append(C0)
                 cstart.append(cstart[-1] + k)
                 block_to_cand.append(slot)
                 max_iters.append(int(merged["max_iter"]))
@@ -217,11 +231,15 @@ class _HostKMeansExecutor:
                                         axis=1)))
 
     def init_sample(self, rng) -> np.ndarray:
-        n = self.X.shape[0]
-        if n <= INIT_SAMPLE_ROWS:
-            return self.X
-        rows = rng.choice(n, size=INIT_SAMPLE_ROWS, replace=False)
-        return self.X[rows]
+        if getattr(self, "_init_sample", None) is None:
+            n = self.X.shape[0]
+            if n <= INIT_SAMPLE_ROWS:
+                self._init_sample = self.X
+            else:
+                rows = np.random.default_rng(0).choice(
+                    n, size=INIT_SAMPLE_ROWS, replace=False)
+                self._init_sample = self.X[rows]
+        return self._init_sample
 
     def fit(self, train_idx, plan):
         C_all, n_iter, inertia = kmeans_fit_ref(

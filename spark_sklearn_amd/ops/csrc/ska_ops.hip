@@ -1246,6 +1246,125 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
       if (sCnt[i] != 0.f) atomicAdd(&counts[i], sCnt[i]);
 }
 
+
+// wide assign: 128-row tiles, 512 threads (2 row-halves x 4 col-waves),
+// for Kp <= 128 (zbuf LDS bound).  4x the MFMA per staged byte of the
+// 32-row kernel.
+template <int NF>   // col-frags per wave: Kp = NF * 64
+__global__ __launch_bounds__(512)
+void kmeans_assign_w_kernel(const __bf16* __restrict__ X,
+                            const int* __restrict__ idx,    // [NTp]
+                            const __bf16* __restrict__ Ct,  // [Kp][Fp]
+                            const float* __restrict__ cnorm,
+                            const float* __restrict__ xnorm,
+                            const int* __restrict__ cstart,
+                            unsigned char* __restrict__ labels,
+                            float* __restrict__ inertia,
+                            float* __restrict__ counts,
+                            int nt, int NTp, int Fp, int Kp, int nb,
+                            int want_labels) {
+  constexpr int cpw = NF * 16;
+  const int Kpad = NF * 64;
+  extern __shared__ char kw_smem[];
+  float* zbuf = reinterpret_cast<float*>(kw_smem);           // [128][Kpad+8]
+  float* sIn = zbuf + 128 * (Kpad + 8);                      // [256]
+  float* sCnt = sIn + 256;                                   // [Kpad]
+  int* sIdx = reinterpret_cast<int*>(sCnt + Kpad);           // [128]
+  __bf16* sX = reinterpret_cast<__bf16*>(sIdx + 128);        // [128][64]
+  __bf16* sC = sX + 128 * 64;                                // [Kpad][64]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rhalf = wave >> 2;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+
+  if (tid < 128) sIdx[tid] = idx[rt * 128 + tid];
+  for (int i = tid; i < nb; i += 512) sIn[i] = 0.f;
+  for (int i = tid; i < Kpad; i += 512) sCnt[i] = 0.f;
+  __syncthreads();
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int kc = 0; kc < Fp; kc += 64) {
+    {
+      int r = tid >> 2, seg = tid & 3;
+      const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+          X + (long)sIdx[r] * Fp + kc + seg * 16);
+      store_row8(&sX[r * 64], r, seg * 16, gx[0]);
+      store_row8(&sX[r * 64], r, seg * 16 + 8, gx[1]);
+    }
+    for (int e = tid; e < Kpad * 8; e += 512) {
+      int c = e >> 3, seg = e & 7;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          Ct + (long)c * Fp + kc + seg * 8);
+      store_row8(&sC[c * 64], c, seg * 8, v);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rhalf * 64 + m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * cpw + fn * 16 + (lane & 15);
+        bf16x8 b = lds_frag_p(&sC[c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], b, acc[m][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col = cwave * cpw + fn * 16 + (lane & 15);
+    float cn = cnorm[col];
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int rowb = rhalf * 64 + m * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        zbuf[(rowb + r) * (Kpad + 8) + col] = cn - 2.f * acc[m][fn][r];
+    }
+  }
+  __syncthreads();
+
+  for (int p = tid; p < 128 * nb; p += 512) {
+    int row = p & 127;
+    int b = p >> 7;
+    int gr = rt * 128 + row;
+    if (gr >= nt) continue;
+    int lo = cstart[b], hi = cstart[b + 1];
+    const float* q = &zbuf[row * (Kpad + 8)];
+    float best = q[lo];
+    int bj = 0;
+    for (int j = lo + 1; j < hi; ++j)
+      if (q[j] < best) { best = q[j]; bj = j - lo; }
+    if (want_labels)
+      labels[(long)b * NTp + gr] = (unsigned char)bj;
+    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], 1.f);
+    atomicAdd(&sIn[b], xnorm[sIdx[row]] + best);
+  }
+  __syncthreads();
+  for (int b = tid; b < nb; b += 512) atomicAdd(&inertia[b], sIn[b]);
+  if (counts != nullptr)
+    for (int i = tid; i < Kp; i += 512)
+      if (sCnt[i] != 0.f) atomicAdd(&counts[i], sCnt[i]);
+}
+
 // one-hot membership matrix: D[cstart[b]+label_b(r)][r] = 1 for train
 // rows.  Feeds the MFMA bwd kernel: sums = X^T . onehot -- replaces the
 // LDS-atomic update kernel (same-label rows serialized its atomics).
@@ -2455,6 +2574,23 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
                                  float* counts, int nt, int NTp, int Fp,
                                  int Kp, int nb, int want_labels,
                                  hipStream_t s) {
+  if (Kp <= 128 && NTp % 128 == 0) {
+    size_t wsm = (size_t)128 * (Kp + 8) * 4 + 256 * 4 + (size_t)Kp * 4 +
+                 128 * 4 + (size_t)128 * 64 * 2 + (size_t)Kp * 64 * 2;
+    dim3 wgrid((unsigned)(NTp / 128));
+    auto gow = [&](auto k) {
+      HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(k),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)wsm));
+      hipLaunchKernelGGL(k, wgrid, dim3(512), wsm, s, X, idx, Ct, cnorm,
+                         xnorm, cstart, labels, inertia, counts, nt, NTp,
+                         Fp, Kp, nb, want_labels);
+    };
+    if (Kp == 64) gow(kmeans_assign_w_kernel<1>);
+    else gow(kmeans_assign_w_kernel<2>);
+    HIP_CHECK(hipGetLastError());
+    return;
+  }
   size_t smem = (size_t)32 * (Kp + 8) * 4 + 256 * 4 + (size_t)Kp * 4 +
                 32 * 4 + (size_t)32 * 64 * 2 + (size_t)Kp * 64 * 2;
   dim3 grid((unsigned)(NTp / 32));

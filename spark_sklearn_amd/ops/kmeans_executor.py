@@ -36,20 +36,26 @@ class DeviceKMeansExecutor:
         self._score_state = None
 
     def init_sample(self, rng) -> np.ndarray:
-        import torch
+        """Row sample for host-side seeding, pulled ONCE per executor (the
+        per-block rng still drives k-means++ randomness)."""
+        if getattr(self, "_init_sample", None) is None:
+            import torch
 
-        if self.n <= INIT_SAMPLE_ROWS:
-            rows = np.arange(self.n)
-        else:
-            rows = rng.choice(self.n, size=INIT_SAMPLE_ROWS, replace=False)
-        rows_t = torch.as_tensor(np.asarray(rows, np.int64)).to(self.device)
-        return self.X[rows_t, :self.F].float().cpu().numpy()
+            if self.n <= INIT_SAMPLE_ROWS:
+                rows = np.arange(self.n)
+            else:
+                rows = np.random.default_rng(0).choice(
+                    self.n, size=INIT_SAMPLE_ROWS, replace=False)
+            rows_t = torch.as_tensor(
+                np.asarray(rows, np.int64)).to(self.device)
+            self._init_sample = self.X[rows_t, :self.F].float().cpu().numpy()
+        return self._init_sample
 
     def _pad_idx(self, idx):
         import torch
 
         n = len(idx)
-        NTp = _roundup(max(n, 32), 32)
+        NTp = _roundup(max(n, 128), 128)
         out = np.zeros(NTp, dtype=np.int32)
         out[:n] = np.asarray(idx, dtype=np.int32)
         return torch.as_tensor(out).to(self.device), n
