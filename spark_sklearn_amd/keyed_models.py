@@ -119,21 +119,122 @@ class KeyedEstimator:
         if missing:
             raise ValueError(f"DataFrame is missing columns {missing}")
 
-        records = []
-        for key, group in df.groupby(self.keyCols, sort=True):
-            if not isinstance(key, tuple):
-                key = (key,)
-            X = _stack_feature_column(group[self.xCol].values)
-            est = clone(self.sklearnEstimator)
-            if self.yCol is not None:
-                est.fit(X, np.asarray(group[self.yCol].values))
-            else:
-                est.fit(X)
-            records.append(key + (SparkSklearnEstimator(est),))
+        groups = [((key if isinstance(key, tuple) else (key,)), group)
+                  for key, group in df.groupby(self.keyCols, sort=True)]
+        fitted = self._fit_keys_batched(groups)
+        if fitted is None:
+            fitted = []
+            for key, group in groups:
+                X = _stack_feature_column(group[self.xCol].values)
+                est = clone(self.sklearnEstimator)
+                if self.yCol is not None:
+                    est.fit(X, np.asarray(group[self.yCol].values))
+                else:
+                    est.fit(X)
+                fitted.append(key + (SparkSklearnEstimator(est),))
         models = pd.DataFrame.from_records(
-            records, columns=self.keyCols + ["estimator"])
+            fitted, columns=self.keyCols + ["estimator"])
         return KeyedModel(models, keyCols=self.keyCols, xCol=self.xCol,
                           yCol=self.yCol, estimatorType=self.estimatorType)
+
+    def _fit_keys_batched(self, groups):
+        """Batched per-key fits for the linear family on GPU: keys become
+        candidate columns of ONE kernel launch, with per-column row masks
+        ("per-key fits batched into single kernel launches,
+        batch dim = keys").  Per-key sample counts are folded into the
+        per-column learning rate and regularization (exact algebra, see
+        the scaling below).  Returns None when the fast path does not
+        apply (non-linear estimator, no GPU, or key-dependent class sets).
+        """
+        proto = self.sklearnEstimator
+        if getattr(type(proto), "_engine_family", None) != "linear" or \
+                self.yCol is None:
+            return None
+        from .engine import _device_or_none
+
+        device = _device_or_none(None)
+        if device is None:
+            return None
+        from sklearn.base import is_classifier
+
+        ys = [np.asarray(g[self.yCol].values) for _, g in groups]
+        if is_classifier(proto):
+            classes = np.unique(np.concatenate(ys))
+            if any(len(np.unique(y)) != len(classes) for y in ys):
+                return None     # keys with missing classes: host loop
+            n_classes = len(classes)
+        else:
+            classes, n_classes = None, 0
+        C = n_classes if n_classes > 2 else 1
+        if C > 64:
+            return None
+
+        Xs = [_stack_feature_column(g[self.xCol].values) for _, g in groups]
+        if any(not isinstance(x, np.ndarray) for x in Xs):
+            return None     # sparse keyed rows: host loop
+        X_all = np.concatenate(Xs).astype(np.float32)
+        if classes is not None:
+            lut = {v: i for i, v in enumerate(classes)}
+            y_all = np.concatenate(
+                [np.vectorize(lut.get)(y) for y in ys]).astype(np.int32)
+        else:
+            y_all = np.concatenate(ys).astype(np.float32)
+        n_total = X_all.shape[0]
+        row_key = np.repeat(np.arange(len(groups), dtype=np.int32),
+                            [len(x) for x in Xs])
+
+        from .models._solver_ref import BatchPlan
+        from .ops.device_executor import DeviceLinearExecutor
+
+        executor = DeviceLinearExecutor(X_all, y_all, classes, device)
+        fitted = []
+        keys_per_chunk = max(1, 256 // C)
+        for lo in range(0, len(groups), keys_per_chunk):
+            chunk = groups[lo:lo + keys_per_chunk]
+            B = len(chunk) * C
+            plan = BatchPlan(
+                loss=np.empty(B, np.int32), lr=np.empty(B, np.float32),
+                l1=np.empty(B, np.float32), l2=np.empty(B, np.float32),
+                momentum=np.empty(B, np.float32),
+                tol=np.empty(B, np.float32),
+                max_iter=np.empty(B, np.int32),
+                target_class=np.empty(B, np.int32),
+                col_to_cand=np.empty(B, np.int32), n_classes=C,
+                fit_intercept=np.empty(B, np.uint8))
+            col_key = np.empty(B, np.int32)
+            for slot, (key, g) in enumerate(chunk):
+                n_k = len(g)
+                spec = proto._column_spec(n_k, executor.msq)
+                # masked columns see only their key's rows, but the kernel
+                # normalizes by n_total: scale lr by n_total/n_k and the
+                # penalties by n_k/n_total so the effective update equals
+                # the per-key fit exactly
+                scale = n_total / max(n_k, 1)
+                for c in range(C):
+                    j = slot * C + c
+                    plan.loss[j] = spec["loss"]
+                    plan.lr[j] = spec["lr"] * scale
+                    plan.l1[j] = spec["l1"] / scale
+                    plan.l2[j] = spec["l2"] / scale
+                    plan.momentum[j] = spec["momentum"]
+                    plan.tol[j] = spec["tol"]
+                    plan.max_iter[j] = spec["max_iter"]
+                    plan.fit_intercept[j] = spec["fit_intercept"]
+                    plan.target_class[j] = c if C > 1 else \
+                        (1 if n_classes else -1)
+                    plan.col_to_cand[j] = slot
+                    col_key[j] = lo + slot
+            (W, b), n_iter, _ = executor.fit(
+                np.arange(n_total, dtype=np.int64), plan,
+                row_key=row_key, col_key=col_key)
+            for slot, (key, g) in enumerate(chunk):
+                est = clone(proto)
+                cols = slice(slot * C, (slot + 1) * C)
+                est._finalize_fit(np.ascontiguousarray(W[:, cols]),
+                                  np.ascontiguousarray(b[cols]),
+                                  n_iter[cols], classes)
+                fitted.append(tuple(key) + (SparkSklearnEstimator(est),))
+        return fitted
 
 
 class KeyedModel:

@@ -235,9 +235,13 @@ void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
                    __bf16* __restrict__ Dt,            // [Bp][NTp]
                    float* __restrict__ loss_sum,       // [Bp]
                    float* __restrict__ db_sum,         // [Bp]
+                   const int* __restrict__ row_key,    // unused (generic)
+                   const int* __restrict__ col_key,    // unused (generic)
                    int nt, int NTp, int Fp, int Bp, int want_loss) {
   constexpr int AMAP = P >> 1, BMAP = P & 1;
   (void)want_loss;  // generic path always computes loss
+  (void)row_key;
+  (void)col_key;
   __shared__ __bf16 sX[64][72];
   __shared__ __bf16 sW[256][72];
   __shared__ float sLoss[256];
@@ -443,6 +447,8 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                    const int* __restrict__ loss_id,
                    const int* __restrict__ target, __bf16* __restrict__ Dt,
                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                   const int* __restrict__ row_key,   // [n] or null
+                   const int* __restrict__ col_key,   // [Bp] or null
                    int nt, int NTp, int Fp, int Bp, int want_loss) {
   __shared__ __bf16 sX[2][64 * 64];
   __shared__ __bf16 sW[2][NF * 64 * 64];
@@ -450,6 +456,7 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   __shared__ float sDb[256];
   __shared__ int sIdx[64];
   __shared__ float sY[64];
+  __shared__ int sKey[64];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -465,6 +472,7 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     int gi = idx[rt * 64 + tid];
     sIdx[tid] = gi;
     sY[tid] = y[gi];
+    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
   }
   for (int i = tid; i < Bp; i += 256) { sLoss[i] = 0.f; sDb[i] = 0.f; }
   __syncthreads();
@@ -536,6 +544,7 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     float bia = bias[col_in];
     int lid = loss_id[col_in];
     int tgt = target[col_in];
+    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
     float lsum = 0.f, dsum = 0.f;
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
@@ -562,6 +571,10 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
           }
         }
         if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;          // keyed batching: column only sees its key
+          ls = 0.f;
+        }
         dpack[r] = (__bf16)d;
         lsum += ls;
         dsum += d;
@@ -1749,6 +1762,8 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                    const int* __restrict__ loss_id,
                    const int* __restrict__ target, __bf16* __restrict__ Dt,
                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                   const int* __restrict__ row_key,
+                   const int* __restrict__ col_key,
                    int nt, int NTp, int Fp, int Bp, int want_loss) {
   constexpr int NF = 4;
   __shared__ __bf16 sX[2][256 * 64];
@@ -1757,6 +1772,7 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   __shared__ float sDb[256];
   __shared__ int sIdx[256];
   __shared__ float sY[256];
+  __shared__ int sKey[256];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -1772,6 +1788,7 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     int gi = idx[rt * 256 + tid];
     sIdx[tid] = gi;
     sY[tid] = y[gi];
+    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
     sLoss[tid] = 0.f;
     sDb[tid] = 0.f;
   }
@@ -1836,6 +1853,7 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     float bia = bias[col_in];
     int lid = loss_id[col_in];
     int tgt = target[col_in];
+    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
     float lsum = 0.f, dsum = 0.f;
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
@@ -1862,6 +1880,10 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
           }
         }
         if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
         dpack[r] = (__bf16)d;
         lsum += ls;
         dsum += d;
@@ -2301,6 +2323,7 @@ std::vector<torch::Tensor> linear_fit(
     torch::Tensor loss_id, torch::Tensor lr, torch::Tensor l1,
     torch::Tensor l2, torch::Tensor mom, torch::Tensor tol,
     torch::Tensor max_iter, torch::Tensor target, torch::Tensor fit_ic,
+    torch::Tensor row_key, torch::Tensor col_key,
     int64_t nt, int64_t max_epochs, int64_t check_every) {
   TORCH_CHECK(g_layout >= 0, "MFMA layout not probed; call ensure_layout()");
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kBFloat16 && X.dim() == 2);
@@ -2359,6 +2382,8 @@ std::vector<torch::Tensor> linear_fit(
 
   const __bf16* Xtp = reinterpret_cast<const __bf16*>(Xt.data_ptr());
   long n_rows = X.size(0);
+  const int* rkp = row_key.numel() ? row_key.data_ptr<int>() : nullptr;
+  const int* ckp = col_key.numel() ? col_key.data_ptr<int>() : nullptr;
   bool any_tol = tol.max().item<float>() > 0.f;
   int want_loss_cur = 1;
   auto launch_epoch = [&](int epoch, auto fwd_k, auto bwd_k) {
@@ -2369,7 +2394,7 @@ std::vector<torch::Tensor> linear_fit(
                        target.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(Dt.data_ptr()),
                        loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
-                       (int)nt, (int)NTp, Fp, Bp, want_loss_cur);
+                       rkp, ckp, (int)nt, (int)NTp, Fp, Bp, want_loss_cur);
     hipLaunchKernelGGL(bwd_k, bwd_grid, dim3(256), 0, s, Xtp,
                        idx.data_ptr<int>(),
                        reinterpret_cast<const __bf16*>(Dt.data_ptr()),
@@ -2408,8 +2433,8 @@ std::vector<torch::Tensor> linear_fit(
                            target.data_ptr<int>(),
                            reinterpret_cast<__bf16*>(Dt.data_ptr()),
                            loss_sum.data_ptr<float>(),
-                           db_sum.data_ptr<float>(), (int)nt, (int)NTp, Fp,
-                           Bp, want_loss_cur);
+                           db_sum.data_ptr<float>(), rkp, ckp, (int)nt,
+                           (int)NTp, Fp, Bp, want_loss_cur);
         if (Fp % 256 == 0) {
           hipLaunchKernelGGL(bwd_g8_kernel,
                              dim3((unsigned)(Fp / 256), (unsigned)nseg),
@@ -3078,8 +3103,8 @@ std::vector<torch::Tensor> linear_search_run(
                            bias.data_ptr<float>(), loss_p, tgt_p,
                            reinterpret_cast<__bf16*>(Dt.data_ptr()),
                            loss_sum.data_ptr<float>(),
-                           db_sum.data_ptr<float>(), (int)nt, (int)NTp, Fp,
-                           Bp, want_loss);
+                           db_sum.data_ptr<float>(), nullptr, nullptr,
+                           (int)nt, (int)NTp, Fp, Bp, want_loss);
         if (Fp % 256 == 0) {
           hipLaunchKernelGGL(bwd_g8_kernel,
                              dim3((unsigned)(Fp / 256), (unsigned)nseg),

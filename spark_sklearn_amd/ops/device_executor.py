@@ -153,7 +153,7 @@ class DeviceLinearExecutor:
         return t
 
     # -- fit ---------------------------------------------------------------
-    def fit(self, train_idx, plan):
+    def fit(self, train_idx, plan, row_key=None, col_key=None):
         torch = self.torch
         nt = len(train_idx)
         idx = self._cached_idx(train_idx, 256, np.int32)
@@ -186,9 +186,17 @@ class DeviceLinearExecutor:
 
         max_epochs = int(plan.max_iter.max()) if B else 0
         check_every = 32 if max_epochs > 48 else 0
+        if row_key is not None:
+            rk_t = torch.as_tensor(
+                np.asarray(row_key, dtype=np.int32)).to(self.device)
+            ck_t = pad_i32(col_key, fill=-1)
+        else:
+            rk_t = torch.empty(0, dtype=torch.int32, device=self.device)
+            ck_t = torch.empty(0, dtype=torch.int32, device=self.device)
         W, bias, n_iter, cur_loss, Wt = self.ext.linear_fit(
             self.X, self.Xt, self.y, idx, loss_t, lr_t, l1_t, l2_t, mom_t,
-            tol_t, mi_t, tgt_t, ic_t, nt, max_epochs, check_every)
+            tol_t, mi_t, tgt_t, ic_t, rk_t, ck_t, nt, max_epochs,
+            check_every)
         torch.cuda.synchronize(self.device)
         self._fit_state = (Wt, bias, loss_t, tgt_t)
         W_np = W[:B, :self.F].float().cpu().numpy().T.copy()   # [F, B]
