@@ -23,7 +23,8 @@ from dataclasses import dataclass
 import numpy as np
 
 __all__ = ["GBTSpec", "quantile_bins", "bin_data", "gbt_fit_ref",
-           "gbt_predict_raw", "sigmoid"]
+           "gbt_fit_ref_multi", "gbt_predict_raw", "gbt_predict_raw_multi",
+           "sigmoid", "softmax_rows"]
 
 LOSS_LOGISTIC = 0
 LOSS_SQUARED = 2
@@ -198,4 +199,115 @@ def gbt_predict_raw(Xb: np.ndarray, trees, f0: float) -> np.ndarray:
     raw = np.full(Xb.shape[0], f0, dtype=np.float32)
     for feat, thr, val in trees:
         raw += _tree_apply(Xb, feat, thr, val)
+    return raw
+
+
+# ---------------------------------------------------------------------------
+# multiclass (K > 2): softmax boosting -- one tree per class per iteration
+# (sklearn GradientBoostingClassifier's multinomial deviance scheme)
+# ---------------------------------------------------------------------------
+
+def softmax_rows(raw: np.ndarray) -> np.ndarray:
+    z = raw - raw.max(axis=1, keepdims=True)
+    e = np.exp(z)
+    return e / e.sum(axis=1, keepdims=True)
+
+
+def _fit_one_tree(Xb, g, h, spec):
+    """One depth-wise tree on the given gradients; returns (feat, thr,
+    val, contribution)."""
+    one = GBTSpec(loss=spec.loss, n_estimators=1,
+                  learning_rate=spec.learning_rate,
+                  max_depth=spec.max_depth,
+                  min_samples_leaf=spec.min_samples_leaf,
+                  l2_regularization=spec.l2_regularization,
+                  max_bins=spec.max_bins)
+    # reuse the single-tree builder by faking the gradient closure
+    return _build_tree(Xb, g, h, one)
+
+
+def _build_tree(Xb, g, h, spec):
+    n, F = Xb.shape
+    n_nodes = 2 ** (spec.max_depth + 1) - 1
+    lam = float(spec.l2_regularization)
+    feat = np.full(n_nodes, -1, dtype=np.int32)
+    thr = np.zeros(n_nodes, dtype=np.uint8)
+    val = np.zeros(n_nodes, dtype=np.float32)
+    node = np.zeros(n, dtype=np.int32)
+    alive = np.ones(n, dtype=bool)
+    for depth in range(spec.max_depth):
+        lvl_lo, lvl_hi = 2 ** depth - 1, 2 ** (depth + 1) - 1
+        live = np.flatnonzero(alive)
+        if len(live) == 0:
+            break
+        nodes_here = np.unique(node[live])
+        nodes_here = nodes_here[(nodes_here >= lvl_lo) &
+                                (nodes_here < lvl_hi)]
+        for nd in nodes_here:
+            rows = live[node[live] == nd]
+            G, H, cnt = float(g[rows].sum()), float(h[rows].sum()), len(rows)
+            parent_score = G * G / (H + lam + 1e-12)
+            best_gain, best_f, best_t = 0.0, -1, 0
+            for f in range(F):
+                bins = Xb[rows, f]
+                gh = np.bincount(bins, weights=g[rows], minlength=256)
+                hh = np.bincount(bins, weights=h[rows], minlength=256)
+                ch = np.bincount(bins, minlength=256)
+                gl, hl, cl = np.cumsum(gh)[:-1], np.cumsum(hh)[:-1],                     np.cumsum(ch)[:-1]
+                gr, hr, cr = G - gl, H - hl, cnt - cl
+                okay = (cl >= spec.min_samples_leaf) &                     (cr >= spec.min_samples_leaf)
+                gain = np.where(
+                    okay, gl * gl / (hl + lam + 1e-12) +
+                    gr * gr / (hr + lam + 1e-12) - parent_score, -np.inf)
+                t = int(np.argmax(gain))
+                if gain[t] > best_gain + 1e-12:
+                    best_gain, best_f, best_t = float(gain[t]), f, t
+            if best_f < 0:
+                feat[nd] = -1
+                val[nd] = -G / (H + lam + 1e-12) * spec.learning_rate
+                alive[rows] = False
+                continue
+            feat[nd] = best_f
+            thr[nd] = best_t
+            node[rows] = 2 * nd + 1 + (Xb[rows, best_f] > best_t
+                                       ).astype(np.int32)
+    live = np.flatnonzero(alive)
+    for nd in np.unique(node[live]):
+        rows = live[node[live] == nd]
+        G, H = float(g[rows].sum()), float(h[rows].sum())
+        feat[nd] = -1
+        val[nd] = -G / (H + lam + 1e-12) * spec.learning_rate
+    return feat, thr, val, _tree_apply(Xb, feat, thr, val)
+
+
+def gbt_fit_ref_multi(Xb: np.ndarray, y_enc: np.ndarray, K: int,
+                      spec: GBTSpec):
+    """Softmax boosting: trees[t][c].  Returns (trees, f0 [K])."""
+    n = Xb.shape[0]
+    y_enc = np.asarray(y_enc, dtype=np.int64)
+    prior = np.bincount(y_enc, minlength=K) / n
+    f0 = np.log(np.clip(prior, 1e-6, None)).astype(np.float32)
+    raw = np.tile(f0, (n, 1))
+    onehot = np.eye(K, dtype=np.float32)[y_enc]
+    trees = []
+    for _ in range(spec.n_estimators):
+        P = softmax_rows(raw)
+        round_trees = []
+        for c in range(K):
+            g = (P[:, c] - onehot[:, c]).astype(np.float32)
+            h = (P[:, c] * (1.0 - P[:, c])).astype(np.float32)
+            feat, thr, val, contrib = _build_tree(Xb, g, h, spec)
+            raw[:, c] += contrib
+            round_trees.append((feat, thr, val))
+        trees.append(round_trees)
+    return trees, f0
+
+
+def gbt_predict_raw_multi(Xb: np.ndarray, trees, f0: np.ndarray):
+    n = Xb.shape[0]
+    K = len(f0)
+    raw = np.tile(np.asarray(f0, dtype=np.float32), (n, 1))
+    for round_trees in trees:
+        for c, (feat, thr, val) in enumerate(round_trees):
+            raw[:, c] += _tree_apply(Xb, feat, thr, val)
     return raw

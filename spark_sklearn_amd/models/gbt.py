@@ -21,7 +21,9 @@ import numpy as np
 from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin
 
 from ._gbt_ref import (GBTSpec, LOSS_LOGISTIC, LOSS_SQUARED, bin_data,
-                       gbt_fit_ref, gbt_predict_raw, quantile_bins, sigmoid)
+                       gbt_fit_ref, gbt_fit_ref_multi, gbt_predict_raw,
+                       gbt_predict_raw_multi, quantile_bins, sigmoid,
+                       softmax_rows)
 
 __all__ = ["GradientBoostingClassifier", "GradientBoostingRegressor"]
 
@@ -65,6 +67,8 @@ class _GBTBase(BaseEstimator):
     def _raw_predict(self, X):
         self._check_fitted()
         Xb = bin_data(np.asarray(X), self._bin_edges)
+        if np.ndim(self._f0) == 1:          # multiclass: [n, K]
+            return gbt_predict_raw_multi(Xb, self._trees, self._f0)
         return gbt_predict_raw(Xb, self._trees, self._f0)
 
     def _check_fitted(self):
@@ -74,16 +78,23 @@ class _GBTBase(BaseEstimator):
 
 
 class GradientBoostingClassifier(ClassifierMixin, _GBTBase):
-    """Binary histogram gradient boosting (logistic loss) on CDNA4."""
+    """Histogram gradient boosting classifier on CDNA4 (binary logistic;
+    multiclass via softmax boosting: one tree per class per round)."""
 
     _loss_id = LOSS_LOGISTIC
 
     def predict_proba(self, X):
-        p = sigmoid(self._raw_predict(X))
+        raw = self._raw_predict(X)
+        if raw.ndim == 2:
+            return softmax_rows(raw)
+        p = sigmoid(raw)
         return np.column_stack([1.0 - p, p])
 
     def predict(self, X):
-        return self.classes_[(self._raw_predict(X) > 0).astype(int)]
+        raw = self._raw_predict(X)
+        if raw.ndim == 2:
+            return self.classes_[np.argmax(raw, axis=1)]
+        return self.classes_[(raw > 0).astype(int)]
 
     def decision_function(self, X):
         return self._raw_predict(X)
@@ -117,11 +128,8 @@ class GBTFamily:
         y = np.asarray(y)
         if is_classifier(estimator):
             classes, y_enc = np.unique(y, return_inverse=True)
-            if len(classes) != 2:
-                raise ValueError(
-                    "GradientBoostingClassifier supports binary targets "
-                    f"(got {len(classes)} classes); multiclass lands with "
-                    "the OVR extension")
+            if len(classes) > 64:
+                raise ValueError("at most 64 classes supported")
             return classes, y_enc.astype(np.float32)
         return None, np.asarray(y, dtype=np.float32)
 
@@ -138,8 +146,8 @@ class GBTFamily:
         if device is not None:
             from ..ops.gbt_executor import DeviceGBTExecutor
 
-            return DeviceGBTExecutor(X, y_enc, device)
-        return _HostGBTExecutor(X, y_enc)
+            return DeviceGBTExecutor(X, y_enc, device, classes=classes)
+        return _HostGBTExecutor(X, y_enc, classes)
 
     @staticmethod
     def build_plan(estimator, params_list, n_samples, executor, n_classes,
@@ -164,7 +172,7 @@ class GBTFamily:
                     raise
                 bad[slot] = exc
                 specs.append(GBTSpec(n_estimators=0, max_depth=1))
-        plan = _GBTPlan(specs)
+        plan = _GBTPlan(specs, n_classes)
         return plan, bad
 
     @staticmethod
@@ -196,10 +204,11 @@ class GBTFamily:
 
 
 class _GBTPlan:
-    def __init__(self, specs):
+    def __init__(self, specs, n_label_classes=0):
         self.specs = specs
         self.col_to_cand = np.arange(len(specs), dtype=np.int32)
-        self.n_classes = 1
+        self.n_classes = 1            # engine chunking unit (per candidate)
+        self.n_label_classes = n_label_classes
 
     @property
     def B(self):
@@ -209,11 +218,12 @@ class _GBTPlan:
 class _HostGBTExecutor:
     """Numpy execution (reference semantics); bins built once per matrix."""
 
-    def __init__(self, X, y_enc):
+    def __init__(self, X, y_enc, classes=None):
         X = np.asarray(X)
         self.edges = quantile_bins(X)
         self.Xb = bin_data(X, self.edges)
         self.y = y_enc
+        self.K = len(classes) if classes is not None else 0
         self.msq = 0.0   # unused by this family
 
     def fit(self, train_idx, plan):
@@ -222,7 +232,11 @@ class _HostGBTExecutor:
         models = []
         n_iter = np.empty(plan.B, np.int32)
         for i, spec in enumerate(plan.specs):
-            trees, f0 = gbt_fit_ref(Xb_tr, y_tr, spec)
+            if self.K > 2:
+                trees, f0 = gbt_fit_ref_multi(
+                    Xb_tr, y_tr.astype(np.int64), self.K, spec)
+            else:
+                trees, f0 = gbt_fit_ref(Xb_tr, y_tr, spec)
             models.append((trees, f0))
             n_iter[i] = len(trees)
         return (models, self.edges), n_iter, None
@@ -233,6 +247,12 @@ class _HostGBTExecutor:
         yv = self.y[idx]
         out = np.empty(len(models))
         for i, (trees, f0) in enumerate(models):
+            if self.K > 2:
+                raw = gbt_predict_raw_multi(Xb, trees, f0)
+                assert metric == "accuracy"
+                out[i] = float(np.mean(raw.argmax(axis=1) ==
+                                       yv.astype(np.int64)))
+                continue
             raw = gbt_predict_raw(Xb, trees, f0)
             if metric == "accuracy":
                 out[i] = float(np.mean((raw > 0) == (yv > 0.5)))

@@ -2037,6 +2037,67 @@ void scores3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 }
 
 
+
+// multiclass softmax gradients: raw [K][nt] class-major
+__global__ void gbt_softmax_grad_kernel(const float* __restrict__ raw,
+                                        const float* __restrict__ y,
+                                        float* __restrict__ g,
+                                        float* __restrict__ h,
+                                        long nt, int K) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    float mx = raw[r];
+    for (int c = 1; c < K; ++c) mx = fmaxf(mx, raw[(long)c * nt + r]);
+    float sum = 0.f;
+    for (int c = 0; c < K; ++c) sum += __expf(raw[(long)c * nt + r] - mx);
+    int yc = (int)y[r];
+    for (int c = 0; c < K; ++c) {
+      float p = __expf(raw[(long)c * nt + r] - mx) / sum;
+      g[(long)c * nt + r] = p - ((c == yc) ? 1.f : 0.f);
+      h[(long)c * nt + r] = p * (1.f - p);
+    }
+  }
+}
+
+// per-class leaf apply into class-major raw
+__global__ void gbt_leaf_apply_c_kernel(float* __restrict__ raw_c,
+                                        int* __restrict__ node_id,
+                                        const float* __restrict__ val,
+                                        long nt) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    raw_c[r] += val[node_id[r]];
+    node_id[r] = 0;
+  }
+}
+
+// multiclass accuracy: argmax over K class-major raw planes
+__global__ void gbt_argmax_acc_kernel(const float* __restrict__ raws,
+                                      const float* __restrict__ y,
+                                      const int* __restrict__ idx,
+                                      long n_idx, int K,
+                                      double* __restrict__ out) {
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  double local = 0.0;
+  for (; r < n_idx; r += (long)gridDim.x * blockDim.x) {
+    int best = 0;
+    float bz = raws[r];
+    for (int c = 1; c < K; ++c) {
+      float z = raws[(long)c * n_idx + r];
+      if (z > bz) { bz = z; best = c; }
+    }
+    local += (best == (int)y[idx[r]]) ? 1.0 : 0.0;
+  }
+  __shared__ double red[256];
+  red[threadIdx.x] = local;
+  __syncthreads();
+  for (int s2 = blockDim.x / 2; s2 > 0; s2 >>= 1) {
+    if (threadIdx.x < s2) red[threadIdx.x] += red[threadIdx.x + s2];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(out, red[0]);
+}
+
 // ===========================================================================
 // K5: CSR-resident linear fits -- for matrices too large or too sparse to
 // densify into HBM.  Layouts:
@@ -3347,6 +3408,134 @@ torch::Tensor csr_linear_scores(
   return out;
 }
 
+
+// multiclass boosting fit: trees [T*K][NN], class-major raw.
+std::vector<torch::Tensor> gbt_fit_multi(
+    torch::Tensor Xb, torch::Tensor idxmap, torch::Tensor y_tr,
+    torch::Tensor f0, int64_t K, int64_t n_estimators, double lrate,
+    int64_t max_depth, int64_t min_leaf, double lam) {
+  const at::cuda::CUDAGuard guard(Xb.device());
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 6);
+  TORCH_CHECK(K >= 2 && K <= 64);
+  int F = (int)Xb.size(1);
+  long nt = idxmap.size(0);
+  int T = (int)n_estimators;
+  int NN = (1 << (max_depth + 1)) - 1;
+
+  auto optF = y_tr.options().dtype(torch::kFloat32);
+  auto optI = y_tr.options().dtype(torch::kInt32);
+  auto raw = f0.to(optF).reshape({K, 1}).repeat({1, nt}).contiguous();
+  auto g = torch::empty({K, nt}, optF);
+  auto h = torch::empty({K, nt}, optF);
+  auto node_id = torch::zeros({nt}, optI);
+  auto feats = torch::full({(long)T * K, NN}, -1, optI);
+  auto thrs = torch::zeros({(long)T * K, NN}, optI);
+  auto vals = torch::zeros({(long)T * K, NN}, optF);
+  int max_lvl_nodes = 1 << (max_depth - 1);
+  auto hist = torch::empty({(long)max_lvl_nodes * F * 256 * 3}, optF);
+  auto hist_prev = torch::empty({(long)std::max(1, max_lvl_nodes / 2) *
+                                 F * 256 * 3}, optF);
+  auto gsum = torch::zeros({NN}, optF);
+  auto hsum = torch::zeros({NN}, optF);
+
+  hipStream_t s = cur_stream();
+  const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+
+  for (int t = 0; t < T; ++t) {
+    hipLaunchKernelGGL(gbt_softmax_grad_kernel, dim3(blocks), dim3(256), 0,
+                       s, raw.data_ptr<float>(), y_tr.data_ptr<float>(),
+                       g.data_ptr<float>(), h.data_ptr<float>(), nt,
+                       (int)K);
+    for (int c = 0; c < (int)K; ++c) {
+      float* gc = g.data_ptr<float>() + (long)c * nt;
+      float* hc = h.data_ptr<float>() + (long)c * nt;
+      int* feat_t = feats.data_ptr<int>() + ((long)t * K + c) * NN;
+      int* thr_t = thrs.data_ptr<int>() + ((long)t * K + c) * NN;
+      float* val_t = vals.data_ptr<float>() + ((long)t * K + c) * NN;
+      for (int d = 0; d < (int)max_depth; ++d) {
+        int lvl_lo = (1 << d) - 1;
+        int lvl_nodes = 1 << d;
+        int f_pb = std::min(8, std::max(1, (96 * 1024) /
+                                        (3072 * lvl_nodes)));
+        while (F % f_pb != 0) f_pb /= 2;
+        int fchunks = F / f_pb;
+        long rowsegs = std::max<long>(1, 2048 / fchunks);
+        long rows_pb = (nt + rowsegs - 1) / rowsegs;
+        rowsegs = (nt + rows_pb - 1) / rows_pb;
+        size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 3 * 4;
+        int even_only = (d > 0) ? 1 : 0;
+        HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0,
+                                 (size_t)lvl_nodes * F * 256 * 3 * 4, s));
+        HIP_CHECK(hipFuncSetAttribute(
+            reinterpret_cast<const void*>(gbt_hist_kernel),
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
+        hipLaunchKernelGGL(gbt_hist_kernel,
+                           dim3((unsigned)rowsegs, (unsigned)fchunks),
+                           dim3(256), lsm, s, Xp, idxmap.data_ptr<int>(),
+                           node_id.data_ptr<int>(), gc, hc,
+                           hist.data_ptr<float>(), nt, F, lvl_lo,
+                           lvl_nodes, f_pb, rows_pb, even_only);
+        if (even_only) {
+          int prev_lo = (1 << (d - 1)) - 1;
+          long cells = (long)F * 256 * 3;
+          unsigned gx = (unsigned)std::min<long>((cells + 255) / 256,
+                                                 1024);
+          hipLaunchKernelGGL(gbt_sibling_kernel,
+                             dim3(gx, (unsigned)(lvl_nodes / 2)),
+                             dim3(256), 0, s, hist_prev.data_ptr<float>(),
+                             hist.data_ptr<float>(), feat_t, prev_lo, F,
+                             lvl_nodes);
+        }
+        hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes),
+                           dim3(256), 0, s, hist.data_ptr<float>(), F,
+                           lvl_lo, (float)lam, (float)lrate,
+                           (int)min_leaf, 0, feat_t, thr_t, val_t);
+        if (d + 1 < (int)max_depth)
+          hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
+              .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
+                     true);
+        hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
+                           0, s, Xp, idxmap.data_ptr<int>(),
+                           node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
+                           lvl_lo, lvl_lo + lvl_nodes);
+        HIP_CHECK(hipGetLastError());
+      }
+      int bl_lo = (1 << max_depth) - 1;
+      HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
+      HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
+      hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0,
+                         s, node_id.data_ptr<int>(), gc, hc,
+                         gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                         nt, bl_lo, NN);
+      hipLaunchKernelGGL(gbt_leaf_vals_kernel,
+                         dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
+                         gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                         val_t, feat_t, (float)lam, (float)lrate, bl_lo,
+                         NN);
+      hipLaunchKernelGGL(gbt_leaf_apply_c_kernel, dim3(blocks), dim3(256),
+                         0, s, raw.data_ptr<float>() + (long)c * nt,
+                         node_id.data_ptr<int>(), val_t, nt);
+      HIP_CHECK(hipGetLastError());
+    }
+  }
+  return {feats, thrs, vals};
+}
+
+torch::Tensor gbt_metric_multi(torch::Tensor raws, torch::Tensor y,
+                               torch::Tensor idx, int64_t K) {
+  const at::cuda::CUDAGuard guard(raws.device());
+  long n_idx = idx.size(0);
+  auto out = torch::zeros({1}, raws.options().dtype(torch::kFloat64));
+  int blocks = (int)std::min<long>((n_idx + 255) / 256, 2048);
+  hipLaunchKernelGGL(gbt_argmax_acc_kernel, dim3(blocks), dim3(256), 0,
+                     cur_stream(), raws.data_ptr<float>(),
+                     y.data_ptr<float>(), idx.data_ptr<int>(), n_idx,
+                     (int)K, out.data_ptr<double>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
@@ -3366,4 +3555,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "CSR-resident batched linear fit (K5)");
   m.def("csr_linear_scores", &csr_linear_scores,
         "CSR-resident per-candidate scores");
+  m.def("gbt_fit_multi", &gbt_fit_multi,
+        "multiclass softmax boosting (one tree per class per round)");
+  m.def("gbt_metric_multi", &gbt_metric_multi,
+        "argmax accuracy over class-major raw planes");
 }

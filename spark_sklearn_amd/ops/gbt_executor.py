@@ -17,11 +17,12 @@ class DeviceGBTExecutor:
     matrix.  Quantile binning runs once per (matrix, device) on the host;
     the uint8 bin matrix then lives in HBM for the whole search."""
 
-    def __init__(self, X, y_enc, device):
+    def __init__(self, X, y_enc, device, classes=None):
         import torch
 
         self.torch = torch
         self.device = device
+        self.K = len(classes) if classes is not None else 0
         self.ext = load_extension()
         key = (id(X), getattr(X, "shape", None), str(device))
         hit = _bin_cache.get(key)
@@ -59,11 +60,31 @@ class DeviceGBTExecutor:
 
         models_host, models_dev = [], []
         n_iter = np.empty(plan.B, np.int32)
+        K = self.K
         for i, spec in enumerate(plan.specs):
             if spec.n_estimators == 0:
                 models_host.append(([], 0.0))
                 models_dev.append(None)
                 n_iter[i] = 0
+                continue
+            if K > 2:
+                prior = np.bincount(y_tr_np.astype(np.int64),
+                                    minlength=K) / len(y_tr_np)
+                f0 = np.log(np.clip(prior, 1e-6, None)).astype(np.float32)
+                f0_t = torch.as_tensor(f0).to(self.device)
+                feats, thrs, vals = self.ext.gbt_fit_multi(
+                    self.Xb, idx_t, y_tr, f0_t, K, spec.n_estimators,
+                    spec.learning_rate, spec.max_depth,
+                    spec.min_samples_leaf, spec.l2_regularization)
+                models_dev.append((feats, thrs, vals, f0, spec))
+                f_np = feats.cpu().numpy()
+                t_np = thrs.cpu().numpy().astype(np.uint8)
+                v_np = vals.cpu().numpy()
+                trees = [[(f_np[t * K + c], t_np[t * K + c],
+                           v_np[t * K + c]) for c in range(K)]
+                         for t in range(spec.n_estimators)]
+                models_host.append((trees, f0))
+                n_iter[i] = spec.n_estimators
                 continue
             f0 = self._f0(spec.loss, y_tr_np)
             feats, thrs, vals = self.ext.gbt_fit(
@@ -97,6 +118,24 @@ class DeviceGBTExecutor:
         for i, m in enumerate(self._models_dev):
             if m is None:
                 out[i] = float("nan")
+                continue
+            if self.K > 2:
+                feats, thrs, vals, f0, spec = m
+                K, T = self.K, spec.n_estimators
+                NN = feats.shape[1]
+                raws = torch.empty((K, n), dtype=torch.float32,
+                                   device=self.device)
+                fv = feats.view(T, K, NN)
+                tv = thrs.view(T, K, NN)
+                vv = vals.view(T, K, NN)
+                for c in range(K):
+                    raws[c] = self.ext.gbt_predict(
+                        self.Xb, idx_t, fv[:, c].contiguous(),
+                        tv[:, c].contiguous(), vv[:, c].contiguous(),
+                        float(f0[c]))
+                sums = self.ext.gbt_metric_multi(raws, self.y, idx_t,
+                                                 self.K)
+                out[i] = float(sums.item()) / n
                 continue
             feats, thrs, vals, f0, loss = m
             raw = self.ext.gbt_predict(self.Xb, idx_t, feats, thrs, vals,

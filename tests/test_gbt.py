@@ -83,11 +83,37 @@ def test_gbt_grid_search(problem):
     assert gs.predict(X[:20]).shape == (20,)
 
 
-def test_multiclass_rejected(problem):
-    X, _ = problem
-    y3 = np.arange(len(X)) % 3
-    with pytest.raises(ValueError, match="binary"):
-        GradientBoostingClassifier(n_estimators=5).fit(X, y3)
+def test_multiclass_softmax(problem):
+    """K>2: one tree per class per round (softmax boosting)."""
+    rng = np.random.default_rng(7)
+    n, f, K = 2500, 10, 4
+    centers = rng.standard_normal((K, f)) * 2.5
+    y = rng.integers(0, K, n)
+    X = (centers[y] + rng.standard_normal((n, f))).astype(np.float32)
+    clf = GradientBoostingClassifier(n_estimators=20, max_depth=3,
+                                     learning_rate=0.3).fit(X, y)
+    assert (clf.predict(X) == y).mean() > 0.95
+    proba = clf.predict_proba(X[:7])
+    assert proba.shape == (7, K)
+    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)
+    # string labels round-trip through classes_
+    labels = np.array(["a", "b", "c", "d"])[y]
+    clf2 = GradientBoostingClassifier(n_estimators=10, max_depth=3,
+                                      learning_rate=0.3).fit(X, labels)
+    assert set(clf2.predict(X[:20])) <= set("abcd")
+
+
+def test_multiclass_grid_search(problem):
+    rng = np.random.default_rng(3)
+    n, f, K = 1500, 8, 3
+    centers = rng.standard_normal((K, f)) * 2
+    y = rng.integers(0, K, n)
+    X = (centers[y] + rng.standard_normal((n, f))).astype(np.float32)
+    gs = ska.GridSearchCV(None, GradientBoostingClassifier(n_estimators=10),
+                          {"max_depth": [2, 3]}, cv=3,
+                          return_train_score=False).fit(X, y)
+    assert gs.best_score_ > 0.9
+    assert gs.predict(X[:5]).shape == (5,)
 
 
 def test_param_validation(problem):

@@ -102,3 +102,60 @@ def test_device_grid_search_end_to_end(problem):
     assert len(gs.cv_results_["params"]) == 4
     assert gs.best_score_ > 0.8
     assert gs.predict(X[:50]).shape == (50,)
+
+
+def test_device_multiclass_matches_reference(problem):
+    """K=3 softmax boosting: device trees vs gbt_fit_ref_multi on the same
+    binned matrix; scores must agree (tree structure may differ only via
+    fp32 sum order in the softmax grads — accuracy must match closely)."""
+    from spark_sklearn_amd.models._gbt_ref import (gbt_fit_ref_multi,
+                                                   gbt_predict_raw_multi)
+
+    rng = np.random.default_rng(11)
+    n, f, K = 20000, 16, 3
+    centers = rng.standard_normal((K, f)) * 2
+    y = rng.integers(0, K, n)
+    X = (centers[y] + rng.standard_normal((n, f))).astype(np.float32)
+
+    dev = torch.device("cuda", 0)
+    classes = np.arange(K)
+    ex = DeviceGBTExecutor(X, y.astype(np.float32), dev, classes=classes)
+    spec = GBTSpec(n_estimators=10, max_depth=3, learning_rate=0.3,
+                   min_samples_leaf=20)
+    plan = _GBTPlan([spec], K)
+    train = np.arange(15000, dtype=np.int64)
+    test = np.arange(15000, n, dtype=np.int64)
+    fitted, n_iter, _ = ex.fit(train, plan)
+    assert n_iter[0] == 10
+    acc_d = ex.scores(test, fitted, plan, "accuracy")[0]
+
+    models, edges = fitted
+    Xb = bin_data(X, edges)
+    trees_r, f0_r = gbt_fit_ref_multi(Xb[train], y[train].astype(np.int64),
+                                      K, spec)
+    raw = gbt_predict_raw_multi(Xb[test], trees_r, f0_r)
+    acc_r = float(np.mean(raw.argmax(axis=1) == y[test]))
+    assert acc_d > 0.9
+    assert abs(acc_d - acc_r) < 0.01
+
+    # downloaded host-format trees predict the same as device scoring
+    trees_h, f0_h = models[0]
+    raw_h = gbt_predict_raw_multi(Xb[test], trees_h, f0_h)
+    acc_h = float(np.mean(raw_h.argmax(axis=1) == y[test]))
+    assert abs(acc_h - acc_d) < 1e-6
+
+
+def test_device_multiclass_grid_search(problem):
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import GradientBoostingClassifier
+
+    rng = np.random.default_rng(5)
+    n, f, K = 12000, 12, 4
+    centers = rng.standard_normal((K, f)) * 2
+    y = rng.integers(0, K, n)
+    X = (centers[y] + rng.standard_normal((n, f))).astype(np.float32)
+    gs = ska.GridSearchCV(
+        None, GradientBoostingClassifier(n_estimators=12),
+        {"max_depth": [2, 3]}, cv=3, return_train_score=False).fit(X, y)
+    assert gs.best_score_ > 0.9
+    assert gs.predict(X[:9]).shape == (9,)
