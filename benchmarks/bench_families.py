@@ -6,6 +6,7 @@ Prints one JSON line per family.  Run on an MI355X box:
 """
 
 import json
+import os
 import sys
 import time
 
@@ -61,6 +62,32 @@ def bench_gbt():
             "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
 
 
+def bench_gbt_multi():
+    """Multiclass GBT: one tree per class per round (softmax boosting)."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import GradientBoostingClassifier
+
+    rng = np.random.default_rng(0)
+    n, f, K = 1_000_000, 64, 8
+    centers = rng.standard_normal((K, f)).astype(np.float32) * 1.5
+    y = rng.integers(0, K, n)
+    X = centers[y] + rng.standard_normal((n, f), dtype=np.float32)
+
+    gs = ska.GridSearchCV(None,
+                          GradientBoostingClassifier(n_estimators=20),
+                          {"max_depth": [3, 5]}, cv=3,
+                          return_train_score=False)
+    t0 = time.perf_counter()
+    gs.fit(X, y)
+    dt = time.perf_counter() - t0
+    trees = gs.n_fits_ * 20 * K
+    return {"family": "gbt-multiclass",
+            "config": f"1Mx{f}, K={K}, 20 rounds, depth 3/5, cv=3",
+            "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
+            "trees_per_sec": round(trees / dt, 1),
+            "best_acc": round(gs.best_score_, 4)}
+
+
 def bench_sweep():
     """Config 5: large randomized sweep with successive-halving pruning."""
     from scipy.stats import loguniform
@@ -75,10 +102,11 @@ def bench_sweep():
     y = ((X @ w + 0.25 * rng.standard_normal(n, dtype=np.float32)) > 0
          ).astype(np.int64)
 
+    n_iter = int(os.environ.get("SWEEP_N", "10000"))
     rs = ska.RandomizedSearchCV(
         None, LogisticRegression(max_iter=25, tol=0.0),
         {"C": loguniform(1e-4, 1e3), "penalty": ["l1", "l2"]},
-        n_iter=10_000, cv=5, random_state=0,
+        n_iter=n_iter, cv=5, random_state=0,
         halving_factor=4, min_resource_frac=0.12,
         return_train_score=False)
     t0 = time.perf_counter()
@@ -86,12 +114,12 @@ def bench_sweep():
     dt = time.perf_counter() - t0
     res = rs.cv_results_
     return {"family": "randomized+halving",
-            "config": "10k candidates, 1Mx256, cv=5, factor 4",
+            "config": f"{n_iter} candidates, 1Mx256, cv=5, factor 4",
             "wall_s": round(dt, 2),
             "final_rung_candidates": int(np.sum(
                 res["resource_iter"] == 25)),
             "best_score": round(rs.best_score_, 4),
-            "candidates_per_sec": round(10_000 / dt, 1)}
+            "candidates_per_sec": round(n_iter / dt, 1)}
 
 
 def bench_sparse():
@@ -131,5 +159,6 @@ if __name__ == "__main__":
     which = sys.argv[1:] or ["kmeans", "gbt", "sweep"]
     for name in which:
         fn = {"kmeans": bench_kmeans, "gbt": bench_gbt,
-              "sweep": bench_sweep, "sparse": bench_sparse}[name]
+              "sweep": bench_sweep, "sparse": bench_sparse,
+              "gbtmulti": bench_gbt_multi}[name]
         print(json.dumps(fn()), flush=True)
