@@ -25,6 +25,7 @@ error_score contract are shared.
 
 from __future__ import annotations
 
+import os
 import time
 import warnings
 from typing import Dict, List, Optional, Sequence
@@ -339,10 +340,16 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
                 for a, b in zip(first, by_fold[f])):
             return None
 
+    dbg = os.environ.get("SKA_TIME") == "1"
+    if dbg:
+        torch.cuda.synchronize()
+        _t0 = time.perf_counter()
     params_list = [t.params for t in first]
     plan, bad = LinearFamily.build_plan(
         estimator, params_list, len(first[0].train), executor,
         n_classes, error_score)
+    if dbg:
+        _t1 = time.perf_counter()
 
     cand_chunk = 256 // C
     n_chunks = (n_cand + cand_chunk - 1) // cand_chunk
@@ -381,6 +388,9 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
 
     from .ops.device_executor import _METRIC_ID
 
+    if dbg:
+        torch.cuda.synchronize()
+        _t2 = time.perf_counter()
     t0 = time.perf_counter()
     out = executor.ext.linear_search_run(
         executor.X, executor.Xt, executor.y, train_ts, nts, test_ts,
@@ -388,6 +398,9 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
         C, _METRIC_ID[metric], n_cand, 1 if return_train_score else 0,
         32)
     test_sums, train_sums, fit_ms, score_ms = out[0], out[1], out[2], out[3]
+    if dbg:
+        torch.cuda.synchronize()
+        _t3 = time.perf_counter()
     test_sums = test_sums.cpu().numpy()
     train_sums = train_sums.cpu().numpy() if return_train_score else None
     fit_ms = fit_ms.numpy()
@@ -431,6 +444,12 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
                 err = None
             results.append(TaskResult(task.index, v, tv, len(task.test),
                                       per_fit, per_score, error=err))
+    if dbg:
+        _t4 = time.perf_counter()
+        print(f"[ska-time] plan_build {1000*(_t1-_t0):.2f}ms "
+              f"plan_upload {1000*(_t2-_t1):.2f}ms "
+              f"native {1000*(_t3-_t2):.2f}ms "
+              f"assemble {1000*(_t4-_t3):.2f}ms", flush=True)
     if verbose > 1:
         print(f"[engine] native search_run: {len(folds)} folds x "
               f"{n_chunks} chunks in {time.perf_counter() - t0:.2f}s")

@@ -1481,9 +1481,13 @@ __global__ void gbt_grad_kernel(const float* __restrict__ raw,
   }
 }
 
-// per-(node, feature, bin) g/h/count histograms in LDS, flushed with
-// global atomics.  Rows keep a heap node_id; settled rows (node at an
-// earlier level) are skipped by the level range check.
+// per-(node, feature, bin) histograms in LDS: g as float atomics, (h,
+// count) packed into ONE u64 atomic -- h fixed-point in the high half
+// (hscale chosen on the host so the sum cannot overflow 32 bits), exact
+// count in the low half.  2 LDS atomics per (row, feature) instead of 3;
+// LDS atomic throughput is the bound (profiles/r01_families_kernels.md).
+// Rows keep a heap node_id; settled rows are skipped by the level range
+// check; odd nodes are skipped under the sibling-subtraction trick.
 __global__ __launch_bounds__(256)
 void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const int* __restrict__ idxmap,          // [nt]
@@ -1492,42 +1496,96 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const float* __restrict__ h,
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
                      long nt, int F, int lvl_lo, int lvl_nodes, int f_pb,
-                     long rows_pb, int even_only) {
-  extern __shared__ float lh[];     // [lvl_nodes][f_pb][256][3]
+                     long rows_pb, int even_only, float hscale) {
+  extern __shared__ char smem[];
+  const int cellsg = lvl_nodes * f_pb * 256;
+  unsigned long long* lhc = reinterpret_cast<unsigned long long*>(smem);
+  float* lg = reinterpret_cast<float*>(smem + (size_t)cellsg * 8);
   const int tid = threadIdx.x;
   const int f0 = blockIdx.y * f_pb;
   const long r0 = (long)blockIdx.x * rows_pb;
   const long r1 = min(r0 + rows_pb, nt);
-  const int cells = lvl_nodes * f_pb * 256 * 3;
 
-  for (int i = tid; i < cells; i += 256) lh[i] = 0.f;
+  for (int i = tid; i < cellsg; i += 256) { lhc[i] = 0ull; lg[i] = 0.f; }
   __syncthreads();
 
+  const bool vec8 = (F % 8 == 0) && (f_pb == 8);
   for (long r = r0 + tid; r < r1; r += 256) {
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
     if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
     long gi = idxmap[r];
-    float gr = g[r], hr = h[r];
+    float gr = g[r];
+    unsigned hfx = __float2uint_rn(h[r] * hscale);
+    unsigned long long hpack = ((unsigned long long)hfx << 32) | 1ull;
+    const int base = (nd - lvl_lo) * f_pb;
     const unsigned char* xrow = Xb + gi * F + f0;
-    for (int ff = 0; ff < f_pb; ++ff) {
-      int bin = xrow[ff];
-      float* cell = &lh[(((nd - lvl_lo) * f_pb + ff) * 256 + bin) * 3];
-      atomicAdd(cell, gr);
-      atomicAdd(cell + 1, hr);
-      atomicAdd(cell + 2, 1.f);
+    if (vec8) {
+      unsigned long long bv =
+          *reinterpret_cast<const unsigned long long*>(xrow);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int cell = (base + j) * 256 + (int)((bv >> (8 * j)) & 0xff);
+        atomicAdd(&lg[cell], gr);
+        atomicAdd(&lhc[cell], hpack);
+      }
+    } else {
+      for (int ff = 0; ff < f_pb; ++ff) {
+        int cell = (base + ff) * 256 + (int)xrow[ff];
+        atomicAdd(&lg[cell], gr);
+        atomicAdd(&lhc[cell], hpack);
+      }
     }
   }
   __syncthreads();
-  for (int i = tid; i < cells; i += 256) {
-    float v = lh[i];
-    if (v == 0.f) continue;
-    int c = i / 3, ch = i % 3;
-    int bin = c % 256;
-    int ff = (c / 256) % f_pb;
-    int ndl = c / (256 * f_pb);
-    atomicAdd(&hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3 + ch], v);
+  const float hinv = 1.f / hscale;
+  for (int i = tid; i < cellsg; i += 256) {
+    unsigned long long p = lhc[i];
+    if (p == 0ull) continue;          // count==0 -> cell never touched
+    float hv = (float)(p >> 32) * hinv;
+    unsigned cnt = (unsigned)(p & 0xffffffffull);
+    int bin = i & 255;
+    int ff = (i >> 8) % f_pb;
+    int ndl = i / (256 * f_pb);
+    float* cell = &hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
+    atomicAdd(cell, lg[i]);
+    atomicAdd(cell + 1, hv);
+    atomicAdd(cell + 2, (float)cnt);
   }
+}
+
+// host helper: one level's histogram launch, shared by gbt_fit and
+// gbt_fit_multi.  Picks features-per-block to fit 96 KB LDS, row
+// segments >=8192 rows (amortizes the global flush), and the largest
+// power-of-two h scale whose worst-case sum stays under 2^31.
+static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
+                              const int* node_id, const float* g,
+                              const float* h, float* hist, long nt, int F,
+                              int d, float hmax, hipStream_t s) {
+  int lvl_lo = (1 << d) - 1;
+  int lvl_nodes = 1 << d;
+  int f_pb = std::min(8, std::max(1, (96 * 1024) / (3072 * lvl_nodes)));
+  while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
+  int fchunks = F / f_pb;
+  long rowsegs = std::max<long>(
+      1, std::min<long>(2048 / fchunks, (nt + 8191) / 8192));
+  long rows_pb = (nt + rowsegs - 1) / rowsegs;
+  rowsegs = (nt + rows_pb - 1) / rows_pb;
+  size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 12;
+  int even_only = (d > 0) ? 1 : 0;       // sibling subtraction
+  float hs = 1.f;
+  while (hs * 2.f * hmax * (float)nt + (float)nt < 1.8e9f &&
+         hs < 65536.f)
+    hs *= 2.f;
+  HIP_CHECK(hipMemsetAsync(hist, 0,
+                           (size_t)lvl_nodes * F * 256 * 3 * 4, s));
+  HIP_CHECK(hipFuncSetAttribute(
+      reinterpret_cast<const void*>(gbt_hist_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
+  hipLaunchKernelGGL(gbt_hist_kernel,
+                     dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(256),
+                     lsm, s, Xp, idxmap, node_id, g, h, hist, nt, F,
+                     lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -2932,25 +2990,11 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
     for (int d = 0; d < (int)max_depth; ++d) {
       int lvl_lo = (1 << d) - 1;
       int lvl_nodes = 1 << d;
-      int f_pb = std::min(8, std::max(1, (96 * 1024) / (3072 * lvl_nodes)));
-      while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
-      int fchunks = F / f_pb;
-      long rowsegs = std::max<long>(1, 2048 / fchunks);
-      long rows_pb = (nt + rowsegs - 1) / rowsegs;
-      rowsegs = (nt + rows_pb - 1) / rows_pb;
-      size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 3 * 4;
       int even_only = (d > 0) ? 1 : 0;    // sibling subtraction
-      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0,
-                               (size_t)lvl_nodes * F * 256 * 3 * 4, s));
-      HIP_CHECK(hipFuncSetAttribute(
-          reinterpret_cast<const void*>(gbt_hist_kernel),
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
-      hipLaunchKernelGGL(gbt_hist_kernel,
-                         dim3((unsigned)rowsegs, (unsigned)fchunks),
-                         dim3(256), lsm, s, Xp, idxmap.data_ptr<int>(),
-                         node_id.data_ptr<int>(), g.data_ptr<float>(),
-                         h.data_ptr<float>(), hist.data_ptr<float>(), nt,
-                         F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only);
+      launch_hist_level(Xp, idxmap.data_ptr<int>(),
+                        node_id.data_ptr<int>(), g.data_ptr<float>(),
+                        h.data_ptr<float>(), hist.data_ptr<float>(), nt,
+                        F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f, s);
       if (even_only) {
         int prev_lo = (1 << (d - 1)) - 1;
         long cells = (long)F * 256 * 3;
@@ -3456,26 +3500,10 @@ std::vector<torch::Tensor> gbt_fit_multi(
       for (int d = 0; d < (int)max_depth; ++d) {
         int lvl_lo = (1 << d) - 1;
         int lvl_nodes = 1 << d;
-        int f_pb = std::min(8, std::max(1, (96 * 1024) /
-                                        (3072 * lvl_nodes)));
-        while (F % f_pb != 0) f_pb /= 2;
-        int fchunks = F / f_pb;
-        long rowsegs = std::max<long>(1, 2048 / fchunks);
-        long rows_pb = (nt + rowsegs - 1) / rowsegs;
-        rowsegs = (nt + rows_pb - 1) / rows_pb;
-        size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 3 * 4;
         int even_only = (d > 0) ? 1 : 0;
-        HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0,
-                                 (size_t)lvl_nodes * F * 256 * 3 * 4, s));
-        HIP_CHECK(hipFuncSetAttribute(
-            reinterpret_cast<const void*>(gbt_hist_kernel),
-            hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
-        hipLaunchKernelGGL(gbt_hist_kernel,
-                           dim3((unsigned)rowsegs, (unsigned)fchunks),
-                           dim3(256), lsm, s, Xp, idxmap.data_ptr<int>(),
-                           node_id.data_ptr<int>(), gc, hc,
-                           hist.data_ptr<float>(), nt, F, lvl_lo,
-                           lvl_nodes, f_pb, rows_pb, even_only);
+        launch_hist_level(Xp, idxmap.data_ptr<int>(),
+                          node_id.data_ptr<int>(), gc, hc,
+                          hist.data_ptr<float>(), nt, F, d, 0.25f, s);
         if (even_only) {
           int prev_lo = (1 << (d - 1)) - 1;
           long cells = (long)F * 256 * 3;
