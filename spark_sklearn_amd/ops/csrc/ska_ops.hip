@@ -1837,18 +1837,11 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   const int wave = tid >> 6;
   const int cwave = wave & 3;
   const int rquad = wave >> 2;        // row quad 0..3
+  const int rt = blockIdx.x;
   const int g = lane >> 4;
   const int srow = tid >> 2;          // staging row/col 0..255
   const int sseg = tid & 3;
-  const int n_rt = NTp >> 8;
 
-  // grid-stride over row tiles: each block runs many tiles back to back,
-  // so the staging pipeline reaches steady state and the per-tile
-  // prologue/epilogue amortizes (4 k-chunks alone never spin it up);
-  // consecutive tiles of one block also reuse the same W from their
-  // XCD's L2.
-  for (int rt = blockIdx.x; rt < n_rt; rt += gridDim.x) {
-  __syncthreads();                    // prior tile done with sY/sDb
   if (tid < 256) {
     int gi = idx[rt * 256 + tid];
     sIdx[tid] = gi;
@@ -1964,7 +1957,6 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
     atomicAdd(&db_sum[tid], sDb[tid]);
   }
-  }   // rt loop
 }
 
 // ---------------------------------------------------------------------------
@@ -2552,8 +2544,7 @@ std::vector<torch::Tensor> linear_fit(
     if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
       if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
         // 1024-thread fwd (256x256 tile, 16 waves/CU)
-        hipLaunchKernelGGL(fwd_d7_kernel,
-                           dim3((unsigned)std::min<long>(NTp / 256, 256)),
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
                            dim3(1024), 0, s, Xp, y.data_ptr<float>(),
                            idx.data_ptr<int>(),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
@@ -3211,8 +3202,7 @@ std::vector<torch::Tensor> linear_search_run(
 
       for (int epoch = 0; epoch < max_ep; ++epoch) {
         int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
-        hipLaunchKernelGGL(fwd_d7_kernel,
-                           dim3((unsigned)std::min<long>(NTp / 256, 256)),
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
                            dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                            bias.data_ptr<float>(), loss_p, tgt_p,
