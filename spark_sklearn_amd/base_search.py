@@ -107,6 +107,37 @@ class TaskResult:
         self.error = error     # formatted exception text for failed fits
 
 
+class _MultiScorer:
+    """Dispatchable wrapper for multi-metric scoring: calls every scorer
+    of the dict and returns ``{name: float}``.  Backends treat it exactly
+    like a single scorer; the dict-valued scores flow through TaskResult
+    into the per-metric ``cv_results_`` assembly (sklearn multimetric
+    contract -- a capability the reference era predates)."""
+
+    def __init__(self, scorers):
+        self.scorers = dict(scorers)
+
+    def keys(self):
+        return self.scorers.keys()
+
+    def __call__(self, est, X, y=None):
+        if y is None:
+            return {k: float(sc(est, X)) for k, sc in self.scorers.items()}
+        return {k: float(sc(est, X, y)) for k, sc in self.scorers.items()}
+
+
+def _coerce_score(v):
+    """float for single-metric, {name: float} for multimetric scorers."""
+    if isinstance(v, dict):
+        for k, val in v.items():
+            if not isinstance(val, numbers.Number):
+                raise ValueError(
+                    f"scoring must return a number, got {val!r} "
+                    f"(key {k!r}) instead.")
+        return {k: float(val) for k, val in v.items()}
+    return float(v)
+
+
 def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
                        fit_params=None, return_train_score=True,
                        error_score="raise") -> TaskResult:
@@ -179,7 +210,10 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
                           n_test, fit_time, 0.0, error=err)
 
     n_test = len(task.test)
-    return TaskResult(task.index, float(test_score), float(train_score),
+    return TaskResult(task.index, _coerce_score(test_score),
+                      _coerce_score(train_score)
+                      if isinstance(train_score, dict)
+                      else float(train_score),
                       n_test, fit_time, score_time)
 
 
@@ -255,18 +289,18 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         estimator = self.estimator
         cv = check_cv(self.cv, y, classifier=is_classifier(estimator))
         if isinstance(self.scoring, (dict, list, tuple, set)):
-            if self.refit:
-                raise ValueError(
-                    "For multi-metric scoring, the parameter refit must be "
-                    "set to a scorer key or a callable to refit an "
-                    "estimator with the best parameter setting on the "
-                    "whole data and make the best_* attributes available "
-                    "for that metric. If this is not needed, refit should "
-                    "be set to False explicitly.")
-            raise NotImplementedError(
-                "multi-metric scoring is not supported by this engine yet; "
-                "run one search per metric")
-        self.scorer_ = check_scoring(estimator, scoring=self.scoring)
+            from sklearn.metrics._scorer import _check_multimetric_scoring
+
+            scoring = self.scoring
+            if isinstance(scoring, (tuple, set)):
+                scoring = list(scoring)
+            scorers_dict = _check_multimetric_scoring(estimator, scoring)
+            self._validate_multimetric_refit(scorers_dict.keys())
+            self.scorer_ = scorers_dict
+            dispatch_scorer = _MultiScorer(scorers_dict)
+        else:
+            self.scorer_ = check_scoring(estimator, scoring=self.scoring)
+            dispatch_scorer = self.scorer_
 
         X, y, groups = indexable(X, y, groups)
         merged_fit_params = dict(self.fit_params or {})
@@ -303,7 +337,7 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         t_run0 = time.perf_counter()
         results = backend.run_search(
             estimator, X, y, tasks,
-            scorer=self.scorer_,
+            scorer=dispatch_scorer,
             fit_params=merged_fit_params,
             return_train_score=self.return_train_score,
             error_score=self.error_score,
@@ -333,10 +367,21 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         if score_log and getattr(backend, "rank", 0) == 0:
             self._append_score_log(score_log, tasks, ordered)
 
+        # multimetric-ness is declared (dict/list scoring) or discovered
+        # (a plain callable that returned a dict of scores)
+        first_ok = next((r for r in ordered if r.error is None), None)
+        metric_names = None
+        if first_ok is not None and isinstance(first_ok.test_score, dict):
+            metric_names = list(first_ok.test_score.keys())
+            self._validate_multimetric_refit(metric_names)
+
         self.cv_results_ = self._assemble_cv_results(
-            candidates, ordered, n_candidates, n_splits)
+            candidates, ordered, n_candidates, n_splits,
+            metric_names=metric_names)
         self.n_splits_ = n_splits
-        self.multimetric_ = False
+        self.multimetric_ = metric_names is not None
+        refit_metric = self.refit if (self.multimetric_ and
+                                      isinstance(self.refit, str))             else "score"
         if callable(self.refit):
             # sklearn contract: refit may be a callable selecting
             # best_index_ from cv_results_; best_score_ is then undefined
@@ -346,12 +391,14 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             if not 0 <= best < n_candidates:
                 raise IndexError("best_index_ index out of range")
             self.best_index_ = int(best)
-        else:
-            self.best_index_ = int(
-                np.flatnonzero(self.cv_results_["rank_test_score"] == 1)[0])
+            self.best_params_ = self.cv_results_["params"][self.best_index_]
+        elif self.refit or not self.multimetric_:
+            self.best_index_ = int(np.flatnonzero(
+                self.cv_results_[f"rank_test_{refit_metric}"] == 1)[0])
             self.best_score_ = float(
-                self.cv_results_["mean_test_score"][self.best_index_])
-        self.best_params_ = self.cv_results_["params"][self.best_index_]
+                self.cv_results_[f"mean_test_{refit_metric}"]
+                [self.best_index_])
+            self.best_params_ = self.cv_results_["params"][self.best_index_]
 
         if self.refit:
             best_estimator = clone(estimator).set_params(**self.best_params_)
@@ -367,6 +414,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self.refit_time_ = time.perf_counter() - t0
             self.best_estimator_ = best_estimator
         return self
+
+    def _validate_multimetric_refit(self, keys):
+        """refit must be False, a callable, or a key of the scorer dict
+        (sklearn multimetric contract, same message)."""
+        if self.refit is False or self.refit is None or                 callable(self.refit):
+            return
+        if isinstance(self.refit, str) and self.refit in set(keys):
+            return
+        raise ValueError(
+            "For multi-metric scoring, the parameter refit must be set to "
+            "a scorer key or a callable to refit an estimator with the "
+            "best parameter setting on the whole data and make the best_* "
+            "attributes available for that metric. If this is not needed, "
+            "refit should be set to False explicitly.")
 
     def _handle_fit_failures(self, tasks, ordered):
         """sklearn failure contract: one aggregated FitFailedWarning when
@@ -417,7 +478,8 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
 
     # -- cv_results_ assembly ---------------------------------------------
     def _assemble_cv_results(self, candidates, ordered_results,
-                             n_candidates, n_splits) -> Dict[str, Any]:
+                             n_candidates, n_splits,
+                             metric_names=None) -> Dict[str, Any]:
         """Build the sklearn-layout ``cv_results_`` dict.
 
         Exact key set: ``mean_fit_time``,
@@ -425,10 +487,18 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         ``split{k}_test_score``, ``mean_test_score``, ``std_test_score``,
         ``rank_test_score``, train-score variants iff
         ``return_train_score``, masked ``param_<p>`` arrays, ``params``.
+        Multimetric (``metric_names``): ``*_test_score`` keys become
+        ``*_test_<name>`` per metric, exactly as sklearn lays them out.
         """
-        test_scores = np.array(
-            [r.test_score for r in ordered_results], dtype=np.float64
-        ).reshape(n_candidates, n_splits)
+
+        def _metric_matrix(get, name):
+            # dict scores from a multimetric scorer; failed fits carry the
+            # scalar error_score, broadcast to every metric
+            vals = [get(r) for r in ordered_results]
+            return np.array(
+                [v[name] if isinstance(v, dict) else v for v in vals],
+                dtype=np.float64).reshape(n_candidates, n_splits)
+
         fit_times = np.array(
             [r.fit_time for r in ordered_results], dtype=np.float64
         ).reshape(n_candidates, n_splits)
@@ -469,13 +539,14 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         score_weights = n_test if self.iid else None
         _store("fit_time", fit_times)
         _store("score_time", score_times)
-        _store("test_score", test_scores, splits=True, rank=True,
-               weights=score_weights)
-        if self.return_train_score:
-            train_scores = np.array(
-                [r.train_score for r in ordered_results], dtype=np.float64
-            ).reshape(n_candidates, n_splits)
-            _store("train_score", train_scores, splits=True)
+        for mname in (metric_names or ["score"]):
+            _store(f"test_{mname}",
+                   _metric_matrix(lambda r: r.test_score, mname),
+                   splits=True, rank=True, weights=score_weights)
+            if self.return_train_score:
+                _store(f"train_{mname}",
+                       _metric_matrix(lambda r: r.train_score, mname),
+                       splits=True)
 
         # masked param_* arrays: mask=True where a candidate lacks that
         # param (heterogeneous grids) -- reference grid_search.py ~L270 [R].
@@ -539,6 +610,13 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
     def score(self, X, y=None):
         if hasattr(self, "scorer_") and self.scorer_ is not None and \
                 hasattr(self, "best_estimator_"):
+            if getattr(self, "multimetric_", False) and \
+                    isinstance(self.refit, str):
+                if isinstance(self.scorer_, dict):
+                    return self.scorer_[self.refit](
+                        self.best_estimator_, X, y)
+                # plain callable that returns a dict of scores
+                return self.scorer_(self.best_estimator_, X, y)[self.refit]
             return self.scorer_(self.best_estimator_, X, y)
         self._check_refit_for("score")
         return self.best_estimator_.score(X, y)

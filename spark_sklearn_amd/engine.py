@@ -344,8 +344,12 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                            f"(candidate {task.params!r} failed plan "
                            f"validation)")
                 else:
-                    ts = float(test_scores[slot])
-                    tr = float(train_scores[slot])
+                    from .base_search import _coerce_score
+
+                    ts = _coerce_score(test_scores[slot])
+                    tr = _coerce_score(train_scores[slot]) \
+                        if isinstance(train_scores[slot], dict) \
+                        else float(train_scores[slot])
                 results.append(TaskResult(task.index, ts, tr,
                                           len(task.test), per_fit,
                                           per_score, error=err))
@@ -504,17 +508,23 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
 
 def _host_score_fallback(family, estimator, X, y, fitted, plan, chunk,
                          scorer, classes, train, test, return_train_score):
-    """Custom scorer: build fitted clones on the host and score each."""
+    """Custom scorer: build fitted clones on the host and score each.
+    Scores may be floats or (multimetric) dicts -- returned as lists."""
     X = np.asarray(X) if not hasattr(X, "cpu") else X.cpu().float().numpy()
-    y = np.asarray(y)
-    test_scores = np.empty(len(chunk))
-    train_scores = np.full(len(chunk), np.nan)
+    y = np.asarray(y) if y is not None else None
+    test_scores = [np.nan] * len(chunk)
+    train_scores = [np.nan] * len(chunk)
     for slot, task in enumerate(chunk):
         est = clone(estimator).set_params(**task.params)
         est = family.install_fitted(est, fitted, slot, plan, classes)
-        test_scores[slot] = scorer(est, X[test], y[test])
-        if return_train_score:
-            train_scores[slot] = scorer(est, X[train], y[train])
+        if y is None:
+            test_scores[slot] = scorer(est, X[test])
+            if return_train_score:
+                train_scores[slot] = scorer(est, X[train])
+        else:
+            test_scores[slot] = scorer(est, X[test], y[test])
+            if return_train_score:
+                train_scores[slot] = scorer(est, X[train], y[train])
     return test_scores, train_scores
 
 

@@ -120,29 +120,58 @@ class DistributedBackend(Backend):
             return_train_score=return_train_score, error_score=error_score,
             verbose=verbose, n_threads=self.n_threads, device=self.device)
 
-        # RC2: dense (n_tasks, 6) float64 table, one sum-all-reduce.
+        # RC2: dense float64 score table, one sum-all-reduce.  Single
+        # metric: (n_tasks, 7) fixed-width records.  Multimetric: the
+        # test/train columns widen to M metrics each; the metric-name
+        # order is agreed across ranks first (sorted keys of any rank's
+        # dict-valued scores -- ranks whose fits all failed have none).
         comm_dev = self.device if (self.device is not None and
                                    self.device.type == "cuda" and
                                    dist.get_backend() == "nccl") else "cpu"
-        table = torch.zeros((len(tasks), 7), dtype=torch.float64,
+        local_keys = next(
+            (sorted(r.test_score) for r in results
+             if isinstance(r.test_score, dict)), None)
+        gathered = [None] * self.world_size
+        dist.all_gather_object(gathered, local_keys)
+        keys = next((k for k in gathered if k), None)
+
+        def _tvals(v):
+            if keys is None:
+                return [v]
+            if isinstance(v, dict):
+                return [v[k] for k in keys]
+            return [v] * len(keys)     # failed fit: scalar error_score
+
+        M = 1 if keys is None else len(keys)
+        table = torch.zeros((len(tasks), 5 + 2 * M), dtype=torch.float64,
                             device=comm_dev)
         for r in results:
             table[r.index] = torch.tensor(
-                [r.index, r.test_score, r.train_score, r.n_test,
-                 r.fit_time, r.score_time,
+                [r.index] + _tvals(r.test_score) + _tvals(r.train_score) +
+                [r.n_test, r.fit_time, r.score_time,
                  1.0 if r.error is not None else 0.0],
                 dtype=torch.float64)
         dist.all_reduce(table, op=dist.ReduceOp.SUM)
         table = table.cpu().numpy()
         local_err = {r.index: r.error for r in results
                      if r.error is not None}
-        return [TaskResult(int(row[0]), row[1], row[2], int(row[3]),
-                           row[4], row[5],
-                           error=(local_err.get(int(row[0]),
-                                  "fit failed on a remote rank "
-                                  "(error text not gathered)")
-                                  if row[6] > 0 else None))
-                for row in table]
+
+        def _unpack(row, lo, failed):
+            if keys is None or failed:
+                return row[lo]         # scalar (error_score when failed)
+            return dict(zip(keys, row[lo:lo + M]))
+
+        out = []
+        for row in table:
+            failed = row[4 + 2 * M] > 0
+            out.append(TaskResult(
+                int(row[0]), _unpack(row, 1, failed),
+                _unpack(row, 1 + M, failed), int(row[1 + 2 * M]),
+                row[2 + 2 * M], row[3 + 2 * M],
+                error=(local_err.get(int(row[0]),
+                       "fit failed on a remote rank "
+                       "(error text not gathered)") if failed else None)))
+        return out
 
     # -- rank-0 refit + parameter broadcast --------------------------------
     def refit(self, estimator, X, y, fit_params):

@@ -39,6 +39,10 @@ def halving_fit(search, X, y=None, groups=None, **fit_params):
     if factor <= 1.0:
         raise ValueError("halving_factor must be > 1")
     estimator = search.estimator
+    if isinstance(search.scoring, (dict, list, tuple, set)):
+        raise ValueError(
+            "halving pruning supports single-metric scoring only; run "
+            "the unpruned search for multimetric sweeps")
     cv = check_cv(search.cv, y, classifier=is_classifier(estimator))
     search.scorer_ = check_scoring(estimator, scoring=search.scoring)
     X, y, groups = indexable(X, y, groups)

@@ -108,3 +108,89 @@ def test_gloo_world2_search_matches_single_process(tmp_path):
     np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
                                r0["mean_test_score"], rtol=1e-10)
     assert gs.best_params_ == r0["best_params"]
+
+
+MULTI_WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["SKA_REPO"])
+import numpy as np
+import torch.distributed as dist
+import spark_sklearn_amd as ska
+
+rank = int(os.environ["RANK"])
+dist.init_process_group("gloo", init_method=os.environ["SKA_INIT"],
+                        rank=rank, world_size=int(os.environ["WORLD_SIZE"]))
+
+rng = np.random.default_rng(1)
+n, f = 400, 12
+X = rng.standard_normal((n, f))
+y = (X[:, 0] + 0.1 * rng.standard_normal(n) > 0).astype(np.int64)
+
+from sklearn.svm import LinearSVC
+gs = ska.GridSearchCV("dist", LinearSVC(),
+                      {"C": [0.01, 0.1, 1.0]}, cv=3,
+                      scoring={"acc": "accuracy",
+                               "ba": "balanced_accuracy"},
+                      refit="acc")
+gs.fit(X, y)
+out = {
+    "rank": rank,
+    "mean_acc": gs.cv_results_["mean_test_acc"].tolist(),
+    "mean_ba": gs.cv_results_["mean_test_ba"].tolist(),
+    "rank_acc": gs.cv_results_["rank_test_acc"].tolist(),
+    "best_params": gs.best_params_,
+    "best_score": gs.best_score_,
+}
+print("RESULT" + json.dumps(out))
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(300)
+def test_gloo_world2_multimetric(tmp_path):
+    """RC2 with widened per-metric score columns: both ranks agree and
+    match a single-process multimetric run."""
+    init_file = tmp_path / "init"
+    script = tmp_path / "worker.py"
+    script.write_text(MULTI_WORKER)
+    env = dict(os.environ,
+               SKA_REPO=str(REPO),
+               SKA_INIT=f"file://{init_file}",
+               WORLD_SIZE="2",
+               MASTER_ADDR="127.0.0.1")
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank))
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=e,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=240)
+        assert p.returncode == 0, f"worker failed:\n{stdout}\n{stderr}"
+        line = [ln for ln in stdout.splitlines()
+                if ln.startswith("RESULT")][0]
+        outs.append(json.loads(line[len("RESULT"):]))
+
+    r0, r1 = sorted(outs, key=lambda o: o["rank"])
+    assert r0["mean_acc"] == r1["mean_acc"]
+    assert r0["mean_ba"] == r1["mean_ba"]
+    assert r0["best_params"] == r1["best_params"]
+
+    import spark_sklearn_amd as ska
+    from sklearn.svm import LinearSVC
+
+    rng = np.random.default_rng(1)
+    n, f = 400, 12
+    X = rng.standard_normal((n, f))
+    y = (X[:, 0] + 0.1 * rng.standard_normal(n) > 0).astype(np.int64)
+    gs = ska.GridSearchCV("sequential", LinearSVC(),
+                          {"C": [0.01, 0.1, 1.0]}, cv=3,
+                          scoring={"acc": "accuracy",
+                                   "ba": "balanced_accuracy"},
+                          refit="acc").fit(X, y)
+    np.testing.assert_allclose(gs.cv_results_["mean_test_acc"],
+                               r0["mean_acc"], rtol=1e-10)
+    np.testing.assert_allclose(gs.cv_results_["mean_test_ba"],
+                               r0["mean_ba"], rtol=1e-10)
+    assert gs.best_params_ == r0["best_params"]

@@ -24,17 +24,6 @@ import spark_sklearn_amd as ska
 # hooks, HTML repr internals, and tests whose wrapper subclass breaks
 # sklearn's get_params introspection (not an engine behavior).
 SKIP = {
-    # multi-metric scoring (dict / list-of-strings / callable-dict)
-    "test_callable_multimetric_confusion_matrix",
-    "test_callable_multimetric_error_failing_clf",
-    "test_callable_multimetric_error_on_invalid_key",
-    "test_callable_multimetric_same_as_list_of_strings",
-    "test_callable_single_metric_same_as_single_string",
-    "test_grid_search_cv_results_multimetric",
-    "test_random_search_cv_results_multimetric",
-    "test_refit_callable_multi_metric",
-    "test_unsupervised_grid_search",
-    "test_no_refit",
     # sklearn-internal extension API (_run_search subclass hook)
     "test_custom_run_search",
     # sklearn HTML repr internals
