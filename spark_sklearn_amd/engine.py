@@ -284,6 +284,17 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     device = _device_or_none(device)
     executor = family.make_executor(X, y_enc, classes, device)
     metric = family.metric_for(scorer, estimator)
+    # multimetric fast path: when EVERY metric of a _MultiScorer maps to a
+    # device metric, score each on-device instead of the host fallback
+    multi_metrics = None
+    if metric is None:
+        from .base_search import _MultiScorer
+
+        if isinstance(scorer, _MultiScorer):
+            mm = {name: family.metric_for(sc, estimator)
+                  for name, sc in scorer.scorers.items()}
+            if all(v is not None for v in mm.values()):
+                multi_metrics = mm
 
     by_fold: Dict[int, List[Task]] = {}
     for t in tasks:
@@ -326,6 +337,20 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                 train_scores = (executor.scores(train, fitted, plan, metric)
                                 if return_train_score else
                                 np.full(len(chunk), np.nan))
+            elif multi_metrics is not None:
+                per_t = {name: executor.scores(test, fitted, plan, m)
+                         for name, m in multi_metrics.items()}
+                test_scores = [
+                    {name: float(per_t[name][i]) for name in per_t}
+                    for i in range(len(chunk))]
+                if return_train_score:
+                    per_r = {name: executor.scores(train, fitted, plan, m)
+                             for name, m in multi_metrics.items()}
+                    train_scores = [
+                        {name: float(per_r[name][i]) for name in per_r}
+                        for i in range(len(chunk))]
+                else:
+                    train_scores = [float("nan")] * len(chunk)
             else:
                 test_scores, train_scores = _host_score_fallback(
                     family, estimator, X, y, fitted, plan, chunk, scorer,

@@ -429,3 +429,28 @@ def test_keyed_batched_fits_match_per_key(dev):
         pred_ref = (Xb @ W_r[:, 0] + b_r[0] > 0).astype(int)
         agree = np.mean(got == pred_ref)
         assert agree > 0.99, (k, agree)
+
+
+def test_multimetric_device_search(problem):
+    """Multimetric scoring on the GPU: every metric device-computable ->
+    per-metric executor.scores; results match two single-metric runs."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import LogisticRegression
+
+    X, y = problem
+    est = LogisticRegression(max_iter=20, tol=0.0)
+    gm = ska.GridSearchCV(None, est, {"C": [0.1, 1.0, 10.0]},
+                          scoring=["accuracy", "neg_log_loss"],
+                          refit="accuracy", cv=3).fit(X, y)
+    assert "mean_test_neg_log_loss" in gm.cv_results_
+    g1 = ska.GridSearchCV(None, est, {"C": [0.1, 1.0, 10.0]},
+                          scoring="accuracy", cv=3).fit(X, y)
+    np.testing.assert_allclose(gm.cv_results_["mean_test_accuracy"],
+                               g1.cv_results_["mean_test_score"],
+                               rtol=1e-9)
+    g2 = ska.GridSearchCV(None, est, {"C": [0.1, 1.0, 10.0]},
+                          scoring="neg_log_loss", cv=3).fit(X, y)
+    np.testing.assert_allclose(gm.cv_results_["mean_test_neg_log_loss"],
+                               g2.cv_results_["mean_test_score"],
+                               rtol=1e-9)
+    assert gm.best_score_ == g1.best_score_

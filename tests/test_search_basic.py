@@ -136,3 +136,40 @@ def test_native_custom_scorer_fallback():
                           scoring="f1").fit(X, y)
     assert 0.0 <= gs.best_score_ <= 1.0
     assert gs.best_score_ > 0.7
+
+
+def test_engine_multimetric_device_metric_loop():
+    """run_batched_tasks with a _MultiScorer whose metrics are ALL
+    device-computable scores each metric through the executor (host tier
+    here; the identical code path runs the HIP executor on GPU)."""
+    from sklearn.metrics import check_scoring
+    from sklearn.model_selection import KFold
+
+    from spark_sklearn_amd.base_search import Task, _MultiScorer
+    from spark_sklearn_amd.engine import run_batched_tasks
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((400, 8)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+    est = LogisticRegression(max_iter=25)
+    scorers = {"acc": check_scoring(est, scoring="accuracy"),
+               "nll": check_scoring(est, scoring="neg_log_loss")}
+    ms = _MultiScorer(scorers)
+    splits = list(KFold(3).split(np.arange(len(X))))
+    tasks = []
+    i = 0
+    for ci, params in enumerate([{"C": 0.1}, {"C": 1.0}]):
+        for fi, (tr, te) in enumerate(splits):
+            tasks.append(Task(i, ci, fi, params, tr, te))
+            i += 1
+    results = run_batched_tasks(
+        est, X, y, tasks, scorer=ms, fit_params=None,
+        return_train_score=True, error_score="raise", verbose=0,
+        device=None)
+    assert len(results) == 6
+    for r in results:
+        assert set(r.test_score) == {"acc", "nll"}
+        assert set(r.train_score) == {"acc", "nll"}
+        assert 0.5 < r.test_score["acc"] <= 1.0
+        assert r.test_score["nll"] <= 0.0
