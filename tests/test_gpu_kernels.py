@@ -447,10 +447,11 @@ def test_multimetric_device_search(problem):
                           scoring="accuracy", cv=3).fit(X, y)
     np.testing.assert_allclose(gm.cv_results_["mean_test_accuracy"],
                                g1.cv_results_["mean_test_score"],
-                               rtol=1e-9)
+                               rtol=1e-6)
     g2 = ska.GridSearchCV(None, est, {"C": [0.1, 1.0, 10.0]},
                           scoring="neg_log_loss", cv=3).fit(X, y)
+    # per-chunk path vs native fast path: fp32 sum-order differs
     np.testing.assert_allclose(gm.cv_results_["mean_test_neg_log_loss"],
                                g2.cv_results_["mean_test_score"],
-                               rtol=1e-9)
+                               rtol=1e-6)
     assert gm.best_score_ == g1.best_score_
