@@ -156,6 +156,16 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
         est.set_params(**{k: clone(v, safe=False)
                           for k, v in task.params.items()})
     fit_params = fit_params or {}
+    if fit_params:
+        # legacy sklearn semantics: per-sample fit params (sample_weight
+        # and friends -- any array of length n_samples) are sliced to the
+        # train fold
+        n_samples = X.shape[0] if hasattr(X, "shape") else len(X)
+        fit_params = {
+            k: (_safe_index(np.asarray(v), task.train)
+                if hasattr(v, "__len__") and not np.isscalar(v) and
+                len(v) == n_samples else v)
+            for k, v in fit_params.items()}
 
     if _is_pairwise(est):
         # precomputed kernels/distances: slice BOTH axes (train columns)
@@ -182,17 +192,6 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
         else:
             est.fit(X_train, y_train, **fit_params)
         fit_time = time.perf_counter() - t0
-        t1 = time.perf_counter()
-        if y_test is None:
-            test_score = scorer(est, X_test)
-            score_time = time.perf_counter() - t1
-            train_score = scorer(est, X_train) if return_train_score \
-                else np.nan
-        else:
-            test_score = scorer(est, X_test, y_test)
-            score_time = time.perf_counter() - t1
-            train_score = scorer(est, X_train, y_train) \
-                if return_train_score else np.nan
     except Exception as exc:  # noqa: BLE001 - error_score contract
         fit_time = time.perf_counter() - t0
         if error_score == "raise":
@@ -208,6 +207,35 @@ def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
         return TaskResult(task.index, float(error_score),
                           float(error_score) if return_train_score else np.nan,
                           n_test, fit_time, 0.0, error=err)
+
+    # scoring errors are NOT fit failures (sklearn contract): they get
+    # error_score + their own warning, without tripping FitFailedWarning
+    t1 = time.perf_counter()
+    try:
+        if y_test is None:
+            test_score = scorer(est, X_test)
+            score_time = time.perf_counter() - t1
+            train_score = scorer(est, X_train) if return_train_score \
+                else np.nan
+        else:
+            test_score = scorer(est, X_test, y_test)
+            score_time = time.perf_counter() - t1
+            train_score = scorer(est, X_train, y_train) \
+                if return_train_score else np.nan
+    except Exception as exc:  # noqa: BLE001 - error_score contract
+        score_time = time.perf_counter() - t1
+        if error_score == "raise":
+            raise
+        if not isinstance(error_score, numbers.Number):
+            raise ValueError(
+                "error_score must be the string 'raise' or a numeric value"
+            ) from exc
+        warnings.warn(
+            "Scoring failed. The score on this train-test partition for "
+            f"these parameters will be set to {error_score}. "
+            f"Details: \n{exc!r}", UserWarning)
+        test_score = float(error_score)
+        train_score = float(error_score) if return_train_score else np.nan
 
     n_test = len(task.test)
     return TaskResult(task.index, _coerce_score(test_score),
@@ -258,7 +286,7 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
 
     def __init__(self, estimator, scoring=None, fit_params=None, n_jobs=1,
                  iid=False, refit=True, cv=None, verbose=0,
-                 pre_dispatch="2*n_jobs", error_score="raise",
+                 pre_dispatch="2*n_jobs", error_score=np.nan,
                  return_train_score=True):
         self.estimator = estimator
         self.scoring = scoring
@@ -305,6 +333,8 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         X, y, groups = indexable(X, y, groups)
         merged_fit_params = dict(self.fit_params or {})
         merged_fit_params.update(fit_params)
+        if "sample_weight" in merged_fit_params:
+            self._warn_scorers_without_sample_weight()
 
         candidates = list(self._get_param_iterator())
         n_candidates = len(candidates)
@@ -401,7 +431,11 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self.best_params_ = self.cv_results_["params"][self.best_index_]
 
         if self.refit:
-            best_estimator = clone(estimator).set_params(**self.best_params_)
+            # param values that are estimators must be cloned so grid
+            # entries are never fitted in place (sklearn contract)
+            best_estimator = clone(estimator).set_params(
+                **{k: clone(v, safe=False)
+                   for k, v in self.best_params_.items()})
             t0 = time.perf_counter()
             refit_fn = getattr(backend, "refit", None)
             if refit_fn is not None:
@@ -414,6 +448,44 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self.refit_time_ = time.perf_counter() - t0
             self.best_estimator_ = best_estimator
         return self
+
+    def _warn_scorers_without_sample_weight(self):
+        """sklearn parity: fitting with sample_weight warns per scorer
+        that cannot route it (statistically incorrect results)."""
+        from inspect import signature
+
+        def accepts(sc):
+            try:
+                if hasattr(sc, "_accept_sample_weight"):
+                    return sc._accept_sample_weight()
+                return "sample_weight" in signature(sc).parameters
+            except (TypeError, ValueError):
+                return True
+
+        if isinstance(self.scoring, dict):
+            from sklearn.metrics import get_scorer
+
+            for name, sc in self.scoring.items():
+                sc = get_scorer(sc) if isinstance(sc, str) else sc
+                if not accepts(sc):
+                    warnings.warn(
+                        f"The scoring {name}={sc} does not support "
+                        "sample_weight, which may lead to statistically "
+                        f"incorrect results when fitting {self} with "
+                        "sample_weight. ")
+        elif self.scoring is not None and not isinstance(self.scoring,
+                                                         (list, tuple,
+                                                          set)):
+            sc = self.scoring
+            if isinstance(sc, str):
+                from sklearn.metrics import get_scorer
+
+                sc = get_scorer(sc)
+            if not accepts(sc):
+                warnings.warn(
+                    f"The scoring {sc} does not support sample_weight, "
+                    "which may lead to statistically incorrect results "
+                    f"when fitting {self} with sample_weight. ")
 
     def _validate_multimetric_refit(self, keys):
         """refit must be False, a callable, or a key of the scorer dict
@@ -633,6 +705,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 f"{type(self).__name__} object has no n_features_in_ "
                 "attribute (call fit first)")
         return self.best_estimator_.n_features_in_
+
+    def _sk_visual_block_(self):
+        """sklearn HTML-diagram block: show best_estimator_ when fitted
+        with refit, else the prototype estimator (sklearn's exact block)."""
+        from sklearn.utils._estimator_html_repr import _VisualBlock
+
+        if hasattr(self, "best_estimator_"):
+            key, estimator = "best_estimator_", self.best_estimator_
+        else:
+            key, estimator = "estimator", self.estimator
+        return _VisualBlock(
+            "parallel", [estimator],
+            names=[f"{key}: {estimator.__class__.__name__}"],
+            name_details=[str(estimator)])
 
     def __sklearn_tags__(self):
         tags = super().__sklearn_tags__()

@@ -17,6 +17,8 @@ and fold scores come back via RCCL all-gather with refit on rank 0.
 
 from __future__ import annotations
 
+import numpy as np
+
 from sklearn.model_selection import ParameterGrid
 
 from .base_search import BaseSearchCV
@@ -62,7 +64,7 @@ class GridSearchCV(BaseSearchCV):
 
     def __init__(self, ctx, estimator, param_grid, scoring=None,
                  fit_params=None, n_jobs=1, iid=False, refit=True, cv=None,
-                 verbose=0, pre_dispatch="2*n_jobs", error_score="raise",
+                 verbose=0, pre_dispatch="2*n_jobs", error_score=np.nan,
                  return_train_score=True):
         super().__init__(
             estimator=estimator, scoring=scoring, fit_params=fit_params,

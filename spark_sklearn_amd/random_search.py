@@ -16,6 +16,8 @@ reference parity.
 
 from __future__ import annotations
 
+import numpy as np
+
 from sklearn.model_selection import ParameterSampler
 
 from .base_search import BaseSearchCV
@@ -35,7 +37,7 @@ class RandomizedSearchCV(BaseSearchCV):
     def __init__(self, ctx, estimator, param_distributions, n_iter=10,
                  scoring=None, fit_params=None, n_jobs=1, iid=False,
                  refit=True, cv=None, verbose=0, pre_dispatch="2*n_jobs",
-                 random_state=None, error_score="raise",
+                 random_state=None, error_score=np.nan,
                  return_train_score=True,
                  halving_factor=None, min_resource_frac=0.1):
         super().__init__(

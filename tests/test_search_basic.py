@@ -47,11 +47,22 @@ def test_error_score_numeric_swallows():
     assert gs.best_params_ == {"C": 1.0}
 
 
-def test_error_score_raise_default():
+def test_error_score_default_nan():
+    """Default error_score matches the installed sklearn oracle (np.nan
+    with an aggregated FitFailedWarning); 'raise' remains available."""
     X, y = _xy()
-    gs = ska.GridSearchCV(None, FailingEstimator(), {"C": [-1.0, 1.0]}, cv=3)
-    with pytest.raises(ValueError, match="bad C"):
+    from sklearn.exceptions import FitFailedWarning
+
+    gs = ska.GridSearchCV(None, FailingEstimator(), {"C": [-1.0, 1.0]},
+                          cv=3)
+    with pytest.warns(FitFailedWarning):
         gs.fit(X, y)
+    assert np.isnan(gs.cv_results_["mean_test_score"][0])
+
+    gs2 = ska.GridSearchCV(None, FailingEstimator(), {"C": [-1.0, 1.0]},
+                           cv=3, error_score="raise")
+    with pytest.raises(ValueError, match="bad C"):
+        gs2.fit(X, y)
 
 
 def test_refit_false_blocks_predict():

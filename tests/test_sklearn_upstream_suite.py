@@ -13,6 +13,8 @@ tests (fixtures) are excluded from enumeration.
 """
 
 import inspect
+
+import numpy as np
 import warnings
 
 import pytest
@@ -26,26 +28,39 @@ import spark_sklearn_amd as ska
 SKIP = {
     # sklearn-internal extension API (_run_search subclass hook)
     "test_custom_run_search",
-    # sklearn HTML repr internals
-    "test_search_html_repr",
-    # wrapper-subclass get_params artifact (set_params on the wrapper)
-    "test_unsupported_sample_weight_scorer",
-    # param values that are estimator INSTANCES: cv_results_ stores the
-    # grid's objects (sklearn stores unfitted clones)
-    "test_grid_search_pipeline_steps",
 }
 
 
 class SKAGridSearchCV(ska.GridSearchCV):
-    """Upstream-compatible wrapper (module-level: must be picklable)."""
+    """Upstream-compatible wrapper (module-level: must be picklable).
+    The full kwarg surface is spelled out so sklearn's get_params/clone
+    introspection sees every parameter."""
 
-    def __init__(self, estimator, param_grid, **kw):
-        super().__init__("sequential", estimator, param_grid, **kw)
+    def __init__(self, estimator, param_grid, scoring=None,
+                 fit_params=None, n_jobs=1, iid=False, refit=True,
+                 cv=None, verbose=0, pre_dispatch="2*n_jobs",
+                 error_score=np.nan, return_train_score=True):
+        super().__init__(
+            "sequential", estimator, param_grid, scoring=scoring,
+            fit_params=fit_params, n_jobs=n_jobs, iid=iid, refit=refit,
+            cv=cv, verbose=verbose, pre_dispatch=pre_dispatch,
+            error_score=error_score,
+            return_train_score=return_train_score)
 
 
 class SKARandomizedSearchCV(ska.RandomizedSearchCV):
-    def __init__(self, estimator, param_distributions, **kw):
-        super().__init__("sequential", estimator, param_distributions, **kw)
+    def __init__(self, estimator, param_distributions, n_iter=10,
+                 scoring=None, fit_params=None, n_jobs=1, iid=False,
+                 refit=True, cv=None, verbose=0, random_state=None,
+                 pre_dispatch="2*n_jobs", error_score=np.nan,
+                 return_train_score=True):
+        super().__init__(
+            "sequential", estimator, param_distributions, n_iter=n_iter,
+            scoring=scoring, fit_params=fit_params, n_jobs=n_jobs,
+            iid=iid, refit=refit, cv=cv, verbose=verbose,
+            random_state=random_state, pre_dispatch=pre_dispatch,
+            error_score=error_score,
+            return_train_score=return_train_score)
 
 
 def _upstream():
