@@ -364,21 +364,50 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 idx += 1
 
         backend = self._get_backend()
+
+        # resume: replay completed (candidate, fold) fits from a prior
+        # run's score log; only the remainder executes
+        resume_path = getattr(backend, "resume_from_log", None)
+        restored: List[TaskResult] = []
+        todo = tasks
+        if resume_path:
+            done = self._load_resume_records(resume_path, n_splits)
+            if done:
+                todo = []
+                for t in tasks:
+                    rec = done.get((self._param_key(t.params),
+                                    t.fold_index))
+                    if rec is None:
+                        todo.append(t)
+                    else:
+                        tr = rec.get("train_score")
+                        restored.append(TaskResult(
+                            t.index, rec["test_score"],
+                            np.nan if tr is None else tr,
+                            rec["n_test"], rec["fit_time"],
+                            rec["score_time"]))
+                if self.verbose > 0:
+                    print(f"[search] resume: {len(restored)} of "
+                          f"{len(tasks)} fits restored from "
+                          f"{resume_path}")
+        self.n_resumed_ = len(restored)
+
         t_run0 = time.perf_counter()
         results = backend.run_search(
-            estimator, X, y, tasks,
+            estimator, X, y, todo,
             scorer=dispatch_scorer,
             fit_params=merged_fit_params,
             return_train_score=self.return_train_score,
             error_score=self.error_score,
             verbose=self.verbose,
-        )
+        ) if todo else []
+        results = list(results) + restored
         run_elapsed = time.perf_counter() - t_run0
         # observability counters: candidate-fits/sec
         # is the north-star metric; exposed on the fitted object
-        self.n_fits_ = len(tasks)
+        self.n_fits_ = len(todo)
         self.fit_duration_ = run_elapsed
-        self.fits_per_sec_ = len(tasks) / max(run_elapsed, 1e-9)
+        self.fits_per_sec_ = len(todo) / max(run_elapsed, 1e-9)
         if self.verbose > 0:
             print(f"[search] {self.n_fits_} fits in {run_elapsed:.2f}s "
                   f"({self.fits_per_sec_:.1f} candidate-fits/s)")
@@ -395,7 +424,11 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
 
         score_log = getattr(backend, "score_log", None)
         if score_log and getattr(backend, "rank", 0) == 0:
-            self._append_score_log(score_log, tasks, ordered)
+            executed = {t.index for t in todo}
+            self._append_score_log(
+                score_log,
+                [t for t in tasks if t.index in executed],
+                [r for r in ordered if r.index in executed])
 
         # multimetric-ness is declared (dict/list scoring) or discovered
         # (a plain callable that returned a dict of scores)
@@ -534,19 +567,63 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             f"Below are more details about the failures:\n{details}",
             FitFailedWarning)
 
+    @staticmethod
+    def _param_key(params) -> str:
+        """Stable text key for a candidate's params (resume matching)."""
+        return repr(sorted((k, repr(v)) for k, v in params.items()))
+
     def _append_score_log(self, path, tasks, ordered):
         """Append-only JSONL score log: one record
         per (candidate, fold) fit so a killed 100k-candidate sweep leaves
-        its completed work on disk."""
+        its completed work on disk; ``resume_from_log`` replays it."""
         import json
 
         with open(path, "a") as f:
             for t, r in zip(tasks, ordered):
+                if r.error is not None:
+                    continue      # failed fits re-run on resume
                 f.write(json.dumps({
                     "cand": t.cand_index, "fold": t.fold_index,
+                    "key": self._param_key(t.params),
                     "params": {k: repr(v) for k, v in t.params.items()},
                     "test_score": r.test_score,
-                    "fit_time": r.fit_time}) + "\n")
+                    "train_score": (None if (isinstance(r.train_score,
+                                                        float) and
+                                             np.isnan(r.train_score))
+                                    else r.train_score),
+                    "n_test": r.n_test,
+                    "fit_time": r.fit_time,
+                    "score_time": r.score_time}) + "\n")
+
+    def _load_resume_records(self, path, n_splits):
+        """Completed (param-key, fold) -> TaskResult fields from a prior
+        run's score log.  Records missing the resume fields (old-format
+        logs) or train scores when this run needs them are skipped, so
+        those tasks simply re-run."""
+        import json
+        import os as _os
+
+        done = {}
+        if not path or not _os.path.exists(path):
+            return done
+        with open(path) as f:
+            for line in f:
+                line = line.strip()
+                if not line:
+                    continue
+                try:
+                    rec = json.loads(line)
+                except ValueError:
+                    continue      # partial last line of a killed run
+                if "key" not in rec or rec.get("fold") is None:
+                    continue
+                if rec["fold"] >= n_splits:
+                    continue      # different cv: ignore the whole record
+                if self.return_train_score and rec.get("train_score") \
+                        is None:
+                    continue
+                done[(rec["key"], rec["fold"])] = rec
+        return done
 
     # -- cv_results_ assembly ---------------------------------------------
     def _assemble_cv_results(self, candidates, ordered_results,

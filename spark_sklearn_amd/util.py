@@ -29,7 +29,7 @@ class HipContext:
     def __init__(self, backend=None, device=None, candidate_batch: int = 128,
                  n_streams: int = 2, dtype: str = "bfloat16",
                  n_threads: Optional[int] = None, score_log: str = None,
-                 progress=None):
+                 resume_from_log: str = None, progress=None):
         from .parallel import resolve_backend
 
         self.candidate_batch = candidate_batch
@@ -37,12 +37,13 @@ class HipContext:
         self.dtype = dtype
         self.n_threads = n_threads
         self.score_log = score_log
+        self.resume_from_log = resume_from_log
         self.progress = progress
         self.device = device if device is not None else _default_device()
         self.backend = resolve_backend(backend)
         # push engine knobs down where backends look for them
         for attr in ("candidate_batch", "n_streams", "dtype", "score_log",
-                     "progress"):
+                     "resume_from_log", "progress"):
             setattr(self.backend, attr, getattr(self, attr))
 
     @property

@@ -195,3 +195,47 @@ def test_device_csr_cpu_roundtrip():
     # bf16 storage tolerance
     np.testing.assert_allclose(back.toarray(), mat.toarray(),
                                atol=0.01, rtol=0.01)
+
+
+def test_score_log_resume(tmp_path):
+    """A killed sweep's JSONL score log replays on the next fit: completed
+    (candidate, fold) fits are restored, only the remainder executes, and
+    cv_results_ matches an uninterrupted run exactly."""
+    import numpy as np
+
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd import HipContext
+    from sklearn.svm import LinearSVC
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 8))
+    y = (X[:, 0] > 0).astype(np.int64)
+    log = str(tmp_path / "scores.jsonl")
+    grid = {"C": [0.01, 0.1, 1.0, 10.0]}
+
+    # phase 1: "killed" run that only saw half the grid
+    ctx1 = HipContext(backend="sequential", score_log=log)
+    ska.GridSearchCV(ctx1, LinearSVC(), {"C": [0.01, 0.1]}, cv=3).fit(X, y)
+
+    # phase 2: full grid, resuming -- only the 2 new candidates run
+    ctx2 = HipContext(backend="sequential", score_log=log,
+                      resume_from_log=log)
+    gs = ska.GridSearchCV(ctx2, LinearSVC(), grid, cv=3)
+    gs.fit(X, y)
+    assert gs.n_resumed_ == 6          # 2 candidates x 3 folds restored
+    assert gs.n_fits_ == 6             # 2 new candidates x 3 folds run
+
+    ref = ska.GridSearchCV("sequential", LinearSVC(), grid, cv=3).fit(X, y)
+    np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
+                               ref.cv_results_["mean_test_score"],
+                               rtol=1e-12)
+    assert gs.best_params_ == ref.best_params_
+
+    # phase 3: everything restored, zero fits executed
+    gs2 = ska.GridSearchCV(HipContext(backend="sequential",
+                                      resume_from_log=log),
+                           LinearSVC(), grid, cv=3).fit(X, y)
+    assert gs2.n_fits_ == 0 and gs2.n_resumed_ == 12
+    np.testing.assert_allclose(gs2.cv_results_["mean_test_score"],
+                               ref.cv_results_["mean_test_score"],
+                               rtol=1e-12)
