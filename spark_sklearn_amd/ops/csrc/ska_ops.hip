@@ -1481,6 +1481,29 @@ __global__ void gbt_grad_kernel(const float* __restrict__ raw,
   }
 }
 
+// quantile binning on device: bin = searchsorted(edges[f], x, 'left')
+// (first edge >= x).  Edges live in global memory (F x 255 floats, hot in
+// L2); one thread per element, 8-step binary search -- replaces a
+// 10+-second host bin_data pass at 1M x 128.
+__global__ void gbt_bin_kernel(const float* __restrict__ X,
+                               const float* __restrict__ edges,  // [F][255]
+                               const int* __restrict__ ecnt,     // [F]
+                               unsigned char* __restrict__ Xb,
+                               long total, int F) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < total; i += (long)gridDim.x * blockDim.x) {
+    int f = (int)(i % F);
+    float x = X[i];
+    const float* e = edges + (long)f * 255;
+    int lo = 0, hi = ecnt[f];
+    while (lo < hi) {
+      int mid = (lo + hi) >> 1;
+      if (e[mid] < x) lo = mid + 1; else hi = mid;
+    }
+    Xb[i] = (unsigned char)lo;
+  }
+}
+
 // per-(node, feature, bin) histograms in LDS: g as float atomics, (h,
 // count) packed into ONE u64 atomic -- h fixed-point in the high half
 // (hscale chosen on the host so the sum cannot overflow 32 bits), exact
@@ -2944,6 +2967,23 @@ torch::Tensor row_norms(torch::Tensor X) {
 
 // One full boosting fit on device.  Returns (feats [T][NN] i32,
 // thrs [T][NN] i32, vals [T][NN] f32).
+torch::Tensor gbt_bin(torch::Tensor X, torch::Tensor edges,
+                      torch::Tensor ecnt) {
+  const at::cuda::CUDAGuard guard(X.device());
+  TORCH_CHECK(X.is_contiguous() && X.dtype() == torch::kFloat32);
+  long n = X.size(0), F = X.size(1);
+  auto Xb = torch::empty({n, F}, X.options().dtype(torch::kUInt8));
+  long total = n * F;
+  hipStream_t s = cur_stream();
+  int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(gbt_bin_kernel, dim3(blocks), dim3(256), 0, s,
+                     X.data_ptr<float>(), edges.data_ptr<float>(),
+                     ecnt.data_ptr<int>(),
+                     Xb.data_ptr<unsigned char>(), total, (int)F);
+  HIP_CHECK(hipGetLastError());
+  return Xb;
+}
+
 std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
                                    torch::Tensor y_tr, int64_t loss,
                                    int64_t n_estimators, double lrate,
@@ -3576,6 +3616,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_fit", &kmeans_fit, "batched Lloyd loop (centroid blocks)");
   m.def("kmeans_assign", &kmeans_assign, "assignment/inertia under centers");
   m.def("row_norms", &row_norms, "per-row squared norms");
+  m.def("gbt_bin", &gbt_bin, "device quantile binning");
   m.def("gbt_fit", &gbt_fit, "histogram boosting fit (depth-wise trees)");
   m.def("gbt_predict", &gbt_predict, "tree-walk raw predictions");
   m.def("gbt_metric", &gbt_metric, "accuracy/ss_res sums");
