@@ -50,19 +50,31 @@ class SparseLinearExecutor:
             cols = torch.as_tensor(
                 Xc.indices.astype(np.int32)).to(dev)
             vals = torch.as_tensor(Xc.data).to(dev).to(torch.bfloat16)
-            # bwd structure: entries sorted by (row chunk, feature, row)
+            # bwd structure: entries sorted by (row chunk, feature, row).
+            # The sort runs ON DEVICE: a host np.lexsort over 100M nnz
+            # costs ~10 s; torch.argsort on a unique composite int64 key
+            # costs ~50 ms.  key = ((chunk*F + col)*n + row) is unique per
+            # entry and fits int64 far beyond the 288 GB HBM shapes.
             coo = Xc.tocoo()
-            chunk = (coo.row // BWD_CHUNK_ROWS).astype(np.int64)
-            order = np.lexsort((coo.row, coo.col, chunk))
+            nchunks = (n - 1) // BWD_CHUNK_ROWS + 1 if n else 1
+            col64 = coo.col.astype(np.int64)
+            row64 = coo.row.astype(np.int64)
+            key = ((row64 // BWD_CHUNK_ROWS) * F + col64) * n + row64
+            key_t = torch.as_tensor(key).to(dev)
+            order_t = torch.argsort(key_t)
+            ks = key_t[order_t]
             tf = torch.as_tensor(
-                coo.col[order].astype(np.int32)).to(dev)
+                coo.col.astype(np.int32)).to(dev)[order_t].contiguous()
             tr = torch.as_tensor(
-                coo.row[order].astype(np.int32)).to(dev)
-            tv = torch.as_tensor(
-                coo.data[order]).to(dev).to(torch.bfloat16)
-            nchunks = int(chunk.max()) + 1 if len(chunk) else 1
-            co = np.searchsorted(chunk[order], np.arange(nchunks + 1))
-            chunk_off = torch.as_tensor(co.astype(np.int64)).to(dev)
+                coo.row.astype(np.int32)).to(dev)[order_t].contiguous()
+            tv = torch.as_tensor(coo.data).to(dev)[order_t] \
+                .to(torch.bfloat16).contiguous()
+            chunk_sorted = ks // (np.int64(F) * np.int64(n))
+            chunk_off = torch.searchsorted(
+                chunk_sorted,
+                torch.arange(nchunks + 1, dtype=torch.int64,
+                             device=dev)).to(torch.int64).contiguous()
+            del key_t, order_t, ks, chunk_sorted
             sq = np.asarray(Xc.multiply(Xc).sum(axis=1)).ravel()
             msq = float(np.mean(sq))
             if len(_sparse_cache) > 2:
