@@ -59,8 +59,8 @@ class SparseLinearExecutor:
             nchunks = (n - 1) // BWD_CHUNK_ROWS + 1 if n else 1
             col64 = coo.col.astype(np.int64)
             row64 = coo.row.astype(np.int64)
-            key = ((row64 // BWD_CHUNK_ROWS) * F + col64) * n + row64
-            key_t = torch.as_tensor(key).to(dev)
+            skey = ((row64 // BWD_CHUNK_ROWS) * F + col64) * n + row64
+            key_t = torch.as_tensor(skey).to(dev)
             order_t = torch.argsort(key_t)
             ks = key_t[order_t]
             tf = torch.as_tensor(
