@@ -1629,7 +1629,12 @@ __global__ void gbt_sibling_kernel(const float* __restrict__ prev,
   }
 }
 
-// best split per node: 256 threads scan features; LDS argmax reduce.
+// best split per node: one WAVE per feature.  Each lane owns 4
+// consecutive bins; cumulative (G,H,C) up to every threshold comes from a
+// wave-level shfl prefix scan, so the 256-bin walk that used to run
+// serially per thread is 64-lane parallel and the histogram is read ONCE
+// (totals = last prefix element).  Deterministic tie-break: lowest
+// feature, then lowest threshold, within a +-1e-12 gain band.
 __global__ __launch_bounds__(256)
 void gbt_split_find_kernel(const float* __restrict__ hist, int F,
                            int lvl_lo, float lam, float lr, int min_leaf,
@@ -1638,67 +1643,89 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
   const int nd_local = blockIdx.x;
   const int nd = lvl_lo + nd_local;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
   const float* nh = hist + (long)nd_local * F * 256 * 3;
 
-  __shared__ float sGain[256];
-  __shared__ int sF[256], sT[256];
+  __shared__ float sGain[4];
+  __shared__ int sFt[4], sTt[4];
   __shared__ float sG, sH;
+
   float bgain = 0.f;
   int bf = -1, bt = 0;
-  float Gtot = 0.f, Htot = 0.f;
 
-  for (int f = tid; f < F; f += 256) {
+  for (int f = wave; f < F; f += 4) {
     const float* fh = nh + (long)f * 256 * 3;
-    float G = 0.f, H = 0.f, C = 0.f;
-    for (int b2 = 0; b2 < 256; ++b2) {
-      G += fh[b2 * 3];
-      H += fh[b2 * 3 + 1];
-      C += fh[b2 * 3 + 2];
+    float g[4], h[4], c[4];
+    float lg = 0.f, lh = 0.f, lc = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int b = 4 * lane + j;
+      g[j] = fh[b * 3];
+      h[j] = fh[b * 3 + 1];
+      c[j] = fh[b * 3 + 2];
+      lg += g[j]; lh += h[j]; lc += c[j];
     }
-    if (f == tid) { Gtot = G; Htot = H; }
+    // inclusive prefix over lanes of the per-lane chunk sums
+    float pg = lg, ph = lh, pc = lc;
+    for (int off = 1; off < 64; off <<= 1) {
+      float tg = __shfl_up(pg, off);
+      float th = __shfl_up(ph, off);
+      float tc = __shfl_up(pc, off);
+      if (lane >= off) { pg += tg; ph += th; pc += tc; }
+    }
+    float G = __shfl(pg, 63), H = __shfl(ph, 63), C = __shfl(pc, 63);
+    if (f == 0 && lane == 0) { sG = G; sH = H; }
     if (!force_leaf) {
       float ps = G * G / (H + lam + 1e-12f);
-      float gl = 0.f, hl = 0.f, cl = 0.f;
-      for (int b2 = 0; b2 < 255; ++b2) {
-        gl += fh[b2 * 3];
-        hl += fh[b2 * 3 + 1];
-        cl += fh[b2 * 3 + 2];
+      float gl = pg - lg, hl = ph - lh, cl = pc - lc;   // exclusive base
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int b = 4 * lane + j;
+        gl += g[j]; hl += h[j]; cl += c[j];
+        if (b >= 255) continue;
         float cr = C - cl;
         if (cl < (float)min_leaf || cr < (float)min_leaf) continue;
         float gr2 = G - gl, hr2 = H - hl;
         float gain = gl * gl / (hl + lam + 1e-12f) +
                      gr2 * gr2 / (hr2 + lam + 1e-12f) - ps;
-        if (gain > bgain + 1e-12f) { bgain = gain; bf = f; bt = b2; }
+        if (gain > bgain + 1e-12f ||
+            (gain > bgain - 1e-12f &&
+             (bf < 0 || f < bf || (f == bf && b < bt)))) {
+          bgain = gain; bf = f; bt = b;
+        }
       }
     }
   }
-  sGain[tid] = bgain;
-  sF[tid] = bf;
-  sT[tid] = bt;
-  if (tid == 0) { sG = Gtot; sH = Htot; }
+  // wave argmax (same tie-break), then cross-wave via LDS
+  for (int off = 32; off > 0; off >>= 1) {
+    float og = __shfl_down(bgain, off);
+    int of = __shfl_down(bf, off);
+    int ot = __shfl_down(bt, off);
+    if (og > bgain + 1e-12f ||
+        (og > bgain - 1e-12f && of >= 0 &&
+         (bf < 0 || of < bf || (of == bf && ot < bt)))) {
+      bgain = og; bf = of; bt = ot;
+    }
+  }
+  if (lane == 0) { sGain[wave] = bgain; sFt[wave] = bf; sTt[wave] = bt; }
   __syncthreads();
-  for (int s2 = 128; s2 > 0; s2 >>= 1) {
-    if (tid < s2) {
-      // deterministic tie-break: keep the LOWER feature index
-      if (sGain[tid + s2] > sGain[tid] + 1e-12f ||
-          (sGain[tid + s2] > sGain[tid] - 1e-12f &&
-           sF[tid + s2] >= 0 &&
-           (sF[tid] < 0 || sF[tid + s2] < sF[tid]))) {
-        sGain[tid] = sGain[tid + s2];
-        sF[tid] = sF[tid + s2];
-        sT[tid] = sT[tid + s2];
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w) {
+      if (sGain[w] > sGain[0] + 1e-12f ||
+          (sGain[w] > sGain[0] - 1e-12f && sFt[w] >= 0 &&
+           (sFt[0] < 0 || sFt[w] < sFt[0] ||
+            (sFt[w] == sFt[0] && sTt[w] < sTt[0])))) {
+        sGain[0] = sGain[w]; sFt[0] = sFt[w]; sTt[0] = sTt[w];
       }
     }
-    __syncthreads();
-  }
-  if (tid == 0) {
-    if (sF[0] < 0 || sGain[0] <= 1e-12f) {
+    if (sFt[0] < 0 || sGain[0] <= 1e-12f) {
       feat[nd] = -1;
       if (sH > 0.f || sG != 0.f)
         val[nd] = -sG / (sH + lam + 1e-12f) * lr;
     } else {
-      feat[nd] = sF[0];
-      thr[nd] = sT[0];
+      feat[nd] = sFt[0];
+      thr[nd] = sTt[0];
       val[nd] = 0.f;
     }
   }
