@@ -2009,166 +2009,6 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   }
 }
 
-// fwd_d10_kernel: 512-thread variant of fwd_d7 -- same 256x256 block
-// tile, 8 waves of 64 rows x 128 cols (NF=8).  Each wave's A-fragments
-// feed 8 column groups instead of 4, so the LDS-read count per MFMA
-// drops 33% (24 reads per 64 MFMA vs 16 per 32) -- the fwd kernel is
-// LDS-throughput-bound (profiles/r01_families_kernels.md).
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(512)
-void fwd_d10_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
-                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
-                    const float* __restrict__ bias,
-                    const int* __restrict__ loss_id,
-                    const int* __restrict__ target, __bf16* __restrict__ Dt,
-                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
-                    const int* __restrict__ row_key,
-                    const int* __restrict__ col_key,
-                    int nt, int NTp, int Fp, int Bp, int want_loss) {
-  constexpr int NF = 8;
-  __shared__ __bf16 sX[2][256 * 64];
-  __shared__ __bf16 sW[2][256 * 64];
-  __shared__ float sLoss[256];
-  __shared__ float sDb[256];
-  __shared__ int sIdx[256];
-  __shared__ float sY[256];
-  __shared__ int sKey[256];
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int chalf = wave & 1;
-  const int rquad = wave >> 1;        // row quad 0..3
-  const int rt = blockIdx.x;
-  const int g = lane >> 4;
-  const int srow = tid >> 1;          // staging row/col 0..255
-  const int sseg = (tid & 1) * 2;     // two 32B halves per row pair
-
-  if (tid < 256) {
-    int gi = idx[rt * 256 + tid];
-    sIdx[tid] = gi;
-    sY[tid] = y[gi];
-    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
-    sLoss[tid] = 0.f;
-    sDb[tid] = 0.f;
-  }
-  __syncthreads();
-
-  bf16x8 rx[4], rw[4];
-  auto issue = [&](int kc) {
-    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
-        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
-    rx[0] = gx[0];
-    rx[1] = gx[1];
-    rx[2] = gx[2];
-    rx[3] = gx[3];
-    const bf16x8* gw = reinterpret_cast<const bf16x8*>(
-        Wt + (long)srow * Fp + kc + sseg * 16);
-    rw[0] = gw[0];
-    rw[1] = gw[1];
-    rw[2] = gw[2];
-    rw[3] = gw[3];
-  };
-  auto commit = [&](int b) {
-    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 16, rx[2]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 24, rx[3]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 16, rw[2]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 24, rw[3]);
-  };
-
-  f32x4 acc[4][NF];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  issue(0);
-  commit(0);
-  __syncthreads();
-  const int nkc = Fp >> 6;
-  for (int kc = 0; kc < nkc; ++kc) {
-    int b = kc & 1;
-    if (kc + 1 < nkc) issue((kc + 1) << 6);
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      bf16x8 a[4];
-#pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        int r = rquad * 64 + m * 16 + (lane & 15);
-        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
-      }
-#pragma unroll
-      for (int fn = 0; fn < NF; ++fn) {
-        int c = chalf * 128 + fn * 16 + (lane & 15);
-        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
-#pragma unroll
-        for (int m = 0; m < 4; ++m)
-          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[m], bb, acc[m][fn], 0, 0, 0);
-      }
-    }
-    if (kc + 1 < nkc) commit(b ^ 1);
-    __syncthreads();
-  }
-
-#pragma unroll
-  for (int fn = 0; fn < NF; ++fn) {
-    int col_in = chalf * 128 + fn * 16 + (lane & 15);
-    float bia = bias[col_in];
-    int lid = loss_id[col_in];
-    int tgt = target[col_in];
-    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
-    float lsum = 0.f, dsum = 0.f;
-#pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
-      bf16x4 dpack;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int gr = rt * 256 + row_base + r;
-        float z = acc[m][fn][r] + bia;
-        float d, ls = 0.f;
-        if (want_loss) {
-          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
-        } else {
-          float yv = sY[row_base + r];
-          if (lid == LOSS_LOGISTIC) {
-            float t = (yv == (float)tgt) ? 1.f : 0.f;
-            d = 1.f / (1.f + __expf(-z)) - t;
-          } else if (lid == LOSS_HINGE) {
-            float sg = (yv == (float)tgt) ? 1.f : -1.f;
-            d = (sg * z < 1.f) ? -sg : 0.f;
-          } else {
-            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
-            d = z - t;
-          }
-        }
-        if (gr >= nt) { d = 0.f; ls = 0.f; }
-        if (row_key != nullptr && sKey[row_base + r] != ck) {
-          d = 0.f;
-          ls = 0.f;
-        }
-        dpack[r] = (__bf16)d;
-        lsum += ls;
-        dsum += d;
-      }
-      *reinterpret_cast<bf16x4*>(
-          Dt + (long)col_in * NTp + rt * 256 + row_base) = dpack;
-    }
-    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
-    atomicAdd(&sDb[col_in], dsum);
-  }
-  __syncthreads();
-  if (tid < 256) {
-    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
-    atomicAdd(&db_sum[tid], sDb[tid]);
-  }
-}
-
 // ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
@@ -2754,8 +2594,8 @@ std::vector<torch::Tensor> linear_fit(
     if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
       if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
         // 1024-thread fwd (256x256 tile, 16 waves/CU)
-        hipLaunchKernelGGL(fwd_d10_kernel, dim3((unsigned)(NTp / 256)),
-                           dim3(512), 0, s, Xp, y.data_ptr<float>(),
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+                           dim3(1024), 0, s, Xp, y.data_ptr<float>(),
                            idx.data_ptr<int>(),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                            bias.data_ptr<float>(), loss_id.data_ptr<int>(),
@@ -3429,8 +3269,8 @@ std::vector<torch::Tensor> linear_search_run(
 
       for (int epoch = 0; epoch < max_ep; ++epoch) {
         int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
-        hipLaunchKernelGGL(fwd_d10_kernel, dim3((unsigned)(NTp / 256)),
-                           dim3(512), 0, s, Xp, y.data_ptr<float>(), idxp,
+        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+                           dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                            bias.data_ptr<float>(), loss_p, tgt_p,
                            reinterpret_cast<__bf16*>(Dt.data_ptr()),
