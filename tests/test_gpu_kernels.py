@@ -455,3 +455,49 @@ def test_multimetric_device_search(problem):
                                g2.cv_results_["mean_test_score"],
                                rtol=1e-6)
     assert gm.best_score_ == g1.best_score_
+
+
+def test_determinism_and_stability(problem):
+    """Race canary: two identical searches on the
+    same resident data must agree to float-atomic reordering noise
+    (split-K G-sums make bit-exactness impossible by design; scores must
+    match to ~1e-6 and the ranking exactly)."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import LogisticRegression
+
+    X, y = problem
+    est = LogisticRegression(max_iter=15, tol=0.0)
+    grid = {"C": [0.05, 0.5, 5.0]}
+    a = ska.GridSearchCV(None, est, grid, cv=3,
+                         return_train_score=False).fit(X, y)
+    b = ska.GridSearchCV(None, est, grid, cv=3,
+                         return_train_score=False).fit(X, y)
+    np.testing.assert_allclose(a.cv_results_["mean_test_score"],
+                               b.cv_results_["mean_test_score"],
+                               rtol=0, atol=2e-5)
+    assert list(a.cv_results_["rank_test_score"]) == \
+        list(b.cv_results_["rank_test_score"])
+    assert a.best_params_ == b.best_params_
+
+
+def test_gbt_structure_determinism(problem):
+    """GBT tree STRUCTURES are integer outputs: two fits on identical
+    data must produce identical (feature, threshold) trees."""
+    from spark_sklearn_amd.models._gbt_ref import GBTSpec
+    from spark_sklearn_amd.models.gbt import _GBTPlan
+    from spark_sklearn_amd.ops.gbt_executor import DeviceGBTExecutor
+
+    X, y = problem
+    dev = torch.device("cuda", 0)
+    spec = GBTSpec(n_estimators=6, max_depth=4, learning_rate=0.3,
+                   min_samples_leaf=20)
+    train = np.arange(6000, dtype=np.int64)
+    outs = []
+    for _ in range(2):
+        ex = DeviceGBTExecutor(np.ascontiguousarray(X), y.astype(np.float32),
+                               dev)
+        (models, edges), _, _ = ex.fit(train, _GBTPlan([spec]))
+        trees, f0 = models[0]
+        outs.append([(tuple(f.tolist()), tuple(t.tolist()))
+                     for f, t, v in trees])
+    assert outs[0] == outs[1]
