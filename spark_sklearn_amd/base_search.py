@@ -411,6 +411,9 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         if self.verbose > 0:
             print(f"[search] {self.n_fits_} fits in {run_elapsed:.2f}s "
                   f"({self.fits_per_sec_:.1f} candidate-fits/s)")
+        from .util import record_search_metrics
+
+        record_search_metrics(self.n_fits_, self.fits_per_sec_)
         # restore deterministic submit order regardless of backend ordering
         by_index = {r.index: r for r in results}
         missing = [t.index for t in tasks if t.index not in by_index]

@@ -267,8 +267,8 @@ def _encode_cached(family, estimator, y):
 def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                       fit_params, return_train_score, error_score, verbose,
                       device=None,
-                      column_batch: int = DEFAULT_COLUMN_BATCH
-                      ) -> List[TaskResult]:
+                      column_batch: int = DEFAULT_COLUMN_BATCH,
+                      progress=None) -> List[TaskResult]:
     """Execute (candidate, fold) tasks through batched kernel launches.
 
     Tasks are grouped by fold (all candidates of a fold share the same
@@ -328,7 +328,17 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             plan, bad = family.build_plan(
                 estimator, params_list, len(train), executor,
                 n_classes, error_score)
-            fitted, n_iter, _ = executor.fit(train, plan)
+            try:
+                fitted, n_iter, _ = executor.fit(train, plan)
+            except RuntimeError as exc:
+                # per-chunk retry on a transient HIP error (SURVEY.md
+                # section 5.3a); a second failure aborts the job
+                if "HIP" not in str(exc) and "hip" not in str(exc):
+                    raise
+                warnings.warn(
+                    f"retrying candidate chunk after HIP error: {exc}",
+                    RuntimeWarning, stacklevel=2)
+                fitted, n_iter, _ = executor.fit(train, plan)
             fit_time = time.perf_counter() - t0
 
             t1 = time.perf_counter()
@@ -359,6 +369,11 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
 
             per_fit = fit_time / len(chunk)
             per_score = score_time / len(chunk)
+            if progress is not None:
+                try:
+                    progress(len(results) + len(chunk), len(tasks))
+                except Exception:  # noqa: BLE001 - observability only
+                    pass
             for slot, task in enumerate(chunk):
                 err = None
                 if slot in bad:

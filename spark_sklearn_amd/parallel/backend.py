@@ -48,7 +48,7 @@ def _device_available() -> bool:
 def execute_local_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                         fit_params, return_train_score, error_score, verbose,
                         n_threads: Optional[int] = None,
-                        device=None) -> List[TaskResult]:
+                        device=None, progress=None) -> List[TaskResult]:
     """Execute this process's share of tasks, on device or host.
 
     Device path: candidates batched per fold into single HIP kernel launches
@@ -64,22 +64,40 @@ def execute_local_tasks(estimator, X, y, tasks: List[Task], *, scorer,
         return run_batched_tasks(
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
-            verbose=verbose, device=device)
+            verbose=verbose, device=device, progress=progress)
 
     n_threads = n_threads or min(len(tasks), os.cpu_count() or 4)
     if n_threads <= 1 or len(tasks) == 1:
-        return [fit_and_score_task(estimator, X, y, scorer, t,
-                                   fit_params=fit_params,
-                                   return_train_score=return_train_score,
-                                   error_score=error_score)
-                for t in tasks]
+        out = []
+        for i, t in enumerate(tasks):
+            out.append(fit_and_score_task(
+                estimator, X, y, scorer, t, fit_params=fit_params,
+                return_train_score=return_train_score,
+                error_score=error_score))
+            _report(progress, i + 1, len(tasks))
+        return out
     with ThreadPoolExecutor(max_workers=n_threads) as pool:
         futs = [pool.submit(fit_and_score_task, estimator, X, y, scorer, t,
                             fit_params=fit_params,
                             return_train_score=return_train_score,
                             error_score=error_score)
                 for t in tasks]
-        return [f.result() for f in futs]
+        out = []
+        for i, f in enumerate(futs):
+            out.append(f.result())
+            _report(progress, i + 1, len(futs))
+        return out
+
+
+def _report(progress, done, total):
+    """Invoke the context's progress callback; callbacks must never kill
+    a search."""
+    if progress is None:
+        return
+    try:
+        progress(done, total)
+    except Exception:  # noqa: BLE001 - observability must not break work
+        pass
 
 
 class Backend:
@@ -101,7 +119,8 @@ class SequentialBackend(Backend):
         return execute_local_tasks(
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
-            verbose=verbose, n_threads=1)
+            verbose=verbose, n_threads=1,
+            progress=getattr(self, "progress", None))
 
 
 class ThreadBackend(Backend):
@@ -115,4 +134,5 @@ class ThreadBackend(Backend):
         return execute_local_tasks(
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
-            verbose=verbose, n_threads=self.n_threads)
+            verbose=verbose, n_threads=self.n_threads,
+            progress=getattr(self, "progress", None))

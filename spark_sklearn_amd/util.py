@@ -13,7 +13,8 @@ from __future__ import annotations
 import os
 from typing import Optional
 
-__all__ = ["HipContext", "createLocalContext", "init_distributed"]
+__all__ = ["HipContext", "createLocalContext", "init_distributed",
+           "enable_prometheus"]
 
 
 class HipContext:
@@ -105,3 +106,37 @@ def init_distributed(backend: Optional[str] = None,
             backend=backend,
             timeout=datetime.timedelta(seconds=timeout_seconds))
     return HipContext(backend="dist")
+
+
+# -- optional Prometheus metrics -------------------
+_PROM: dict = {"enabled": False}
+
+
+def enable_prometheus(port: int = 9109):
+    """Expose search counters on a Prometheus endpoint (optional --
+    requires prometheus_client, present in this image's wheelhouse).
+
+    Metrics: ``ska_candidate_fits_total``, ``ska_searches_total``,
+    ``ska_fits_per_sec`` (last search's rate -- the north-star metric).
+    """
+    if _PROM["enabled"]:
+        return
+    from prometheus_client import Counter, Gauge, start_http_server
+
+    start_http_server(port)
+    _PROM.update(
+        enabled=True,
+        fits=Counter("ska_candidate_fits_total",
+                     "completed (candidate, fold) fits"),
+        searches=Counter("ska_searches_total", "completed searches"),
+        fps=Gauge("ska_fits_per_sec",
+                  "candidate-fits/sec of the last search"))
+
+
+def record_search_metrics(n_fits: int, fits_per_sec: float):
+    """Called by BaseSearchCV.fit; no-op unless enable_prometheus ran."""
+    if not _PROM["enabled"]:
+        return
+    _PROM["fits"].inc(n_fits)
+    _PROM["searches"].inc()
+    _PROM["fps"].set(fits_per_sec)

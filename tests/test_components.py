@@ -239,3 +239,98 @@ def test_score_log_resume(tmp_path):
     np.testing.assert_allclose(gs2.cv_results_["mean_test_score"],
                                ref.cv_results_["mean_test_score"],
                                rtol=1e-12)
+
+
+def test_progress_callback():
+    """HipContext(progress=fn) receives (done, total) ticks ending at
+    (total, total); a raising callback must not break the search."""
+    import numpy as np
+
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd import HipContext
+    from sklearn.svm import LinearSVC
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((200, 6))
+    y = (X[:, 0] > 0).astype(np.int64)
+
+    calls = []
+    ctx = HipContext(backend="sequential",
+                     progress=lambda d, t: calls.append((d, t)))
+    ska.GridSearchCV(ctx, LinearSVC(), {"C": [0.1, 1.0]}, cv=3).fit(X, y)
+    assert calls and calls[-1] == (6, 6)
+    assert all(t == 6 and 1 <= d <= 6 for d, t in calls)
+
+    def boom(d, t):
+        raise RuntimeError("callback bug")
+
+    ctx2 = HipContext(backend="sequential", progress=boom)
+    gs = ska.GridSearchCV(ctx2, LinearSVC(), {"C": [1.0]}, cv=3).fit(X, y)
+    assert hasattr(gs, "best_score_")
+
+
+def test_hip_error_retry(monkeypatch):
+    """A transient HIP RuntimeError on a candidate chunk retries once
+    (SURVEY.md section 5.3a); persistent failure propagates."""
+    import numpy as np
+
+    from sklearn.model_selection import KFold
+
+    from spark_sklearn_amd.base_search import Task
+    from spark_sklearn_amd.engine import (LinearFamily, run_batched_tasks,
+                                          _HostLinearExecutor)
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((200, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+
+    fails = {"n": 1}
+    orig_fit = _HostLinearExecutor.fit
+
+    def flaky_fit(self, train_idx, plan):
+        if fails["n"] > 0:
+            fails["n"] -= 1
+            raise RuntimeError("HIP error: out of resources (fake)")
+        return orig_fit(self, train_idx, plan)
+
+    monkeypatch.setattr(_HostLinearExecutor, "fit", flaky_fit)
+    splits = list(KFold(3).split(np.arange(len(X))))
+    tasks = [Task(i, 0, f, {"C": 1.0}, tr, te)
+             for i, (f, (tr, te)) in enumerate(enumerate(splits))]
+    with pytest.warns(RuntimeWarning, match="retrying candidate chunk"):
+        results = run_batched_tasks(
+            LogisticRegression(max_iter=5), X, y, tasks, scorer=None,
+            fit_params=None, return_train_score=False,
+            error_score="raise", verbose=0, device=None)
+    assert len(results) == 3 and all(r.error is None for r in results)
+
+
+def test_prometheus_metrics():
+    """Optional observability endpoint."""
+    pytest.importorskip("prometheus_client")
+    import socket
+    import urllib.request
+
+    import numpy as np
+
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.util import enable_prometheus
+    from sklearn.svm import LinearSVC
+
+    sock = socket.socket()
+    sock.bind(("127.0.0.1", 0))
+    port = sock.getsockname()[1]
+    sock.close()
+    enable_prometheus(port)
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((150, 5))
+    y = (X[:, 0] > 0).astype(np.int64)
+    ska.GridSearchCV("sequential", LinearSVC(), {"C": [0.1, 1.0]},
+                     cv=3).fit(X, y)
+    body = urllib.request.urlopen(
+        f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+    assert "ska_candidate_fits_total 6.0" in body
+    assert "ska_searches_total 1.0" in body
+    assert "ska_fits_per_sec" in body
