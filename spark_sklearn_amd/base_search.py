@@ -30,8 +30,7 @@ import numbers
 import time
 import warnings
 from collections import defaultdict
-from functools import partial
-from typing import Any, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Dict, List
 
 import numpy as np
 from numpy.ma import MaskedArray

@@ -28,7 +28,7 @@ from __future__ import annotations
 import os
 import time
 import warnings
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Optional
 
 import numpy as np
 

@@ -31,7 +31,7 @@ Update:   V = momentum*V - lr*g;  W += V
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import numpy as np

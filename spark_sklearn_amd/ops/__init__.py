@@ -11,7 +11,6 @@ always exercise the native kernels.
 
 from __future__ import annotations
 
-import os
 
 _ext = None
 _ext_err: Exception | None = None

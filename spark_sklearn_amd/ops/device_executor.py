@@ -14,8 +14,6 @@ one is installed with set_layout().
 
 from __future__ import annotations
 
-import math
-from typing import Optional
 
 import numpy as np
 

@@ -5,7 +5,7 @@ from __future__ import annotations
 import numpy as np
 
 from . import load_extension
-from ..models._gbt_ref import (LOSS_LOGISTIC, bin_data, quantile_bins)
+from ..models._gbt_ref import LOSS_LOGISTIC, quantile_bins
 
 __all__ = ["DeviceGBTExecutor"]
 
