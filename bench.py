@@ -139,11 +139,15 @@ def main():
             device=device if use_gpu else None,
             column_batch=256)
         if dist is not None:
-            # RC2: fold-score records, fixed width, one all-reduce per drain
-            table = torch.zeros((len(tasks), 6), dtype=torch.float64,
+            # RC2: the GLOBAL fold-score table (world x local tasks rows);
+            # each rank fills its own row block, one sum-all-reduce gives
+            # every rank the full table -- the real search path's shape
+            table = torch.zeros((world * len(tasks), 6),
+                                dtype=torch.float64,
                                 device=device if use_gpu else "cpu")
+            base = rank * len(tasks)
             for r in results:
-                table[r.index, 1] = r.test_score
+                table[base + r.index, 1] = r.test_score
             dist.all_reduce(table)
         return results
 
