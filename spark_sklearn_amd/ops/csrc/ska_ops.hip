@@ -1511,6 +1511,11 @@ __global__ void gbt_bin_kernel(const float* __restrict__ X,
 // LDS atomic throughput is the bound (profiles/r01_families_kernels.md).
 // Rows keep a heap node_id; settled rows are skipped by the level range
 // check; odd nodes are skipped under the sibling-subtraction trick.
+// gscale > 0 selects the DETERMINISTIC mode: g is accumulated as a
+// biased fixed-point u64 (g+1 >= 0, so plain unsigned adds; unbiased at
+// flush using the exact count) -- integer atomics commute exactly, so
+// tree structures are bit-stable run to run.  gscale == 0 keeps the
+// default float-atomic g (faster: 12 B/cell vs 16 B).
 __global__ __launch_bounds__(256)
 void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const int* __restrict__ idxmap,          // [nt]
@@ -1519,17 +1524,24 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const float* __restrict__ h,
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
                      long nt, int F, int lvl_lo, int lvl_nodes, int f_pb,
-                     long rows_pb, int even_only, float hscale) {
+                     long rows_pb, int even_only, float hscale,
+                     float gscale) {
   extern __shared__ char smem[];
   const int cellsg = lvl_nodes * f_pb * 256;
+  const bool gfix = gscale > 0.f;
   unsigned long long* lhc = reinterpret_cast<unsigned long long*>(smem);
+  unsigned long long* lgq = lhc + cellsg;            // gfix mode only
   float* lg = reinterpret_cast<float*>(smem + (size_t)cellsg * 8);
   const int tid = threadIdx.x;
   const int f0 = blockIdx.y * f_pb;
   const long r0 = (long)blockIdx.x * rows_pb;
   const long r1 = min(r0 + rows_pb, nt);
 
-  for (int i = tid; i < cellsg; i += 256) { lhc[i] = 0ull; lg[i] = 0.f; }
+  if (gfix) {
+    for (int i = tid; i < 2 * cellsg; i += 256) lhc[i] = 0ull;
+  } else {
+    for (int i = tid; i < cellsg; i += 256) { lhc[i] = 0ull; lg[i] = 0.f; }
+  }
   __syncthreads();
 
   const bool vec8 = (F % 8 == 0) && (f_pb == 8);
@@ -1539,6 +1551,9 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
     if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
     long gi = idxmap[r];
     float gr = g[r];
+    unsigned long long gq = 0ull;
+    if (gfix) gq = (unsigned long long)((double)(gr + 1.f) *
+                                        (double)gscale + 0.5);
     unsigned hfx = __float2uint_rn(h[r] * hscale);
     unsigned long long hpack = ((unsigned long long)hfx << 32) | 1ull;
     const int base = (nd - lvl_lo) * f_pb;
@@ -1549,13 +1564,15 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int cell = (base + j) * 256 + (int)((bv >> (8 * j)) & 0xff);
-        atomicAdd(&lg[cell], gr);
+        if (gfix) atomicAdd(&lgq[cell], gq);
+        else atomicAdd(&lg[cell], gr);
         atomicAdd(&lhc[cell], hpack);
       }
     } else {
       for (int ff = 0; ff < f_pb; ++ff) {
         int cell = (base + ff) * 256 + (int)xrow[ff];
-        atomicAdd(&lg[cell], gr);
+        if (gfix) atomicAdd(&lgq[cell], gq);
+        else atomicAdd(&lg[cell], gr);
         atomicAdd(&lhc[cell], hpack);
       }
     }
@@ -1567,11 +1584,16 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
     if (p == 0ull) continue;          // count==0 -> cell never touched
     float hv = (float)(p >> 32) * hinv;
     unsigned cnt = (unsigned)(p & 0xffffffffull);
+    float gv;
+    if (gfix)
+      gv = (float)((double)lgq[i] / (double)gscale - (double)cnt);
+    else
+      gv = lg[i];
     int bin = i & 255;
     int ff = (i >> 8) % f_pb;
     int ndl = i / (256 * f_pb);
     float* cell = &hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
-    atomicAdd(cell, lg[i]);
+    atomicAdd(cell, gv);
     atomicAdd(cell + 1, hv);
     atomicAdd(cell + 2, (float)cnt);
   }
@@ -1581,25 +1603,52 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
 // gbt_fit_multi.  Picks features-per-block to fit 96 KB LDS, row
 // segments >=8192 rows (amortizes the global flush), and the largest
 // power-of-two h scale whose worst-case sum stays under 2^31.
+// SKA_DETERMINISTIC_GBT=1 selects bit-stable histograms for losses with
+// |g| <= 1 (logistic, softmax): g accumulates as biased fixed-point u64,
+// so integer-add commutativity makes tree structures identical run to
+// run.  Costs 16 B/LDS cell instead of 12 (fewer features per block at
+// deep levels) -- opt-in.
+static bool det_gbt_mode() {
+  // read per call (cheap, once per level launch) so tests can toggle
+  // the env var inside one process
+  const char* e = getenv("SKA_DETERMINISTIC_GBT");
+  return e != nullptr && e[0] == '1';
+}
+
+static float det_fix_scale(long nt) {
+  double cap = 4.0e18 / (2.0 * (double)nt + 1.0);
+  float gs = 1.f;
+  while ((double)gs * 2.0 < cap && gs < 1.0e12f) gs *= 2.f;
+  return gs;
+}
+
 static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
                               const int* node_id, const float* g,
                               const float* h, float* hist, long nt, int F,
-                              int d, float hmax, hipStream_t s) {
+                              int d, float hmax, bool g_bounded,
+                              hipStream_t s) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
-  int f_pb = std::min(8, std::max(1, (96 * 1024) / (3072 * lvl_nodes)));
+  bool gfix = g_bounded && det_gbt_mode();
+  int cell_b = gfix ? 16 : 12;
+  int f_pb = std::min(8, std::max(1, (96 * 1024) /
+                                  (cell_b * 256 * lvl_nodes)));
   while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
   int fchunks = F / f_pb;
   long rowsegs = std::max<long>(
       1, std::min<long>(2048 / fchunks, (nt + 8191) / 8192));
+  if (gfix) rowsegs = 1;   // one writer per global cell: the float flush
+                           // adds stay single-sourced, so bit-stable
   long rows_pb = (nt + rowsegs - 1) / rowsegs;
   rowsegs = (nt + rows_pb - 1) / rows_pb;
-  size_t lsm = (size_t)lvl_nodes * f_pb * 256 * 12;
+  size_t lsm = (size_t)lvl_nodes * f_pb * 256 * cell_b;
   int even_only = (d > 0) ? 1 : 0;       // sibling subtraction
   float hs = 1.f;
   while (hs * 2.f * hmax * (float)nt + (float)nt < 1.8e9f &&
          hs < 65536.f)
     hs *= 2.f;
+  // biased g in [0,2]: keep nt*(2*gs + 1) well under 2^62
+  float gs = gfix ? det_fix_scale(nt) : 0.f;
   HIP_CHECK(hipMemsetAsync(hist, 0,
                            (size_t)lvl_nodes * F * 256 * 3 * 4, s));
   HIP_CHECK(hipFuncSetAttribute(
@@ -1608,7 +1657,7 @@ static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
   hipLaunchKernelGGL(gbt_hist_kernel,
                      dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(256),
                      lsm, s, Xp, idxmap, node_id, g, h, hist, nt, F,
-                     lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs);
+                     lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs, gs);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -1752,30 +1801,56 @@ __global__ void gbt_split_apply_kernel(const unsigned char* __restrict__ Xb,
 // bottom-level leaf values: per-node g/h sums.  LDS-staged per block --
 // direct global atomics serialized 1M adds onto <=64 node addresses
 // (4.8 ms/tree measured; this form is ~100x cheaper).
+// gscale > 0 selects deterministic fixed-point accumulation (biased g,
+// exact count to unbias; integer atomics commute) -- launched with a
+// single block in that mode so the global add has one writer.
 __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
                                      const float* __restrict__ g,
                                      const float* __restrict__ h,
                                      float* __restrict__ gsum,
                                      float* __restrict__ hsum,
-                                     long nt, int lvl_lo, int lvl_hi) {
+                                     long nt, int lvl_lo, int lvl_hi,
+                                     float gscale, float hscale) {
   __shared__ float lg[128], lh[128];
+  __shared__ unsigned long long lgq[128], lhq[128], lcq[128];
+  const bool fix = gscale > 0.f;
   int nn = lvl_hi - lvl_lo;
   for (int i = threadIdx.x; i < nn; i += blockDim.x) {
-    lg[i] = 0.f;
-    lh[i] = 0.f;
+    if (fix) { lgq[i] = 0ull; lhq[i] = 0ull; lcq[i] = 0ull; }
+    else { lg[i] = 0.f; lh[i] = 0.f; }
   }
   __syncthreads();
   long r = blockIdx.x * blockDim.x + threadIdx.x;
   for (; r < nt; r += (long)gridDim.x * blockDim.x) {
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_hi) continue;
-    atomicAdd(&lg[nd - lvl_lo], g[r]);
-    atomicAdd(&lh[nd - lvl_lo], h[r]);
+    if (fix) {
+      atomicAdd(&lgq[nd - lvl_lo],
+                (unsigned long long)((double)(g[r] + 1.f) *
+                                     (double)gscale + 0.5));
+      atomicAdd(&lhq[nd - lvl_lo],
+                (unsigned long long)((double)h[r] *
+                                     (double)hscale + 0.5));
+      atomicAdd(&lcq[nd - lvl_lo], 1ull);
+    } else {
+      atomicAdd(&lg[nd - lvl_lo], g[r]);
+      atomicAdd(&lh[nd - lvl_lo], h[r]);
+    }
   }
   __syncthreads();
   for (int i = threadIdx.x; i < nn; i += blockDim.x) {
-    if (lg[i] != 0.f) atomicAdd(&gsum[lvl_lo + i], lg[i]);
-    if (lh[i] != 0.f) atomicAdd(&hsum[lvl_lo + i], lh[i]);
+    if (fix) {
+      if (lcq[i] != 0ull) {
+        atomicAdd(&gsum[lvl_lo + i],
+                  (float)((double)lgq[i] / (double)gscale -
+                          (double)lcq[i]));
+        atomicAdd(&hsum[lvl_lo + i],
+                  (float)((double)lhq[i] / (double)hscale));
+      }
+    } else {
+      if (lg[i] != 0.f) atomicAdd(&gsum[lvl_lo + i], lg[i]);
+      if (lh[i] != 0.f) atomicAdd(&hsum[lvl_lo + i], lh[i]);
+    }
   }
 }
 
@@ -3061,7 +3136,8 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
       launch_hist_level(Xp, idxmap.data_ptr<int>(),
                         node_id.data_ptr<int>(), g.data_ptr<float>(),
                         h.data_ptr<float>(), hist.data_ptr<float>(), nt,
-                        F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f, s);
+                        F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f,
+                        loss == LOSS_LOGISTIC, s);
       if (even_only) {
         int prev_lo = (1 << (d - 1)) - 1;
         long cells = (long)F * 256 * 3;
@@ -3090,10 +3166,15 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
     int bl_lo = (1 << max_depth) - 1;
     HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
     HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
-    hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0, s,
-                       node_id.data_ptr<int>(), g.data_ptr<float>(),
-                       h.data_ptr<float>(), gsum.data_ptr<float>(),
-                       hsum.data_ptr<float>(), nt, bl_lo, NN);
+    {
+      bool fix = det_gbt_mode() && loss == LOSS_LOGISTIC;
+      float fsc = fix ? det_fix_scale(nt) : 0.f;
+      hipLaunchKernelGGL(gbt_node_sums_kernel,
+                         dim3(fix ? 1 : blocks), dim3(256), 0, s,
+                         node_id.data_ptr<int>(), g.data_ptr<float>(),
+                         h.data_ptr<float>(), gsum.data_ptr<float>(),
+                         hsum.data_ptr<float>(), nt, bl_lo, NN, fsc, fsc);
+    }
     hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                        dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
                        gsum.data_ptr<float>(), hsum.data_ptr<float>(),
@@ -3570,7 +3651,8 @@ std::vector<torch::Tensor> gbt_fit_multi(
         int even_only = (d > 0) ? 1 : 0;
         launch_hist_level(Xp, idxmap.data_ptr<int>(),
                           node_id.data_ptr<int>(), gc, hc,
-                          hist.data_ptr<float>(), nt, F, d, 0.25f, s);
+                          hist.data_ptr<float>(), nt, F, d, 0.25f,
+                          /*g_bounded=*/true, s);
         if (even_only) {
           int prev_lo = (1 << (d - 1)) - 1;
           long cells = (long)F * 256 * 3;
@@ -3599,10 +3681,15 @@ std::vector<torch::Tensor> gbt_fit_multi(
       int bl_lo = (1 << max_depth) - 1;
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
-      hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0,
-                         s, node_id.data_ptr<int>(), gc, hc,
-                         gsum.data_ptr<float>(), hsum.data_ptr<float>(),
-                         nt, bl_lo, NN);
+      {
+        bool fix = det_gbt_mode();
+        float fsc = fix ? det_fix_scale(nt) : 0.f;
+        hipLaunchKernelGGL(gbt_node_sums_kernel,
+                           dim3(fix ? 1 : blocks), dim3(256), 0, s,
+                           node_id.data_ptr<int>(), gc, hc,
+                           gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                           nt, bl_lo, NN, fsc, fsc);
+      }
       hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                          dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
                          gsum.data_ptr<float>(), hsum.data_ptr<float>(),

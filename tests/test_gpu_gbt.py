@@ -159,3 +159,37 @@ def test_device_multiclass_grid_search(problem):
         {"max_depth": [2, 3]}, cv=3, return_train_score=False).fit(X, y)
     assert gs.best_score_ > 0.9
     assert gs.predict(X[:9]).shape == (9,)
+
+
+def test_deterministic_mode(problem, monkeypatch):
+    """SKA_DETERMINISTIC_GBT=1: fixed-point histogram/node-sum
+    accumulation makes tree structures bit-stable run to run for
+    classification losses, and the trees still match the reference."""
+    import os
+
+    from spark_sklearn_amd.models._gbt_ref import (GBTSpec, gbt_fit_ref)
+
+    X, y = problem
+    dev = torch.device("cuda", 0)
+    spec = GBTSpec(n_estimators=6, max_depth=4, learning_rate=0.3,
+                   min_samples_leaf=20)
+    train = np.arange(15000, dtype=np.int64)
+
+    monkeypatch.setenv("SKA_DETERMINISTIC_GBT", "1")
+    outs = []
+    for _ in range(2):
+        ex = DeviceGBTExecutor(np.ascontiguousarray(X),
+                               y.astype(np.float32), dev)
+        (models, edges), _, _ = ex.fit(train, _GBTPlan([spec]))
+        trees, f0 = models[0]
+        outs.append([(tuple(f.tolist()), tuple(t.tolist()),
+                      tuple(v.tolist())) for f, t, v in trees])
+    assert outs[0] == outs[1]          # bit-identical, values included
+
+    # and still structure-identical to the numpy reference
+    Xb = bin_data(X, edges)
+    trees_r, _ = gbt_fit_ref(Xb[train], y[train].astype(np.float32), spec)
+    ref_struct = [(tuple(np.asarray(f).tolist()),
+                   tuple(np.asarray(t).tolist())) for f, t, v in trees_r]
+    dev_struct = [(f, t) for f, t, v in outs[0]]
+    assert dev_struct == ref_struct
