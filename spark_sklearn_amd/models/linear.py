@@ -50,6 +50,13 @@ class _LinearBase(BaseEstimator):
 
     def _spec_common(self, l1: float, l2: float, n: int,
                      mean_sq_row_norm: float) -> dict:
+        # NOTE (documented divergence): inside a search the auto lr uses
+        # the FULL matrix's mean squared row norm (fold-invariant, so B
+        # candidates x 5 folds share one plan); a standalone .fit() sees
+        # only its own rows.  Label-free scale statistic -- no leakage of
+        # fold targets -- but trajectories differ slightly between the
+        # two entry points (tests/test_fuzz.py pins the invariant that
+        # DOES matter: column independence within a batch).
         lr = self.learning_rate
         if lr in (None, "auto"):
             lr = auto_lr(mean_sq_row_norm, self._loss_id, l2)
