@@ -58,6 +58,18 @@ def ensure_layout(device) -> int:
     return best
 
 
+def _pad_features(F: int) -> int:
+    """Feature padding to a kernel-supported width: the fit kernels accept
+    Fp in {64, 128, k*256} (TORCH_CHECK in linear_fit).  Zero-padded
+    feature columns are inert: their gradients are exactly zero, so the
+    extra weights stay zero through L1/L2 too."""
+    if F <= 64:
+        return 64
+    if F <= 128:
+        return 128
+    return _roundup(F, 256)
+
+
 def _as_device_bf16(X, device):
     """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array.
 
@@ -80,7 +92,7 @@ def _as_device_bf16(X, device):
     if sp.issparse(X):
         Xc = sp.csr_matrix(X)
         n, F = Xc.shape
-        Fp = max(64, _roundup(F, 64))
+        Fp = _pad_features(F)
         dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
         step = max(1, (1 << 28) // max(F, 1))   # ~256M elems per chunk
         for lo in range(0, n, step):
@@ -93,7 +105,7 @@ def _as_device_bf16(X, device):
         t = X if isinstance(X, torch.Tensor) else torch.as_tensor(
             np.ascontiguousarray(np.asarray(X, dtype=np.float32)))
         n, F = t.shape
-        Fp = max(64, _roundup(F, 64))
+        Fp = _pad_features(F)
         dev_t = torch.zeros((n, Fp), dtype=torch.bfloat16, device=device)
         dev_t[:, :F] = t.to(device).to(torch.bfloat16)
     # resident transpose: the bwd kernel's B-operand stages coalesced from
@@ -158,7 +170,10 @@ class DeviceLinearExecutor:
         NTp = idx.shape[0]
 
         B = plan.B
-        Bp = max(64, _roundup(B, 64))
+        # kernel contract: Bp in {64, 128, 256}; padded columns are inert
+        Bp = 64 if B <= 64 else (128 if B <= 128 else 256)
+        if B > 256:
+            raise RuntimeError(f"candidate batch {B} > 256 columns")
 
         def pad_i32(a, fill=0):
             out = np.full(Bp, fill, dtype=np.int32)
