@@ -105,3 +105,18 @@ def test_fuzz_device_kmeans_matches_host(n, f, k, seed):
         rel = abs(dev[idx][0] - host[idx][0]) / max(abs(host[idx][0]),
                                                     1e-9)
         assert rel < 0.02, (n, f, k, idx, dev[idx], host[idx])
+
+
+@pytest.mark.parametrize("f", [129, 192, 300])
+def test_odd_feature_widths_regression(f):
+    """Pinned fuzz findings: feature counts padding past 128 must route
+    through a kernel-supported Fp (129->256, 300->512)."""
+    from spark_sklearn_amd.models import LogisticRegression
+
+    X, y = _problem(2000, f, 3, seed=1)
+    est = LogisticRegression(max_iter=6, tol=0.0)
+    cands = [{"C": 1.0}]
+    dev = _run(est, X, y, cands, torch.device("cuda", 0))
+    host = _run(est, X, y, cands, None)
+    for idx in dev:
+        assert abs(dev[idx][0] - host[idx][0]) < 0.03
