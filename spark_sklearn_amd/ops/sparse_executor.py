@@ -105,8 +105,9 @@ class SparseLinearExecutor:
         torch = self.torch
         nt = len(train_idx)
         B = plan.B
-        Bp = max(64, _roundup(B, 64))
-        if Bp > 256:
+        # csr_linear_fit contract: Bp in {64, 128, 256}
+        Bp = 64 if B <= 64 else (128 if B <= 128 else 256)
+        if B > 256:
             raise RuntimeError("chunk larger batches to <= 256 columns")
 
         def pad_i32(a, fill=0):
