@@ -2221,6 +2221,140 @@ void scores3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 
 
 
+// ---------------------------------------------------------------------------
+// scores7_kernel: 1024-thread scoring for C==1 candidates at Bp=256 --
+// fwd_d7's 256x256 GEMM tile with a metric-reduction epilogue instead of
+// the D write.  scores3's 32-row tiles re-read the whole W per block
+// (~800 MB of L2 traffic per dispatch at 200k test rows); 256-row tiles
+// cut that 8x.  Multiclass (C>1) needs the per-row zbuf for argmax and
+// stays on scores3.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void scores7_kernel(const __bf16* __restrict__ X,
+                    const float* __restrict__ y,
+                    const long* __restrict__ idx,
+                    const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    int metric, long n_idx, int Fp, int Bp, int n_cand,
+                    double* __restrict__ out_sum) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sVal[256];
+  __shared__ long sRow[256];
+  __shared__ float sY[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int srow = tid >> 2;
+  const int sseg = tid & 3;
+
+  if (tid < 256) {
+    long r = (long)rt * 256 + tid;
+    long gi = (r < n_idx) ? idx[r] : idx[n_idx - 1];
+    sRow[tid] = gi;
+    sY[tid] = y[gi];
+    sVal[tid] = 0.f;
+  }
+  __syncthreads();
+
+  bf16x8 rx[2], rw[2];
+  auto issue = [&](int kc) {
+    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+        X + sRow[srow] * Fp + kc + sseg * 16);
+    rx[0] = gx[0];
+    rx[1] = gx[1];
+    const bf16x8* gw = reinterpret_cast<const bf16x8*>(
+        Wt + (long)srow * Fp + kc + sseg * 16);
+    rw[0] = gw[0];
+    rw[1] = gw[1];
+  };
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  issue(0);
+  commit(0);
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rquad * 64 + m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (kc + 1 < nkc) commit(b ^ 1);
+    __syncthreads();
+  }
+
+  const float eps = 1e-12f;
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    float vsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long gr = (long)rt * 256 + row_base + r;
+        if (gr >= n_idx) continue;
+        float z = acc[m][fn][r] + bia;
+        float yv = sY[row_base + r];
+        float val;
+        if (metric == 2) {
+          float d = yv - z;
+          val = d * d;
+        } else if (metric == 0) {
+          int pred = (z > 0.f) ? 1 : 0;
+          val = (pred == (int)yv) ? 1.f : 0.f;
+        } else {
+          float pr = 1.f / (1.f + __expf(-z));
+          float t = (yv == 1.f) ? 1.f : 0.f;
+          val = t * __logf(pr + eps) +
+                (1.f - t) * __logf(1.f - pr + eps);
+        }
+        vsum += val;
+      }
+    }
+    if (col_in < n_cand) atomicAdd(&sVal[col_in], vsum);
+  }
+  __syncthreads();
+  if (tid < n_cand)
+    atomicAdd(&out_sum[tid], (double)sVal[tid]);
+}
+
+
 // multiclass softmax gradients: raw [K][nt] class-major
 __global__ void gbt_softmax_grad_kernel(const float* __restrict__ raw,
                                         const float* __restrict__ y,
@@ -2805,6 +2939,20 @@ torch::Tensor linear_scores(torch::Tensor X, torch::Tensor y,
   };
   int nf = Bp / 64;
   if (g_layout == 3) {
+    if (C == 1 && Bp == 256 && n_idx >= 4096 && Fp % 128 == 0) {
+      // wide tile: 8x less W traffic than the 32-row scores3 blocks
+      hipLaunchKernelGGL(scores7_kernel,
+                         dim3((unsigned)((n_idx + 255) / 256)), dim3(1024),
+                         0, s,
+                         reinterpret_cast<const __bf16*>(X.data_ptr()),
+                         y.data_ptr<float>(), idx.data_ptr<long>(),
+                         reinterpret_cast<const __bf16*>(
+                             Wt_bf16.data_ptr()),
+                         bias.data_ptr<float>(), (int)metric, n_idx, Fp,
+                         Bp, (int)n_cand, out.data_ptr<double>());
+      HIP_CHECK(hipGetLastError());
+      return out;
+    }
     switch (nf) {
       case 1: launch3(scores3_kernel<1>); break;
       case 2: launch3(scores3_kernel<2>); break;
@@ -3407,39 +3555,42 @@ std::vector<torch::Tensor> linear_search_run(
         size_t scand_pad = ((size_t)n_cand_chunk + 3) & ~(size_t)3;
         size_t smem3 = 32 * 8 + 32 * (Bp + 8) * 4 + scand_pad * 4 +
                        32 * 64 * 2 + (size_t)Bp * 64 * 2;
-        HIP_CHECK(hipFuncSetAttribute(
-            reinterpret_cast<const void*>(scores3_kernel<4>),
-            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem3));
-        hipLaunchKernelGGL(scores3_kernel<4>,
-                           dim3((unsigned)((n_t + 31) / 32)), dim3(256),
-                           smem3, s,
-                           reinterpret_cast<const __bf16*>(X.data_ptr()),
-                           y.data_ptr<float>(),
-                           test_idx[f].data_ptr<long>(),
-                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
-                           bias.data_ptr<float>(), loss_p, tgt_p, (int)C,
-                           (int)metric, n_t, Fp, Bp, n_cand_chunk,
-                           test_sums.data_ptr<double>() +
-                               (long)f * n_cand_total +
-                               (long)c * cands_per_chunk);
-        if (want_train_scores) {
-          // score the train fold with the padded i32 indices reused as i64
-          // is not possible; train scoring reuses the i64 view built by
-          // the caller in test-idx style (train_idx64 appended after the
-          // n_folds test entries)
-          long n_tr = test_idx[n_folds + f].size(0);
+        bool wide = (C == 1 && Bp == 256 && Fp % 128 == 0);
+        auto score_one = [&](torch::Tensor& it, long ni, double* outp) {
+          if (wide && ni >= 4096) {
+            hipLaunchKernelGGL(scores7_kernel,
+                               dim3((unsigned)((ni + 255) / 256)),
+                               dim3(1024), 0, s,
+                               reinterpret_cast<const __bf16*>(
+                                   X.data_ptr()),
+                               y.data_ptr<float>(), it.data_ptr<long>(),
+                               reinterpret_cast<const __bf16*>(
+                                   Wt.data_ptr()),
+                               bias.data_ptr<float>(), (int)metric, ni,
+                               Fp, Bp, n_cand_chunk, outp);
+            return;
+          }
+          HIP_CHECK(hipFuncSetAttribute(
+              reinterpret_cast<const void*>(scores3_kernel<4>),
+              hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem3));
           hipLaunchKernelGGL(scores3_kernel<4>,
-                             dim3((unsigned)((n_tr + 31) / 32)), dim3(256),
+                             dim3((unsigned)((ni + 31) / 32)), dim3(256),
                              smem3, s,
                              reinterpret_cast<const __bf16*>(X.data_ptr()),
-                             y.data_ptr<float>(),
-                             test_idx[n_folds + f].data_ptr<long>(),
+                             y.data_ptr<float>(), it.data_ptr<long>(),
                              reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                              bias.data_ptr<float>(), loss_p, tgt_p, (int)C,
-                             (int)metric, n_tr, Fp, Bp, n_cand_chunk,
-                             train_sums.data_ptr<double>() +
-                                 (long)f * n_cand_total +
-                                 (long)c * cands_per_chunk);
+                             (int)metric, ni, Fp, Bp, n_cand_chunk, outp);
+        };
+        score_one(test_idx[f], n_t,
+                  test_sums.data_ptr<double>() + (long)f * n_cand_total +
+                      (long)c * cands_per_chunk);
+        if (want_train_scores) {
+          long n_tr = test_idx[n_folds + f].size(0);
+          score_one(test_idx[n_folds + f], n_tr,
+                    train_sums.data_ptr<double>() +
+                        (long)f * n_cand_total +
+                        (long)c * cands_per_chunk);
         }
       }
       HIP_CHECK(hipEventRecord(ev_s1, s));
