@@ -48,7 +48,9 @@ def _device_available() -> bool:
 def execute_local_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                         fit_params, return_train_score, error_score, verbose,
                         n_threads: Optional[int] = None,
-                        device=None, progress=None) -> List[TaskResult]:
+                        device=None, progress=None,
+                        column_batch: Optional[int] = None
+                        ) -> List[TaskResult]:
     """Execute this process's share of tasks, on device or host.
 
     Device path: candidates batched per fold into single HIP kernel launches
@@ -61,10 +63,13 @@ def execute_local_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     if _is_hip_batched(estimator) and (device is not None or _device_available()):
         from ..engine import run_batched_tasks
 
+        kw = {}
+        if column_batch:
+            kw["column_batch"] = int(column_batch)
         return run_batched_tasks(
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
-            verbose=verbose, device=device, progress=progress)
+            verbose=verbose, device=device, progress=progress, **kw)
 
     n_threads = n_threads or min(len(tasks), os.cpu_count() or 4)
     if n_threads <= 1 or len(tasks) == 1:
@@ -120,7 +125,8 @@ class SequentialBackend(Backend):
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
             verbose=verbose, n_threads=1,
-            progress=getattr(self, "progress", None))
+            progress=getattr(self, "progress", None),
+            column_batch=getattr(self, "candidate_batch", None))
 
 
 class ThreadBackend(Backend):
@@ -135,4 +141,5 @@ class ThreadBackend(Backend):
             estimator, X, y, tasks, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
             verbose=verbose, n_threads=self.n_threads,
-            progress=getattr(self, "progress", None))
+            progress=getattr(self, "progress", None),
+            column_batch=getattr(self, "candidate_batch", None))

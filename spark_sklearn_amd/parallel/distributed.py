@@ -119,7 +119,8 @@ class DistributedBackend(Backend):
             estimator, X, y, local, scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
             verbose=verbose, n_threads=self.n_threads, device=self.device,
-            progress=getattr(self, "progress", None))
+            progress=getattr(self, "progress", None),
+            column_batch=getattr(self, "candidate_batch", None))
 
         # RC2: dense float64 score table, one sum-all-reduce.  Single
         # metric: (n_tasks, 7) fixed-width records.  Multimetric: the

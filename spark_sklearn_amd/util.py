@@ -27,7 +27,7 @@ class HipContext:
     resident matrix.
     """
 
-    def __init__(self, backend=None, device=None, candidate_batch: int = 128,
+    def __init__(self, backend=None, device=None, candidate_batch: int = 256,
                  n_streams: int = 2, dtype: str = "bfloat16",
                  n_threads: Optional[int] = None, score_log: str = None,
                  resume_from_log: str = None, progress=None):
