@@ -28,11 +28,24 @@ class HipContext:
     """
 
     def __init__(self, backend=None, device=None, candidate_batch: int = 256,
-                 n_streams: int = 2, dtype: str = "bfloat16",
+                 n_streams: int = 1, dtype: str = "bfloat16",
                  n_threads: Optional[int] = None, score_log: str = None,
                  resume_from_log: str = None, progress=None):
         from .parallel import resolve_backend
 
+        if dtype != "bfloat16":
+            raise ValueError(
+                "dtype='bfloat16' is the only resident dtype in round 1 "
+                "(the MFMA fit kernels are bf16 with fp32 accumulation); "
+                "fp8 residency is on the roadmap")
+        if n_streams != 1:
+            raise ValueError(
+                "n_streams is reserved: the native runner currently "
+                "drives one compute stream per GPU (copies are async on "
+                "the same stream); multi-stream fold overlap is on the "
+                "roadmap")
+        if not (1 <= int(candidate_batch) <= 256):
+            raise ValueError("candidate_batch must be in [1, 256]")
         self.candidate_batch = candidate_batch
         self.n_streams = n_streams
         self.dtype = dtype
