@@ -504,3 +504,38 @@ def test_gbt_run_to_run_stability(problem):
         fitted, _, _ = ex.fit(train, plan)
         accs.append(ex.scores(test, fitted, plan, "accuracy")[0])
     assert abs(accs[0] - accs[1]) < 1e-3
+
+
+def test_cuda_tensor_input_matches_numpy(problem):
+    """bench.py hands ranks a CUDA-resident X after the RC1 broadcast;
+    that tensor input path must score identically to numpy input."""
+    from sklearn.model_selection import KFold
+
+    from spark_sklearn_amd.base_search import Task
+    from spark_sklearn_amd.engine import run_batched_tasks
+    from spark_sklearn_amd.models import LogisticRegression
+
+    X, y = problem
+    dev = torch.device("cuda", 0)
+    Xt = torch.as_tensor(X).to(dev)
+    est = LogisticRegression(max_iter=10, tol=0.0)
+    splits = list(KFold(3).split(np.arange(len(X))))
+
+    def run(data):
+        tasks = []
+        i = 0
+        for ci, p in enumerate([{"C": 0.5}, {"C": 5.0}]):
+            for fi, (tr, te) in enumerate(splits):
+                tasks.append(Task(i, ci, fi, p, tr, te))
+                i += 1
+        out = run_batched_tasks(
+            est, data, y, tasks, scorer=None, fit_params=None,
+            return_train_score=False, error_score="raise", verbose=0,
+            device=dev)
+        return {r.index: r.test_score for r in out}
+
+    a = run(X)
+    b = run(Xt)
+    assert a.keys() == b.keys()
+    for k in a:
+        assert abs(a[k] - b[k]) < 1e-12, (k, a[k], b[k])
