@@ -5,9 +5,9 @@ databricks/spark-sklearn (SURVEY.md is the spec of record): a drop-in
 ``GridSearchCV`` / ``RandomizedSearchCV`` whose (param-candidate x CV-fold)
 task graph is scheduled onto the 8 GPUs of one MI355X node instead of a
 Spark cluster -- training matrix broadcast once into each GPU's 288 GB HBM
-via RCCL over xGMI, per-GPU task queues on independent HIP streams,
+via RCCL over xGMI, candidate-round-robin task partition per GPU,
 candidate fits batched into hand-written CDNA4 MFMA kernels, fold scores
-all-gathered, refit on rank 0.
+all-reduced, refit on rank 0.
 
 Public surface mirrors the reference's ``__init__`` exports:
 GridSearchCV, RandomizedSearchCV, Converter, KeyedEstimator, KeyedModel,
@@ -22,7 +22,8 @@ from .group_apply import gapply
 from .keyed_models import KeyedEstimator, KeyedModel, SparkSklearnEstimator
 from .random_search import RandomizedSearchCV
 from .udt import CSRVectorUDT, DeviceCSRMatrix
-from .util import HipContext, createLocalContext, init_distributed
+from .util import (HipContext, createLocalContext, enable_prometheus,
+                   init_distributed)
 
 __version__ = "0.1.0"
 
@@ -31,4 +32,5 @@ __all__ = [
     "KeyedEstimator", "KeyedModel", "SparkSklearnEstimator",
     "gapply", "CSRVectorUDT", "DeviceCSRMatrix",
     "HipContext", "createLocalContext", "init_distributed",
+    "enable_prometheus",
 ]
