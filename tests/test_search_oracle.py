@@ -142,3 +142,23 @@ def test_pickle_roundtrip(digits_std):
     gs2 = pickle.loads(blob)
     assert np.array_equal(gs2.predict(X[:10]), gs.predict(X[:10]))
     assert gs2.best_params_ == gs.best_params_
+
+
+def test_reference_readme_example(digits_std):
+    """The reference README's signature workload: digits x RandomForest
+    grid (scaled down) through our search, vs the sklearn oracle."""
+    from sklearn.ensemble import RandomForestClassifier
+    from sklearn.model_selection import GridSearchCV as SKGS
+
+    X, y = digits_std
+    grid = {"max_depth": [3, None], "n_estimators": [10, 20],
+            "min_samples_split": [2, 10]}
+    est = RandomForestClassifier(random_state=0)
+    ours = ska.GridSearchCV(None, est, grid, cv=3,
+                            return_train_score=False).fit(X, y)
+    ref = SKGS(est, grid, cv=3, return_train_score=False).fit(X, y)
+    np.testing.assert_allclose(ours.cv_results_["mean_test_score"],
+                               ref.cv_results_["mean_test_score"],
+                               rtol=1e-12)
+    assert ours.best_params_ == ref.best_params_
+    assert ours.predict(X[:10]).shape == (10,)
