@@ -128,7 +128,9 @@ def halving_fit(search, X, y=None, groups=None, **fit_params):
     search.multimetric_ = False
 
     if search.refit:
-        best_estimator = clone(estimator).set_params(**search.best_params_)
+        best_estimator = clone(estimator).set_params(
+            **{k: clone(v, safe=False)
+               for k, v in search.best_params_.items()})
         t0 = time.perf_counter()
         refit_fn = getattr(backend, "refit", None)
         if refit_fn is not None:
