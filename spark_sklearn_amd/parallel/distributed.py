@@ -145,10 +145,13 @@ class DistributedBackend(Backend):
             return [v] * len(keys)     # failed fit: scalar error_score
 
         M = 1 if keys is None else len(keys)
+        # rows are POSITIONS in the task list passed in (task indices are
+        # global and may be sparse, e.g. under score-log resume)
+        row_of = {t.index: i for i, t in enumerate(tasks)}
         table = torch.zeros((len(tasks), 5 + 2 * M), dtype=torch.float64,
                             device=comm_dev)
         for r in results:
-            table[r.index] = torch.tensor(
+            table[row_of[r.index]] = torch.tensor(
                 [r.index] + _tvals(r.test_score) + _tvals(r.train_score) +
                 [r.n_test, r.fit_time, r.score_time,
                  1.0 if r.error is not None else 0.0],
