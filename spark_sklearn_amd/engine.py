@@ -309,6 +309,11 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             estimator, executor, by_fold, n_classes, metric,
             return_train_score, error_score, verbose)
         if fast is not None:
+            if progress is not None:
+                try:
+                    progress(len(fast), len(tasks))
+                except Exception:  # noqa: BLE001 - observability only
+                    pass
             return fast
 
     cols_per_cand = family.cols_per_cand(n_classes)

@@ -54,6 +54,13 @@ def halving_fit(search, X, y=None, groups=None, **fit_params):
     budget = max(1, int(full_budget * float(search.min_resource_frac)))
 
     backend = search._get_backend()
+    if getattr(backend, "resume_from_log", None):
+        import warnings
+
+        warnings.warn(
+            "resume_from_log is ignored by the halving path (rungs are "
+            "re-sharded per budget); run the unpruned search to resume",
+            UserWarning, stacklevel=2)
     merged_fit_params = dict(search.fit_params or {})
     merged_fit_params.update(fit_params)
 
