@@ -1977,25 +1977,22 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   }
   __syncthreads();
 
-  // two register sets: loads for chunk kc+2 issue while kc's MFMA runs
-  // and kc+1's data still waits in the other set -- one chunk more
-  // global-load latency cover for 8 extra VGPRs
-  bf16x8 rx[2][2], rw[2][2];
-  auto issue = [&](int rs, int kc) {
+  bf16x8 rx[2], rw[2];
+  auto issue = [&](int kc) {
     const bf16x8* gx = reinterpret_cast<const bf16x8*>(
         X + (long)sIdx[srow] * Fp + kc + sseg * 16);
-    rx[rs][0] = gx[0];
-    rx[rs][1] = gx[1];
+    rx[0] = gx[0];
+    rx[1] = gx[1];
     const bf16x8* gw = reinterpret_cast<const bf16x8*>(
         Wt + (long)srow * Fp + kc + sseg * 16);
-    rw[rs][0] = gw[0];
-    rw[rs][1] = gw[1];
+    rw[0] = gw[0];
+    rw[1] = gw[1];
   };
-  auto commit = [&](int rs, int b) {
-    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[rs][0]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[rs][1]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[rs][0]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[rs][1]);
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
   };
 
   f32x4 acc[4][NF];
@@ -2004,14 +2001,13 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  issue(0, 0);
-  commit(0, 0);
+  issue(0);
+  commit(0);
   __syncthreads();
   const int nkc = Fp >> 6;
-  if (nkc > 1) issue(1, 64);
   for (int kc = 0; kc < nkc; ++kc) {
     int b = kc & 1;
-    if (kc + 2 < nkc) issue(kc & 1, (kc + 2) << 6);
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       bf16x8 a[4];
@@ -2030,7 +2026,7 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
               a[m], bb, acc[m][fn], 0, 0, 0);
       }
     }
-    if (kc + 1 < nkc) commit((kc + 1) & 1, b ^ 1);
+    if (kc + 1 < nkc) commit(b ^ 1);
     __syncthreads();
   }
 
