@@ -1977,43 +1977,22 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   }
   __syncthreads();
 
-  // TWO register sets (A even chunks, B odd) with compile-time indices
-  // only: loads for chunk kc+2 issue during chunk kc's MFMA while kc+1
-  // waits in the other set -- one extra chunk of global-load latency
-  // cover.  Runtime-indexed sets would spill to scratch (guide rule 20),
-  // so the chunk loop is unrolled 2x with literal sets instead.
-  bf16x8 rxA[2], rwA[2], rxB[2], rwB[2];
-  auto issueA = [&](int kc) {
+  bf16x8 rx[2], rw[2];
+  auto issue = [&](int kc) {
     const bf16x8* gx = reinterpret_cast<const bf16x8*>(
         X + (long)sIdx[srow] * Fp + kc + sseg * 16);
-    rxA[0] = gx[0];
-    rxA[1] = gx[1];
+    rx[0] = gx[0];
+    rx[1] = gx[1];
     const bf16x8* gw = reinterpret_cast<const bf16x8*>(
         Wt + (long)srow * Fp + kc + sseg * 16);
-    rwA[0] = gw[0];
-    rwA[1] = gw[1];
+    rw[0] = gw[0];
+    rw[1] = gw[1];
   };
-  auto issueB = [&](int kc) {
-    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
-        X + (long)sIdx[srow] * Fp + kc + sseg * 16);
-    rxB[0] = gx[0];
-    rxB[1] = gx[1];
-    const bf16x8* gw = reinterpret_cast<const bf16x8*>(
-        Wt + (long)srow * Fp + kc + sseg * 16);
-    rwB[0] = gw[0];
-    rwB[1] = gw[1];
-  };
-  auto commitA = [&](int b) {
-    store_row8(&sX[b][srow * 64], srow, sseg * 16, rxA[0]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rxA[1]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16, rwA[0]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rwA[1]);
-  };
-  auto commitB = [&](int b) {
-    store_row8(&sX[b][srow * 64], srow, sseg * 16, rxB[0]);
-    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rxB[1]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16, rwB[0]);
-    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rwB[1]);
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
   };
 
   f32x4 acc[4][NF];
@@ -2022,12 +2001,13 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  issueA(0);
-  commitA(0);
+  issue(0);
+  commit(0);
   __syncthreads();
   const int nkc = Fp >> 6;
-  if (nkc > 1) issueB(64);
-  auto mfma_buf = [&](int b) {
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       bf16x8 a[4];
@@ -2046,18 +2026,7 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
               a[m], bb, acc[m][fn], 0, 0, 0);
       }
     }
-  };
-  for (int kc = 0; kc < nkc; kc += 2) {
-    // even chunk kc: buffer 0; refill set A with kc+2
-    if (kc + 2 < nkc) issueA((kc + 2) << 6);
-    mfma_buf(0);
-    if (kc + 1 < nkc) commitB(1);
-    __syncthreads();
-    if (kc + 1 >= nkc) break;
-    // odd chunk kc+1: buffer 1; refill set B with kc+3
-    if (kc + 3 < nkc) issueB((kc + 3) << 6);
-    mfma_buf(1);
-    if (kc + 2 < nkc) commitA(0);
+    if (kc + 1 < nkc) commit(b ^ 1);
     __syncthreads();
   }
 
