@@ -64,6 +64,23 @@ class _GBTBase(BaseEstimator):
 
         return fit_single(self, X, y)
 
+    @property
+    def feature_importances_(self):
+        """Split-frequency importances, normalized to sum to 1 (documented
+        divergence: sklearn weighs splits by impurity decrease; the device
+        trees store (feature, threshold, value) only, so frequency is the
+        faithful statistic -- same ranking on well-separated problems)."""
+        self._check_fitted()
+        counts = np.zeros(self.n_features_in_, dtype=np.float64)
+        for tree in self._trees:
+            groups = tree if isinstance(tree, list) else [tree]
+            for feats, thrs, vals in groups:
+                for f in np.asarray(feats).ravel():
+                    if f >= 0:
+                        counts[int(f)] += 1.0
+        tot = counts.sum()
+        return counts / tot if tot > 0 else counts
+
     def _raw_predict(self, X):
         self._check_fitted()
         Xb = bin_data(np.asarray(X), self._bin_edges)

@@ -122,3 +122,14 @@ def test_param_validation(problem):
         GradientBoostingClassifier(max_depth=0).fit(X, y)
     with pytest.raises(ValueError):
         GradientBoostingClassifier(n_estimators=0).fit(X, y)
+
+
+def test_feature_importances(problem):
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((2000, 8)).astype(np.float32)
+    y = (X[:, 2] > 0).astype(np.int64)
+    clf = GradientBoostingClassifier(n_estimators=10, max_depth=2).fit(X, y)
+    fi = clf.feature_importances_
+    assert fi.shape == (8,)
+    assert abs(fi.sum() - 1.0) < 1e-9
+    assert fi.argmax() == 2
