@@ -93,6 +93,10 @@ class KMeans(ClusterMixin, BaseEstimator):
     def fit_predict(self, X, y=None):
         return self.fit(X, y).labels_
 
+    def fit_transform(self, X, y=None):
+        """Fit, then distances to the fitted centers (sklearn contract)."""
+        return self.fit(X, y).transform(X)
+
     def score(self, X, y=None):
         """Negative inertia of X under the fitted centers (sklearn)."""
         self._check_fitted()
