@@ -43,7 +43,7 @@ def bench_gbt():
     from spark_sklearn_amd.models import GradientBoostingClassifier
 
     rng = np.random.default_rng(0)
-    n, f = 1_000_000, 128
+    n, f = int(os.environ.get("GBT_ROWS", "1000000")), 128
     X = rng.standard_normal((n, f), dtype=np.float32)
     y = ((X[:, 0] * X[:, 1] + X[:, 2] + X[:, 3] ** 2 - 1 +
           0.5 * rng.standard_normal(n)) > 0).astype(np.int64)
@@ -55,7 +55,7 @@ def bench_gbt():
     t0 = time.perf_counter()
     gs.fit(X, y)
     dt = time.perf_counter() - t0
-    return {"family": "gbt", "config": "1Mx128, 50 trees, depth 3/5, cv=3",
+    return {"family": "gbt", "config": f"{n}x128, 50 trees, depth 3/5, cv=3",
             "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
             "fits_per_sec": round(gs.n_fits_ / dt, 3),
             "trees_per_sec": round(gs.n_fits_ * 50 / dt, 1),
