@@ -96,7 +96,7 @@ def bench_sweep():
     from spark_sklearn_amd.models import LogisticRegression
 
     rng = np.random.default_rng(0)
-    n, f = 1_000_000, 256
+    n, f = int(os.environ.get("KM_ROWS", "1000000")), 256
     X = rng.standard_normal((n, f), dtype=np.float32)
     w = rng.standard_normal(f, dtype=np.float32) / np.sqrt(f)
     y = ((X @ w + 0.25 * rng.standard_normal(n, dtype=np.float32)) > 0
