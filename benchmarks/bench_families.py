@@ -20,7 +20,7 @@ def bench_kmeans():
     from spark_sklearn_amd.models import KMeans
 
     rng = np.random.default_rng(0)
-    n, f, true_k = 1_000_000, 256, 16
+    n, f, true_k = int(os.environ.get("KM_ROWS", "1000000")), 256, 16
     centers = rng.standard_normal((true_k, f)).astype(np.float32) * 3
     lab = rng.integers(0, true_k, n)
     X = centers[lab] + rng.standard_normal((n, f), dtype=np.float32)
@@ -31,8 +31,9 @@ def bench_kmeans():
     t0 = time.perf_counter()
     gs.fit(X)
     dt = time.perf_counter() - t0
-    return {"family": "kmeans", "config": "1Mx256, k in 4..32 x n_init=2, "
-            "cv=5, 20 Lloyd iters", "n_fits": gs.n_fits_,
+    return {"family": "kmeans",
+            "config": f"{n}x256, k in 4..32 x n_init=2, cv=5, 20 Lloyd "
+                      "iters", "n_fits": gs.n_fits_,
             "wall_s": round(dt, 2),
             "fits_per_sec": round(gs.n_fits_ / dt, 2),
             "best": gs.best_params_}
