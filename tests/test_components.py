@@ -381,3 +381,13 @@ def test_candidate_batch_knob():
     finally:
         eng.run_batched_tasks = old
     assert seen == [64]
+
+
+def test_keyed_models_doctest():
+    """The reference carried doctest examples in keyed_models; ours run."""
+    import doctest
+
+    import spark_sklearn_amd.keyed_models as km
+
+    res = doctest.testmod(km)
+    assert res.attempted >= 1 and res.failed == 0
