@@ -128,6 +128,32 @@ __device__ __forceinline__ bf16x8 lds_frag_p(const __bf16* row, int r,
 }
 
 // ---------------------------------------------------------------------------
+// v12 staging: async global->LDS (global_load_lds_dwordx4), no register
+// round-trip.  LDS layout is LINEAR rows of 64 elements whose 16B granules
+// are XOR-swizzled by (row&7); the swizzle is applied to the per-lane
+// global SOURCE address (the LDS destination of global_load_lds is
+// hardware-fixed at wave-uniform-base + lane*16, guide m104/m173), so LDS
+// granule j of row r holds logical k-granule (j ^ (r&7)).  Fragment reads
+// un-swizzle and deliver CONTIGUOUS logical k per lane-group (map 0).
+// That differs from the HW-probed per-slot k map, but the GEMM contracts
+// A and B slot-wise: feeding both operands through the SAME logical-k
+// permutation leaves sum_k A[m][k]*B[n][k] unchanged, so no layout probe
+// is needed on this path.  Bank math: 64 lanes of one ds_read_b128 hit
+// (row, granule) pairs whose granule cycles over all 8 slots -> exactly
+// 2 lanes/bank (free, guide m136).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void gload_lds16(const void* g, void* lds) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)lds, 16, 0, 0);
+}
+// fragment read: tile base, row r, logical granule gg (= h*4 + (lane>>4))
+__device__ __forceinline__ bf16x8 frag12(const __bf16* tile, int r, int gg) {
+  return *reinterpret_cast<const bf16x8*>(tile + r * 64 +
+                                          ((gg ^ (r & 7)) << 3));
+}
+
+// ---------------------------------------------------------------------------
 // layout probe: D = A(16x32) . B(32x16), one wave, LDS-staged like the real
 // kernels.  Host test compares all 4 (AMAP, BMAP) combos vs reference.
 // ---------------------------------------------------------------------------
@@ -2085,6 +2111,259 @@ void fwd_d7_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 }
 
 // ---------------------------------------------------------------------------
+// fwd_d12_kernel: fwd_d7 with async global->LDS staging (gload_lds16 +
+// frag12 layout; see the v12 comment block up top).  Removes the register
+// round-trip + perm_pos VALU path that made fwd_d7 VALU-bound
+// (PMC round 1: MfmaUtil 10.8, VALUBusy 47.9).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void fwd_d12_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    const int* __restrict__ loss_id,
+                    const int* __restrict__ target, __bf16* __restrict__ Dt,
+                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                    const int* __restrict__ row_key,
+                    const int* __restrict__ col_key,
+                    int nt, int NTp, int Fp, int Bp, int want_loss) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[256];
+  __shared__ float sY[256];
+  __shared__ int sKey[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;        // row quad 0..3
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int rl8 = lane >> 3;          // staging: row-in-group 0..7
+  const int slot = lane & 7;          // staging: 16B granule 0..7
+
+  if (tid < 256) {
+    int gi = idx[rt * 256 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+  __syncthreads();
+
+  auto stage = [&](int b, int kc) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int rbase = q * 128 + wave * 8;
+      int r = rbase + rl8;
+      int ksw = kc + ((slot ^ (r & 7)) << 3);
+      gload_lds16(X + (long)sIdx[r] * Fp + ksw, &sX[b][rbase * 64]);
+      gload_lds16(Wt + (long)r * Fp + ksw, &sW[b][rbase * 64]);
+    }
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) stage(b ^ 1, (kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rquad * 64 + m * 16 + (lane & 15);
+        a[m] = frag12(sX[b], r, h * 4 + g);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = frag12(sW[b], c, h * 4 + g);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)");
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 256 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = 1.f / (1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 256 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+    atomicAdd(&sDb[col_in], dsum);
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bwd_g12_kernel: bwd_g8 with async global->LDS staging (same v12 layout).
+// The Xt tile is gathered by train-row index; when the 64-sample window is
+// contiguous and 16B-aligned (the common KFold case away from the single
+// fold boundary) it streams through global_load_lds, otherwise a register
+// gather writes the identical LDS layout.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void bwd_g12_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                    const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                    long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sD[2][256 * 64];
+  __shared__ __bf16 sXt[2][256 * 64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cquad = wave >> 2;       // candidate quad 0..3
+  const int fwave = wave & 3;        // feature wave 0..3
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  const int g = lane >> 4;
+  const int rl8 = lane >> 3;
+  const int slot = lane & 7;
+
+  f32x4 acc[NF][4];
+#pragma unroll
+  for (int i = 0; i < NF; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  auto stage = [&](int b, int rc) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int cbase = q * 128 + wave * 8;
+      int c = cbase + rl8;
+      gload_lds16(Dt + (long)c * NTp + rc + ((slot ^ (c & 7)) << 3),
+                  &sD[b][cbase * 64]);
+    }
+    int i0 = idx[rc];
+    if ((idx[rc + 63] - i0) == 63 && (((i0 | (int)n_rows) & 7) == 0)) {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        int fbase = q * 128 + wave * 8;
+        int f = fbase + rl8;
+        gload_lds16(Xt + (long)(ft * 256 + f) * n_rows + i0 +
+                        ((slot ^ (f & 7)) << 3),
+                    &sXt[b][fbase * 64]);
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        int gi = tid + (q << 10);
+        int f = gi >> 3, sl = gi & 7;
+        int kb = rc + ((sl ^ (f & 7)) << 3);
+        const __bf16* src = Xt + (long)(ft * 256 + f) * n_rows;
+        bf16x8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = src[idx[kb + j]];
+        *reinterpret_cast<bf16x8*>(&sXt[b][f * 64 + (sl << 3)]) = v;
+      }
+    }
+  };
+
+  stage(0, r0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+  for (int rc = r0; rc < r1; rc += 64) {
+    int b = (int)(((rc - r0) >> 6) & 1);
+    if (rc + 64 < r1) stage(b ^ 1, rc + 64);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fwave * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = frag12(sXt[b], f, h * 4 + g);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = cquad * 64 + m * 16 + (lane & 15);
+        bf16x8 a = frag12(sD[b], c, h * 4 + g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)");
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fwave * 64 + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cquad * 64 + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 256 + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
 // ---------------------------------------------------------------------------
@@ -2648,6 +2927,22 @@ void csr_scores_kernel(const int* __restrict__ indptr,
 
 static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 
+// Flagship fwd/bwd kernel version: 12 (async global->LDS staging, default)
+// or 7 (register-staged round-1 kernels) for A/B via SKA_FLAGSHIP_V=7.
+static int flagship_v() {
+  static int v = [] {
+    const char* e = getenv("SKA_FLAGSHIP_V");
+    return (e && *e) ? atoi(e) : 12;
+  }();
+  return v;
+}
+static auto* flagship_fwd() {
+  return flagship_v() >= 12 ? fwd_d12_kernel : fwd_d7_kernel;
+}
+static auto* flagship_bwd() {
+  return flagship_v() >= 12 ? bwd_g12_kernel : bwd_g8_kernel;
+}
+
 static hipStream_t cur_stream() {
   return at::cuda::getCurrentCUDAStream().stream();
 }
@@ -2803,7 +3098,7 @@ std::vector<torch::Tensor> linear_fit(
     if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
       if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
         // 1024-thread fwd (256x256 tile, 16 waves/CU)
-        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+        hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
                            dim3(1024), 0, s, Xp, y.data_ptr<float>(),
                            idx.data_ptr<int>(),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
@@ -2814,7 +3109,7 @@ std::vector<torch::Tensor> linear_fit(
                            db_sum.data_ptr<float>(), rkp, ckp, (int)nt,
                            (int)NTp, Fp, Bp, want_loss_cur);
         if (Fp % 256 == 0) {
-          hipLaunchKernelGGL(bwd_g8_kernel,
+          hipLaunchKernelGGL(flagship_bwd(),
                              dim3((unsigned)(Fp / 256), (unsigned)nseg),
                              dim3(1024), 0, s, Xtp, idx.data_ptr<int>(),
                              reinterpret_cast<const __bf16*>(Dt.data_ptr()),
@@ -3498,7 +3793,7 @@ std::vector<torch::Tensor> linear_search_run(
 
       for (int epoch = 0; epoch < max_ep; ++epoch) {
         int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
-        hipLaunchKernelGGL(fwd_d7_kernel, dim3((unsigned)(NTp / 256)),
+        hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
                            dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
                            bias.data_ptr<float>(), loss_p, tgt_p,
@@ -3507,7 +3802,7 @@ std::vector<torch::Tensor> linear_search_run(
                            db_sum.data_ptr<float>(), nullptr, nullptr,
                            (int)nt, (int)NTp, Fp, Bp, want_loss);
         if (Fp % 256 == 0) {
-          hipLaunchKernelGGL(bwd_g8_kernel,
+          hipLaunchKernelGGL(flagship_bwd(),
                              dim3((unsigned)(Fp / 256), (unsigned)nseg),
                              dim3(1024), 0, s, Xtp, idxp,
                              reinterpret_cast<const __bf16*>(Dt.data_ptr()),
