@@ -154,6 +154,33 @@ __device__ __forceinline__ bf16x8 frag12(const __bf16* tile, int r, int gg) {
 }
 
 // ---------------------------------------------------------------------------
+// v13: K=32 chunks, 4-buffer LDS ring, counted s_waitcnt vmcnt(N) with raw
+// s_barrier (guide T3+T4) -- loads for chunks kc+1/kc+2 stay in flight
+// across the barrier instead of draining to 0, which is the structural
+// stall of the 2-phase schedule.  Rows are 32 elements = 4 granules of 16B;
+// granule swizzle uses ((r>>1)&3) so a 16-row column read spreads over all
+// 32 banks at 2 lanes/bank (parity bit of r supplies the 16-word offset).
+// Iteration shape (one barrier per chunk):
+//   wait own chunk-kc loads (vmcnt counted) -> s_barrier -> stage kc+3
+//   (its buffer was last read at kc-1, before everyone's barrier) -> MFMA.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int swz32(int r) { return (r >> 1) & 3; }
+// read logical granule g (= lane>>4) of row r; tile rows are 32 elements
+__device__ __forceinline__ bf16x8 frag13(const __bf16* tile, int r, int g) {
+  return *reinterpret_cast<const bf16x8*>(tile + r * 32 +
+                                          ((g ^ swz32(r)) << 3));
+}
+#define SKA_WAIT_VMCNT(beyond)                                   \
+  do {                                                           \
+    if ((beyond) >= 2)                                           \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");           \
+    else if ((beyond) == 1)                                      \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");           \
+    else                                                         \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           \
+  } while (0)
+
+// ---------------------------------------------------------------------------
 // layout probe: D = A(16x32) . B(32x16), one wave, LDS-staged like the real
 // kernels.  Host test compares all 4 (AMAP, BMAP) combos vs reference.
 // ---------------------------------------------------------------------------
@@ -221,12 +248,20 @@ __global__ void msq_kernel(const __bf16* __restrict__ X, long n, int Fp,
 #define LOSS_HINGE 1
 #define LOSS_SQUARED 2
 
+// 1-ulp hardware reciprocal: D is stored as bf16 (8 mantissa bits) and the
+// loss sum is a convergence heuristic, so v_rcp_f32 replaces the
+// correctly-rounded v_div_scale chain (~6 VALU ops saved per element in a
+// 64x-unrolled epilogue).
+__device__ __forceinline__ float fast_rcp(float x) {
+  return __builtin_amdgcn_rcpf(x);
+}
+
 __device__ __forceinline__ void loss_grad(int lid, float z, float yv,
                                           int tgt, float* d, float* ls) {
   const float eps = 1e-12f;
   if (lid == LOSS_LOGISTIC) {
     float t = (yv == (float)tgt) ? 1.f : 0.f;
-    float p = 1.f / (1.f + __expf(-z));
+    float p = fast_rcp(1.f + __expf(-z));
     *d = p - t;
     *ls = -(t * __logf(p + eps) + (1.f - t) * __logf(1.f - p + eps));
   } else if (lid == LOSS_HINGE) {
@@ -2364,6 +2399,264 @@ void bwd_g12_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 }
 
 // ---------------------------------------------------------------------------
+// fwd_d13_kernel: fwd with the v13 counted-vmcnt 4-ring pipeline (see the
+// v13 comment block up top).  K=32 chunks, one raw s_barrier per chunk,
+// chunk kc+1/kc+2 loads stay in flight across it.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void fwd_d13_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    const int* __restrict__ loss_id,
+                    const int* __restrict__ target, __bf16* __restrict__ Dt,
+                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                    const int* __restrict__ row_key,
+                    const int* __restrict__ col_key,
+                    int nt, int NTp, int Fp, int Bp, int want_loss) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[4][256 * 32];
+  __shared__ __bf16 sW[4][256 * 32];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[256];
+  __shared__ float sY[256];
+  __shared__ int sKey[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;        // row quad 0..3
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int srow = wave * 16 + (lane >> 2);   // staging row 0..255
+  const int slot = lane & 3;                  // staging 16B granule 0..3
+
+  if (tid < 256) {
+    int gi = idx[rt * 256 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+  __syncthreads();
+
+  auto stage = [&](int buf, int kc) {
+    int ksw = (kc << 5) + ((slot ^ swz32(srow)) << 3);
+    gload_lds16(X + (long)sIdx[srow] * Fp + ksw, &sX[buf][wave * 16 * 32]);
+    gload_lds16(Wt + (long)srow * Fp + ksw, &sW[buf][wave * 16 * 32]);
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nkc = Fp >> 5;
+  stage(0, 0);
+  if (nkc > 1) stage(1, 1);
+  if (nkc > 2) stage(2, 2);
+  for (int kc = 0; kc < nkc; ++kc) {
+    SKA_WAIT_VMCNT(min(nkc - 1 - kc, 2));
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (kc + 3 < nkc) stage((kc + 3) & 3, kc + 3);
+    const __bf16* bx = sX[kc & 3];
+    const __bf16* bw = sW[kc & 3];
+    bf16x8 a[4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int r = rquad * 64 + m * 16 + (lane & 15);
+      a[m] = frag13(bx, r, g);
+    }
+#pragma unroll
+    for (int fn = 0; fn < NF; ++fn) {
+      int c = cwave * 64 + fn * 16 + (lane & 15);
+      bf16x8 bb = frag13(bw, c, g);
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[m], bb, acc[m][fn], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 256 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = fast_rcp(1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 256 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+    atomicAdd(&sDb[col_in], dsum);
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bwd_g13_kernel: bwd with the v13 counted-vmcnt 4-ring pipeline.  The
+// pipelined path requires the segment's train indices to be one contiguous
+// aligned run (the common KFold case); other segments (one per fold
+// boundary) take a single-buffer gather loop.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void bwd_g13_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                    const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                    long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sD[4][256 * 32];
+  __shared__ __bf16 sXt[4][256 * 32];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cquad = wave >> 2;       // candidate quad 0..3
+  const int fwave = wave & 3;        // feature wave 0..3
+  const int ft = blockIdx.x;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  const int g = lane >> 4;
+  const int srow = wave * 16 + (lane >> 2);
+  const int slot = lane & 3;
+
+  f32x4 acc[NF][4];
+#pragma unroll
+  for (int i = 0; i < NF; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ibase = idx[r0];
+  bool seg_cont = (idx[r1 - 1] - ibase == r1 - 1 - r0) &&
+                  (((ibase | (int)n_rows) & 7) == 0);
+  if (seg_cont) {
+    auto stage = [&](int buf, int rc) {
+      int gsw = ((slot ^ swz32(srow)) << 3);
+      gload_lds16(Dt + (long)srow * NTp + rc + gsw,
+                  &sD[buf][wave * 16 * 32]);
+      gload_lds16(Xt + (long)(ft * 256 + srow) * n_rows + ibase + (rc - r0) +
+                      gsw,
+                  &sXt[buf][wave * 16 * 32]);
+    };
+    const int nkc = (r1 - r0) >> 5;
+    stage(0, r0);
+    if (nkc > 1) stage(1, r0 + 32);
+    if (nkc > 2) stage(2, r0 + 64);
+    for (int i = 0; i < nkc; ++i) {
+      SKA_WAIT_VMCNT(min(nkc - 1 - i, 2));
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+      if (i + 3 < nkc) stage((i + 3) & 3, r0 + ((i + 3) << 5));
+      const __bf16* bd = sD[i & 3];
+      const __bf16* bx = sXt[i & 3];
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fwave * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = frag13(bx, f, g);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = cquad * 64 + m * 16 + (lane & 15);
+        bf16x8 a = frag13(bd, c, g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+  } else {
+    // gather fallback: 1024 threads cover 256 rows x 4 granules for both
+    // tiles; full-drain single buffer (fold-boundary segments only).
+    const int fr = tid >> 2, sl = tid & 3;
+    for (int rc = r0; rc < r1; rc += 32) {
+      int kb = rc + ((sl ^ swz32(fr)) << 3);
+      const __bf16* srcx = Xt + (long)(ft * 256 + fr) * n_rows;
+      bf16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = srcx[idx[kb + j]];
+      *reinterpret_cast<bf16x8*>(&sXt[0][fr * 32 + (sl << 3)]) = v;
+      bf16x8 vd = *reinterpret_cast<const bf16x8*>(Dt + (long)fr * NTp + kb);
+      *reinterpret_cast<bf16x8*>(&sD[0][fr * 32 + (sl << 3)]) = vd;
+      __syncthreads();
+      const __bf16* bd = sD[0];
+      const __bf16* bx = sXt[0];
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fwave * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = frag13(bx, f, g);
+      }
+#pragma unroll
+      for (int m = 0; m < NF; ++m) {
+        int c = cquad * 64 + m * 16 + (lane & 15);
+        bf16x8 a = frag13(bd, c, g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int m = 0; m < NF; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = fwave * 64 + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cquad * 64 + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + ft * 256 + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
 // ---------------------------------------------------------------------------
@@ -2927,20 +3220,23 @@ void csr_scores_kernel(const int* __restrict__ indptr,
 
 static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 
-// Flagship fwd/bwd kernel version: 12 (async global->LDS staging, default)
-// or 7 (register-staged round-1 kernels) for A/B via SKA_FLAGSHIP_V=7.
+// Flagship fwd/bwd kernel version: 13 (counted-vmcnt 4-ring pipeline,
+// default), 12 (async-staged 2-phase) or 7 (register-staged round-1
+// kernels); A/B via SKA_FLAGSHIP_V.
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
-    return (e && *e) ? atoi(e) : 12;
+    return (e && *e) ? atoi(e) : 13;
   }();
   return v;
 }
 static auto* flagship_fwd() {
-  return flagship_v() >= 12 ? fwd_d12_kernel : fwd_d7_kernel;
+  int v = flagship_v();
+  return v >= 13 ? fwd_d13_kernel : (v == 12 ? fwd_d12_kernel : fwd_d7_kernel);
 }
 static auto* flagship_bwd() {
-  return flagship_v() >= 12 ? bwd_g12_kernel : bwd_g8_kernel;
+  int v = flagship_v();
+  return v >= 13 ? bwd_g13_kernel : (v == 12 ? bwd_g12_kernel : bwd_g8_kernel);
 }
 
 static hipStream_t cur_stream() {
