@@ -2657,6 +2657,240 @@ void bwd_g13_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 }
 
 // ---------------------------------------------------------------------------
+// fused_fb14_kernel: ONE kernel per epoch computing the fwd GEMM, the loss
+// epilogue AND the gradient GEMM, with the D (dual/residual) tile resident
+// in LDS between the two -- the bf16 D matrix (Bp x NTp, ~0.4 GB at the
+// flagship shape) is never written to HBM.  PMC round 2 measured the
+// separate fwd/bwd pair moving 1.64 GB/epoch of which 0.82 GB was the D
+// write + re-read; fusing nearly halves the HBM traffic and removes the
+// scatter write that stalled fwd.
+//   Per block: loop over S subtiles of 64 rows.  Per subtile:
+//     fwd   : acc_f[4] = X[64r x Fp] . Wt^T       (K=64 chunks via LDS)
+//     epilog: d = dloss(z), transposed store into sD[256c][64r] (frag12
+//             swizzle), loss/db partial sums into LDS
+//     bwd   : acc_g[4][NFW] += D^T . Xt           (two r-chunks of 32)
+//   Block end: one atomicAdd flush of G (and loss/db) -- G atomic traffic
+//   is 256 KB per block, amortized over S subtiles.
+// The 64-row subtile keeps live registers at acc_g(64) + acc_f(16) +
+// fragments, inside the 128-VGPR budget __launch_bounds__(1024) implies
+// (the 128-row variant spilled to scratch in the hot loop).
+// NFW = Fp/64 (1, 2 or 4); requires Bp==256, NTp%256==0, Fp in {64,128,256}.
+// ---------------------------------------------------------------------------
+template <int NFW>
+__global__ __launch_bounds__(1024)
+void fused_fb14_kernel(const __bf16* __restrict__ X,
+                       const __bf16* __restrict__ Xt,
+                       const float* __restrict__ y,
+                       const int* __restrict__ idx,
+                       const __bf16* __restrict__ Wt,
+                       const float* __restrict__ bias,
+                       const int* __restrict__ loss_id,
+                       const int* __restrict__ target,
+                       float* __restrict__ Gt,
+                       float* __restrict__ loss_sum,
+                       float* __restrict__ db_sum,
+                       const int* __restrict__ row_key,
+                       const int* __restrict__ col_key,
+                       int nt, int NTp, int Fp, int Bp, int want_loss,
+                       int S, long n_rows) {
+  __shared__ __bf16 sD[256 * 64];         // D^T tile [cand][row], 32 KB
+  __shared__ __bf16 sA[64 * 64];          // X chunk [64r][64k], 8 KB
+  __shared__ __bf16 sW[256 * 64];         // W chunk [256c][64k] / Xt chunk
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[64];
+  __shared__ float sY[64];
+  __shared__ int sKey[64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;         // fwd: candidate wave (64 cols)
+  const int rquad = wave >> 2;        // fwd: row quad (16 rows)
+  const int cquad = wave >> 2;        // bwd: candidate quad (64 cands)
+  const int fwave = wave & 3;         // bwd: feature wave (16*NFW feats)
+  const int g = lane >> 4;
+
+  if (tid < 256) {
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+
+  f32x4 acc_g[4][NFW];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NFW; ++j) acc_g[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const long row0 = (long)blockIdx.x * S * 64;
+  const int nsub = (int)min((long)S, (NTp - row0 + 63) / 64);
+
+  for (int s = 0; s < nsub; ++s) {
+    const int rbase = (int)row0 + s * 64;
+    __syncthreads();     // sIdx/sA/sW reuse vs previous subtile
+    if (tid < 64) {
+      int gi = idx[rbase + tid];
+      sIdx[tid] = gi;
+      sY[tid] = y[gi];
+      sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    }
+    __syncthreads();
+
+    // ---- fwd GEMM: acc_f = X . W^T -------------------------------------
+    f32x4 acc_f[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc_f[j] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int kc = 0; kc < Fp; kc += 64) {
+      if (tid < 512) {  // stage X chunk [64r][64k]: waves 0-7
+        int r = tid >> 3, sl = tid & 7;
+        gload_lds16(X + (long)sIdx[r] * Fp + kc + ((sl ^ (r & 7)) << 3),
+                    &sA[wave * 8 * 64]);
+      }
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {  // stage W chunk [256c][64k]: 2/thread
+        int l = tid + (q << 10);
+        int c = l >> 3, sl = l & 7;
+        gload_lds16(Wt + (long)c * Fp + kc + ((sl ^ (c & 7)) << 3),
+                    &sW[(q * 128 + wave * 8) * 64]);
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        int r = rquad * 16 + (lane & 15);
+        bf16x8 a = frag12(sA, r, h * 4 + g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int c = cwave * 64 + fn * 16 + (lane & 15);
+          bf16x8 bb = frag12(sW, c, h * 4 + g);
+          acc_f[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bb, acc_f[fn], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- epilogue: d values -> sD (transposed, swizzled) + loss/db -----
+#pragma unroll 1
+    for (int fn = 0; fn < 4; ++fn) {
+      int col_in = cwave * 64 + fn * 16 + (lane & 15);
+      float bia = bias[col_in];
+      int lid = loss_id[col_in];
+      int tgt = target[col_in];
+      int ck = (col_key != nullptr) ? col_key[col_in] : 0;
+      float lsum = 0.f, dsum = 0.f;
+      int row_base = rquad * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rbase + row_base + r;
+        float z = acc_f[fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = fast_rcp(1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      {
+        int j = row_base >> 3, half = (row_base >> 2) & 1;
+        *reinterpret_cast<bf16x4*>(
+            &sD[col_in * 64 + ((j ^ (col_in & 7)) << 3) + (half << 2)]) =
+            dpack;
+      }
+      if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+      atomicAdd(&sDb[col_in], dsum);
+    }
+    __syncthreads();
+
+    // ---- bwd GEMM: acc_g += D^T . Xt (two r-chunks of 32) --------------
+    for (int rc = 0; rc < 64; rc += 32) {
+      {  // stage Xt chunk [Fp f][32 r] into sW (16 KB at NFW=4)
+        int i0 = sIdx[rc];
+        bool cont = (sIdx[rc + 31] - i0 == 31) &&
+                    (((i0 | (int)n_rows) & 7) == 0);
+        if (cont) {
+          int fr = wave * 16 + (lane >> 2), sl = lane & 3;
+          if (fr < NFW * 64)
+            gload_lds16(Xt + (long)fr * n_rows + i0 + ((sl ^ swz32(fr)) << 3),
+                        &sW[(wave * 16) * 32]);
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        } else {
+          int fr = tid >> 2, sl = tid & 3;
+          if (fr < NFW * 64) {
+            int kb = rc + ((sl ^ swz32(fr)) << 3);
+            const __bf16* src = Xt + (long)fr * n_rows;
+            bf16x8 v;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v[j] = src[idx[rbase + kb + j]];
+            *reinterpret_cast<bf16x8*>(&sW[fr * 32 + (sl << 3)]) = v;
+          }
+        }
+        __syncthreads();
+        constexpr int FH = NFW > 2 ? 2 : NFW;   // bfr regs held at once
+#pragma unroll
+        for (int fh = 0; fh < NFW / FH; ++fh) {
+          bf16x8 bfr[FH];
+#pragma unroll
+          for (int fj = 0; fj < FH; ++fj) {
+            int f = fwave * (16 * NFW) + (fh * FH + fj) * 16 + (lane & 15);
+            bfr[fj] = frag13(sW, f, g);
+          }
+#pragma unroll
+          for (int mm = 0; mm < 4; ++mm) {
+            int c = cquad * 64 + mm * 16 + (lane & 15);
+            bf16x8 a = frag12(sD, c, (rc >> 3) + g);
+#pragma unroll
+            for (int fj = 0; fj < FH; ++fj)
+              acc_g[mm][fh * FH + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bfr[fj], acc_g[mm][fh * FH + fj], 0, 0, 0);
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+  // ---- block end: flush G, loss, db ------------------------------------
+#pragma unroll
+  for (int mm = 0; mm < 4; ++mm) {
+#pragma unroll
+    for (int fn = 0; fn < NFW; ++fn) {
+      int f_in = fwave * (16 * NFW) + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cquad * 64 + mm * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + f_in], acc_g[mm][fn][r]);
+      }
+    }
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss && sLoss[tid] != 0.f) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
 // ---------------------------------------------------------------------------
@@ -3226,7 +3460,7 @@ static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
-    return (e && *e) ? atoi(e) : 13;
+    return (e && *e) ? atoi(e) : 14;
   }();
   return v;
 }
@@ -3317,7 +3551,9 @@ std::vector<torch::Tensor> linear_fit(
   auto V = torch::zeros({Bp, Fp}, optF);
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::zeros({Bp, Fp}, X.options());
-  auto Dt = torch::zeros({(long)Bp, NTp}, X.options());
+  bool use_fused14 = flagship_v() >= 14 && Bp == 256 && NTp % 256 == 0 &&
+                     (Fp == 64 || Fp == 128 || Fp == 256);
+  auto Dt = torch::zeros({(long)Bp, use_fused14 ? 1L : NTp}, X.options());
   auto bias = torch::zeros({Bp}, optF);
   auto Vb = torch::zeros({Bp}, optF);
   auto loss_sum = torch::zeros({Bp}, optF);
@@ -3392,6 +3628,51 @@ std::vector<torch::Tensor> linear_fit(
   for (int epoch = 0; epoch < (int)max_epochs; ++epoch) {
     want_loss_cur = (any_tol || epoch == (int)max_epochs - 1) ? 1 : 0;
     if (g_layout == 3) {   // HW-probed fast path: permuted/swizzled LDS
+      bool fused = use_fused14 && nf == 4;
+      if (fused) {
+        long subt = NTp / 64;
+        long Ssub = (subt + 255) / 256;
+        unsigned nblk = (unsigned)((subt + Ssub - 1) / Ssub);
+        auto* fk = Fp == 256   ? fused_fb14_kernel<4>
+                   : Fp == 128 ? fused_fb14_kernel<2>
+                               : fused_fb14_kernel<1>;
+        hipLaunchKernelGGL(fk, dim3(nblk), dim3(1024), 0, s, Xp, Xtp,
+                           y.data_ptr<float>(), idx.data_ptr<int>(),
+                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                           bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+                           target.data_ptr<int>(), Gt.data_ptr<float>(),
+                           loss_sum.data_ptr<float>(),
+                           db_sum.data_ptr<float>(), rkp, ckp, (int)nt,
+                           (int)NTp, Fp, Bp, want_loss_cur, (int)Ssub,
+                           n_rows);
+        hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
+                           loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
+                           bias.data_ptr<float>(), Vb.data_ptr<float>(),
+                           prev_loss.data_ptr<float>(), cur_loss.data_ptr<float>(),
+                           active.data_ptr<int>(), epoch_mask.data_ptr<int>(),
+                           n_iter.data_ptr<int>(), max_iter.data_ptr<int>(),
+                           lr.data_ptr<float>(), mom.data_ptr<float>(),
+                           tol.data_ptr<float>(),
+                           fit_ic.data_ptr<unsigned char>(), inv_n, epoch, Bp,
+                           want_loss_cur, n_active.data_ptr<int>());
+        hipLaunchKernelGGL(update_w_kernel, upd_grid2, dim3(upd_threads), 0, s,
+                           W.data_ptr<float>(), V.data_ptr<float>(),
+                           Gt.data_ptr<float>(),
+                           reinterpret_cast<__bf16*>(Wt.data_ptr()),
+                           epoch_mask.data_ptr<int>(), lr.data_ptr<float>(),
+                           mom.data_ptr<float>(), l1.data_ptr<float>(),
+                           l2.data_ptr<float>(), inv_n, Fp);
+        HIP_CHECK(hipGetLastError());
+        if (check_every > 0 && (epoch + 1) % check_every == 0 &&
+            epoch + 1 < max_epochs) {
+          int na = n_active.item<int>();
+          if (na == 0) break;
+          n_active.zero_();
+        } else {
+          HIP_CHECK(hipMemsetAsync(n_active.data_ptr(), 0, sizeof(int), s));
+        }
+        continue;
+      }
       if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
         // 1024-thread fwd (256x256 tile, 16 waves/CU)
         hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
@@ -4006,7 +4287,9 @@ std::vector<torch::Tensor> linear_search_run(
   auto V = torch::empty({Bp, Fp}, optF);
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::empty({Bp, Fp}, X.options());
-  auto Dt = torch::empty({(long)Bp, max_NTp}, X.options());
+  // fused v14 epoch kernel keeps D in LDS -- no global D buffer needed
+  bool use_fused = flagship_v() >= 14 && (Fp == 64 || Fp == 128 || Fp == 256);
+  auto Dt = torch::empty({(long)Bp, use_fused ? 1L : max_NTp}, X.options());
   auto bias = torch::empty({Bp}, optF);
   auto Vb = torch::empty({Bp}, optF);
   auto loss_sum = torch::zeros({Bp}, optF);
@@ -4087,8 +4370,26 @@ std::vector<torch::Tensor> linear_search_run(
           fit_ic.data_ptr<unsigned char>() + off;
       int* ni_p = n_iter.data_ptr<int>() + (long)f * Bt + off;
 
+      bool fused = use_fused && NTp % 256 == 0;
+      long subt = NTp / 64;
+      long Ssub = (subt + 255) / 256;
+      unsigned fnblk = (unsigned)((subt + Ssub - 1) / Ssub);
+      auto* fk = Fp == 256   ? fused_fb14_kernel<4>
+                 : Fp == 128 ? fused_fb14_kernel<2>
+                             : fused_fb14_kernel<1>;
       for (int epoch = 0; epoch < max_ep; ++epoch) {
         int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
+        if (fused) {
+          hipLaunchKernelGGL(fk, dim3(fnblk), dim3(1024), 0, s, Xp, Xtp,
+                             y.data_ptr<float>(), idxp,
+                             reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                             bias.data_ptr<float>(), loss_p, tgt_p,
+                             Gt.data_ptr<float>(),
+                             loss_sum.data_ptr<float>(),
+                             db_sum.data_ptr<float>(), nullptr, nullptr,
+                             (int)nt, (int)NTp, Fp, Bp, want_loss,
+                             (int)Ssub, n_rows);
+        } else {
         hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
                            dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
@@ -4111,6 +4412,7 @@ std::vector<torch::Tensor> linear_search_run(
                              reinterpret_cast<const __bf16*>(Dt.data_ptr()),
                              Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
                              Bp, seg_rows);
+        }
         }
         hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
                            loss_sum.data_ptr<float>(),
