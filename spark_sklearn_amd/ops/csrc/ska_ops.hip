@@ -2891,6 +2891,249 @@ void fused_fb14_kernel(const __bf16* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
+// v15: block-level latency hiding.  Every earlier flagship variant (2-phase
+// register staging, async staging, counted-vmcnt ring, fused) ran at ONE
+// resident 16-wave block per CU (LDS 96-136 KB), so each barrier-
+// synchronized stage burst was latency-exposed with nothing to overlap it
+// -- measured ~4-7 us per phase across all of them.  v15 shrinks to
+// 512-thread blocks with 128x128 tiles and ~34 KB LDS / ~100 VGPR so TWO+
+// independent blocks are resident per CU: one block's MFMA phase overlaps
+// the other's staging drain, no intra-block pipelining needed.
+// fwd grid: (NTp/128, Bp/128); bwd grid: ((Fp/128)*(Bp/128), nseg).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512, 4)
+void fwd_d15_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    const int* __restrict__ loss_id,
+                    const int* __restrict__ target, __bf16* __restrict__ Dt,
+                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                    const int* __restrict__ row_key,
+                    const int* __restrict__ col_key,
+                    int nt, int NTp, int Fp, int Bp, int want_loss) {
+  __shared__ __bf16 sX[128 * 64];
+  __shared__ __bf16 sW[128 * 64];
+  __shared__ float sLoss[128];
+  __shared__ float sDb[128];
+  __shared__ int sIdx[128];
+  __shared__ float sY[128];
+  __shared__ int sKey[128];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;          // 0..7
+  const int rquad = wave >> 1;        // 32-row quad 0..3
+  const int cwave = wave & 1;         // 64-col wave 0..1
+  const int rt = blockIdx.x;
+  const int col0 = blockIdx.y * 128;
+  const int g = lane >> 4;
+
+  if (tid < 128) {
+    int gi = idx[rt * 128 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+  __syncthreads();
+
+  auto stage = [&](int kc) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int l = tid + (q << 9);
+      int r = l >> 3, sl = l & 7;
+      gload_lds16(X + (long)sIdx[r] * Fp + kc + ((sl ^ (r & 7)) << 3),
+                  &sX[((q << 6) + wave * 8) * 64]);
+      gload_lds16(Wt + (long)(col0 + r) * Fp + kc + ((sl ^ (r & 7)) << 3),
+                  &sW[((q << 6) + wave * 8) * 64]);
+    }
+  };
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int kc = 0; kc < Fp; kc += 64) {
+    stage(kc);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        int r = rquad * 32 + m * 16 + (lane & 15);
+        a[m] = frag12(sX, r, h * 4 + g);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = frag12(sW, c, h * 4 + g);
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll 1
+  for (int fn = 0; fn < 4; ++fn) {
+    int c_loc = cwave * 64 + fn * 16 + (lane & 15);
+    int col_in = col0 + c_loc;
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      int row_base = rquad * 32 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 128 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = fast_rcp(1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      *reinterpret_cast<bf16x4*>(
+          Dt + (long)col_in * NTp + rt * 128 + row_base) = dpack;
+    }
+    if (want_loss) atomicAdd(&sLoss[c_loc], lsum);
+    atomicAdd(&sDb[c_loc], dsum);
+  }
+  __syncthreads();
+  if (tid < 128) {
+    if (want_loss) atomicAdd(&loss_sum[col0 + tid], sLoss[tid]);
+    atomicAdd(&db_sum[col0 + tid], sDb[tid]);
+  }
+}
+
+__global__ __launch_bounds__(512, 4)
+void bwd_g15_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
+                    const __bf16* __restrict__ Dt, float* __restrict__ Gt,
+                    long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+  __shared__ __bf16 sD[128 * 64];
+  __shared__ __bf16 sXt[128 * 64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cquad = wave >> 1;       // 32-cand quad 0..3
+  const int fwave = wave & 1;        // 64-feature wave 0..1
+  const int nft = Fp >> 7;
+  const int f0 = (blockIdx.x % nft) * 128;
+  const int c0 = (blockIdx.x / nft) * 128;
+  const int r0 = blockIdx.y * seg_rows;
+  const int r1 = min(r0 + seg_rows, NTp);
+  const int g = lane >> 4;
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  auto stage = [&](int rc) {
+    int i0 = idx[rc];
+    bool cont = (idx[rc + 63] - i0 == 63) && (((i0 | (int)n_rows) & 7) == 0);
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int l = tid + (q << 9);
+      int r = l >> 3, sl = l & 7;
+      gload_lds16(Dt + (long)(c0 + r) * NTp + rc + ((sl ^ (r & 7)) << 3),
+                  &sD[((q << 6) + wave * 8) * 64]);
+    }
+    if (cont) {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        int l = tid + (q << 9);
+        int r = l >> 3, sl = l & 7;
+        gload_lds16(Xt + (long)(f0 + r) * n_rows + i0 + ((sl ^ (r & 7)) << 3),
+                    &sXt[((q << 6) + wave * 8) * 64]);
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        int l = tid + (q << 9);
+        int f = l >> 3, sl = l & 7;
+        int kb = rc + ((sl ^ (f & 7)) << 3);
+        const __bf16* src = Xt + (long)(f0 + f) * n_rows;
+        bf16x8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = src[idx[kb + j]];
+        *reinterpret_cast<bf16x8*>(&sXt[f * 64 + (sl << 3)]) = v;
+      }
+    }
+  };
+
+  for (int rc = r0; rc < r1; rc += 64) {
+    stage(rc);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int f = fwave * 64 + fn * 16 + (lane & 15);
+        bfr[fn] = frag12(sXt, f, h * 4 + g);
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        int c = cquad * 32 + m * 16 + (lane & 15);
+        bf16x8 a = frag12(sD, c, h * 4 + g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bfr[fn], acc[m][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int f_in = f0 + fwave * 64 + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = c0 + cquad * 32 + m * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + f_in], acc[m][fn][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
 // ---------------------------------------------------------------------------
@@ -3460,7 +3703,7 @@ static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
-    return (e && *e) ? atoi(e) : 14;
+    return (e && *e) ? atoi(e) : 15;
   }();
   return v;
 }
@@ -3472,6 +3715,44 @@ static auto* flagship_bwd() {
   int v = flagship_v();
   return v >= 13 ? bwd_g13_kernel : (v == 12 ? bwd_g12_kernel : bwd_g8_kernel);
 }
+
+// Launch the flagship fwd+bwd pair for the current SKA_FLAGSHIP_V.
+static void launch_flagship_pair(
+    hipStream_t s, const __bf16* Xp, const __bf16* Xtp, const float* yp,
+    const int* idxp, const __bf16* Wtp, float* biasp, const int* lossp,
+    const int* tgtp, __bf16* Dtp, float* losssum, float* dbsum,
+    const int* rkp, const int* ckp, long nt, long NTp, int Fp, int Bp,
+    int want_loss, float* Gtp, long n_rows, long nseg, int seg_rows) {
+  if (flagship_v() >= 15 && Bp % 128 == 0 && NTp % 128 == 0) {
+    hipLaunchKernelGGL(fwd_d15_kernel,
+                       dim3((unsigned)(NTp / 128), (unsigned)(Bp / 128)),
+                       dim3(512), 0, s, Xp, yp, idxp, Wtp, biasp, lossp,
+                       tgtp, Dtp, losssum, dbsum, rkp, ckp, (int)nt,
+                       (int)NTp, Fp, Bp, want_loss);
+    hipLaunchKernelGGL(bwd_g15_kernel,
+                       dim3((unsigned)((Fp / 128) * (Bp / 128)),
+                            (unsigned)nseg),
+                       dim3(512), 0, s, Xtp, idxp, Dtp, Gtp, n_rows,
+                       (int)NTp, Fp, Bp, seg_rows);
+    return;
+  }
+  hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)), dim3(1024),
+                     0, s, Xp, yp, idxp, Wtp, biasp, lossp, tgtp, Dtp,
+                     losssum, dbsum, rkp, ckp, (int)nt, (int)NTp, Fp, Bp,
+                     want_loss);
+  if (Fp % 256 == 0) {
+    hipLaunchKernelGGL(flagship_bwd(),
+                       dim3((unsigned)(Fp / 256), (unsigned)nseg), dim3(1024),
+                       0, s, Xtp, idxp, Dtp, Gtp, n_rows, (int)NTp, Fp, Bp,
+                       seg_rows);
+  } else {
+    hipLaunchKernelGGL(bwd_g5_kernel,
+                       dim3((unsigned)(Fp / 128), (unsigned)nseg), dim3(512),
+                       0, s, Xtp, idxp, Dtp, Gtp, n_rows, (int)NTp, Fp, Bp,
+                       seg_rows);
+  }
+}
+
 
 static hipStream_t cur_stream() {
   return at::cuda::getCurrentCUDAStream().stream();
@@ -3551,7 +3832,7 @@ std::vector<torch::Tensor> linear_fit(
   auto V = torch::zeros({Bp, Fp}, optF);
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::zeros({Bp, Fp}, X.options());
-  bool use_fused14 = flagship_v() >= 14 && Bp == 256 && NTp % 256 == 0 &&
+  bool use_fused14 = flagship_v() == 14 && Bp == 256 && NTp % 256 == 0 &&
                      (Fp == 64 || Fp == 128 || Fp == 256);
   auto Dt = torch::zeros({(long)Bp, use_fused14 ? 1L : NTp}, X.options());
   auto bias = torch::zeros({Bp}, optF);
@@ -3674,32 +3955,14 @@ std::vector<torch::Tensor> linear_fit(
         continue;
       }
       if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
-        // 1024-thread fwd (256x256 tile, 16 waves/CU)
-        hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
-                           dim3(1024), 0, s, Xp, y.data_ptr<float>(),
-                           idx.data_ptr<int>(),
-                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
-                           bias.data_ptr<float>(), loss_id.data_ptr<int>(),
-                           target.data_ptr<int>(),
-                           reinterpret_cast<__bf16*>(Dt.data_ptr()),
-                           loss_sum.data_ptr<float>(),
-                           db_sum.data_ptr<float>(), rkp, ckp, (int)nt,
-                           (int)NTp, Fp, Bp, want_loss_cur);
-        if (Fp % 256 == 0) {
-          hipLaunchKernelGGL(flagship_bwd(),
-                             dim3((unsigned)(Fp / 256), (unsigned)nseg),
-                             dim3(1024), 0, s, Xtp, idx.data_ptr<int>(),
-                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
-                             Bp, seg_rows);
-        } else {
-          hipLaunchKernelGGL(bwd_g5_kernel,
-                             dim3((unsigned)(Fp / 128), (unsigned)nseg),
-                             dim3(512), 0, s, Xtp, idx.data_ptr<int>(),
-                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
-                             Bp, seg_rows);
-        }
+        launch_flagship_pair(
+            s, Xp, Xtp, y.data_ptr<float>(), idx.data_ptr<int>(),
+            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+            bias.data_ptr<float>(), loss_id.data_ptr<int>(),
+            target.data_ptr<int>(), reinterpret_cast<__bf16*>(Dt.data_ptr()),
+            loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(), rkp, ckp,
+            nt, NTp, Fp, Bp, want_loss_cur, Gt.data_ptr<float>(), n_rows,
+            nseg, seg_rows);
         hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
                            loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
                            bias.data_ptr<float>(), Vb.data_ptr<float>(),
@@ -4288,7 +4551,7 @@ std::vector<torch::Tensor> linear_search_run(
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::empty({Bp, Fp}, X.options());
   // fused v14 epoch kernel keeps D in LDS -- no global D buffer needed
-  bool use_fused = flagship_v() >= 14 && (Fp == 64 || Fp == 128 || Fp == 256);
+  bool use_fused = flagship_v() == 14 && (Fp == 64 || Fp == 128 || Fp == 256);
   auto Dt = torch::empty({(long)Bp, use_fused ? 1L : max_NTp}, X.options());
   auto bias = torch::empty({Bp}, optF);
   auto Vb = torch::empty({Bp}, optF);
@@ -4390,29 +4653,14 @@ std::vector<torch::Tensor> linear_search_run(
                              (int)nt, (int)NTp, Fp, Bp, want_loss,
                              (int)Ssub, n_rows);
         } else {
-        hipLaunchKernelGGL(flagship_fwd(), dim3((unsigned)(NTp / 256)),
-                           dim3(1024), 0, s, Xp, y.data_ptr<float>(), idxp,
-                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
-                           bias.data_ptr<float>(), loss_p, tgt_p,
-                           reinterpret_cast<__bf16*>(Dt.data_ptr()),
-                           loss_sum.data_ptr<float>(),
-                           db_sum.data_ptr<float>(), nullptr, nullptr,
-                           (int)nt, (int)NTp, Fp, Bp, want_loss);
-        if (Fp % 256 == 0) {
-          hipLaunchKernelGGL(flagship_bwd(),
-                             dim3((unsigned)(Fp / 256), (unsigned)nseg),
-                             dim3(1024), 0, s, Xtp, idxp,
-                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
-                             Bp, seg_rows);
-        } else {
-          hipLaunchKernelGGL(bwd_g5_kernel,
-                             dim3((unsigned)(Fp / 128), (unsigned)nseg),
-                             dim3(512), 0, s, Xtp, idxp,
-                             reinterpret_cast<const __bf16*>(Dt.data_ptr()),
-                             Gt.data_ptr<float>(), n_rows, (int)NTp, Fp,
-                             Bp, seg_rows);
-        }
+          launch_flagship_pair(
+              s, Xp, Xtp, y.data_ptr<float>(), idxp,
+              reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+              bias.data_ptr<float>(), loss_p, tgt_p,
+              reinterpret_cast<__bf16*>(Dt.data_ptr()),
+              loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(), nullptr,
+              nullptr, nt, NTp, Fp, Bp, want_loss, Gt.data_ptr<float>(),
+              n_rows, nseg, seg_rows);
         }
         hipLaunchKernelGGL(col_finalize_kernel, fin_grid, dim3(256), 0, s,
                            loss_sum.data_ptr<float>(),
