@@ -5010,7 +5010,210 @@ torch::Tensor gbt_metric_multi(torch::Tensor raws, torch::Tensor y,
   return out;
 }
 
+// ---------------------------------------------------------------------------
+// flagship_ablate: time the fwd kernel with pieces removed, to locate the
+// wall empirically (guide common-mistake #8: ablate before optimizing).
+// bits: 1 = skip Dt store, 2 = trivial epilogue (d = z), 4 = skip X stage,
+// 8 = skip W stage, 16 = skip MFMA loop.
+// ---------------------------------------------------------------------------
+template <int ABL>
+__global__ __launch_bounds__(1024)
+void fwd_dAB_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
+                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    const int* __restrict__ loss_id,
+                    const int* __restrict__ target, __bf16* __restrict__ Dt,
+                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
+                    int nt, int NTp, int Fp, int Bp, int want_loss) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[256];
+  __shared__ float sY[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int rl8 = lane >> 3;
+  const int slot = lane & 7;
+
+  if (tid < 256) {
+    int gi = idx[rt * 256 + tid];
+    sIdx[tid] = gi;
+    sY[tid] = y[gi];
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+  __syncthreads();
+
+  auto stage = [&](int b, int kc) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int rbase = q * 128 + wave * 8;
+      int r = rbase + rl8;
+      int ksw = kc + ((slot ^ (r & 7)) << 3);
+      if (!(ABL & 4))
+        gload_lds16(X + (long)sIdx[r] * Fp + ksw, &sX[b][rbase * 64]);
+      if (!(ABL & 8))
+        gload_lds16(Wt + (long)r * Fp + ksw, &sW[b][rbase * 64]);
+    }
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) stage(b ^ 1, (kc + 1) << 6);
+    if (!(ABL & 16)) {
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        bf16x8 a[4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int r = rquad * 64 + m * 16 + (lane & 15);
+          a[m] = frag12(sX[b], r, h * 4 + g);
+        }
+#pragma unroll
+        for (int fn = 0; fn < NF; ++fn) {
+          int c = cwave * 64 + fn * 16 + (lane & 15);
+          bf16x8 bb = frag12(sW[b], c, h * 4 + g);
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+            acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[m], bb, acc[m][fn], 0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fn = 0; fn < NF; ++fn) {
+    int col_in = cwave * 64 + fn * 16 + (lane & 15);
+    float bia = bias[col_in];
+    int lid = loss_id[col_in];
+    int tgt = target[col_in];
+    float lsum = 0.f, dsum = 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rt * 256 + row_base + r;
+        float z = acc[m][fn][r] + bia;
+        float d;
+        if (ABL & 2) {
+          d = z;
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = fast_rcp(1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            d = z - yv;
+          }
+          if (gr >= nt) d = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        dsum += d;
+      }
+      if (ABL & 1) {
+        asm volatile("" ::"v"(dpack[0]), "v"(dpack[2]));  // keep live
+      } else {
+        *reinterpret_cast<bf16x4*>(
+            Dt + (long)col_in * NTp + rt * 256 + row_base) = dpack;
+      }
+    }
+    atomicAdd(&sDb[col_in], dsum + lsum);
+  }
+  __syncthreads();
+  if (tid < 256) atomicAdd(&db_sum[tid], sDb[tid]);
+}
+
+std::vector<double> flagship_ablate(int64_t rows, int64_t Fp, int64_t iters) {
+  TORCH_CHECK(g_layout >= 0, "probe first");
+  auto opt = torch::TensorOptions().dtype(torch::kBFloat16).device(torch::kCUDA);
+  long NTp = (rows / 256) * 256;
+  auto X = torch::randn({NTp, Fp}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA)).to(torch::kBFloat16);
+  auto Wt = torch::randn({256, Fp}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA)).to(torch::kBFloat16) * 0.01;
+  auto y = torch::ones({NTp}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
+  auto idx = torch::arange((long)NTp, torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA));
+  auto bias = torch::zeros({256}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
+  auto lossi = torch::zeros({256}, torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA));
+  auto tgt = torch::ones({256}, torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA));
+  auto Dt = torch::zeros({256L, NTp}, opt);
+  auto ls = torch::zeros({256}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
+  auto db = torch::zeros({256}, torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA));
+  hipStream_t s = cur_stream();
+  std::vector<double> out;
+  auto run = [&](auto kern) {
+    // warmup
+    for (int i = 0; i < 3; ++i)
+      hipLaunchKernelGGL(kern, dim3((unsigned)(NTp / 256)), dim3(1024), 0, s,
+                         reinterpret_cast<const __bf16*>(X.data_ptr()),
+                         y.data_ptr<float>(), idx.data_ptr<int>(),
+                         reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                         bias.data_ptr<float>(), lossi.data_ptr<int>(),
+                         tgt.data_ptr<int>(),
+                         reinterpret_cast<__bf16*>(Dt.data_ptr()),
+                         ls.data_ptr<float>(), db.data_ptr<float>(),
+                         (int)NTp, (int)NTp, (int)Fp, 256, 0);
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    HIP_CHECK(hipEventRecord(e0, s));
+    for (int i = 0; i < (int)iters; ++i)
+      hipLaunchKernelGGL(kern, dim3((unsigned)(NTp / 256)), dim3(1024), 0, s,
+                         reinterpret_cast<const __bf16*>(X.data_ptr()),
+                         y.data_ptr<float>(), idx.data_ptr<int>(),
+                         reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                         bias.data_ptr<float>(), lossi.data_ptr<int>(),
+                         tgt.data_ptr<int>(),
+                         reinterpret_cast<__bf16*>(Dt.data_ptr()),
+                         ls.data_ptr<float>(), db.data_ptr<float>(),
+                         (int)NTp, (int)NTp, (int)Fp, 256, 0);
+    HIP_CHECK(hipEventRecord(e1, s));
+    HIP_CHECK(hipEventSynchronize(e1));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+    hipEventDestroy(e0);
+    hipEventDestroy(e1);
+    out.push_back((double)ms / iters);
+  };
+  run(fwd_dAB_kernel<0>);    // full
+  run(fwd_dAB_kernel<1>);    // no Dt store
+  run(fwd_dAB_kernel<2>);    // trivial epilogue
+  run(fwd_dAB_kernel<3>);    // no store + trivial epilogue
+  run(fwd_dAB_kernel<4>);    // no X stage
+  run(fwd_dAB_kernel<8>);    // no W stage
+  run(fwd_dAB_kernel<16>);   // no MFMA
+  run(fwd_dAB_kernel<17>);   // no MFMA + no store
+  run(fwd_dAB_kernel<29>);   // stages only (no mfma/store/epilogue... 1|4 off: 16|8|4|1 = X only? 29 = 16+8+4+1: no mfma, no W, no X, no store)
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flagship_ablate", &flagship_ablate,
+        "ablation timings for the fwd kernel (ms/launch)");
   m.def("gemm_probe", &gemm_probe, "MFMA layout probe (4 combos)");
   m.def("set_layout", &set_layout);
   m.def("get_layout", &get_layout);
