@@ -2426,6 +2426,7 @@ void fwd_d16_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                     int nt, int NTp, int Fp, int Bp, int want_loss) {
   constexpr int NF = 4;
   __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
   __shared__ float sLoss[256];
   __shared__ float sDb[256];
   __shared__ int sIdx[256];
@@ -2457,8 +2458,9 @@ void fwd_d16_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     for (int q = 0; q < 2; ++q) {
       int rbase = q * 128 + wave * 8;
       int r = rbase + rl8;
-      gload_lds16(X + (long)sIdx[r] * Fp + kc + ((slot ^ (r & 7)) << 3),
-                  &sX[b][rbase * 64]);
+      int ksw = kc + ((slot ^ (r & 7)) << 3);
+      gload_lds16(X + (long)sIdx[r] * Fp + ksw, &sX[b][rbase * 64]);
+      gload_lds16(Wt + (long)r * Fp + ksw, &sW[b][rbase * 64]);
     }
   };
 
@@ -2486,9 +2488,7 @@ void fwd_d16_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 #pragma unroll
       for (int fn = 0; fn < NF; ++fn) {
         int c = cwave * 64 + fn * 16 + (lane & 15);
-        // W fragment straight from L2-resident global (map-0 logical k)
-        bf16x8 bb = *reinterpret_cast<const bf16x8*>(
-            Wt + (long)c * Fp + (kc << 6) + h * 32 + g * 8);
+        bf16x8 bb = frag12(sW[b], c, h * 4 + g);
 #pragma unroll
         for (int m = 0; m < 4; ++m)
           acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
