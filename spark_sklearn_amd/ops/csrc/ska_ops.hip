@@ -2501,7 +2501,7 @@ void fwd_d16_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 
   const bool interior =
       (rt * 256 + 256 <= nt) && (row_key == nullptr) && !want_loss;
-#pragma unroll 1
+#pragma unroll
   for (int fn = 0; fn < NF; ++fn) {
     int col_in = cwave * 64 + fn * 16 + (lane & 15);
     float bia = bias[col_in];
