@@ -888,7 +888,7 @@ void bwd_g5_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 __global__ __launch_bounds__(1024)
 void bwd_g8_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
                    const __bf16* __restrict__ Dt, float* __restrict__ Gt,
-                   long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+                   long n_rows, int NTp, int n_take, int Fp, int Bp, int seg_rows) {
   constexpr int NF = 4;
   __shared__ __bf16 sD[2][256 * 64];
   __shared__ __bf16 sXt[2][256 * 64];
@@ -900,7 +900,7 @@ void bwd_g8_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   const int fwave = wave & 3;        // feature wave 0..3
   const int ft = blockIdx.x;
   const int r0 = blockIdx.y * seg_rows;
-  const int r1 = min(r0 + seg_rows, NTp);
+  const int r1 = min(r0 + seg_rows, n_take);
   const int g = lane >> 4;
   const int srow = tid >> 2;         // staging row 0..255
   const int sseg = tid & 3;
@@ -2300,7 +2300,7 @@ void fwd_d12_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 __global__ __launch_bounds__(1024)
 void bwd_g12_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
                     const __bf16* __restrict__ Dt, float* __restrict__ Gt,
-                    long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+                    long n_rows, int NTp, int n_take, int Fp, int Bp, int seg_rows) {
   constexpr int NF = 4;
   __shared__ __bf16 sD[2][256 * 64];
   __shared__ __bf16 sXt[2][256 * 64];
@@ -2312,7 +2312,7 @@ void bwd_g12_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   const int fwave = wave & 3;        // feature wave 0..3
   const int ft = blockIdx.x;
   const int r0 = blockIdx.y * seg_rows;
-  const int r1 = min(r0 + seg_rows, NTp);
+  const int r1 = min(r0 + seg_rows, n_take);
   const int g = lane >> 4;
   const int rl8 = lane >> 3;
   const int slot = lane & 7;
@@ -2723,7 +2723,7 @@ void fwd_d13_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 __global__ __launch_bounds__(1024)
 void bwd_g13_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
                     const __bf16* __restrict__ Dt, float* __restrict__ Gt,
-                    long n_rows, int NTp, int Fp, int Bp, int seg_rows) {
+                    long n_rows, int NTp, int n_take, int Fp, int Bp, int seg_rows) {
   constexpr int NF = 4;
   __shared__ __bf16 sD[4][256 * 32];
   __shared__ __bf16 sXt[4][256 * 32];
@@ -2735,7 +2735,7 @@ void bwd_g13_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
   const int fwave = wave & 3;        // feature wave 0..3
   const int ft = blockIdx.x;
   const int r0 = blockIdx.y * seg_rows;
-  const int r1 = min(r0 + seg_rows, NTp);
+  const int r1 = min(r0 + seg_rows, n_take);
   const int g = lane >> 4;
   const int srow = wave * 16 + (lane >> 2);
   const int slot = lane & 3;
@@ -3881,7 +3881,7 @@ static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
-    return (e && *e) ? atoi(e) : 16;
+    return (e && *e) ? atoi(e) : 17;
   }();
   return v;
 }
@@ -3896,6 +3896,16 @@ static auto* flagship_bwd() {
   return v == 13 ? bwd_g13_kernel : (v == 12 ? bwd_g12_kernel : bwd_g8_kernel);
 }
 
+// Side stream for overlapped bwd passes (v17 pipelined epoch).
+static hipStream_t side_stream() {
+  static hipStream_t t = [] {
+    hipStream_t s2;
+    HIP_CHECK(hipStreamCreateWithFlags(&s2, hipStreamNonBlocking));
+    return s2;
+  }();
+  return t;
+}
+
 // Launch the flagship fwd+bwd pair for the current SKA_FLAGSHIP_V.
 static void launch_flagship_pair(
     hipStream_t s, const __bf16* Xp, const __bf16* Xtp, const float* yp,
@@ -3903,6 +3913,43 @@ static void launch_flagship_pair(
     const int* tgtp, __bf16* Dtp, float* losssum, float* dbsum,
     const int* rkp, const int* ckp, long nt, long NTp, int Fp, int Bp,
     int want_loss, float* Gtp, long n_rows, long nseg, int seg_rows) {
+  if (flagship_v() >= 17 && Bp == 256 && NTp % 256 == 0 && Fp % 256 == 0 &&
+      NTp > 2 * (long)seg_rows) {
+    // v17: epoch split into row passes small enough that each pass's D
+    // stripe (256 cands x pass_rows bf16) stays L3-resident between its
+    // fwd (producer) and bwd (consumer); bwd pass k runs on a side stream
+    // concurrently with fwd pass k+1.  PMC round 2: the D re-read was
+    // ~25% of epoch HBM traffic and the two kernels' pipe usage is
+    // complementary (fwd VALU-heavy epilogue, bwd pure streams).
+    hipStream_t sb = side_stream();
+    long segs_per_pass =
+        std::max<long>(1, (96L << 20) / (512L * (long)seg_rows));
+    long pass_rows = segs_per_pass * (long)seg_rows;
+    for (long off = 0; off < NTp; off += pass_rows) {
+      long rows = std::min<long>(pass_rows, NTp - off);
+      long nsegp = (rows + seg_rows - 1) / seg_rows;
+      long nt_loc = std::max<long>(0, nt - off);
+      hipLaunchKernelGGL(fwd_d16_kernel, dim3((unsigned)(rows / 256)),
+                         dim3(1024), 0, s, Xp, yp, idxp + off, Wtp, biasp,
+                         lossp, tgtp, Dtp + off, losssum, dbsum, rkp, ckp,
+                         (int)nt_loc, (int)NTp, Fp, Bp, want_loss);
+      hipEvent_t ev;
+      HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+      HIP_CHECK(hipEventRecord(ev, s));
+      HIP_CHECK(hipStreamWaitEvent(sb, ev, 0));
+      HIP_CHECK(hipEventDestroy(ev));
+      hipLaunchKernelGGL(flagship_bwd(),
+                         dim3((unsigned)(Fp / 256), (unsigned)nsegp),
+                         dim3(1024), 0, sb, Xtp, idxp + off, Dtp + off, Gtp,
+                         n_rows, (int)NTp, (int)rows, Fp, Bp, seg_rows);
+    }
+    hipEvent_t evb;
+    HIP_CHECK(hipEventCreateWithFlags(&evb, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(evb, sb));
+    HIP_CHECK(hipStreamWaitEvent(s, evb, 0));
+    HIP_CHECK(hipEventDestroy(evb));
+    return;
+  }
   if (flagship_v() == 15 && Bp % 128 == 0 && NTp % 128 == 0) {
     hipLaunchKernelGGL(fwd_d15_kernel,
                        dim3((unsigned)(NTp / 128), (unsigned)(Bp / 128)),
@@ -3923,8 +3970,8 @@ static void launch_flagship_pair(
   if (Fp % 256 == 0) {
     hipLaunchKernelGGL(flagship_bwd(),
                        dim3((unsigned)(Fp / 256), (unsigned)nseg), dim3(1024),
-                       0, s, Xtp, idxp, Dtp, Gtp, n_rows, (int)NTp, Fp, Bp,
-                       seg_rows);
+                       0, s, Xtp, idxp, Dtp, Gtp, n_rows, (int)NTp, (int)NTp,
+                       Fp, Bp, seg_rows);
   } else {
     hipLaunchKernelGGL(bwd_g5_kernel,
                        dim3((unsigned)(Fp / 128), (unsigned)nseg), dim3(512),
