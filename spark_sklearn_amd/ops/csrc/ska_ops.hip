@@ -3881,7 +3881,7 @@ static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
-    return (e && *e) ? atoi(e) : 17;
+    return (e && *e) ? atoi(e) : 16;
   }();
   return v;
 }
