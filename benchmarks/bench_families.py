@@ -156,10 +156,70 @@ def bench_sparse():
             "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
 
 
+
+def bench_rf():
+    """The reference's motivating workload shape scaled to 1M rows: a
+    RandomForest hyperparameter grid."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import RandomForestClassifier
+
+    rng = np.random.default_rng(0)
+    n, f = int(os.environ.get("RF_ROWS", "1000000")), 64
+    X = rng.standard_normal((n, f), dtype=np.float32)
+    y = ((X[:, 0] * X[:, 1] + X[:, 2] + X[:, 3] ** 2 - 1 +
+          0.5 * rng.standard_normal(n)) > 0).astype(np.int64)
+
+    grid = {"max_depth": [4, 6], "max_features": [8, 16]}
+    gs = ska.GridSearchCV(None,
+                          RandomForestClassifier(n_estimators=50,
+                                                 random_state=0),
+                          grid, cv=3, return_train_score=False)
+    t0 = time.perf_counter()
+    gs.fit(X, y)
+    dt = time.perf_counter() - t0
+    return {"family": "rf",
+            "config": f"{n}x{f}, 50 trees, depth 4/6 x mf 8/16, cv=3",
+            "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
+            "fits_per_sec": round(gs.n_fits_ / dt, 3),
+            "trees_per_sec": round(gs.n_fits_ * 50 / dt, 1),
+            "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
+
+
+def bench_rf_digits():
+    """The literal reference README workload: digits + RF grid (~864
+    candidates x cv=3 in the original; here the same digits matrix with a
+    compact grid plus an 864-candidate variant via RF_DIGITS_FULL=1)."""
+    import spark_sklearn_amd as ska
+    from sklearn.datasets import load_digits
+    from spark_sklearn_amd.models import RandomForestClassifier
+
+    X, y = load_digits(return_X_y=True)
+    if os.environ.get("RF_DIGITS_FULL"):
+        grid = {"max_depth": [2, 3, 4, 5, 6], "max_features": [2, 4, 6, 8],
+                "n_estimators": [20, 40, 80],
+                "min_samples_leaf": [1, 2, 4],
+                "bootstrap": [True, False]}     # 360 candidates
+    else:
+        grid = {"max_depth": [4, 6], "max_features": [4, 8],
+                "n_estimators": [20, 40]}
+    gs = ska.GridSearchCV(None,
+                          RandomForestClassifier(random_state=0), grid,
+                          cv=3, return_train_score=False)
+    t0 = time.perf_counter()
+    gs.fit(X, y)
+    dt = time.perf_counter() - t0
+    return {"family": "rf_digits", "config": "digits 1797x64, cv=3",
+            "n_cands": len(gs.cv_results_["params"]),
+            "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
+            "fits_per_sec": round(gs.n_fits_ / dt, 2),
+            "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
+
+
 if __name__ == "__main__":
     which = sys.argv[1:] or ["kmeans", "gbt", "sweep"]
     for name in which:
         fn = {"kmeans": bench_kmeans, "gbt": bench_gbt,
               "sweep": bench_sweep, "sparse": bench_sparse,
-              "gbtmulti": bench_gbt_multi}[name]
+              "gbtmulti": bench_gbt_multi, "rf": bench_rf,
+              "rfdigits": bench_rf_digits}[name]
         print(json.dumps(fn()), flush=True)

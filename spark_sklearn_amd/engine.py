@@ -596,8 +596,9 @@ def fit_single(estimator, X, y, device=None):
 
 
 from .models.gbt import GBTFamily  # noqa: E402
+from .models.rf import RFFamily  # noqa: E402
 from .models.kmeans import KMeansFamily  # noqa: E402  (no import cycle:
 # model modules only import engine lazily inside methods)
 
 FAMILIES = {"linear": LinearFamily, "kmeans": KMeansFamily,
-            "gbt": GBTFamily}
+            "gbt": GBTFamily, "rf": RFFamily}

@@ -1745,11 +1745,89 @@ __global__ void gbt_sibling_kernel(const float* __restrict__ prev,
 // serially per thread is 64-lane parallel and the histogram is read ONCE
 // (totals = last prefix element).  Deterministic tie-break: lowest
 // feature, then lowest threshold, within a +-1e-12 gain band.
+// ---------------------------------------------------------------------------
+// RandomForest (K3b) device pieces.  Shares the binned matrix, histogram,
+// sibling-subtraction, split-apply and leaf kernels with GBT; RF adds a
+// stateless-hash bootstrap resample, exact-k per-node feature subsampling
+// inside the split finder, and (for K>2) a summed-per-class gini gain.
+// Binary-gini and MSE split criteria are EXACTLY the xgboost-style gain
+// with g=y, h=1, lam=0 (the -2n1+n terms telescope in the gain
+// difference), so the single-channel path reuses gbt_split_find verbatim;
+// leaf value = mean(y) falls out of -g/(h)*lr with lr=-1.
+// The numpy reference (models/_rf_ref.py) replicates rf_hash bit-for-bit.
+// ---------------------------------------------------------------------------
+__host__ __device__ __forceinline__ unsigned rf_hash(unsigned s) {
+  s ^= s >> 16;
+  s *= 0x7feb352dU;
+  s ^= s >> 15;
+  s *= 0x846ca68bU;
+  s ^= s >> 16;
+  return s;
+}
+__device__ __forceinline__ unsigned rf_feat_hash(unsigned fseed, int nd,
+                                                 int f) {
+  return rf_hash(fseed ^ (unsigned)(nd * 2654435761u) ^
+                 (unsigned)(f * 40503u));
+}
+
+// bootstrap resample: position i draws source position (hash * nt) >> 32
+__global__ void rf_bootstrap_kernel(const int* __restrict__ idx_in,
+                                    const float* __restrict__ y_in,
+                                    int* __restrict__ idx_out,
+                                    float* __restrict__ y_out, long nt,
+                                    unsigned seed) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  for (; i < nt; i += (long)gridDim.x * blockDim.x) {
+    unsigned r = rf_hash(seed ^ (unsigned)i);
+    long p = (long)(((unsigned long long)r * (unsigned long long)nt) >> 32);
+    idx_out[i] = idx_in[p];
+    y_out[i] = y_in[p];
+  }
+}
+
+// per-class indicator planes for multiclass gini
+__global__ void rf_planes_kernel(const float* __restrict__ y_boot,
+                                 float* __restrict__ G, long nt, int K) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  for (; i < nt; i += (long)gridDim.x * blockDim.x) {
+    int cls = (int)y_boot[i];
+    for (int c = 0; c < K; ++c) G[(long)c * nt + i] = (c == cls) ? 1.f : 0.f;
+  }
+}
+
+// exact-k feature selection: block-wide binary search for the largest
+// 32-bit threshold t with |{f : rf_feat_hash(f) < t}| <= kfeat; with
+// collision-free hashes the selected set is exactly the kfeat smallest.
+__device__ unsigned rf_feat_threshold(int F, int kfeat, unsigned fseed,
+                                      int nd, int* sred) {
+  unsigned t = 0;
+  const int tid = threadIdx.x;
+  for (int bit = 31; bit >= 0; --bit) {
+    unsigned cand = t | (1u << bit);
+    int cnt = 0;
+    for (int f = tid; f < F; f += (int)blockDim.x)
+      cnt += (rf_feat_hash(fseed, nd, f) < cand) ? 1 : 0;
+    for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+    if ((tid & 63) == 0) sred[tid >> 6] = cnt;
+    __syncthreads();
+    if (tid == 0) {
+      int tot = 0;
+      for (int w = 0; w < (int)(blockDim.x >> 6); ++w) tot += sred[w];
+      sred[4] = tot;
+    }
+    __syncthreads();
+    if (sred[4] <= kfeat) t = cand;
+    __syncthreads();
+  }
+  return t;
+}
+
 __global__ __launch_bounds__(256)
 void gbt_split_find_kernel(const float* __restrict__ hist, int F,
                            int lvl_lo, float lam, float lr, int min_leaf,
                            int force_leaf, int* __restrict__ feat,
-                           int* __restrict__ thr, float* __restrict__ val) {
+                           int* __restrict__ thr, float* __restrict__ val,
+                           int kfeat, unsigned fseed) {
   const int nd_local = blockIdx.x;
   const int nd = lvl_lo + nd_local;
   const int tid = threadIdx.x;
@@ -1760,6 +1838,10 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
   __shared__ float sGain[4];
   __shared__ int sFt[4], sTt[4];
   __shared__ float sG, sH;
+  __shared__ int sRed[5];
+
+  unsigned fthr = 0xffffffffu;
+  if (kfeat < F) fthr = rf_feat_threshold(F, kfeat, fseed, nd, sRed);
 
   float bgain = 0.f;
   int bf = -1, bt = 0;
@@ -1786,7 +1868,8 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
     }
     float G = __shfl(pg, 63), H = __shfl(ph, 63), C = __shfl(pc, 63);
     if (f == 0 && lane == 0) { sG = G; sH = H; }
-    if (!force_leaf) {
+    if (!force_leaf &&
+        (kfeat >= F || rf_feat_hash(fseed, nd, f) < fthr)) {
       float ps = G * G / (H + lam + 1e-12f);
       float gl = pg - lg, hl = ph - lh, cl = pc - lc;   // exclusive base
 #pragma unroll
@@ -4661,7 +4744,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
       hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
                          0, s, hist.data_ptr<float>(), F, lvl_lo,
                          (float)lam, (float)lrate, (int)min_leaf, 0,
-                         feat_t, thr_t, val_t);
+                         feat_t, thr_t, val_t, 1 << 30, 0u);
       if (d + 1 < (int)max_depth)
         hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
             .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
@@ -5186,7 +5269,8 @@ std::vector<torch::Tensor> gbt_fit_multi(
         hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes),
                            dim3(256), 0, s, hist.data_ptr<float>(), F,
                            lvl_lo, (float)lam, (float)lrate,
-                           (int)min_leaf, 0, feat_t, thr_t, val_t);
+                           (int)min_leaf, 0, feat_t, thr_t, val_t,
+                           1 << 30, 0u);
         if (d + 1 < (int)max_depth)
           hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
               .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
@@ -5235,6 +5319,291 @@ torch::Tensor gbt_metric_multi(torch::Tensor raws, torch::Tensor y,
                      (int)K, out.data_ptr<double>());
   HIP_CHECK(hipGetLastError());
   return out;
+}
+
+// ---------------------------------------------------------------------------
+// rf_split_find_multi_kernel: summed per-class gini gain over K histogram
+// copies (class c's g-channel = count of class c; the count channel is
+// identical in every copy, class 0's is used).  Writes the winning
+// (feature, threshold) into all K planes of the tree so gbt_predict can
+// walk each plane unchanged; no-split nodes get per-class leaf
+// probabilities G_c / C.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256)
+void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
+                                int K, int F, int lvl_lo, int min_leaf,
+                                int kfeat, unsigned fseed,
+                                int* __restrict__ feats,
+                                int* __restrict__ thrs,
+                                float* __restrict__ vals, long NN) {
+  const int nd_local = blockIdx.x;
+  const int nd = lvl_lo + nd_local;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  __shared__ float sGain[4];
+  __shared__ int sFt[4], sTt[4];
+  __shared__ int sRed[5];
+  __shared__ float sCnt;
+  __shared__ float sPC[64];
+
+  unsigned fthr = 0xffffffffu;
+  if (kfeat < F) fthr = rf_feat_threshold(F, kfeat, fseed, nd, sRed);
+
+  float bgain = 0.f;
+  int bf = -1, bt = 0;
+  const float eps = 1e-12f;
+
+  for (int f = wave; f < F; f += 4) {
+    bool active = (kfeat >= F) || (rf_feat_hash(fseed, nd, f) < fthr);
+    const float* fh0 =
+        hist + (long)nd_local * F * 256 * 3 + (long)f * 256 * 3;
+    float c4[4];
+    float lc = 0.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      c4[j] = fh0[(4 * lane + j) * 3 + 2];
+      lc += c4[j];
+    }
+    float pc = lc;
+    for (int off = 1; off < 64; off <<= 1) {
+      float t = __shfl_up(pc, off);
+      if (lane >= off) pc += t;
+    }
+    float C = __shfl(pc, 63);
+    float cl[4];
+    {
+      float a = pc - lc;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        a += c4[j];
+        cl[j] = a;
+      }
+    }
+    if (f == 0 && lane == 0) sCnt = C;
+    float gainsum[4] = {0.f, 0.f, 0.f, 0.f};
+    float parent = 0.f;
+    for (int c = 0; c < K; ++c) {
+      const float* fhc = fh0 + (long)c * cstride;
+      float g4[4];
+      float lg = 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        g4[j] = fhc[(4 * lane + j) * 3];
+        lg += g4[j];
+      }
+      float pg = lg;
+      for (int off = 1; off < 64; off <<= 1) {
+        float t = __shfl_up(pg, off);
+        if (lane >= off) pg += t;
+      }
+      float G = __shfl(pg, 63);
+      if (f == 0 && lane == 0) sPC[c] = G;
+      float a = pg - lg;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        a += g4[j];
+        float gl = a, gr = G - gl;
+        gainsum[j] +=
+            gl * gl / (cl[j] + eps) + gr * gr / ((C - cl[j]) + eps);
+      }
+      parent += G * G / (C + eps);
+    }
+    if (active) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int b = 4 * lane + j;
+        if (b >= 255) continue;
+        if (cl[j] < (float)min_leaf || C - cl[j] < (float)min_leaf)
+          continue;
+        float gain = gainsum[j] - parent;
+        if (gain > bgain + 1e-12f ||
+            (gain > bgain - 1e-12f &&
+             (bf < 0 || f < bf || (f == bf && b < bt)))) {
+          bgain = gain;
+          bf = f;
+          bt = b;
+        }
+      }
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    float og = __shfl_down(bgain, off);
+    int of = __shfl_down(bf, off);
+    int ot = __shfl_down(bt, off);
+    if (og > bgain + 1e-12f ||
+        (og > bgain - 1e-12f && of >= 0 &&
+         (bf < 0 || of < bf || (of == bf && ot < bt)))) {
+      bgain = og;
+      bf = of;
+      bt = ot;
+    }
+  }
+  if (lane == 0) { sGain[wave] = bgain; sFt[wave] = bf; sTt[wave] = bt; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w) {
+      if (sGain[w] > sGain[0] + 1e-12f ||
+          (sGain[w] > sGain[0] - 1e-12f && sFt[w] >= 0 &&
+           (sFt[0] < 0 || sFt[w] < sFt[0] ||
+            (sFt[w] == sFt[0] && sTt[w] < sTt[0])))) {
+        sGain[0] = sGain[w];
+        sFt[0] = sFt[w];
+        sTt[0] = sTt[w];
+      }
+    }
+    if (sFt[0] < 0 || sGain[0] <= 1e-12f) {
+      for (int c = 0; c < K; ++c) {
+        feats[(long)c * NN + nd] = -1;
+        if (sCnt > 0.f)
+          vals[(long)c * NN + nd] = sPC[c] / (sCnt + eps);
+      }
+    } else {
+      for (int c = 0; c < K; ++c) {
+        feats[(long)c * NN + nd] = sFt[0];
+        thrs[(long)c * NN + nd] = sTt[0];
+        vals[(long)c * NN + nd] = 0.f;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// rf_fit: RandomForest fit over the shared binned matrix.  Trees are
+// independent: per tree a stateless-hash bootstrap resample of the train
+// positions, level-wise growth with gini (classification) / MSE
+// (regression) splits and exact-k per-node feature subsampling.  Returns
+// (feats, thrs, vals) in the gbt_fit layout ([T*planes][NN]); leaf values
+// are class probabilities / means ALREADY divided by n_estimators so the
+// summing gbt_predict yields the forest average.
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor idxmap0,
+                                  torch::Tensor y_tr, int64_t K,
+                                  int64_t n_estimators, int64_t max_depth,
+                                  int64_t min_leaf, int64_t max_features,
+                                  int64_t seed, int64_t bootstrap) {
+  const at::cuda::CUDAGuard guard(Xb.device());
+  TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 6,
+              "device RF supports max_depth in [1, 6]");
+  TORCH_CHECK(K <= 64, "device RF supports at most 64 classes");
+  int F = (int)Xb.size(1);
+  long nt = idxmap0.size(0);
+  int T = (int)n_estimators;
+  int NN = (1 << (max_depth + 1)) - 1;
+  int KP = K > 2 ? (int)K : 1;          // histogram / tree planes
+  int kfeat = (max_features <= 0 || max_features >= F) ? (1 << 30)
+                                                       : (int)max_features;
+
+  auto optF = y_tr.options().dtype(torch::kFloat32);
+  auto optI = y_tr.options().dtype(torch::kInt32);
+  auto node_id = torch::zeros({nt}, optI);
+  auto feats = torch::full({(long)T * KP, (long)NN}, -1, optI);
+  auto thrs = torch::zeros({(long)T * KP, (long)NN}, optI);
+  auto vals = torch::zeros({(long)T * KP, (long)NN}, optF);
+  int max_lvl_nodes = 1 << (max_depth - 1);
+  long ccells = (long)max_lvl_nodes * F * 256 * 3;
+  auto hist = torch::empty({ccells * KP}, optF);
+  auto hist_prev = torch::empty(
+      {(long)std::max(1, max_lvl_nodes / 2) * F * 256 * 3 * KP}, optF);
+  long pcells = (long)std::max(1, max_lvl_nodes / 2) * F * 256 * 3;
+  auto gsum = torch::zeros({NN}, optF);
+  auto hsum = torch::zeros({NN}, optF);
+  auto ones = torch::ones({nt}, optF);
+  auto idx_t = bootstrap ? torch::empty({nt}, optI) : idxmap0;
+  auto y_boot = bootstrap ? torch::empty({nt}, optF) : y_tr;
+  auto G = (KP > 1) ? torch::empty({(long)KP, nt}, optF)
+                    : torch::empty({0}, optF);
+
+  hipStream_t s = cur_stream();
+  const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+
+  for (int t = 0; t < T; ++t) {
+    unsigned st = rf_hash((unsigned)seed ^ (unsigned)(t * 0x9e3779b9u));
+    unsigned fs = rf_hash(st ^ 0x85ebca6bu);
+    if (bootstrap) {
+      hipLaunchKernelGGL(rf_bootstrap_kernel, dim3(blocks), dim3(256), 0, s,
+                         idxmap0.data_ptr<int>(), y_tr.data_ptr<float>(),
+                         idx_t.data_ptr<int>(), y_boot.data_ptr<float>(),
+                         nt, st);
+    }
+    if (KP > 1)
+      hipLaunchKernelGGL(rf_planes_kernel, dim3(blocks), dim3(256), 0, s,
+                         y_boot.data_ptr<float>(), G.data_ptr<float>(), nt,
+                         KP);
+    HIP_CHECK(hipMemsetAsync(node_id.data_ptr(), 0, nt * 4, s));
+    int* feat_t = feats.data_ptr<int>() + (long)t * KP * NN;
+    int* thr_t = thrs.data_ptr<int>() + (long)t * KP * NN;
+    float* val_t = vals.data_ptr<float>() + (long)t * KP * NN;
+    for (int d = 0; d < (int)max_depth; ++d) {
+      int lvl_lo = (1 << d) - 1;
+      int lvl_nodes = 1 << d;
+      for (int c = 0; c < KP; ++c) {
+        const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
+                                   : y_boot.data_ptr<float>();
+        launch_hist_level(Xp, idx_t.data_ptr<int>(),
+                          node_id.data_ptr<int>(), gp,
+                          ones.data_ptr<float>(),
+                          hist.data_ptr<float>() + (long)c * ccells, nt, F,
+                          d, 1.f, false, s);
+        if (d > 0) {
+          int prev_lo = (1 << (d - 1)) - 1;
+          long cells = (long)F * 256 * 3;
+          unsigned gx = (unsigned)std::min<long>((cells + 255) / 256, 1024);
+          hipLaunchKernelGGL(gbt_sibling_kernel,
+                             dim3(gx, (unsigned)(lvl_nodes / 2)), dim3(256),
+                             0, s,
+                             hist_prev.data_ptr<float>() + (long)c * pcells,
+                             hist.data_ptr<float>() + (long)c * ccells,
+                             feat_t, prev_lo, F, lvl_nodes);
+        }
+      }
+      if (KP > 1) {
+        hipLaunchKernelGGL(rf_split_find_multi_kernel, dim3(lvl_nodes),
+                           dim3(256), 0, s, hist.data_ptr<float>(), ccells,
+                           KP, F, lvl_lo, (int)min_leaf, kfeat, fs, feat_t,
+                           thr_t, val_t, (long)NN);
+      } else {
+        hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
+                           0, s, hist.data_ptr<float>(), F, lvl_lo, 0.f,
+                           -1.f, (int)min_leaf, 0, feat_t, thr_t, val_t,
+                           kfeat, fs);
+      }
+      if (d + 1 < (int)max_depth) {
+        long lvlcells = (long)lvl_nodes * F * 256 * 3;
+        for (int c = 0; c < KP; ++c)
+          hist_prev.narrow(0, (long)c * pcells, lvlcells)
+              .copy_(hist.narrow(0, (long)c * ccells, lvlcells),
+                     /*non_blocking=*/true);
+      }
+      hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
+                         0, s, Xp, idx_t.data_ptr<int>(),
+                         node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
+                         lvl_lo, lvl_lo + lvl_nodes);
+      HIP_CHECK(hipGetLastError());
+    }
+    int bl_lo = (1 << max_depth) - 1;
+    for (int c = 0; c < KP; ++c) {
+      const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
+                                 : y_boot.data_ptr<float>();
+      HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
+      HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
+      hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0,
+                         s, node_id.data_ptr<int>(), gp,
+                         ones.data_ptr<float>(), gsum.data_ptr<float>(),
+                         hsum.data_ptr<float>(), nt, bl_lo, NN, 0.f, 0.f);
+      hipLaunchKernelGGL(gbt_leaf_vals_kernel,
+                         dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
+                         gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                         val_t + (long)c * NN, feat_t + (long)c * NN, 0.f,
+                         -1.f, bl_lo, NN);
+    }
+    HIP_CHECK(hipGetLastError());
+  }
+  vals.mul_(1.0 / (double)T);           // forest average via summing walk
+  return {feats, thrs, vals};
 }
 
 // ---------------------------------------------------------------------------
@@ -5454,6 +5823,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_norms", &row_norms, "per-row squared norms");
   m.def("gbt_bin", &gbt_bin, "device quantile binning");
   m.def("gbt_fit", &gbt_fit, "histogram boosting fit (depth-wise trees)");
+  m.def("rf_fit", &rf_fit,
+        "RandomForest fit (bootstrap + gini/mse splits, feature subsampling)");
   m.def("gbt_predict", &gbt_predict, "tree-walk raw predictions");
   m.def("gbt_metric", &gbt_metric, "accuracy/ss_res sums");
   m.def("csr_linear_fit", &csr_linear_fit,

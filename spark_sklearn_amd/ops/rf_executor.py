@@ -1,0 +1,102 @@
+"""Device execution of RandomForest candidate chunks (K3b kernels).
+
+Shares the resident uint8 binned matrix (and its cache) with the GBT
+executor; per candidate, `rf_fit` runs the whole forest on device --
+bootstrap resample, per-level histograms with sibling subtraction,
+gini/MSE split scan with exact-k feature subsampling, leaf
+probabilities -- with no host round-trips between trees.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from .gbt_executor import DeviceGBTExecutor
+
+__all__ = ["DeviceRFExecutor"]
+
+
+class DeviceRFExecutor(DeviceGBTExecutor):
+    def fit(self, train_idx, plan):
+        torch = self.torch
+        idx_np = np.asarray(train_idx, dtype=np.int32)
+        idx_t = torch.as_tensor(idx_np).to(self.device)
+        y_tr_np = self.y_np[idx_np]
+        y_tr = torch.as_tensor(y_tr_np).to(self.device)
+
+        models_host, models_dev = [], []
+        n_iter = np.empty(plan.B, np.int32)
+        K = self.K
+        for i, spec in enumerate(plan.specs):
+            if spec.n_estimators == 0:
+                models_host.append([])
+                models_dev.append(None)
+                n_iter[i] = 0
+                continue
+            feats, thrs, vals = self.ext.rf_fit(
+                self.Xb, idx_t, y_tr, K, spec.n_estimators, spec.max_depth,
+                spec.min_samples_leaf, spec.max_features,
+                spec.random_state, 1 if spec.bootstrap else 0)
+            T = spec.n_estimators
+            f_np = feats.cpu().numpy()
+            t_np = thrs.cpu().numpy().astype(np.uint8)
+            v_np = vals.cpu().numpy()
+            if K > 2:
+                models_dev.append(("multi", feats, thrs, vals, spec))
+                trees = [[(f_np[t * K + c], t_np[t * K + c],
+                           v_np[t * K + c]) for c in range(K)]
+                         for t in range(T)]
+            else:
+                models_dev.append(("flat", feats, thrs, vals, spec))
+                trees = [(f_np[t], t_np[t], v_np[t]) for t in range(T)]
+            models_host.append(trees)
+            n_iter[i] = T
+        torch.cuda.synchronize(self.device)
+        self._models_dev = models_dev
+        return (models_host, self.edges), n_iter, None
+
+    def scores(self, idx, fitted, plan, metric):
+        torch = self.torch
+        if self._models_dev is None:
+            raise RuntimeError("scores() requires a preceding fit()")
+        idx_t = torch.as_tensor(
+            np.asarray(idx, dtype=np.int32)).to(self.device)
+        n = len(idx)
+        out = np.empty(len(self._models_dev))
+        mid = 0 if metric == "accuracy" else 1
+        if metric == "r2":
+            y_sel = self.y_np[np.asarray(idx)]
+            ss_tot = float(np.sum((y_sel - y_sel.mean()) ** 2))
+        for i, m in enumerate(self._models_dev):
+            if m is None:
+                out[i] = float("nan")
+                continue
+            kind, feats, thrs, vals, spec = m
+            T = spec.n_estimators
+            if kind == "multi":
+                K = self.K
+                NN = feats.shape[1]
+                raws = torch.empty((K, n), dtype=torch.float32,
+                                   device=self.device)
+                fv = feats.view(T, K, NN)
+                tv = thrs.view(T, K, NN)
+                vv = vals.view(T, K, NN)
+                for c in range(K):
+                    raws[c] = self.ext.gbt_predict(
+                        self.Xb, idx_t, fv[:, c].contiguous(),
+                        tv[:, c].contiguous(), vv[:, c].contiguous(), 0.0)
+                sums = self.ext.gbt_metric_multi(raws, self.y, idx_t,
+                                                 self.K)
+                out[i] = float(sums.item()) / n
+                continue
+            # binary: leaf vals are p/T; raw + f0(-0.5) > 0 <=> mean p > .5
+            f0 = -0.5 if metric == "accuracy" else 0.0
+            raw = self.ext.gbt_predict(self.Xb, idx_t, feats, thrs, vals,
+                                       f0)
+            sums = self.ext.gbt_metric(raw, self.y, idx_t, mid)
+            v = float(sums.item())
+            if metric == "accuracy":
+                out[i] = v / n
+            else:
+                out[i] = 1.0 - v / max(ss_tot, 1e-12)
+        return out

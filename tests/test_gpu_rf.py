@@ -1,0 +1,114 @@
+"""GPU tests: device RandomForest kernels vs the numpy reference."""
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+pytestmark = pytest.mark.gpu
+
+from sklearn.datasets import make_classification  # noqa: E402
+
+from spark_sklearn_amd import GridSearchCV  # noqa: E402
+from spark_sklearn_amd.models import RandomForestClassifier  # noqa: E402
+from spark_sklearn_amd.models._rf_ref import (RFSpec,  # noqa: E402
+                                              rf_fit_ref, rf_predict_ref)
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    return torch.device("cuda:0")
+
+
+def _ext():
+    from spark_sklearn_amd.ops import load_extension
+
+    return load_extension()
+
+
+def _dev_trees(ext, dev, Xb_np, y_np, spec, K=0):
+    Xb = torch.as_tensor(Xb_np).to(dev)
+    idx = torch.arange(Xb_np.shape[0], dtype=torch.int32, device=dev)
+    y = torch.as_tensor(y_np.astype(np.float32)).to(dev)
+    feats, thrs, vals = ext.rf_fit(
+        Xb, idx, y, K, spec.n_estimators, spec.max_depth,
+        spec.min_samples_leaf, spec.max_features, spec.random_state,
+        1 if spec.bootstrap else 0)
+    return (feats.cpu().numpy(), thrs.cpu().numpy().astype(np.uint8),
+            vals.cpu().numpy())
+
+
+def test_rf_device_matches_ref_binary(dev):
+    rng = np.random.default_rng(0)
+    n = 4000
+    Xb = rng.integers(0, 64, size=(n, 16), dtype=np.uint8)
+    # well-separated target: split margins are clear, so float-atomic
+    # ordering cannot flip them
+    y = ((Xb[:, 0] > 30) ^ (Xb[:, 3] > 12)).astype(np.float32)
+    spec = RFSpec(n_estimators=6, max_depth=4, min_samples_leaf=2,
+                  max_features=8, random_state=11)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec)
+    ref = rf_fit_ref(Xb, y, spec)
+    for t, (f_r, t_r, v_r) in enumerate(ref):
+        assert np.array_equal(f_d[t], f_r), f"tree {t} feats differ"
+        split = f_d[t] >= 0
+        assert np.array_equal(t_d[t][split], t_r[split]), f"tree {t} thrs"
+        np.testing.assert_allclose(v_d[t], v_r, atol=2e-5)
+
+
+def test_rf_device_matches_ref_multiclass(dev):
+    rng = np.random.default_rng(2)
+    n, K = 3000, 5
+    Xb = rng.integers(0, 32, size=(n, 12), dtype=np.uint8)
+    y = (Xb[:, 1] // 7 + (Xb[:, 5] > 16)).astype(np.float32) % K
+    spec = RFSpec(n_estimators=4, max_depth=3, min_samples_leaf=2,
+                  max_features=6, random_state=5, n_classes=K)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec, K=K)
+    ref = rf_fit_ref(Xb, y, spec)
+    for t, planes in enumerate(ref):
+        for c, (f_r, t_r, v_r) in enumerate(planes):
+            i = t * K + c
+            assert np.array_equal(f_d[i], f_r), f"tree {t} class {c}"
+            split = f_d[i] >= 0
+            assert np.array_equal(t_d[i][split], t_r[split])
+            np.testing.assert_allclose(v_d[i], v_r, atol=2e-5)
+
+
+def test_rf_device_no_bootstrap_no_mask_matches(dev):
+    rng = np.random.default_rng(3)
+    Xb = rng.integers(0, 64, size=(2000, 8), dtype=np.uint8)
+    y = (Xb[:, 2] > 40).astype(np.float32)
+    spec = RFSpec(n_estimators=2, max_depth=3, bootstrap=False,
+                  max_features=0, random_state=0)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec)
+    ref = rf_fit_ref(Xb, y, spec)
+    for t, (f_r, t_r, v_r) in enumerate(ref):
+        assert np.array_equal(f_d[t], f_r)
+        np.testing.assert_allclose(v_d[t], v_r, atol=2e-5)
+
+
+def test_rf_grid_search_device_end_to_end(dev):
+    X, y = make_classification(n_samples=20_000, n_features=32,
+                               n_informative=12, random_state=0)
+    gs = GridSearchCV(None, RandomForestClassifier(random_state=0),
+                      {"max_depth": [4, 6], "n_estimators": [10, 25]},
+                      cv=3)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.85
+    assert gs.cv_results_["mean_test_score"].shape == (4,)
+    # device executor actually ran
+    from spark_sklearn_amd.engine import get_family
+
+    fam = get_family(RandomForestClassifier())
+    assert fam.name == "rf"
+
+
+def test_rf_device_digits_shape_multiclass_accuracy(dev):
+    from sklearn.datasets import load_digits
+
+    X, y = load_digits(return_X_y=True)
+    gs = GridSearchCV(None, RandomForestClassifier(random_state=0),
+                      {"max_depth": [6], "n_estimators": [20, 40]}, cv=3)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.85
