@@ -43,9 +43,12 @@ def test_rf_device_matches_ref_binary(dev):
     rng = np.random.default_rng(0)
     n = 4000
     Xb = rng.integers(0, 64, size=(n, 16), dtype=np.uint8)
-    # well-separated target: split margins are clear, so float-atomic
-    # ordering cannot flip them
-    y = ((Xb[:, 0] > 30) ^ (Xb[:, 3] > 12)).astype(np.float32)
+    # additive, well-separated target: every candidate split has a clear
+    # margin, so fp32 (device) vs fp64 (numpy) gain rounding cannot flip
+    # the argmax (an XOR target makes all first-level gains ~0 and is
+    # decided by noise)
+    y = (Xb[:, 0].astype(np.float32) + 0.5 * Xb[:, 3] > 40).astype(
+        np.float32)
     spec = RFSpec(n_estimators=6, max_depth=4, min_samples_leaf=2,
                   max_features=8, random_state=11)
     f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec)
