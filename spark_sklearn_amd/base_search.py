@@ -303,6 +303,26 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
     def _get_param_iterator(self):
         raise NotImplementedError
 
+    def _maybe_broadcast_input(self, backend, X, y):
+        if getattr(backend, "world_size", 1) <= 1 or \
+                not hasattr(backend, "broadcast_data"):
+            if X is None:
+                raise ValueError(
+                    "X is None: only ranks of a distributed backend may "
+                    "omit X (rank 0 must always provide it)")
+            return X, y
+        dist = backend._dist
+        flags = [None] * backend.world_size
+        dist.all_gather_object(flags, X is not None)
+        if all(flags):
+            return X, y
+        X_t, y_t = backend.broadcast_data(X, y)
+        dev = getattr(backend, "device", None)
+        if dev is None or getattr(dev, "type", "cpu") != "cuda":
+            X_t = X_t.numpy()
+            y_t = y_t.numpy() if y_t is not None else None
+        return X_t, y_t
+
     def _get_backend(self):
         """Resolve the execution backend from the context handle."""
         ctx = getattr(self, "ctx", None)
@@ -314,6 +334,12 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
     def fit(self, X, y=None, groups=None, **fit_params):
         """Run the full search: build tasks, dispatch, aggregate, refit."""
         estimator = self.estimator
+        backend = self._get_backend()
+        # RC1: the reference's _fit broadcasts X/y itself (SURVEY.md
+        # section 3.1); under a distributed backend, non-root ranks may
+        # pass X=None and receive rank 0's arrays (resident in HBM on the
+        # RCCL tier) before any splitting happens.
+        X, y = self._maybe_broadcast_input(backend, X, y)
         cv = check_cv(self.cv, y, classifier=is_classifier(estimator))
         if isinstance(self.scoring, (dict, list, tuple, set)):
             from sklearn.metrics._scorer import _check_multimetric_scoring
@@ -361,8 +387,6 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 tasks.append(Task(idx, cand_i, fold_i, params,
                                   np.asarray(train), np.asarray(test)))
                 idx += 1
-
-        backend = self._get_backend()
 
         # resume: replay completed (candidate, fold) fits from a prior
         # run's score log; only the remainder executes

@@ -25,7 +25,8 @@ def resolve_backend(ctx=None) -> Backend:
     * a :class:`Backend` instance -- used as-is.
     * a :class:`spark_sklearn_amd.HipContext` -- its backend.
     * strings: ``"sequential"``, ``"threads"`` / ``"threads:N"``,
-      ``"dist"`` / ``"distributed"``.
+      ``"gpu"`` (single-process sequential dispatch on the default
+      device), ``"dist"`` / ``"distributed"``.
     """
     if isinstance(ctx, Backend):
         return ctx
@@ -40,6 +41,10 @@ def resolve_backend(ctx=None) -> Backend:
             return ThreadBackend(n)
         if name in ("dist", "distributed"):
             return DistributedBackend()
+        if name == "gpu":
+            # single-process GPU execution: sequential dispatch, device
+            # resolved by the context (cuda:0 by default when visible)
+            return SequentialBackend()
         if name == "auto":
             ctx = None
         else:

@@ -46,7 +46,8 @@ class DistributedBackend(Backend):
         Thread-pool width for host fits on each rank.
     """
 
-    def __init__(self, device=None, n_threads: Optional[int] = None):
+    def __init__(self, device=None, n_threads: Optional[int] = None,
+                 partition: str = "static", dynamic_block: int = 8):
         import torch
         import torch.distributed as dist
 
@@ -54,9 +55,14 @@ class DistributedBackend(Backend):
             raise RuntimeError(
                 "DistributedBackend requires torch.distributed to be "
                 "initialized (launch with torchrun / init_process_group)")
+        if partition not in ("static", "dynamic"):
+            raise ValueError("partition must be 'static' or 'dynamic'")
         self.rank = dist.get_rank()
         self.world_size = dist.get_world_size()
         self.n_threads = n_threads
+        self.partition = partition
+        self.dynamic_block = int(dynamic_block)
+        self._fit_seq = 0
         if device is None and torch.cuda.is_available():
             import os
 
@@ -108,19 +114,62 @@ class DistributedBackend(Backend):
         on one rank so batched kernels see whole candidates."""
         return [t for t in tasks if t.cand_index % self.world_size == self.rank]
 
+    def _run_dynamic(self, estimator, X, y, tasks, exec_kwargs):
+        """Bounded dynamic partition: whole-candidate blocks are pulled
+        from a shared atomic counter (the process group's rendezvous
+        store -- works on gloo AND nccl) as each rank finishes its
+        previous block, so skewed per-candidate costs balance.  Trades
+        cross-candidate kernel batching width for balance; block size is
+        the bound (HipContext(dynamic_block=...))."""
+        dist = self._dist
+        self._fit_seq += 1
+        key = f"ska_dynq_{self._fit_seq}"
+        store = dist.distributed_c10d._get_default_store()
+        by_cand = {}
+        order = []
+        for t in tasks:
+            if t.cand_index not in by_cand:
+                by_cand[t.cand_index] = []
+                order.append(t.cand_index)
+            by_cand[t.cand_index].append(t)
+        blk = max(1, self.dynamic_block)
+        nblocks = (len(order) + blk - 1) // blk
+        results = []
+        pulled = 0
+        while True:
+            i = int(store.add(key, 1)) - 1
+            if i >= nblocks:
+                break
+            pulled += 1
+            block = []
+            for c in order[i * blk:(i + 1) * blk]:
+                block.extend(by_cand[c])
+            results.extend(execute_local_tasks(
+                estimator, X, y, block, **exec_kwargs))
+        self.n_blocks_pulled_ = pulled
+        return results
+
     def run_search(self, estimator, X, y, tasks, *, scorer, fit_params,
                    return_train_score, error_score, verbose):
         torch, dist = self._torch, self._dist
-        local = self.local_task_filter(tasks)
-        if verbose > 0 and self.rank == 0:
-            print(f"[dist] {self.world_size} ranks, "
-                  f"{len(tasks)} tasks (~{len(local)} per rank)")
-        results = execute_local_tasks(
-            estimator, X, y, local, scorer=scorer, fit_params=fit_params,
+        exec_kwargs = dict(
+            scorer=scorer, fit_params=fit_params,
             return_train_score=return_train_score, error_score=error_score,
             verbose=verbose, n_threads=self.n_threads, device=self.device,
             progress=getattr(self, "progress", None),
             column_batch=getattr(self, "candidate_batch", None))
+        if self.partition == "dynamic":
+            if verbose > 0 and self.rank == 0:
+                print(f"[dist] {self.world_size} ranks, {len(tasks)} tasks "
+                      f"(dynamic blocks of {self.dynamic_block} candidates)")
+            results = self._run_dynamic(estimator, X, y, tasks, exec_kwargs)
+        else:
+            local = self.local_task_filter(tasks)
+            if verbose > 0 and self.rank == 0:
+                print(f"[dist] {self.world_size} ranks, "
+                      f"{len(tasks)} tasks (~{len(local)} per rank)")
+            results = execute_local_tasks(estimator, X, y, local,
+                                          **exec_kwargs)
 
         # RC2: dense float64 score table, one sum-all-reduce.  Single
         # metric: (n_tasks, 7) fixed-width records.  Multimetric: the
@@ -158,8 +207,17 @@ class DistributedBackend(Backend):
                 dtype=torch.float64)
         dist.all_reduce(table, op=dist.ReduceOp.SUM)
         table = table.cpu().numpy()
-        local_err = {r.index: r.error for r in results
+        local_err = {r.index: str(r.error) for r in results
                      if r.error is not None}
+        # gather real error text from every rank so the FitFailedWarning
+        # keeps its diagnostic payload (round-1 gap: remote failures
+        # surfaced as 'error text not gathered')
+        err_shards = [None] * self.world_size
+        dist.all_gather_object(err_shards, local_err)
+        all_err = {}
+        for shard in err_shards:
+            if shard:
+                all_err.update(shard)
 
         def _unpack(row, lo, failed):
             if keys is None or failed:
@@ -173,7 +231,7 @@ class DistributedBackend(Backend):
                 int(row[0]), _unpack(row, 1, failed),
                 _unpack(row, 1 + M, failed), int(row[1 + 2 * M]),
                 row[2 + 2 * M], row[3 + 2 * M],
-                error=(local_err.get(int(row[0]),
+                error=(all_err.get(int(row[0]),
                        "fit failed on a remote rank "
                        "(error text not gathered)") if failed else None)))
         return out

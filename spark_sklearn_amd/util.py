@@ -30,7 +30,8 @@ class HipContext:
     def __init__(self, backend=None, device=None, candidate_batch: int = 256,
                  n_streams: int = 1, dtype: str = "bfloat16",
                  n_threads: Optional[int] = None, score_log: str = None,
-                 resume_from_log: str = None, progress=None):
+                 resume_from_log: str = None, progress=None,
+                 partition: str = "static", dynamic_block: int = 8):
         from .parallel import resolve_backend
 
         if dtype != "bfloat16":
@@ -46,6 +47,10 @@ class HipContext:
                 "roadmap")
         if not (1 <= int(candidate_batch) <= 256):
             raise ValueError("candidate_batch must be in [1, 256]")
+        if partition not in ("static", "dynamic"):
+            raise ValueError("partition must be 'static' or 'dynamic'")
+        self.partition = partition
+        self.dynamic_block = int(dynamic_block)
         self.candidate_batch = candidate_batch
         self.n_streams = n_streams
         self.dtype = dtype
@@ -57,7 +62,8 @@ class HipContext:
         self.backend = resolve_backend(backend)
         # push engine knobs down where backends look for them
         for attr in ("candidate_batch", "n_streams", "dtype", "score_log",
-                     "resume_from_log", "progress"):
+                     "resume_from_log", "progress", "partition",
+                     "dynamic_block"):
             setattr(self.backend, attr, getattr(self, attr))
 
     @property
