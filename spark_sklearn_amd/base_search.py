@@ -393,8 +393,10 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         resume_path = getattr(backend, "resume_from_log", None)
         restored: List[TaskResult] = []
         todo = tasks
+        log_header = self._score_log_header(X, y)
         if resume_path:
-            done = self._load_resume_records(resume_path, n_splits)
+            done = self._load_resume_records(resume_path, n_splits,
+                                             log_header)
             if done:
                 todo = []
                 for t in tasks:
@@ -454,7 +456,8 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             self._append_score_log(
                 score_log,
                 [t for t in tasks if t.index in executed],
-                [r for r in ordered if r.index in executed])
+                [r for r in ordered if r.index in executed],
+                header=log_header)
 
         # multimetric-ness is declared (dict/list scoring) or discovered
         # (a plain callable that returned a dict of scores)
@@ -598,12 +601,61 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         """Stable text key for a candidate's params (resume matching)."""
         return repr(sorted((k, repr(v)) for k, v in params.items()))
 
-    def _append_score_log(self, path, tasks, ordered):
+    def _score_log_header(self, X, y):
+        """Identity of this search for score-log resume validation: a
+        cheap strided fingerprint of (X, y) plus the splitter, estimator
+        class and scoring.  A log written against different data or CV
+        must not be silently replayed into cv_results_."""
+        import hashlib
+
+        h = hashlib.sha1()
+
+        def upd(a):
+            if a is None:
+                h.update(b"none")
+                return
+            flat = None
+            try:
+                if hasattr(a, "detach"):          # torch tensor (any device)
+                    t = a.detach().reshape(-1)
+                    step = max(1, t.shape[0] // 1024)
+                    flat = t[::step][:1024].cpu().numpy()
+                    h.update(str(tuple(a.shape)).encode())
+                    h.update(str(a.dtype).encode())
+                else:
+                    arr = np.asarray(a)
+                    h.update(str(arr.shape).encode())
+                    h.update(str(arr.dtype).encode())
+                    if arr.size:
+                        f2 = arr.reshape(-1)
+                        step = max(1, f2.shape[0] // 1024)
+                        flat = f2[::step][:1024]
+                if flat is not None:
+                    h.update(np.ascontiguousarray(flat).tobytes())
+            except Exception:
+                h.update(str(type(a)).encode())
+
+        upd(X)
+        upd(y)
+        return {
+            "header": 1,
+            "data": h.hexdigest(),
+            "cv": repr(self.cv),
+            "estimator": type(self.estimator).__name__,
+            "scoring": repr(self.scoring),
+        }
+
+    def _append_score_log(self, path, tasks, ordered, header=None):
         """Append-only JSONL score log: one record
         per (candidate, fold) fit so a killed 100k-candidate sweep leaves
         its completed work on disk; ``resume_from_log`` replays it."""
         import json
+        import os as _os
 
+        if header is not None and (not _os.path.exists(path) or
+                                   _os.path.getsize(path) == 0):
+            with open(path, "a") as f:
+                f.write(json.dumps(header) + "\n")
         with open(path, "a") as f:
             for t, r in zip(tasks, ordered):
                 if r.error is not None:
@@ -621,17 +673,23 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                     "fit_time": r.fit_time,
                     "score_time": r.score_time}) + "\n")
 
-    def _load_resume_records(self, path, n_splits):
+    def _load_resume_records(self, path, n_splits, expected_header=None):
         """Completed (param-key, fold) -> TaskResult fields from a prior
         run's score log.  Records missing the resume fields (old-format
         logs) or train scores when this run needs them are skipped, so
-        those tasks simply re-run."""
+        those tasks simply re-run.  A header record written by
+        ``_append_score_log`` identifies the search (data fingerprint,
+        cv, estimator class, scoring); a mismatching log is refused
+        rather than silently replayed, and a header-less (old-format)
+        log only warns."""
         import json
         import os as _os
+        import warnings as _warnings
 
         done = {}
         if not path or not _os.path.exists(path):
             return done
+        saw_header = False
         with open(path) as f:
             for line in f:
                 line = line.strip()
@@ -641,6 +699,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                     rec = json.loads(line)
                 except ValueError:
                     continue      # partial last line of a killed run
+                if rec.get("header") == 1:
+                    saw_header = True
+                    if expected_header is not None:
+                        for field in ("data", "cv", "estimator", "scoring"):
+                            if rec.get(field) != expected_header[field]:
+                                raise ValueError(
+                                    "resume_from_log refused: the score "
+                                    f"log at {path!r} was written for a "
+                                    f"different search ({field} differs: "
+                                    f"log {rec.get(field)!r} vs current "
+                                    f"{expected_header[field]!r}). Delete "
+                                    "the log or point resume_from_log at "
+                                    "the matching one.")
+                    continue
                 if "key" not in rec or rec.get("fold") is None:
                     continue
                 if rec["fold"] >= n_splits:
@@ -649,6 +721,12 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                         is None:
                     continue
                 done[(rec["key"], rec["fold"])] = rec
+        if done and not saw_header:
+            _warnings.warn(
+                "resume_from_log: score log has no header record (written "
+                "by an older version); resuming without validating that "
+                "it matches this search's data/cv/estimator/scoring.",
+                UserWarning)
         return done
 
     # -- cv_results_ assembly ---------------------------------------------

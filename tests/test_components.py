@@ -391,3 +391,46 @@ def test_keyed_models_doctest():
 
     res = doctest.testmod(km)
     assert res.attempted >= 1 and res.failed == 0
+
+
+def test_resume_log_header_mismatch_refused(tmp_path):
+    """ADVICE round-1 (medium): a score log written against different
+    data / cv / estimator must be refused, not silently replayed."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd import HipContext
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    X1 = rng.standard_normal((200, 6))
+    y1 = (X1[:, 0] > 0).astype(np.int64)
+    log = str(tmp_path / "scores.jsonl")
+    ctx = HipContext(backend="sequential", score_log=log)
+    ska.GridSearchCV(ctx, LogisticRegression(max_iter=20),
+                     {"C": [0.1, 1.0]}, cv=2).fit(X1, y1)
+    # header record present
+    import json
+    first = json.loads(open(log).readline())
+    assert first.get("header") == 1 and "data" in first
+
+    # same search + same data resumes cleanly (everything restored)
+    ctx2 = HipContext(backend="sequential", resume_from_log=log)
+    gs2 = ska.GridSearchCV(ctx2, LogisticRegression(max_iter=20),
+                           {"C": [0.1, 1.0]}, cv=2)
+    gs2.fit(X1, y1)
+    assert gs2.n_resumed_ == 4
+
+    # different data -> refused
+    X2 = rng.standard_normal((200, 6))
+    y2 = (X2[:, 1] > 0).astype(np.int64)
+    ctx3 = HipContext(backend="sequential", resume_from_log=log)
+    gs3 = ska.GridSearchCV(ctx3, LogisticRegression(max_iter=20),
+                           {"C": [0.1, 1.0]}, cv=2)
+    with pytest.raises(ValueError, match="resume_from_log refused"):
+        gs3.fit(X2, y2)
+
+    # different cv -> refused
+    ctx4 = HipContext(backend="sequential", resume_from_log=log)
+    gs4 = ska.GridSearchCV(ctx4, LogisticRegression(max_iter=20),
+                           {"C": [0.1, 1.0]}, cv=3)
+    with pytest.raises(ValueError, match="resume_from_log refused"):
+        gs4.fit(X1, y1)
