@@ -51,6 +51,12 @@ def halving_fit(search, X, y=None, groups=None, **fit_params):
 
     candidates = list(search._get_param_iterator())
     full_budget = int(estimator.get_params().get("max_iter", 100))
+    # grids may sweep max_iter themselves; the rung budget must clamp the
+    # candidate value (not be overridden by it), and the final rung must
+    # reach the largest requested budget
+    grid_mi = [int(p["max_iter"]) for p in candidates if "max_iter" in p]
+    if grid_mi:
+        full_budget = max(full_budget, max(grid_mi))
     budget = max(1, int(full_budget * float(search.min_resource_frac)))
 
     backend = search._get_backend()
@@ -77,8 +83,12 @@ def halving_fit(search, X, y=None, groups=None, **fit_params):
         tasks: List[Task] = []
         idx = 0
         for slot, cand_i in enumerate(alive):
+            params = candidates[cand_i]
+            if "max_iter" in params:
+                params = dict(params,
+                              max_iter=min(int(params["max_iter"]), budget))
             for fold_i, (train, test) in enumerate(splits):
-                tasks.append(Task(idx, slot, fold_i, candidates[cand_i],
+                tasks.append(Task(idx, slot, fold_i, params,
                                   np.asarray(train), np.asarray(test)))
                 idx += 1
         results = backend.run_search(

@@ -63,3 +63,21 @@ def test_halving_validation(problem):
         halving_factor=1.0)
     with pytest.raises(ValueError, match="halving_factor"):
         rs.fit(X, y)
+
+
+def test_halving_clamps_grid_max_iter(problem):
+    """ADVICE round-1: a grid that sweeps max_iter must not override the
+    rung budget, and the schedule must reach the grid's largest value."""
+    X, y = problem
+    rs = ska.RandomizedSearchCV(
+        None, LogisticRegression(max_iter=10, tol=0.0),
+        {"C": loguniform(1e-3, 1e2), "max_iter": [5, 60]},
+        n_iter=12, cv=2, random_state=0, halving_factor=3,
+        min_resource_frac=0.1, return_train_score=False)
+    rs.fit(X, y)
+    res = rs.cv_results_
+    # the final rung reached the largest requested budget
+    assert res["resource_iter"].max() == 60
+    # early rungs ran at a fraction of it
+    assert res["resource_iter"].min() < 60
+    assert rs.best_score_ > 0.8
