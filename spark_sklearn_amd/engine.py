@@ -428,6 +428,14 @@ def _linear_device_search(estimator, executor, by_fold, n_classes, metric,
                 a.cand_index != b.cand_index
                 for a, b in zip(first, by_fold[f])):
             return None
+    # the whole-search plan encodes per-sample scales (1/(C*n_train)) from
+    # fold 0; for KFold-style splits fold sizes differ by at most one row
+    # (O(1/n) scale deviation, documented divergence), but splitters with
+    # materially unequal train folds fall back to the per-fold generic
+    # path, which rebuilds the plan per fold.
+    sizes = [len(by_fold[f][0].train) for f in folds]
+    if max(sizes) > min(sizes) * 1.001:
+        return None
 
     dbg = os.environ.get("SKA_TIME") == "1"
     if dbg:

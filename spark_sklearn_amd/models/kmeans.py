@@ -177,7 +177,15 @@ class KMeansFamily:
         mean_var = executor.feature_mean_var
         # centers are deterministic in (k, init, seed): cache them on the
         # executor so the 5 CV folds (and repeated chunks) don't re-run
-        # host-side k-means++ seeding
+        # host-side k-means++ seeding.
+        # DOCUMENTED DIVERGENCE (ADVICE round-1): the seeding sample (and
+        # the tol variance scale) is drawn from the FULL matrix, not the
+        # train fold, so during CV a few test-fold rows can influence the
+        # INITIAL centers a fold's score is computed from.  This mirrors
+        # the one-shared-binning divergence of the GBT family (bins from
+        # the full matrix); Lloyd iterations themselves see only train
+        # rows.  Exact per-fold seeding would defeat the cross-fold C0
+        # cache (the family's largest round-1 win, 8.8x).
         c0_cache = getattr(executor, "_c0_cache", None)
         if c0_cache is None:
             c0_cache = executor._c0_cache = {}
