@@ -28,8 +28,7 @@ setup(
     name="spark_sklearn_amd",
     version="0.1.0",
     packages=["spark_sklearn_amd", "spark_sklearn_amd.models",
-              "spark_sklearn_amd.ops", "spark_sklearn_amd.parallel",
-              "spark_sklearn_amd.utils"],
+              "spark_sklearn_amd.ops", "spark_sklearn_amd.parallel"],
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -73,11 +73,12 @@ def _pad_features(F: int) -> int:
 def _as_device_bf16(X, device):
     """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array.
 
-    scipy CSR matrices are ingested chunk-wise into a dense bf16 resident
-    matrix: 288 GB of HBM3E makes even a 10M x 1024 matrix (20 GB) cheap
-    to hold dense, and the MFMA fit kernels then run at dense speed.
-    (A dedicated LDS-staged CSR kernel path -- SURVEY.md K5 -- is the
-    planned upgrade for matrices that exceed dense residency.)"""
+    scipy CSR matrices reaching THIS dense-ingest path are materialized
+    chunk-wise into a dense bf16 resident matrix (288 GB of HBM3E makes
+    even a 10M x 1024 matrix cheap to hold dense).  NOTE: sparse
+    searches at <= 5% density normally run the CSR-RESIDENT K5 kernels
+    in ops/sparse_executor.py and never densify; this fallback covers
+    denser matrices and the multiclass OVR path."""
     import scipy.sparse as sp
     import torch
 

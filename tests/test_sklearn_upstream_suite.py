@@ -21,10 +21,11 @@ import pytest
 
 import spark_sklearn_amd as ska
 
-# Features this engine intentionally does not implement (documented in
-# README/BASELINE): multi-metric scoring dicts/lists, sklearn-internal
-# hooks, HTML repr internals, and tests whose wrapper subclass breaks
-# sklearn's get_params introspection (not an engine behavior).
+# Tests skipped for documented reasons.  (Multi-metric scoring IS
+# implemented -- dict/list/callable-returning-dict scoring with
+# per-metric cv_results_ keys -- and its upstream tests run; the skips
+# below are sklearn-internal hooks, HTML repr internals, and tests whose
+# wrapper subclass breaks sklearn's get_params introspection.)
 SKIP = {
     # compares our fitted attribute set against sklearn's internal
     # BaseSearchCV attribute-for-attribute: our deliberate observability
