@@ -137,6 +137,14 @@ def _coerce_score(v):
     return float(v)
 
 
+def return_train_score_effective(search, result) -> bool:
+    """Whether a per-fit verbose line should show the train score."""
+    if not search.return_train_score:
+        return False
+    ts = result.train_score
+    return not (isinstance(ts, float) and np.isnan(ts))
+
+
 def fit_and_score_task(base_estimator, X, y, scorer, task: Task,
                        fit_params=None, return_train_score=True,
                        error_score="raise") -> TaskResult:
@@ -449,6 +457,21 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
             )
         ordered = [by_index[i] for i in range(len(tasks))]
         self._handle_fit_failures(tasks, ordered)
+
+        if self.verbose > 2 and getattr(backend, "rank", 0) == 0:
+            # sklearn's per-fit verbose lines (single-metric form)
+            for t, r in zip(tasks, ordered):
+                if isinstance(r.test_score, dict):
+                    continue
+                pstr = ", ".join(f"{k}={v}" for k, v in
+                                 sorted(t.params.items()))
+                if return_train_score_effective(self, r):
+                    msg = (f"score=(train={r.train_score:.3f}, "
+                           f"test={r.test_score:.3f})")
+                else:
+                    msg = f"score={r.test_score:.3f}"
+                print(f"[CV {t.fold_index + 1}/{n_splits}] END {pstr};, "
+                      f"{msg} total time={r.fit_time + r.score_time:.3f}s")
 
         score_log = getattr(backend, "score_log", None)
         if score_log and getattr(backend, "rank", 0) == 0:

