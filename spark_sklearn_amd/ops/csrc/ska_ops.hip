@@ -1577,7 +1577,7 @@ __global__ void gbt_bin_kernel(const float* __restrict__ X,
 // flush using the exact count) -- integer atomics commute exactly, so
 // tree structures are bit-stable run to run.  gscale == 0 keeps the
 // default float-atomic g (faster: 12 B/cell vs 16 B).
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(512)
 void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      const int* __restrict__ idxmap,          // [nt]
                      const int* __restrict__ node_id,         // [nt]
@@ -1588,25 +1588,30 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                      long rows_pb, int even_only, float hscale,
                      float gscale) {
   extern __shared__ char smem[];
-  const int cellsg = lvl_nodes * f_pb * 256;
+  // under sibling subtraction only EVEN nodes are histogrammed: LDS
+  // covers lvl_nodes/2 node slots, doubling resident blocks at deep
+  // levels (round-1 PMC: 11.8% occupancy, WAIT-bound on LDS atomics)
+  const int lvl_eff = even_only ? (lvl_nodes >> 1) : lvl_nodes;
+  const int cellsg = lvl_eff * f_pb * 256;
   const bool gfix = gscale > 0.f;
   unsigned long long* lhc = reinterpret_cast<unsigned long long*>(smem);
   unsigned long long* lgq = lhc + cellsg;            // gfix mode only
   float* lg = reinterpret_cast<float*>(smem + (size_t)cellsg * 8);
   const int tid = threadIdx.x;
+  const int nthr = blockDim.x;
   const int f0 = blockIdx.y * f_pb;
   const long r0 = (long)blockIdx.x * rows_pb;
   const long r1 = min(r0 + rows_pb, nt);
 
   if (gfix) {
-    for (int i = tid; i < 2 * cellsg; i += 256) lhc[i] = 0ull;
+    for (int i = tid; i < 2 * cellsg; i += nthr) lhc[i] = 0ull;
   } else {
-    for (int i = tid; i < cellsg; i += 256) { lhc[i] = 0ull; lg[i] = 0.f; }
+    for (int i = tid; i < cellsg; i += nthr) { lhc[i] = 0ull; lg[i] = 0.f; }
   }
   __syncthreads();
 
   const bool vec8 = (F % 8 == 0) && (f_pb == 8);
-  for (long r = r0 + tid; r < r1; r += 256) {
+  for (long r = r0 + tid; r < r1; r += nthr) {
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
     if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
@@ -1617,7 +1622,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
                                         (double)gscale + 0.5);
     unsigned hfx = __float2uint_rn(h[r] * hscale);
     unsigned long long hpack = ((unsigned long long)hfx << 32) | 1ull;
-    const int base = (nd - lvl_lo) * f_pb;
+    const int base = ((nd - lvl_lo) >> even_only) * f_pb;
     const unsigned char* xrow = Xb + gi * F + f0;
     if (vec8) {
       unsigned long long bv =
@@ -1640,7 +1645,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
   }
   __syncthreads();
   const float hinv = 1.f / hscale;
-  for (int i = tid; i < cellsg; i += 256) {
+  for (int i = tid; i < cellsg; i += nthr) {
     unsigned long long p = lhc[i];
     if (p == 0ull) continue;          // count==0 -> cell never touched
     float hv = (float)(p >> 32) * hinv;
@@ -1652,7 +1657,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
       gv = lg[i];
     int bin = i & 255;
     int ff = (i >> 8) % f_pb;
-    int ndl = i / (256 * f_pb);
+    int ndl = (i / (256 * f_pb)) << even_only;   // back to full-level slot
     float* cell = &hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
     atomicAdd(cell, gv);
     atomicAdd(cell + 1, hv);
@@ -1690,10 +1695,11 @@ static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
                               hipStream_t s) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
+  int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
   bool gfix = g_bounded && det_gbt_mode();
   int cell_b = gfix ? 16 : 12;
-  int f_pb = std::min(8, std::max(1, (96 * 1024) /
-                                  (cell_b * 256 * lvl_nodes)));
+  int f_pb = std::min(8, std::max(1, (48 * 1024) /
+                                  (cell_b * 256 * lvl_eff)));
   while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
   int fchunks = F / f_pb;
   long rowsegs = std::max<long>(
@@ -1702,7 +1708,7 @@ static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
                            // adds stay single-sourced, so bit-stable
   long rows_pb = (nt + rowsegs - 1) / rowsegs;
   rowsegs = (nt + rows_pb - 1) / rows_pb;
-  size_t lsm = (size_t)lvl_nodes * f_pb * 256 * cell_b;
+  size_t lsm = (size_t)lvl_eff * f_pb * 256 * cell_b;
   int even_only = (d > 0) ? 1 : 0;       // sibling subtraction
   float hs = 1.f;
   while (hs * 2.f * hmax * (float)nt + (float)nt < 1.8e9f &&
@@ -1716,7 +1722,7 @@ static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
       reinterpret_cast<const void*>(gbt_hist_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
   hipLaunchKernelGGL(gbt_hist_kernel,
-                     dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(256),
+                     dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(512),
                      lsm, s, Xp, idxmap, node_id, g, h, hist, nt, F,
                      lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs, gs);
 }
