@@ -1578,15 +1578,15 @@ __global__ void gbt_bin_kernel(const float* __restrict__ X,
 // tree structures are bit-stable run to run.  gscale == 0 keeps the
 // default float-atomic g (faster: 12 B/cell vs 16 B).
 __global__ __launch_bounds__(512)
-void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
+void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      const int* __restrict__ idxmap,          // [nt]
                      const int* __restrict__ node_id,         // [nt]
                      const float* __restrict__ g,
                      const float* __restrict__ h,
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
-                     long nt, int F, int lvl_lo, int lvl_nodes, int f_pb,
-                     long rows_pb, int even_only, float hscale,
-                     float gscale) {
+                     long nt, long n_rows, int F, int lvl_lo,
+                     int lvl_nodes, int f_pb, long rows_pb, int even_only,
+                     float hscale, float gscale) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1623,20 +1623,22 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xb,   // [n][F]
     unsigned hfx = __float2uint_rn(h[r] * hscale);
     unsigned long long hpack = ((unsigned long long)hfx << 32) | 1ull;
     const int base = ((nd - lvl_lo) >> even_only) * f_pb;
-    const unsigned char* xrow = Xb + gi * F + f0;
+    // feature-major bins: each feature is a contiguous [n_rows] stream,
+    // so a wave's consecutive-ish gi values coalesce into full 64B
+    // lines per feature chunk (the row-major layout pulled a line per
+    // row PER FEATURE-CHUNK BLOCK -- F/f_pb-fold read amplification)
+    const unsigned char* xcol = Xbt + gi;
     if (vec8) {
-      unsigned long long bv =
-          *reinterpret_cast<const unsigned long long*>(xrow);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int cell = (base + j) * 256 + (int)((bv >> (8 * j)) & 0xff);
+        int cell = (base + j) * 256 + (int)xcol[(long)(f0 + j) * n_rows];
         if (gfix) atomicAdd(&lgq[cell], gq);
         else atomicAdd(&lg[cell], gr);
         atomicAdd(&lhc[cell], hpack);
       }
     } else {
       for (int ff = 0; ff < f_pb; ++ff) {
-        int cell = (base + ff) * 256 + (int)xrow[ff];
+        int cell = (base + ff) * 256 + (int)xcol[(long)(f0 + ff) * n_rows];
         if (gfix) atomicAdd(&lgq[cell], gq);
         else atomicAdd(&lg[cell], gr);
         atomicAdd(&lhc[cell], hpack);
@@ -1688,11 +1690,11 @@ static float det_fix_scale(long nt) {
   return gs;
 }
 
-static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
+static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                               const int* node_id, const float* g,
-                              const float* h, float* hist, long nt, int F,
-                              int d, float hmax, bool g_bounded,
-                              hipStream_t s) {
+                              const float* h, float* hist, long nt,
+                              long n_rows, int F, int d, float hmax,
+                              bool g_bounded, hipStream_t s) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
@@ -1723,8 +1725,9 @@ static void launch_hist_level(const unsigned char* Xp, const int* idxmap,
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
   hipLaunchKernelGGL(gbt_hist_kernel,
                      dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(512),
-                     lsm, s, Xp, idxmap, node_id, g, h, hist, nt, F,
-                     lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs, gs);
+                     lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
+                     F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs,
+                     gs);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -4685,7 +4688,8 @@ torch::Tensor gbt_bin(torch::Tensor X, torch::Tensor edges,
   return Xb;
 }
 
-std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
+std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
+                                   torch::Tensor idxmap,
                                    torch::Tensor y_tr, int64_t loss,
                                    int64_t n_estimators, double lrate,
                                    int64_t max_depth, int64_t min_leaf,
@@ -4718,6 +4722,10 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
 
   hipStream_t s = cur_stream();
   const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  TORCH_CHECK(Xbt.dtype() == torch::kUInt8 && Xbt.size(0) == Xb.size(1) &&
+              Xbt.size(1) == Xb.size(0) && Xbt.stride(1) == 1);
+  const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
+  long n_rows = Xb.size(0);
   int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
 
   for (int t = 0; t < T; ++t) {
@@ -4732,9 +4740,9 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor idxmap,
       int lvl_lo = (1 << d) - 1;
       int lvl_nodes = 1 << d;
       int even_only = (d > 0) ? 1 : 0;    // sibling subtraction
-      launch_hist_level(Xp, idxmap.data_ptr<int>(),
+      launch_hist_level(Xtp, idxmap.data_ptr<int>(),
                         node_id.data_ptr<int>(), g.data_ptr<float>(),
-                        h.data_ptr<float>(), hist.data_ptr<float>(), nt,
+                        h.data_ptr<float>(), hist.data_ptr<float>(), nt, n_rows,
                         F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f,
                         loss == LOSS_LOGISTIC, s);
       if (even_only) {
@@ -5211,13 +5219,16 @@ torch::Tensor csr_linear_scores(
 
 // multiclass boosting fit: trees [T*K][NN], class-major raw.
 std::vector<torch::Tensor> gbt_fit_multi(
-    torch::Tensor Xb, torch::Tensor idxmap, torch::Tensor y_tr,
+    torch::Tensor Xb, torch::Tensor Xbt, torch::Tensor idxmap,
+    torch::Tensor y_tr,
     torch::Tensor f0, int64_t K, int64_t n_estimators, double lrate,
     int64_t max_depth, int64_t min_leaf, double lam) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(max_depth >= 1 && max_depth <= 6);
   TORCH_CHECK(K >= 2 && K <= 64);
   int F = (int)Xb.size(1);
+  const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
+  long n_rows = Xb.size(0);
   long nt = idxmap.size(0);
   int T = (int)n_estimators;
   int NN = (1 << (max_depth + 1)) - 1;
@@ -5257,9 +5268,9 @@ std::vector<torch::Tensor> gbt_fit_multi(
         int lvl_lo = (1 << d) - 1;
         int lvl_nodes = 1 << d;
         int even_only = (d > 0) ? 1 : 0;
-        launch_hist_level(Xp, idxmap.data_ptr<int>(),
+        launch_hist_level(Xtp, idxmap.data_ptr<int>(),
                           node_id.data_ptr<int>(), gc, hc,
-                          hist.data_ptr<float>(), nt, F, d, 0.25f,
+                          hist.data_ptr<float>(), nt, n_rows, F, d, 0.25f,
                           /*g_bounded=*/true, s);
         if (even_only) {
           int prev_lo = (1 << (d - 1)) - 1;
@@ -5484,7 +5495,8 @@ void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
 // are class probabilities / means ALREADY divided by n_estimators so the
 // summing gbt_predict yields the forest average.
 // ---------------------------------------------------------------------------
-std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor idxmap0,
+std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
+                                  torch::Tensor idxmap0,
                                   torch::Tensor y_tr, int64_t K,
                                   int64_t n_estimators, int64_t max_depth,
                                   int64_t min_leaf, int64_t max_features,
@@ -5524,6 +5536,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor idxmap0,
 
   hipStream_t s = cur_stream();
   const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
+  long n_rows = Xb.size(0);
   int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
 
   for (int t = 0; t < T; ++t) {
@@ -5549,10 +5563,10 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor idxmap0,
       for (int c = 0; c < KP; ++c) {
         const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
                                    : y_boot.data_ptr<float>();
-        launch_hist_level(Xp, idx_t.data_ptr<int>(),
+        launch_hist_level(Xtp, idx_t.data_ptr<int>(),
                           node_id.data_ptr<int>(), gp,
                           ones.data_ptr<float>(),
-                          hist.data_ptr<float>() + (long)c * ccells, nt, F,
+                          hist.data_ptr<float>() + (long)c * ccells, nt, n_rows, F,
                           d, 1.f, false, s);
         if (d > 0) {
           int prev_lo = (1 << (d - 1)) - 1;

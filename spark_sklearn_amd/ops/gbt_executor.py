@@ -58,12 +58,16 @@ class DeviceGBTExecutor:
                 X_dev, torch.as_tensor(eflat).to(device),
                 torch.as_tensor(ecnt).to(device))
             del X_dev
+            # feature-major copy for the histogram kernel (coalesced
+            # per-feature streams; the row-major layout read-amplified
+            # X by F/f_pb per level)
+            Xbt_t = Xb_t.t().contiguous()
             if len(_bin_cache) > 2:
                 _bin_cache.clear()
             # the entry HOLDS X: id() alone is unsound after GC recycling
-            _bin_cache[key] = (X, (Xb_t, edges))
+            _bin_cache[key] = (X, (Xb_t, Xbt_t, edges))
             hit = _bin_cache[key]
-        self.Xb, self.edges = hit[1]
+        self.Xb, self.Xbt, self.edges = hit[1]
         self.F = self.Xb.shape[1]
         self.y = torch.as_tensor(
             np.asarray(y_enc, dtype=np.float32)).to(device)
@@ -99,7 +103,8 @@ class DeviceGBTExecutor:
                 f0 = np.log(np.clip(prior, 1e-6, None)).astype(np.float32)
                 f0_t = torch.as_tensor(f0).to(self.device)
                 feats, thrs, vals = self.ext.gbt_fit_multi(
-                    self.Xb, idx_t, y_tr, f0_t, K, spec.n_estimators,
+                    self.Xb, self.Xbt, idx_t, y_tr, f0_t, K,
+                    spec.n_estimators,
                     spec.learning_rate, spec.max_depth,
                     spec.min_samples_leaf, spec.l2_regularization)
                 models_dev.append((feats, thrs, vals, f0, spec))
@@ -114,7 +119,8 @@ class DeviceGBTExecutor:
                 continue
             f0 = self._f0(spec.loss, y_tr_np)
             feats, thrs, vals = self.ext.gbt_fit(
-                self.Xb, idx_t, y_tr, spec.loss, spec.n_estimators,
+                self.Xb, self.Xbt, idx_t, y_tr, spec.loss,
+                spec.n_estimators,
                 spec.learning_rate, spec.max_depth, spec.min_samples_leaf,
                 spec.l2_regularization, f0)
             models_dev.append((feats, thrs, vals, f0, spec.loss))

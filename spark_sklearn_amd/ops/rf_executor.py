@@ -34,7 +34,8 @@ class DeviceRFExecutor(DeviceGBTExecutor):
                 n_iter[i] = 0
                 continue
             feats, thrs, vals = self.ext.rf_fit(
-                self.Xb, idx_t, y_tr, K, spec.n_estimators, spec.max_depth,
+                self.Xb, self.Xbt, idx_t, y_tr, K, spec.n_estimators,
+                spec.max_depth,
                 spec.min_samples_leaf, spec.max_features,
                 spec.random_state, 1 if spec.bootstrap else 0)
             T = spec.n_estimators
