@@ -5543,15 +5543,23 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
   for (int t = 0; t < T; ++t) {
     unsigned st = rf_hash((unsigned)seed ^ (unsigned)(t * 0x9e3779b9u));
     unsigned fs = rf_hash(st ^ 0x85ebca6bu);
+    torch::Tensor idx_u = idx_t, y_u = y_boot;
     if (bootstrap) {
       hipLaunchKernelGGL(rf_bootstrap_kernel, dim3(blocks), dim3(256), 0, s,
                          idxmap0.data_ptr<int>(), y_tr.data_ptr<float>(),
                          idx_t.data_ptr<int>(), y_boot.data_ptr<float>(),
                          nt, st);
+      // sort the resample by row id: histogram/sum kernels are order-
+      // invariant, and sorted gathers coalesce the feature-major bin
+      // reads (a random-order resample made every 1-byte load its own
+      // 64 B line)
+      auto srt = torch::sort(idx_t);
+      idx_u = std::get<0>(srt).contiguous();
+      y_u = y_boot.index_select(0, std::get<1>(srt)).contiguous();
     }
     if (KP > 1)
       hipLaunchKernelGGL(rf_planes_kernel, dim3(blocks), dim3(256), 0, s,
-                         y_boot.data_ptr<float>(), G.data_ptr<float>(), nt,
+                         y_u.data_ptr<float>(), G.data_ptr<float>(), nt,
                          KP);
     HIP_CHECK(hipMemsetAsync(node_id.data_ptr(), 0, nt * 4, s));
     int* feat_t = feats.data_ptr<int>() + (long)t * KP * NN;
@@ -5562,8 +5570,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
       int lvl_nodes = 1 << d;
       for (int c = 0; c < KP; ++c) {
         const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
-                                   : y_boot.data_ptr<float>();
-        launch_hist_level(Xtp, idx_t.data_ptr<int>(),
+                                   : y_u.data_ptr<float>();
+        launch_hist_level(Xtp, idx_u.data_ptr<int>(),
                           node_id.data_ptr<int>(), gp,
                           ones.data_ptr<float>(),
                           hist.data_ptr<float>() + (long)c * ccells, nt, n_rows, F,
@@ -5599,7 +5607,7 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                      /*non_blocking=*/true);
       }
       hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
-                         0, s, Xp, idx_t.data_ptr<int>(),
+                         0, s, Xp, idx_u.data_ptr<int>(),
                          node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
                          lvl_lo, lvl_lo + lvl_nodes);
       HIP_CHECK(hipGetLastError());
@@ -5607,7 +5615,7 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
     int bl_lo = (1 << max_depth) - 1;
     for (int c = 0; c < KP; ++c) {
       const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
-                                 : y_boot.data_ptr<float>();
+                                 : y_u.data_ptr<float>();
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
       hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0,

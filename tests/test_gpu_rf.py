@@ -29,10 +29,11 @@ def _ext():
 
 def _dev_trees(ext, dev, Xb_np, y_np, spec, K=0):
     Xb = torch.as_tensor(Xb_np).to(dev)
+    Xbt = Xb.t().contiguous()
     idx = torch.arange(Xb_np.shape[0], dtype=torch.int32, device=dev)
     y = torch.as_tensor(y_np.astype(np.float32)).to(dev)
     feats, thrs, vals = ext.rf_fit(
-        Xb, idx, y, K, spec.n_estimators, spec.max_depth,
+        Xb, Xbt, idx, y, K, spec.n_estimators, spec.max_depth,
         spec.min_samples_leaf, spec.max_features, spec.random_state,
         1 if spec.bootstrap else 0)
     return (feats.cpu().numpy(), thrs.cpu().numpy().astype(np.uint8),
