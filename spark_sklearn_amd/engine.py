@@ -103,7 +103,7 @@ class LinearFamily:
     def _new_device_executor(X, y_enc, classes, device):
         import scipy.sparse as sp
 
-        if sp.issparse(X) and (classes is None or len(classes) <= 2):
+        if sp.issparse(X):
             from .ops.sparse_executor import (CSR_DENSITY_LIMIT,
                                               SparseLinearExecutor)
 

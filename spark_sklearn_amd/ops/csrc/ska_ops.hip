@@ -3961,6 +3961,71 @@ void csr_scores_kernel(const int* __restrict__ indptr,
   }
 }
 
+// multiclass OVR scores over CSR rows: per candidate (= C adjacent
+// columns) argmax for accuracy / normalized OVR log-likelihood.  A wave
+// stages one row's Bp z-values in LDS (wave-lockstep, no barrier), then
+// lanes reduce candidate groups.
+__global__ __launch_bounds__(256)
+void csr_scores_multi_kernel(const int* __restrict__ indptr,
+                             const int* __restrict__ cols,
+                             const __bf16* __restrict__ vals,
+                             const float* __restrict__ y,
+                             const long* __restrict__ idx,
+                             const __bf16* __restrict__ Wf,
+                             const float* __restrict__ bias,
+                             int C, int metric, long n_idx, int Bp,
+                             int n_cand, double* __restrict__ out_sum) {
+  __shared__ float zrow[4][256];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long gwave = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const long nwaves = ((long)gridDim.x * blockDim.x) >> 6;
+  const int cpl = Bp >> 6;
+  const float eps = 1e-12f;
+
+  float accm[2] = {0.f, 0.f};
+  for (long i = gwave; i < n_idx; i += nwaves) {
+    long r = idx[i];
+    float z[4] = {0.f, 0.f, 0.f, 0.f};
+    int k1 = indptr[r + 1];
+    for (int k = indptr[r]; k < k1; ++k) {
+      float v = (float)vals[k];
+      const __bf16* wrow = Wf + (long)cols[k] * Bp + lane * cpl;
+      for (int j = 0; j < cpl; ++j) z[j] += v * (float)wrow[j];
+    }
+    for (int j = 0; j < cpl; ++j) {
+      int c = lane * cpl + j;
+      zrow[wave][c] = z[j] + bias[c];
+    }
+    float yv = y[r];
+    for (int q = 0; q < 2; ++q) {
+      int cd = lane + (q << 6);
+      if (cd >= n_cand) continue;
+      const float* zr = &zrow[wave][cd * C];
+      if (metric == 0) {
+        int best = 0;
+        float bz = zr[0];
+        for (int c = 1; c < C; ++c)
+          if (zr[c] > bz) { bz = zr[c]; best = c; }
+        accm[q] += (best == (int)yv) ? 1.f : 0.f;
+      } else {
+        float ssum = 0.f, sy = 0.f;
+        for (int c = 0; c < C; ++c) {
+          float s2 = 1.f / (1.f + __expf(-zr[c]));
+          ssum += s2;
+          if (c == (int)yv) sy = s2;
+        }
+        accm[q] += __logf(sy / fmaxf(ssum, eps) + eps);
+      }
+    }
+  }
+  for (int q = 0; q < 2; ++q) {
+    int cd = lane + (q << 6);
+    if (cd < n_cand && accm[q] != 0.f)
+      atomicAdd(&out_sum[cd], (double)accm[q]);
+  }
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -5191,12 +5256,24 @@ torch::Tensor csr_linear_scores(
     torch::Tensor indptr, torch::Tensor ccols, torch::Tensor vals,
     torch::Tensor y, torch::Tensor idx, torch::Tensor Wf,
     torch::Tensor bias, torch::Tensor loss_id, torch::Tensor target,
-    int64_t metric, int64_t n_cand) {
+    int64_t metric, int64_t n_cand, int64_t C) {
   const at::cuda::CUDAGuard guard(vals.device());
   int Bp = (int)Wf.size(1);
   long n_idx = idx.size(0);
   auto out = torch::zeros({n_cand}, y.options().dtype(torch::kFloat64));
   int blocks = (int)std::min<long>((n_idx + 3) / 4, 4096);
+  if (C > 1) {   // multiclass OVR: per-candidate argmax over C columns
+    hipLaunchKernelGGL(csr_scores_multi_kernel, dim3(blocks), dim3(256), 0,
+                       cur_stream(), indptr.data_ptr<int>(),
+                       ccols.data_ptr<int>(),
+                       reinterpret_cast<const __bf16*>(vals.data_ptr()),
+                       y.data_ptr<float>(), idx.data_ptr<long>(),
+                       reinterpret_cast<const __bf16*>(Wf.data_ptr()),
+                       bias.data_ptr<float>(), (int)C, (int)metric, n_idx,
+                       Bp, (int)n_cand, out.data_ptr<double>());
+    HIP_CHECK(hipGetLastError());
+    return out;
+  }
   auto go = [&](auto k) {
     hipLaunchKernelGGL(k, dim3(blocks), dim3(256), 0, cur_stream(),
                        indptr.data_ptr<int>(), ccols.data_ptr<int>(),

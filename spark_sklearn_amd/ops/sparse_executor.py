@@ -3,7 +3,7 @@
 Used when the input is scipy-sparse and low-density: the matrix stays in
 HBM as CSR (indptr/indices/data) plus a (row-chunk, feature, row)-sorted
 transpose entry list for the backward pass -- never densified.  Binary /
-regression targets only (multiclass OVR sparse searches densify instead;
+regression and multiclass-OVR targets (accuracy / neg_log_loss;
 the per-row argmax would span lanes).
 """
 
@@ -34,9 +34,8 @@ class SparseLinearExecutor:
         self.ext = load_extension()
         if ensure_layout(device) != 3:
             raise RuntimeError("CSR kernels require the probed layout P=3")
-        if classes is not None and len(classes) > 2:
-            raise RuntimeError(
-                "sparse executor supports binary/regression targets")
+        self.C = len(classes) if (classes is not None and
+                                  len(classes) > 2) else 1
 
         key = (id(X), X.shape, str(device))
         hit = _sparse_cache.get(key)
@@ -159,13 +158,14 @@ class SparseLinearExecutor:
             raise RuntimeError("scores() requires a preceding fit()")
         Wf, bias, loss_t, tgt_t = self._fit_state
         n_cand = int(plan.col_to_cand.max()) + 1
-        if plan.n_classes != 1:
-            raise RuntimeError("sparse scoring is binary/regression only")
+        C = plan.n_classes
+        if C != 1 and metric == "r2":
+            raise RuntimeError("r2 scoring is single-output only")
         idx_t = torch.as_tensor(
             np.asarray(idx, dtype=np.int64)).to(self.device)
         sums = self.ext.csr_linear_scores(
             self.indptr, self.cols, self.vals, self.y, idx_t, Wf, bias,
-            loss_t, tgt_t, _METRIC_ID[metric], n_cand).cpu().numpy()
+            loss_t, tgt_t, _METRIC_ID[metric], n_cand, C).cpu().numpy()
         n = len(idx)
         if metric in ("accuracy", "neg_log_loss"):
             return sums / n

@@ -539,3 +539,32 @@ def test_cuda_tensor_input_matches_numpy(problem):
     assert a.keys() == b.keys()
     for k in a:
         assert abs(a[k] - b[k]) < 1e-12, (k, a[k], b[k])
+
+
+def test_csr_multiclass_ovr_matches_dense(dev):
+    """CSR-resident OVR multiclass (round-2): same scores as the dense
+    executor on the same data."""
+    import scipy.sparse as sp
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    n, f, K = 40_000, 256, 4
+    X = sp.random(n, f, density=0.02, format="csr", random_state=1,
+                  dtype=np.float32)
+    w = rng.standard_normal((f, K)).astype(np.float32)
+    y = np.asarray((X @ w) + 0.1 * rng.standard_normal((n, K)),
+                   dtype=np.float32).argmax(axis=1)
+
+    grid = {"C": [0.1, 1.0, 10.0]}
+    import spark_sklearn_amd as ska
+
+    gs_sp = ska.GridSearchCV(None, LogisticRegression(max_iter=25, tol=0.0),
+                             grid, cv=2)
+    gs_sp.fit(X, y)
+    gs_d = ska.GridSearchCV(None, LogisticRegression(max_iter=25, tol=0.0),
+                            grid, cv=2)
+    gs_d.fit(np.asarray(X.todense()), y)
+    np.testing.assert_allclose(gs_sp.cv_results_["mean_test_score"],
+                               gs_d.cv_results_["mean_test_score"],
+                               atol=5e-3)
+    assert gs_sp.best_score_ > 0.6
