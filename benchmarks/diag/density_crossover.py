@@ -5,7 +5,11 @@ dense-ingest executor at several densities; prints a JSON line per cell.
 GPU box:  python benchmarks/diag/density_crossover.py
 """
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 
 import numpy as np
 import scipy.sparse as sp
