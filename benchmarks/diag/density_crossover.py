@@ -19,8 +19,10 @@ from spark_sklearn_amd.models import LogisticRegression
 
 
 def run(X, y, label):
+    # 64 candidates: amortizes one-time prep (CSR transpose argsort /
+    # dense ingest) so the cell measures steady-state per-epoch cost
     gs = ska.GridSearchCV(None, LogisticRegression(max_iter=15, tol=0.0),
-                          {"C": [0.03, 0.3, 3.0, 30.0]}, cv=3)
+                          {"C": list(np.logspace(-3, 3, 64))}, cv=3)
     t0 = time.perf_counter()
     gs.fit(X, y)
     dt = time.perf_counter() - t0
@@ -31,7 +33,7 @@ def run(X, y, label):
 def main():
     rng = np.random.default_rng(0)
     n, f = 2_000_000, 512
-    for dens in (0.005, 0.01, 0.02, 0.05, 0.10, 0.20):
+    for dens in (0.005, 0.01, 0.02, 0.05, 0.10):
         Xs = sp.random(n, f, density=dens, format="csr", random_state=1,
                        dtype=np.float32)
         w = rng.standard_normal(f).astype(np.float32)
