@@ -141,17 +141,29 @@ def bench_sparse():
     w = rng.standard_normal(f).astype(np.float32)
     y = (np.asarray(X @ w).ravel() > 0).astype(np.int64)
 
-    gs = ska.GridSearchCV(
-        None, SGDClassifier(max_iter=15, tol=0.0),
-        {"alpha": [1e-6, 1e-5, 1e-4, 1e-3],
-         "loss": ["hinge", "log_loss"]}, cv=3,
-        return_train_score=False)
-    t0 = time.perf_counter()
-    gs.fit(X, y)
-    dt = time.perf_counter() - t0
+    def one(force_csr):
+        import spark_sklearn_amd.ops.sparse_executor as se
+
+        old = se.CSR_MIN_DENSE_BYTES
+        se.CSR_MIN_DENSE_BYTES = 0 if force_csr else float("inf")
+        try:
+            gs = ska.GridSearchCV(
+                None, SGDClassifier(max_iter=15, tol=0.0),
+                {"alpha": [1e-6, 1e-5, 1e-4, 1e-3],
+                 "loss": ["hinge", "log_loss"]}, cv=3,
+                return_train_score=False)
+            t0 = time.perf_counter()
+            gs.fit(X, y)
+            return gs, time.perf_counter() - t0
+        finally:
+            se.CSR_MIN_DENSE_BYTES = old
+
+    gs, dt = one(force_csr=True)            # the K5 CSR-resident kernels
+    _, dt_dense = one(force_csr=False)      # dense-ingest comparison point
     return {"family": "sparse-csr", "config": "10Mx1024 CSR d=1% (100M "
             "nnz), mixed hinge/log grid, cv=3, 15 epochs",
             "n_fits": gs.n_fits_, "wall_s": round(dt, 2),
+            "dense_ingest_wall_s": round(dt_dense, 2),
             "fits_per_sec": round(gs.n_fits_ / dt, 2),
             "best": gs.best_params_, "best_acc": round(gs.best_score_, 4)}
 

@@ -104,13 +104,18 @@ class LinearFamily:
         import scipy.sparse as sp
 
         if sp.issparse(X):
-            from .ops.sparse_executor import (CSR_DENSITY_LIMIT,
-                                              SparseLinearExecutor)
+            from .ops import sparse_executor as se
 
             density = X.nnz / max(1, X.shape[0] * X.shape[1])
-            if density <= CSR_DENSITY_LIMIT:
-                # CSR-resident K5 kernels: never densified
-                return SparseLinearExecutor(X, y_enc, classes, device)
+            # dense X + transpose, bf16
+            dense_bytes = 2.0 * X.shape[0] * X.shape[1] * 2
+            if density <= se.CSR_DENSITY_LIMIT and \
+                    dense_bytes > se.CSR_MIN_DENSE_BYTES:
+                # CSR-resident K5 kernels: never densified.  Chosen only
+                # when the dense form would strain HBM -- the measured
+                # crossover (profiles/r02_csr_crossover.md) shows dense-
+                # ingest is faster whenever it fits.
+                return se.SparseLinearExecutor(X, y_enc, classes, device)
         from .ops import DeviceLinearExecutor
 
         return DeviceLinearExecutor(X, y_enc, classes, device)

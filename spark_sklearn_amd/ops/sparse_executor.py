@@ -17,7 +17,15 @@ from .device_executor import _METRIC_ID, _roundup, ensure_layout
 
 __all__ = ["SparseLinearExecutor", "CSR_DENSITY_LIMIT"]
 
-CSR_DENSITY_LIMIT = 0.05      # auto-pick the CSR path at or below this
+CSR_DENSITY_LIMIT = 0.05      # CSR candidate at or below this density
+# Round-2 measurement (profiles/r02_csr_crossover.md): the dense-ingest
+# path beat the CSR-resident kernels at EVERY tested (shape, density) --
+# 2M x 512 at 0.5-10% and the headline 10M x 1024 at 1% (2.16 s vs
+# 4.99 s) -- because the batched bf16 MFMA kernels amortize the dense
+# matrix over 256 candidate columns.  The CSR path's remaining role is
+# CAPACITY: matrices whose dense bf16 form (plus transpose) would not
+# fit comfortably in 288 GB HBM.
+CSR_MIN_DENSE_BYTES = 64e9    # use CSR only above this dense footprint
 BWD_CHUNK_ROWS = 4096         # D rows touched by a chunk stay L2-resident
 
 _sparse_cache: dict = {}

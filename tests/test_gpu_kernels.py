@@ -350,14 +350,18 @@ def test_csr_resident_kernels_match_ref(dev):
     np.testing.assert_allclose(s_d, s_r, atol=5e-3)
 
 
-def test_csr_grid_search_end_to_end(dev):
+def test_csr_grid_search_end_to_end(dev, monkeypatch):
     """Low-density sparse input routes through the CSR kernels inside a
-    full GridSearchCV."""
+    full GridSearchCV (capacity gate forced open: the default now picks
+    dense-ingest whenever the dense form fits -- see
+    profiles/r02_csr_crossover.md)."""
     import scipy.sparse as sp
 
     import spark_sklearn_amd as ska
+    import spark_sklearn_amd.ops.sparse_executor as se
     from spark_sklearn_amd.models import LogisticRegression
 
+    monkeypatch.setattr(se, "CSR_MIN_DENSE_BYTES", 0)
     rng = np.random.default_rng(6)
     n, f = 30000, 256
     mask = rng.random((n, f)) < 0.02
@@ -541,12 +545,14 @@ def test_cuda_tensor_input_matches_numpy(problem):
         assert abs(a[k] - b[k]) < 1e-12, (k, a[k], b[k])
 
 
-def test_csr_multiclass_ovr_matches_dense(dev):
+def test_csr_multiclass_ovr_matches_dense(dev, monkeypatch):
     """CSR-resident OVR multiclass (round-2): same scores as the dense
-    executor on the same data."""
+    executor on the same data (CSR gate forced open)."""
     import scipy.sparse as sp
+    import spark_sklearn_amd.ops.sparse_executor as se
     from spark_sklearn_amd.models import LogisticRegression
 
+    monkeypatch.setattr(se, "CSR_MIN_DENSE_BYTES", 0)
     rng = np.random.default_rng(0)
     n, f, K = 40_000, 256, 4
     X = sp.random(n, f, density=0.02, format="csr", random_state=1,
