@@ -3404,6 +3404,243 @@ void bwd_g15_kernel(const __bf16* __restrict__ Xt, const int* __restrict__ idx,
 }
 
 // ---------------------------------------------------------------------------
+// fused_fb18_kernel: fused_fb14 with double-buffered async staging -- the
+// v14 measurement showed the fused epoch is correct and moves ~45% less
+// HBM but loses 1.4x to serialized stage->drain phases; v18 overlaps
+// chunk k+1's loads with chunk k's MFMA (fwd) and the first Xt chunk
+// with the loss epilogue (bwd), keeping the traffic win without the
+// drain serialization.
+// ---------------------------------------------------------------------------
+template <int NFW>
+__global__ __launch_bounds__(1024)
+void fused_fb18_kernel(const __bf16* __restrict__ X,
+                       const __bf16* __restrict__ Xt,
+                       const float* __restrict__ y,
+                       const int* __restrict__ idx,
+                       const __bf16* __restrict__ Wt,
+                       const float* __restrict__ bias,
+                       const int* __restrict__ loss_id,
+                       const int* __restrict__ target,
+                       float* __restrict__ Gt,
+                       float* __restrict__ loss_sum,
+                       float* __restrict__ db_sum,
+                       const int* __restrict__ row_key,
+                       const int* __restrict__ col_key,
+                       int nt, int NTp, int Fp, int Bp, int want_loss,
+                       int S, long n_rows) {
+  __shared__ __bf16 sD[256 * 64];         // D^T tile [cand][row], 32 KB
+  __shared__ __bf16 sA[2][64 * 64];       // X chunk dbuf, 2 x 8 KB
+  __shared__ __bf16 sW[2][256 * 64];      // W chunk dbuf / Xt chunks
+  __shared__ float sLoss[256];
+  __shared__ float sDb[256];
+  __shared__ int sIdx[64];
+  __shared__ float sY[64];
+  __shared__ int sKey[64];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;
+  const int cquad = wave >> 2;
+  const int fwave = wave & 3;
+  const int g = lane >> 4;
+
+  if (tid < 256) {
+    sLoss[tid] = 0.f;
+    sDb[tid] = 0.f;
+  }
+
+  f32x4 acc_g[4][NFW];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NFW; ++j) acc_g[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const long row0 = (long)blockIdx.x * S * 64;
+  const int nsub = (int)min((long)S, (NTp - row0 + 63) / 64);
+  const int nkc = Fp >> 6;
+
+  auto stage_fwd = [&](int buf, int kc) {
+    if (tid < 512) {
+      int r = tid >> 3, sl = tid & 7;
+      gload_lds16(X + (long)sIdx[r] * Fp + kc + ((sl ^ (r & 7)) << 3),
+                  &sA[buf][wave * 8 * 64]);
+    }
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int l = tid + (q << 10);
+      int c = l >> 3, sl = l & 7;
+      gload_lds16(Wt + (long)c * Fp + kc + ((sl ^ (c & 7)) << 3),
+                  &sW[buf][(q * 128 + wave * 8) * 64]);
+    }
+  };
+  auto stage_xt = [&](int buf, int rbase, int rc) {
+    int i0 = sIdx[rc];
+    bool cont = (sIdx[rc + 31] - i0 == 31) &&
+                (((i0 | (int)n_rows) & 7) == 0);
+    if (cont) {
+      int fr = wave * 16 + (lane >> 2), sl = lane & 3;
+      if (fr < NFW * 64)
+        gload_lds16(Xt + (long)fr * n_rows + i0 + ((sl ^ swz32(fr)) << 3),
+                    &sW[buf][(wave * 16) * 32]);
+    } else {
+      int fr = tid >> 2, sl = tid & 3;
+      if (fr < NFW * 64) {
+        int kb = rc + ((sl ^ swz32(fr)) << 3);
+        const __bf16* src = Xt + (long)fr * n_rows;
+        bf16x8 v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] = src[idx[rbase + kb + j]];
+        *reinterpret_cast<bf16x8*>(&sW[buf][fr * 32 + (sl << 3)]) = v;
+      }
+    }
+  };
+
+  for (int s = 0; s < nsub; ++s) {
+    const int rbase = (int)row0 + s * 64;
+    __syncthreads();
+    if (tid < 64) {
+      int gi = idx[rbase + tid];
+      sIdx[tid] = gi;
+      sY[tid] = y[gi];
+      sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    }
+    __syncthreads();
+
+    // ---- fwd GEMM, chunk k+1 staged during chunk k's MFMA -------------
+    f32x4 acc_f[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc_f[j] = {0.f, 0.f, 0.f, 0.f};
+
+    stage_fwd(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    for (int kc = 0; kc < nkc; ++kc) {
+      int b = kc & 1;
+      if (kc + 1 < nkc) stage_fwd(b ^ 1, (kc + 1) << 6);
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        int r = rquad * 16 + (lane & 15);
+        bf16x8 a = frag12(sA[b], r, h * 4 + g);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int c = cwave * 64 + fn * 16 + (lane & 15);
+          bf16x8 bb = frag12(sW[b], c, h * 4 + g);
+          acc_f[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bb, acc_f[fn], 0, 0, 0);
+        }
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
+
+    // first Xt chunk flies during the loss epilogue (reuses the W buffer
+    // the last fwd chunk did NOT occupy)
+    const int xb0 = nkc & 1;
+    stage_xt(xb0, rbase, 0);
+
+    // ---- epilogue: d values -> sD + loss/db ----------------------------
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      int col_in = cwave * 64 + fn * 16 + (lane & 15);
+      float bia = bias[col_in];
+      int lid = loss_id[col_in];
+      int tgt = target[col_in];
+      int ck = (col_key != nullptr) ? col_key[col_in] : 0;
+      float lsum = 0.f, dsum = 0.f;
+      int row_base = rquad * 16 + ((lane >> 4) << 2);
+      bf16x4 dpack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gr = rbase + row_base + r;
+        float z = acc_f[fn][r] + bia;
+        float d, ls = 0.f;
+        if (want_loss) {
+          loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
+        } else {
+          float yv = sY[row_base + r];
+          if (lid == LOSS_LOGISTIC) {
+            float t = (yv == (float)tgt) ? 1.f : 0.f;
+            d = fast_rcp(1.f + __expf(-z)) - t;
+          } else if (lid == LOSS_HINGE) {
+            float sg = (yv == (float)tgt) ? 1.f : -1.f;
+            d = (sg * z < 1.f) ? -sg : 0.f;
+          } else {
+            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
+            d = z - t;
+          }
+        }
+        if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (row_key != nullptr && sKey[row_base + r] != ck) {
+          d = 0.f;
+          ls = 0.f;
+        }
+        dpack[r] = (__bf16)d;
+        lsum += ls;
+        dsum += d;
+      }
+      {
+        int j = row_base >> 3, half = (row_base >> 2) & 1;
+        *reinterpret_cast<bf16x4*>(
+            &sD[col_in * 64 + ((j ^ (col_in & 7)) << 3) + (half << 2)]) =
+            dpack;
+      }
+      if (want_loss) atomicAdd(&sLoss[col_in], lsum);
+      atomicAdd(&sDb[col_in], dsum);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    // ---- bwd GEMM: second Xt chunk staged during the first's MFMA ------
+    for (int rc = 0; rc < 64; rc += 32) {
+      int xb = xb0 ^ ((rc >> 5) & 1);
+      if (rc + 32 < 64) stage_xt(xb ^ 1, rbase, rc + 32);
+      constexpr int FH = NFW > 2 ? 2 : NFW;
+#pragma unroll
+      for (int fh = 0; fh < NFW / FH; ++fh) {
+        bf16x8 bfr[FH];
+#pragma unroll
+        for (int fj = 0; fj < FH; ++fj) {
+          int f = fwave * (16 * NFW) + (fh * FH + fj) * 16 + (lane & 15);
+          bfr[fj] = frag13(sW[xb], f, g);
+        }
+#pragma unroll
+        for (int mm = 0; mm < 4; ++mm) {
+          int c = cquad * 64 + mm * 16 + (lane & 15);
+          bf16x8 a = frag12(sD, c, (rc >> 3) + g);
+#pragma unroll
+          for (int fj = 0; fj < FH; ++fj)
+            acc_g[mm][fh * FH + fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, bfr[fj], acc_g[mm][fh * FH + fj], 0, 0, 0);
+        }
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
+  }
+
+  // ---- block end: flush G, loss, db ------------------------------------
+#pragma unroll
+  for (int mm = 0; mm < 4; ++mm) {
+#pragma unroll
+    for (int fn = 0; fn < NFW; ++fn) {
+      int f_in = fwave * (16 * NFW) + fn * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c_in = cquad * 64 + mm * 16 + ((lane >> 4) << 2) + r;
+        atomicAdd(&Gt[(long)c_in * Fp + f_in], acc_g[mm][fn][r]);
+      }
+    }
+  }
+  __syncthreads();
+  if (tid < 256) {
+    if (want_loss && sLoss[tid] != 0.f) atomicAdd(&loss_sum[tid], sLoss[tid]);
+    atomicAdd(&db_sum[tid], sDb[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // scores3_kernel: permuted-LDS (P=3) port of scores_kernel -- removes the
 // 22M bank conflicts per dispatch of the generic [72]-stride tiles.
 // ---------------------------------------------------------------------------
@@ -4216,7 +4453,8 @@ std::vector<torch::Tensor> linear_fit(
   auto V = torch::zeros({Bp, Fp}, optF);
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::zeros({Bp, Fp}, X.options());
-  bool use_fused14 = flagship_v() == 14 && Bp == 256 && NTp % 256 == 0 &&
+  bool use_fused14 = (flagship_v() == 14 || flagship_v() >= 18) &&
+                     Bp == 256 && NTp % 256 == 0 &&
                      (Fp == 64 || Fp == 128 || Fp == 256);
   auto Dt = torch::zeros({(long)Bp, use_fused14 ? 1L : NTp}, X.options());
   auto bias = torch::zeros({Bp}, optF);
@@ -4298,9 +4536,13 @@ std::vector<torch::Tensor> linear_fit(
         long subt = NTp / 64;
         long Ssub = (subt + 255) / 256;
         unsigned nblk = (unsigned)((subt + Ssub - 1) / Ssub);
-        auto* fk = Fp == 256   ? fused_fb14_kernel<4>
-                   : Fp == 128 ? fused_fb14_kernel<2>
-                               : fused_fb14_kernel<1>;
+        auto* fk = flagship_v() >= 18
+                       ? (Fp == 256   ? fused_fb18_kernel<4>
+                          : Fp == 128 ? fused_fb18_kernel<2>
+                                      : fused_fb18_kernel<1>)
+                       : (Fp == 256   ? fused_fb14_kernel<4>
+                          : Fp == 128 ? fused_fb14_kernel<2>
+                                      : fused_fb14_kernel<1>);
         hipLaunchKernelGGL(fk, dim3(nblk), dim3(1024), 0, s, Xp, Xtp,
                            y.data_ptr<float>(), idx.data_ptr<int>(),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
@@ -4940,7 +5182,8 @@ std::vector<torch::Tensor> linear_search_run(
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::empty({Bp, Fp}, X.options());
   // fused v14 epoch kernel keeps D in LDS -- no global D buffer needed
-  bool use_fused = flagship_v() == 14 && (Fp == 64 || Fp == 128 || Fp == 256);
+  bool use_fused = (flagship_v() == 14 || flagship_v() >= 18) &&
+                   (Fp == 64 || Fp == 128 || Fp == 256);
   auto Dt = torch::empty({(long)Bp, use_fused ? 1L : max_NTp}, X.options());
   auto bias = torch::empty({Bp}, optF);
   auto Vb = torch::empty({Bp}, optF);
@@ -5026,9 +5269,13 @@ std::vector<torch::Tensor> linear_search_run(
       long subt = NTp / 64;
       long Ssub = (subt + 255) / 256;
       unsigned fnblk = (unsigned)((subt + Ssub - 1) / Ssub);
-      auto* fk = Fp == 256   ? fused_fb14_kernel<4>
-                 : Fp == 128 ? fused_fb14_kernel<2>
-                             : fused_fb14_kernel<1>;
+      auto* fk = flagship_v() >= 18
+                     ? (Fp == 256   ? fused_fb18_kernel<4>
+                        : Fp == 128 ? fused_fb18_kernel<2>
+                                    : fused_fb18_kernel<1>)
+                     : (Fp == 256   ? fused_fb14_kernel<4>
+                        : Fp == 128 ? fused_fb14_kernel<2>
+                                    : fused_fb14_kernel<1>);
       for (int epoch = 0; epoch < max_ep; ++epoch) {
         int want_loss = (any_tol || epoch == max_ep - 1) ? 1 : 0;
         if (fused) {
