@@ -68,6 +68,10 @@ class DistributedBackend(Backend):
 
             local = int(os.environ.get("LOCAL_RANK", self.rank))
             device = torch.device("cuda", local % torch.cuda.device_count())
+        if device is not None and getattr(device, "type", None) == "cuda":
+            # pin the rank's device so nccl collectives (incl. the object
+            # gathers) use the right GPU instead of everyone's cuda:0
+            torch.cuda.set_device(device)
         self.device = device
         self._torch = torch
         self._dist = dist
