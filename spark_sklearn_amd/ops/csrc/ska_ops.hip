@@ -3911,6 +3911,151 @@ void scores7_kernel(const __bf16* __restrict__ X,
 }
 
 
+// ---------------------------------------------------------------------------
+// scores8_kernel: wide-tile MULTICLASS scoring (VERDICT round-1 item 6) --
+// scores7's 256x256 GEMM tile, with the per-row argmax/OVR epilogue done
+// through a z-tile that REUSES the staging LDS (the 2x32 KB sX buffers
+// become a [64][256] f32 z chunk after the GEMM loop; 4 row-chunks).
+// Replaces the 32-row scores3 tiles that re-read all of W per block.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024)
+void scores8_kernel(const __bf16* __restrict__ X,
+                    const float* __restrict__ y,
+                    const long* __restrict__ idx,
+                    const __bf16* __restrict__ Wt,
+                    const float* __restrict__ bias,
+                    int C, int metric, long n_idx, int Fp, int Bp,
+                    int n_cand, double* __restrict__ out_sum) {
+  constexpr int NF = 4;
+  __shared__ __bf16 sX[2][256 * 64];
+  __shared__ __bf16 sW[2][256 * 64];
+  __shared__ float sVal[256];
+  __shared__ long sRow[256];
+  __shared__ float sY[256];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int cwave = wave & 3;
+  const int rquad = wave >> 2;
+  const int rt = blockIdx.x;
+  const int g = lane >> 4;
+  const int srow = tid >> 2;
+  const int sseg = tid & 3;
+
+  if (tid < 256) {
+    long r = (long)rt * 256 + tid;
+    long gi = (r < n_idx) ? idx[r] : idx[n_idx - 1];
+    sRow[tid] = gi;
+    sY[tid] = y[gi];
+    sVal[tid] = 0.f;
+  }
+  __syncthreads();
+
+  bf16x8 rx[2], rw[2];
+  auto issue = [&](int kc) {
+    const bf16x8* gx = reinterpret_cast<const bf16x8*>(
+        X + sRow[srow] * Fp + kc + sseg * 16);
+    rx[0] = gx[0];
+    rx[1] = gx[1];
+    const bf16x8* gw = reinterpret_cast<const bf16x8*>(
+        Wt + (long)srow * Fp + kc + sseg * 16);
+    rw[0] = gw[0];
+    rw[1] = gw[1];
+  };
+  auto commit = [&](int b) {
+    store_row8(&sX[b][srow * 64], srow, sseg * 16, rx[0]);
+    store_row8(&sX[b][srow * 64], srow, sseg * 16 + 8, rx[1]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16, rw[0]);
+    store_row8(&sW[b][srow * 64], srow, sseg * 16 + 8, rw[1]);
+  };
+
+  f32x4 acc[4][NF];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  issue(0);
+  commit(0);
+  __syncthreads();
+  const int nkc = Fp >> 6;
+  for (int kc = 0; kc < nkc; ++kc) {
+    int b = kc & 1;
+    if (kc + 1 < nkc) issue((kc + 1) << 6);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 a[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        int r = rquad * 64 + m * 16 + (lane & 15);
+        a[m] = lds_frag_p(&sX[b][r * 64], r, g, h);
+      }
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int c = cwave * 64 + fn * 16 + (lane & 15);
+        bf16x8 bb = lds_frag_p(&sW[b][c * 64], c, g, h);
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[m], bb, acc[m][fn], 0, 0, 0);
+      }
+    }
+    if (kc + 1 < nkc) commit(b ^ 1);
+    __syncthreads();
+  }
+
+  // z-tile through the retired staging LDS, 64 rows per pass
+  float* zbuf = reinterpret_cast<float*>(sX);   // [64][256] f32 = 64 KB
+  const float eps = 1e-12f;
+  for (int q = 0; q < 4; ++q) {
+    __syncthreads();      // previous pass's reads complete
+    if (rquad == q) {
+#pragma unroll
+      for (int fn = 0; fn < NF; ++fn) {
+        int col = cwave * 64 + fn * 16 + (lane & 15);
+        float bia = bias[col];
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+          int rb = m * 16 + ((lane >> 4) << 2);
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            zbuf[(rb + r) * 256 + col] = acc[m][fn][r] + bia;
+        }
+      }
+    }
+    __syncthreads();
+    for (int p = tid; p < 64 * n_cand; p += 1024) {
+      int cd = p % n_cand;      // cand fastest: spreads LDS banks
+      int row = p / n_cand;
+      long gr = (long)rt * 256 + q * 64 + row;
+      if (gr >= n_idx) continue;
+      float yv = sY[q * 64 + row];
+      const float* zr = &zbuf[row * 256 + cd * C];
+      float val;
+      if (metric == 0) {
+        int best = 0;
+        float bz = zr[0];
+        for (int c = 1; c < C; ++c)
+          if (zr[c] > bz) { bz = zr[c]; best = c; }
+        val = (best == (int)yv) ? 1.f : 0.f;
+      } else {
+        float ssum = 0.f, sy = 0.f;
+        for (int c = 0; c < C; ++c) {
+          float s2 = 1.f / (1.f + __expf(-zr[c]));
+          ssum += s2;
+          if (c == (int)yv) sy = s2;
+        }
+        val = __logf(sy / fmaxf(ssum, eps) + eps);
+      }
+      atomicAdd(&sVal[cd], val);
+    }
+  }
+  __syncthreads();
+  if (tid < n_cand)
+    atomicAdd(&out_sum[tid], (double)sVal[tid]);
+}
+
 // multiclass softmax gradients: raw [K][nt] class-major
 __global__ void gbt_softmax_grad_kernel(const float* __restrict__ raw,
                                         const float* __restrict__ y,
@@ -4700,6 +4845,21 @@ torch::Tensor linear_scores(torch::Tensor X, torch::Tensor y,
   };
   int nf = Bp / 64;
   if (g_layout == 3) {
+    if (C > 1 && Bp == 256 && n_idx >= 4096 && Fp % 128 == 0) {
+      // wide multiclass tile (z chunk through the retired staging LDS)
+      hipLaunchKernelGGL(scores8_kernel,
+                         dim3((unsigned)((n_idx + 255) / 256)), dim3(1024),
+                         0, s,
+                         reinterpret_cast<const __bf16*>(X.data_ptr()),
+                         y.data_ptr<float>(), idx.data_ptr<long>(),
+                         reinterpret_cast<const __bf16*>(
+                             Wt_bf16.data_ptr()),
+                         bias.data_ptr<float>(), (int)C, (int)metric,
+                         n_idx, Fp, Bp, (int)n_cand,
+                         out.data_ptr<double>());
+      HIP_CHECK(hipGetLastError());
+      return out;
+    }
     if (C == 1 && Bp == 256 && n_idx >= 4096 && Fp % 128 == 0) {
       // wide tile: 8x less W traffic than the 32-row scores3 blocks
       hipLaunchKernelGGL(scores7_kernel,
@@ -5333,6 +5493,7 @@ std::vector<torch::Tensor> linear_search_run(
         size_t smem3 = 32 * 8 + 32 * (Bp + 8) * 4 + scand_pad * 4 +
                        32 * 64 * 2 + (size_t)Bp * 64 * 2;
         bool wide = (C == 1 && Bp == 256 && Fp % 128 == 0);
+        bool wide_mc = (C > 1 && Bp == 256 && Fp % 128 == 0);
         auto score_one = [&](torch::Tensor& it, long ni, double* outp) {
           if (wide && ni >= 4096) {
             hipLaunchKernelGGL(scores7_kernel,
@@ -5345,6 +5506,20 @@ std::vector<torch::Tensor> linear_search_run(
                                    Wt.data_ptr()),
                                bias.data_ptr<float>(), (int)metric, ni,
                                Fp, Bp, n_cand_chunk, outp);
+            return;
+          }
+          if (wide_mc && ni >= 4096) {
+            hipLaunchKernelGGL(scores8_kernel,
+                               dim3((unsigned)((ni + 255) / 256)),
+                               dim3(1024), 0, s,
+                               reinterpret_cast<const __bf16*>(
+                                   X.data_ptr()),
+                               y.data_ptr<float>(), it.data_ptr<long>(),
+                               reinterpret_cast<const __bf16*>(
+                                   Wt.data_ptr()),
+                               bias.data_ptr<float>(), (int)C,
+                               (int)metric, ni, Fp, Bp, n_cand_chunk,
+                               outp);
             return;
           }
           HIP_CHECK(hipFuncSetAttribute(

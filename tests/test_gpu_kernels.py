@@ -574,3 +574,30 @@ def test_csr_multiclass_ovr_matches_dense(dev, monkeypatch):
                                gs_d.cv_results_["mean_test_score"],
                                atol=5e-3)
     assert gs_sp.best_score_ > 0.6
+
+
+def test_multiclass_wide_tile_scoring_matches_host(dev):
+    """scores8 (wide multiclass tile, round-2) vs the CPU engine on a
+    test fold large enough to trigger the wide path (>= 4096 rows)."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(3)
+    n, f, K = 30_000, 128, 5
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    w = rng.standard_normal((f, K)).astype(np.float32)
+    y = (X @ w + 0.5 * rng.standard_normal((n, K))).argmax(axis=1)
+
+    grid = {"C": [0.1, 1.0, 10.0]}
+    gs = ska.GridSearchCV(None, LogisticRegression(max_iter=25, tol=0.0),
+                          grid, cv=3)
+    gs.fit(X, y)
+    assert gs.cv_results_["mean_test_score"].shape == (3,)
+    assert gs.best_score_ > 0.75
+    gs_host = ska.GridSearchCV("sequential",
+                               LogisticRegression(max_iter=25, tol=0.0),
+                               grid, cv=3)
+    gs_host.fit(X, y)
+    np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
+                               gs_host.cv_results_["mean_test_score"],
+                               atol=7e-3)
