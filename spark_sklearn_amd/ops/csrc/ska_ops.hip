@@ -2669,207 +2669,6 @@ void fwd_d16_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
 }
 
 // ---------------------------------------------------------------------------
-// fwd_d19_kernel: fwd_d16 plus a COALESCED D store: the 256x256 D tile
-// is transposed through the retired staging LDS (sX+sW = 128 KB become
-// the [cand][row] tile after the K loop) and written as contiguous
-// 512 B column runs, replacing the 8 B scattered bf16x4 stores
-// (ablation: store was ~36 us of 355).  Otherwise identical to fwd_d16.  The
-// round-2 ablation (flagship_ablate) showed the kernel is throughput-
-// additive -- X stream ~90us, W staging ~57us, MFMA+LDS reads ~65us,
-// epilogue VALU ~77us, D store ~36us of a 355us launch -- so this variant
-// REMOVES work instead of pipelining it:
-//  * W fragments are read DIRECTLY from global memory.  W is 128 KB and
-//    L2/L3-resident (every XCD caches it), a 4-lane group reads one full
-//    64 B line per fragment, and this deletes the whole W staging path
-//    (2 global_load_lds + 32 KB LDS write + 8 LDS fragment reads per
-//    chunk) -- LDS read traffic halves.
-//  * The epilogue's tail (gr >= nt) and row-key masks are hoisted into a
-//    block-uniform fast path: interior blocks of key-less fits skip 4+
-//    VALU ops per element (210M elements/epoch).
-//  * X keeps the v12 async global->LDS staging (one buffer pair).
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(1024)
-void fwd_d19_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
-                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
-                    const float* __restrict__ bias,
-                    const int* __restrict__ loss_id,
-                    const int* __restrict__ target, __bf16* __restrict__ Dt,
-                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
-                    const int* __restrict__ row_key,
-                    const int* __restrict__ col_key,
-                    int nt, int NTp, int Fp, int Bp, int want_loss) {
-  constexpr int NF = 4;
-  __shared__ __bf16 sBuf[4][256 * 64];   // sX = [0..1], sW = [2..3];
-  __bf16 (*sX)[256 * 64] = &sBuf[0];     // after the K loop the whole
-  __bf16 (*sW)[256 * 64] = &sBuf[2];     // 128 KB becomes the D tile
-  __shared__ float sLoss[256];
-  __shared__ float sDb[256];
-  __shared__ int sIdx[256];
-  __shared__ float sY[256];
-  __shared__ int sKey[256];
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int cwave = wave & 3;
-  const int rquad = wave >> 2;
-  const int rt = blockIdx.x;
-  const int g = lane >> 4;
-  const int rl8 = lane >> 3;
-  const int slot = lane & 7;
-
-  if (tid < 256) {
-    int gi = idx[rt * 256 + tid];
-    sIdx[tid] = gi;
-    sY[tid] = y[gi];
-    sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
-    sLoss[tid] = 0.f;
-    sDb[tid] = 0.f;
-  }
-  __syncthreads();
-
-  auto stage = [&](int b, int kc) {
-#pragma unroll
-    for (int q = 0; q < 2; ++q) {
-      int rbase = q * 128 + wave * 8;
-      int r = rbase + rl8;
-      int ksw = kc + ((slot ^ (r & 7)) << 3);
-      gload_lds16(X + (long)sIdx[r] * Fp + ksw, &sX[b][rbase * 64]);
-      gload_lds16(Wt + (long)r * Fp + ksw, &sW[b][rbase * 64]);
-    }
-  };
-
-  f32x4 acc[4][NF];
-#pragma unroll
-  for (int i = 0; i < 4; ++i)
-#pragma unroll
-    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  const int nkc = Fp >> 6;
-  for (int kc = 0; kc < nkc; ++kc) {
-    int b = kc & 1;
-    if (kc + 1 < nkc) stage(b ^ 1, (kc + 1) << 6);
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      bf16x8 a[4];
-#pragma unroll
-      for (int m = 0; m < 4; ++m) {
-        int r = rquad * 64 + m * 16 + (lane & 15);
-        a[m] = frag12(sX[b], r, h * 4 + g);
-      }
-#pragma unroll
-      for (int fn = 0; fn < NF; ++fn) {
-        int c = cwave * 64 + fn * 16 + (lane & 15);
-        bf16x8 bb = frag12(sW[b], c, h * 4 + g);
-#pragma unroll
-        for (int m = 0; m < 4; ++m)
-          acc[m][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a[m], bb, acc[m][fn], 0, 0, 0);
-      }
-    }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-  }
-
-  const bool interior =
-      (rt * 256 + 256 <= nt) && (row_key == nullptr) && !want_loss;
-#pragma unroll
-  for (int fn = 0; fn < NF; ++fn) {
-    int col_in = cwave * 64 + fn * 16 + (lane & 15);
-    float bia = bias[col_in];
-    int lid = loss_id[col_in];
-    int tgt = target[col_in];
-    int ck = (col_key != nullptr) ? col_key[col_in] : 0;
-    float lsum = 0.f, dsum = 0.f;
-#pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      int row_base = rquad * 64 + m * 16 + ((lane >> 4) << 2);
-      bf16x4 dpack;
-      if (interior) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float z = acc[m][fn][r] + bia;
-          float yv = sY[row_base + r];
-          float d;
-          if (lid == LOSS_LOGISTIC) {
-            float t = (yv == (float)tgt) ? 1.f : 0.f;
-            d = fast_rcp(1.f + __expf(-z)) - t;
-          } else if (lid == LOSS_HINGE) {
-            float sg = (yv == (float)tgt) ? 1.f : -1.f;
-            d = (sg * z < 1.f) ? -sg : 0.f;
-          } else {
-            float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
-            d = z - t;
-          }
-          dpack[r] = (__bf16)d;
-          dsum += d;
-        }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int gr = rt * 256 + row_base + r;
-          float z = acc[m][fn][r] + bia;
-          float d, ls = 0.f;
-          if (want_loss) {
-            loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
-          } else {
-            float yv = sY[row_base + r];
-            if (lid == LOSS_LOGISTIC) {
-              float t = (yv == (float)tgt) ? 1.f : 0.f;
-              d = fast_rcp(1.f + __expf(-z)) - t;
-            } else if (lid == LOSS_HINGE) {
-              float sg = (yv == (float)tgt) ? 1.f : -1.f;
-              d = (sg * z < 1.f) ? -sg : 0.f;
-            } else {
-              float t = (tgt < 0) ? yv : ((yv == (float)tgt) ? 1.f : -1.f);
-              d = z - t;
-            }
-          }
-          if (gr >= nt) { d = 0.f; ls = 0.f; }
-          if (row_key != nullptr && sKey[row_base + r] != ck) {
-            d = 0.f;
-            ls = 0.f;
-          }
-          dpack[r] = (__bf16)d;
-          lsum += ls;
-          dsum += d;
-        }
-      }
-      {  // D tile into LDS [cand][row], 32-granule XOR swizzle
-        __bf16* dT = &sBuf[0][0];
-        int rbg = row_base >> 3, half = (row_base >> 2) & 1;
-        *reinterpret_cast<bf16x4*>(
-            &dT[col_in * 256 + ((rbg ^ (col_in & 31)) << 3) +
-                (half << 2)]) = dpack;
-      }
-    }
-    if (want_loss) atomicAdd(&sLoss[col_in], lsum);
-    atomicAdd(&sDb[col_in], dsum);
-  }
-  __syncthreads();
-  {  // coalesced D write: 512 B contiguous per candidate column
-    const __bf16* dT = &sBuf[0][0];
-#pragma unroll
-    for (int w = 0; w < 8; ++w) {
-      int gidx = tid + (w << 10);      // granule id 0..8191
-      int col = gidx >> 5, rg = gidx & 31;
-      bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          &dT[col * 256 + ((rg ^ (col & 31)) << 3)]);
-      *reinterpret_cast<bf16x8*>(
-          Dt + (long)col * NTp + rt * 256 + (rg << 3)) = v;
-    }
-  }
-  if (tid < 256) {
-    if (want_loss) atomicAdd(&loss_sum[tid], sLoss[tid]);
-    atomicAdd(&db_sum[tid], sDb[tid]);
-  }
-}
-
-
-// ---------------------------------------------------------------------------
 // fwd_d13_kernel: fwd with the v13 counted-vmcnt 4-ring pipeline (see the
 // v13 comment block up top).  K=32 chunks, one raw s_barrier per chunk,
 // chunk kc+1/kc+2 loads stay in flight across it.
@@ -4627,7 +4426,6 @@ static int flagship_v() {
 }
 static auto* flagship_fwd() {
   int v = flagship_v();
-  if (v >= 19) return fwd_d19_kernel;
   if (v >= 16) return fwd_d16_kernel;
   return v == 13 ? fwd_d13_kernel : (v == 12 ? fwd_d12_kernel : fwd_d7_kernel);
 }
