@@ -15,7 +15,7 @@ Documented divergences from sklearn's RandomForest*:
 - max_features selects the k features with the smallest per-(tree, node)
   hash values -- the same uniform k-subset distribution as sklearn's
   permutation draw, but a different (deterministic) stream;
-- tree growth is depth-wise with max_depth required (device cap 6).
+- tree growth is depth-wise with max_depth required (device cap 10).
 """
 
 from __future__ import annotations
@@ -32,7 +32,7 @@ _U32 = np.uint32
 @dataclass
 class RFSpec:
     n_estimators: int = 100
-    max_depth: int = 6
+    max_depth: int = 10
     min_samples_leaf: int = 1
     max_features: int = 0          # 0 => all features
     bootstrap: bool = True

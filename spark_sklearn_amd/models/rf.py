@@ -15,7 +15,7 @@ binned matrix.
 
 Documented divergences from sklearn (see _rf_ref module docstring):
 binned splits, hash-stream feature subsampling, max_depth required on the
-device path (cap 6; higher depths run the numpy/host reference path).
+device path (cap 10, node-grouped deep-level histograms).
 """
 
 from __future__ import annotations
@@ -31,14 +31,14 @@ from ._rf_ref import RFSpec, rf_fit_ref, rf_predict_ref
 
 __all__ = ["RandomForestClassifier", "RandomForestRegressor"]
 
-_DEVICE_MAX_DEPTH = 6
+_DEVICE_MAX_DEPTH = 10
 
 
 class _RFBase(BaseEstimator):
     _hip_batched = True
     _engine_family = "rf"
 
-    def __init__(self, n_estimators=100, max_depth=6, min_samples_leaf=1,
+    def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
                  max_features="sqrt", bootstrap=True, random_state=None):
         self.n_estimators = n_estimators
         self.max_depth = max_depth
@@ -136,7 +136,7 @@ class RandomForestClassifier(ClassifierMixin, _RFBase):
 class RandomForestRegressor(RegressorMixin, _RFBase):
     """Bagged histogram-tree regressor (MSE splits)."""
 
-    def __init__(self, n_estimators=100, max_depth=6, min_samples_leaf=1,
+    def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
                  max_features=1.0, bootstrap=True, random_state=None):
         super().__init__(n_estimators=n_estimators, max_depth=max_depth,
                          min_samples_leaf=min_samples_leaf,

@@ -1586,13 +1586,17 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
                      long nt, long n_rows, int F, int lvl_lo,
                      int lvl_nodes, int f_pb, long rows_pb, int even_only,
-                     float hscale, float gscale) {
+                     int ng_nodes, float hscale, float gscale) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
-  // levels (round-1 PMC: 11.8% occupancy, WAIT-bound on LDS atomics)
-  const int lvl_eff = even_only ? (lvl_nodes >> 1) : lvl_nodes;
-  const int cellsg = lvl_eff * f_pb * 256;
+  // levels (round-1 PMC: 11.8% occupancy, WAIT-bound on LDS atomics).
+  // Deeper levels than LDS can hold are split into node GROUPS of
+  // ng_nodes along grid.z (each z-block scans the rows, histogramming
+  // only its nodes) -- this is what lifts the device tree-depth cap
+  // from 6 to 10.
+  const int ng0 = (int)blockIdx.z * ng_nodes;
+  const int cellsg = ng_nodes * f_pb * 256;
   const bool gfix = gscale > 0.f;
   unsigned long long* lhc = reinterpret_cast<unsigned long long*>(smem);
   unsigned long long* lgq = lhc + cellsg;            // gfix mode only
@@ -1615,6 +1619,8 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
     if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
+    int local = (nd - lvl_lo) >> even_only;
+    if (local < ng0 || local >= ng0 + ng_nodes) continue;
     long gi = idxmap[r];
     float gr = g[r];
     unsigned long long gq = 0ull;
@@ -1622,7 +1628,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                                         (double)gscale + 0.5);
     unsigned hfx = __float2uint_rn(h[r] * hscale);
     unsigned long long hpack = ((unsigned long long)hfx << 32) | 1ull;
-    const int base = ((nd - lvl_lo) >> even_only) * f_pb;
+    const int base = (local - ng0) * f_pb;
     // feature-major bins: each feature is a contiguous [n_rows] stream,
     // so a wave's consecutive-ish gi values coalesce into full 64B
     // lines per feature chunk (the row-major layout pulled a line per
@@ -1659,7 +1665,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
       gv = lg[i];
     int bin = i & 255;
     int ff = (i >> 8) % f_pb;
-    int ndl = (i / (256 * f_pb)) << even_only;   // back to full-level slot
+    int ndl = ((i / (256 * f_pb)) + ng0) << even_only;  // full-level slot
     float* cell = &hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
     atomicAdd(cell, gv);
     atomicAdd(cell + 1, hv);
@@ -1701,16 +1707,19 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
   bool gfix = g_bounded && det_gbt_mode();
   int cell_b = gfix ? 16 : 12;
   int f_pb = std::min(8, std::max(1, (48 * 1024) /
-                                  (cell_b * 256 * lvl_eff)));
+                                  (cell_b * 256 * std::min(lvl_eff, 16))));
   while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
   int fchunks = F / f_pb;
+  int ng_nodes = std::min(lvl_eff,
+                          std::max(1, (48 * 1024) / (cell_b * 256 * f_pb)));
+  int ngroups = (lvl_eff + ng_nodes - 1) / ng_nodes;
   long rowsegs = std::max<long>(
       1, std::min<long>(2048 / fchunks, (nt + 8191) / 8192));
   if (gfix) rowsegs = 1;   // one writer per global cell: the float flush
                            // adds stay single-sourced, so bit-stable
   long rows_pb = (nt + rowsegs - 1) / rowsegs;
   rowsegs = (nt + rows_pb - 1) / rows_pb;
-  size_t lsm = (size_t)lvl_eff * f_pb * 256 * cell_b;
+  size_t lsm = (size_t)ng_nodes * f_pb * 256 * cell_b;
   int even_only = (d > 0) ? 1 : 0;       // sibling subtraction
   float hs = 1.f;
   while (hs * 2.f * hmax * (float)nt + (float)nt < 1.8e9f &&
@@ -1724,10 +1733,11 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
       reinterpret_cast<const void*>(gbt_hist_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
   hipLaunchKernelGGL(gbt_hist_kernel,
-                     dim3((unsigned)rowsegs, (unsigned)fchunks), dim3(512),
+                     dim3((unsigned)rowsegs, (unsigned)fchunks,
+                          (unsigned)ngroups), dim3(512),
                      lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
-                     F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only, hs,
-                     gs);
+                     F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only,
+                     ng_nodes, hs, gs);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -1962,11 +1972,14 @@ __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
                                      const float* __restrict__ h,
                                      float* __restrict__ gsum,
                                      float* __restrict__ hsum,
-                                     long nt, int lvl_lo, int lvl_hi,
+                                     long nt, int lvl_lo_full, int lvl_hi_full,
                                      float gscale, float hscale) {
   __shared__ float lg[128], lh[128];
   __shared__ unsigned long long lgq[128], lhq[128], lcq[128];
   const bool fix = gscale > 0.f;
+  // grid.y chunks the node range in 128s (depth > 7 levels)
+  const int lvl_lo = lvl_lo_full + (int)blockIdx.y * 128;
+  const int lvl_hi = min(lvl_hi_full, lvl_lo + 128);
   int nn = lvl_hi - lvl_lo;
   for (int i = threadIdx.x; i < nn; i += blockDim.x) {
     if (fix) { lgq[i] = 0ull; lhq[i] = 0ull; lcq[i] = 0ull; }
@@ -5163,8 +5176,8 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                    double lam, double f0) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
-  TORCH_CHECK(max_depth >= 1 && max_depth <= 6,
-              "device GBT supports max_depth in [1, 6]");
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 10,
+              "device GBT supports max_depth in [1, 10]");
   int F = (int)Xb.size(1);
   long nt = idxmap.size(0);
   int T = (int)n_estimators;
@@ -5244,7 +5257,8 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
       bool fix = det_gbt_mode() && loss == LOSS_LOGISTIC;
       float fsc = fix ? det_fix_scale(nt) : 0.f;
       hipLaunchKernelGGL(gbt_node_sums_kernel,
-                         dim3(fix ? 1 : blocks), dim3(256), 0, s,
+                         dim3(fix ? 1 : blocks,
+                              (unsigned)(((NN - bl_lo) + 127) / 128)), dim3(256), 0, s,
                          node_id.data_ptr<int>(), g.data_ptr<float>(),
                          h.data_ptr<float>(), gsum.data_ptr<float>(),
                          hsum.data_ptr<float>(), nt, bl_lo, NN, fsc, fsc);
@@ -5723,7 +5737,7 @@ std::vector<torch::Tensor> gbt_fit_multi(
     torch::Tensor f0, int64_t K, int64_t n_estimators, double lrate,
     int64_t max_depth, int64_t min_leaf, double lam) {
   const at::cuda::CUDAGuard guard(Xb.device());
-  TORCH_CHECK(max_depth >= 1 && max_depth <= 6);
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 10);
   TORCH_CHECK(K >= 2 && K <= 64);
   int F = (int)Xb.size(1);
   const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
@@ -5804,7 +5818,8 @@ std::vector<torch::Tensor> gbt_fit_multi(
         bool fix = det_gbt_mode();
         float fsc = fix ? det_fix_scale(nt) : 0.f;
         hipLaunchKernelGGL(gbt_node_sums_kernel,
-                           dim3(fix ? 1 : blocks), dim3(256), 0, s,
+                           dim3(fix ? 1 : blocks,
+                                (unsigned)(((NN - bl_lo) + 127) / 128)), dim3(256), 0, s,
                            node_id.data_ptr<int>(), gc, hc,
                            gsum.data_ptr<float>(), hsum.data_ptr<float>(),
                            nt, bl_lo, NN, fsc, fsc);
@@ -6002,8 +6017,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                   int64_t seed, int64_t bootstrap) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
-  TORCH_CHECK(max_depth >= 1 && max_depth <= 6,
-              "device RF supports max_depth in [1, 6]");
+  TORCH_CHECK(max_depth >= 1 && max_depth <= 10,
+              "device RF supports max_depth in [1, 10]");
   TORCH_CHECK(K <= 64, "device RF supports at most 64 classes");
   int F = (int)Xb.size(1);
   long nt = idxmap0.size(0);
@@ -6117,7 +6132,10 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                  : y_u.data_ptr<float>();
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
-      hipLaunchKernelGGL(gbt_node_sums_kernel, dim3(blocks), dim3(256), 0,
+      hipLaunchKernelGGL(gbt_node_sums_kernel,
+                         dim3(blocks,
+                              (unsigned)(((NN - bl_lo) + 127) / 128)),
+                         dim3(256), 0,
                          s, node_id.data_ptr<int>(), gp,
                          ones.data_ptr<float>(), gsum.data_ptr<float>(),
                          hsum.data_ptr<float>(), nt, bl_lo, NN, 0.f, 0.f);

@@ -193,3 +193,19 @@ def test_deterministic_mode(problem, monkeypatch):
                    tuple(np.asarray(t).tolist())) for f, t, v in trees_r]
     dev_struct = [(f, t) for f, t, v in outs[0]]
     assert dev_struct == ref_struct
+
+
+@pytest.mark.gpu
+def test_gbt_depth8_device(problem):
+    """Depth-8 boosting exercises the node-grouped histogram path."""
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import GradientBoostingClassifier
+
+    X, y = problem
+    gs = ska.GridSearchCV(None,
+                          GradientBoostingClassifier(n_estimators=10,
+                                                     max_depth=8,
+                                                     min_samples_leaf=5),
+                          {"learning_rate": [0.2]}, cv=2)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.85

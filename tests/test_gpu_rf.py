@@ -141,3 +141,24 @@ def test_rf_device_digits_shape_multiclass_accuracy(dev):
                       {"max_depth": [6], "n_estimators": [20, 40]}, cv=3)
     gs.fit(X, y)
     assert gs.best_score_ > 0.85
+
+
+def test_rf_deep_trees_depth8(dev):
+    """Depth-8 forests exercise the node-grouped histogram path (levels
+    whose histograms exceed one block's LDS)."""
+    rng = np.random.default_rng(7)
+    n = 20_000
+    Xb = rng.integers(0, 64, size=(n, 16), dtype=np.uint8)
+    y = (Xb[:, 0].astype(np.float32) + 0.5 * Xb[:, 3] +
+         0.25 * Xb[:, 7] + 0.125 * Xb[:, 11] > 52).astype(np.float32)
+    spec = RFSpec(n_estimators=8, max_depth=8, min_samples_leaf=2,
+                  max_features=8, random_state=1)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec)
+    ref = rf_fit_ref(Xb, y, spec)
+    trees_d = [(f_d[t], t_d[t], v_d[t]) for t in range(len(ref))]
+    p_dev = rf_predict_ref(Xb, trees_d)
+    p_ref = rf_predict_ref(Xb, ref)
+    acc_d = float(np.mean((p_dev > 0.5) == (y > 0.5)))
+    acc_r = float(np.mean((p_ref > 0.5) == (y > 0.5)))
+    assert acc_d > 0.97
+    assert abs(acc_d - acc_r) < 0.01
