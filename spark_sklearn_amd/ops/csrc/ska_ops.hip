@@ -1706,8 +1706,10 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
   bool gfix = g_bounded && det_gbt_mode();
   int cell_b = gfix ? 16 : 12;
-  int f_pb = std::min(8, std::max(1, (48 * 1024) /
-                                  (cell_b * 256 * std::min(lvl_eff, 16))));
+  // feature-major bins have no per-row vector-load constraint: f_pb up
+  // to 16 halves the (node_id, idxmap, g, h) re-reads per level vs 8
+  int f_pb = std::min(16, std::max(1, (48 * 1024) /
+                                   (cell_b * 256 * std::min(lvl_eff, 16))));
   while (F % f_pb != 0) f_pb /= 2;       // F is a multiple of f_pb
   int fchunks = F / f_pb;
   int ng_nodes = std::min(lvl_eff,
