@@ -73,12 +73,13 @@ def _pad_features(F: int) -> int:
 def _as_device_bf16(X, device):
     """Upload X as bf16 [n, Fp] (Fp zero-padded to 64), cached per array.
 
-    scipy CSR matrices reaching THIS dense-ingest path are materialized
+    scipy CSR matrices reaching this dense-ingest path are materialized
     chunk-wise into a dense bf16 resident matrix (288 GB of HBM3E makes
-    even a 10M x 1024 matrix cheap to hold dense).  NOTE: sparse
-    searches at <= 5% density normally run the CSR-RESIDENT K5 kernels
-    in ops/sparse_executor.py and never densify; this fallback covers
-    denser matrices and the multiclass OVR path."""
+    even a 10M x 1024 matrix cheap to hold dense, and the measured
+    crossover -- profiles/r02_csr_crossover.md -- shows dense-ingest is
+    FASTER than the CSR kernels wherever the dense form fits).  Sparse
+    inputs whose dense footprint exceeds the capacity gate run the
+    CSR-resident K5 kernels in ops/sparse_executor.py instead."""
     import scipy.sparse as sp
     import torch
 

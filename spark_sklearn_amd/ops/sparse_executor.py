@@ -1,10 +1,13 @@
 """CSR-resident device execution for the linear family (K5 kernels).
 
-Used when the input is scipy-sparse and low-density: the matrix stays in
-HBM as CSR (indptr/indices/data) plus a (row-chunk, feature, row)-sorted
-transpose entry list for the backward pass -- never densified.  Binary /
-regression and multiclass-OVR targets (accuracy / neg_log_loss;
-the per-row argmax would span lanes).
+Used when the input is scipy-sparse, low-density AND too large for dense
+residency (the capacity gate in engine._new_device_executor; see
+profiles/r02_csr_crossover.md -- dense-ingest measured faster wherever
+the dense form fits).  The matrix stays in HBM as CSR
+(indptr/indices/data) plus a (row-chunk, feature, row)-sorted transpose
+entry list for the backward pass -- never densified.  Targets: binary,
+regression, and multiclass-OVR (accuracy / neg_log_loss via the
+per-candidate argmax kernel csr_scores_multi_kernel).
 """
 
 from __future__ import annotations
