@@ -281,8 +281,13 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     within a fold are chunked to ``column_batch`` columns per launch.
     """
     if fit_params:
-        warnings.warn("fit_params are ignored by the batched device path",
-                      stacklevel=2)
+        warnings.warn(
+            "fit_params "
+            f"({', '.join(sorted(fit_params))}) are IGNORED by the batched "
+            "kernel path: fits run unweighted/unmodified, so results "
+            "differ from an estimator that honors them (sample_weight "
+            "support is a roadmap item; see ROADMAP.md)",
+            stacklevel=2)
     family = get_family(estimator)
     classes, y_enc = _encode_cached(family, estimator, y)
     n_classes = len(classes) if classes is not None else 0
