@@ -1,7 +1,8 @@
 """MI355X-native estimator families (batched CDNA4 kernel fits)."""
 
 from .gbt import GradientBoostingClassifier, GradientBoostingRegressor
-from .rf import RandomForestClassifier, RandomForestRegressor
+from .rf import (ExtraTreesClassifier, ExtraTreesRegressor,
+                 RandomForestClassifier, RandomForestRegressor)
 from .kmeans import KMeans
 from .linear import (LinearRegression, LinearSVC, LogisticRegression, Ridge,
                      SGDClassifier)
@@ -9,4 +10,5 @@ from .linear import (LinearRegression, LinearSVC, LogisticRegression, Ridge,
 __all__ = ["LogisticRegression", "LinearSVC", "SGDClassifier",
            "LinearRegression", "Ridge", "KMeans",
            "GradientBoostingClassifier", "GradientBoostingRegressor",
-           "RandomForestClassifier", "RandomForestRegressor"]
+           "RandomForestClassifier", "RandomForestRegressor",
+           "ExtraTreesClassifier", "ExtraTreesRegressor"]

@@ -38,6 +38,7 @@ class RFSpec:
     bootstrap: bool = True
     random_state: int = 0
     n_classes: int = 0             # 0/2 => single channel; >2 => planes
+    splitter: str = "best"         # "random" => ExtraTrees thresholds
 
 
 def rf_hash(s):
@@ -65,6 +66,12 @@ def _bootstrap_positions(nt, seed):
     r = rf_hash((_U32(seed) ^ i.astype(_U32)))
     return ((r.astype(np.uint64) * np.uint64(nt)) >> np.uint64(32)).astype(
         np.int64)
+
+
+def _thr_hash(fseed, nd, f):
+    with np.errstate(over="ignore"):
+        return rf_hash(_feat_hash(fseed, nd, np.asarray([f]))[0] ^
+                       _U32(0x9E3779B9))
 
 
 def _selected_features(F, kfeat, fseed, nd):
@@ -119,6 +126,18 @@ def _build_rf_tree(Xb, gplanes, spec, fseed):
                 gain -= parent
                 okay = (cl >= spec.min_samples_leaf) & \
                     (cr >= spec.min_samples_leaf)
+                if spec.splitter == "random":
+                    # ExtraTrees: one uniform threshold bin over the
+                    # node's occupied [minbin, maxbin) range
+                    nz = np.flatnonzero(ch)
+                    if len(nz) < 2:
+                        continue
+                    lo, hi = int(nz[0]), int(nz[-1])
+                    tstar = lo + int(_thr_hash(fseed, nd, f) %
+                                     _U32(hi - lo))
+                    mask = np.zeros(255, dtype=bool)
+                    mask[tstar] = True
+                    okay &= mask
                 gain = np.where(okay, gain, -np.inf)
                 t = int(np.argmax(gain))
                 if gain[t] > best_gain + 1e-12:

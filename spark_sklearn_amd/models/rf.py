@@ -29,7 +29,8 @@ from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin
 from ._gbt_ref import bin_data, quantile_bins
 from ._rf_ref import RFSpec, rf_fit_ref, rf_predict_ref
 
-__all__ = ["RandomForestClassifier", "RandomForestRegressor"]
+__all__ = ["RandomForestClassifier", "RandomForestRegressor",
+           "ExtraTreesClassifier", "ExtraTreesRegressor"]
 
 _DEVICE_MAX_DEPTH = 10
 
@@ -37,6 +38,7 @@ _DEVICE_MAX_DEPTH = 10
 class _RFBase(BaseEstimator):
     _hip_batched = True
     _engine_family = "rf"
+    _splitter = "best"
 
     def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
                  max_features="sqrt", bootstrap=True, random_state=None):
@@ -81,6 +83,7 @@ class _RFBase(BaseEstimator):
             bootstrap=bool(self.bootstrap),
             random_state=int(self.random_state or 0),
             n_classes=n_classes,
+            splitter=self._splitter,
         )
 
     def fit(self, X, y):
@@ -133,6 +136,21 @@ class RandomForestClassifier(ClassifierMixin, _RFBase):
         return self.classes_[(raw > 0.5).astype(np.int64)]
 
 
+class ExtraTreesClassifier(RandomForestClassifier):
+    """Extremely-randomized trees: one uniformly-drawn threshold bin per
+    (tree, node, feature) instead of the best-gain scan; no bootstrap by
+    default (sklearn semantics on the shared binned-forest machinery)."""
+
+    _splitter = "random"
+
+    def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
+                 max_features="sqrt", bootstrap=False, random_state=None):
+        super().__init__(n_estimators=n_estimators, max_depth=max_depth,
+                         min_samples_leaf=min_samples_leaf,
+                         max_features=max_features, bootstrap=bootstrap,
+                         random_state=random_state)
+
+
 class RandomForestRegressor(RegressorMixin, _RFBase):
     """Bagged histogram-tree regressor (MSE splits)."""
 
@@ -145,6 +163,19 @@ class RandomForestRegressor(RegressorMixin, _RFBase):
 
     def predict(self, X):
         return self._forest_raw(X).astype(np.float64)
+
+
+class ExtraTreesRegressor(RandomForestRegressor):
+    """Extremely-randomized regression trees (MSE, random thresholds)."""
+
+    _splitter = "random"
+
+    def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
+                 max_features=1.0, bootstrap=False, random_state=None):
+        super().__init__(n_estimators=n_estimators, max_depth=max_depth,
+                         min_samples_leaf=min_samples_leaf,
+                         max_features=max_features, bootstrap=bootstrap,
+                         random_state=random_state)
 
 
 class RFFamily:

@@ -1804,6 +1804,11 @@ __device__ __forceinline__ unsigned rf_feat_hash(unsigned fseed, int nd,
   return rf_hash(fseed ^ (unsigned)(nd * 2654435761u) ^
                  (unsigned)(f * 40503u));
 }
+// ExtraTrees random-threshold stream (distinct from the selection stream)
+__device__ __forceinline__ unsigned rf_thr_hash(unsigned fseed, int nd,
+                                                int f) {
+  return rf_hash(rf_feat_hash(fseed, nd, f) ^ 0x9E3779B9u);
+}
 
 // bootstrap resample: position i draws source position (hash * nt) >> 32
 __global__ void rf_bootstrap_kernel(const int* __restrict__ idx_in,
@@ -1862,7 +1867,7 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
                            int lvl_lo, float lam, float lr, int min_leaf,
                            int force_leaf, int* __restrict__ feat,
                            int* __restrict__ thr, float* __restrict__ val,
-                           int kfeat, unsigned fseed) {
+                           int kfeat, unsigned fseed, int etr) {
   const int nd_local = blockIdx.x;
   const int nd = lvl_lo + nd_local;
   const int tid = threadIdx.x;
@@ -1903,6 +1908,29 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
     }
     float G = __shfl(pg, 63), H = __shfl(ph, 63), C = __shfl(pc, 63);
     if (f == 0 && lane == 0) { sG = G; sH = H; }
+    // ExtraTrees: one RANDOM threshold bin per (node, feature), drawn
+    // uniformly over the node's occupied bin range [minbin, maxbin)
+    int tstar = -1;
+    if (etr) {
+      int lmin = 256, lmax = -1;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (c[j] > 0.f) {
+          int b = 4 * lane + j;
+          if (b < lmin) lmin = b;
+          if (b > lmax) lmax = b;
+        }
+      }
+      for (int off = 32; off > 0; off >>= 1) {
+        lmin = min(lmin, __shfl_down(lmin, off));
+        lmax = max(lmax, __shfl_down(lmax, off));
+      }
+      lmin = __shfl(lmin, 0);
+      lmax = __shfl(lmax, 0);
+      if (lmax > lmin)
+        tstar = lmin + (int)(rf_thr_hash(fseed, nd, f) %
+                             (unsigned)(lmax - lmin));
+    }
     if (!force_leaf &&
         (kfeat >= F || rf_feat_hash(fseed, nd, f) < fthr)) {
       float ps = G * G / (H + lam + 1e-12f);
@@ -1912,6 +1940,7 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
         int b = 4 * lane + j;
         gl += g[j]; hl += h[j]; cl += c[j];
         if (b >= 255) continue;
+        if (etr && b != tstar) continue;
         float cr = C - cl;
         if (cl < (float)min_leaf || cr < (float)min_leaf) continue;
         float gr2 = G - gl, hr2 = H - hl;
@@ -5254,7 +5283,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
       hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
                          0, s, hist.data_ptr<float>(), F, lvl_lo,
                          (float)lam, (float)lrate, (int)min_leaf, 0,
-                         feat_t, thr_t, val_t, 1 << 30, 0u);
+                         feat_t, thr_t, val_t, 1 << 30, 0u, 0);
       if (d + 1 < (int)max_depth)
         hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
             .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
@@ -5816,7 +5845,7 @@ std::vector<torch::Tensor> gbt_fit_multi(
                            dim3(256), 0, s, hist.data_ptr<float>(), F,
                            lvl_lo, (float)lam, (float)lrate,
                            (int)min_leaf, 0, feat_t, thr_t, val_t,
-                           1 << 30, 0u);
+                           1 << 30, 0u, 0);
         if (d + 1 < (int)max_depth)
           hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
               .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
@@ -5879,7 +5908,7 @@ torch::Tensor gbt_metric_multi(torch::Tensor raws, torch::Tensor y,
 __global__ __launch_bounds__(256)
 void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
                                 int K, int F, int lvl_lo, int min_leaf,
-                                int kfeat, unsigned fseed,
+                                int kfeat, unsigned fseed, int etr,
                                 int* __restrict__ feats,
                                 int* __restrict__ thrs,
                                 float* __restrict__ vals, long NN) {
@@ -5929,6 +5958,27 @@ void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
       }
     }
     if (f == 0 && lane == 0) sCnt = C;
+    int tstar = -1;
+    if (etr) {
+      int lmin = 256, lmax = -1;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (c4[j] > 0.f) {
+          int b = 4 * lane + j;
+          if (b < lmin) lmin = b;
+          if (b > lmax) lmax = b;
+        }
+      }
+      for (int off = 32; off > 0; off >>= 1) {
+        lmin = min(lmin, __shfl_down(lmin, off));
+        lmax = max(lmax, __shfl_down(lmax, off));
+      }
+      lmin = __shfl(lmin, 0);
+      lmax = __shfl(lmax, 0);
+      if (lmax > lmin)
+        tstar = lmin + (int)(rf_thr_hash(fseed, nd, f) %
+                             (unsigned)(lmax - lmin));
+    }
     float gainsum[4] = {0.f, 0.f, 0.f, 0.f};
     float parent = 0.f;
     for (int c = 0; c < K; ++c) {
@@ -5962,6 +6012,7 @@ void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
       for (int j = 0; j < 4; ++j) {
         int b = 4 * lane + j;
         if (b >= 255) continue;
+        if (etr && b != tstar) continue;
         if (cl[j] < (float)min_leaf || C - cl[j] < (float)min_leaf)
           continue;
         float gain = gainsum[j] - parent;
@@ -6030,7 +6081,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                   torch::Tensor y_tr, int64_t K,
                                   int64_t n_estimators, int64_t max_depth,
                                   int64_t min_leaf, int64_t max_features,
-                                  int64_t seed, int64_t bootstrap) {
+                                  int64_t seed, int64_t bootstrap,
+                                  int64_t splitter_random) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
   TORCH_CHECK(max_depth >= 1 && max_depth <= 10,
@@ -6123,13 +6175,14 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
       if (KP > 1) {
         hipLaunchKernelGGL(rf_split_find_multi_kernel, dim3(lvl_nodes),
                            dim3(256), 0, s, hist.data_ptr<float>(), ccells,
-                           KP, F, lvl_lo, (int)min_leaf, kfeat, fs, feat_t,
+                           KP, F, lvl_lo, (int)min_leaf, kfeat, fs,
+                           (int)splitter_random, feat_t,
                            thr_t, val_t, (long)NN);
       } else {
         hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes), dim3(256),
                            0, s, hist.data_ptr<float>(), F, lvl_lo, 0.f,
                            -1.f, (int)min_leaf, 0, feat_t, thr_t, val_t,
-                           kfeat, fs);
+                           kfeat, fs, (int)splitter_random);
       }
       if (d + 1 < (int)max_depth) {
         long lvlcells = (long)lvl_nodes * F * 256 * 3;

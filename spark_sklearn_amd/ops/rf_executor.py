@@ -37,7 +37,8 @@ class DeviceRFExecutor(DeviceGBTExecutor):
                 self.Xb, self.Xbt, idx_t, y_tr, K, spec.n_estimators,
                 spec.max_depth,
                 spec.min_samples_leaf, spec.max_features,
-                spec.random_state, 1 if spec.bootstrap else 0)
+                spec.random_state, 1 if spec.bootstrap else 0,
+                1 if spec.splitter == "random" else 0)
             T = spec.n_estimators
             f_np = feats.cpu().numpy()
             t_np = thrs.cpu().numpy().astype(np.uint8)

@@ -162,3 +162,29 @@ def test_rf_deep_trees_depth8(dev):
     acc_r = float(np.mean((p_ref > 0.5) == (y > 0.5)))
     assert acc_d > 0.97
     assert abs(acc_d - acc_r) < 0.01
+
+
+def test_extratrees_device_matches_ref(dev):
+    rng = np.random.default_rng(9)
+    n = 4000
+    Xb = rng.integers(0, 64, size=(n, 12), dtype=np.uint8)
+    y = (Xb[:, 0].astype(np.float32) + 0.5 * Xb[:, 4] > 40).astype(
+        np.float32)
+    spec = RFSpec(n_estimators=5, max_depth=3, min_samples_leaf=2,
+                  max_features=6, bootstrap=False, random_state=4,
+                  splitter="random")
+    ext = _ext()
+    Xb_t = torch.as_tensor(Xb).to(dev)
+    idx = torch.arange(n, dtype=torch.int32, device=dev)
+    yt = torch.as_tensor(y).to(dev)
+    feats, thrs, vals = ext.rf_fit(Xb_t, Xb_t.t().contiguous(), idx, yt, 0,
+                                   5, 3, 2, 6, 4, 0, 1)
+    ref = rf_fit_ref(Xb, y, spec)
+    f_d = feats.cpu().numpy()
+    t_d = thrs.cpu().numpy().astype(np.uint8)
+    v_d = vals.cpu().numpy()
+    for t, (f_r, t_r, v_r) in enumerate(ref):
+        assert np.array_equal(f_d[t], f_r), f"tree {t}"
+        split = f_d[t] >= 0
+        assert np.array_equal(t_d[t][split], t_r[split])
+        np.testing.assert_allclose(v_d[t], v_r, atol=2e-5)

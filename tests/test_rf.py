@@ -141,3 +141,40 @@ def test_rf_ref_multiclass_planes_share_structure():
     assert raw.shape == (500, 4)
     # forest-average probabilities sum to ~1
     assert np.allclose(raw.sum(axis=1), 1.0, atol=1e-4)
+
+
+def test_extratrees_classifier_and_regressor():
+    X, y = make_classification(n_samples=1500, n_features=16,
+                               n_informative=8, random_state=0)
+    from spark_sklearn_amd.models import (ExtraTreesClassifier,
+                                          ExtraTreesRegressor)
+
+    et = ExtraTreesClassifier(n_estimators=20, max_depth=8,
+                              random_state=0).fit(X, y)
+    acc = (et.predict(X) == y).mean()
+    from sklearn.ensemble import ExtraTreesClassifier as SkET
+
+    sk = SkET(n_estimators=20, max_depth=8, random_state=0).fit(X, y)
+    assert acc > 0.85
+    assert abs(acc - (sk.predict(X) == y).mean()) < 0.08
+    # no bootstrap by default, random thresholds still vary trees
+    roots = {int(t[0][0]) for t in et._trees}
+    assert len(roots) >= 1
+
+    Xr, yr = make_regression(n_samples=1200, n_features=10, noise=3.0,
+                             random_state=1)
+    er = ExtraTreesRegressor(n_estimators=20, max_depth=8,
+                             random_state=0).fit(Xr, yr)
+    from sklearn.metrics import r2_score
+
+    assert r2_score(yr, er.predict(Xr)) > 0.5
+
+
+def test_extratrees_grid_search():
+    X, y = load_digits(return_X_y=True)
+    from spark_sklearn_amd.models import ExtraTreesClassifier
+
+    gs = GridSearchCV(None, ExtraTreesClassifier(random_state=0),
+                      {"max_depth": [8], "n_estimators": [15, 30]}, cv=3)
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.85
