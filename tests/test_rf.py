@@ -171,10 +171,12 @@ def test_extratrees_classifier_and_regressor():
 
 
 def test_extratrees_grid_search():
-    X, y = load_digits(return_X_y=True)
+    # small shapes: the CPU tier runs the numpy reference forest
+    X, y = make_classification(n_samples=600, n_features=12,
+                               n_informative=6, random_state=2)
     from spark_sklearn_amd.models import ExtraTreesClassifier
 
     gs = GridSearchCV(None, ExtraTreesClassifier(random_state=0),
-                      {"max_depth": [8], "n_estimators": [15, 30]}, cv=3)
+                      {"max_depth": [6], "n_estimators": [10, 20]}, cv=2)
     gs.fit(X, y)
-    assert gs.best_score_ > 0.85
+    assert gs.best_score_ > 0.8
