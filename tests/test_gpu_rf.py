@@ -35,7 +35,8 @@ def _dev_trees(ext, dev, Xb_np, y_np, spec, K=0):
     feats, thrs, vals = ext.rf_fit(
         Xb, Xbt, idx, y, K, spec.n_estimators, spec.max_depth,
         spec.min_samples_leaf, spec.max_features, spec.random_state,
-        1 if spec.bootstrap else 0)
+        1 if spec.bootstrap else 0,
+        1 if spec.splitter == "random" else 0)
     return (feats.cpu().numpy(), thrs.cpu().numpy().astype(np.uint8),
             vals.cpu().numpy())
 
