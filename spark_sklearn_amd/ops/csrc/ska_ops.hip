@@ -4472,9 +4472,17 @@ void csr_scores_multi_kernel(const int* __restrict__ indptr,
 
 static int g_layout = -1;   // P = AMAP*2+BMAP chosen by the runtime probe
 
-// Flagship fwd/bwd kernel version: 13 (counted-vmcnt 4-ring pipeline,
-// default), 12 (async-staged 2-phase) or 7 (register-staged round-1
-// kernels); A/B via SKA_FLAGSHIP_V.
+// Flagship fwd/bwd kernel version selector (SKA_FLAGSHIP_V; the round-2
+// measured ladder lives in profiles/r02_flagship_ablation.md):
+//   16 (DEFAULT) fwd_d16 + bwd_g12: async-staged 2-phase, block-uniform
+//      fast epilogue -- 17.0-17.2k candidate-fits/s
+//   7  round-1 register-staged kernels (baseline)
+//   12 async global->LDS staging only
+//   13 counted-vmcnt 4-buffer ring      (measured slower; kept for A/B)
+//   14 fused epoch, D resident in LDS   (measured slower; kept for A/B)
+//   15 512-thread 128x128 multi-block   (measured slower; kept for A/B)
+//   17 pass-pipelined epoch, side stream(measured slower; kept for A/B)
+//   18 fused + double-buffered staging  (measured slower; kept for A/B)
 static int flagship_v() {
   static int v = [] {
     const char* e = getenv("SKA_FLAGSHIP_V");
