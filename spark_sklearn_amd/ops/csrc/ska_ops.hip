@@ -1586,7 +1586,8 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      float* __restrict__ hist,  // [lvl_nodes][F][256][3]
                      long nt, long n_rows, int F, int lvl_lo,
                      int lvl_nodes, int f_pb, long rows_pb, int even_only,
-                     int ng_nodes, int g01, float hscale, float gscale) {
+                     int ng_nodes, int g01, float hscale, float gscale,
+                     int KP, long g_stride, long hist_stride) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1607,6 +1608,13 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   const long r0 = (long)blockIdx.x * rows_pb;
   const long r1 = min(r0 + rows_pb, nt);
 
+  // KP > 1: one launch histograms ALL class planes (g + c*g_stride ->
+  // hist + c*hist_stride) with the block's row range cache-hot across
+  // classes -- replaces K separate launches (the digits-style K=10
+  // forests were launch-bound, and the row-metadata HBM re-read was xK)
+  for (int cls = 0; cls < KP; ++cls) {
+  const float* gcls = g + (long)cls * g_stride;
+  float* hist_c = hist + (long)cls * hist_stride;
   if (gfix) {
     for (int i = tid; i < 2 * cellsg; i += nthr) lhc[i] = 0ull;
   } else if (g01) {
@@ -1627,7 +1635,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     int local = (nd - lvl_lo) >> even_only;
     if (local < ng0 || local >= ng0 + ng_nodes) continue;
     long gi = idxmap[r];
-    float gr = g[r];
+    float gr = gcls[r];
     unsigned long long gq = 0ull;
     if (gfix) gq = (unsigned long long)((double)(gr + 1.f) *
                                         (double)gscale + 0.5);
@@ -1679,10 +1687,12 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     int bin = i & 255;
     int ff = (i >> 8) % f_pb;
     int ndl = ((i / (256 * f_pb)) + ng0) << even_only;  // full-level slot
-    float* cell = &hist[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
+    float* cell = &hist_c[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
     atomicAdd(cell, gv);
     atomicAdd(cell + 1, hv);
     atomicAdd(cell + 2, (float)cnt);
+  }
+  if (KP > 1) __syncthreads();   // LDS reuse across class planes
   }
 }
 
@@ -1713,7 +1723,9 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                               const int* node_id, const float* g,
                               const float* h, float* hist, long nt,
                               long n_rows, int F, int d, float hmax,
-                              bool g_bounded, bool g01, hipStream_t s) {
+                              bool g_bounded, bool g01, hipStream_t s,
+                              int KP = 1, long g_stride = 0,
+                              long hist_stride = 0) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
@@ -1743,8 +1755,10 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
     hs *= 2.f;
   // biased g in [0,2]: keep nt*(2*gs + 1) well under 2^62
   float gs = gfix ? det_fix_scale(nt) : 0.f;
-  HIP_CHECK(hipMemsetAsync(hist, 0,
-                           (size_t)lvl_nodes * F * 256 * 3 * 4, s));
+  HIP_CHECK(hipMemsetAsync(
+      hist, 0,
+      ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
+       (size_t)lvl_nodes * F * 256 * 3) * 4, s));
   HIP_CHECK(hipFuncSetAttribute(
       reinterpret_cast<const void*>(gbt_hist_kernel),
       hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
@@ -1753,7 +1767,8 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                           (unsigned)ngroups), dim3(512),
                      lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
                      F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only,
-                     ng_nodes, g01 ? 1 : 0, hs, gs);
+                     ng_nodes, g01 ? 1 : 0, hs, gs, KP, g_stride,
+                     hist_stride);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -6158,17 +6173,20 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
     for (int d = 0; d < (int)max_depth; ++d) {
       int lvl_lo = (1 << d) - 1;
       int lvl_nodes = 1 << d;
-      for (int c = 0; c < KP; ++c) {
-        const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
+      {
+        // ONE launch histograms every class plane (rows stay cache-hot
+        // across classes inside each block); classification g-planes are
+        // 0/1 indicators -> packed single-atomic cells
+        const float* gp = (KP > 1) ? G.data_ptr<float>()
                                    : y_u.data_ptr<float>();
-        // classification g-planes are 0/1 indicators: packed single-
-        // atomic cells (regression keeps the float g channel)
         launch_hist_level(Xtp, idx_u.data_ptr<int>(),
                           node_id.data_ptr<int>(), gp,
-                          ones.data_ptr<float>(),
-                          hist.data_ptr<float>() + (long)c * ccells, nt,
-                          n_rows, F, d, 1.f, false, K != 0, s);
-        if (d > 0) {
+                          ones.data_ptr<float>(), hist.data_ptr<float>(),
+                          nt, n_rows, F, d, 1.f, false, K != 0, s, KP,
+                          nt, ccells);
+      }
+      if (d > 0) {
+        for (int c = 0; c < KP; ++c) {
           int prev_lo = (1 << (d - 1)) - 1;
           long cells = (long)F * 256 * 3;
           unsigned gx = (unsigned)std::min<long>((cells + 255) / 256, 1024);
