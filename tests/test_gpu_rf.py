@@ -189,3 +189,31 @@ def test_extratrees_device_matches_ref(dev):
         split = f_d[t] >= 0
         assert np.array_equal(t_d[t][split], t_r[split])
         np.testing.assert_allclose(v_d[t], v_r, atol=2e-5)
+
+
+def test_extratrees_multiclass_device_matches_ref(dev):
+    """Random-threshold splitter through the K-class gini finder."""
+    rng = np.random.default_rng(12)
+    n, K = 3000, 4
+    Xb = rng.integers(0, 32, size=(n, 10), dtype=np.uint8)
+    y = ((Xb[:, 2] // 8) % K).astype(np.float32)
+    spec = RFSpec(n_estimators=4, max_depth=3, min_samples_leaf=2,
+                  max_features=5, bootstrap=False, random_state=6,
+                  splitter="random", n_classes=K)
+    ext = _ext()
+    Xb_t = torch.as_tensor(Xb).to(dev)
+    idx = torch.arange(n, dtype=torch.int32, device=dev)
+    yt = torch.as_tensor(y).to(dev)
+    feats, thrs, vals = ext.rf_fit(Xb_t, Xb_t.t().contiguous(), idx, yt, K,
+                                   4, 3, 2, 5, 6, 0, 1)
+    ref = rf_fit_ref(Xb, y, spec)
+    f_d = feats.cpu().numpy()
+    t_d = thrs.cpu().numpy().astype(np.uint8)
+    v_d = vals.cpu().numpy()
+    for t, planes in enumerate(ref):
+        for c, (f_r, t_r, v_r) in enumerate(planes):
+            i = t * K + c
+            assert np.array_equal(f_d[i], f_r), f"tree {t} class {c}"
+            split = f_d[i] >= 0
+            assert np.array_equal(t_d[i][split], t_r[split])
+            np.testing.assert_allclose(v_d[i], v_r, atol=2e-5)
