@@ -1587,7 +1587,8 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      long nt, long n_rows, int F, int lvl_lo,
                      int lvl_nodes, int f_pb, long rows_pb, int even_only,
                      int ng_nodes, int g01, float hscale, float gscale,
-                     int KP, long g_stride, long hist_stride) {
+                     float gbias, int KP, long g_stride,
+                     long hist_stride) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1637,7 +1638,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     long gi = idxmap[r];
     float gr = gcls[r];
     unsigned long long gq = 0ull;
-    if (gfix) gq = (unsigned long long)((double)(gr + 1.f) *
+    if (gfix) gq = (unsigned long long)((double)(gr + gbias) *
                                         (double)gscale + 0.5);
     unsigned long long hpack;
     if (g01) {
@@ -1681,7 +1682,8 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
       hv = (float)cnt;                // h == 1 per row
     } else {
       hv = (float)(p >> 32) * hinv;
-      gv = gfix ? (float)((double)lgq[i] / (double)gscale - (double)cnt)
+      gv = gfix ? (float)((double)lgq[i] / (double)gscale -
+                          (double)cnt * (double)gbias)
                 : lg[i];
     }
     int bin = i & 255;
@@ -1712,11 +1714,33 @@ static bool det_gbt_mode() {
   return e != nullptr && e[0] == '1';
 }
 
-static float det_fix_scale(long nt) {
-  double cap = 4.0e18 / (2.0 * (double)nt + 1.0);
+static float det_fix_scale(long nt, float gbias = 1.f) {
+  // biased g in [0, 2*gbias]: keep nt*(2*gbias*gs + 1) well under 2^62
+  double cap = 4.0e18 / (2.0 * (double)gbias * (double)nt + 1.0);
   float gs = 1.f;
   while ((double)gs * 2.0 < cap && gs < 1.0e12f) gs *= 2.f;
+  while ((double)gs > cap && gs > 1e-12f) gs *= 0.5f;
   return gs;
+}
+
+// deterministic per-round |g| bound for the squared-loss det mode:
+// bit-level atomicMax on non-negative floats is order-independent
+__global__ void absmax_kernel(const float* __restrict__ g, long n,
+                              unsigned* __restrict__ out_bits) {
+  float m = 0.f;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    m = fmaxf(m, fabsf(g[i]));
+  for (int off = 32; off > 0; off >>= 1)
+    m = fmaxf(m, __shfl_down(m, off));
+  __shared__ float wm[16];
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) wm[wave] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) m = fmaxf(m, wm[w]);
+    atomicMax(out_bits, __float_as_uint(m));
+  }
 }
 
 static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
@@ -1725,7 +1749,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                               long n_rows, int F, int d, float hmax,
                               bool g_bounded, bool g01, hipStream_t s,
                               int KP = 1, long g_stride = 0,
-                              long hist_stride = 0) {
+                              long hist_stride = 0, float gbias = 1.f) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
@@ -1754,7 +1778,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
          hs < 65536.f)
     hs *= 2.f;
   // biased g in [0,2]: keep nt*(2*gs + 1) well under 2^62
-  float gs = gfix ? det_fix_scale(nt) : 0.f;
+  float gs = gfix ? det_fix_scale(nt, gbias) : 0.f;
   HIP_CHECK(hipMemsetAsync(
       hist, 0,
       ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
@@ -1767,7 +1791,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                           (unsigned)ngroups), dim3(512),
                      lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
                      F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only,
-                     ng_nodes, g01 ? 1 : 0, hs, gs, KP, g_stride,
+                     ng_nodes, g01 ? 1 : 0, hs, gs, gbias, KP, g_stride,
                      hist_stride);
 }
 
@@ -2033,7 +2057,8 @@ __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
                                      float* __restrict__ gsum,
                                      float* __restrict__ hsum,
                                      long nt, int lvl_lo_full, int lvl_hi_full,
-                                     float gscale, float hscale) {
+                                     float gscale, float hscale,
+                                     float gbias) {
   __shared__ float lg[128], lh[128];
   __shared__ unsigned long long lgq[128], lhq[128], lcq[128];
   const bool fix = gscale > 0.f;
@@ -2052,7 +2077,7 @@ __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
     if (nd < lvl_lo || nd >= lvl_hi) continue;
     if (fix) {
       atomicAdd(&lgq[nd - lvl_lo],
-                (unsigned long long)((double)(g[r] + 1.f) *
+                (unsigned long long)((double)(g[r] + gbias) *
                                      (double)gscale + 0.5));
       atomicAdd(&lhq[nd - lvl_lo],
                 (unsigned long long)((double)h[r] *
@@ -2069,7 +2094,7 @@ __global__ void gbt_node_sums_kernel(const int* __restrict__ node_id,
       if (lcq[i] != 0ull) {
         atomicAdd(&gsum[lvl_lo + i],
                   (float)((double)lgq[i] / (double)gscale -
-                          (double)lcq[i]));
+                          (double)lcq[i] * (double)gbias));
         atomicAdd(&hsum[lvl_lo + i],
                   (float)((double)lhq[i] / (double)hscale));
       }
@@ -5275,6 +5300,13 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
   const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
   long n_rows = Xb.size(0);
   int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+  // deterministic squared loss: g is unbounded, so each round measures
+  // max|g| (bit-level atomicMax, order-independent) to set the
+  // fixed-point bias/scale; the one host sync per round is the price of
+  // the opt-in mode
+  bool det_sq = det_gbt_mode() && loss == LOSS_SQUARED;
+  auto gmax_bits = torch::zeros({1}, optI);
+  float gbias_r = 1.f;
 
   for (int t = 0; t < T; ++t) {
     int* feat_t = feats.data_ptr<int>() + (long)t * NN;
@@ -5284,6 +5316,17 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
                        raw.data_ptr<float>(), y_tr.data_ptr<float>(),
                        g.data_ptr<float>(), h.data_ptr<float>(), nt,
                        (int)loss);
+    if (det_sq) {
+      HIP_CHECK(hipMemsetAsync(gmax_bits.data_ptr(), 0, 4, s));
+      hipLaunchKernelGGL(absmax_kernel, dim3(blocks), dim3(256), 0, s,
+                         g.data_ptr<float>(), nt,
+                         reinterpret_cast<unsigned*>(
+                             gmax_bits.data_ptr<int>()));
+      int bits = gmax_bits.item<int>();            // syncs the stream
+      float gm;
+      memcpy(&gm, &bits, 4);
+      gbias_r = std::max(gm, 1e-3f);
+    }
     for (int d = 0; d < (int)max_depth; ++d) {
       int lvl_lo = (1 << d) - 1;
       int lvl_nodes = 1 << d;
@@ -5292,7 +5335,8 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
                         node_id.data_ptr<int>(), g.data_ptr<float>(),
                         h.data_ptr<float>(), hist.data_ptr<float>(), nt, n_rows,
                         F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f,
-                        loss == LOSS_LOGISTIC, false, s);
+                        loss == LOSS_LOGISTIC || det_sq, false, s,
+                        1, 0, 0, det_sq ? gbias_r : 1.f);
       if (even_only) {
         int prev_lo = (1 << (d - 1)) - 1;
         long cells = (long)F * 256 * 3;
@@ -5322,14 +5366,17 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
     HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
     HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
     {
-      bool fix = det_gbt_mode() && loss == LOSS_LOGISTIC;
-      float fsc = fix ? det_fix_scale(nt) : 0.f;
+      bool fix = det_gbt_mode() &&
+                 (loss == LOSS_LOGISTIC || loss == LOSS_SQUARED);
+      float gb = det_sq ? gbias_r : 1.f;
+      float fsc = fix ? det_fix_scale(nt, gb) : 0.f;
       hipLaunchKernelGGL(gbt_node_sums_kernel,
                          dim3(fix ? 1 : blocks,
                               (unsigned)(((NN - bl_lo) + 127) / 128)), dim3(256), 0, s,
                          node_id.data_ptr<int>(), g.data_ptr<float>(),
                          h.data_ptr<float>(), gsum.data_ptr<float>(),
-                         hsum.data_ptr<float>(), nt, bl_lo, NN, fsc, fsc);
+                         hsum.data_ptr<float>(), nt, bl_lo, NN, fsc, fsc,
+                         gb);
     }
     hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                        dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
@@ -5890,7 +5937,7 @@ std::vector<torch::Tensor> gbt_fit_multi(
                                 (unsigned)(((NN - bl_lo) + 127) / 128)), dim3(256), 0, s,
                            node_id.data_ptr<int>(), gc, hc,
                            gsum.data_ptr<float>(), hsum.data_ptr<float>(),
-                           nt, bl_lo, NN, fsc, fsc);
+                           nt, bl_lo, NN, fsc, fsc, 1.f);
       }
       hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                          dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
@@ -6235,7 +6282,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                          dim3(256), 0,
                          s, node_id.data_ptr<int>(), gp,
                          ones.data_ptr<float>(), gsum.data_ptr<float>(),
-                         hsum.data_ptr<float>(), nt, bl_lo, NN, 0.f, 0.f);
+                         hsum.data_ptr<float>(), nt, bl_lo, NN, 0.f, 0.f,
+                         1.f);
       hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                          dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
                          gsum.data_ptr<float>(), hsum.data_ptr<float>(),

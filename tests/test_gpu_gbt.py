@@ -209,3 +209,28 @@ def test_gbt_depth8_device(problem):
                           {"learning_rate": [0.2]}, cv=2)
     gs.fit(X, y)
     assert gs.best_score_ > 0.85
+
+
+@pytest.mark.gpu
+def test_deterministic_mode_squared_loss(problem, monkeypatch):
+    """Round-2: SKA_DETERMINISTIC_GBT=1 now covers squared loss via a
+    per-round max-|g| reduction setting the fixed-point bias/scale."""
+    from spark_sklearn_amd.models._gbt_ref import LOSS_SQUARED, GBTSpec
+
+    X, y = problem
+    dev = torch.device("cuda", 0)
+    yr = (y.astype(np.float32) * 3.0 +
+          X[:, 0].astype(np.float32))          # real-valued target
+    spec = GBTSpec(loss=LOSS_SQUARED, n_estimators=5, max_depth=4,
+                   learning_rate=0.3, min_samples_leaf=20)
+    train = np.arange(15000, dtype=np.int64)
+
+    monkeypatch.setenv("SKA_DETERMINISTIC_GBT", "1")
+    outs = []
+    for _ in range(2):
+        ex = DeviceGBTExecutor(np.ascontiguousarray(X), yr, dev)
+        (models, edges), _, _ = ex.fit(train, _GBTPlan([spec]))
+        trees, f0 = models[0]
+        outs.append([(tuple(f.tolist()), tuple(t.tolist()),
+                      tuple(v.tolist())) for f, t, v in trees])
+    assert outs[0] == outs[1]          # bit-identical incl. leaf values
