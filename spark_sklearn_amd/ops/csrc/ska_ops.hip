@@ -1588,7 +1588,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      int lvl_nodes, int f_pb, long rows_pb, int even_only,
                      int ng_nodes, int g01, float hscale, float gscale,
                      float gbias, int KP, long g_stride,
-                     long hist_stride, long node_stride, long h_stride) {
+                     long hist_stride) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1615,8 +1615,6 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   // forests were launch-bound, and the row-metadata HBM re-read was xK)
   for (int cls = 0; cls < KP; ++cls) {
   const float* gcls = g + (long)cls * g_stride;
-  const float* hcls = h + (long)cls * h_stride;
-  const int* ncls = node_id + (long)cls * node_stride;
   float* hist_c = hist + (long)cls * hist_stride;
   if (gfix) {
     for (int i = tid; i < 2 * cellsg; i += nthr) lhc[i] = 0ull;
@@ -1632,7 +1630,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
 
   const bool vec8 = (F % 8 == 0) && (f_pb == 8);
   for (long r = r0 + tid; r < r1; r += nthr) {
-    int nd = ncls[r];
+    int nd = node_id[r];
     if (nd < lvl_lo || nd >= lvl_lo + lvl_nodes) continue;
     if (even_only && ((nd - lvl_lo) & 1)) continue;   // sibling trick
     int local = (nd - lvl_lo) >> even_only;
@@ -1646,7 +1644,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     if (g01) {
       hpack = ((unsigned long long)(gr != 0.f ? 1u : 0u) << 32) | 1ull;
     } else {
-      unsigned hfx = __float2uint_rn(hcls[r] * hscale);
+      unsigned hfx = __float2uint_rn(h[r] * hscale);
       hpack = ((unsigned long long)hfx << 32) | 1ull;
     }
     const int base = (local - ng0) * f_pb;
@@ -1751,8 +1749,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                               long n_rows, int F, int d, float hmax,
                               bool g_bounded, bool g01, hipStream_t s,
                               int KP = 1, long g_stride = 0,
-                              long hist_stride = 0, float gbias = 1.f,
-                              long node_stride = 0, long h_stride = 0) {
+                              long hist_stride = 0, float gbias = 1.f) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
@@ -1795,7 +1792,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                      lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
                      F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only,
                      ng_nodes, g01 ? 1 : 0, hs, gs, gbias, KP, g_stride,
-                     hist_stride, node_stride, h_stride);
+                     hist_stride);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -5869,18 +5866,14 @@ std::vector<torch::Tensor> gbt_fit_multi(
   auto raw = f0.to(optF).reshape({K, 1}).repeat({1, nt}).contiguous();
   auto g = torch::empty({K, nt}, optF);
   auto h = torch::empty({K, nt}, optF);
-  // per-class tree partitions: the K class-trees of a round are
-  // independent, so all K level-histograms run in ONE launch (per-class
-  // node/g/h/hist strides) with the block's row range cache-hot
-  auto node_id = torch::zeros({K, nt}, optI);
+  auto node_id = torch::zeros({nt}, optI);
   auto feats = torch::full({(long)T * K, NN}, -1, optI);
   auto thrs = torch::zeros({(long)T * K, NN}, optI);
   auto vals = torch::zeros({(long)T * K, NN}, optF);
   int max_lvl_nodes = 1 << (max_depth - 1);
-  long ccells = (long)max_lvl_nodes * F * 256 * 3;
-  long pcells = (long)std::max(1, max_lvl_nodes / 2) * F * 256 * 3;
-  auto hist = torch::empty({(long)K * ccells}, optF);
-  auto hist_prev = torch::empty({(long)K * pcells}, optF);
+  auto hist = torch::empty({(long)max_lvl_nodes * F * 256 * 3}, optF);
+  auto hist_prev = torch::empty({(long)std::max(1, max_lvl_nodes / 2) *
+                                 F * 256 * 3}, optF);
   auto gsum = torch::zeros({NN}, optF);
   auto hsum = torch::zeros({NN}, optF);
 
@@ -5893,27 +5886,20 @@ std::vector<torch::Tensor> gbt_fit_multi(
                        s, raw.data_ptr<float>(), y_tr.data_ptr<float>(),
                        g.data_ptr<float>(), h.data_ptr<float>(), nt,
                        (int)K);
-    HIP_CHECK(hipMemsetAsync(node_id.data_ptr(), 0, (size_t)K * nt * 4, s));
-    auto fptr = [&](int c) {
-      return feats.data_ptr<int>() + ((long)t * K + c) * NN;
-    };
-    auto tptr = [&](int c) {
-      return thrs.data_ptr<int>() + ((long)t * K + c) * NN;
-    };
-    auto vptr = [&](int c) {
-      return vals.data_ptr<float>() + ((long)t * K + c) * NN;
-    };
-    for (int d = 0; d < (int)max_depth; ++d) {
-      int lvl_lo = (1 << d) - 1;
-      int lvl_nodes = 1 << d;
-      int even_only = (d > 0) ? 1 : 0;
-      // all K class planes in one launch
-      launch_hist_level(Xtp, idxmap.data_ptr<int>(),
-                        node_id.data_ptr<int>(), g.data_ptr<float>(),
-                        h.data_ptr<float>(), hist.data_ptr<float>(), nt,
-                        n_rows, F, d, 0.25f, /*g_bounded=*/true, false, s,
-                        (int)K, nt, ccells, 1.f, nt, nt);
-      for (int c = 0; c < (int)K; ++c) {
+    for (int c = 0; c < (int)K; ++c) {
+      float* gc = g.data_ptr<float>() + (long)c * nt;
+      float* hc = h.data_ptr<float>() + (long)c * nt;
+      int* feat_t = feats.data_ptr<int>() + ((long)t * K + c) * NN;
+      int* thr_t = thrs.data_ptr<int>() + ((long)t * K + c) * NN;
+      float* val_t = vals.data_ptr<float>() + ((long)t * K + c) * NN;
+      for (int d = 0; d < (int)max_depth; ++d) {
+        int lvl_lo = (1 << d) - 1;
+        int lvl_nodes = 1 << d;
+        int even_only = (d > 0) ? 1 : 0;
+        launch_hist_level(Xtp, idxmap.data_ptr<int>(),
+                          node_id.data_ptr<int>(), gc, hc,
+                          hist.data_ptr<float>(), nt, n_rows, F, d, 0.25f,
+                          /*g_bounded=*/true, false, s);
         if (even_only) {
           int prev_lo = (1 << (d - 1)) - 1;
           long cells = (long)F * 256 * 3;
@@ -5921,36 +5907,26 @@ std::vector<torch::Tensor> gbt_fit_multi(
                                                  1024);
           hipLaunchKernelGGL(gbt_sibling_kernel,
                              dim3(gx, (unsigned)(lvl_nodes / 2)),
-                             dim3(256), 0, s,
-                             hist_prev.data_ptr<float>() + (long)c * pcells,
-                             hist.data_ptr<float>() + (long)c * ccells,
-                             fptr(c), prev_lo, F, lvl_nodes);
+                             dim3(256), 0, s, hist_prev.data_ptr<float>(),
+                             hist.data_ptr<float>(), feat_t, prev_lo, F,
+                             lvl_nodes);
         }
         hipLaunchKernelGGL(gbt_split_find_kernel, dim3(lvl_nodes),
-                           dim3(256), 0, s,
-                           hist.data_ptr<float>() + (long)c * ccells, F,
+                           dim3(256), 0, s, hist.data_ptr<float>(), F,
                            lvl_lo, (float)lam, (float)lrate,
-                           (int)min_leaf, 0, fptr(c), tptr(c), vptr(c),
+                           (int)min_leaf, 0, feat_t, thr_t, val_t,
                            1 << 30, 0u, 0);
-      }
-      if (d + 1 < (int)max_depth) {
-        long lvlcells = (long)lvl_nodes * F * 256 * 3;
-        for (int c = 0; c < (int)K; ++c)
-          hist_prev.narrow(0, (long)c * pcells, lvlcells)
-              .copy_(hist.narrow(0, (long)c * ccells, lvlcells), true);
-      }
-      for (int c = 0; c < (int)K; ++c)
+        if (d + 1 < (int)max_depth)
+          hist_prev.narrow(0, 0, (long)lvl_nodes * F * 256 * 3)
+              .copy_(hist.narrow(0, 0, (long)lvl_nodes * F * 256 * 3),
+                     true);
         hipLaunchKernelGGL(gbt_split_apply_kernel, dim3(blocks), dim3(256),
                            0, s, Xp, idxmap.data_ptr<int>(),
-                           node_id.data_ptr<int>() + (long)c * nt,
-                           fptr(c), tptr(c), nt, F,
+                           node_id.data_ptr<int>(), feat_t, thr_t, nt, F,
                            lvl_lo, lvl_lo + lvl_nodes);
-      HIP_CHECK(hipGetLastError());
-    }
-    int bl_lo = (1 << max_depth) - 1;
-    for (int c = 0; c < (int)K; ++c) {
-      float* gc = g.data_ptr<float>() + (long)c * nt;
-      float* hc = h.data_ptr<float>() + (long)c * nt;
+        HIP_CHECK(hipGetLastError());
+      }
+      int bl_lo = (1 << max_depth) - 1;
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
       {
@@ -5959,19 +5935,18 @@ std::vector<torch::Tensor> gbt_fit_multi(
         hipLaunchKernelGGL(gbt_node_sums_kernel,
                            dim3(fix ? 1 : blocks,
                                 (unsigned)(((NN - bl_lo) + 127) / 128)), dim3(256), 0, s,
-                           node_id.data_ptr<int>() + (long)c * nt, gc, hc,
+                           node_id.data_ptr<int>(), gc, hc,
                            gsum.data_ptr<float>(), hsum.data_ptr<float>(),
                            nt, bl_lo, NN, fsc, fsc, 1.f);
       }
       hipLaunchKernelGGL(gbt_leaf_vals_kernel,
                          dim3(((NN - bl_lo) + 255) / 256), dim3(256), 0, s,
                          gsum.data_ptr<float>(), hsum.data_ptr<float>(),
-                         vptr(c), fptr(c), (float)lam, (float)lrate, bl_lo,
+                         val_t, feat_t, (float)lam, (float)lrate, bl_lo,
                          NN);
       hipLaunchKernelGGL(gbt_leaf_apply_c_kernel, dim3(blocks), dim3(256),
                          0, s, raw.data_ptr<float>() + (long)c * nt,
-                         node_id.data_ptr<int>() + (long)c * nt, vptr(c),
-                         nt);
+                         node_id.data_ptr<int>(), val_t, nt);
       HIP_CHECK(hipGetLastError());
     }
   }
