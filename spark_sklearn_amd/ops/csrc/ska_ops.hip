@@ -1577,6 +1577,8 @@ __global__ void gbt_bin_kernel(const float* __restrict__ X,
 // flush using the exact count) -- integer atomics commute exactly, so
 // tree structures are bit-stable run to run.  gscale == 0 keeps the
 // default float-atomic g (faster: 12 B/cell vs 16 B).
+template <int MULTI>    // 0: single plane, loop-free (the hot path: a
+                        // runtime class loop cost ~30% at KP == 1)
 __global__ __launch_bounds__(512)
 void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      const int* __restrict__ idxmap,          // [nt]
@@ -1613,9 +1615,9 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   // hist + c*hist_stride) with the block's row range cache-hot across
   // classes -- replaces K separate launches (the digits-style K=10
   // forests were launch-bound, and the row-metadata HBM re-read was xK)
-  for (int cls = 0; cls < KP; ++cls) {
-  const float* gcls = g + (long)cls * g_stride;
-  float* hist_c = hist + (long)cls * hist_stride;
+  for (int cls = 0; cls < (MULTI ? KP : 1); ++cls) {
+  const float* gcls = MULTI ? g + (long)cls * g_stride : g;
+  float* hist_c = MULTI ? hist + (long)cls * hist_stride : hist;
   if (gfix) {
     for (int i = tid; i < 2 * cellsg; i += nthr) lhc[i] = 0ull;
   } else if (g01) {
@@ -1694,7 +1696,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     atomicAdd(cell + 1, hv);
     atomicAdd(cell + 2, (float)cnt);
   }
-  if (KP > 1) __syncthreads();   // LDS reuse across class planes
+  if (MULTI) __syncthreads();    // LDS reuse across class planes
   }
 }
 
@@ -1783,16 +1785,20 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
       hist, 0,
       ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
        (size_t)lvl_nodes * F * 256 * 3) * 4, s));
-  HIP_CHECK(hipFuncSetAttribute(
-      reinterpret_cast<const void*>(gbt_hist_kernel),
-      hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
-  hipLaunchKernelGGL(gbt_hist_kernel,
-                     dim3((unsigned)rowsegs, (unsigned)fchunks,
-                          (unsigned)ngroups), dim3(512),
-                     lsm, s, Xtp, idxmap, node_id, g, h, hist, nt, n_rows,
-                     F, lvl_lo, lvl_nodes, f_pb, rows_pb, even_only,
-                     ng_nodes, g01 ? 1 : 0, hs, gs, gbias, KP, g_stride,
-                     hist_stride);
+  auto hl = [&](auto kern) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(kern),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
+    hipLaunchKernelGGL(kern,
+                       dim3((unsigned)rowsegs, (unsigned)fchunks,
+                            (unsigned)ngroups), dim3(512),
+                       lsm, s, Xtp, idxmap, node_id, g, h, hist, nt,
+                       n_rows, F, lvl_lo, lvl_nodes, f_pb, rows_pb,
+                       even_only, ng_nodes, g01 ? 1 : 0, hs, gs, gbias,
+                       KP, g_stride, hist_stride);
+  };
+  if (KP > 1) hl(gbt_hist_kernel<1>);
+  else hl(gbt_hist_kernel<0>);
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
