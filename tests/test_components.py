@@ -434,3 +434,30 @@ def test_resume_log_header_mismatch_refused(tmp_path):
                            {"C": [0.1, 1.0]}, cv=3)
     with pytest.raises(ValueError, match="resume_from_log refused"):
         gs4.fit(X1, y1)
+
+
+def test_keyed_estimator_with_rf():
+    """KeyedEstimator's generic per-key path works with the RF family."""
+    import pandas as pd
+
+    from spark_sklearn_amd import KeyedEstimator
+    from spark_sklearn_amd.models import RandomForestClassifier
+
+    rng = np.random.default_rng(0)
+    rows = []
+    for key in ("a", "b"):
+        X = rng.standard_normal((200, 6))
+        y = (X[:, 0] > 0).astype(int)
+        for i in range(200):
+            rows.append({"k": key, "features": X[i], "label": y[i]})
+    df = pd.DataFrame(rows)
+    ke = KeyedEstimator(
+        sklearnEstimator=RandomForestClassifier(n_estimators=5,
+                                                max_depth=4,
+                                                random_state=0),
+        keyCols=["k"], xCol="features", yCol="label")
+    model = ke.fit(df)
+    out = model.transform(df)
+    assert "prediction" in out.columns
+    acc = (out["prediction"] == df["label"]).mean()
+    assert acc > 0.8
