@@ -36,15 +36,19 @@ class HipContext:
 
         if dtype != "bfloat16":
             raise ValueError(
-                "dtype='bfloat16' is the only resident dtype in round 1 "
-                "(the MFMA fit kernels are bf16 with fp32 accumulation); "
-                "fp8 residency is on the roadmap")
+                "dtype='bfloat16' is the only resident dtype (the MFMA "
+                "fit kernels are bf16 with fp32 accumulation); an fp8 "
+                "opt-in is a roadmap item -- see ROADMAP.md and the "
+                "round-2 ablation study for why it is not the default")
         if n_streams != 1:
             raise ValueError(
-                "n_streams is reserved: the native runner currently "
-                "drives one compute stream per GPU (copies are async on "
-                "the same stream); multi-stream fold overlap is on the "
-                "roadmap")
+                "n_streams is reserved: one compute stream per GPU. "
+                "Multi-stream overlap was MEASURED in round 2 (the "
+                "pass-pipelined v17 epoch, profiles/"
+                "r02_flagship_ablation.md) and lost 23% -- the big-LDS "
+                "fwd/bwd blocks cannot co-reside on a CU, so concurrent "
+                "kernels partition the chip. Honest single value until "
+                "a co-residable design exists")
         if not (1 <= int(candidate_batch) <= 256):
             raise ValueError("candidate_batch must be in [1, 256]")
         if partition not in ("static", "dynamic"):
