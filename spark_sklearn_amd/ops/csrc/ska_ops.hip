@@ -1643,8 +1643,14 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
     if (gfix) gq = (unsigned long long)((double)(gr + gbias) *
                                         (double)gscale + 0.5);
     unsigned long long hpack;
-    if (g01) {
+    if (g01 == 1) {
       hpack = ((unsigned long long)(gr != 0.f ? 1u : 0u) << 32) | 1ull;
+    } else if (g01 == 2) {
+      // (g+1) in [0,2] at 2^-13; h in [0,0.25] at 2^-11; count 14 bits
+      unsigned gq28 = __float2uint_rn((gr + 1.f) * 8192.f);
+      unsigned hq22 = __float2uint_rn(h[r] * 2048.f);
+      hpack = ((unsigned long long)gq28 << 36) |
+              ((unsigned long long)hq22 << 14) | 1ull;
     } else {
       unsigned hfx = __float2uint_rn(h[r] * hscale);
       hpack = ((unsigned long long)hfx << 32) | 1ull;
@@ -1677,12 +1683,18 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   for (int i = tid; i < cellsg; i += nthr) {
     unsigned long long p = lhc[i];
     if (p == 0ull) continue;          // count==0 -> cell never touched
-    unsigned cnt = (unsigned)(p & 0xffffffffull);
     float hv, gv;
-    if (g01) {
+    unsigned cnt;
+    if (g01 == 2) {
+      cnt = (unsigned)(p & 0x3fffull);
+      hv = (float)((p >> 14) & 0x3fffffull) * (1.f / 2048.f);
+      gv = (float)(p >> 36) * (1.f / 8192.f) - (float)cnt;
+    } else if (g01 == 1) {
+      cnt = (unsigned)(p & 0xffffffffull);
       gv = (float)(p >> 32);          // integer class-1 count
       hv = (float)cnt;                // h == 1 per row
     } else {
+      cnt = (unsigned)(p & 0xffffffffull);
       hv = (float)(p >> 32) * hinv;
       gv = gfix ? (float)((double)lgq[i] / (double)gscale -
                           (double)cnt * (double)gbias)
@@ -1756,8 +1768,24 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
   bool gfix = g_bounded && det_gbt_mode();
-  if (gfix) g01 = false;              // det mode owns the packing
-  int cell_b = gfix ? 16 : (g01 ? 8 : 12);
+  // cell mode: 1 = RF 0/1-g single-atomic; 2 = bounded-g (logistic/
+  // softmax) fixed-point single-atomic; 0 = float g channel.
+  // mode 2's packed fields (count 14 b, h-sum 22 b at 2^11 scale) bound
+  // the rows a block may scan to 8192 -- the block cap below can push
+  // rows_pb beyond that at large nt, so mode 2 only engages when the
+  // resulting segmentation stays within the field widths.
+  int mode = gfix ? 0 : (g01 ? 1 : (g_bounded ? 2 : 0));
+  if (mode == 2) {
+    long fp2 = std::min(16, std::max(1, (48 * 1024) /
+                                     (8 * 256 * std::min(lvl_eff, 16))));
+    while (F % fp2 != 0) fp2 /= 2;
+    long fch2 = F / fp2;
+    long rs2 = std::max<long>(1, std::min<long>(2048 / fch2,
+                                                (nt + 8191) / 8192));
+    long rpb2 = (nt + rs2 - 1) / rs2;
+    if (rpb2 > 8192) mode = 0;
+  }
+  int cell_b = gfix ? 16 : (mode ? 8 : 12);
   // feature-major bins have no per-row vector-load constraint: f_pb up
   // to 16 halves the (node_id, idxmap, g, h) re-reads per level vs 8
   int f_pb = std::min(16, std::max(1, (48 * 1024) /
@@ -1794,7 +1822,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                             (unsigned)ngroups), dim3(512),
                        lsm, s, Xtp, idxmap, node_id, g, h, hist, nt,
                        n_rows, F, lvl_lo, lvl_nodes, f_pb, rows_pb,
-                       even_only, ng_nodes, g01 ? 1 : 0, hs, gs, gbias,
+                       even_only, ng_nodes, mode, hs, gs, gbias,
                        KP, g_stride, hist_stride);
   };
   if (KP > 1) hl(gbt_hist_kernel<1>);
