@@ -8,10 +8,15 @@ MI355X-native communication plan of SURVEY.md section 5.8:
   out over the root's 7 links).  This replaces Spark's TorrentBroadcast.
 * task dispatch -- needs no network at all on one node: tasks are
   partitioned statically, candidate-round-robin, so every rank derives its
-  share from the task list deterministically (replaces the Spark scheduler).
-* RC2 -- fold scores are fixed-width 6-float records summed into a dense
-  (n_tasks, 6) table with one all-reduce (each rank fills only its rows;
-  zeros elsewhere), so every rank assembles identical ``cv_results_``.
+  share from the task list deterministically (replaces the Spark
+  scheduler).  ``partition="dynamic"`` switches to a bounded work queue:
+  whole-candidate blocks pulled from the rendezvous store's atomic
+  counter, balancing skewed per-candidate costs (round 2).
+* RC2 -- fold scores are fixed-width records summed into a dense score
+  table with one all-reduce (each rank fills only its rows; zeros
+  elsewhere), so every rank assembles identical ``cv_results_``; an
+  object all-gather carries the error TEXT of failed fits so
+  FitFailedWarning keeps its diagnostic payload on every rank (round 2).
   This replaces ``collect()``.
 * refit -- rank 0 refits on the full resident matrix (it already holds all
   of X after RC1; RC4 in SURVEY.md needs no collective) and broadcasts the
