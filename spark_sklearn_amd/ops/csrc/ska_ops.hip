@@ -1577,8 +1577,12 @@ __global__ void gbt_bin_kernel(const float* __restrict__ X,
 // flush using the exact count) -- integer atomics commute exactly, so
 // tree structures are bit-stable run to run.  gscale == 0 keeps the
 // default float-atomic g (faster: 12 B/cell vs 16 B).
-template <int MULTI>    // 0: single plane, loop-free (the hot path: a
+template <int MULTI,    // 0: single plane, loop-free (the hot path: a
                         // runtime class loop cost ~30% at KP == 1)
+          int TBT>      // 1: TREE-BATCHED small-n mode -- grid.z carries
+                        // (tree, node-group): the launch-bound digits-
+                        // style forests fold a whole tree chunk's level
+                        // into ONE launch (per-tree buffer strides below)
 __global__ __launch_bounds__(512)
 void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      const int* __restrict__ idxmap,          // [nt]
@@ -1590,7 +1594,8 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      int lvl_nodes, int f_pb, long rows_pb, int even_only,
                      int ng_nodes, int g01, float hscale, float gscale,
                      float gbias, int KP, long g_stride,
-                     long hist_stride) {
+                     long hist_stride, int n_tgroups, long idx_tstride,
+                     long g_tstride, long hist_tstride) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1599,7 +1604,15 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   // ng_nodes along grid.z (each z-block scans the rows, histogramming
   // only its nodes) -- this is what lifts the device tree-depth cap
   // from 6 to 10.
-  const int ng0 = (int)blockIdx.z * ng_nodes;
+  if (TBT) {
+    const int t = (int)blockIdx.z / n_tgroups;
+    idxmap += (long)t * idx_tstride;    // 0 when all trees share the map
+    node_id += (long)t * nt;
+    g += (long)t * g_tstride;
+    hist += (long)t * hist_tstride;
+  }
+  const int ng0 =
+      (TBT ? ((int)blockIdx.z % n_tgroups) : (int)blockIdx.z) * ng_nodes;
   const int cellsg = ng_nodes * f_pb * 256;
   const bool gfix = gscale > 0.f;
   unsigned long long* lhc = reinterpret_cast<unsigned long long*>(smem);
@@ -1764,7 +1777,9 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                               bool g_bounded, bool g01, hipStream_t s,
                               int KP = 1, long g_stride = 0,
                               long hist_stride = 0, float gbias = 1.f,
-                              bool packed_ok = false) {
+                              bool packed_ok = false, int n_trees = 1,
+                              long idx_tstride = 0, long g_tstride = 0,
+                              long hist_tstride = 0) {
   int lvl_lo = (1 << d) - 1;
   int lvl_nodes = 1 << d;
   int lvl_eff = (d > 0) ? (lvl_nodes >> 1) : lvl_nodes;  // even nodes only
@@ -1815,24 +1830,41 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
     hs *= 2.f;
   // biased g in [0,2]: keep nt*(2*gs + 1) well under 2^62
   float gs = gfix ? det_fix_scale(nt, gbias) : 0.f;
-  HIP_CHECK(hipMemsetAsync(
-      hist, 0,
-      ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
-       (size_t)lvl_nodes * F * 256 * 3) * 4, s));
+  long lvlcells = (long)lvl_nodes * F * 256 * 3;
+  if (n_trees > 1) {
+    // zero exactly the level's cells of EVERY (tree, class) plane with
+    // one strided memset (planes are contiguous: hist_tstride == KP *
+    // hist_stride)
+    long planes = (long)n_trees * std::max(1, KP);
+    long pitch = (KP > 1) ? hist_stride : hist_tstride;
+    HIP_CHECK(hipMemset2DAsync(hist, (size_t)pitch * 4, 0,
+                               (size_t)lvlcells * 4, (size_t)planes, s));
+  } else {
+    HIP_CHECK(hipMemsetAsync(
+        hist, 0,
+        ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
+         (size_t)lvlcells) * 4, s));
+  }
   auto hl = [&](auto kern) {
     HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(kern),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lsm));
     hipLaunchKernelGGL(kern,
                        dim3((unsigned)rowsegs, (unsigned)fchunks,
-                            (unsigned)ngroups), dim3(512),
+                            (unsigned)(ngroups * n_trees)), dim3(512),
                        lsm, s, Xtp, idxmap, node_id, g, h, hist, nt,
                        n_rows, F, lvl_lo, lvl_nodes, f_pb, rows_pb,
                        even_only, ng_nodes, mode, hs, gs, gbias,
-                       KP, g_stride, hist_stride);
+                       KP, g_stride, hist_stride, ngroups, idx_tstride,
+                       g_tstride, hist_tstride);
   };
-  if (KP > 1) hl(gbt_hist_kernel<1>);
-  else hl(gbt_hist_kernel<0>);
+  if (n_trees > 1) {
+    if (KP > 1) hl(gbt_hist_kernel<1, 1>);
+    else hl(gbt_hist_kernel<0, 1>);
+  } else {
+    if (KP > 1) hl(gbt_hist_kernel<1, 0>);
+    else hl(gbt_hist_kernel<0, 0>);
+  }
 }
 
 // sibling subtraction: right-child histogram = parent - left.  Parents
@@ -1941,18 +1973,15 @@ __device__ unsigned rf_feat_threshold(int F, int kfeat, unsigned fseed,
   return t;
 }
 
-__global__ __launch_bounds__(256)
-void gbt_split_find_kernel(const float* __restrict__ hist, int F,
-                           int lvl_lo, float lam, float lr, int min_leaf,
-                           int force_leaf, int* __restrict__ feat,
-                           int* __restrict__ thr, float* __restrict__ val,
-                           int kfeat, unsigned fseed, int etr) {
-  const int nd_local = blockIdx.x;
-  const int nd = lvl_lo + nd_local;
+__device__ __forceinline__
+void gbt_split_find_body(const float* __restrict__ nh, int F, int nd,
+                         float lam, float lr, int min_leaf,
+                         int force_leaf, int* __restrict__ feat,
+                         int* __restrict__ thr, float* __restrict__ val,
+                         int kfeat, unsigned fseed, int etr) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const float* nh = hist + (long)nd_local * F * 256 * 3;
 
   __shared__ float sGain[4];
   __shared__ int sFt[4], sTt[4];
@@ -2065,6 +2094,40 @@ void gbt_split_find_kernel(const float* __restrict__ hist, int F,
       val[nd] = 0.f;
     }
   }
+}
+
+__global__ __launch_bounds__(256)
+void gbt_split_find_kernel(const float* __restrict__ hist, int F,
+                           int lvl_lo, float lam, float lr, int min_leaf,
+                           int force_leaf, int* __restrict__ feat,
+                           int* __restrict__ thr, float* __restrict__ val,
+                           int kfeat, unsigned fseed, int etr) {
+  const int nd_local = blockIdx.x;
+  gbt_split_find_body(hist + (long)nd_local * F * 256 * 3, F,
+                      lvl_lo + nd_local, lam, lr, min_leaf, force_leaf,
+                      feat, thr, val, kfeat, fseed, etr);
+}
+
+// tree-batched split find (small-n RF): grid.y = tree within the chunk;
+// the per-tree feature-selection seed is recomputed from the forest seed
+// exactly as the host loop does, so trees match the per-tree path (and
+// the numpy reference) bit for bit.
+__global__ __launch_bounds__(256)
+void rf_bsplit_find_kernel(const float* __restrict__ hist,
+                           long tree_hstride, int F, int lvl_lo,
+                           int min_leaf, int* __restrict__ feat,
+                           int* __restrict__ thr, float* __restrict__ val,
+                           long plane_tstride, int kfeat, unsigned seed,
+                           int etr, int t0) {
+  const int tl = blockIdx.y;
+  const unsigned st =
+      rf_hash(seed ^ (unsigned)((t0 + tl) * 0x9e3779b9u));
+  const unsigned fs = rf_hash(st ^ 0x85ebca6bu);
+  const long po = (long)tl * plane_tstride;
+  gbt_split_find_body(
+      hist + (long)tl * tree_hstride + (long)blockIdx.x * F * 256 * 3, F,
+      lvl_lo + (int)blockIdx.x, 0.f, -1.f, min_leaf, 0, feat + po,
+      thr + po, val + po, kfeat, fs, etr);
 }
 
 __global__ void gbt_split_apply_kernel(const unsigned char* __restrict__ Xb,
@@ -6016,15 +6079,13 @@ torch::Tensor gbt_metric_multi(torch::Tensor raws, torch::Tensor y,
 // walk each plane unchanged; no-split nodes get per-class leaf
 // probabilities G_c / C.
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256)
-void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
-                                int K, int F, int lvl_lo, int min_leaf,
-                                int kfeat, unsigned fseed, int etr,
-                                int* __restrict__ feats,
-                                int* __restrict__ thrs,
-                                float* __restrict__ vals, long NN) {
-  const int nd_local = blockIdx.x;
-  const int nd = lvl_lo + nd_local;
+__device__ __forceinline__
+void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
+                              int K, int F, int nd, int min_leaf,
+                              int kfeat, unsigned fseed, int etr,
+                              int* __restrict__ feats,
+                              int* __restrict__ thrs,
+                              float* __restrict__ vals, long NN) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -6044,8 +6105,7 @@ void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
 
   for (int f = wave; f < F; f += 4) {
     bool active = (kfeat >= F) || (rf_feat_hash(fseed, nd, f) < fthr);
-    const float* fh0 =
-        hist + (long)nd_local * F * 256 * 3 + (long)f * 256 * 3;
+    const float* fh0 = ndh + (long)f * 256 * 3;
     float c4[4];
     float lc = 0.f;
 #pragma unroll
@@ -6178,6 +6238,186 @@ void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
   }
 }
 
+__global__ __launch_bounds__(256)
+void rf_split_find_multi_kernel(const float* __restrict__ hist, long cstride,
+                                int K, int F, int lvl_lo, int min_leaf,
+                                int kfeat, unsigned fseed, int etr,
+                                int* __restrict__ feats,
+                                int* __restrict__ thrs,
+                                float* __restrict__ vals, long NN) {
+  const int nd_local = blockIdx.x;
+  rf_split_find_multi_body(hist + (long)nd_local * F * 256 * 3, cstride,
+                           K, F, lvl_lo + nd_local, min_leaf, kfeat,
+                           fseed, etr, feats, thrs, vals, NN);
+}
+
+// tree-batched multiclass split find: grid.y = tree within the chunk
+// (hist layout [tree][K][nodes][F][256][3]; outputs offset by the tree's
+// K*NN plane block)
+__global__ __launch_bounds__(256)
+void rf_bsplit_find_multi_kernel(const float* __restrict__ hist,
+                                 long tree_hstride, long cstride, int K,
+                                 int F, int lvl_lo, int min_leaf,
+                                 int kfeat, unsigned seed, int etr,
+                                 int* __restrict__ feats,
+                                 int* __restrict__ thrs,
+                                 float* __restrict__ vals, long NN,
+                                 int t0) {
+  const int tl = blockIdx.y;
+  const unsigned st =
+      rf_hash(seed ^ (unsigned)((t0 + tl) * 0x9e3779b9u));
+  const unsigned fs = rf_hash(st ^ 0x85ebca6bu);
+  const long po = (long)tl * (long)K * NN;
+  rf_split_find_multi_body(
+      hist + (long)tl * tree_hstride + (long)blockIdx.x * F * 256 * 3,
+      cstride, K, F, lvl_lo + (int)blockIdx.x, min_leaf, kfeat, fs, etr,
+      feats + po, thrs + po, vals + po, NN);
+}
+
+// ---------------------------------------------------------------------------
+// tree-batched small-n RF pieces: for launch-bound forests (digits-scale
+// nt) a chunk of TB whole trees advances one level per kernel launch.
+// grid.y carries the tree (bootstrap/planes/apply), grid.z carries
+// (tree, class) (sibling/sums) or (tree, node-group) (histograms, see
+// gbt_hist_kernel<MULTI, 1>).
+// ---------------------------------------------------------------------------
+__global__ void rf_bbootstrap_kernel(const int* __restrict__ idx_in,
+                                     const float* __restrict__ y_in,
+                                     int* __restrict__ idx_out,
+                                     float* __restrict__ y_out, long nt,
+                                     unsigned seed, int t0) {
+  const int tl = blockIdx.y;
+  const unsigned st =
+      rf_hash(seed ^ (unsigned)((t0 + tl) * 0x9e3779b9u));
+  idx_out += (long)tl * nt;
+  y_out += (long)tl * nt;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  for (; i < nt; i += (long)gridDim.x * blockDim.x) {
+    unsigned r = rf_hash(st ^ (unsigned)i);
+    long p = (long)(((unsigned long long)r * (unsigned long long)nt) >> 32);
+    idx_out[i] = idx_in[p];
+    y_out[i] = y_in[p];
+  }
+}
+
+__global__ void rf_bplanes_kernel(const float* __restrict__ y_boot,
+                                  float* __restrict__ G, long nt, int K) {
+  const int tl = blockIdx.y;
+  y_boot += (long)tl * nt;
+  G += (long)tl * (long)K * nt;
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  for (; i < nt; i += (long)gridDim.x * blockDim.x) {
+    int cls = (int)y_boot[i];
+    for (int c = 0; c < K; ++c) G[(long)c * nt + i] = (c == cls) ? 1.f : 0.f;
+  }
+}
+
+// sibling subtraction over every (tree, class) plane in one launch:
+// grid.z = tl * KP + c; prev/cur are SAME-SHAPE alternating full-level
+// buffers ([TB][KP][ccells]), so both offset by z * ccells
+__global__ void rf_bsibling_kernel(const float* __restrict__ prev,
+                                   float* __restrict__ cur,
+                                   const int* __restrict__ feat,
+                                   int prev_lo, int F, int lvl_nodes,
+                                   long ccells, long plane_tstride,
+                                   int KP) {
+  const int z = blockIdx.z;
+  prev += (long)z * ccells;
+  cur += (long)z * ccells;
+  feat += (long)(z / KP) * plane_tstride;     // plane 0 of the tree
+  int pair = blockIdx.y;
+  long cells = (long)F * 256 * 3;
+  long base_l = (long)(2 * pair) * cells;
+  long base_r = (long)(2 * pair + 1) * cells;
+  long base_p = (long)pair * cells;
+  bool split = feat[prev_lo + pair] >= 0;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < cells;
+       i += (long)gridDim.x * blockDim.x) {
+    cur[base_r + i] = split ? prev[base_p + i] - cur[base_l + i] : 0.f;
+  }
+}
+
+__global__ void rf_bapply_kernel(const unsigned char* __restrict__ Xb,
+                                 const int* __restrict__ idxmap,
+                                 int* __restrict__ node_id,
+                                 const int* __restrict__ feat,
+                                 const int* __restrict__ thr, long nt,
+                                 int F, int lvl_lo, int lvl_hi,
+                                 long idx_tstride, long plane_tstride) {
+  const int tl = blockIdx.y;
+  idxmap += (long)tl * idx_tstride;
+  node_id += (long)tl * nt;
+  feat += (long)tl * plane_tstride;
+  thr += (long)tl * plane_tstride;
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    int nd = node_id[r];
+    if (nd < lvl_lo || nd >= lvl_hi) continue;
+    int f = feat[nd];
+    if (f < 0) continue;
+    int bin = Xb[(long)idxmap[r] * F + f];
+    node_id[r] = 2 * nd + 1 + (bin > thr[nd] ? 1 : 0);
+  }
+}
+
+// bottom-level (g, count) sums per (tree, class): h == 1 in RF, so the
+// count IS the h channel.  grid (row blocks, 128-node chunks, tl*KP+c).
+__global__ void rf_bsums_kernel(const int* __restrict__ node_id,
+                                const float* __restrict__ g,
+                                float* __restrict__ gsum,
+                                float* __restrict__ hsum, long nt,
+                                int lvl_lo_full, int lvl_hi_full,
+                                long g_tstride, long g_cstride, int KP,
+                                int NN) {
+  __shared__ float lg[128], lc[128];
+  const int z = blockIdx.z;
+  node_id += (long)(z / KP) * nt;
+  g += (long)(z / KP) * g_tstride + (long)(z % KP) * g_cstride;
+  gsum += (long)z * NN;
+  hsum += (long)z * NN;
+  const int lvl_lo = lvl_lo_full + (int)blockIdx.y * 128;
+  const int lvl_hi = min(lvl_hi_full, lvl_lo + 128);
+  int nn = lvl_hi - lvl_lo;
+  for (int i = threadIdx.x; i < nn; i += blockDim.x) {
+    lg[i] = 0.f;
+    lc[i] = 0.f;
+  }
+  __syncthreads();
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    int nd = node_id[r];
+    if (nd < lvl_lo || nd >= lvl_hi) continue;
+    atomicAdd(&lg[nd - lvl_lo], g[r]);
+    atomicAdd(&lc[nd - lvl_lo], 1.f);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < nn; i += blockDim.x) {
+    if (lc[i] != 0.f) {
+      atomicAdd(&gsum[lvl_lo + i], lg[i]);
+      atomicAdd(&hsum[lvl_lo + i], lc[i]);
+    }
+  }
+}
+
+// bottom-level leaf values for every (tree, class) plane in one launch;
+// vals/feats offset per plane (z = tl*KP + c matches the [T*KP][NN]
+// output layout when the host pre-offsets by t0*KP*NN)
+__global__ void rf_bleafs_kernel(const float* __restrict__ gsum,
+                                 const float* __restrict__ hsum,
+                                 float* __restrict__ vals,
+                                 int* __restrict__ feats, int lvl_lo,
+                                 int lvl_hi, int NN) {
+  const int z = blockIdx.y;
+  const long po = (long)z * NN;
+  int nd = lvl_lo + blockIdx.x * blockDim.x + threadIdx.x;
+  if (nd >= lvl_hi) return;
+  float h = hsum[po + nd];
+  if (h > 0.f) {
+    feats[po + nd] = -1;
+    vals[po + nd] = gsum[po + nd] / (h + 1e-12f);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // rf_fit: RandomForest fit over the shared binned matrix.  Trees are
 // independent: per tree a stateless-hash bootstrap resample of the train
@@ -6209,12 +6449,145 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
 
   auto optF = y_tr.options().dtype(torch::kFloat32);
   auto optI = y_tr.options().dtype(torch::kInt32);
-  auto node_id = torch::zeros({nt}, optI);
   auto feats = torch::full({(long)T * KP, (long)NN}, -1, optI);
   auto thrs = torch::zeros({(long)T * KP, (long)NN}, optI);
   auto vals = torch::zeros({(long)T * KP, (long)NN}, optF);
   int max_lvl_nodes = 1 << (max_depth - 1);
   long ccells = (long)max_lvl_nodes * F * 256 * 3;
+
+  hipStream_t s = cur_stream();
+  const unsigned char* Xp = Xb.data_ptr<unsigned char>();
+  const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
+  long n_rows = Xb.size(0);
+  int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+  int bl_lo = (1 << max_depth) - 1;
+
+  if (nt <= 65536) {
+    // ---- tree-batched small-n path (digits-scale forests are LAUNCH
+    // bound: ~4 launches per tree-LEVEL here vs ~4 launches per tree-
+    // level PER TREE in the large-n loop below).  A chunk of TB whole
+    // trees shares every launch; per-tree seeds are recomputed inside
+    // the kernels, so trees are identical to the per-tree path.  Two
+    // full-size histogram buffers alternate as (current, parent) so the
+    // sibling subtraction needs no parent copy.  The coalescing sort of
+    // the large-n path is skipped: at <=64k rows the binned matrix is
+    // L2-resident and gather order does not matter. ----
+    double per_tree = 2.0 * (double)KP * (double)ccells * 4.0;
+    int TB = (int)std::max<long>(
+        1, std::min<long>({(long)T, 2048L,
+                           (long)(6.0e9 / std::max(per_tree, 1.0))}));
+    auto hista = torch::empty({(long)TB * KP * ccells}, optF);
+    auto histb = torch::empty({(long)TB * KP * ccells}, optF);
+    auto node_id = torch::empty({(long)TB * nt}, optI);
+    auto gsum = torch::zeros({(long)TB * KP, (long)NN}, optF);
+    auto hsum = torch::zeros({(long)TB * KP, (long)NN}, optF);
+    auto ones = torch::ones({nt}, optF);
+    torch::Tensor idx_b, y_b, G;
+    long idx_ts = 0, g_ts = 0;
+    if (bootstrap) {
+      idx_b = torch::empty({(long)TB * nt}, optI);
+      y_b = torch::empty({(long)TB * nt}, optF);
+      idx_ts = nt;
+      g_ts = (KP > 1) ? (long)KP * nt : nt;
+    }
+    if (KP > 1) {
+      G = torch::empty({(long)(bootstrap ? TB : 1) * KP, nt}, optF);
+      if (!bootstrap)      // identical planes for every tree: share them
+        hipLaunchKernelGGL(rf_planes_kernel, dim3(blocks), dim3(256), 0,
+                           s, y_tr.data_ptr<float>(), G.data_ptr<float>(),
+                           nt, KP);
+    }
+    for (int t0 = 0; t0 < T; t0 += TB) {
+      int tb = std::min(TB, T - t0);
+      if (bootstrap) {
+        hipLaunchKernelGGL(rf_bbootstrap_kernel,
+                           dim3(blocks, (unsigned)tb), dim3(256), 0, s,
+                           idxmap0.data_ptr<int>(),
+                           y_tr.data_ptr<float>(), idx_b.data_ptr<int>(),
+                           y_b.data_ptr<float>(), nt, (unsigned)seed, t0);
+        if (KP > 1)
+          hipLaunchKernelGGL(rf_bplanes_kernel,
+                             dim3(blocks, (unsigned)tb), dim3(256), 0, s,
+                             y_b.data_ptr<float>(), G.data_ptr<float>(),
+                             nt, KP);
+      }
+      HIP_CHECK(hipMemsetAsync(node_id.data_ptr(), 0,
+                               (size_t)tb * nt * 4, s));
+      const int* idxp = bootstrap ? idx_b.data_ptr<int>()
+                                  : idxmap0.data_ptr<int>();
+      const float* gp = (KP > 1)
+                            ? G.data_ptr<float>()
+                            : (bootstrap ? y_b.data_ptr<float>()
+                                         : y_tr.data_ptr<float>());
+      int* feat_t = feats.data_ptr<int>() + (long)t0 * KP * NN;
+      int* thr_t = thrs.data_ptr<int>() + (long)t0 * KP * NN;
+      float* val_t = vals.data_ptr<float>() + (long)t0 * KP * NN;
+      for (int d = 0; d < (int)max_depth; ++d) {
+        int lvl_lo = (1 << d) - 1;
+        int lvl_nodes = 1 << d;
+        float* cur = (d & 1) ? histb.data_ptr<float>()
+                             : hista.data_ptr<float>();
+        float* prev = (d & 1) ? hista.data_ptr<float>()
+                              : histb.data_ptr<float>();
+        launch_hist_level(Xtp, idxp, node_id.data_ptr<int>(), gp,
+                          ones.data_ptr<float>(), cur, nt, n_rows, F, d,
+                          1.f, false, K != 0, s, KP, nt, ccells, 1.f,
+                          false, tb, idx_ts, g_ts, (long)KP * ccells);
+        if (d > 0) {
+          long cells = (long)F * 256 * 3;
+          unsigned gx =
+              (unsigned)std::min<long>((cells + 255) / 256, 256);
+          hipLaunchKernelGGL(rf_bsibling_kernel,
+                             dim3(gx, (unsigned)(lvl_nodes / 2),
+                                  (unsigned)(tb * KP)), dim3(256), 0, s,
+                             prev, cur, feat_t, (1 << (d - 1)) - 1, F,
+                             lvl_nodes, ccells, (long)KP * NN, KP);
+        }
+        if (KP > 1) {
+          hipLaunchKernelGGL(rf_bsplit_find_multi_kernel,
+                             dim3((unsigned)lvl_nodes, (unsigned)tb),
+                             dim3(256), 0, s, cur, (long)KP * ccells,
+                             ccells, KP, F, lvl_lo, (int)min_leaf, kfeat,
+                             (unsigned)seed, (int)splitter_random,
+                             feat_t, thr_t, val_t, (long)NN, t0);
+        } else {
+          hipLaunchKernelGGL(rf_bsplit_find_kernel,
+                             dim3((unsigned)lvl_nodes, (unsigned)tb),
+                             dim3(256), 0, s, cur, ccells, F, lvl_lo,
+                             (int)min_leaf, feat_t, thr_t, val_t,
+                             (long)KP * NN, kfeat, (unsigned)seed,
+                             (int)splitter_random, t0);
+        }
+        hipLaunchKernelGGL(rf_bapply_kernel,
+                           dim3(blocks, (unsigned)tb), dim3(256), 0, s,
+                           Xp, idxp, node_id.data_ptr<int>(), feat_t,
+                           thr_t, nt, F, lvl_lo, lvl_lo + lvl_nodes,
+                           idx_ts, (long)KP * NN);
+        HIP_CHECK(hipGetLastError());
+      }
+      HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0,
+                               (size_t)tb * KP * NN * 4, s));
+      HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0,
+                               (size_t)tb * KP * NN * 4, s));
+      hipLaunchKernelGGL(
+          rf_bsums_kernel,
+          dim3(blocks, (unsigned)(((NN - bl_lo) + 127) / 128),
+               (unsigned)(tb * KP)), dim3(256), 0, s,
+          node_id.data_ptr<int>(), gp, gsum.data_ptr<float>(),
+          hsum.data_ptr<float>(), nt, bl_lo, NN, g_ts,
+          (KP > 1) ? nt : 0L, KP, NN);
+      hipLaunchKernelGGL(rf_bleafs_kernel,
+                         dim3((unsigned)(((NN - bl_lo) + 255) / 256),
+                              (unsigned)(tb * KP)), dim3(256), 0, s,
+                         gsum.data_ptr<float>(), hsum.data_ptr<float>(),
+                         val_t, feat_t, bl_lo, NN, NN);
+      HIP_CHECK(hipGetLastError());
+    }
+    vals.mul_(1.0 / (double)T);
+    return {feats, thrs, vals};
+  }
+
+  auto node_id = torch::zeros({nt}, optI);
   auto hist = torch::empty({ccells * KP}, optF);
   auto hist_prev = torch::empty(
       {(long)std::max(1, max_lvl_nodes / 2) * F * 256 * 3 * KP}, optF);
@@ -6226,12 +6599,6 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
   auto y_boot = bootstrap ? torch::empty({nt}, optF) : y_tr;
   auto G = (KP > 1) ? torch::empty({(long)KP, nt}, optF)
                     : torch::empty({0}, optF);
-
-  hipStream_t s = cur_stream();
-  const unsigned char* Xp = Xb.data_ptr<unsigned char>();
-  const unsigned char* Xtp = Xbt.data_ptr<unsigned char>();
-  long n_rows = Xb.size(0);
-  int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
 
   for (int t = 0; t < T; ++t) {
     unsigned st = rf_hash((unsigned)seed ^ (unsigned)(t * 0x9e3779b9u));
@@ -6311,7 +6678,6 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                          lvl_lo, lvl_lo + lvl_nodes);
       HIP_CHECK(hipGetLastError());
     }
-    int bl_lo = (1 << max_depth) - 1;
     for (int c = 0; c < KP; ++c) {
       const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
                                  : y_u.data_ptr<float>();
