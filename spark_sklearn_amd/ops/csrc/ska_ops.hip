@@ -1595,7 +1595,7 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
                      int ng_nodes, int g01, float hscale, float gscale,
                      float gbias, int KP, long g_stride,
                      long hist_stride, int n_tgroups, long idx_tstride,
-                     long g_tstride, long hist_tstride) {
+                     long g_tstride, long hist_tstride, int store_flush) {
   extern __shared__ char smem[];
   // under sibling subtraction only EVEN nodes are histogrammed: LDS
   // covers lvl_nodes/2 node slots, doubling resident blocks at deep
@@ -1694,32 +1694,48 @@ void gbt_hist_kernel(const unsigned char* __restrict__ Xbt,  // [F][n]
   __syncthreads();
   const float hinv = 1.f / hscale;
   for (int i = tid; i < cellsg; i += nthr) {
+    int ndl = ((i / (256 * f_pb)) + ng0) << even_only;  // full-level slot
+    // the last node group can overrun the level (lvl_eff % ng_nodes):
+    // those LDS cells are necessarily empty, but the store flush below
+    // must not write them
+    if (ndl >= lvl_nodes) continue;
     unsigned long long p = lhc[i];
-    if (p == 0ull) continue;          // count==0 -> cell never touched
-    float hv, gv;
-    unsigned cnt;
-    if (g01 == 2) {
-      cnt = (unsigned)(p & 0x3fffull);
-      hv = (float)((p >> 14) & 0x3fffffull) * (1.f / 2048.f);
-      gv = (float)(p >> 36) * (1.f / 8192.f) - (float)cnt;
-    } else if (g01 == 1) {
-      cnt = (unsigned)(p & 0xffffffffull);
-      gv = (float)(p >> 32);          // integer class-1 count
-      hv = (float)cnt;                // h == 1 per row
-    } else {
-      cnt = (unsigned)(p & 0xffffffffull);
-      hv = (float)(p >> 32) * hinv;
-      gv = gfix ? (float)((double)lgq[i] / (double)gscale -
-                          (double)cnt * (double)gbias)
-                : lg[i];
+    // count==0 -> cell never touched.  store_flush (host-guaranteed: ONE
+    // block owns every global cell) writes zeros instead, replacing the
+    // pre-launch global memset -- the 2D fill was 27% of device time on
+    // launch-bound small-n forests.
+    if (p == 0ull && !store_flush) continue;
+    float hv = 0.f, gv = 0.f;
+    unsigned cnt = 0;
+    if (p != 0ull) {
+      if (g01 == 2) {
+        cnt = (unsigned)(p & 0x3fffull);
+        hv = (float)((p >> 14) & 0x3fffffull) * (1.f / 2048.f);
+        gv = (float)(p >> 36) * (1.f / 8192.f) - (float)cnt;
+      } else if (g01 == 1) {
+        cnt = (unsigned)(p & 0xffffffffull);
+        gv = (float)(p >> 32);          // integer class-1 count
+        hv = (float)cnt;                // h == 1 per row
+      } else {
+        cnt = (unsigned)(p & 0xffffffffull);
+        hv = (float)(p >> 32) * hinv;
+        gv = gfix ? (float)((double)lgq[i] / (double)gscale -
+                            (double)cnt * (double)gbias)
+                  : lg[i];
+      }
     }
     int bin = i & 255;
     int ff = (i >> 8) % f_pb;
-    int ndl = ((i / (256 * f_pb)) + ng0) << even_only;  // full-level slot
     float* cell = &hist_c[(((long)ndl * F + f0 + ff) * 256 + bin) * 3];
-    atomicAdd(cell, gv);
-    atomicAdd(cell + 1, hv);
-    atomicAdd(cell + 2, (float)cnt);
+    if (store_flush) {
+      cell[0] = gv;
+      cell[1] = hv;
+      cell[2] = (float)cnt;
+    } else {
+      atomicAdd(cell, gv);
+      atomicAdd(cell + 1, hv);
+      atomicAdd(cell + 2, (float)cnt);
+    }
   }
   if (MULTI) __syncthreads();    // LDS reuse across class planes
   }
@@ -1831,19 +1847,25 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
   // biased g in [0,2]: keep nt*(2*gs + 1) well under 2^62
   float gs = gfix ? det_fix_scale(nt, gbias) : 0.f;
   long lvlcells = (long)lvl_nodes * F * 256 * 3;
-  if (n_trees > 1) {
-    // zero exactly the level's cells of EVERY (tree, class) plane with
-    // one strided memset (planes are contiguous: hist_tstride == KP *
-    // hist_stride)
-    long planes = (long)n_trees * std::max(1, KP);
-    long pitch = (KP > 1) ? hist_stride : hist_tstride;
-    HIP_CHECK(hipMemset2DAsync(hist, (size_t)pitch * 4, 0,
-                               (size_t)lvlcells * 4, (size_t)planes, s));
-  } else {
-    HIP_CHECK(hipMemsetAsync(
-        hist, 0,
-        ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
-         (size_t)lvlcells) * 4, s));
+  // rowsegs == 1 means exactly ONE block owns every global cell: the
+  // flush can plain-store (empty cells write zeros), and the pre-launch
+  // zeroing of the level is skipped entirely
+  int store_flush = (rowsegs == 1) ? 1 : 0;
+  if (!store_flush) {
+    if (n_trees > 1) {
+      // zero exactly the level's cells of EVERY (tree, class) plane with
+      // one strided memset (planes are contiguous: hist_tstride == KP *
+      // hist_stride)
+      long planes = (long)n_trees * std::max(1, KP);
+      long pitch = (KP > 1) ? hist_stride : hist_tstride;
+      HIP_CHECK(hipMemset2DAsync(hist, (size_t)pitch * 4, 0,
+                                 (size_t)lvlcells * 4, (size_t)planes, s));
+    } else {
+      HIP_CHECK(hipMemsetAsync(
+          hist, 0,
+          ((size_t)std::max(0L, (long)(KP - 1)) * (size_t)hist_stride +
+           (size_t)lvlcells) * 4, s));
+    }
   }
   auto hl = [&](auto kern) {
     HIP_CHECK(hipFuncSetAttribute(
@@ -1856,7 +1878,7 @@ static void launch_hist_level(const unsigned char* Xtp, const int* idxmap,
                        n_rows, F, lvl_lo, lvl_nodes, f_pb, rows_pb,
                        even_only, ng_nodes, mode, hs, gs, gbias,
                        KP, g_stride, hist_stride, ngroups, idx_tstride,
-                       g_tstride, hist_tstride);
+                       g_tstride, hist_tstride, store_flush);
   };
   if (n_trees > 1) {
     if (KP > 1) hl(gbt_hist_kernel<1, 1>);
@@ -1979,6 +2001,10 @@ void gbt_split_find_body(const float* __restrict__ nh, int F, int nd,
                          int force_leaf, int* __restrict__ feat,
                          int* __restrict__ thr, float* __restrict__ val,
                          int kfeat, unsigned fseed, int etr) {
+  // dead subtree: a non-split parent leaves no rows (and, under the
+  // store-flush + skipped-sibling scheme, garbage cells) here.  feat
+  // buffers are -1-initialized, so returning IS the leaf/dead marking.
+  if (nd > 0 && feat[(nd - 1) >> 1] < 0) return;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -2256,6 +2282,37 @@ __global__ void gbt_predict_kernel(const unsigned char* __restrict__ Xb,
       acc += vals[(long)t * NN + nd];
     }
     out[r] = acc;
+  }
+}
+
+// multiclass predict: ONE launch walks every class plane (grid.y = c)
+// directly in the interleaved [T*K][NN] layout -- replaces K separate
+// gbt_predict launches on K .contiguous() plane copies (the digits-style
+// K=10 scoring was launch/copy bound)
+__global__ void gbt_predict_multi_kernel(
+    const unsigned char* __restrict__ Xb, const int* __restrict__ idx,
+    const int* __restrict__ feats, const int* __restrict__ thrs,
+    const float* __restrict__ vals, int T, int K, int NN, int F,
+    long n_idx, const float* __restrict__ f0s,
+    float* __restrict__ out) {
+  const int c = blockIdx.y;
+  long r = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; r < n_idx; r += (long)gridDim.x * blockDim.x) {
+    const unsigned char* xrow = Xb + (long)idx[r] * F;
+    float acc = f0s[c];
+    for (int t = 0; t < T; ++t) {
+      const long pl = ((long)t * K + c) * NN;
+      const int* tf = feats + pl;
+      const int* tt = thrs + pl;
+      int nd = 0;
+      int guard = 0;
+      while (tf[nd] >= 0 && guard++ < 16) {
+        nd = 2 * nd + 1 + (xrow[tf[nd]] > tt[nd] ? 1 : 0);
+        if (nd >= NN) { nd = (nd - 1) / 2; break; }
+      }
+      acc += vals[pl + nd];
+    }
+    out[(long)c * n_idx + r] = acc;
   }
 }
 
@@ -5513,6 +5570,30 @@ torch::Tensor gbt_predict(torch::Tensor Xb, torch::Tensor idx,
   return out;
 }
 
+torch::Tensor gbt_predict_multi(torch::Tensor Xb, torch::Tensor idx,
+                                torch::Tensor feats, torch::Tensor thrs,
+                                torch::Tensor vals, int64_t K,
+                                torch::Tensor f0s) {
+  const at::cuda::CUDAGuard guard(Xb.device());
+  TORCH_CHECK(f0s.size(0) == K && f0s.dtype() == torch::kFloat32);
+  int T = (int)(feats.size(0) / K);
+  int NN = (int)feats.size(1);
+  int F = (int)Xb.size(1);
+  long n_idx = idx.size(0);
+  auto out = torch::empty({(long)K, n_idx},
+                          vals.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<long>((n_idx + 255) / 256, 2048);
+  hipLaunchKernelGGL(gbt_predict_multi_kernel,
+                     dim3(blocks, (unsigned)K), dim3(256), 0,
+                     cur_stream(), Xb.data_ptr<unsigned char>(),
+                     idx.data_ptr<int>(), feats.data_ptr<int>(),
+                     thrs.data_ptr<int>(), vals.data_ptr<float>(), T,
+                     (int)K, NN, F, n_idx, f0s.data_ptr<float>(),
+                     out.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 torch::Tensor gbt_metric(torch::Tensor raw, torch::Tensor y,
                          torch::Tensor idx, int64_t metric) {
   const at::cuda::CUDAGuard guard(raw.device());
@@ -6086,6 +6167,9 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
                               int* __restrict__ feats,
                               int* __restrict__ thrs,
                               float* __restrict__ vals, long NN) {
+  // dead subtree (see gbt_split_find_body): plane-0 feat carries the
+  // tree structure for every class
+  if (nd > 0 && feats[(nd - 1) >> 1] < 0) return;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -6330,10 +6414,13 @@ __global__ void rf_bsibling_kernel(const float* __restrict__ prev,
   long base_l = (long)(2 * pair) * cells;
   long base_r = (long)(2 * pair + 1) * cells;
   long base_p = (long)pair * cells;
-  bool split = feat[prev_lo + pair] >= 0;
+  // non-split parents are skipped outright (instead of zero-filling the
+  // right child): the split finders early-out on dead subtrees, so the
+  // stale cells are never read
+  if (feat[prev_lo + pair] < 0) return;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < cells;
        i += (long)gridDim.x * blockDim.x) {
-    cur[base_r + i] = split ? prev[base_p + i] - cur[base_l + i] : 0.f;
+    cur[base_r + i] = prev[base_p + i] - cur[base_l + i];
   }
 }
 
@@ -6923,6 +7010,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_fit", &rf_fit,
         "RandomForest fit (bootstrap + gini/mse splits, feature subsampling)");
   m.def("gbt_predict", &gbt_predict, "tree-walk raw predictions");
+  m.def("gbt_predict_multi", &gbt_predict_multi,
+        "one-launch all-class tree walk over [T*K][NN] planes");
   m.def("gbt_metric", &gbt_metric, "accuracy/ss_res sums");
   m.def("csr_linear_fit", &csr_linear_fit,
         "CSR-resident batched linear fit (K5)");

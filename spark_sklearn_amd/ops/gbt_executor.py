@@ -153,18 +153,10 @@ class DeviceGBTExecutor:
                 continue
             if self.K > 2:
                 feats, thrs, vals, f0, spec = m
-                K, T = self.K, spec.n_estimators
-                NN = feats.shape[1]
-                raws = torch.empty((K, n), dtype=torch.float32,
-                                   device=self.device)
-                fv = feats.view(T, K, NN)
-                tv = thrs.view(T, K, NN)
-                vv = vals.view(T, K, NN)
-                for c in range(K):
-                    raws[c] = self.ext.gbt_predict(
-                        self.Xb, idx_t, fv[:, c].contiguous(),
-                        tv[:, c].contiguous(), vv[:, c].contiguous(),
-                        float(f0[c]))
+                f0_t = torch.as_tensor(
+                    np.asarray(f0, dtype=np.float32)).to(self.device)
+                raws = self.ext.gbt_predict_multi(
+                    self.Xb, idx_t, feats, thrs, vals, self.K, f0_t)
                 sums = self.ext.gbt_metric_multi(raws, self.y, idx_t,
                                                  self.K)
                 out[i] = float(sums.item()) / n

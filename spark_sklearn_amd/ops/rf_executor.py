@@ -74,19 +74,13 @@ class DeviceRFExecutor(DeviceGBTExecutor):
                 out[i] = float("nan")
                 continue
             kind, feats, thrs, vals, spec = m
-            T = spec.n_estimators
             if kind == "multi":
-                K = self.K
-                NN = feats.shape[1]
-                raws = torch.empty((K, n), dtype=torch.float32,
-                                   device=self.device)
-                fv = feats.view(T, K, NN)
-                tv = thrs.view(T, K, NN)
-                vv = vals.view(T, K, NN)
-                for c in range(K):
-                    raws[c] = self.ext.gbt_predict(
-                        self.Xb, idx_t, fv[:, c].contiguous(),
-                        tv[:, c].contiguous(), vv[:, c].contiguous(), 0.0)
+                if getattr(self, "_zeros_k", None) is None:
+                    self._zeros_k = torch.zeros(
+                        self.K, dtype=torch.float32, device=self.device)
+                raws = self.ext.gbt_predict_multi(
+                    self.Xb, idx_t, feats, thrs, vals, self.K,
+                    self._zeros_k)
                 sums = self.ext.gbt_metric_multi(raws, self.y, idx_t,
                                                  self.K)
                 out[i] = float(sums.item()) / n
