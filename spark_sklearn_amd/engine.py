@@ -122,7 +122,7 @@ class LinearFamily:
 
     @staticmethod
     def build_plan(estimator, params_list, n_samples, executor, n_classes,
-                   error_score):
+                   error_score, msq_scale=1.0):
         from .models._solver_ref import BatchPlan
 
         cols_per_cand = LinearFamily.cols_per_cand(n_classes)
@@ -131,6 +131,10 @@ class LinearFamily:
         # ~30us each: seconds of host overhead at 10k candidates)
         proto = clone(estimator)
         base_params = proto.get_params()
+        # msq_scale > 1 under sample_weight: the auto-lr curvature bound
+        # becomes c * max(w) * E[||x||^2] (w is mean-1 normalized), a safe
+        # upper bound computable identically on host and device tiers
+        eff_msq = executor.msq * msq_scale
         for slot, params in enumerate(params_list):
             try:
                 unknown = set(params) - set(base_params)
@@ -140,7 +144,7 @@ class LinearFamily:
                         f"{type(proto).__name__}")
                 proto.__dict__.update(base_params)
                 proto.__dict__.update(params)
-                spec = proto._column_spec(n_samples, executor.msq)
+                spec = proto._column_spec(n_samples, eff_msq)
             except Exception as exc:  # noqa: BLE001 - error_score contract
                 if error_score == "raise":
                     raise
@@ -221,11 +225,15 @@ class _HostLinearExecutor:
         self.y = y_enc
         self.classes = classes
 
-    def fit(self, train_idx, plan):
+    supports_sample_weight = True
+
+    def fit(self, train_idx, plan, sample_weight=None):
         from .models._solver_ref import linear_fit_ref
 
+        sw = (np.asarray(sample_weight, np.float32)[train_idx]
+              if sample_weight is not None else None)
         W, b, n_iter, loss = linear_fit_ref(
-            self.X[train_idx], self.y[train_idx], plan)
+            self.X[train_idx], self.y[train_idx], plan, sample_weight=sw)
         return (W, b), n_iter, loss
 
     def scores(self, idx, fitted, plan, metric):
@@ -269,6 +277,46 @@ def _encode_cached(family, estimator, y):
     return out
 
 
+def warn_sample_weight_ignored(sample_weight, name):
+    """Loud divergence marker for families without device weighting
+    (KMeans / GBT / RandomForest): the argument is accepted for sklearn
+    signature parity but the fit runs unweighted."""
+    if sample_weight is not None:
+        warnings.warn(
+            f"sample_weight is IGNORED by {name}: the batched device "
+            "path fits unweighted (linear-family estimators honor it; "
+            "tree/kmeans support is a roadmap item)", stacklevel=3)
+
+
+def _route_fit_params(fit_params, executor, X):
+    """Split fit_params into (normalized sample_weight, msq_scale,
+    ignored dict).  sample_weight is honored when the executor supports
+    it (linear family, dense device + host tiers): validated, normalized
+    to MEAN 1 over the full dataset (so the (1/n)-scaled objective keeps
+    its magnitude and the auto-lr bound only needs max(w)), float32."""
+    if not fit_params:
+        return None, 1.0, {}
+    ignored = {k: v for k, v in fit_params.items() if k != "sample_weight"}
+    sw = fit_params.get("sample_weight")
+    if sw is None:
+        return None, 1.0, ignored
+    if not getattr(executor, "supports_sample_weight", False):
+        ignored["sample_weight"] = sw
+        return None, 1.0, ignored
+    n_rows = X.shape[0] if hasattr(X, "shape") else len(X)
+    sw = np.asarray(sw, dtype=np.float64).reshape(-1)
+    if sw.shape[0] != n_rows:
+        raise ValueError(
+            f"sample_weight has {sw.shape[0]} entries for {n_rows} rows")
+    if not np.isfinite(sw).all() or (sw < 0).any():
+        raise ValueError("sample_weight must be finite and non-negative")
+    tot = float(sw.sum())
+    if tot <= 0:
+        raise ValueError("sample_weight sums to zero")
+    w = (sw * (n_rows / tot)).astype(np.float32)
+    return w, float(w.max()), ignored
+
+
 def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                       fit_params, return_train_score, error_score, verbose,
                       device=None,
@@ -280,19 +328,22 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     train/test row sets, hence the same resident data view); candidates
     within a fold are chunked to ``column_batch`` columns per launch.
     """
-    if fit_params:
-        warnings.warn(
-            "fit_params "
-            f"({', '.join(sorted(fit_params))}) are IGNORED by the batched "
-            "kernel path: fits run unweighted/unmodified, so results "
-            "differ from an estimator that honors them (sample_weight "
-            "support is a roadmap item; see ROADMAP.md)",
-            stacklevel=2)
     family = get_family(estimator)
     classes, y_enc = _encode_cached(family, estimator, y)
     n_classes = len(classes) if classes is not None else 0
     device = _device_or_none(device)
     executor = family.make_executor(X, y_enc, classes, device)
+    sample_weight, msq_scale, ignored_fp = _route_fit_params(
+        fit_params, executor, X)
+    if ignored_fp:
+        warnings.warn(
+            "fit_params "
+            f"({', '.join(sorted(ignored_fp))}) are IGNORED by the batched "
+            "kernel path: fits run unweighted/unmodified, so results "
+            "differ from an estimator that honors them (sample_weight is "
+            "honored for the linear family; other fit_params are a "
+            "roadmap item -- see ROADMAP.md)",
+            stacklevel=2)
     metric = family.metric_for(scorer, estimator)
     # multimetric fast path: when EVERY metric of a _MultiScorer maps to a
     # device metric, score each on-device instead of the host fallback
@@ -340,11 +391,18 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             chunk = fold_tasks[start:start + cand_chunk]
             params_list = [t.params for t in chunk]
             t0 = time.perf_counter()
-            plan, bad = family.build_plan(
-                estimator, params_list, len(train), executor,
-                n_classes, error_score)
+            if msq_scale != 1.0 and family is LinearFamily:
+                plan, bad = family.build_plan(
+                    estimator, params_list, len(train), executor,
+                    n_classes, error_score, msq_scale=msq_scale)
+            else:
+                plan, bad = family.build_plan(
+                    estimator, params_list, len(train), executor,
+                    n_classes, error_score)
+            fit_kw = ({"sample_weight": sample_weight}
+                      if sample_weight is not None else {})
             try:
-                fitted, n_iter, _ = executor.fit(train, plan)
+                fitted, n_iter, _ = executor.fit(train, plan, **fit_kw)
             except RuntimeError as exc:
                 # per-chunk retry on a transient HIP error (SURVEY.md
                 # section 5.3a); a second failure aborts the job
@@ -353,7 +411,7 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                 warnings.warn(
                     f"retrying candidate chunk after HIP error: {exc}",
                     RuntimeWarning, stacklevel=2)
-                fitted, n_iter, _ = executor.fit(train, plan)
+                fitted, n_iter, _ = executor.fit(train, plan, **fit_kw)
             fit_time = time.perf_counter() - t0
 
             t1 = time.perf_counter()
@@ -595,7 +653,7 @@ def _host_score_fallback(family, estimator, X, y, fitted, plan, chunk,
 # single-estimator fit (standalone .fit(), and rank-0 refit)
 # --------------------------------------------------------------------------
 
-def fit_single(estimator, X, y, device=None):
+def fit_single(estimator, X, y, device=None, sample_weight=None):
     """Fit one estimator (its OVR classes still batch into one launch)."""
     family = get_family(estimator)
     classes, y_enc = _encode_cached(family, estimator, y)
@@ -603,10 +661,23 @@ def fit_single(estimator, X, y, device=None):
     device = _device_or_none(device)
     executor = family.make_executor(X, y_enc, classes, device)
     n = X.shape[0] if hasattr(X, "shape") else len(X)
-    plan, _ = family.build_plan(estimator, [{}], n, executor, n_classes,
-                                "raise")
+    fp = {"sample_weight": sample_weight} if sample_weight is not None \
+        else {}
+    sw, msq_scale, ignored = _route_fit_params(fp, executor, X)
+    if ignored:
+        warnings.warn(
+            "sample_weight is IGNORED by this estimator family's batched "
+            "path (honored for the linear family only)", stacklevel=2)
+    if msq_scale != 1.0 and family is LinearFamily:
+        plan, _ = family.build_plan(estimator, [{}], n, executor,
+                                    n_classes, "raise",
+                                    msq_scale=msq_scale)
+    else:
+        plan, _ = family.build_plan(estimator, [{}], n, executor,
+                                    n_classes, "raise")
     all_idx = np.arange(n, dtype=np.int64)
-    fitted, n_iter, _ = executor.fit(all_idx, plan)
+    fit_kw = {"sample_weight": sw} if sw is not None else {}
+    fitted, n_iter, _ = executor.fit(all_idx, plan, **fit_kw)
     family.install_fitted(estimator, fitted, 0, plan, classes)
     if n_iter is not None:
         estimator.n_iter_ = n_iter

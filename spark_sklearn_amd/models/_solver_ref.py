@@ -93,9 +93,16 @@ def _column_targets(y: np.ndarray, plan: BatchPlan) -> np.ndarray:
 
 
 def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
-                   sample_stride: int = 1):
+                   sample_stride: int = 1,
+                   sample_weight: Optional[np.ndarray] = None):
     """Fit B columns simultaneously.  Returns (W [F,B], b [B], n_iter [B],
-    final_loss [B]).  X may be dense or scipy CSR."""
+    final_loss [B]).  X may be dense or scipy CSR.
+
+    sample_weight (already normalized to mean ~1 by the engine) scales
+    each row's loss AND gradient: d_i -> w_i * d_i, l_i -> w_i * l_i.
+    Everything downstream (G = X^T D / n, gb = mean(D), the convergence
+    loss) then weights automatically.  The device kernels apply the same
+    per-row multiply in the fwd epilogue (fwd_d3/fwd_d2 WT=1)."""
     import scipy.sparse as sp
 
     if sp.issparse(X):
@@ -105,6 +112,9 @@ def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
     n, F = X.shape
     B = plan.B
     T = _column_targets(np.asarray(y), plan)
+    w_row = None
+    if sample_weight is not None:
+        w_row = np.asarray(sample_weight, dtype=np.float32).reshape(n, 1)
 
     W = np.zeros((F, B), dtype=np.float32)
     b = np.zeros(B, dtype=np.float32)
@@ -127,17 +137,20 @@ def linear_fit_ref(X: np.ndarray, y: np.ndarray, plan: BatchPlan,
                 p = 1.0 / (1.0 + np.exp(-z))
                 D[:, cols] = p - t
                 eps = 1e-12
-                loss_sum[cols] = -np.sum(
-                    t * np.log(p + eps) + (1 - t) * np.log(1 - p + eps),
-                    axis=0)
+                le = -(t * np.log(p + eps) + (1 - t) * np.log(1 - p + eps))
             elif lid == LOSS_HINGE:
                 margin = t * z
                 D[:, cols] = np.where(margin < 1.0, -t, 0.0)
-                loss_sum[cols] = np.sum(np.maximum(0.0, 1.0 - margin), axis=0)
+                le = np.maximum(0.0, 1.0 - margin)
             else:
                 r = z - t
                 D[:, cols] = r
-                loss_sum[cols] = 0.5 * np.sum(r * r, axis=0)
+                le = 0.5 * r * r
+            if w_row is not None:
+                le = le * w_row
+            loss_sum[cols] = np.sum(le, axis=0)
+        if w_row is not None:
+            D *= w_row
 
         # data-loss only (no reg term): the convergence criterion the HIP
         # finalize kernel uses -- keeps device/host n_iter_ identical

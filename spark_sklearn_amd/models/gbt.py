@@ -59,7 +59,10 @@ class _GBTBase(BaseEstimator):
                        l2_regularization=float(self.l2_regularization),
                        max_bins=int(self.max_bins))
 
-    def fit(self, X, y):
+    def fit(self, X, y, sample_weight=None):
+        from ..engine import warn_sample_weight_ignored
+
+        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import fit_single
 
         return fit_single(self, X, y)

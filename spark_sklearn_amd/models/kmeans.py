@@ -56,7 +56,10 @@ class KMeans(ClusterMixin, BaseEstimator):
         self.random_state = random_state
 
     # -- standalone fit ----------------------------------------------------
-    def fit(self, X, y=None):
+    def fit(self, X, y=None, sample_weight=None):
+        from ..engine import warn_sample_weight_ignored
+
+        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import _device_or_none, get_family
 
         family = get_family(self)

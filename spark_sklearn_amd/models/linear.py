@@ -66,10 +66,10 @@ class _LinearBase(BaseEstimator):
                     fit_intercept=1 if self.fit_intercept else 0)
 
     # -- single-estimator fit (CPU numpy; GPU via engine helper) -----------
-    def fit(self, X, y):
+    def fit(self, X, y, sample_weight=None):
         from ..engine import fit_single
 
-        return fit_single(self, X, y)
+        return fit_single(self, X, y, sample_weight=sample_weight)
 
     def _finalize_fit(self, W: np.ndarray, b: np.ndarray,
                       n_iter: np.ndarray, classes: Optional[np.ndarray]):

@@ -86,7 +86,10 @@ class _RFBase(BaseEstimator):
             splitter=self._splitter,
         )
 
-    def fit(self, X, y):
+    def fit(self, X, y, sample_weight=None):
+        from ..engine import warn_sample_weight_ignored
+
+        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import fit_single
 
         fit_single(self, X, y)

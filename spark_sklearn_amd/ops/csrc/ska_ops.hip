@@ -284,7 +284,7 @@ __device__ __forceinline__ void loss_grad(int lid, float z, float yv,
 // 64x64 v1 tile, and W is staged once per k-chunk for ALL columns (no
 // cross-block X re-reads within a row tile).
 // ---------------------------------------------------------------------------
-template <int P, int NF>
+template <int P, int NF, int WT = 0>
 __global__ __launch_bounds__(256)
 void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
                    const float* __restrict__ y,        // [n]
@@ -298,7 +298,8 @@ void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
                    float* __restrict__ db_sum,         // [Bp]
                    const int* __restrict__ row_key,    // unused (generic)
                    const int* __restrict__ col_key,    // unused (generic)
-                   int nt, int NTp, int Fp, int Bp, int want_loss) {
+                   int nt, int NTp, int Fp, int Bp, int want_loss,
+                   const float* __restrict__ w = nullptr) {
   constexpr int AMAP = P >> 1, BMAP = P & 1;
   (void)want_loss;  // generic path always computes loss
   (void)row_key;
@@ -309,6 +310,7 @@ void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
   __shared__ float sDb[256];
   __shared__ int sIdx[64];
   __shared__ float sY[64];
+  __shared__ float sWgt[WT ? 64 : 1];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -321,6 +323,7 @@ void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
     int gi = idx[r];
     sIdx[tid] = gi;
     sY[tid] = y[gi];
+    if (WT) sWgt[tid] = w[gi];
   }
   for (int i = tid; i < Bp; i += 256) { sLoss[i] = 0.f; sDb[i] = 0.f; }
   __syncthreads();
@@ -382,6 +385,11 @@ void fwd_d2_kernel(const __bf16* __restrict__ X,       // [n][Fp]
         float d, ls;
         loss_grad(lid, z, sY[row_base + r], tgt, &d, &ls);
         if (gr >= nt) { d = 0.f; ls = 0.f; }
+        if (WT) {
+          float wt = sWgt[row_base + r];
+          d *= wt;
+          ls *= wt;
+        }
         dpack[r] = (__bf16)d;
         lsum += ls;
         dsum += d;
@@ -500,7 +508,7 @@ void bwd_g2_kernel(const __bf16* __restrict__ Xt,    // [Fp][n]
 // registers while MFMA consumes the current LDS buffer; guide T3 minimum
 // recipe) + want_loss gating.
 // ---------------------------------------------------------------------------
-template <int NF>
+template <int NF, int WT = 0>   // WT: per-row sample_weight (w != null)
 __global__ __launch_bounds__(256)
 void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                    const int* __restrict__ idx, const __bf16* __restrict__ Wt,
@@ -510,7 +518,8 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
                    float* __restrict__ loss_sum, float* __restrict__ db_sum,
                    const int* __restrict__ row_key,   // [n] or null
                    const int* __restrict__ col_key,   // [Bp] or null
-                   int nt, int NTp, int Fp, int Bp, int want_loss) {
+                   int nt, int NTp, int Fp, int Bp, int want_loss,
+                   const float* __restrict__ w = nullptr) {
   __shared__ __bf16 sX[2][64 * 64];
   __shared__ __bf16 sW[2][NF * 64 * 64];
   __shared__ float sLoss[256];
@@ -518,6 +527,7 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
   __shared__ int sIdx[64];
   __shared__ float sY[64];
   __shared__ int sKey[64];
+  __shared__ float sWgt[WT ? 64 : 1];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -534,6 +544,7 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
     sIdx[tid] = gi;
     sY[tid] = y[gi];
     sKey[tid] = (row_key != nullptr) ? row_key[gi] : 0;
+    if (WT) sWgt[tid] = w[gi];
   }
   for (int i = tid; i < Bp; i += 256) { sLoss[i] = 0.f; sDb[i] = 0.f; }
   __syncthreads();
@@ -635,6 +646,11 @@ void fwd_d3_kernel(const __bf16* __restrict__ X, const float* __restrict__ y,
         if (row_key != nullptr && sKey[row_base + r] != ck) {
           d = 0.f;          // keyed batching: column only sees its key
           ls = 0.f;
+        }
+        if (WT) {           // sample_weight: scale grad AND loss per row
+          float wt = sWgt[row_base + r];
+          d *= wt;
+          ls *= wt;
         }
         dpack[r] = (__bf16)d;
         lsum += ls;
@@ -4865,7 +4881,8 @@ std::vector<torch::Tensor> linear_fit(
     torch::Tensor l2, torch::Tensor mom, torch::Tensor tol,
     torch::Tensor max_iter, torch::Tensor target, torch::Tensor fit_ic,
     torch::Tensor row_key, torch::Tensor col_key,
-    int64_t nt, int64_t max_epochs, int64_t check_every) {
+    int64_t nt, int64_t max_epochs, int64_t check_every,
+    torch::Tensor sw) {
   TORCH_CHECK(g_layout >= 0, "MFMA layout not probed; call ensure_layout()");
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kBFloat16 && X.dim() == 2);
   TORCH_CHECK(X.stride(1) == 1 && X.stride(0) == X.size(1));
@@ -4885,11 +4902,24 @@ std::vector<torch::Tensor> linear_fit(
 
   auto optF = X.options().dtype(torch::kFloat32);
   auto optI = X.options().dtype(torch::kInt32);
+  // sample_weight: per-row weight over the FULL resident matrix (kernels
+  // index it by the global row id).  Weighted fits run the generic
+  // fwd_d3/fwd_d2 epoch path with the WT=1 epilogue (the fused/flagship
+  // variants stay weight-free).
+  const bool weighted = sw.numel() > 0;
+  const float* wp = nullptr;
+  if (weighted) {
+    TORCH_CHECK(sw.dtype() == torch::kFloat32 && sw.is_cuda() &&
+                sw.size(0) == X.size(0),
+                "sample_weight must be float32 [n_rows] on device");
+    wp = sw.data_ptr<float>();
+  }
   auto W = torch::zeros({Bp, Fp}, optF);
   auto V = torch::zeros({Bp, Fp}, optF);
   auto Gt = torch::zeros({Bp, Fp}, optF);
   auto Wt = torch::zeros({Bp, Fp}, X.options());
   bool use_fused14 = (flagship_v() == 14 || flagship_v() >= 18) &&
+                     !weighted &&
                      Bp == 256 && NTp % 256 == 0 &&
                      (Fp == 64 || Fp == 128 || Fp == 256);
   auto Dt = torch::zeros({(long)Bp, use_fused14 ? 1L : NTp}, X.options());
@@ -4938,7 +4968,8 @@ std::vector<torch::Tensor> linear_fit(
                        target.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(Dt.data_ptr()),
                        loss_sum.data_ptr<float>(), db_sum.data_ptr<float>(),
-                       rkp, ckp, (int)nt, (int)NTp, Fp, Bp, want_loss_cur);
+                       rkp, ckp, (int)nt, (int)NTp, Fp, Bp, want_loss_cur,
+                       wp);
     hipLaunchKernelGGL(bwd_k, bwd_grid, dim3(256), 0, s, Xtp,
                        idx.data_ptr<int>(),
                        reinterpret_cast<const __bf16*>(Dt.data_ptr()),
@@ -5016,7 +5047,7 @@ std::vector<torch::Tensor> linear_fit(
         }
         continue;
       }
-      if (nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
+      if (!weighted && nf == 4 && NTp % 256 == 0 && Fp % 128 == 0) {
         launch_flagship_pair(
             s, Xp, Xtp, y.data_ptr<float>(), idx.data_ptr<int>(),
             reinterpret_cast<const __bf16*>(Wt.data_ptr()),
@@ -5042,6 +5073,16 @@ std::vector<torch::Tensor> linear_fit(
                            epoch_mask.data_ptr<int>(), lr.data_ptr<float>(),
                            mom.data_ptr<float>(), l1.data_ptr<float>(),
                            l2.data_ptr<float>(), inv_n, Fp);
+      } else if (weighted) switch (nf) {
+        case 1:
+          launch_epoch(epoch, fwd_d3_kernel<1, 1>, bwd_g3_kernel<1>);
+          break;
+        case 2:
+          launch_epoch(epoch, fwd_d3_kernel<2, 1>, bwd_g3_kernel<2>);
+          break;
+        default:
+          launch_epoch(epoch, fwd_d3_kernel<4, 1>, bwd_g3_kernel<4>);
+          break;
       } else switch (nf) {
         case 1: launch_epoch(epoch, fwd_d3_kernel<1>, bwd_g3_kernel<1>); break;
         case 2: launch_epoch(epoch, fwd_d3_kernel<2>, bwd_g3_kernel<2>); break;
@@ -5058,7 +5099,20 @@ std::vector<torch::Tensor> linear_fit(
       }
       continue;
     }
-    switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
+    if (weighted) switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
+      case 0:  launch_epoch(epoch, fwd_d2_kernel<0, 1, 1>, bwd_g2_kernel<0, 1>); break;
+      case 1:  launch_epoch(epoch, fwd_d2_kernel<0, 2, 1>, bwd_g2_kernel<0, 2>); break;
+      case 2:  launch_epoch(epoch, fwd_d2_kernel<0, 4, 1>, bwd_g2_kernel<0, 4>); break;
+      case 4:  launch_epoch(epoch, fwd_d2_kernel<1, 1, 1>, bwd_g2_kernel<1, 1>); break;
+      case 5:  launch_epoch(epoch, fwd_d2_kernel<1, 2, 1>, bwd_g2_kernel<1, 2>); break;
+      case 6:  launch_epoch(epoch, fwd_d2_kernel<1, 4, 1>, bwd_g2_kernel<1, 4>); break;
+      case 8:  launch_epoch(epoch, fwd_d2_kernel<2, 1, 1>, bwd_g2_kernel<2, 1>); break;
+      case 9:  launch_epoch(epoch, fwd_d2_kernel<2, 2, 1>, bwd_g2_kernel<2, 2>); break;
+      case 10: launch_epoch(epoch, fwd_d2_kernel<2, 4, 1>, bwd_g2_kernel<2, 4>); break;
+      case 12: launch_epoch(epoch, fwd_d2_kernel<3, 1, 1>, bwd_g2_kernel<3, 1>); break;
+      case 13: launch_epoch(epoch, fwd_d2_kernel<3, 2, 1>, bwd_g2_kernel<3, 2>); break;
+      default: launch_epoch(epoch, fwd_d2_kernel<3, 4, 1>, bwd_g2_kernel<3, 4>); break;
+    } else switch (g_layout * 4 + (nf == 4 ? 2 : nf - 1)) {
       case 0:  launch_epoch(epoch, fwd_d2_kernel<0, 1>, bwd_g2_kernel<0, 1>); break;
       case 1:  launch_epoch(epoch, fwd_d2_kernel<0, 2>, bwd_g2_kernel<0, 2>); break;
       case 2:  launch_epoch(epoch, fwd_d2_kernel<0, 4>, bwd_g2_kernel<0, 4>); break;

@@ -164,8 +164,25 @@ class DeviceLinearExecutor:
         self._idx_cache[key] = (arr, t)
         return t
 
+    supports_sample_weight = True
+
+    def _cached_weights(self, w):
+        """Upload the (engine-normalized) per-row weight vector once per
+        array; kernels index it by GLOBAL row id, so the full vector is
+        shared by every fold."""
+        key = id(w)
+        hit = getattr(self, "_w_cache", None)
+        if hit is not None and hit[0] is w:
+            return hit[1]
+        t = self.torch.as_tensor(
+            np.ascontiguousarray(np.asarray(w, dtype=np.float32))
+        ).to(self.device)
+        self._w_cache = (w, t)
+        return t
+
     # -- fit ---------------------------------------------------------------
-    def fit(self, train_idx, plan, row_key=None, col_key=None):
+    def fit(self, train_idx, plan, row_key=None, col_key=None,
+            sample_weight=None):
         torch = self.torch
         nt = len(train_idx)
         idx = self._cached_idx(train_idx, 256, np.int32)
@@ -208,10 +225,13 @@ class DeviceLinearExecutor:
         else:
             rk_t = torch.empty(0, dtype=torch.int32, device=self.device)
             ck_t = torch.empty(0, dtype=torch.int32, device=self.device)
+        sw_t = (self._cached_weights(sample_weight)
+                if sample_weight is not None else
+                torch.empty(0, dtype=torch.float32, device=self.device))
         W, bias, n_iter, cur_loss, Wt = self.ext.linear_fit(
             self.X, self.Xt, self.y, idx, loss_t, lr_t, l1_t, l2_t, mom_t,
             tol_t, mi_t, tgt_t, ic_t, rk_t, ck_t, nt, max_epochs,
-            check_every)
+            check_every, sw_t)
         torch.cuda.synchronize(self.device)
         self._fit_state = (Wt, bias, loss_t, tgt_t)
         W_np = W[:B, :self.F].float().cpu().numpy().T.copy()   # [F, B]

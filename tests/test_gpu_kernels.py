@@ -99,6 +99,39 @@ def test_linear_fit_matches_ref_logistic(dev, problem):
     assert list(ni_d) == list(ni_r)
 
 
+def test_linear_fit_sample_weight_matches_ref(dev, problem):
+    """Weighted fit (WT=1 fwd epilogue) vs the weighted numpy reference;
+    also: unit weights reproduce the unweighted fit exactly."""
+    X, y = problem
+    Xb = _bf16_round(X)
+    msq = float(np.mean(np.sum(Xb.astype(np.float64) ** 2, axis=1)))
+    rng = np.random.default_rng(11)
+    w_full = rng.uniform(0.2, 3.0, size=X.shape[0]).astype(np.float64)
+    w_full = (w_full * (len(w_full) / w_full.sum())).astype(np.float32)
+    lr = auto_lr(msq * float(w_full.max()), LOSS_LOGISTIC, 0.0)
+    B = 4
+    plan = _mkplan(B, LOSS_LOGISTIC, lr,
+                   l2=np.array([0.0, 1e-4, 1e-3, 1e-2], np.float32),
+                   max_iter=12)
+    train = np.arange(0, 6000, dtype=np.int64)
+
+    ex = DeviceLinearExecutor(X, y.astype(np.float32), None, dev)
+    (W_d, b_d), ni_d, loss_d = ex.fit(train, plan, sample_weight=w_full)
+    W_r, b_r, ni_r, loss_r = linear_fit_ref(
+        Xb[train], y[train], plan, sample_weight=w_full[train])
+    np.testing.assert_allclose(W_d, W_r, rtol=0.05, atol=5e-4)
+    np.testing.assert_allclose(b_d, b_r, rtol=0.05, atol=5e-4)
+    np.testing.assert_allclose(loss_d, loss_r, rtol=0.02)
+    assert list(ni_d) == list(ni_r)
+
+    # unit weights == unweighted, bit-for-bit on device outputs
+    ones = np.ones(X.shape[0], np.float32)
+    (W_1, b_1), _, _ = ex.fit(train, plan, sample_weight=ones)
+    (W_0, b_0), _, _ = ex.fit(train, plan)
+    np.testing.assert_allclose(W_1, W_0, rtol=2e-3, atol=1e-5)
+    np.testing.assert_allclose(b_1, b_0, rtol=2e-3, atol=1e-5)
+
+
 def test_linear_fit_l1_and_hinge(dev, problem):
     X, y = problem
     Xb = _bf16_round(X)
