@@ -70,38 +70,52 @@ def bin_data(X: np.ndarray, edges) -> np.ndarray:
     return out
 
 
-def _grad_hess(loss, raw, y):
+def _grad_hess(loss, raw, y, w=None):
     if loss == LOSS_LOGISTIC:
         p = sigmoid(raw)
-        return (p - y).astype(np.float32), (p * (1.0 - p)).astype(np.float32)
-    return (raw - y).astype(np.float32), np.ones_like(raw, dtype=np.float32)
+        g, h = p - y, p * (1.0 - p)
+    else:
+        g, h = raw - y, np.ones_like(raw)
+    if w is not None:
+        g, h = g * w, h * w
+    return g.astype(np.float32), h.astype(np.float32)
 
 
-def _init_raw(loss, y):
+def _init_raw(loss, y, w=None):
+    m = float(np.mean(y)) if w is None else \
+        float(np.sum(w * y) / max(np.sum(w), 1e-12))
     if loss == LOSS_LOGISTIC:
-        p = np.clip(np.mean(y), 1e-6, 1 - 1e-6)
+        p = np.clip(m, 1e-6, 1 - 1e-6)
         return float(np.log(p / (1.0 - p)))
-    return float(np.mean(y))
+    return m
 
 
-def gbt_fit_ref(Xb: np.ndarray, y: np.ndarray, spec: GBTSpec):
+def gbt_fit_ref(Xb: np.ndarray, y: np.ndarray, spec: GBTSpec,
+                sample_weight=None):
     """Fit one model on binned data.  Returns (trees, f0).
 
     Tree heap layout per tree (size 2^(max_depth+1) - 1):
       feature int32 (-1 => leaf), threshold_bin uint8 (go left if
       bin <= thr), value float32 (leaf contribution, already scaled by
       the learning rate).
+
+    sample_weight scales g AND h per row (xgboost semantics: leaf value
+    -G/H and split gains become weighted automatically) plus the f0
+    base score; min_samples_leaf stays a plain row COUNT (sklearn
+    semantics).
     """
     n, F = Xb.shape
     y = np.asarray(y, dtype=np.float32)
-    f0 = _init_raw(spec.loss, y)
+    w = None if sample_weight is None else \
+        np.asarray(sample_weight, dtype=np.float32)
+    f0 = _init_raw(spec.loss, y, w)
     raw = np.full(n, f0, dtype=np.float32)
     n_nodes = 2 ** (spec.max_depth + 1) - 1
     lam = float(spec.l2_regularization)
     trees = []
 
     for _ in range(spec.n_estimators):
-        g, h = _grad_hess(spec.loss, raw, y)
+        g, h = _grad_hess(spec.loss, raw, y, w)
         feat = np.full(n_nodes, -1, dtype=np.int32)
         thr = np.zeros(n_nodes, dtype=np.uint8)
         val = np.zeros(n_nodes, dtype=np.float32)
@@ -281,11 +295,17 @@ def _build_tree(Xb, g, h, spec):
 
 
 def gbt_fit_ref_multi(Xb: np.ndarray, y_enc: np.ndarray, K: int,
-                      spec: GBTSpec):
+                      spec: GBTSpec, sample_weight=None):
     """Softmax boosting: trees[t][c].  Returns (trees, f0 [K])."""
     n = Xb.shape[0]
     y_enc = np.asarray(y_enc, dtype=np.int64)
-    prior = np.bincount(y_enc, minlength=K) / n
+    w = None if sample_weight is None else \
+        np.asarray(sample_weight, dtype=np.float32)
+    if w is None:
+        prior = np.bincount(y_enc, minlength=K) / n
+    else:
+        prior = np.bincount(y_enc, weights=w, minlength=K) / \
+            max(float(w.sum()), 1e-12)
     f0 = np.log(np.clip(prior, 1e-6, None)).astype(np.float32)
     raw = np.tile(f0, (n, 1))
     onehot = np.eye(K, dtype=np.float32)[y_enc]
@@ -296,6 +316,8 @@ def gbt_fit_ref_multi(Xb: np.ndarray, y_enc: np.ndarray, K: int,
         for c in range(K):
             g = (P[:, c] - onehot[:, c]).astype(np.float32)
             h = (P[:, c] * (1.0 - P[:, c])).astype(np.float32)
+            if w is not None:
+                g, h = g * w, h * w
             feat, thr, val, contrib = _build_tree(Xb, g, h, spec)
             raw[:, c] += contrib
             round_trees.append((feat, thr, val))

@@ -60,12 +60,9 @@ class _GBTBase(BaseEstimator):
                        max_bins=int(self.max_bins))
 
     def fit(self, X, y, sample_weight=None):
-        from ..engine import warn_sample_weight_ignored
-
-        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import fit_single
 
-        return fit_single(self, X, y)
+        return fit_single(self, X, y, sample_weight=sample_weight)
 
     @property
     def feature_importances_(self):
@@ -246,17 +243,23 @@ class _HostGBTExecutor:
         self.K = len(classes) if classes is not None else 0
         self.msq = 0.0   # unused by this family
 
-    def fit(self, train_idx, plan):
+    supports_sample_weight = True
+
+    def fit(self, train_idx, plan, sample_weight=None):
         Xb_tr = self.Xb[train_idx]
         y_tr = self.y[train_idx]
+        sw = (np.asarray(sample_weight, np.float32)[train_idx]
+              if sample_weight is not None else None)
         models = []
         n_iter = np.empty(plan.B, np.int32)
         for i, spec in enumerate(plan.specs):
             if self.K > 2:
                 trees, f0 = gbt_fit_ref_multi(
-                    Xb_tr, y_tr.astype(np.int64), self.K, spec)
+                    Xb_tr, y_tr.astype(np.int64), self.K, spec,
+                    sample_weight=sw)
             else:
-                trees, f0 = gbt_fit_ref(Xb_tr, y_tr, spec)
+                trees, f0 = gbt_fit_ref(Xb_tr, y_tr, spec,
+                                        sample_weight=sw)
             models.append((trees, f0))
             n_iter[i] = len(trees)
         return (models, self.edges), n_iter, None

@@ -1544,16 +1544,18 @@ __global__ void kmeans_block_finalize_kernel(
 __global__ void gbt_grad_kernel(const float* __restrict__ raw,
                                 const float* __restrict__ y,
                                 float* __restrict__ g, float* __restrict__ h,
-                                long nt, int loss) {
+                                long nt, int loss,
+                                const float* __restrict__ w) {
   long r = blockIdx.x * blockDim.x + threadIdx.x;
   for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    float wt = (w != nullptr) ? w[r] : 1.f;
     if (loss == LOSS_LOGISTIC) {
       float p = 1.f / (1.f + __expf(-raw[r]));
-      g[r] = p - y[r];
-      h[r] = p * (1.f - p);
+      g[r] = wt * (p - y[r]);
+      h[r] = wt * (p * (1.f - p));
     } else {
-      g[r] = raw[r] - y[r];
-      h[r] = 1.f;
+      g[r] = wt * (raw[r] - y[r]);
+      h[r] = wt;
     }
   }
 }
@@ -4360,7 +4362,8 @@ __global__ void gbt_softmax_grad_kernel(const float* __restrict__ raw,
                                         const float* __restrict__ y,
                                         float* __restrict__ g,
                                         float* __restrict__ h,
-                                        long nt, int K) {
+                                        long nt, int K,
+                                        const float* __restrict__ w) {
   long r = blockIdx.x * blockDim.x + threadIdx.x;
   for (; r < nt; r += (long)gridDim.x * blockDim.x) {
     float mx = raw[r];
@@ -4368,10 +4371,11 @@ __global__ void gbt_softmax_grad_kernel(const float* __restrict__ raw,
     float sum = 0.f;
     for (int c = 0; c < K; ++c) sum += __expf(raw[(long)c * nt + r] - mx);
     int yc = (int)y[r];
+    float wt = (w != nullptr) ? w[r] : 1.f;
     for (int c = 0; c < K; ++c) {
       float p = __expf(raw[(long)c * nt + r] - mx) / sum;
-      g[(long)c * nt + r] = p - ((c == yc) ? 1.f : 0.f);
-      h[(long)c * nt + r] = p * (1.f - p);
+      g[(long)c * nt + r] = wt * (p - ((c == yc) ? 1.f : 0.f));
+      h[(long)c * nt + r] = wt * (p * (1.f - p));
     }
   }
 }
@@ -5505,11 +5509,24 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                    torch::Tensor y_tr, int64_t loss,
                                    int64_t n_estimators, double lrate,
                                    int64_t max_depth, int64_t min_leaf,
-                                   double lam, double f0) {
+                                   double lam, double f0,
+                                   torch::Tensor sw) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
   TORCH_CHECK(max_depth >= 1 && max_depth <= 10,
               "device GBT supports max_depth in [1, 10]");
+  // sample_weight (fold-sliced [nt], engine-normalized): scales g and h
+  // in the grad kernel; the packed/deterministic histogram modes assume
+  // unweighted gradient bounds and are disabled under weighting
+  const bool weighted = sw.numel() > 0;
+  const float* wp = nullptr;
+  float wmax = 1.f;
+  if (weighted) {
+    TORCH_CHECK(sw.dtype() == torch::kFloat32 &&
+                sw.size(0) == idxmap.size(0));
+    wp = sw.data_ptr<float>();
+    wmax = sw.max().item<float>();
+  }
   int F = (int)Xb.size(1);
   long nt = idxmap.size(0);
   int T = (int)n_estimators;
@@ -5543,7 +5560,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
   // max|g| (bit-level atomicMax, order-independent) to set the
   // fixed-point bias/scale; the one host sync per round is the price of
   // the opt-in mode
-  bool det_sq = det_gbt_mode() && loss == LOSS_SQUARED;
+  bool det_sq = det_gbt_mode() && loss == LOSS_SQUARED && !weighted;
   auto gmax_bits = torch::zeros({1}, optI);
   float gbias_r = 1.f;
 
@@ -5554,7 +5571,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
     hipLaunchKernelGGL(gbt_grad_kernel, dim3(blocks), dim3(256), 0, s,
                        raw.data_ptr<float>(), y_tr.data_ptr<float>(),
                        g.data_ptr<float>(), h.data_ptr<float>(), nt,
-                       (int)loss);
+                       (int)loss, wp);
     if (det_sq) {
       HIP_CHECK(hipMemsetAsync(gmax_bits.data_ptr(), 0, 4, s));
       hipLaunchKernelGGL(absmax_kernel, dim3(blocks), dim3(256), 0, s,
@@ -5573,9 +5590,10 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
       launch_hist_level(Xtp, idxmap.data_ptr<int>(),
                         node_id.data_ptr<int>(), g.data_ptr<float>(),
                         h.data_ptr<float>(), hist.data_ptr<float>(), nt, n_rows,
-                        F, d, (loss == LOSS_LOGISTIC) ? 0.25f : 1.f,
-                        loss == LOSS_LOGISTIC || det_sq, false, s,
-                        1, 0, 0, det_sq ? gbias_r : 1.f);
+                        F, d,
+                        ((loss == LOSS_LOGISTIC) ? 0.25f : 1.f) * wmax,
+                        (loss == LOSS_LOGISTIC && !weighted) || det_sq,
+                        false, s, 1, 0, 0, det_sq ? gbias_r : 1.f);
       if (even_only) {
         int prev_lo = (1 << (d - 1)) - 1;
         long cells = (long)F * 256 * 3;
@@ -5605,7 +5623,7 @@ std::vector<torch::Tensor> gbt_fit(torch::Tensor Xb, torch::Tensor Xbt,
     HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
     HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
     {
-      bool fix = det_gbt_mode() &&
+      bool fix = det_gbt_mode() && !weighted &&
                  (loss == LOSS_LOGISTIC || loss == LOSS_SQUARED);
       float gb = det_sq ? gbias_r : 1.f;
       float fsc = fix ? det_fix_scale(nt, gb) : 0.f;
@@ -6112,7 +6130,8 @@ torch::Tensor csr_linear_scores(
 std::vector<torch::Tensor> gbt_fit_multi(
     torch::Tensor Xb, torch::Tensor Xbt, torch::Tensor idxmap,
     torch::Tensor y_tr,
-    torch::Tensor f0, int64_t K, int64_t n_estimators, double lrate,
+    torch::Tensor f0, torch::Tensor sw,
+    int64_t K, int64_t n_estimators, double lrate,
     int64_t max_depth, int64_t min_leaf, double lam) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(max_depth >= 1 && max_depth <= 10);
@@ -6143,12 +6162,20 @@ std::vector<torch::Tensor> gbt_fit_multi(
   hipStream_t s = cur_stream();
   const unsigned char* Xp = Xb.data_ptr<unsigned char>();
   int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
+  const bool weighted = sw.numel() > 0;
+  const float* wp = nullptr;
+  float wmax = 1.f;
+  if (weighted) {
+    TORCH_CHECK(sw.dtype() == torch::kFloat32 && sw.size(0) == nt);
+    wp = sw.data_ptr<float>();
+    wmax = sw.max().item<float>();
+  }
 
   for (int t = 0; t < T; ++t) {
     hipLaunchKernelGGL(gbt_softmax_grad_kernel, dim3(blocks), dim3(256), 0,
                        s, raw.data_ptr<float>(), y_tr.data_ptr<float>(),
                        g.data_ptr<float>(), h.data_ptr<float>(), nt,
-                       (int)K);
+                       (int)K, wp);
     for (int c = 0; c < (int)K; ++c) {
       float* gc = g.data_ptr<float>() + (long)c * nt;
       float* hc = h.data_ptr<float>() + (long)c * nt;
@@ -6161,9 +6188,9 @@ std::vector<torch::Tensor> gbt_fit_multi(
         int even_only = (d > 0) ? 1 : 0;
         launch_hist_level(Xtp, idxmap.data_ptr<int>(),
                           node_id.data_ptr<int>(), gc, hc,
-                          hist.data_ptr<float>(), nt, n_rows, F, d, 0.25f,
-                          /*g_bounded=*/true, false, s, 1, 0, 0, 1.f,
-                          /*packed_ok=*/true);
+                          hist.data_ptr<float>(), nt, n_rows, F, d,
+                          0.25f * wmax, /*g_bounded=*/!weighted, false, s,
+                          1, 0, 0, 1.f, /*packed_ok=*/!weighted);
         if (even_only) {
           int prev_lo = (1 << (d - 1)) - 1;
           long cells = (long)F * 256 * 3;
@@ -6194,7 +6221,7 @@ std::vector<torch::Tensor> gbt_fit_multi(
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
       {
-        bool fix = det_gbt_mode();
+        bool fix = det_gbt_mode() && !weighted;
         float fsc = fix ? det_fix_scale(nt) : 0.f;
         hipLaunchKernelGGL(gbt_node_sums_kernel,
                            dim3(fix ? 1 : blocks,

@@ -75,18 +75,28 @@ class DeviceGBTExecutor:
         self.msq = 0.0
         self._models_dev = None
 
-    def _f0(self, loss, y_tr):
-        if loss == LOSS_LOGISTIC:
-            p = float(np.clip(np.mean(y_tr), 1e-6, 1 - 1e-6))
-            return float(np.log(p / (1.0 - p)))
-        return float(np.mean(y_tr))
+    supports_sample_weight = True
 
-    def fit(self, train_idx, plan):
+    def _f0(self, loss, y_tr, w=None):
+        m = float(np.mean(y_tr)) if w is None else \
+            float(np.sum(w * y_tr) / max(np.sum(w), 1e-12))
+        if loss == LOSS_LOGISTIC:
+            p = float(np.clip(m, 1e-6, 1 - 1e-6))
+            return float(np.log(p / (1.0 - p)))
+        return m
+
+    def fit(self, train_idx, plan, sample_weight=None):
         torch = self.torch
         idx_np = np.asarray(train_idx, dtype=np.int32)
         idx_t = torch.as_tensor(idx_np).to(self.device)
         y_tr_np = self.y_np[idx_np]
         y_tr = torch.as_tensor(y_tr_np).to(self.device)
+        # fold-sliced weights: the grad kernels index rows of the fold
+        w_np = None
+        sw_t = torch.empty(0, dtype=torch.float32, device=self.device)
+        if sample_weight is not None:
+            w_np = np.asarray(sample_weight, np.float32)[idx_np]
+            sw_t = torch.as_tensor(w_np).to(self.device)
 
         models_host, models_dev = [], []
         n_iter = np.empty(plan.B, np.int32)
@@ -98,12 +108,17 @@ class DeviceGBTExecutor:
                 n_iter[i] = 0
                 continue
             if K > 2:
-                prior = np.bincount(y_tr_np.astype(np.int64),
-                                    minlength=K) / len(y_tr_np)
+                if w_np is None:
+                    prior = np.bincount(y_tr_np.astype(np.int64),
+                                        minlength=K) / len(y_tr_np)
+                else:
+                    prior = np.bincount(
+                        y_tr_np.astype(np.int64), weights=w_np,
+                        minlength=K) / max(float(w_np.sum()), 1e-12)
                 f0 = np.log(np.clip(prior, 1e-6, None)).astype(np.float32)
                 f0_t = torch.as_tensor(f0).to(self.device)
                 feats, thrs, vals = self.ext.gbt_fit_multi(
-                    self.Xb, self.Xbt, idx_t, y_tr, f0_t, K,
+                    self.Xb, self.Xbt, idx_t, y_tr, f0_t, sw_t, K,
                     spec.n_estimators,
                     spec.learning_rate, spec.max_depth,
                     spec.min_samples_leaf, spec.l2_regularization)
@@ -117,12 +132,12 @@ class DeviceGBTExecutor:
                 models_host.append((trees, f0))
                 n_iter[i] = spec.n_estimators
                 continue
-            f0 = self._f0(spec.loss, y_tr_np)
+            f0 = self._f0(spec.loss, y_tr_np, w_np)
             feats, thrs, vals = self.ext.gbt_fit(
                 self.Xb, self.Xbt, idx_t, y_tr, spec.loss,
                 spec.n_estimators,
                 spec.learning_rate, spec.max_depth, spec.min_samples_leaf,
-                spec.l2_regularization, f0)
+                spec.l2_regularization, f0, sw_t)
             models_dev.append((feats, thrs, vals, f0, spec.loss))
             f_np = feats.cpu().numpy()
             t_np = thrs.cpu().numpy().astype(np.uint8)

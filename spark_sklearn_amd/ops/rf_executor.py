@@ -55,6 +55,8 @@ class _LazyForests:
 
 
 class DeviceRFExecutor(DeviceGBTExecutor):
+    supports_sample_weight = False    # rf_fit has no weighted gini path
+
     def fit(self, train_idx, plan):
         torch = self.torch
         idx_np = np.asarray(train_idx, dtype=np.int32)

@@ -53,6 +53,42 @@ def test_device_trees_match_reference(problem):
     assert agree >= 6, f"only {agree}/8 trees match the reference"
 
 
+def test_device_weighted_trees_match_reference(problem):
+    """sample_weight on device (grad-kernel w scaling) vs the weighted
+    numpy reference: same structures, same weighted f0."""
+    X, y = problem
+    dev = torch.device("cuda", 0)
+    rng = np.random.default_rng(4)
+    w = rng.uniform(0.25, 4.0, size=X.shape[0]).astype(np.float64)
+    w = (w * (len(w) / w.sum()))
+    ex = DeviceGBTExecutor(X, y.astype(np.float32), dev)
+    spec = GBTSpec(n_estimators=6, max_depth=4, learning_rate=0.3,
+                   min_samples_leaf=20)
+    plan = _GBTPlan([spec])
+    train = np.arange(15000, dtype=np.int64)
+    (models, edges), n_iter, _ = ex.fit(train, plan, sample_weight=w)
+    trees_d, f0_d = models[0]
+
+    Xb = bin_data(X, ex.edges)
+    trees_r, f0_r = gbt_fit_ref(
+        Xb[train], y[train].astype(np.float32), spec,
+        sample_weight=w[train].astype(np.float32))
+    assert abs(f0_d - f0_r) < 1e-5
+    agree = 0
+    for (fd, td, vd), (fr, tr, vr) in zip(trees_d, trees_r):
+        if np.array_equal(fd, fr) and np.array_equal(td, tr):
+            agree += 1
+            np.testing.assert_allclose(vd, vr, rtol=1e-3, atol=1e-5)
+    assert agree >= 4, f"only {agree}/6 weighted trees match"
+
+    # unit weights reproduce the unweighted device model exactly
+    (m1, _), _, _ = ex.fit(train, plan,
+                           sample_weight=np.ones(X.shape[0]))
+    (m0, _), _, _ = ex.fit(train, plan)
+    for (fa, ta, va), (fb, tb, vb) in zip(m1[0][0], m0[0][0]):
+        assert np.array_equal(fa, fb) and np.array_equal(ta, tb)
+
+
 def test_device_scores_match_reference(problem):
     X, y = problem
     dev = torch.device("cuda", 0)

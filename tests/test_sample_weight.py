@@ -93,6 +93,79 @@ def test_grid_search_routes_sample_weight_without_ignore_warning():
                            gs0.cv_results_["mean_test_score"])
 
 
+def test_gbt_sample_weight_reference_semantics():
+    """GBT honors sample_weight: w=1 reproduces the unweighted model
+    exactly, w=2 equals duplicated rows (histogram sums are linear in
+    w), and upweighting a class moves predictions toward it."""
+    from spark_sklearn_amd.models._gbt_ref import (
+        GBTSpec, bin_data, gbt_fit_ref, gbt_predict_raw, quantile_bins)
+
+    X, y = _data(n=900, f=8, seed=5)
+    edges = quantile_bins(X)
+    Xb = bin_data(X, edges)
+    spec = GBTSpec(loss=0, n_estimators=5, learning_rate=0.3, max_depth=3,
+                   min_samples_leaf=1, l2_regularization=0.0)
+    yf = y.astype(np.float32)
+    t0, f00 = gbt_fit_ref(Xb, yf, spec)
+    t1, f01 = gbt_fit_ref(Xb, yf, spec, sample_weight=np.ones(900,
+                                                             np.float32))
+    assert f00 == f01
+    for (fa, ta, va), (fb, tb, vb) in zip(t0, t1):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_array_equal(ta, tb)
+        np.testing.assert_allclose(va, vb)
+
+    # w=2 on a subset == duplicating those rows (min_samples_leaf=1 so
+    # the count channel cannot diverge the split sets)
+    dup = np.arange(0, 900, 3)
+    Xb2 = np.vstack([Xb, Xb[dup]])
+    y2 = np.concatenate([yf, yf[dup]])
+    w = np.ones(900, np.float32)
+    w[dup] = 2.0
+    tw, f0w = gbt_fit_ref(Xb, yf, spec, sample_weight=w)
+    td, f0d = gbt_fit_ref(Xb2, y2, spec)
+    assert abs(f0w - f0d) < 1e-6
+    for (fa, ta, va), (fb, tb, vb) in zip(tw, td):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_array_equal(ta, tb)
+        np.testing.assert_allclose(va, vb, rtol=1e-4, atol=1e-6)
+
+    # upweighting class 1 moves raw scores up
+    wup = np.where(y == 1, 8.0, 1.0).astype(np.float32)
+    tu, f0u = gbt_fit_ref(Xb, yf, spec, sample_weight=wup)
+    r_base = gbt_predict_raw(Xb, t0, f00)
+    r_up = gbt_predict_raw(Xb, tu, f0u)
+    assert r_up.mean() > r_base.mean()
+
+
+def test_gbt_grid_routes_sample_weight():
+    from spark_sklearn_amd.models import GradientBoostingClassifier
+
+    X, y = _data(n=800, f=8, seed=2)
+    w = np.where(y == 1, 5.0, 1.0)
+    gs = ska.GridSearchCV(
+        None, GradientBoostingClassifier(n_estimators=5, max_depth=2),
+        {"learning_rate": [0.1, 0.3]}, cv=2)
+    with warnings.catch_warnings():
+        warnings.filterwarnings("error", message=".*IGNORED.*")
+        gs.fit(X, y, sample_weight=w)
+    gs0 = ska.GridSearchCV(
+        None, GradientBoostingClassifier(n_estimators=5, max_depth=2),
+        {"learning_rate": [0.1, 0.3]}, cv=2)
+    gs0.fit(X, y)
+    assert not np.allclose(gs.cv_results_["mean_test_score"],
+                           gs0.cv_results_["mean_test_score"])
+
+
+def test_rf_sample_weight_warns_ignored():
+    from spark_sklearn_amd.models import RandomForestClassifier
+
+    X, y = _data(n=600, f=8)
+    with pytest.warns(UserWarning, match="IGNORED"):
+        RandomForestClassifier(n_estimators=3, max_depth=3).fit(
+            X, y, sample_weight=np.ones(600))
+
+
 def test_kmeans_sample_weight_warns_ignored():
     X, _ = _data(n=800, f=8)
     w = np.ones(800)
