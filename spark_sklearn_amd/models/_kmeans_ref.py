@@ -100,8 +100,9 @@ def bf16_round_arr(a: np.ndarray) -> np.ndarray:
     return r.astype(np.uint32).view(np.float32)
 
 
-def _assign(X, C_all, plan):
-    """labels [n, n_blocks] int32 + per-block inertia (float64)."""
+def _assign(X, C_all, plan, w=None):
+    """labels [n, n_blocks] int32 + per-block inertia (float64);
+    w weights the inertia terms (assignment itself is weight-free)."""
     C_all = bf16_round_arr(C_all)
     S = X @ C_all.T                                   # [n, K_total]
     cn = np.sum(C_all.astype(np.float64) ** 2, axis=1)
@@ -114,18 +115,29 @@ def _assign(X, C_all, plan):
         q = cn[lo:hi][np.newaxis, :] - 2.0 * S[:, lo:hi]
         lab = np.argmin(q, axis=1)
         labels[:, b] = lab
-        inertia[b] = float(np.sum(xn + q[np.arange(n), lab]))
+        terms = xn + q[np.arange(n), lab]
+        if w is not None:
+            terms = terms * w
+        inertia[b] = float(np.sum(terms))
     return labels, inertia
 
 
-def kmeans_fit_ref(X: np.ndarray, C0_all: np.ndarray, plan: KMeansPlan):
+def kmeans_fit_ref(X: np.ndarray, C0_all: np.ndarray, plan: KMeansPlan,
+                   sample_weight=None):
     """Run Lloyd iterations for all blocks simultaneously.
 
     Returns (C_all [K_total, F] fp32, n_iter [n_blocks], inertia
     [n_blocks] -- inertia of the FINAL assignment).
+
+    sample_weight: weighted centroid means / inertia (assignment is
+    weight-free).  The device path quantizes w to bf16-representable
+    values for its one-hot GEMM; pass the same through bf16_round_arr
+    when checking parity.
     """
     X = np.ascontiguousarray(X, dtype=np.float32)
     C_all = np.array(C0_all, dtype=np.float32, copy=True)
+    w = None if sample_weight is None else \
+        np.asarray(sample_weight, dtype=np.float64)
     nb = plan.n_blocks
     active = np.ones(nb, dtype=bool)
     n_iter = np.zeros(nb, dtype=np.int32)
@@ -136,15 +148,21 @@ def kmeans_fit_ref(X: np.ndarray, C0_all: np.ndarray, plan: KMeansPlan):
         epoch_active = active & (it < plan.max_iter)
         if not epoch_active.any():
             break
-        labels, cur_inertia = _assign(X, C_all, plan)
+        labels, cur_inertia = _assign(X, C_all, plan, w)
         for b in range(nb):
             if not epoch_active[b]:
                 continue
             lo, hi = int(plan.cstart[b]), int(plan.cstart[b + 1])
             k = hi - lo
             sums = np.zeros((k, X.shape[1]), dtype=np.float64)
-            counts = np.bincount(labels[:, b], minlength=k).astype(np.float64)
-            np.add.at(sums, labels[:, b], X)
+            if w is None:
+                counts = np.bincount(labels[:, b],
+                                     minlength=k).astype(np.float64)
+                np.add.at(sums, labels[:, b], X)
+            else:
+                counts = np.bincount(labels[:, b], weights=w,
+                                     minlength=k).astype(np.float64)
+                np.add.at(sums, labels[:, b], X * w[:, np.newaxis])
             new = C_all[lo:hi].astype(np.float64)
             nz = counts > 0
             new[nz] = sums[nz] / counts[nz, np.newaxis]
@@ -155,7 +173,7 @@ def kmeans_fit_ref(X: np.ndarray, C0_all: np.ndarray, plan: KMeansPlan):
             if shift <= plan.tol_scaled[b]:
                 active[b] = False
     # final inertia under the FINAL centers
-    _, inertia = _assign(X, C_all, plan)
+    _, inertia = _assign(X, C_all, plan, w)
     return C_all, n_iter, inertia
 
 

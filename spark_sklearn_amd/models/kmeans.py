@@ -57,18 +57,26 @@ class KMeans(ClusterMixin, BaseEstimator):
 
     # -- standalone fit ----------------------------------------------------
     def fit(self, X, y=None, sample_weight=None):
-        from ..engine import warn_sample_weight_ignored
-
-        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import _device_or_none, get_family
+
+        from ..engine import _route_fit_params
 
         family = get_family(self)
         device = _device_or_none(None)
         executor = family.make_executor(X, None, None, device)
         n = X.shape[0]
+        fp = {"sample_weight": sample_weight} \
+            if sample_weight is not None else {}
+        sw, _, ignored = _route_fit_params(fp, executor, X)
+        if ignored:
+            import warnings
+
+            warnings.warn("sample_weight is IGNORED by this KMeans "
+                          "executor tier", stacklevel=2)
         plan = family.build_plan(self, [{}], n, executor, 0, "raise")[0]
         all_idx = np.arange(n, dtype=np.int64)
-        fitted, n_iter, _ = executor.fit(all_idx, plan)
+        fit_kw = {"sample_weight": sw} if sw is not None else {}
+        fitted, n_iter, _ = executor.fit(all_idx, plan, **fit_kw)
         family.install_fitted(self, fitted, 0, plan, None)
         C_all, best_block, _ = fitted
         self.labels_ = executor.assign_labels(all_idx, self.cluster_centers_)
@@ -93,12 +101,12 @@ class KMeans(ClusterMixin, BaseEstimator):
               np.sum(C ** 2, axis=1)[None, :])
         return np.sqrt(np.maximum(d2, 0.0))
 
-    def fit_predict(self, X, y=None):
-        return self.fit(X, y).labels_
+    def fit_predict(self, X, y=None, sample_weight=None):
+        return self.fit(X, y, sample_weight=sample_weight).labels_
 
-    def fit_transform(self, X, y=None):
+    def fit_transform(self, X, y=None, sample_weight=None):
         """Fit, then distances to the fitted centers (sklearn contract)."""
-        return self.fit(X, y).transform(X)
+        return self.fit(X, y, sample_weight=sample_weight).transform(X)
 
     def score(self, X, y=None):
         """Negative inertia of X under the fitted centers (sklearn)."""
@@ -256,9 +264,13 @@ class _HostKMeansExecutor:
                 self._init_sample = self.X[rows]
         return self._init_sample
 
-    def fit(self, train_idx, plan):
+    supports_sample_weight = True
+
+    def fit(self, train_idx, plan, sample_weight=None):
+        sw = (np.asarray(sample_weight, np.float64)[train_idx]
+              if sample_weight is not None else None)
         C_all, n_iter, inertia = kmeans_fit_ref(
-            self.X[train_idx], plan.C0, plan)
+            self.X[train_idx], plan.C0, plan, sample_weight=sw)
         best_block = np.empty(plan.n_cand, np.int32)
         for c in range(plan.n_cand):
             blocks = np.flatnonzero(plan.block_to_cand == c)

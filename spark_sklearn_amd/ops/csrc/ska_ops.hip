@@ -1237,7 +1237,8 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
                           float* __restrict__ inertia,      // [nb]
                           float* __restrict__ counts,       // [Kp] or null
                           int nt, int NTp, int Fp, int Kp, int nb,
-                          int want_labels) {
+                          int want_labels,
+                          const float* __restrict__ w) {
   constexpr int cpw = NF * 16;
   const int Kpad = NF * 64;
   extern __shared__ char km_smem[];
@@ -1326,8 +1327,9 @@ void kmeans_assign_kernel(const __bf16* __restrict__ X,
       if (q[j] < best) { best = q[j]; bj = j - lo; }
     if (want_labels)
       labels[(long)b * NTp + gr] = (unsigned char)bj;
-    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], 1.f);
-    atomicAdd(&sIn[b], xnorm[sIdx[row]] + best);
+    float wt = (w != nullptr) ? w[sIdx[row]] : 1.f;
+    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], wt);
+    atomicAdd(&sIn[b], wt * (xnorm[sIdx[row]] + best));
   }
   __syncthreads();
   for (int b = tid; b < nb; b += 256) atomicAdd(&inertia[b], sIn[b]);
@@ -1352,7 +1354,8 @@ void kmeans_assign_w_kernel(const __bf16* __restrict__ X,
                             float* __restrict__ inertia,
                             float* __restrict__ counts,
                             int nt, int NTp, int Fp, int Kp, int nb,
-                            int want_labels) {
+                            int want_labels,
+                            const float* __restrict__ w) {
   constexpr int cpw = NF * 16;
   const int Kpad = NF * 64;
   extern __shared__ char kw_smem[];
@@ -1445,8 +1448,9 @@ void kmeans_assign_w_kernel(const __bf16* __restrict__ X,
       if (q[j] < best) { best = q[j]; bj = j - lo; }
     if (want_labels)
       labels[(long)b * NTp + gr] = (unsigned char)bj;
-    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], 1.f);
-    atomicAdd(&sIn[b], xnorm[sIdx[row]] + best);
+    float wt = (w != nullptr) ? w[sIdx[row]] : 1.f;
+    if (counts != nullptr) atomicAdd(&sCnt[lo + bj], wt);
+    atomicAdd(&sIn[b], wt * (xnorm[sIdx[row]] + best));
   }
   __syncthreads();
   for (int b = tid; b < nb; b += 512) atomicAdd(&inertia[b], sIn[b]);
@@ -1461,12 +1465,18 @@ void kmeans_assign_w_kernel(const __bf16* __restrict__ X,
 __global__ void kmeans_onehot_kernel(const unsigned char* __restrict__ labels,
                                      const int* __restrict__ cstart,
                                      __bf16* __restrict__ Dt,  // [Kp][NTp]
-                                     long nt, long NTp, int nb) {
+                                     long nt, long NTp, int nb,
+                                     const int* __restrict__ idx,
+                                     const float* __restrict__ w) {
   long r = blockIdx.x * blockDim.x + threadIdx.x;
   for (; r < nt; r += (long)gridDim.x * blockDim.x) {
+    // sample_weight: the one-hot entry becomes w (pre-quantized to a
+    // bf16-representable value on the host, so the cast is exact and
+    // the weighted-count denominator matches the GEMM numerator)
+    float wt = (w != nullptr) ? w[idx[r]] : 1.f;
     for (int b = 0; b < nb; ++b) {
       int c = cstart[b] + labels[(long)b * NTp + r];
-      Dt[(long)c * NTp + r] = (__bf16)1.0f;
+      Dt[(long)c * NTp + r] = (__bf16)wt;
     }
   }
 }
@@ -5260,7 +5270,8 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
                                  unsigned char* labels, float* inertia,
                                  float* counts, int nt, int NTp, int Fp,
                                  int Kp, int nb, int want_labels,
-                                 hipStream_t s) {
+                                 hipStream_t s,
+                                 const float* w = nullptr) {
   if (Kp <= 128 && NTp % 128 == 0) {
     size_t wsm = (size_t)128 * (Kp + 8) * 4 + 256 * 4 + (size_t)Kp * 4 +
                  128 * 4 + (size_t)128 * 64 * 2 + (size_t)Kp * 64 * 2;
@@ -5271,7 +5282,7 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)wsm));
       hipLaunchKernelGGL(k, wgrid, dim3(512), wsm, s, X, idx, Ct, cnorm,
                          xnorm, cstart, labels, inertia, counts, nt, NTp,
-                         Fp, Kp, nb, want_labels);
+                         Fp, Kp, nb, want_labels, w);
     };
     if (Kp == 64) gow(kmeans_assign_w_kernel<1>);
     else gow(kmeans_assign_w_kernel<2>);
@@ -5287,7 +5298,7 @@ static void launch_kmeans_assign(const __bf16* X, const int* idx,
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem));
     hipLaunchKernelGGL(k, grid, dim3(256), smem, s, X, idx, Ct, cnorm,
                        xnorm, cstart, labels, inertia, counts, nt, NTp,
-                       Fp, Kp, nb, want_labels);
+                       Fp, Kp, nb, want_labels, w);
   };
   switch (Kp / 64) {
     case 1: go(kmeans_assign_kernel<1>); break;
@@ -5305,9 +5316,16 @@ std::vector<torch::Tensor> kmeans_fit(
     torch::Tensor X, torch::Tensor Xt, torch::Tensor idx, torch::Tensor C0,
     torch::Tensor cstart, torch::Tensor cen_block, torch::Tensor max_iter,
     torch::Tensor tol, torch::Tensor xnorm, int64_t nt, int64_t max_epochs,
-    int64_t check_every) {
+    int64_t check_every, torch::Tensor sw) {
   TORCH_CHECK(g_layout == 3,
               "kmeans kernels require the probed layout P=3 (MI355X)");
+  // sample_weight [n_rows] (bf16-representable values, engine-normalized):
+  // weighted counts/sums/inertia; assignment is weight-free
+  const float* wp = nullptr;
+  if (sw.numel() > 0) {
+    TORCH_CHECK(sw.dtype() == torch::kFloat32 && sw.size(0) == X.size(0));
+    wp = sw.data_ptr<float>();
+  }
   const at::cuda::CUDAGuard guard(X.device());
   int Fp = (int)X.size(1);
   int Kp = (int)C0.size(0);
@@ -5366,7 +5384,7 @@ std::vector<torch::Tensor> kmeans_fit(
                          labels.data_ptr<unsigned char>(),
                          inertia.data_ptr<float>(),
                          counts.data_ptr<float>(), (int)nt, (int)NTp, Fp,
-                         Kp, nb, 1, s);
+                         Kp, nb, 1, s, wp);
     // sums = X^T . onehot(labels): the same MFMA bwd kernel as the
     // linear family (one-hot rows are exact in bf16)
     HIP_CHECK(hipMemsetAsync(Dhot.data_ptr(), 0,
@@ -5375,7 +5393,7 @@ std::vector<torch::Tensor> kmeans_fit(
                        dim3(256), 0, s, labels.data_ptr<unsigned char>(),
                        cstart.data_ptr<int>(),
                        reinterpret_cast<__bf16*>(Dhot.data_ptr()), nt,
-                       (long)NTp, nb);
+                       (long)NTp, nb, idx.data_ptr<int>(), wp);
     switch (Kp / 64) {
       case 1:
         hipLaunchKernelGGL(bwd_g3_kernel<1>, dim3(ftiles, (unsigned)nseg),
@@ -5437,7 +5455,7 @@ std::vector<torch::Tensor> kmeans_fit(
                        cstart.data_ptr<int>(),
                        labels.data_ptr<unsigned char>(),
                        inertia.data_ptr<float>(), nullptr, (int)nt,
-                       (int)NTp, Fp, Kp, nb, 1, s);
+                       (int)NTp, Fp, Kp, nb, 1, s, wp);
   return {C, n_iter, inertia, Ct, cnorm, labels};
 }
 

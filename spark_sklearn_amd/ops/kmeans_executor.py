@@ -60,7 +60,23 @@ class DeviceKMeansExecutor:
         out[:n] = np.asarray(idx, dtype=np.int32)
         return torch.as_tensor(out).to(self.device), n
 
-    def fit(self, train_idx, plan):
+    supports_sample_weight = True
+
+    def _cached_weights(self, w):
+        """Full-length per-row weights, quantized to bf16-representable
+        values so the one-hot GEMM numerator and the fp32 weighted-count
+        denominator see the SAME w (kernels index by global row id)."""
+        hit = getattr(self, "_w_cache", None)
+        if hit is not None and hit[0] is w:
+            return hit[1]
+        torch = self.torch
+        t = torch.as_tensor(
+            np.ascontiguousarray(np.asarray(w, dtype=np.float32))
+        ).to(self.device).to(torch.bfloat16).to(torch.float32)
+        self._w_cache = (w, t)
+        return t
+
+    def fit(self, train_idx, plan, sample_weight=None):
         torch = self.torch
         idx_t, nt = self._pad_idx(train_idx)
         K_total = int(plan.cstart[-1])
@@ -89,9 +105,12 @@ class DeviceKMeansExecutor:
 
         max_epochs = int(plan.max_iter.max()) if nb else 0
         check_every = 8 if max_epochs > 12 else 0
+        sw_t = (self._cached_weights(sample_weight)
+                if sample_weight is not None else
+                torch.empty(0, dtype=torch.float32, device=dev))
         C, n_iter, inertia, Ct, cnorm, labels = self.ext.kmeans_fit(
             self.X, self.Xt, idx_t, C0_t, cstart_t, cen_block_t, mi_t,
-            tol_t, self.xnorm, nt, max_epochs, check_every)
+            tol_t, self.xnorm, nt, max_epochs, check_every, sw_t)
         torch.cuda.synchronize(dev)
         self._score_state = (Ct, cnorm, cstart_t)
 

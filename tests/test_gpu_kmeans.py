@@ -61,6 +61,33 @@ def test_device_matches_ref_short(blobs):
     assert list(ni_d) == list(ni_r)
 
 
+def test_device_weighted_matches_ref(blobs):
+    """sample_weight on device (weighted one-hot GEMM + weighted counts/
+    inertia) vs the weighted reference with the SAME bf16-quantized w."""
+    X, _ = blobs
+    Xb = _bf16_round(X)
+    rng = np.random.default_rng(9)
+    w = rng.uniform(0.25, 4.0, size=X.shape[0])
+    w_q = _bf16_round(w.astype(np.float32)).astype(np.float64)
+    plan = _plan([5, 8], max_iter=2, X=Xb)
+    dev = torch.device("cuda", 0)
+    ex = DeviceKMeansExecutor(X, dev)
+    train = np.arange(15000, dtype=np.int64)
+    (C_d, bb_d, in_d), ni_d, _ = ex.fit(train, plan, sample_weight=w)
+    C_r, ni_r, in_r = kmeans_fit_ref(Xb[train], plan.C0, plan,
+                                     sample_weight=w_q[train])
+    np.testing.assert_allclose(in_d, in_r, rtol=5e-3)
+    np.testing.assert_allclose(C_d, C_r, rtol=0.05, atol=0.05)
+    assert list(ni_d) == list(ni_r)
+
+    # unit weights == unweighted, same device tier
+    (C_1, _, in_1), _, _ = ex.fit(train, plan,
+                                  sample_weight=np.ones(X.shape[0]))
+    (C_0, _, in_0), _, _ = ex.fit(train, plan)
+    np.testing.assert_allclose(C_1, C_0, rtol=1e-5, atol=1e-5)
+    np.testing.assert_allclose(in_1, in_0, rtol=1e-5)
+
+
 def test_device_converged_quality(blobs):
     """At convergence, device and reference land at equal-quality optima
     (labels near ambiguous splits may flip under bf16, so trajectories --

@@ -166,14 +166,36 @@ def test_rf_sample_weight_warns_ignored():
             X, y, sample_weight=np.ones(600))
 
 
-def test_kmeans_sample_weight_warns_ignored():
-    X, _ = _data(n=800, f=8)
-    w = np.ones(800)
-    gs = ska.GridSearchCV(None, KMeans(n_clusters=3, max_iter=5,
+def test_kmeans_sample_weight_weighted_centroids():
+    """KMeans honors sample_weight: w=1 == unweighted exactly; a heavy
+    cluster pulls its centroid toward the heavy points (weighted-mean
+    update, reference tier)."""
+    rng = np.random.default_rng(6)
+    base = np.concatenate([rng.normal(-3, 0.3, (200, 2)),
+                           rng.normal(+3, 0.3, (200, 2))]).astype(np.float32)
+    m0 = KMeans(n_clusters=2, max_iter=20, random_state=0).fit(base)
+    m1 = KMeans(n_clusters=2, max_iter=20, random_state=0).fit(
+        base, sample_weight=np.ones(400))
+    np.testing.assert_array_equal(m0.cluster_centers_, m1.cluster_centers_)
+
+    # upweight the 20 right-most points of the right blob: that
+    # centroid's x moves right
+    w = np.ones(400)
+    right = np.argsort(base[:, 0])[-20:]
+    w[right] = 50.0
+    m2 = KMeans(n_clusters=2, max_iter=20, random_state=0).fit(
+        base, sample_weight=w)
+    cx0 = sorted(m0.cluster_centers_[:, 0])
+    cx2 = sorted(m2.cluster_centers_[:, 0])
+    assert cx2[1] > cx0[1] + 0.1
+
+    # grid routing: no ignored-warning
+    gs = ska.GridSearchCV(None, KMeans(n_clusters=2, max_iter=5,
                                        random_state=0),
                           {"n_clusters": [2, 3]}, cv=2)
-    with pytest.warns(UserWarning, match="IGNORED"):
-        gs.fit(X, sample_weight=w)
+    with warnings.catch_warnings():
+        warnings.filterwarnings("error", message=".*IGNORED.*")
+        gs.fit(base, sample_weight=w)
 
 
 def test_sample_weight_validation():
