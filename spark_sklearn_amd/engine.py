@@ -278,14 +278,13 @@ def _encode_cached(family, estimator, y):
 
 
 def warn_sample_weight_ignored(sample_weight, name):
-    """Loud divergence marker for families without device weighting
-    (KMeans / GBT / RandomForest): the argument is accepted for sklearn
+    """Loud divergence marker for executors without weighting (today:
+    the CSR-capacity linear tier): the argument is accepted for sklearn
     signature parity but the fit runs unweighted."""
     if sample_weight is not None:
         warnings.warn(
-            f"sample_weight is IGNORED by {name}: the batched device "
-            "path fits unweighted (linear-family estimators honor it; "
-            "tree/kmeans support is a roadmap item)", stacklevel=3)
+            f"sample_weight is IGNORED by {name}: this executor tier "
+            "fits unweighted", stacklevel=3)
 
 
 def _route_fit_params(fit_params, executor, X):

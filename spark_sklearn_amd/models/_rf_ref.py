@@ -83,8 +83,13 @@ def _selected_features(F, kfeat, fseed, nd):
     return sel
 
 
-def _build_rf_tree(Xb, gplanes, spec, fseed):
-    """One tree; gplanes [KP][n] class-indicator (or y) channels."""
+def _build_rf_tree(Xb, gplanes, spec, fseed, w=None):
+    """One tree; gplanes [KP][n] class-indicator (or y) channels.
+
+    w (per-row sample_weight, already applied to gplanes by the caller):
+    gini/MSE denominators and leaf values use the weighted mass sum(w),
+    while min_samples_leaf keeps plain row COUNTS (sklearn semantics --
+    weighted leaf mass is min_weight_fraction_leaf territory)."""
     n, F = Xb.shape
     KP = gplanes.shape[0]
     NN = 2 ** (spec.max_depth + 1) - 1
@@ -106,7 +111,18 @@ def _build_rf_tree(Xb, gplanes, spec, fseed):
             rows = live[node[live] == nd]
             cnt = len(rows)
             G = gplanes[:, rows].sum(axis=1)            # [KP]
-            parent = float(np.sum(G * G) / (cnt + eps))
+            if KP > 1:
+                # classes partition the mass: the denominator is the SUM
+                # of the class numerators (same accumulation), so pure
+                # nodes cancel to exactly zero gain -- mirrors the
+                # device split kernel, which derives the mass from the
+                # class-g prefixes for the same reason
+                wmass = float(np.sum(G))
+            elif w is None:
+                wmass = cnt
+            else:
+                wmass = float(np.sum(w[rows]))
+            parent = float(np.sum(G * G) / (wmass + eps))
             sel = _selected_features(F, spec.max_features, fseed, nd)
             best_gain, best_f, best_t = 0.0, -1, 0
             for f in range(F):
@@ -116,13 +132,25 @@ def _build_rf_tree(Xb, gplanes, spec, fseed):
                 ch = np.bincount(bins, minlength=256)
                 cl = np.cumsum(ch)[:-1].astype(np.float64)
                 cr = cnt - cl
-                gain = np.zeros(255)
+                gls = []
                 for c in range(KP):
                     gh = np.bincount(bins, weights=gplanes[c, rows],
                                      minlength=256)
-                    gl = np.cumsum(gh)[:-1]
+                    gls.append(np.cumsum(gh)[:-1])
+                if KP > 1:
+                    hl = np.sum(gls, axis=0)
+                    hr = wmass - hl
+                elif w is None:
+                    hl, hr = cl, cr
+                else:
+                    hh = np.bincount(bins, weights=w[rows], minlength=256)
+                    hl = np.cumsum(hh)[:-1]
+                    hr = wmass - hl
+                gain = np.zeros(255)
+                for c in range(KP):
+                    gl = gls[c]
                     gr = G[c] - gl
-                    gain += gl * gl / (cl + eps) + gr * gr / (cr + eps)
+                    gain += gl * gl / (hl + eps) + gr * gr / (hr + eps)
                 gain -= parent
                 okay = (cl >= spec.min_samples_leaf) & \
                     (cr >= spec.min_samples_leaf)
@@ -144,7 +172,7 @@ def _build_rf_tree(Xb, gplanes, spec, fseed):
                     best_gain, best_f, best_t = float(gain[t]), f, t
             if best_f < 0:
                 feat[nd] = -1
-                val[:, nd] = (G / (cnt + eps)).astype(np.float32)
+                val[:, nd] = (G / (wmass + eps)).astype(np.float32)
                 alive[rows] = False
                 continue
             feat[nd] = best_f
@@ -154,14 +182,20 @@ def _build_rf_tree(Xb, gplanes, spec, fseed):
     live = np.flatnonzero(alive)
     for nd in np.unique(node[live]):
         rows = live[node[live] == nd]
-        cnt = len(rows)
+        G = gplanes[:, rows].sum(axis=1)
+        if KP > 1:
+            wmass = float(np.sum(G))
+        elif w is None:
+            wmass = len(rows)
+        else:
+            wmass = float(np.sum(w[rows]))
         feat[nd] = -1
-        val[:, nd] = (gplanes[:, rows].sum(axis=1) / (cnt + eps)).astype(
-            np.float32)
+        val[:, nd] = (G / (wmass + eps)).astype(np.float32)
     return feat, thr, val
 
 
-def rf_fit_ref(Xb: np.ndarray, y: np.ndarray, spec: RFSpec):
+def rf_fit_ref(Xb: np.ndarray, y: np.ndarray, spec: RFSpec,
+               sample_weight=None):
     """Returns trees in the gbt walk format.
 
     Binary/regression: list of (feat, thr, val); K>2: list of per-class
@@ -182,12 +216,16 @@ def rf_fit_ref(Xb: np.ndarray, y: np.ndarray, spec: RFSpec):
             pos = np.arange(n)
         Xb_t = Xb[pos]
         y_t = y[pos]
+        w_t = None if sample_weight is None else \
+            np.asarray(sample_weight, np.float64)[pos]
         if KP > 1:
             gpl = np.stack([(y_t == c).astype(np.float32)
                             for c in range(KP)])
         else:
             gpl = y_t.astype(np.float32)[None, :]
-        feat, thr, val = _build_rf_tree(Xb_t, gpl, spec, fs)
+        if w_t is not None:
+            gpl = (gpl * w_t[np.newaxis, :]).astype(np.float32)
+        feat, thr, val = _build_rf_tree(Xb_t, gpl, spec, fs, w_t)
         val = val / np.float32(spec.n_estimators)
         if KP > 1:
             trees.append([(feat, thr, val[c]) for c in range(KP)])

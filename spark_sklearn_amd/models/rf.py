@@ -87,12 +87,9 @@ class _RFBase(BaseEstimator):
         )
 
     def fit(self, X, y, sample_weight=None):
-        from ..engine import warn_sample_weight_ignored
-
-        warn_sample_weight_ignored(sample_weight, type(self).__name__)
         from ..engine import fit_single
 
-        fit_single(self, X, y)
+        fit_single(self, X, y, sample_weight=sample_weight)
         return self
 
     @property
@@ -294,9 +291,13 @@ class _HostRFExecutor:
         self.K = len(classes) if classes is not None else 0
         self.msq = 0.0
 
-    def fit(self, train_idx, plan):
+    supports_sample_weight = True
+
+    def fit(self, train_idx, plan, sample_weight=None):
         Xb_tr = self.Xb[train_idx]
         y_tr = self.y[train_idx]
+        sw = (np.asarray(sample_weight, np.float64)[train_idx]
+              if sample_weight is not None else None)
         models = []
         n_iter = np.empty(plan.B, np.int32)
         for i, spec in enumerate(plan.specs):
@@ -304,7 +305,7 @@ class _HostRFExecutor:
                 models.append([])
                 n_iter[i] = 0
                 continue
-            trees = rf_fit_ref(Xb_tr, y_tr, spec)
+            trees = rf_fit_ref(Xb_tr, y_tr, spec, sample_weight=sw)
             models.append(trees)
             n_iter[i] = len(trees)
         return (models, self.edges), n_iter, None

@@ -1999,23 +1999,29 @@ __global__ void rf_bootstrap_kernel(const int* __restrict__ idx_in,
                                     const float* __restrict__ y_in,
                                     int* __restrict__ idx_out,
                                     float* __restrict__ y_out, long nt,
-                                    unsigned seed) {
+                                    unsigned seed,
+                                    const float* __restrict__ w_in,
+                                    float* __restrict__ w_out) {
   long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   for (; i < nt; i += (long)gridDim.x * blockDim.x) {
     unsigned r = rf_hash(seed ^ (unsigned)i);
     long p = (long)(((unsigned long long)r * (unsigned long long)nt) >> 32);
     idx_out[i] = idx_in[p];
     y_out[i] = y_in[p];
+    if (w_in != nullptr) w_out[i] = w_in[p];
   }
 }
 
 // per-class indicator planes for multiclass gini
 __global__ void rf_planes_kernel(const float* __restrict__ y_boot,
-                                 float* __restrict__ G, long nt, int K) {
+                                 float* __restrict__ G, long nt, int K,
+                                 const float* __restrict__ w) {
   long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
   for (; i < nt; i += (long)gridDim.x * blockDim.x) {
     int cls = (int)y_boot[i];
-    for (int c = 0; c < K; ++c) G[(long)c * nt + i] = (c == cls) ? 1.f : 0.f;
+    float wt = (w != nullptr) ? w[i] : 1.f;
+    for (int c = 0; c < K; ++c)
+      G[(long)c * nt + i] = (c == cls) ? wt : 0.f;
   }
 }
 
@@ -6304,6 +6310,7 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
   __shared__ int sFt[4], sTt[4];
   __shared__ int sRed[5];
   __shared__ float sCnt;
+  __shared__ float sH;
   __shared__ float sPC[64];
 
   unsigned fthr = 0xffffffffu;
@@ -6340,6 +6347,37 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
       }
     }
     if (f == 0 && lane == 0) sCnt = C;
+    // weighted mass per threshold: DERIVED as the sum of the class-g
+    // prefixes (classes partition the mass: sum_c w*1{y=c} == w).
+    // Reading a separate h channel instead would mix float-atomic g
+    // sums with fixed-point h sums -- pure nodes then get tiny nonzero
+    // gains and split spuriously.  Unweighted fits have integer-exact
+    // class counts, so this equals the count channel bit-for-bit.
+    float hl[4] = {0.f, 0.f, 0.f, 0.f};
+    float H = 0.f;
+    for (int c = 0; c < K; ++c) {
+      const float* fhc = FMT ? (fh0 + c) : (fh0 + (long)c * cstride);
+      float g4[4];
+      float lg = 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        g4[j] = fhc[(4 * lane + j) * cw];
+        lg += g4[j];
+      }
+      float pg = lg;
+      for (int off = 1; off < 64; off <<= 1) {
+        float t = __shfl_up(pg, off);
+        if (lane >= off) pg += t;
+      }
+      H += __shfl(pg, 63);
+      float a = pg - lg;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        a += g4[j];
+        hl[j] += a;
+      }
+    }
+    if (f == 0 && lane == 0) sH = H;
     int tstar = -1;
     if (etr) {
       int lmin = 256, lmax = -1;
@@ -6385,9 +6423,9 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
         a += g4[j];
         float gl = a, gr = G - gl;
         gainsum[j] +=
-            gl * gl / (cl[j] + eps) + gr * gr / ((C - cl[j]) + eps);
+            gl * gl / (hl[j] + eps) + gr * gr / ((H - hl[j]) + eps);
       }
-      parent += G * G / (C + eps);
+      parent += G * G / (H + eps);
     }
     if (active) {
 #pragma unroll
@@ -6433,11 +6471,18 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
         sTt[0] = sTt[w];
       }
     }
-    if (sFt[0] < 0 || sGain[0] <= 1e-12f) {
+    // PURE node: one class holds the whole mass (exact in fp32: the
+    // other class sums are sums of 0.0f).  The reference's f64 eps
+    // makes pure-node gains negative; fp32's x*x/x rounding noise can
+    // make them spuriously positive under weighted masses, so decide
+    // purity explicitly.
+    float gmx = 0.f;
+    for (int c = 0; c < K; ++c) gmx = fmaxf(gmx, sPC[c]);
+    if (sFt[0] < 0 || sGain[0] <= 1e-12f || gmx == sH) {
       for (int c = 0; c < K; ++c) {
         feats[(long)c * NN + nd] = -1;
         if (sCnt > 0.f)
-          vals[(long)c * NN + nd] = sPC[c] / (sCnt + eps);
+          vals[(long)c * NN + nd] = sPC[c] / (sH + eps);
       }
     } else {
       for (int c = 0; c < K; ++c) {
@@ -6648,12 +6693,24 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                                   int64_t n_estimators, int64_t max_depth,
                                   int64_t min_leaf, int64_t max_features,
                                   int64_t seed, int64_t bootstrap,
-                                  int64_t splitter_random) {
+                                  int64_t splitter_random,
+                                  torch::Tensor sw) {
   const at::cuda::CUDAGuard guard(Xb.device());
   TORCH_CHECK(Xb.dtype() == torch::kUInt8 && Xb.dim() == 2);
   TORCH_CHECK(max_depth >= 1 && max_depth <= 10,
               "device RF supports max_depth in [1, 10]");
   TORCH_CHECK(K <= 64, "device RF supports at most 64 classes");
+  // sample_weight (fold-sliced [nt]): weighted gini/MSE via w-scaled g
+  // planes + an h channel carrying sum(w) (min_samples_leaf keeps plain
+  // counts).  Weighted fits run the per-tree CLASSIC-layout path: the
+  // tree-batched compact cells have no h channel.
+  const bool weighted = sw.numel() > 0;
+  float wmax = 1.f;
+  if (weighted) {
+    TORCH_CHECK(sw.dtype() == torch::kFloat32 &&
+                sw.size(0) == idxmap0.size(0));
+    wmax = sw.max().item<float>();
+  }
   int F = (int)Xb.size(1);
   long nt = idxmap0.size(0);
   int T = (int)n_estimators;
@@ -6677,7 +6734,7 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
   int blocks = (int)std::min<long>((nt + 255) / 256, 2048);
   int bl_lo = (1 << max_depth) - 1;
 
-  if (nt <= 65536) {
+  if (nt <= 65536 && !weighted) {
     // ---- tree-batched small-n path (digits-scale forests are LAUNCH
     // bound: ~4 launches per tree-LEVEL here vs ~4 launches per tree-
     // level PER TREE in the large-n loop below).  A chunk of TB whole
@@ -6716,7 +6773,7 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
       if (!bootstrap)      // identical planes for every tree: share them
         hipLaunchKernelGGL(rf_planes_kernel, dim3(blocks), dim3(256), 0,
                            s, y_tr.data_ptr<float>(), G.data_ptr<float>(),
-                           nt, KP);
+                           nt, KP, nullptr);
     }
     for (int t0 = 0; t0 < T; t0 += TB) {
       int tb = std::min(TB, T - t0);
@@ -6818,30 +6875,39 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
   auto ones = torch::ones({nt}, optF);
   auto idx_t = bootstrap ? torch::empty({nt}, optI) : idxmap0;
   auto y_boot = bootstrap ? torch::empty({nt}, optF) : y_tr;
+  auto w_boot = (weighted && bootstrap) ? torch::empty({nt}, optF) : sw;
   auto G = (KP > 1) ? torch::empty({(long)KP, nt}, optF)
                     : torch::empty({0}, optF);
+  auto y_w = (weighted && KP == 1) ? torch::empty({nt}, optF)
+                                   : torch::empty({0}, optF);
 
   for (int t = 0; t < T; ++t) {
     unsigned st = rf_hash((unsigned)seed ^ (unsigned)(t * 0x9e3779b9u));
     unsigned fs = rf_hash(st ^ 0x85ebca6bu);
-    torch::Tensor idx_u = idx_t, y_u = y_boot;
+    torch::Tensor idx_u = idx_t, y_u = y_boot, w_u = w_boot;
     if (bootstrap) {
       hipLaunchKernelGGL(rf_bootstrap_kernel, dim3(blocks), dim3(256), 0, s,
                          idxmap0.data_ptr<int>(), y_tr.data_ptr<float>(),
                          idx_t.data_ptr<int>(), y_boot.data_ptr<float>(),
-                         nt, st);
+                         nt, st,
+                         weighted ? sw.data_ptr<float>() : nullptr,
+                         weighted ? w_boot.data_ptr<float>() : nullptr);
       // sort the resample by row id: histogram/sum kernels are order-
       // invariant, and sorted gathers coalesce the feature-major bin
       // reads (a random-order resample made every 1-byte load its own
       // 64 B line)
       auto srt = torch::sort(idx_t);
       idx_u = std::get<0>(srt).contiguous();
-      y_u = y_boot.index_select(0, std::get<1>(srt)).contiguous();
+      auto perm = std::get<1>(srt);
+      y_u = y_boot.index_select(0, perm).contiguous();
+      if (weighted) w_u = w_boot.index_select(0, perm).contiguous();
     }
+    const float* wp_u = weighted ? w_u.data_ptr<float>() : nullptr;
     if (KP > 1)
       hipLaunchKernelGGL(rf_planes_kernel, dim3(blocks), dim3(256), 0, s,
                          y_u.data_ptr<float>(), G.data_ptr<float>(), nt,
-                         KP);
+                         KP, wp_u);
+    if (weighted && KP == 1) y_w.copy_(y_u).mul_(w_u);
     HIP_CHECK(hipMemsetAsync(node_id.data_ptr(), 0, nt * 4, s));
     int* feat_t = feats.data_ptr<int>() + (long)t * KP * NN;
     int* thr_t = thrs.data_ptr<int>() + (long)t * KP * NN;
@@ -6853,12 +6919,16 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
         // ONE launch histograms every class plane (rows stay cache-hot
         // across classes inside each block); classification g-planes are
         // 0/1 indicators -> packed single-atomic cells
-        const float* gp = (KP > 1) ? G.data_ptr<float>()
-                                   : y_u.data_ptr<float>();
+        const float* gp = (KP > 1)
+                              ? G.data_ptr<float>()
+                              : (weighted ? y_w.data_ptr<float>()
+                                          : y_u.data_ptr<float>());
         launch_hist_level(Xtp, idx_u.data_ptr<int>(),
                           node_id.data_ptr<int>(), gp,
-                          ones.data_ptr<float>(), hist.data_ptr<float>(),
-                          nt, n_rows, F, d, 1.f, false, K != 0, s, KP,
+                          weighted ? wp_u : ones.data_ptr<float>(),
+                          hist.data_ptr<float>(),
+                          nt, n_rows, F, d, weighted ? wmax : 1.f, false,
+                          (K != 0) && !weighted, s, KP,
                           nt, ccells);
       }
       if (d > 0) {
@@ -6900,8 +6970,10 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
       HIP_CHECK(hipGetLastError());
     }
     for (int c = 0; c < KP; ++c) {
-      const float* gp = (KP > 1) ? G.data_ptr<float>() + (long)c * nt
-                                 : y_u.data_ptr<float>();
+      const float* gp = (KP > 1)
+                            ? G.data_ptr<float>() + (long)c * nt
+                            : (weighted ? y_w.data_ptr<float>()
+                                        : y_u.data_ptr<float>());
       HIP_CHECK(hipMemsetAsync(gsum.data_ptr(), 0, NN * 4, s));
       HIP_CHECK(hipMemsetAsync(hsum.data_ptr(), 0, NN * 4, s));
       hipLaunchKernelGGL(gbt_node_sums_kernel,
@@ -6909,7 +6981,8 @@ std::vector<torch::Tensor> rf_fit(torch::Tensor Xb, torch::Tensor Xbt,
                               (unsigned)(((NN - bl_lo) + 127) / 128)),
                          dim3(256), 0,
                          s, node_id.data_ptr<int>(), gp,
-                         ones.data_ptr<float>(), gsum.data_ptr<float>(),
+                         weighted ? wp_u : ones.data_ptr<float>(),
+                         gsum.data_ptr<float>(),
                          hsum.data_ptr<float>(), nt, bl_lo, NN, 0.f, 0.f,
                          1.f);
       hipLaunchKernelGGL(gbt_leaf_vals_kernel,

@@ -55,14 +55,19 @@ class _LazyForests:
 
 
 class DeviceRFExecutor(DeviceGBTExecutor):
-    supports_sample_weight = False    # rf_fit has no weighted gini path
+    supports_sample_weight = True
 
-    def fit(self, train_idx, plan):
+    def fit(self, train_idx, plan, sample_weight=None):
         torch = self.torch
         idx_np = np.asarray(train_idx, dtype=np.int32)
         idx_t = torch.as_tensor(idx_np).to(self.device)
         y_tr_np = self.y_np[idx_np]
         y_tr = torch.as_tensor(y_tr_np).to(self.device)
+        sw_t = torch.empty(0, dtype=torch.float32, device=self.device)
+        if sample_weight is not None:
+            sw_t = torch.as_tensor(
+                np.asarray(sample_weight, np.float32)[idx_np]
+            ).to(self.device)
 
         models_dev = []
         n_iter = np.empty(plan.B, np.int32)
@@ -77,7 +82,7 @@ class DeviceRFExecutor(DeviceGBTExecutor):
                 spec.max_depth,
                 spec.min_samples_leaf, spec.max_features,
                 spec.random_state, 1 if spec.bootstrap else 0,
-                1 if spec.splitter == "random" else 0)
+                1 if spec.splitter == "random" else 0, sw_t)
             kind = "multi" if K > 2 else "flat"
             models_dev.append((kind, feats, thrs, vals, spec))
             n_iter[i] = spec.n_estimators

@@ -27,18 +27,65 @@ def _ext():
     return load_extension()
 
 
-def _dev_trees(ext, dev, Xb_np, y_np, spec, K=0):
+def _dev_trees(ext, dev, Xb_np, y_np, spec, K=0, sample_weight=None):
     Xb = torch.as_tensor(Xb_np).to(dev)
     Xbt = Xb.t().contiguous()
     idx = torch.arange(Xb_np.shape[0], dtype=torch.int32, device=dev)
     y = torch.as_tensor(y_np.astype(np.float32)).to(dev)
+    sw = (torch.as_tensor(np.asarray(sample_weight, np.float32)).to(dev)
+          if sample_weight is not None else
+          torch.empty(0, dtype=torch.float32, device=dev))
     feats, thrs, vals = ext.rf_fit(
         Xb, Xbt, idx, y, K, spec.n_estimators, spec.max_depth,
         spec.min_samples_leaf, spec.max_features, spec.random_state,
         1 if spec.bootstrap else 0,
-        1 if spec.splitter == "random" else 0)
+        1 if spec.splitter == "random" else 0, sw)
     return (feats.cpu().numpy(), thrs.cpu().numpy().astype(np.uint8),
             vals.cpu().numpy())
+
+
+def test_rf_device_weighted_matches_ref(dev):
+    """Weighted gini on device (w-scaled planes + sum(w) h channel,
+    classic-layout per-tree path) vs the weighted numpy reference;
+    binary and multiclass."""
+    rng = np.random.default_rng(3)
+    n = 4000
+    Xb = rng.integers(0, 64, size=(n, 16), dtype=np.uint8)
+    y = (Xb[:, 0].astype(np.float32) + 0.5 * Xb[:, 3] > 40).astype(
+        np.float32)
+    w = rng.uniform(0.5, 2.0, size=n)
+    spec = RFSpec(n_estimators=6, max_depth=3, min_samples_leaf=2,
+                  max_features=8, random_state=11)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, y, spec, sample_weight=w)
+    ref = rf_fit_ref(Xb, y, spec, sample_weight=w)
+    agree = 0
+    for t, (f_r, t_r, v_r) in enumerate(ref):
+        if np.array_equal(f_d[t], f_r) and np.array_equal(t_d[t], t_r):
+            agree += 1
+            np.testing.assert_allclose(v_d[t], v_r, rtol=2e-3, atol=1e-5)
+    # the device h channel is fixed-point-quantized (hscale), so rare
+    # near-tied splits may flip vs the fp64 reference
+    assert agree >= 4, f"only {agree}/6 weighted trees match"
+
+    ym = (Xb[:, 1] // 22).astype(np.float32)       # 3 classes
+    K = 3
+    spec_m = RFSpec(n_estimators=4, max_depth=3, min_samples_leaf=2,
+                    max_features=8, random_state=5, n_classes=K)
+    f_d, t_d, v_d = _dev_trees(_ext(), dev, Xb, ym, spec_m, K=K,
+                               sample_weight=w)
+    ref = rf_fit_ref(Xb, ym, spec_m, sample_weight=w)
+    agree = 0
+    for t, planes in enumerate(ref):
+        f_r = planes[0][0]
+        if np.array_equal(f_d[t * K], f_r):
+            agree += 1
+    assert agree >= 3, f"only {agree}/4 weighted multiclass trees match"
+
+    # unit weights reproduce the unweighted forest structures
+    f_1, t_1, _ = _dev_trees(_ext(), dev, Xb, y, spec,
+                             sample_weight=np.ones(n))
+    f_0, t_0, _ = _dev_trees(_ext(), dev, Xb, y, spec)
+    assert np.array_equal(f_1, f_0) and np.array_equal(t_1, t_0)
 
 
 def test_rf_device_matches_ref_binary(dev):
@@ -178,8 +225,9 @@ def test_extratrees_device_matches_ref(dev):
     Xb_t = torch.as_tensor(Xb).to(dev)
     idx = torch.arange(n, dtype=torch.int32, device=dev)
     yt = torch.as_tensor(y).to(dev)
-    feats, thrs, vals = ext.rf_fit(Xb_t, Xb_t.t().contiguous(), idx, yt, 0,
-                                   5, 3, 2, 6, 4, 0, 1)
+    feats, thrs, vals = ext.rf_fit(
+        Xb_t, Xb_t.t().contiguous(), idx, yt, 0, 5, 3, 2, 6, 4, 0, 1,
+        torch.empty(0, dtype=torch.float32, device=dev))
     ref = rf_fit_ref(Xb, y, spec)
     f_d = feats.cpu().numpy()
     t_d = thrs.cpu().numpy().astype(np.uint8)
@@ -204,8 +252,9 @@ def test_extratrees_multiclass_device_matches_ref(dev):
     Xb_t = torch.as_tensor(Xb).to(dev)
     idx = torch.arange(n, dtype=torch.int32, device=dev)
     yt = torch.as_tensor(y).to(dev)
-    feats, thrs, vals = ext.rf_fit(Xb_t, Xb_t.t().contiguous(), idx, yt, K,
-                                   4, 3, 2, 5, 6, 0, 1)
+    feats, thrs, vals = ext.rf_fit(
+        Xb_t, Xb_t.t().contiguous(), idx, yt, K, 4, 3, 2, 5, 6, 0, 1,
+        torch.empty(0, dtype=torch.float32, device=dev))
     ref = rf_fit_ref(Xb, y, spec)
     f_d = feats.cpu().numpy()
     t_d = thrs.cpu().numpy().astype(np.uint8)

@@ -157,13 +157,59 @@ def test_gbt_grid_routes_sample_weight():
                            gs0.cv_results_["mean_test_score"])
 
 
-def test_rf_sample_weight_warns_ignored():
+def test_rf_sample_weight_reference_semantics():
+    """RF honors sample_weight: weighted gini via w-scaled class planes
+    + a sum(w) mass channel.  w=1 == unweighted exactly; w=2 ==
+    duplicated rows (bootstrap=False, min_samples_leaf=1 so the
+    count-based leaf gate cannot diverge)."""
+    from spark_sklearn_amd.models._gbt_ref import bin_data, quantile_bins
+    from spark_sklearn_amd.models._rf_ref import RFSpec, rf_fit_ref
+
+    X, y = _data(n=600, f=8, seed=7)
+    Xb = bin_data(X, quantile_bins(X))
+    yf = y.astype(np.float32)
+    spec = RFSpec(n_estimators=4, max_depth=3, min_samples_leaf=1,
+                  max_features=4, bootstrap=False, random_state=0,
+                  n_classes=0)
+    t0 = rf_fit_ref(Xb, yf, spec)
+    t1 = rf_fit_ref(Xb, yf, spec, sample_weight=np.ones(600))
+    for (fa, ta, va), (fb, tb, vb) in zip(t0, t1):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_array_equal(ta, tb)
+        np.testing.assert_allclose(va, vb)
+
+    dup = np.arange(0, 600, 3)
+    Xb2 = np.vstack([Xb, Xb[dup]])
+    y2 = np.concatenate([yf, yf[dup]])
+    w = np.ones(600)
+    w[dup] = 2.0
+    tw = rf_fit_ref(Xb, yf, spec, sample_weight=w)
+    td = rf_fit_ref(Xb2, y2, spec)
+    for (fa, ta, va), (fb, tb, vb) in zip(tw, td):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_array_equal(ta, tb)
+        np.testing.assert_allclose(va, vb, rtol=1e-5, atol=1e-7)
+
+
+def test_rf_grid_routes_sample_weight():
     from spark_sklearn_amd.models import RandomForestClassifier
 
-    X, y = _data(n=600, f=8)
-    with pytest.warns(UserWarning, match="IGNORED"):
-        RandomForestClassifier(n_estimators=3, max_depth=3).fit(
-            X, y, sample_weight=np.ones(600))
+    X, y = _data(n=700, f=8, seed=8)
+    w = np.where(y == 1, 6.0, 1.0)
+    gs = ska.GridSearchCV(
+        None, RandomForestClassifier(n_estimators=4, max_depth=3,
+                                     random_state=0),
+        {"max_features": [2, 4]}, cv=2)
+    with warnings.catch_warnings():
+        warnings.filterwarnings("error", message=".*IGNORED.*")
+        gs.fit(X, y, sample_weight=w)
+    gs0 = ska.GridSearchCV(
+        None, RandomForestClassifier(n_estimators=4, max_depth=3,
+                                     random_state=0),
+        {"max_features": [2, 4]}, cv=2)
+    gs0.fit(X, y)
+    assert not np.allclose(gs.cv_results_["mean_test_score"],
+                           gs0.cv_results_["mean_test_score"])
 
 
 def test_kmeans_sample_weight_weighted_centroids():
