@@ -110,6 +110,80 @@ def test_gloo_world2_search_matches_single_process(tmp_path):
     assert gs.best_params_ == r0["best_params"]
 
 
+SW_WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["SKA_REPO"])
+import numpy as np
+import torch.distributed as dist
+import spark_sklearn_amd as ska
+
+rank = int(os.environ["RANK"])
+dist.init_process_group("gloo", init_method=os.environ["SKA_INIT"],
+                        rank=rank, world_size=int(os.environ["WORLD_SIZE"]))
+
+rng = np.random.default_rng(2)
+n, f = 500, 12
+X = rng.standard_normal((n, f))
+y = (X[:, 0] + 0.2 * rng.standard_normal(n) > 0).astype(np.int64)
+w = np.where(y == 1, 5.0, 1.0)     # passed on EVERY rank (see docs/api.md)
+
+from spark_sklearn_amd.models import LogisticRegression
+gs = ska.GridSearchCV("dist", LogisticRegression(max_iter=25),
+                      {"C": [0.1, 1.0]}, cv=3)
+gs.fit(X, y, sample_weight=w)
+out = {
+    "rank": rank,
+    "mean_test_score": gs.cv_results_["mean_test_score"].tolist(),
+    "best_params": gs.best_params_,
+}
+print("RESULT" + json.dumps(out))
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(300)
+def test_gloo_world2_sample_weight(tmp_path):
+    """Weighted search through the distributed tier: both ranks agree,
+    and the result matches a weighted single-process run exactly."""
+    init_file = tmp_path / "init"
+    script = tmp_path / "worker_sw.py"
+    script.write_text(SW_WORKER)
+    env = dict(os.environ, SKA_REPO=str(REPO),
+               SKA_INIT=f"file://{init_file}", WORLD_SIZE="2",
+               MASTER_ADDR="127.0.0.1")
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank))
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=e,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=240)
+        assert p.returncode == 0, f"worker failed:\n{stdout}\n{stderr}"
+        line = [ln for ln in stdout.splitlines()
+                if ln.startswith("RESULT")][0]
+        outs.append(json.loads(line[len("RESULT"):]))
+    r0, r1 = sorted(outs, key=lambda o: o["rank"])
+    assert r0["mean_test_score"] == r1["mean_test_score"]
+    assert r0["best_params"] == r1["best_params"]
+
+    import spark_sklearn_amd as ska
+    from spark_sklearn_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(2)
+    n, f = 500, 12
+    X = rng.standard_normal((n, f))
+    y = (X[:, 0] + 0.2 * rng.standard_normal(n) > 0).astype(np.int64)
+    w = np.where(y == 1, 5.0, 1.0)
+    gs = ska.GridSearchCV("sequential", LogisticRegression(max_iter=25),
+                          {"C": [0.1, 1.0]}, cv=3).fit(X, y,
+                                                       sample_weight=w)
+    np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
+                               r0["mean_test_score"], rtol=1e-10)
+
+
 MULTI_WORKER = r"""
 import json, os, sys
 sys.path.insert(0, os.environ["SKA_REPO"])
