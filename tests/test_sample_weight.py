@@ -244,6 +244,37 @@ def test_kmeans_sample_weight_weighted_centroids():
         gs.fit(base, sample_weight=w)
 
 
+def test_sample_weight_scale_invariance():
+    """The engine normalizes w to mean 1, so a global rescale of the
+    weights must not change any family's fitted model."""
+    from spark_sklearn_amd.models import (
+        GradientBoostingClassifier, RandomForestClassifier)
+
+    X, y = _data(n=700, f=8, seed=9)
+    w = np.where(y == 1, 3.0, 1.0)
+
+    m1 = LogisticRegression(max_iter=15).fit(X, y, sample_weight=w)
+    m2 = LogisticRegression(max_iter=15).fit(X, y, sample_weight=7.5 * w)
+    np.testing.assert_allclose(m1.coef_, m2.coef_, rtol=1e-5)
+
+    g1 = GradientBoostingClassifier(n_estimators=4, max_depth=2).fit(
+        X, y, sample_weight=w)
+    g2 = GradientBoostingClassifier(n_estimators=4, max_depth=2).fit(
+        X, y, sample_weight=7.5 * w)
+    for (fa, ta, va), (fb, tb, vb) in zip(g1._trees, g2._trees):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_allclose(va, vb, rtol=1e-4, atol=1e-6)
+
+    r1 = RandomForestClassifier(n_estimators=3, max_depth=3,
+                                random_state=0).fit(X, y, sample_weight=w)
+    r2 = RandomForestClassifier(n_estimators=3, max_depth=3,
+                                random_state=0).fit(X, y,
+                                                    sample_weight=7.5 * w)
+    for (fa, ta, va), (fb, tb, vb) in zip(r1._trees, r2._trees):
+        np.testing.assert_array_equal(fa, fb)
+        np.testing.assert_allclose(va, vb, rtol=1e-4, atol=1e-6)
+
+
 def test_sample_weight_validation():
     X, y = _data(n=500, f=8)
     with pytest.raises(ValueError, match="entries"):
