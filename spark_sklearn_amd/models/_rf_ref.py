@@ -146,12 +146,16 @@ def _build_rf_tree(Xb, gplanes, spec, fseed, w=None):
                     hh = np.bincount(bins, weights=w[rows], minlength=256)
                     hl = np.cumsum(hh)[:-1]
                     hr = wmass - hl
-                gain = np.zeros(255)
+                # factored form (matches the device kernel): the gini
+                # denominators are class-independent, so divide once
+                A = np.zeros(255)
+                Bq = np.zeros(255)
                 for c in range(KP):
                     gl = gls[c]
                     gr = G[c] - gl
-                    gain += gl * gl / (hl + eps) + gr * gr / (hr + eps)
-                gain -= parent
+                    A += gl * gl
+                    Bq += gr * gr
+                gain = A / (hl + eps) + Bq / (hr + eps) - parent
                 okay = (cl >= spec.min_samples_leaf) & \
                     (cr >= spec.min_samples_leaf)
                 if spec.splitter == "random":

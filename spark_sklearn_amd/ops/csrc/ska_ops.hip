@@ -6353,8 +6353,14 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
     // sums with fixed-point h sums -- pure nodes then get tiny nonzero
     // gains and split spuriously.  Unweighted fits have integer-exact
     // class counts, so this equals the count channel bit-for-bit.
+    // ONE pass accumulates mass hl, sum-of-squares A (left) / B
+    // (right) and SG2: the gini denominators are class-INDEPENDENT,
+    // so the division factors out of the class loop (2 divisions per
+    // threshold instead of 2K, and the g channels are read once).
     float hl[4] = {0.f, 0.f, 0.f, 0.f};
-    float H = 0.f;
+    float A[4] = {0.f, 0.f, 0.f, 0.f};
+    float B[4] = {0.f, 0.f, 0.f, 0.f};
+    float H = 0.f, SG2 = 0.f;
     for (int c = 0; c < K; ++c) {
       const float* fhc = FMT ? (fh0 + c) : (fh0 + (long)c * cstride);
       float g4[4];
@@ -6369,12 +6375,18 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
         float t = __shfl_up(pg, off);
         if (lane >= off) pg += t;
       }
-      H += __shfl(pg, 63);
+      float G = __shfl(pg, 63);
+      H += G;
+      SG2 += G * G;
+      if (f == 0 && lane == 0) sPC[c] = G;
       float a = pg - lg;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         a += g4[j];
-        hl[j] += a;
+        float gl = a, gr = G - gl;
+        hl[j] += gl;
+        A[j] += gl * gl;
+        B[j] += gr * gr;
       }
     }
     if (f == 0 && lane == 0) sH = H;
@@ -6399,34 +6411,11 @@ void rf_split_find_multi_body(const float* __restrict__ ndh, long cstride,
         tstar = lmin + (int)(rf_thr_hash(fseed, nd, f) %
                              (unsigned)(lmax - lmin));
     }
-    float gainsum[4] = {0.f, 0.f, 0.f, 0.f};
-    float parent = 0.f;
-    for (int c = 0; c < K; ++c) {
-      const float* fhc = FMT ? (fh0 + c) : (fh0 + (long)c * cstride);
-      float g4[4];
-      float lg = 0.f;
+    float gainsum[4];
+    const float parent = SG2 / (H + eps);
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        g4[j] = fhc[(4 * lane + j) * cw];
-        lg += g4[j];
-      }
-      float pg = lg;
-      for (int off = 1; off < 64; off <<= 1) {
-        float t = __shfl_up(pg, off);
-        if (lane >= off) pg += t;
-      }
-      float G = __shfl(pg, 63);
-      if (f == 0 && lane == 0) sPC[c] = G;
-      float a = pg - lg;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        a += g4[j];
-        float gl = a, gr = G - gl;
-        gainsum[j] +=
-            gl * gl / (hl[j] + eps) + gr * gr / ((H - hl[j]) + eps);
-      }
-      parent += G * G / (H + eps);
-    }
+    for (int j = 0; j < 4; ++j)
+      gainsum[j] = A[j] / (hl[j] + eps) + B[j] / ((H - hl[j]) + eps);
     if (active) {
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
