@@ -244,6 +244,30 @@ def test_kmeans_sample_weight_weighted_centroids():
         gs.fit(base, sample_weight=w)
 
 
+def test_halving_sweep_routes_sample_weight():
+    """Successive-halving (RandomizedSearchCV + halving_factor) carries
+    fit_params into every rung."""
+    from scipy.stats import loguniform
+
+    X, y = _data(n=900, f=12, seed=4)
+    w = np.where(y == 1, 5.0, 1.0)
+    rs = ska.RandomizedSearchCV(
+        None, LogisticRegression(max_iter=16),
+        {"C": loguniform(1e-3, 1e3)}, n_iter=8, cv=2, random_state=0,
+        halving_factor=2)
+    with warnings.catch_warnings():
+        warnings.filterwarnings("error", message=".*IGNORED.*")
+        rs.fit(X, y, sample_weight=w)
+    rs0 = ska.RandomizedSearchCV(
+        None, LogisticRegression(max_iter=16),
+        {"C": loguniform(1e-3, 1e3)}, n_iter=8, cv=2, random_state=0,
+        halving_factor=2)
+    rs0.fit(X, y)
+    assert not np.allclose(
+        np.nan_to_num(rs.cv_results_["mean_test_score"], nan=-1),
+        np.nan_to_num(rs0.cv_results_["mean_test_score"], nan=-1))
+
+
 def test_sample_weight_scale_invariance():
     """The engine normalizes w to mean 1, so a global rescale of the
     weights must not change any family's fitted model."""
