@@ -12,6 +12,45 @@ __all__ = ["DeviceGBTExecutor"]
 _bin_cache: dict = {}
 
 
+class _LazyModels:
+    """List-like view over the device GBT models: the numpy
+    (trees, f0) tuples that install_fitted / host scoring need
+    materialize per slot ON DEMAND (the search scores on device, so
+    eager D2H pulls + per-tree python tuples were pure overhead --
+    same pattern as the RF executor's _LazyForests)."""
+
+    def __init__(self, models_dev, K):
+        self._dev = models_dev
+        self._K = K
+        self._cache = {}
+
+    def __len__(self):
+        return len(self._dev)
+
+    def __getitem__(self, i):
+        if i in self._cache:
+            return self._cache[i]
+        m = self._dev[i]
+        if m is None:
+            out = ([], 0.0)
+        else:
+            feats, thrs, vals, f0, spec_or_loss = m
+            K = self._K
+            T = feats.shape[0] // max(K, 1) if K > 2 else feats.shape[0]
+            f_np = feats.cpu().numpy()
+            t_np = thrs.cpu().numpy().astype(np.uint8)
+            v_np = vals.cpu().numpy()
+            if K > 2:
+                trees = [[(f_np[t * K + c], t_np[t * K + c],
+                           v_np[t * K + c]) for c in range(K)]
+                         for t in range(T)]
+            else:
+                trees = [(f_np[t], t_np[t], v_np[t]) for t in range(T)]
+            out = (trees, f0)
+        self._cache[i] = out
+        return out
+
+
 class DeviceGBTExecutor:
     """Sequential per-candidate boosting over a shared resident binned
     matrix.  Quantile binning runs once per (matrix, device) on the host;
@@ -98,12 +137,11 @@ class DeviceGBTExecutor:
             w_np = np.asarray(sample_weight, np.float32)[idx_np]
             sw_t = torch.as_tensor(w_np).to(self.device)
 
-        models_host, models_dev = [], []
+        models_dev = []
         n_iter = np.empty(plan.B, np.int32)
         K = self.K
         for i, spec in enumerate(plan.specs):
             if spec.n_estimators == 0:
-                models_host.append(([], 0.0))
                 models_dev.append(None)
                 n_iter[i] = 0
                 continue
@@ -123,13 +161,6 @@ class DeviceGBTExecutor:
                     spec.learning_rate, spec.max_depth,
                     spec.min_samples_leaf, spec.l2_regularization)
                 models_dev.append((feats, thrs, vals, f0, spec))
-                f_np = feats.cpu().numpy()
-                t_np = thrs.cpu().numpy().astype(np.uint8)
-                v_np = vals.cpu().numpy()
-                trees = [[(f_np[t * K + c], t_np[t * K + c],
-                           v_np[t * K + c]) for c in range(K)]
-                         for t in range(spec.n_estimators)]
-                models_host.append((trees, f0))
                 n_iter[i] = spec.n_estimators
                 continue
             f0 = self._f0(spec.loss, y_tr_np, w_np)
@@ -139,16 +170,10 @@ class DeviceGBTExecutor:
                 spec.learning_rate, spec.max_depth, spec.min_samples_leaf,
                 spec.l2_regularization, f0, sw_t)
             models_dev.append((feats, thrs, vals, f0, spec.loss))
-            f_np = feats.cpu().numpy()
-            t_np = thrs.cpu().numpy().astype(np.uint8)
-            v_np = vals.cpu().numpy()
-            trees = [(f_np[t], t_np[t], v_np[t])
-                     for t in range(spec.n_estimators)]
-            models_host.append((trees, f0))
             n_iter[i] = spec.n_estimators
         torch.cuda.synchronize(self.device)
         self._models_dev = models_dev
-        return (models_host, self.edges), n_iter, None
+        return (_LazyModels(models_dev, K), self.edges), n_iter, None
 
     def scores(self, idx, fitted, plan, metric):
         torch = self.torch
