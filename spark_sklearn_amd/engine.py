@@ -277,16 +277,6 @@ def _encode_cached(family, estimator, y):
     return out
 
 
-def warn_sample_weight_ignored(sample_weight, name):
-    """Loud divergence marker for executors without weighting (today:
-    the CSR-capacity linear tier): the argument is accepted for sklearn
-    signature parity but the fit runs unweighted."""
-    if sample_weight is not None:
-        warnings.warn(
-            f"sample_weight is IGNORED by {name}: this executor tier "
-            "fits unweighted", stacklevel=3)
-
-
 def _route_fit_params(fit_params, executor, X):
     """Split fit_params into (normalized sample_weight, msq_scale,
     ignored dict).  sample_weight is honored when the executor supports
