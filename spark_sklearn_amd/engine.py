@@ -306,6 +306,79 @@ def _route_fit_params(fit_params, executor, X):
     return w, float(w.max()), ignored
 
 
+def _class_weight_rows(cw, classes, y_enc, train_idx):
+    """Per-row multiplier from class_weight.  'balanced' uses the TRAIN
+    FOLD's class frequencies (n_fold / (K * count_c) -- sklearn computes
+    it on the data each fit sees); a dict maps ORIGINAL class labels,
+    missing classes default to 1.0."""
+    K = len(classes)
+    yi = np.asarray(y_enc).astype(np.int64)
+    if isinstance(cw, str):
+        if cw != "balanced":
+            raise ValueError(
+                f"class_weight must be None, 'balanced' or a dict, "
+                f"got {cw!r}")
+        cnt = np.bincount(yi[train_idx], minlength=K)
+        nt = len(train_idx)
+        wmap = np.where(cnt > 0, nt / (K * np.maximum(cnt, 1)), 1.0)
+    elif isinstance(cw, dict):
+        wmap = np.ones(K)
+        for k, v in cw.items():
+            hits = np.flatnonzero(classes == k)
+            if len(hits) == 0:
+                raise ValueError(
+                    f"class_weight key {k!r} is not a class label")
+            wmap[hits[0]] = float(v)
+    else:
+        raise ValueError(
+            f"class_weight must be None, 'balanced' or a dict, got "
+            f"{type(cw).__name__}")
+    return wmap[yi]
+
+
+def _uniform_class_weight(estimator, tasks, classes, executor):
+    """Resolve the search's class_weight (constructor value, optionally
+    overridden UNIFORMLY by the candidates).  Per-candidate class_weight
+    grids would need per-candidate row weights inside one kernel batch;
+    they raise with a pointer at the sklearn-estimator fallback."""
+    base = getattr(estimator, "class_weight", None)
+    seen = {repr(sorted(p.items())) if isinstance(p, dict) else repr(p)
+            for p in (t.params.get("class_weight", base) for t in tasks)}
+    if len(seen) > 1:
+        raise ValueError(
+            "per-candidate class_weight grids are not supported on the "
+            "batched kernel path (one weight vector per kernel batch); "
+            "use sklearn's own estimator for that grid, or fix "
+            "class_weight in the constructor")
+    cw = (tasks[0].params.get("class_weight", base) if tasks else base)
+    if cw is None:
+        return None
+    if classes is None:
+        raise ValueError("class_weight is only valid for classifiers")
+    if not getattr(executor, "supports_sample_weight", False):
+        warnings.warn(
+            "class_weight is IGNORED by this executor tier (fits run "
+            "unweighted)", stacklevel=3)
+        return None
+    return cw
+
+
+def _fold_weights(sample_weight, msq_scale, cw, classes, y_enc, train):
+    """Effective per-row weights for ONE fold: user sample_weight (mean-1
+    normalized) times the class_weight row multiplier, renormalized to
+    mean 1 over the fold's train rows."""
+    if cw is None:
+        return sample_weight, msq_scale
+    wrow = _class_weight_rows(cw, classes, y_enc, train)
+    eff = wrow if sample_weight is None else sample_weight * wrow
+    tr = eff[train]
+    tot = float(tr.sum())
+    if tot <= 0:
+        raise ValueError("class_weight zeroes out every training row")
+    eff = (eff * (len(train) / tot)).astype(np.float32)
+    return eff, float(eff[train].max())
+
+
 def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
                       fit_params, return_train_score, error_score, verbose,
                       device=None,
@@ -324,6 +397,8 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     executor = family.make_executor(X, y_enc, classes, device)
     sample_weight, msq_scale, ignored_fp = _route_fit_params(
         fit_params, executor, X)
+    class_weight = _uniform_class_weight(estimator, tasks, classes,
+                                         executor)
     if ignored_fp:
         warnings.warn(
             "fit_params "
@@ -354,7 +429,8 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
 
     if family is LinearFamily and device is not None and \
             isinstance(executor, _DLE) and \
-            metric is not None and not fit_params:
+            metric is not None and not fit_params and \
+            class_weight is None:
         fast = _linear_device_search(
             estimator, executor, by_fold, n_classes, metric,
             return_train_score, error_score, verbose)
@@ -376,20 +452,23 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     for fold_index in sorted(by_fold):
         fold_tasks = by_fold[fold_index]
         train, test = fold_tasks[0].train, fold_tasks[0].test
+        fold_sw, fold_msq = _fold_weights(sample_weight, msq_scale,
+                                          class_weight, classes, y_enc,
+                                          train)
         for start in range(0, len(fold_tasks), cand_chunk):
             chunk = fold_tasks[start:start + cand_chunk]
             params_list = [t.params for t in chunk]
             t0 = time.perf_counter()
-            if msq_scale != 1.0 and family is LinearFamily:
+            if fold_msq != 1.0 and family is LinearFamily:
                 plan, bad = family.build_plan(
                     estimator, params_list, len(train), executor,
-                    n_classes, error_score, msq_scale=msq_scale)
+                    n_classes, error_score, msq_scale=fold_msq)
             else:
                 plan, bad = family.build_plan(
                     estimator, params_list, len(train), executor,
                     n_classes, error_score)
-            fit_kw = ({"sample_weight": sample_weight}
-                      if sample_weight is not None else {})
+            fit_kw = ({"sample_weight": fold_sw}
+                      if fold_sw is not None else {})
             try:
                 fitted, n_iter, _ = executor.fit(train, plan, **fit_kw)
             except RuntimeError as exc:
@@ -655,8 +734,12 @@ def fit_single(estimator, X, y, device=None, sample_weight=None):
     sw, msq_scale, ignored = _route_fit_params(fp, executor, X)
     if ignored:
         warnings.warn(
-            "sample_weight is IGNORED by this estimator family's batched "
-            "path (honored for the linear family only)", stacklevel=2)
+            "sample_weight is IGNORED by this executor tier (fits run "
+            "unweighted)", stacklevel=2)
+    all_idx = np.arange(n, dtype=np.int64)
+    cw = _uniform_class_weight(estimator, [], classes, executor)
+    sw, msq_scale = _fold_weights(sw, msq_scale, cw, classes, y_enc,
+                                  all_idx)
     if msq_scale != 1.0 and family is LinearFamily:
         plan, _ = family.build_plan(estimator, [{}], n, executor,
                                     n_classes, "raise",
@@ -664,7 +747,6 @@ def fit_single(estimator, X, y, device=None, sample_weight=None):
     else:
         plan, _ = family.build_plan(estimator, [{}], n, executor,
                                     n_classes, "raise")
-    all_idx = np.arange(n, dtype=np.int64)
     fit_kw = {"sample_weight": sw} if sw is not None else {}
     fitted, n_iter, _ = executor.fit(all_idx, plan, **fit_kw)
     family.install_fitted(estimator, fitted, 0, plan, classes)

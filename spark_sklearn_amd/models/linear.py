@@ -128,7 +128,7 @@ class LogisticRegression(_LinearClassifier):
 
     def __init__(self, C=1.0, penalty="l2", l1_ratio=0.5, tol=1e-4,
                  max_iter=100, fit_intercept=True, learning_rate="auto",
-                 momentum=0.9):
+                 momentum=0.9, class_weight=None):
         self.C = C
         self.penalty = penalty
         self.l1_ratio = l1_ratio
@@ -137,6 +137,7 @@ class LogisticRegression(_LinearClassifier):
         self.fit_intercept = fit_intercept
         self.learning_rate = learning_rate
         self.momentum = momentum
+        self.class_weight = class_weight
 
     def _column_spec(self, n, mean_sq_row_norm):
         # sklearn objective: sum(logloss) + ||w||^2/(2C); per-sample scaled
@@ -162,7 +163,8 @@ class LinearSVC(_LinearClassifier):
     _loss_id = LOSS_HINGE
 
     def __init__(self, C=1.0, penalty="l2", tol=1e-4, max_iter=200,
-                 fit_intercept=True, learning_rate="auto", momentum=0.9):
+                 fit_intercept=True, learning_rate="auto", momentum=0.9,
+                 class_weight=None):
         self.C = C
         self.penalty = penalty
         self.tol = tol
@@ -170,6 +172,7 @@ class LinearSVC(_LinearClassifier):
         self.fit_intercept = fit_intercept
         self.learning_rate = learning_rate
         self.momentum = momentum
+        self.class_weight = class_weight
 
     def _column_spec(self, n, mean_sq_row_norm):
         base = 1.0 / (max(float(self.C), 1e-12) * n)
@@ -183,7 +186,7 @@ class SGDClassifier(_LinearClassifier):
 
     def __init__(self, loss="hinge", penalty="l2", alpha=1e-4, l1_ratio=0.15,
                  tol=1e-3, max_iter=100, fit_intercept=True,
-                 learning_rate="auto", momentum=0.9):
+                 learning_rate="auto", momentum=0.9, class_weight=None):
         self.loss = loss
         self.penalty = penalty
         self.alpha = alpha
@@ -193,6 +196,7 @@ class SGDClassifier(_LinearClassifier):
         self.fit_intercept = fit_intercept
         self.learning_rate = learning_rate
         self.momentum = momentum
+        self.class_weight = class_weight
 
     @property
     def _loss_id(self):

@@ -121,6 +121,15 @@ class _RFBase(BaseEstimator):
 class RandomForestClassifier(ClassifierMixin, _RFBase):
     """Bagged histogram-tree classifier (gini splits; binary + K-class)."""
 
+    def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
+                 max_features="sqrt", bootstrap=True, random_state=None,
+                 class_weight=None):
+        super().__init__(n_estimators=n_estimators, max_depth=max_depth,
+                         min_samples_leaf=min_samples_leaf,
+                         max_features=max_features, bootstrap=bootstrap,
+                         random_state=random_state)
+        self.class_weight = class_weight
+
     def predict_proba(self, X):
         raw = self._forest_raw(X)
         if raw.ndim == 2:                      # K > 2: per-class probs
@@ -144,11 +153,13 @@ class ExtraTreesClassifier(RandomForestClassifier):
     _splitter = "random"
 
     def __init__(self, n_estimators=100, max_depth=10, min_samples_leaf=1,
-                 max_features="sqrt", bootstrap=False, random_state=None):
+                 max_features="sqrt", bootstrap=False, random_state=None,
+                 class_weight=None):
         super().__init__(n_estimators=n_estimators, max_depth=max_depth,
                          min_samples_leaf=min_samples_leaf,
                          max_features=max_features, bootstrap=bootstrap,
                          random_state=random_state)
+        self.class_weight = class_weight
 
 
 class RandomForestRegressor(RegressorMixin, _RFBase):
