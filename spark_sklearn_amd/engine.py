@@ -405,8 +405,8 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
             f"({', '.join(sorted(ignored_fp))}) are IGNORED by the batched "
             "kernel path: fits run unweighted/unmodified, so results "
             "differ from an estimator that honors them (sample_weight is "
-            "honored for the linear family; other fit_params are a "
-            "roadmap item -- see ROADMAP.md)",
+            "honored by every batched family on the dense tiers; other "
+            "fit_params are a roadmap item -- see ROADMAP.md)",
             stacklevel=2)
     metric = family.metric_for(scorer, estimator)
     # multimetric fast path: when EVERY metric of a _MultiScorer maps to a
