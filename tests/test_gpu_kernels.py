@@ -636,3 +636,19 @@ def test_multiclass_wide_tile_scoring_matches_host(dev):
     np.testing.assert_allclose(gs.cv_results_["mean_test_score"],
                                gs_host.cv_results_["mean_test_score"],
                                atol=7e-3)
+
+
+def test_class_weight_maps_to_sample_weight_on_device(dev, problem):
+    """class_weight (engine-composed onto the weighted kernels) equals
+    the manual sample_weight fit on the device tier."""
+    from spark_sklearn_amd.engine import fit_single
+    from spark_sklearn_amd.models import LogisticRegression as SkaLR
+
+    X, y = problem
+    cw = {0: 1.0, 1: 3.0}
+    m1 = SkaLR(max_iter=8, class_weight=cw)
+    fit_single(m1, X, y, device=dev)
+    w = np.where(y == 1, 3.0, 1.0).astype(np.float64)
+    m2 = SkaLR(max_iter=8)
+    fit_single(m2, X, y, device=dev, sample_weight=w)
+    np.testing.assert_allclose(m1.coef_, m2.coef_, rtol=1e-4, atol=1e-6)
