@@ -336,25 +336,29 @@ def _class_weight_rows(cw, classes, y_enc, train_idx):
     return wmap[yi]
 
 
+def _cw_key(v):
+    """Hashable identity for a class_weight value (dicts normalized)."""
+    return repr(sorted(v.items())) if isinstance(v, dict) else repr(v)
+
+
 def _uniform_class_weight(estimator, tasks, classes, executor):
-    """Resolve the search's class_weight (constructor value, optionally
-    overridden UNIFORMLY by the candidates).  Per-candidate class_weight
-    grids would need per-candidate row weights inside one kernel batch;
-    they raise with a pointer at the sklearn-estimator fallback."""
+    """Resolve one candidate batch's class_weight (constructor value,
+    optionally overridden by the candidates -- run_batched_tasks groups
+    tasks so each batch is uniform: one weight vector per kernel
+    batch)."""
     base = getattr(estimator, "class_weight", None)
-    seen = {repr(sorted(p.items())) if isinstance(p, dict) else repr(p)
-            for p in (t.params.get("class_weight", base) for t in tasks)}
+    seen = {_cw_key(t.params.get("class_weight", base)) for t in tasks}
     if len(seen) > 1:
+        # callers group tasks by class_weight before reaching here
         raise ValueError(
-            "per-candidate class_weight grids are not supported on the "
-            "batched kernel path (one weight vector per kernel batch); "
-            "use sklearn's own estimator for that grid, or fix "
-            "class_weight in the constructor")
+            "internal: mixed class_weight values in one candidate batch")
     cw = (tasks[0].params.get("class_weight", base) if tasks else base)
     if cw is None:
         return None
     if classes is None:
-        raise ValueError("class_weight is only valid for classifiers")
+        # regressor grids naming class_weight fall through to the
+        # build_plan unknown-parameter path (error_score contract)
+        return None
     if not getattr(executor, "supports_sample_weight", False):
         warnings.warn(
             "class_weight is IGNORED by this executor tier (fits run "
@@ -395,6 +399,33 @@ def run_batched_tasks(estimator, X, y, tasks: List[Task], *, scorer,
     n_classes = len(classes) if classes is not None else 0
     device = _device_or_none(device)
     executor = family.make_executor(X, y_enc, classes, device)
+    # per-candidate class_weight grids: one weight vector feeds a kernel
+    # batch, so tasks are GROUPED by class_weight value and each group
+    # runs as its own batch sequence (executors/binning are cached, so
+    # the split costs only smaller batches)
+    base_cw = getattr(estimator, "class_weight", None)
+    if classes is not None and tasks:
+        keys = [_cw_key(t.params.get("class_weight", base_cw))
+                for t in tasks]
+        if len(set(keys)) > 1:
+            groups: Dict[str, List[Task]] = {}
+            for k, t in zip(keys, tasks):
+                groups.setdefault(k, []).append(t)
+            out: List[TaskResult] = []
+            done_base = 0
+            for g in groups.values():
+                def _prog(done, total, _b=done_base):
+                    if progress is not None:
+                        progress(_b + done, len(tasks))
+                out.extend(run_batched_tasks(
+                    estimator, X, y, g, scorer=scorer,
+                    fit_params=fit_params,
+                    return_train_score=return_train_score,
+                    error_score=error_score, verbose=verbose,
+                    device=device, column_batch=column_batch,
+                    progress=_prog if progress is not None else None))
+                done_base += len(g)
+            return out
     sample_weight, msq_scale, ignored_fp = _route_fit_params(
         fit_params, executor, X)
     class_weight = _uniform_class_weight(estimator, tasks, classes,

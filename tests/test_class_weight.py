@@ -94,22 +94,36 @@ def _engine_tasks(X, y, params_list, cv=2):
     return tasks
 
 
-def test_per_candidate_class_weight_raises_on_batched_path():
-    """The BATCHED kernel path (one weight vector per batch) rejects
-    per-candidate class_weight grids; the per-task thread tier handles
-    them fine (each fit applies its own estimator params)."""
+def test_per_candidate_class_weight_grid_on_batched_path():
+    """Per-candidate class_weight grids run on the batched path via
+    task GROUPING (one weight vector per kernel batch): each group's
+    scores equal the same candidate run alone."""
     from spark_sklearn_amd.engine import run_batched_tasks
 
     X, y = _imbalanced(seed=5)
-    tasks = _engine_tasks(X, y, [{"class_weight": None},
-                                 {"class_weight": "balanced"}])
-    with pytest.raises(ValueError, match="per-candidate class_weight"):
-        run_batched_tasks(
-            LogisticRegression(max_iter=10), X, y, tasks, scorer=None,
-            fit_params=None, return_train_score=False,
-            error_score="raise", verbose=0, device=None)
+    grid = [{"class_weight": None}, {"class_weight": "balanced"},
+            {"class_weight": {0: 1.0, 1: 2.0}}]
+    tasks = _engine_tasks(X, y, grid)
+    res = run_batched_tasks(
+        LogisticRegression(max_iter=10), X, y, tasks, scorer=None,
+        fit_params=None, return_train_score=False,
+        error_score="raise", verbose=0, device=None)
+    by_idx = {r.index: r.test_score for r in res}
+    assert len(by_idx) == len(tasks)
 
-    # thread tier: the same grid works (per-task fits)
+    for params in grid:
+        solo = _engine_tasks(X, y, [params])
+        res_s = run_batched_tasks(
+            LogisticRegression(max_iter=10, **params), X, y, solo,
+            scorer=None, fit_params=None, return_train_score=False,
+            error_score="raise", verbose=0, device=None)
+        # match fold-for-fold against the grouped run
+        gi = grid.index(params)
+        for fi, r in enumerate(res_s):
+            grouped = by_idx[gi * 2 + fi]
+            np.testing.assert_allclose(grouped, r.test_score, rtol=1e-9)
+
+    # thread tier: the same grid also works (per-task fits)
     gs = ska.GridSearchCV(
         "threads", LogisticRegression(max_iter=10),
         {"class_weight": [None, "balanced"]}, cv=2)
