@@ -401,7 +401,7 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         resume_path = getattr(backend, "resume_from_log", None)
         restored: List[TaskResult] = []
         todo = tasks
-        log_header = self._score_log_header(X, y)
+        log_header = self._score_log_header(X, y, merged_fit_params)
         if resume_path:
             done = self._load_resume_records(resume_path, n_splits,
                                              log_header)
@@ -624,11 +624,13 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
         """Stable text key for a candidate's params (resume matching)."""
         return repr(sorted((k, repr(v)) for k, v in params.items()))
 
-    def _score_log_header(self, X, y):
+    def _score_log_header(self, X, y, fit_params=None):
         """Identity of this search for score-log resume validation: a
-        cheap strided fingerprint of (X, y) plus the splitter, estimator
-        class and scoring.  A log written against different data or CV
-        must not be silently replayed into cv_results_."""
+        cheap strided fingerprint of (X, y) AND of any fit-time
+        sample_weight, plus the splitter, the estimator class WITH its
+        constructor params (a log written with class_weight='balanced'
+        or momentum=0.5 must not replay into a differently-configured
+        search) and the scoring spec."""
         import hashlib
 
         h = hashlib.sha1()
@@ -660,11 +662,20 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
 
         upd(X)
         upd(y)
+        upd((fit_params or {}).get("sample_weight"))
+        try:
+            est_params = repr(sorted(
+                (k, repr(v))
+                for k, v in self.estimator.get_params(deep=False).items()))
+        except Exception:  # noqa: BLE001 - header must never break fit
+            est_params = ""
         return {
             "header": 1,
             "data": h.hexdigest(),
             "cv": repr(self.cv),
             "estimator": type(self.estimator).__name__,
+            "estimator_params": est_params,
+            "fit_param_keys": sorted((fit_params or {}).keys()),
             "scoring": repr(self.scoring),
         }
 

@@ -435,6 +435,21 @@ def test_resume_log_header_mismatch_refused(tmp_path):
     with pytest.raises(ValueError, match="resume_from_log refused"):
         gs4.fit(X1, y1)
 
+    # weighted rerun of an unweighted log -> refused (scores differ)
+    ctx5 = HipContext(backend="sequential", resume_from_log=log)
+    gs5 = ska.GridSearchCV(ctx5, LogisticRegression(max_iter=20),
+                           {"C": [0.1, 1.0]}, cv=2)
+    with pytest.raises(ValueError, match="resume_from_log refused"):
+        gs5.fit(X1, y1, sample_weight=np.where(y1 == 1, 3.0, 1.0))
+
+    # different CONSTRUCTOR params (not in the grid) -> refused
+    ctx6 = HipContext(backend="sequential", resume_from_log=log)
+    gs6 = ska.GridSearchCV(ctx6, LogisticRegression(max_iter=20,
+                                                    momentum=0.5),
+                           {"C": [0.1, 1.0]}, cv=2)
+    with pytest.raises(ValueError, match="resume_from_log refused"):
+        gs6.fit(X1, y1)
+
 
 def test_keyed_estimator_with_rf():
     """KeyedEstimator's generic per-key path works with the RF family."""
