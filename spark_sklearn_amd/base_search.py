@@ -736,7 +736,9 @@ class BaseSearchCV(MetaEstimatorMixin, BaseEstimator):
                 if rec.get("header") == 1:
                     saw_header = True
                     if expected_header is not None:
-                        for field in ("data", "cv", "estimator", "scoring"):
+                        for field in expected_header:
+                            if field == "header":
+                                continue
                             if rec.get(field) != expected_header[field]:
                                 raise ValueError(
                                     "resume_from_log refused: the score "
