@@ -108,14 +108,18 @@ class KMeans(ClusterMixin, BaseEstimator):
         """Fit, then distances to the fitted centers (sklearn contract)."""
         return self.fit(X, y, sample_weight=sample_weight).transform(X)
 
-    def score(self, X, y=None):
-        """Negative inertia of X under the fitted centers (sklearn)."""
+    def score(self, X, y=None, sample_weight=None):
+        """Negative (optionally weighted) inertia of X under the fitted
+        centers (sklearn contract)."""
         self._check_fitted()
         X = np.asarray(X, dtype=np.float64)
         C = self.cluster_centers_.astype(np.float64)
         d2 = (np.sum(X ** 2, axis=1)[:, None] - 2.0 * X @ C.T +
               np.sum(C ** 2, axis=1)[None, :])
-        return -float(np.sum(np.min(d2, axis=1)))
+        terms = np.min(d2, axis=1)
+        if sample_weight is not None:
+            terms = terms * np.asarray(sample_weight, dtype=np.float64)
+        return -float(np.sum(terms))
 
     def _check_fitted(self):
         if not hasattr(self, "cluster_centers_"):
